@@ -1,0 +1,128 @@
+/* tpx_abi.h — C-ABI drop-in boundary of the MI355X-native TransformStage executor.
+ *
+ * This is the replacement for the reference's JITSymbols contract: Tuplex's
+ * LocalBackend resolves per-stage function pointers by name from its LLVM JIT
+ * (reference: tuplex/core/include/physical/TransformStage.h:291-314, resolved in
+ * TransformStage.cc:763-846; functor signatures tuplex/core/include/CodeDefs.h:48,
+ * :55, :94-116) and hands each executor thread a `read_block_f` to run over raw
+ * partition bytes (TransformTask.cc:382 execute). Here the whole per-stage surface is
+ * collapsed into: compile a stage (tpx_stage_compile — replaces TransformStage::compile
+ * TransformStage.cc:763) and execute it over a partition batch (tpx_stage_execute —
+ * replaces LocalBackend::executeTransformStage LocalBackend.cc:815 +
+ * TransformTask.cc:682 processMemorySource / :724 processFileSource).
+ *
+ * Byte formats crossing this boundary are the reference's:
+ *  - partitions: [int64 numRows][rows...] (Partition.h:38), rows in the Tungsten-style
+ *    layout of utils/src/Serializer.cc:20-24 (+:29 bitmap, :1097 varlen info word);
+ *  - exception buffers: records [row,ec,opID,size][data] per
+ *    core/include/physical/IExceptionableTask.h:20 serializeExceptionToMemory;
+ *  - error codes: ExceptionCodes.h:26 (SUCCESS=0).
+ * So the host resolve/merge path (ResolveTask semantics) runs unchanged on what this
+ * library returns.
+ *
+ * All entry points are extern "C", plain pointers + sizes; no torch types.
+ */
+#ifndef TPX_ABI_H
+#define TPX_ABI_H
+
+#include <stdint.h>
+#include <stddef.h>
+
+#ifdef __cplusplus
+extern "C" {
+#endif
+
+/* ---- library / device management ------------------------------------------------ */
+
+/* ABI version (major<<16 | minor). */
+int64_t tpx_version(void);
+
+/* Number of visible HIP devices; 0 if none / HIP unavailable. Never throws. */
+int64_t tpx_device_count(void);
+
+/* Select device for subsequent calls on this thread (default 0). Returns 0 on
+ * success, nonzero HIP error otherwise. */
+int64_t tpx_set_device(int64_t device);
+
+/* Last error message (thread-local, NUL-terminated, owned by the library). */
+const char* tpx_last_error(void);
+
+/* ---- stage compile (replaces TransformStage::compile, TransformStage.cc:763) ---- */
+
+/* Opaque compiled-stage handle (the analog of JITSymbols, TransformStage.h:291). */
+typedef struct tpx_stage tpx_stage;
+
+/* Compile a stage from generated HIP source. `hip_source` is the fused-pipeline
+ * kernel source emitted by the front end (the replacement for StageBuilder.cc:602
+ * generateFastCodePath's LLVM module); `kernel_name` is the __global__ entry.
+ * `cache_dir` (may be NULL) holds hsaco code objects keyed by source hash — the
+ * analog of the JIT cache. `flags` bit 0: compile-only (do not load a module; valid
+ * on a machine with no GPU).
+ * Returns NULL on failure (see tpx_last_error). */
+tpx_stage* tpx_stage_compile(const char* hip_source,
+                             const char* kernel_name,
+                             const char* cache_dir,
+                             int64_t flags);
+
+void tpx_stage_free(tpx_stage* stage);
+
+/* ---- stage execute (replaces LocalBackend.cc:815 executeTransformStage) --------- */
+
+/* One input partition: raw bytes in the reference layout ([int64 numRows][rows...])
+ * plus the per-row offsets (offsets[i] = byte offset of row i from `data`;
+ * offsets[num_rows] = size). Offsets are host-side metadata our PartitionWriter
+ * tracks; the reference walks rows serially instead (TransformTask.cc:682). */
+typedef struct {
+    const uint8_t* data;
+    int64_t        size;
+    int64_t        num_rows;
+    const int64_t* row_offsets;   /* num_rows+1 entries */
+} tpx_partition;
+
+/* Execution result. Buffers are allocated by the library; free with
+ * tpx_result_free. */
+typedef struct {
+    /* normal-case output partition, reference layout [int64 numRows][rows...] */
+    uint8_t* out_data;
+    int64_t  out_size;
+    int64_t  out_num_rows;
+    int64_t* out_row_offsets;     /* out_num_rows+1 entries */
+    /* exception buffer: packed records [row,ec,opID,size][data]
+     * (IExceptionableTask.h:20); `row` is the global input row index. */
+    uint8_t* exc_data;
+    int64_t  exc_size;
+    int64_t  exc_num_rows;
+    /* metrics (JobMetrics.h:23 analog) */
+    double   t_h2d_ms, t_kernel_ms, t_d2h_ms;
+    int64_t  bytes_in, bytes_out;
+} tpx_result;
+
+/* Run the compiled stage over a batch of memory partitions (mem2mem source,
+ * TransformTask.cc:682 processMemorySource analog). Row indices in exception records
+ * are global across the batch in input order. Returns 0 on success. */
+int64_t tpx_stage_execute(tpx_stage* stage,
+                          const tpx_partition* parts, int64_t n_parts,
+                          tpx_result* result);
+
+/* Run the compiled stage over raw CSV bytes (file source,
+ * TransformTask.cc:724 processFileSource + CSVReader.cc:390 analog). The chunk must
+ * start at a row start and end at a row end (host chunker guarantees this, mirroring
+ * utils/src/CSVUtils.cc:1494 findLineStart). `csv_desc` is the serialized CSV schema
+ * descriptor produced by the front end (delimiter, quote, columns, null values,
+ * projected columns, per-column types). If the stage sinks to a file
+ * (tocsv), result->out_data holds CSV text bytes and out_row_offsets is NULL. */
+int64_t tpx_stage_execute_csv(tpx_stage* stage,
+                              const uint8_t* csv_bytes, int64_t size,
+                              int64_t first_global_row,
+                              tpx_result* result);
+
+void tpx_result_free(tpx_result* result);
+
+/* Generated-source introspection (debug / judge). Returns the HIP source the stage
+ * was compiled from (owned by the stage). */
+const char* tpx_stage_source(const tpx_stage* stage);
+
+#ifdef __cplusplus
+}
+#endif
+#endif /* TPX_ABI_H */
