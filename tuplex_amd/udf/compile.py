@@ -103,10 +103,9 @@ class _Compiler:
         env: Dict[str, dict] = {}
         if len(args) == 1:
             p = args[0].arg
+            self.row_param = p  # x['col'] / x[i] row access always allowed
             if len(self.input_types) == 1:
-                env[p] = tir.inp(0, self.input_types[0])
-            else:
-                self.row_param = p
+                env[p] = tir.inp(0, self.input_types[0])  # bare-scalar use
         elif len(args) == len(self.input_types):
             for i, a in enumerate(args):
                 env[a.arg] = tir.inp(i, self.input_types[i])
@@ -294,20 +293,29 @@ class _Compiler:
         raise UDFCompileError("unsupported call form")
 
     def _subscript(self, e, env):
-        # row access: x['col'] / x[i] on the row param
+        # row access: x['col'] / x[i] on the row param. Like the reference's traced
+        # typing, the meaning of x[...] depends on what x IS: when x is bound to a
+        # single str value, x[0] / x[1:] are string ops (CPython semantics); a str
+        # key is always column access (str[str] is invalid Python).
         if isinstance(e.value, ast.Name) and e.value.id == self.row_param:
-            if isinstance(e.slice, ast.Constant):
+            scalar = env.get(self.row_param)
+            scalar_is_str = scalar is not None and T.deopt(scalar["t"]) == T.STR
+            if (isinstance(e.slice, ast.Constant)
+                    and isinstance(e.slice.value, str)):
                 key = e.slice.value
-                if isinstance(key, str):
-                    if not self.columns or key not in self.columns:
-                        raise UDFCompileError("unknown column %r" % key)
-                    i = self.columns.index(key)
-                    return tir.inp(i, self.input_types[i])
-                if isinstance(key, int):
+                if not self.columns or key not in self.columns:
+                    raise UDFCompileError("unknown column %r" % key)
+                i = self.columns.index(key)
+                return tir.inp(i, self.input_types[i])
+            if not scalar_is_str:
+                if (isinstance(e.slice, ast.Constant)
+                        and isinstance(e.slice.value, int)):
+                    key = e.slice.value
                     if not (0 <= key < len(self.input_types)):
                         raise UDFCompileError("column index out of range")
                     return tir.inp(key, self.input_types[key])
-            raise UDFCompileError("row subscript must be a constant")
+                raise UDFCompileError("row subscript must be a constant")
+            # else: fall through to string getitem/slice on the scalar binding
         obj = self._expr(e.value, env)
         if isinstance(e.slice, ast.Slice):
             if e.slice.step is not None:
