@@ -54,13 +54,15 @@ typedef struct tpx_stage tpx_stage;
 
 /* Compile a stage from generated HIP source. `hip_source` is the fused-pipeline
  * kernel source emitted by the front end (the replacement for StageBuilder.cc:602
- * generateFastCodePath's LLVM module); `kernel_name` is the __global__ entry.
- * `cache_dir` (may be NULL) holds hsaco code objects keyed by source hash — the
- * analog of the JIT cache. `flags` bit 0: compile-only (do not load a module; valid
- * on a machine with no GPU).
+ * generateFastCodePath's LLVM module); it defines the fixed-name kernels
+ * tpx_stage_main / tpx_stage_write plus the runtime's scan kernels. `stage_desc`
+ * is a small key=value text descriptor (source/sink kind, in/out column types)
+ * the executor uses to size buffers. `cache_dir` (may be NULL) holds hsaco code
+ * objects keyed by source hash — the analog of the JIT cache. `flags` bit 0:
+ * compile-only (do not load a module; valid on a machine with no GPU).
  * Returns NULL on failure (see tpx_last_error). */
 tpx_stage* tpx_stage_compile(const char* hip_source,
-                             const char* kernel_name,
+                             const char* stage_desc,
                              const char* cache_dir,
                              int64_t flags);
 
@@ -82,11 +84,15 @@ typedef struct {
 /* Execution result. Buffers are allocated by the library; free with
  * tpx_result_free. */
 typedef struct {
-    /* normal-case output partition, reference layout [int64 numRows][rows...] */
+    /* normal-case output partition, reference layout [int64 numRows][rows...]
+     * (csv sink: raw CSV text and out_row_offsets are text offsets) */
     uint8_t* out_data;
     int64_t  out_size;
     int64_t  out_num_rows;
     int64_t* out_row_offsets;     /* out_num_rows+1 entries */
+    /* global input row index of each output row (order-merge support; the
+     * reference keeps row order via task ordering + ResolveTask.cc:878) */
+    int64_t* out_row_indices;     /* out_num_rows entries */
     /* exception buffer: packed records [row,ec,opID,size][data]
      * (IExceptionableTask.h:20); `row` is the global input row index. */
     uint8_t* exc_data;

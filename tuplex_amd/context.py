@@ -1,0 +1,103 @@
+"""Context — drop-in surface of the reference's tuplex/python/tuplex/context.py.
+
+Construction (context.py:51), parallelize (:246), csv (:288), options (:407).
+The execution backend behind it is the MI355X engine (engine.py) instead of the
+LLVM LocalBackend.
+"""
+from typing import Any, List, Optional
+
+from . import engine
+from .dataset import DataSet
+from .options import Options
+
+
+class ParallelizeSource:
+    kind = "mem"
+
+    def __init__(self, data, columns):
+        self.data = data
+        self.columns = columns
+
+
+class CsvSource:
+    kind = "csv"
+
+    def __init__(self, pattern, columns, delimiter, quotechar, null_values,
+                 header, type_hints):
+        self.pattern = pattern
+        self.columns = columns
+        self.delimiter = delimiter
+        self.quotechar = quotechar
+        self.null_values = null_values
+        self.header = header
+        self.type_hints = type_hints
+
+
+class Metrics:
+    def __init__(self):
+        self.data = {}
+
+    def as_json(self):
+        import json
+        return json.dumps(self.data)
+
+
+class Context:
+    def __init__(self, conf: Optional[dict] = None, name: str = "", **kwargs):
+        conf = dict(conf) if conf else {}
+        conf.update(kwargs)
+        self.options_obj = Options(conf)
+        self.metrics = Metrics()
+
+    # ---- sources ------------------------------------------------------------
+    def parallelize(self, value_list: List[Any], columns: Optional[List[str]] = None,
+                    schema=None, auto_unpack: bool = True) -> DataSet:
+        assert isinstance(value_list, list), "data must be given as a list of objects"
+        # normalize lists to tuples (reference maps python lists to tuple rows for
+        # parallelize of records)
+        return DataSet(self, ParallelizeSource(value_list, columns))
+
+    def csv(self, pattern: str, columns: Optional[List[str]] = None,
+            header: Optional[bool] = None, delimiter: Optional[str] = None,
+            quotechar: str = '"', null_values: Optional[List[str]] = None,
+            type_hints: Optional[dict] = None) -> DataSet:
+        if null_values is None:
+            null_values = [""]
+        src = CsvSource(pattern, columns, delimiter, quotechar, null_values,
+                        header, type_hints or {})
+        return DataSet(self, src)
+
+    def text(self, pattern: str, null_values: Optional[List[str]] = None) -> DataSet:
+        raise NotImplementedError("text source: next slice")
+
+    def orc(self, pattern, columns=None):
+        raise NotImplementedError("ORC is out of scope this round (SURVEY.md §8f-2)")
+
+    # ---- config -------------------------------------------------------------
+    def options(self, nested: bool = False) -> dict:
+        return self.options_obj.as_dict()
+
+    def optionsToYAML(self, file_path: str = "config.yaml") -> None:
+        import yaml
+        with open(file_path, "w") as f:
+            yaml.safe_dump(self.options(), f)
+
+    def uiWebURL(self) -> str:
+        return ""
+
+    # ---- execution ----------------------------------------------------------
+    def _execute(self, ds: DataSet, sink=None):
+        src = ds._source
+        if src.kind == "mem":
+            outcome = engine.run_collect(src.data, ds._ops, src.columns,
+                                         self.options_obj)
+        else:
+            from . import csvio
+            outcome = csvio.run_csv(self, src, ds._ops, sink)
+        self.metrics.data.update({
+            "mode": outcome.mode,
+            **{k: v for k, v in outcome.metrics.items()},
+        })
+        if sink is not None and sink[0] == "csv" and src.kind == "mem":
+            raise NotImplementedError("tocsv from parallelize: next slice")
+        return outcome

@@ -1,0 +1,117 @@
+"""DataSet — drop-in surface of the reference's tuplex/python/tuplex/dataset.py.
+
+Operator chaining builds the logical op list consumed by plan.build_stage; the same
+method names, argument shapes and semantics (map dataset.py:49, filter :83,
+collect :113, take :125, resolve :162, withColumn :201, mapColumn :231,
+selectColumns :262, renameColumn :293, ignore :319, aggregate :593, tocsv :500).
+"""
+from typing import Any, Callable, List, Optional, Union
+
+
+class DataSet:
+    def __init__(self, context, source, ops=None):
+        self._context = context
+        self._source = source          # ParallelizeSource | CsvSource
+        self._ops = ops or []
+        self._last_outcome = None
+
+    def _chain(self, op) -> "DataSet":
+        return DataSet(self._context, self._source, self._ops + [op])
+
+    # ---- transformations ----------------------------------------------------
+    def map(self, ftor: Callable) -> "DataSet":
+        assert callable(ftor), "need a function"
+        return self._chain(("map", ftor))
+
+    def filter(self, ftor: Callable) -> "DataSet":
+        assert callable(ftor), "need a function"
+        return self._chain(("filter", ftor))
+
+    def withColumn(self, column: str, ftor: Callable) -> "DataSet":
+        assert isinstance(column, str) and callable(ftor)
+        return self._chain(("withColumn", column, ftor))
+
+    def mapColumn(self, column: Union[int, str], ftor: Callable) -> "DataSet":
+        assert callable(ftor)
+        return self._chain(("mapColumn", column, ftor))
+
+    def selectColumns(self, columns: List[Union[str, int]]) -> "DataSet":
+        if not isinstance(columns, list):
+            columns = [columns]
+        return self._chain(("selectColumns", columns))
+
+    def renameColumn(self, key, newColumnName: str) -> "DataSet":
+        return self._chain(("renameColumn", key, newColumnName))
+
+    def resolve(self, eclass, ftor: Callable) -> "DataSet":
+        assert callable(ftor)
+        return self._chain(("resolve", eclass, ftor))
+
+    def ignore(self, eclass) -> "DataSet":
+        return self._chain(("ignore", eclass))
+
+    def aggregate(self, combine: Callable, aggregate: Callable,
+                  initial_value: Any) -> "DataSet":
+        return self._chain(("aggregate", combine, aggregate, initial_value))
+
+    def aggregateByKey(self, combine, aggregate, initial_value, key_columns):
+        raise NotImplementedError("aggregateByKey: next slice (SURVEY.md §7.6)")
+
+    def unique(self) -> "DataSet":
+        raise NotImplementedError("unique: next slice")
+
+    def join(self, *a, **kw):
+        raise NotImplementedError("join is out of scope (SURVEY.md §2 OOS)")
+
+    def leftJoin(self, *a, **kw):
+        raise NotImplementedError("join is out of scope (SURVEY.md §2 OOS)")
+
+    def cache(self, store_specialized: bool = True) -> "DataSet":
+        return self  # device-resident cache: later round
+
+    # ---- actions ------------------------------------------------------------
+    def collect(self) -> List[Any]:
+        outcome = self._context._execute(self)
+        self._last_outcome = outcome
+        return outcome.rows
+
+    def take(self, nrows: int = 5) -> List[Any]:
+        return self.collect()[:nrows]
+
+    def show(self, nrows: Optional[int] = None) -> None:
+        rows = self.collect()
+        if nrows is not None:
+            rows = rows[:nrows]
+        for r in rows:
+            print(r)
+
+    def tocsv(self, path: str, **kw) -> None:
+        outcome = self._context._execute(self, sink=("csv", path))
+        self._last_outcome = outcome
+
+    def toorc(self, *a, **kw):
+        raise NotImplementedError("ORC output is out of scope this round")
+
+    # ---- introspection ------------------------------------------------------
+    @property
+    def columns(self) -> Optional[List[str]]:
+        cols = self._source.columns
+        for op in self._ops:
+            if op[0] == "map":
+                cols = None
+            elif op[0] == "withColumn":
+                if cols and op[1] not in cols:
+                    cols = cols + [op[1]]
+            elif op[0] == "selectColumns":
+                cols = [c if isinstance(c, str) else (cols[c] if cols else None)
+                        for c in op[1]]
+            elif op[0] == "renameColumn":
+                if cols:
+                    cols = [op[2] if c == op[1] else c for c in cols]
+        return cols
+
+    @property
+    def exception_counts(self) -> dict:
+        if self._last_outcome is None:
+            return {}
+        return dict(self._last_outcome.exception_counts)

@@ -1,0 +1,242 @@
+"""Host execution engine: drives the C-ABI GPU executor and the host resolve path.
+
+Replaces the reference's LocalBackend orchestration (LocalBackend.cc:815
+executeTransformStage: compile, task fan-out, perform, exception resolution, sink)
+for the single-node hot path. The normal case runs ONLY on the GPU (fails loudly if
+the HIP library or a device is missing); exception/fallback rows are replayed on the
+CPython interpreter (resolve.py) and merged in order — the reference's dual-mode
+contract (ResolveTask.cc:702/:878).
+"""
+import ctypes
+import os
+import time
+from typing import Any, List, Optional
+
+from . import codegen, plan, rowfmt, resolve
+from . import ttypes as T
+from . import ec as EC
+
+_PKG = os.path.dirname(os.path.abspath(__file__))
+_LIB_PATH = os.path.join(_PKG, "libtpx_gpu.so")
+_CACHE_DIR = os.path.join(_PKG, ".kernel_cache")
+
+
+class TpxResult(ctypes.Structure):
+    _fields_ = [
+        ("out_data", ctypes.POINTER(ctypes.c_uint8)),
+        ("out_size", ctypes.c_int64),
+        ("out_num_rows", ctypes.c_int64),
+        ("out_row_offsets", ctypes.POINTER(ctypes.c_int64)),
+        ("out_row_indices", ctypes.POINTER(ctypes.c_int64)),
+        ("exc_data", ctypes.POINTER(ctypes.c_uint8)),
+        ("exc_size", ctypes.c_int64),
+        ("exc_num_rows", ctypes.c_int64),
+        ("t_h2d_ms", ctypes.c_double),
+        ("t_kernel_ms", ctypes.c_double),
+        ("t_d2h_ms", ctypes.c_double),
+        ("bytes_in", ctypes.c_int64),
+        ("bytes_out", ctypes.c_int64),
+    ]
+
+
+class TpxPartition(ctypes.Structure):
+    _fields_ = [
+        ("data", ctypes.POINTER(ctypes.c_uint8)),
+        ("size", ctypes.c_int64),
+        ("num_rows", ctypes.c_int64),
+        ("row_offsets", ctypes.POINTER(ctypes.c_int64)),
+    ]
+
+
+class GpuLib:
+    _instance = None
+
+    def __init__(self):
+        if not os.path.exists(_LIB_PATH):
+            raise RuntimeError(
+                "libtpx_gpu.so not built — run python tuplex_amd/csrc/build.py "
+                "(the GPU normal-case path has no CPU fallback by design)")
+        lib = ctypes.CDLL(_LIB_PATH)
+        lib.tpx_version.restype = ctypes.c_int64
+        lib.tpx_device_count.restype = ctypes.c_int64
+        lib.tpx_set_device.argtypes = [ctypes.c_int64]
+        lib.tpx_set_device.restype = ctypes.c_int64
+        lib.tpx_last_error.restype = ctypes.c_char_p
+        lib.tpx_stage_compile.restype = ctypes.c_void_p
+        lib.tpx_stage_compile.argtypes = [ctypes.c_char_p, ctypes.c_char_p,
+                                          ctypes.c_char_p, ctypes.c_int64]
+        lib.tpx_stage_free.argtypes = [ctypes.c_void_p]
+        lib.tpx_stage_source.argtypes = [ctypes.c_void_p]
+        lib.tpx_stage_source.restype = ctypes.c_char_p
+        lib.tpx_stage_execute.restype = ctypes.c_int64
+        lib.tpx_stage_execute.argtypes = [ctypes.c_void_p,
+                                          ctypes.POINTER(TpxPartition),
+                                          ctypes.c_int64,
+                                          ctypes.POINTER(TpxResult)]
+        lib.tpx_stage_execute_csv.restype = ctypes.c_int64
+        lib.tpx_stage_execute_csv.argtypes = [ctypes.c_void_p,
+                                              ctypes.POINTER(ctypes.c_uint8),
+                                              ctypes.c_int64, ctypes.c_int64,
+                                              ctypes.POINTER(TpxResult)]
+        lib.tpx_result_free.argtypes = [ctypes.POINTER(TpxResult)]
+        self.lib = lib
+        os.makedirs(_CACHE_DIR, exist_ok=True)
+        self._stage_cache = {}
+
+    @classmethod
+    def get(cls):
+        if cls._instance is None:
+            cls._instance = GpuLib()
+        return cls._instance
+
+    def device_count(self) -> int:
+        return int(self.lib.tpx_device_count())
+
+    def err(self) -> str:
+        return self.lib.tpx_last_error().decode("utf-8", "replace")
+
+    def compile_stage(self, src: str, desc: str, compile_only=False):
+        key = (src, compile_only)
+        h = self._stage_cache.get(key)
+        if h:
+            return h
+        h = self.lib.tpx_stage_compile(src.encode(), desc.encode(),
+                                       _CACHE_DIR.encode(),
+                                       1 if compile_only else 0)
+        if not h:
+            raise RuntimeError("stage compile failed: " + self.err())
+        self._stage_cache[key] = h
+        return h
+
+
+class ExecResult:
+    def __init__(self):
+        self.rows: List[tuple] = []         # normal-case output rows (value tuples)
+        self.row_indices: List[int] = []    # local input row index per output row
+        self.exceptions = []                # [(global_row, ec, opid)]
+        self.metrics = {}
+
+
+def execute_stage_mem(sp: plan.StageProgram, norm_rows: List[tuple],
+                      device: int = 0) -> ExecResult:
+    """Run the fused stage on the GPU over serialized normal-case rows."""
+    glib = GpuLib.get()
+    if glib.device_count() == 0:
+        raise RuntimeError("no HIP device visible — the normal-case path runs only "
+                           "on GPU (no CPU fallback by design)")
+    glib.lib.tpx_set_device(device)
+    in_row_type = T.tup(sp.input_types)
+    t0 = time.perf_counter()
+    buf, offs = rowfmt.serialize_partition(norm_rows, in_row_type)
+    t_ser = time.perf_counter() - t0
+
+    src, desc = codegen.generate_stage(sp, source="mem", sink="mem")
+    stage = glib.compile_stage(src, desc)
+
+    cbuf = (ctypes.c_uint8 * len(buf)).from_buffer_copy(buf)
+    coffs = (ctypes.c_int64 * len(offs))(*offs)
+    part = TpxPartition(
+        data=ctypes.cast(cbuf, ctypes.POINTER(ctypes.c_uint8)),
+        size=len(buf), num_rows=len(norm_rows),
+        row_offsets=ctypes.cast(coffs, ctypes.POINTER(ctypes.c_int64)))
+    res = TpxResult()
+    rc = glib.lib.tpx_stage_execute(stage, ctypes.byref(part), 1, ctypes.byref(res))
+    if rc != 0:
+        raise RuntimeError("stage execute failed: " + glib.err())
+    try:
+        out = ExecResult()
+        out_bytes = ctypes.string_at(res.out_data, res.out_size)
+        out_row_type = T.tup(sp.output_types)
+        out.rows = rowfmt.deserialize_partition(out_bytes, out_row_type)
+        out.row_indices = [res.out_row_indices[i] for i in range(res.out_num_rows)] \
+            if res.out_num_rows else []
+        if res.exc_num_rows:
+            import struct as _s
+            eb = ctypes.string_at(res.exc_data, res.exc_size)
+            pos = 0
+            for _ in range(res.exc_num_rows):
+                row, ecode, opid, size = _s.unpack_from("<4q", eb, pos)
+                pos += 32 + size
+                out.exceptions.append((row, ecode, opid))
+        out.metrics = {
+            "t_serialize_s": t_ser,
+            "t_h2d_ms": res.t_h2d_ms, "t_kernel_ms": res.t_kernel_ms,
+            "t_d2h_ms": res.t_d2h_ms,
+            "bytes_in": res.bytes_in, "bytes_out": res.bytes_out,
+        }
+        return out
+    finally:
+        glib.lib.tpx_result_free(ctypes.byref(res))
+
+
+def run_collect(data: List[Any], logical_ops: List[tuple],
+                columns: Optional[List[str]], options) -> "CollectOutcome":
+    """Full dual-mode collect: majority-type inference, GPU normal case,
+    interpreter replay of fallback + exception rows, in-order merge
+    (mirrors LocalBackend.cc:963-1085 + ResolveTask.cc:878)."""
+    out = CollectOutcome()
+    maj = T.infer_majority_type(data, optional_threshold=options.optional_threshold)
+    row_maj = T.row_type_of(maj)
+    scalar_input = not T.is_tuple(maj)
+
+    norm_idx, norm_rows, fallback = [], [], []
+    for i, v in enumerate(data):
+        if T.value_conforms(v, maj):
+            norm_idx.append(i)
+            norm_rows.append((v,) if scalar_input else v)
+        else:
+            fallback.append((i, v))
+
+    input_types = list(T.tuple_params(row_maj))
+    sp = plan.build_stage(input_types, columns, logical_ops)
+
+    results = {}  # original index -> row value
+    if sp.compilable and norm_rows:
+        er = execute_stage_mem(sp, norm_rows)
+        out.mode = "gpu"
+        out.metrics = er.metrics
+        # outputs carry their exact local row index (out_row_indices) -> map back
+        # to original data indices for the in-order merge (ResolveTask.cc:878)
+        for row, local in zip(er.rows, er.row_indices):
+            results[norm_idx[local]] = _unwrap_row(row)
+        replay = [(norm_idx[r], data[norm_idx[r]]) for (r, _, _) in er.exceptions]
+    elif not sp.compilable:
+        out.mode = "fallback"
+        out.fallback_reason = sp.why_not_compilable
+        replay = [(i, data[i]) for i in norm_idx]
+    else:
+        out.mode = "gpu"
+        replay = []
+
+    replay += fallback
+    for i, v in sorted(replay):
+        r = resolve.replay_row(v, logical_ops, columns, scalar_input)
+        if r[0] == "row":
+            results[i] = r[1]
+        elif r[0] == "exc":
+            name = type(r[1]).__name__
+            out.exception_counts[name] = out.exception_counts.get(name, 0) + 1
+
+    merged = [results[i] for i in sorted(results)]
+    agg = next((op for op in logical_ops if op[0] == "aggregate"), None)
+    if agg is not None:
+        _, combine_fn, agg_fn, initial = agg
+        a = initial
+        for v in merged:
+            a = agg_fn(a, v)
+        merged = [a]
+    out.rows = merged
+    return out
+
+
+def _unwrap_row(row: tuple):
+    return row[0] if len(row) == 1 else row
+
+
+class CollectOutcome:
+    def __init__(self):
+        self.rows = []
+        self.exception_counts = {}
+        self.mode = None
+        self.fallback_reason = None
+        self.metrics = {}

@@ -1,0 +1,196 @@
+"""Logical plan -> TransformStage program.
+
+The reference builds a DAG of LogicalOperators (core/src/logical/), splits it into
+TransformStages at pipeline breakers (PhysicalPlan.cc:423 splitIntoAndPlanStages) and
+fuses each stage's operators into one function (StageBuilder.cc:602). The hot path
+here has a single stage: source -> row ops (map/filter/withColumn/mapColumn/select/
+rename) -> optional aggregate -> sink. This module turns the DataSet op chain into a
+StageProgram: per-op TIR (compiled UDFs), schema threading, opids for exception
+records, and the resolver table used by the host resolve path.
+"""
+from typing import List, Optional
+
+from . import ttypes as T
+from . import ec as EC
+from .udf import compile_udf, UDFCompileError
+
+
+class StageOp:
+    def __init__(self, kind, opid, **kw):
+        self.kind = kind      # map|filter|withColumn|mapColumn|selectColumns|rename
+        self.opid = opid
+        self.fn = kw.get("fn")            # original python callable
+        self.tir = kw.get("tir")          # compiled TIR (None -> not compilable)
+        self.col = kw.get("col")          # withColumn/mapColumn target
+        self.cols = kw.get("cols")        # selectColumns list
+        self.rename = kw.get("rename")    # (old, new)
+        self.resolvers = []               # [(ec_code, exc_class, fn)]
+        self.ignores = []                 # [(ec_code, exc_class)]
+        # schemas threaded at build time
+        self.in_types = None
+        self.in_columns = None
+        self.out_types = None
+        self.out_columns = None
+
+
+class StageProgram:
+    """One fused TransformStage (normal-case path)."""
+
+    def __init__(self, input_types: List, input_columns: Optional[List[str]]):
+        self.input_types = list(input_types)
+        self.input_columns = list(input_columns) if input_columns else None
+        self.ops: List[StageOp] = []
+        self.aggregate = None  # (opid, combine_fn, agg_fn, initial)
+        self.compilable = True
+        self.why_not_compilable = None
+
+    @property
+    def output_types(self):
+        if self.ops:
+            return self.ops[-1].out_types
+        return self.input_types
+
+    @property
+    def output_columns(self):
+        if self.ops:
+            return self.ops[-1].out_columns
+        return self.input_columns
+
+    def signature(self) -> str:
+        """Stable identity for the kernel cache (the analog of the reference's
+        per-stage symbol names Stage_N)."""
+        import hashlib
+        import json
+
+        def node_key(n):
+            if n is None:
+                return None
+            out = {"op": n["op"], "t": repr(n["t"])}
+            for k in ("v", "i", "w"):
+                if k in n:
+                    out[k] = repr(n[k])
+            out["a"] = [node_key(c) for c in n["args"]]
+            return out
+
+        desc = {
+            "in": repr(self.input_types), "cols": self.input_columns,
+            "ops": [{"k": o.kind, "col": o.col, "cols": o.cols,
+                     "r": o.rename, "tir": node_key(o.tir)} for o in self.ops],
+        }
+        return hashlib.sha256(json.dumps(desc, sort_keys=True).encode()).hexdigest()[:16]
+
+
+def build_stage(input_types, input_columns, logical_ops) -> StageProgram:
+    """logical_ops: list of tuples as produced by DataSet:
+    ("map", fn) ("filter", fn) ("withColumn", col, fn) ("mapColumn", col, fn)
+    ("selectColumns", [cols]) ("renameColumn", old, new)
+    ("resolve", exc_class, fn) ("ignore", exc_class)
+    ("aggregate", combine_fn, agg_fn, initial)
+    """
+    sp = StageProgram(input_types, input_columns)
+    cur_types = list(input_types)
+    cur_cols = list(input_columns) if input_columns else None
+    opid = 0
+    for entry in logical_ops:
+        kind = entry[0]
+        opid += 1
+        if kind in ("resolve", "ignore"):
+            if not sp.ops:
+                raise ValueError("%s with no preceding operator" % kind)
+            target = sp.ops[-1]
+            cls = entry[1]
+            code = EC.code_for_class(cls)
+            if kind == "resolve":
+                target.resolvers.append((code, cls, entry[2]))
+            else:
+                target.ignores.append((code, cls))
+            continue
+        if kind == "aggregate":
+            sp.aggregate = (opid, entry[1], entry[2], entry[3])
+            continue
+
+        op = StageOp(kind, opid)
+        op.in_types = list(cur_types)
+        op.in_columns = list(cur_cols) if cur_cols else None
+        if kind == "map":
+            op.fn = entry[1]
+            _compile_into(sp, op, cur_types, cur_cols)
+            if op.tir is not None:
+                rt = T.row_type_of(op.tir["t"])
+                cur_types = list(T.tuple_params(rt))
+            cur_cols = None  # map drops column names (reference DataSet::map)
+        elif kind == "filter":
+            op.fn = entry[1]
+            _compile_into(sp, op, cur_types, cur_cols)
+            if op.tir is not None and op.tir["t"] != T.BOOL:
+                _fallback(sp, "filter UDF returns %r, not bool" % (op.tir["t"],))
+                op.tir = None
+        elif kind == "withColumn":
+            op.col, op.fn = entry[1], entry[2]
+            cols_for_udf = cur_cols or ["column%d" % i for i in range(len(cur_types))]
+            _compile_into(sp, op, cur_types, cols_for_udf)
+            vt = op.tir["t"] if op.tir is not None else T.STR
+            if op.col in cols_for_udf:
+                i = cols_for_udf.index(op.col)
+                cur_types = cur_types[:i] + [vt] + cur_types[i + 1:]
+            else:
+                cols_for_udf = cols_for_udf + [op.col]
+                cur_types = cur_types + [vt]
+            cur_cols = cols_for_udf
+        elif kind == "mapColumn":
+            op.col, op.fn = entry[1], entry[2]
+            if not cur_cols or op.col not in cur_cols:
+                raise ValueError("mapColumn: unknown column %r" % op.col)
+            i = cur_cols.index(op.col)
+            try:
+                op.tir = compile_udf(op.fn, [cur_types[i]], None)
+            except UDFCompileError as e:
+                _fallback(sp, str(e))
+                op.tir = None
+            if op.tir is not None:
+                cur_types = cur_types[:i] + [op.tir["t"]] + cur_types[i + 1:]
+        elif kind == "selectColumns":
+            op.cols = entry[1]
+            if cur_cols is None:
+                idxs = [c for c in op.cols]
+                if not all(isinstance(c, int) for c in idxs):
+                    raise ValueError("selectColumns by name needs named columns")
+            else:
+                idxs = [cur_cols.index(c) if isinstance(c, str) else c
+                        for c in op.cols]
+            op.sel_idxs = idxs
+            cur_types = [cur_types[i] for i in idxs]
+            new_cols = []
+            for c, i in zip(op.cols, idxs):
+                if isinstance(c, str):
+                    new_cols.append(c)
+                elif cur_cols:
+                    new_cols.append(cur_cols[i])
+                else:
+                    new_cols.append("column%d" % i)
+            cur_cols = new_cols
+        elif kind == "renameColumn":
+            op.rename = (entry[1], entry[2])
+            if cur_cols is None or entry[1] not in cur_cols:
+                raise ValueError("renameColumn: unknown column %r" % entry[1])
+            cur_cols = [entry[2] if c == entry[1] else c for c in cur_cols]
+        else:
+            raise ValueError("unknown op %r" % (kind,))
+        op.out_types = list(cur_types)
+        op.out_columns = list(cur_cols) if cur_cols else None
+        sp.ops.append(op)
+    return sp
+
+
+def _compile_into(sp, op, cur_types, cur_cols):
+    try:
+        op.tir = compile_udf(op.fn, cur_types, cur_cols)
+    except UDFCompileError as e:
+        _fallback(sp, "%s (op %d)" % (e, op.opid))
+        op.tir = None
+
+
+def _fallback(sp, why):
+    sp.compilable = False
+    if sp.why_not_compilable is None:
+        sp.why_not_compilable = why

@@ -1,0 +1,120 @@
+"""Host interpreter replay of exception/fallback rows — the product's slow path.
+
+The reference keeps exception handling on the host: ResolveTask.cc:389
+processExceptionRow (try compiled resolver, else CPython interpreter via the
+generated pure-Python pipeline, PythonPipelineBuilder.cc) and :878 executeInOrder
+(in-order merge). Here every diverted row (fallback row, normal-case violation,
+user exception, heap overflow) is replayed through the full operator chain with
+CPython semantics, with resolver/ignore handlers applied at the operator that
+raised — per north_star this path stays CPU-only and semantically identical.
+"""
+from typing import Any, List, Optional
+
+
+def _n_params(fn):
+    return fn.__code__.co_argcount
+
+
+def _call_udf(fn, row, columns):
+    if isinstance(row, tuple):
+        if _n_params(fn) == len(row) and _n_params(fn) > 1:
+            return fn(*row)
+        if columns:
+            return fn(dict(zip(columns, row)))
+        return fn(row)
+    return fn(row)
+
+
+def _as_row(v):
+    return v if isinstance(v, tuple) else (v,)
+
+
+def replay_row(value: Any, logical_ops: List[tuple], columns: Optional[List[str]],
+               scalar_input: bool):
+    """Replay one row through the op chain. Returns ("row", v) | ("drop",) |
+    ("exc", exception)."""
+    cur = value
+    cols = list(columns) if columns else None
+    row_ops = [op for op in logical_ops if op[0] != "aggregate"]
+    k = 0
+    while k < len(row_ops):
+        op = row_ops[k]
+        kind = op[0]
+        if kind in ("resolve", "ignore"):
+            k += 1
+            continue
+        try:
+            cur, cols, dropped = _apply_op(op, cur, cols)
+            if dropped:
+                return ("drop",)
+        except Exception as e:  # noqa: BLE001 — row-level dual-mode machinery
+            j = k + 1
+            handled = False
+            while j < len(row_ops) and row_ops[j][0] in ("resolve", "ignore"):
+                rkind, rcls = row_ops[j][0], row_ops[j][1]
+                if isinstance(e, rcls):
+                    if rkind == "ignore":
+                        return ("drop",)
+                    try:
+                        cur, cols, dropped = _apply_resolver(op, row_ops[j][2],
+                                                             cur, cols)
+                    except Exception as e2:  # resolver raised
+                        return ("exc", e2)
+                    if dropped:
+                        return ("drop",)
+                    handled = True
+                    break
+                j += 1
+            if not handled:
+                return ("exc", e)
+            k = j
+        k += 1
+    return ("row", cur)
+
+
+def _apply_op(op, cur, cols):
+    kind = op[0]
+    if kind == "map":
+        return _call_udf(op[1], cur, cols), None, False
+    if kind == "filter":
+        return cur, cols, not _call_udf(op[1], cur, cols)
+    if kind == "withColumn":
+        col, fn = op[1], op[2]
+        row = _as_row(cur)
+        cols2 = list(cols) if cols else ["column%d" % i for i in range(len(row))]
+        v = _call_udf(fn, row, cols2)
+        if col in cols2:
+            i = cols2.index(col)
+            row = row[:i] + (v,) + row[i + 1:]
+        else:
+            cols2 = cols2 + [col]
+            row = row + (v,)
+        return row, cols2, False
+    if kind == "mapColumn":
+        col, fn = op[1], op[2]
+        row = _as_row(cur)
+        i = cols.index(col)
+        return row[:i] + (fn(row[i]),) + row[i + 1:], cols, False
+    if kind == "selectColumns":
+        sel = op[1]
+        row = _as_row(cur)
+        idxs = [cols.index(c) if isinstance(c, str) else c for c in sel]
+        names = [c if isinstance(c, str) else (cols[c] if cols else "column%d" % c)
+                 for c in sel]
+        out = tuple(row[i] for i in idxs)
+        return (out if len(out) > 1 else out[0]), names, False
+    if kind == "renameColumn":
+        old, new = op[1], op[2]
+        return cur, [new if c == old else c for c in cols], False
+    raise ValueError("unknown op %r" % (kind,))
+
+
+def _apply_resolver(op, resolver, cur, cols):
+    kind = op[0]
+    if kind == "map":
+        return _call_udf(resolver, cur, cols), None, False
+    if kind == "filter":
+        return cur, cols, not _call_udf(resolver, cur, cols)
+    if kind in ("withColumn", "mapColumn"):
+        return _apply_op((kind, op[1], resolver), cur, cols)
+    raise ValueError("resolver after %r unsupported" % (kind,))
