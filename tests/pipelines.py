@@ -1,0 +1,192 @@
+"""Shared pipeline definitions for GPU parity tests and kernel pre-compilation.
+
+Each entry: (name, data, columns, ops) where ops is the logical op list used both
+by tuplex_amd (DataSet chaining) and the oracle. UDFs live at module level so
+inspect.getsource works both here and on the GPU box, and so __graft_entry__.build()
+can pre-compile every stage's hsaco into the in-tree cache (which travels with the
+repo snapshot — the GPU box then loads instead of re-compiling).
+"""
+import random
+import string
+
+from tests.zillow_data import ZILLOW_COLS, make_zillow_rows
+from tests.test_codegen_compile import zillow_ops, zillow_input_types  # noqa: F401
+
+
+# ---- UDFs (module level for source capture) --------------------------------------
+
+def sq_map(x):
+    return (x, x * x)
+
+
+def div_map(x):
+    return 1 // x
+
+
+def div_resolver(x):
+    return -1
+
+
+def div_filter(x):
+    return (1 // x) < 5
+
+
+def str_of(x):
+    return str(x)
+
+
+def mapcol_div(x):
+    return 1 // x
+
+
+def withcol_str(x, y, z):
+    return str(1 // x) + y
+
+
+def withcol_resolver(x, y, z):
+    return "NULL"
+
+
+def swapcase_udf(x):
+    return x.swapcase()
+
+
+def startswith_udf(s, p):
+    return s.startswith(p)
+
+
+def agg_combine(a, b):
+    return a + b
+
+
+def agg_sum(a, x):
+    return a + x
+
+
+def lower_udf(x):
+    return x.lower()
+
+
+def find_def_udf(x):
+    return x.find("def")
+
+
+def floor_udf(x):
+    return (x // 2, x % 3, -x)
+
+
+def divmod_udf(x):
+    return (x, x // 7, x % 13)
+
+
+def strops_udf(x):
+    return (x.find(","), x.upper(), len(x))
+
+
+def int_udf(x):
+    return int(x)
+
+
+def gt1000(x):
+    return x > 1000
+
+
+def ge0(x):
+    return x >= 0
+
+
+def _rand_ints(n=10000, seed=42):
+    rng = random.Random(seed)
+    return [rng.randint(-2**40, 2**40) for _ in range(n)]
+
+
+def _rand_strs(n=5000, seed=7):
+    rng = random.Random(seed)
+    alphabet = string.ascii_letters + string.digits + " ,$/"
+    return ["".join(rng.choice(alphabet) for _ in range(rng.randint(0, 60)))
+            for _ in range(n)]
+
+
+PIPELINES = [
+    ("config1_plumbing", [1, 2, None, 4], None, [("map", sq_map)]),
+    ("map_div_exc", [1, 0, 0, 2], None, [("map", div_map)]),
+    ("map_div_resolved", [1, 0, 0, 2], None,
+     [("map", div_map), ("resolve", ZeroDivisionError, div_resolver)]),
+    ("filter_exc", [1, 0, 0, 2], None, [("filter", div_filter)]),
+    ("option_str", [1, 2, None, 3, None, None], None, [("map", str_of)]),
+    ("mapcol_resolve", [(1, "a"), (0, "b"), (3, "c")], ["int", "str"],
+     [("mapColumn", "int", mapcol_div),
+      ("resolve", ZeroDivisionError, div_resolver)]),
+    ("withcol_replace", [(1, "a", True), (0, "b", False), (3, "c", True)],
+     ["num", "str", "bool"], [("withColumn", "str", withcol_str)]),
+    ("swapcase", ["all lower", "ALL UPPER", "Upper && Lower", "Number *45.67++",
+                  "", "pQ" * 16], None, [("map", swapcase_udf)]),
+    ("startswith", [("hello", "h"), ("hello", "he"), ("Hello", "hello"),
+                    ("abcde", "abcde")], None, [("map", startswith_udf)]),
+    ("agg_sum", [1, 2, 3, 4, 5, 6], None,
+     [("aggregate", agg_combine, agg_sum, 0)]),
+    ("unicode_lower", ["", "abc", "héllo wörld", "x", "ÅBC"], None,
+     [("map", lower_udf)]),
+    ("unicode_find", ["abc def", "héllo wörld def", "no match here"],
+     None, [("map", find_def_udf)]),
+    ("floor_semantics", [-7, -1, 1, 7, -8, 8], None, [("map", floor_udf)]),
+    ("rand_ints", _rand_ints(), None, [("map", divmod_udf)]),
+    ("rand_strs", _rand_strs(), None, [("map", strops_udf)]),
+    ("int_parse", ["42", " 17 ", "-5", "bogus", "", "123456789012", "007"], None,
+     [("map", int_udf)]),
+    ("filter_all", list(range(100)), None, [("filter", gt1000)]),
+    ("filter_none", list(range(100)), None, [("filter", ge0)]),
+    ("zillow_mem", make_zillow_rows(2000, seed=42), ZILLOW_COLS, zillow_ops()),
+]
+
+
+def apply_ops(ds, ops):
+    for op in ops:
+        kind = op[0]
+        if kind == "map":
+            ds = ds.map(op[1])
+        elif kind == "filter":
+            ds = ds.filter(op[1])
+        elif kind == "withColumn":
+            ds = ds.withColumn(op[1], op[2])
+        elif kind == "mapColumn":
+            ds = ds.mapColumn(op[1], op[2])
+        elif kind == "selectColumns":
+            ds = ds.selectColumns(op[1])
+        elif kind == "renameColumn":
+            ds = ds.renameColumn(op[1], op[2])
+        elif kind == "resolve":
+            ds = ds.resolve(op[1], op[2])
+        elif kind == "ignore":
+            ds = ds.ignore(op[1])
+        elif kind == "aggregate":
+            ds = ds.aggregate(op[1], op[2], op[3])
+        else:
+            raise ValueError(kind)
+    return ds
+
+
+def precompile_all(verbose=False):
+    """Generate + hipRTC-compile (compile-only) every pipeline's stage so the
+    hsaco cache is warm. Works with no GPU."""
+    from tuplex_amd import codegen, engine, plan
+    from tuplex_amd import ttypes as T
+    from tuplex_amd.options import Options
+
+    glib = engine.GpuLib.get()
+    opts = Options()
+    n = 0
+    for name, data, columns, ops in PIPELINES:
+        maj = T.infer_majority_type(data, optional_threshold=opts.optional_threshold)
+        row_maj = T.row_type_of(maj)
+        sp = plan.build_stage(list(T.tuple_params(row_maj)), columns, ops)
+        if not sp.compilable:
+            if verbose:
+                print("skip (fallback):", name, sp.why_not_compilable)
+            continue
+        src, desc = codegen.generate_stage(sp, source="mem", sink="mem")
+        glib.compile_stage(src, desc, compile_only=True)
+        n += 1
+        if verbose:
+            print("precompiled:", name)
+    return n
