@@ -339,51 +339,11 @@ def run_pipeline(data: List[Any], ops: List[tuple], columns: Optional[List[str]]
         name = type(e).__name__
         exc_counts[name] = exc_counts.get(name, 0) + 1
 
-    def process(idx, value, fast):
-        """Run the row through row_ops. Returns ("row", value) | ("drop",) |
-        ("exc", exception)."""
-        cur = value
-        cols = list(columns) if columns else None
-        k = 0
-        while k < len(row_ops):
-            op = row_ops[k]
-            kind = op[0]
-            if kind in ("resolve", "ignore"):
-                k += 1
-                continue
-            try:
-                cur, cols, dropped = _apply_op(op, cur, cols, fast)
-                if dropped:
-                    return ("drop",)
-            except Exception as e:  # noqa: BLE001 — row-level exception machinery
-                # scan following resolve/ignore ops (ResolveTask.cc:389)
-                j = k + 1
-                while j < len(row_ops) and row_ops[j][0] in ("resolve", "ignore"):
-                    rkind = row_ops[j][0]
-                    rcls = row_ops[j][1]
-                    if isinstance(e, rcls):
-                        if rkind == "ignore":
-                            return ("drop",)
-                        try:
-                            cur, cols, dropped = _apply_resolver(op, row_ops[j][2],
-                                                                 cur, cols, fast)
-                        except Exception as e2:  # resolver itself raised
-                            return ("exc", e2)
-                        if dropped:
-                            return ("drop",)
-                        break
-                    j += 1
-                else:
-                    return ("exc", e)
-                k = j  # continue after the matched resolver
-            k += 1
-        return ("row", cur)
-
     stream = sorted(normal + fallback) if merge_rows_in_order else normal + fallback
     fb_idx = {i for i, _ in fallback}
     for idx, v in stream:
         fast = idx not in fb_idx
-        r = process(idx, v, fast)
+        r = process_row(v, row_ops, columns, fast)
         if r[0] == "row":
             results.append((idx, r[1]))
         elif r[0] == "exc":
@@ -401,6 +361,47 @@ def run_pipeline(data: List[Any], ops: List[tuple], columns: Optional[List[str]]
         out = [a]
 
     return {"output": out, "exception_counts": exc_counts}
+
+
+def process_row(value, row_ops, columns, fast):
+    """Run one row through row_ops with the reference's dual-mode semantics.
+    Returns ("row", value) | ("drop",) | ("exc", exception).
+    fast=True: compiled-path int()/float() semantics; False: interpreter."""
+    cur = value
+    cols = list(columns) if columns else None
+    k = 0
+    while k < len(row_ops):
+        op = row_ops[k]
+        kind = op[0]
+        if kind in ("resolve", "ignore"):
+            k += 1
+            continue
+        try:
+            cur, cols, dropped = _apply_op(op, cur, cols, fast)
+            if dropped:
+                return ("drop",)
+        except Exception as e:  # noqa: BLE001 — row-level exception machinery
+            j = k + 1  # scan following resolve/ignore ops (ResolveTask.cc:389)
+            while j < len(row_ops) and row_ops[j][0] in ("resolve", "ignore"):
+                rkind = row_ops[j][0]
+                rcls = row_ops[j][1]
+                if isinstance(e, rcls):
+                    if rkind == "ignore":
+                        return ("drop",)
+                    try:
+                        cur, cols, dropped = _apply_resolver(op, row_ops[j][2],
+                                                             cur, cols, fast)
+                    except Exception as e2:  # resolver itself raised
+                        return ("exc", e2)
+                    if dropped:
+                        return ("drop",)
+                    break
+                j += 1
+            else:
+                return ("exc", e)
+            k = j  # continue after the matched resolver
+        k += 1
+    return ("row", cur)
 
 
 def _apply_op(op, cur, cols, fast):

@@ -1,0 +1,96 @@
+"""CPU-side CSV tests: oracle CSV restatement self-consistency, sniffing, row/cell
+split rules, and hipRTC compile of the CSV-source Zillow stage."""
+import ctypes
+import os
+
+import pytest
+
+from oracle import pyoracle, pyoracle_csv
+from tuplex_amd import codegen, csvio, plan
+from tuplex_amd import ttypes as T
+from tests.zillow_data import ZILLOW_COLS, make_zillow_csv_bytes
+from tests.test_codegen_compile import zillow_ops, _compile_only
+
+
+def test_split_rows_quoted_newline():
+    data = b'a,"x\ny",b\nc,d,e\n'
+    rows = pyoracle_csv.split_rows(data)
+    assert len(rows) == 2
+    assert rows[0] == b'a,"x\ny",b\n'
+
+
+def test_split_cells_quotes_and_escapes():
+    cells, flags = pyoracle_csv.split_cells(b'plain,"quoted, cell",tail\n')
+    assert [c.decode() for c in cells] == ["plain", "quoted, cell", "tail"]
+    assert flags == 0
+    cells, flags = pyoracle_csv.split_cells(b'a,"he said ""hi""",b\n')
+    assert flags & 2  # escape -> divert
+    cells, flags = pyoracle_csv.split_cells(b'a,"unterminated\n')
+    assert flags & 4
+
+
+def test_product_split_matches_oracle():
+    import random
+    rng = random.Random(3)
+    pool = 'abc,"\n'
+    for _ in range(200):
+        s = "".join(rng.choice(pool) for _ in range(rng.randint(0, 40))) + "\n"
+        b = s.encode()
+        assert csvio.split_rows(b) == pyoracle_csv.split_rows(b)
+        assert csvio.split_cells(b) == pyoracle_csv.split_cells(b)
+
+
+def test_sniff_zillow_schema():
+    data, _rows = make_zillow_csv_bytes(500, seed=42)
+    has_header, names, types = pyoracle_csv.sniff(data, [""], 0.9, None,
+                                                  None)
+    assert has_header
+    assert names == ZILLOW_COLS
+    assert types[4] == "f64"  # postal_code
+    assert all(t == "str" for i, t in enumerate(types) if i != 4)
+    # product sniffer agrees
+    h2, n2, t2 = csvio.sniff(data, [""], 0.9, None, None)
+    assert (h2, n2) == (has_header, names)
+    assert [T.deopt(t) if not T.is_opt(t) else t for t in t2] == \
+        ["str", "str", "str", "str", "f64", "str", "str", "str", "str", "str"]
+
+
+def test_oracle_csv_matches_oracle_mem_on_zillow():
+    """Internal consistency: the oracle's CSV path over serialized rows must equal
+    the oracle's mem path over the same typed rows."""
+    data, rows = make_zillow_csv_bytes(800, seed=11, dirty_frac=0.03)
+    r_csv = pyoracle_csv.run_csv_pipeline(data, zillow_ops(), columns=None,
+                                          header=None, null_values=[""])
+    r_mem = pyoracle.run_pipeline(rows, zillow_ops(), columns=ZILLOW_COLS)
+    assert r_csv["output"] == r_mem["output"]
+    assert r_csv["exception_counts"] == r_mem["exception_counts"]
+    assert len(r_csv["output"]) > 20
+
+
+def test_oracle_csv_tocsv_text():
+    data, rows = make_zillow_csv_bytes(200, seed=5, dirty_frac=0.0)
+    r = pyoracle_csv.run_csv_pipeline(data, zillow_ops(), sink="csv")
+    text = r["csv_text"].decode()
+    lines = text.strip().split("\n")
+    assert lines[0].startswith("url,zipcode,address,city,state,bedrooms")
+    assert len(lines) == len(r["output"]) + 1
+
+
+def test_compile_zillow_csv_source():
+    sp = plan.build_stage(
+        [T.STR, T.STR, T.STR, T.STR, T.F64, T.STR, T.STR, T.STR, T.STR, T.STR],
+        ZILLOW_COLS, zillow_ops())
+    assert sp.compilable, sp.why_not_compilable
+    _compile_only(sp, source="csv", sink="csv")
+    _compile_only(sp, source="csv", sink="mem")
+
+
+def test_try_f64_quirks():
+    assert csvio.try_f64("-")      # fast_atod quirk: '-' -> -0.0
+    assert csvio.try_f64(".")
+    assert csvio.try_f64("e5")
+    assert csvio.try_f64("nan") and csvio.try_f64("INF")
+    assert not csvio.try_f64("-nan")   # sign consumes -> nan match disabled
+    assert not csvio.try_f64("abc")
+    assert not csvio.try_f64("")
+    assert csvio.try_f64("1801.0") and csvio.try_f64("-1e3")

@@ -104,10 +104,11 @@ class StageCodegen:
     """Emits the full stage source. `row_ctx` maps TIR input index -> (var, type,
     nullvar or None) for the operator currently being emitted."""
 
-    def __init__(self, sp, source="mem", sink="mem"):
+    def __init__(self, sp, source="mem", sink="mem", csv_info=None):
         self.sp = sp
         self.source = source
         self.sink = sink
+        self.csv_info = csv_info or {}  # {"null_values": [...]}
         self.lits = {}  # python str -> lit var name
         self.lit_defs = []
 
@@ -528,6 +529,9 @@ class StageCodegen:
         body.w("o.keep = true;")
         body.w("return 0;")
 
+        # generate kernels BEFORE assembling (they may add string literals)
+        main_src = self._main_kernel(in_types, out_types)
+        write_src = self._write_kernel(out_types)
         src = [
             "// generated by tuplex_amd.codegen — stage %s" % sp.signature(),
             runtime_header(),
@@ -544,8 +548,8 @@ class StageCodegen:
         ] + body.lines + [
             "}",
             "",
-            self._main_kernel(in_types, out_types),
-            self._write_kernel(out_types),
+            main_src,
+            write_src,
         ]
         desc = self._desc(in_types, out_types)
         return "\n".join(src), desc
@@ -579,8 +583,41 @@ class StageCodegen:
         L.append("  long long stride = (long long)gridDim.x * blockDim.x;")
         L.append("  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;"
                  " i < n; i += stride) {")
+        if self.source == "csv":
+            L.extend(self._load_inputs_csv(in_types))
+        else:
+            L.extend(self._load_inputs_mem(in_types))
+        L.append("    Out o;")
+        args = []
+        for idx, t in enumerate(in_types):
+            args.append("c%d" % idx)
+            if T.is_opt(t):
+                args.append("c%d_n" % idx)
+        L.append("    long long rc = prc;")
+        L.append("    if (!rc) rc = tpx_process(%s, heap, o);" % ", ".join(args))
+        L.append("    if (rc != 0) {")
+        L.append("      unsigned long long e = atomicAdd(exc_count, 1ULL);")
+        L.append("      if (e < exc_cap) {")
+        L.append("        exc_buf[e*5+0] = row0 + i;")
+        L.append("        exc_buf[e*5+1] = rc & 0xFFFFFFFFLL;")
+        L.append("        exc_buf[e*5+2] = rc >> 32;")
+        L.append("        exc_buf[e*5+3] = in_offs[i];")
+        L.append("        exc_buf[e*5+4] = in_offs[i+1];")
+        L.append("      }")
+        L.append("      keep[i] = 0; keep01[i] = 0; sizes[i] = 0; continue;")
+        L.append("    }")
+        L.append("    if (!o.keep) { keep[i] = 0; keep01[i] = 0; sizes[i] = 0; continue; }")
+        L.append("    keep[i] = 1; keep01[i] = 1;")
+        L.extend(self._store_columnar(out_types))
+        L.append("  }")
+        L.append("}")
+        return "\n".join(L)
+
+    def _load_inputs_mem(self, in_types):
+        """Deserialize one reference-layout row (Serializer.cc:20-24) into typed
+        locals c0..cN."""
+        L = ["    long long prc = 0;  // no pre-parse errors on the mem path"]
         L.append("    const unsigned char* row = in_data + in_offs[i];")
-        # deserialize per input schema (Serializer.cc layout)
         n_opt = sum(1 for t in in_types if T.is_opt(t))
         bitmap_size = ((n_opt + 63) // 64) * 8 if n_opt else 0
         L.append("    // deserialize (bitmap %dB, %d slots)" % (bitmap_size, len(in_types)))
@@ -609,28 +646,67 @@ class StageCodegen:
                     L.append("    if (c%d_n) c%d = tstr{(const char*)row, 0};" % (idx, idx))
             else:
                 raise CodegenError("input type %r" % (t,))
-        L.append("    Out o;")
-        args = []
-        for idx, t in enumerate(in_types):
-            args.append("c%d" % idx)
-            if T.is_opt(t):
-                args.append("c%d_n" % idx)
-        L.append("    long long rc = tpx_process(%s, heap, o);" % ", ".join(args + []))
-        L.append("    if (rc != 0) {")
-        L.append("      unsigned long long e = atomicAdd(exc_count, 1ULL);")
-        L.append("      if (e < exc_cap) {")
-        L.append("        exc_buf[e*3+0] = row0 + i;")
-        L.append("        exc_buf[e*3+1] = rc & 0xFFFFFFFFLL;")
-        L.append("        exc_buf[e*3+2] = rc >> 32;")
-        L.append("      }")
-        L.append("      keep[i] = 0; keep01[i] = 0; sizes[i] = 0; continue;")
+        return L
+
+    def _load_inputs_csv(self, in_types):
+        """Single-pass cell split + typed parse specialised to the sniffed schema —
+        the CSVParseRowGenerator.cc replacement. Structure/parse failures set prc
+        (BADPARSE/UNDERRUN/OVERRUN; raw line becomes the exception payload, like
+        the reference's BADPARSE_STRING_INPUT rows)."""
+        nc = len(in_types)
+        null_values = self.csv_info.get("null_values", [""])
+        L = []
+        L.append("    long long prc = 0;")
+        L.append("    const char* rp = (const char*)in_data + in_offs[i];")
+        L.append("    const char* rend = (const char*)in_data + in_offs[i+1];")
+        L.append("    if (rend > rp && rend[-1] == '\\n') --rend;")
+        L.append("    if (rend > rp && rend[-1] == '\\r') --rend;")
+        L.append("    tpx_cell cells[%d];" % nc)
+        L.append("    { bool m = true; int k = 0;")
+        L.append("      const char* cur = rp;")
+        L.append("      while (k < %d) { cur = tpx_csv_next_cell(cur, rend, &cells[k], &m);"
+                 " ++k; if (!m) break; }" % nc)
+        L.append("      if (k < %d) prc = %d;          // CSV_UNDERRUN" % (nc, 20))
+        L.append("      else if (m) prc = %d;          // CSV_OVERRUN" % 21)
+        L.append("      int badf = 0;")
+        L.append("      for (int j = 0; j < k; ++j) badf |= cells[j].flags;")
+        L.append("      if (!prc && (badf & 6)) prc = %d;  // BADPARSE (escapes -> host)" % 70)
         L.append("    }")
-        L.append("    if (!o.keep) { keep[i] = 0; keep01[i] = 0; sizes[i] = 0; continue; }")
-        L.append("    keep[i] = 1; keep01[i] = 1;")
-        L.extend(self._store_columnar(out_types))
-        L.append("  }")
-        L.append("}")
-        return "\n".join(L)
+
+        def null_check(idx):
+            cv = "tstr{cells[%d].p, cells[%d].n}" % (idx, idx)
+            checks = " || ".join("tpx_streq(%s, %s)" % (cv, self.lit(nv))
+                                 for nv in null_values) or "false"
+            return checks
+
+        for idx, t in enumerate(in_types):
+            base = T.deopt(t)
+            opt = T.is_opt(t)
+            if opt:
+                L.append("    bool c%d_n = !prc && (%s);" % (idx, null_check(idx)))
+            if base == T.STR:
+                L.append("    tstr c%d{cells[%d].p, cells[%d].n};" % (idx, idx, idx))
+                L.append("    if (prc) c%d = tstr{rp, 0};" % idx)
+                if opt:
+                    L.append("    if (c%d_n) c%d = tstr{rp, 0};" % (idx, idx))
+            elif base == T.I64:
+                L.append("    long long c%d = 0;" % idx)
+                guard = ("!prc && !c%d_n" % idx) if opt else "!prc"
+                L.append("    if (%s && tpx_cell_i64(cells[%d], &c%d) != 0) prc = %d;"
+                         % (guard, idx, idx, 70))
+            elif base == T.F64:
+                L.append("    double c%d = 0.0;" % idx)
+                guard = ("!prc && !c%d_n" % idx) if opt else "!prc"
+                L.append("    if (%s && tpx_cell_f64(cells[%d], &c%d) != 0) prc = %d;"
+                         % (guard, idx, idx, 70))
+            elif base == T.BOOL:
+                L.append("    bool c%d = false;" % idx)
+                guard = ("!prc && !c%d_n" % idx) if opt else "!prc"
+                L.append("    if (%s && tpx_cell_bool(cells[%d], &c%d) != 0) prc = %d;"
+                         % (guard, idx, idx, 70))
+            else:
+                raise CodegenError("csv input type %r" % (t,))
+        return L
 
     def _store_columnar(self, out_types):
         """Store Out o -> columnar arrays + per-row serialized size (mem sink) or
@@ -828,5 +904,5 @@ class StageCodegen:
         return "\n".join(lines) + "\n"
 
 
-def generate_stage(sp, source="mem", sink="mem"):
-    return StageCodegen(sp, source, sink).generate()
+def generate_stage(sp, source="mem", sink="mem", csv_info=None):
+    return StageCodegen(sp, source, sink, csv_info).generate()

@@ -128,7 +128,7 @@ struct tpx_stage {
     hipModule_t module = nullptr;
     hipFunction_t k_main = nullptr, k_write = nullptr;
     hipFunction_t k_scan_block = nullptr, k_scan_add = nullptr;
-    hipFunction_t k_csv_chunk = nullptr, k_csv_rows = nullptr, k_csv_cells = nullptr;
+    hipFunction_t k_csv_chunk = nullptr, k_csv_sel = nullptr, k_csv_rows = nullptr;
     bool loaded = false;
 };
 
@@ -219,8 +219,8 @@ extern "C" tpx_stage* tpx_stage_compile(const char* hip_source,
         {"tpx_scan_block", &st->k_scan_block, true},
         {"tpx_scan_add", &st->k_scan_add, true},
         {"tpx_csv_chunk_stats", &st->k_csv_chunk, false},
+        {"tpx_csv_select_counts", &st->k_csv_sel, false},
         {"tpx_csv_emit_rows", &st->k_csv_rows, false},
-        {"tpx_csv_cells", &st->k_csv_cells, false},
     };
     for (auto& e : lut) {
         hipError_t r = hipModuleGetFunction(e.fn, st->module, e.name);
@@ -308,65 +308,27 @@ static size_t col_slot_bytes(const ColDesc& c, long long n, int slot) {
 }
 
 struct ExcRec {
-    long long row, ec, opid;
+    long long row, ec, opid, off_start, off_end;  // offsets into the input bytes
 };
 
 // ---------------------------------------------------------------------------------
-// tpx_stage_execute (mem source)
+// shared stage-execution core (mem + csv sources share the fixed kernel signature:
+// tpx_stage_main(in_data, in_offs, n, row0, heap..., keep..., exc..., outv))
 
-extern "C" int64_t tpx_stage_execute(tpx_stage* st, const tpx_partition* parts,
-                                     int64_t n_parts, tpx_result* res) {
-    memset(res, 0, sizeof(*res));
-    if (!st->loaded) { set_err("stage not loaded on a GPU"); return -1; }
+struct PayloadSrc {
+    const uint8_t* bytes;         // flat input bytes (csv) — off_start/off_end index here
+    const tpx_partition* parts;   // or: partition list (mem)
+    int64_t n_parts;
+};
+
+static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
+                        long long row0, long long in_bytes, tpx_result* res,
+                        const PayloadSrc& psrc, hipStream_t stream) {
     const StageDesc& D = st->desc;
-    hipStream_t stream = nullptr;  // default stream
-
-    // concatenate partition geometry
-    long long n = 0;
-    long long in_bytes = 0;
-    for (int64_t p = 0; p < n_parts; ++p) {
-        n += parts[p].num_rows;
-        in_bytes += parts[p].size;
-    }
-    res->bytes_in = in_bytes;
-    if (n == 0) {
-        res->out_data = (uint8_t*)malloc(8);
-        memset(res->out_data, 0, 8);
-        res->out_size = 8;
-        res->out_row_offsets = (int64_t*)malloc(8);
-        res->out_row_offsets[0] = 8;
-        return 0;
-    }
-
-    hipEvent_t ev0, ev1, ev2, ev3;
-    hipEventCreate(&ev0); hipEventCreate(&ev1);
-    hipEventCreate(&ev2); hipEventCreate(&ev3);
-    hipEventRecord(ev0, stream);
-
-    // upload rows + offsets (rebased into the concatenated buffer)
-    DevBuf d_in, d_offs;
-    HIP_CHECK(d_in.alloc((size_t)in_bytes));
-    HIP_CHECK(d_offs.alloc((size_t)(n + 1) * 8));
-    {
-        std::vector<long long> offs((size_t)n + 1);
-        long long byte_base = 0, row_base = 0;
-        for (int64_t p = 0; p < n_parts; ++p) {
-            const tpx_partition& P = parts[p];
-            HIP_CHECK(hipMemcpyAsync((char*)d_in.p + byte_base, P.data,
-                                     (size_t)P.size, hipMemcpyHostToDevice, stream));
-            for (long long r = 0; r < P.num_rows; ++r)
-                offs[(size_t)(row_base + r)] = byte_base + P.row_offsets[r];
-            byte_base += P.size;
-            row_base += P.num_rows;
-        }
-        offs[(size_t)n] = byte_base;
-        HIP_CHECK(hipMemcpyAsync(d_offs.p, offs.data(), ((size_t)n + 1) * 8,
-                                 hipMemcpyHostToDevice, stream));
-        HIP_CHECK(hipStreamSynchronize(stream));
-    }
+    hipEvent_t ev1, ev2, ev3;
+    hipEventCreate(&ev1); hipEventCreate(&ev2); hipEventCreate(&ev3);
     hipEventRecord(ev1, stream);
 
-    // output columnar buffers (3 slots per column)
     int nout = (int)D.out_cols.size();
     std::vector<DevBuf> colbufs((size_t)nout * 3);
     std::vector<void*> outv((size_t)nout * 3, nullptr);
@@ -379,7 +341,7 @@ extern "C" int64_t tpx_stage_execute(tpx_stage* st, const tpx_partition* parts,
             }
         }
     DevBuf d_outv;
-    HIP_CHECK(d_outv.alloc(outv.size() * sizeof(void*)));
+    HIP_CHECK(d_outv.alloc(outv.size() * sizeof(void*) + 8));
     HIP_CHECK(hipMemcpyAsync(d_outv.p, outv.data(), outv.size() * sizeof(void*),
                              hipMemcpyHostToDevice, stream));
 
@@ -388,7 +350,7 @@ extern "C" int64_t tpx_stage_execute(tpx_stage* st, const tpx_partition* parts,
     HIP_CHECK(d_keep01.alloc((size_t)n * 8));
     HIP_CHECK(d_sizes.alloc((size_t)n * 8));
     unsigned long long exc_cap = (unsigned long long)std::min<long long>(n, 1 << 20);
-    HIP_CHECK(d_exc.alloc((size_t)exc_cap * 24));
+    HIP_CHECK(d_exc.alloc((size_t)exc_cap * sizeof(ExcRec)));
     HIP_CHECK(d_exc_count.alloc(8));
     unsigned long long heap_cap =
         (unsigned long long)std::max<long long>(2 * in_bytes + (16 << 20), 1 << 20);
@@ -399,10 +361,8 @@ extern "C" int64_t tpx_stage_execute(tpx_stage* st, const tpx_partition* parts,
     for (int attempt = 0;; ++attempt) {
         HIP_CHECK(hipMemsetAsync(d_exc_count.p, 0, 8, stream));
         HIP_CHECK(hipMemsetAsync(d_heap_cursor.p, 0, 8, stream));
-        long long row0 = 0;
-        unsigned grid =
-            (unsigned)std::min<long long>((n + 255) / 256, 8192);
-        void* args[] = {&d_in.p, &d_offs.p, &n, &row0, &d_heap.p, &d_heap_cursor.p,
+        unsigned grid = (unsigned)std::min<long long>((n + 255) / 256, 8192);
+        void* args[] = {&d_in, &d_offs, &n, &row0, &d_heap.p, &d_heap_cursor.p,
                         &heap_cap, &d_keep.p, &d_keep01.p, &d_sizes.p, &d_exc.p,
                         &d_exc_count.p, &exc_cap, &d_outv.p};
         if (launch(st->k_main, grid, 256, stream, args)) return -1;
@@ -423,11 +383,11 @@ extern "C" int64_t tpx_stage_execute(tpx_stage* st, const tpx_partition* parts,
         if (exc_overflow) {
             hipFree(d_exc.p); d_exc.p = nullptr;
             exc_cap = exc_count + 1024;
-            HIP_CHECK(d_exc.alloc((size_t)exc_cap * 24));
+            HIP_CHECK(d_exc.alloc((size_t)exc_cap * sizeof(ExcRec)));
         }
     }
 
-    // scans
+    // compaction scans
     DevBuf d_keep_scan, d_size_scan;
     HIP_CHECK(d_keep_scan.alloc((size_t)n * 8));
     HIP_CHECK(d_size_scan.alloc((size_t)n * 8));
@@ -439,7 +399,6 @@ extern "C" int64_t tpx_stage_execute(tpx_stage* st, const tpx_partition* parts,
                  &total_bytes))
         return -1;
 
-    // serialize output (mem sink: row container; csv sink: text)
     bool mem_sink = D.sink == "mem";
     long long out_total = mem_sink ? 8 + total_bytes : total_bytes;
     DevBuf d_out, d_out_offs, d_out_rowidx;
@@ -448,7 +407,6 @@ extern "C" int64_t tpx_stage_execute(tpx_stage* st, const tpx_partition* parts,
     HIP_CHECK(d_out_rowidx.alloc(((size_t)total_rows + 1) * 8));
     {
         unsigned grid = (unsigned)std::min<long long>((n + 255) / 256, 8192);
-        long long row0 = 0;
         void* args[] = {&d_keep.p, &d_keep_scan.p, &d_size_scan.p, &n, &row0,
                         &d_outv.p, &d_out.p, &d_out_offs.p, &d_out_rowidx.p,
                         &total_rows, &total_bytes};
@@ -464,41 +422,50 @@ extern "C" int64_t tpx_stage_execute(tpx_stage* st, const tpx_partition* parts,
     res->out_num_rows = total_rows;
     res->out_row_offsets = (int64_t*)malloc(((size_t)total_rows + 1) * 8);
     HIP_CHECK(hipMemcpyAsync(res->out_row_offsets, d_out_offs.p,
-                             ((size_t)total_rows + 1) * 8,
-                             hipMemcpyDeviceToHost, stream));
+                             ((size_t)total_rows + 1) * 8, hipMemcpyDeviceToHost,
+                             stream));
     res->out_row_indices = (int64_t*)malloc(((size_t)total_rows + 1) * 8);
-    HIP_CHECK(hipMemcpyAsync(res->out_row_indices, d_out_rowidx.p,
-                             ((size_t)total_rows) * 8,
-                             hipMemcpyDeviceToHost, stream));
+    if (total_rows)
+        HIP_CHECK(hipMemcpyAsync(res->out_row_indices, d_out_rowidx.p,
+                                 (size_t)total_rows * 8, hipMemcpyDeviceToHost,
+                                 stream));
     std::vector<ExcRec> excs((size_t)exc_count);
     if (exc_count)
-        HIP_CHECK(hipMemcpyAsync(excs.data(), d_exc.p, (size_t)exc_count * 24,
+        HIP_CHECK(hipMemcpyAsync(excs.data(), d_exc.p,
+                                 (size_t)exc_count * sizeof(ExcRec),
                                  hipMemcpyDeviceToHost, stream));
     HIP_CHECK(hipStreamSynchronize(stream));
     hipEventRecord(ev3, stream);
     HIP_CHECK(hipEventSynchronize(ev3));
 
-    // exception buffer in reference format (payload = input row bytes)
+    // exception buffer in the reference record format (IExceptionableTask.h:20)
     if (exc_count) {
         std::sort(excs.begin(), excs.end(),
                   [](const ExcRec& a, const ExcRec& b) { return a.row < b.row; });
-        // row -> partition lookup
-        std::vector<long long> row_base((size_t)n_parts + 1);
-        row_base[0] = 0;
-        for (int64_t p = 0; p < n_parts; ++p)
-            row_base[(size_t)p + 1] = row_base[(size_t)p] + parts[p].num_rows;
+        std::vector<long long> row_base;
+        if (!psrc.bytes) {
+            row_base.resize((size_t)psrc.n_parts + 1);
+            row_base[0] = 0;
+            for (int64_t p = 0; p < psrc.n_parts; ++p)
+                row_base[(size_t)p + 1] = row_base[(size_t)p] + psrc.parts[p].num_rows;
+        }
         size_t total = 0;
         std::vector<std::pair<const uint8_t*, long long>> payloads(excs.size());
         for (size_t i = 0; i < excs.size(); ++i) {
-            long long r = excs[i].row;
-            int64_t p = (int64_t)(std::upper_bound(row_base.begin(), row_base.end(), r)
-                        - row_base.begin()) - 1;
-            long long lr = r - row_base[(size_t)p];
-            const tpx_partition& P = parts[p];
-            const uint8_t* rp = P.data + P.row_offsets[lr];
-            long long rs = P.row_offsets[lr + 1] - P.row_offsets[lr];
-            payloads[i] = {rp, rs};
-            total += 32 + (size_t)rs;
+            if (psrc.bytes) {
+                payloads[i] = {psrc.bytes + excs[i].off_start,
+                               excs[i].off_end - excs[i].off_start};
+            } else {
+                long long r = excs[i].row - row0;
+                int64_t p = (int64_t)(std::upper_bound(row_base.begin(),
+                                                       row_base.end(), r) -
+                                      row_base.begin()) - 1;
+                long long lr = r - row_base[(size_t)p];
+                const tpx_partition& P = psrc.parts[p];
+                payloads[i] = {P.data + P.row_offsets[lr],
+                               P.row_offsets[lr + 1] - P.row_offsets[lr]};
+            }
+            total += 32 + (size_t)payloads[i].second;
         }
         res->exc_data = (uint8_t*)malloc(total);
         uint8_t* w = res->exc_data;
@@ -516,21 +483,143 @@ extern "C" int64_t tpx_stage_execute(tpx_stage* st, const tpx_partition* parts,
     }
     res->bytes_out = out_total;
     float ms = 0;
-    hipEventElapsedTime(&ms, ev0, ev1); res->t_h2d_ms = ms;
     hipEventElapsedTime(&ms, ev1, ev2); res->t_kernel_ms = ms;
     hipEventElapsedTime(&ms, ev2, ev3); res->t_d2h_ms = ms;
-    hipEventDestroy(ev0); hipEventDestroy(ev1);
-    hipEventDestroy(ev2); hipEventDestroy(ev3);
+    (void)hipEventDestroy(ev1); (void)hipEventDestroy(ev2); (void)hipEventDestroy(ev3);
     return 0;
 }
+
+static void empty_result(tpx_result* res, bool mem_sink) {
+    if (mem_sink) {
+        res->out_data = (uint8_t*)malloc(8);
+        memset(res->out_data, 0, 8);
+        res->out_size = 8;
+    } else {
+        res->out_data = (uint8_t*)malloc(1);
+        res->out_size = 0;
+    }
+    res->out_row_offsets = (int64_t*)malloc(8);
+    res->out_row_offsets[0] = mem_sink ? 8 : 0;
+    res->out_row_indices = (int64_t*)malloc(8);
+}
+
+// ---------------------------------------------------------------------------------
+// mem source (TransformTask.cc:682 processMemorySource analog)
+
+extern "C" int64_t tpx_stage_execute(tpx_stage* st, const tpx_partition* parts,
+                                     int64_t n_parts, tpx_result* res) {
+    memset(res, 0, sizeof(*res));
+    if (!st->loaded) { set_err("stage not loaded on a GPU"); return -1; }
+    hipStream_t stream = nullptr;
+
+    long long n = 0, in_bytes = 0;
+    for (int64_t p = 0; p < n_parts; ++p) {
+        n += parts[p].num_rows;
+        in_bytes += parts[p].size;
+    }
+    res->bytes_in = in_bytes;
+    if (n == 0) { empty_result(res, st->desc.sink == "mem"); return 0; }
+
+    hipEvent_t ev0, ev1;
+    hipEventCreate(&ev0); hipEventCreate(&ev1);
+    hipEventRecord(ev0, stream);
+    DevBuf d_in, d_offs;
+    HIP_CHECK(d_in.alloc((size_t)in_bytes));
+    HIP_CHECK(d_offs.alloc((size_t)(n + 1) * 8));
+    {
+        std::vector<long long> offs((size_t)n + 1);
+        long long byte_base = 0, row_base = 0;
+        for (int64_t p = 0; p < n_parts; ++p) {
+            const tpx_partition& P = parts[p];
+            HIP_CHECK(hipMemcpyAsync((char*)d_in.p + byte_base, P.data,
+                                     (size_t)P.size, hipMemcpyHostToDevice, stream));
+            for (long long r = 0; r < P.num_rows; ++r)
+                offs[(size_t)(row_base + r)] = byte_base + P.row_offsets[r];
+            byte_base += P.size;
+            row_base += P.num_rows;
+        }
+        offs[(size_t)n] = byte_base;
+        HIP_CHECK(hipMemcpyAsync(d_offs.p, offs.data(), ((size_t)n + 1) * 8,
+                                 hipMemcpyHostToDevice, stream));
+    }
+    hipEventRecord(ev1, stream);
+    HIP_CHECK(hipEventSynchronize(ev1));
+    float ms = 0;
+    hipEventElapsedTime(&ms, ev0, ev1);
+    res->t_h2d_ms = ms;
+    (void)hipEventDestroy(ev0); (void)hipEventDestroy(ev1);
+
+    PayloadSrc psrc{nullptr, parts, n_parts};
+    return run_core(st, d_in.p, d_offs.p, n, 0, in_bytes, res, psrc, stream);
+}
+
+// ---------------------------------------------------------------------------------
+// csv source (TransformTask.cc:724 processFileSource + CSVReader.cc:390 analog;
+// row boundaries found on-device by the quote-parity scan kernels)
 
 extern "C" int64_t tpx_stage_execute_csv(tpx_stage* st, const uint8_t* csv_bytes,
                                          int64_t size, int64_t first_global_row,
                                          tpx_result* res) {
-    (void)st; (void)csv_bytes; (void)size; (void)first_global_row;
     memset(res, 0, sizeof(*res));
-    set_err("csv source: implemented in a later commit this round");
-    return -1;
+    if (!st->loaded) { set_err("stage not loaded on a GPU"); return -1; }
+    if (!st->k_csv_chunk || !st->k_csv_sel || !st->k_csv_rows) {
+        set_err("stage compiled without csv kernels");
+        return -1;
+    }
+    hipStream_t stream = nullptr;
+    res->bytes_in = size;
+    if (size == 0) { empty_result(res, st->desc.sink == "mem"); return 0; }
+
+    hipEvent_t ev0, ev1;
+    hipEventCreate(&ev0); hipEventCreate(&ev1);
+    hipEventRecord(ev0, stream);
+    DevBuf d_in;
+    HIP_CHECK(d_in.alloc((size_t)size));
+    HIP_CHECK(hipMemcpyAsync(d_in.p, csv_bytes, (size_t)size,
+                             hipMemcpyHostToDevice, stream));
+    hipEventRecord(ev1, stream);
+
+    long long nchunks = (size + 4095) / 4096;  // TPX_CSV_CHUNK
+    DevBuf d_q, d_c0, d_c1, d_qs, d_rc, d_base;
+    HIP_CHECK(d_q.alloc((size_t)nchunks * 8));
+    HIP_CHECK(d_c0.alloc((size_t)nchunks * 8));
+    HIP_CHECK(d_c1.alloc((size_t)nchunks * 8));
+    HIP_CHECK(d_qs.alloc((size_t)nchunks * 8));
+    HIP_CHECK(d_rc.alloc((size_t)nchunks * 8));
+    HIP_CHECK(d_base.alloc((size_t)nchunks * 8));
+    unsigned grid = (unsigned)std::min<long long>((nchunks + 255) / 256, 8192);
+    {
+        void* args[] = {&d_in.p, (void*)&size, &nchunks, &d_q.p, &d_c0.p, &d_c1.p};
+        if (launch(st->k_csv_chunk, grid, 256, stream, args)) return -1;
+    }
+    long long qtotal = 0;
+    if (dev_scan(st, stream, (long long*)d_q.p, (long long*)d_qs.p, nchunks, &qtotal))
+        return -1;
+    {
+        void* args[] = {&d_qs.p, &d_c0.p, &d_c1.p, &d_rc.p, &nchunks};
+        if (launch(st->k_csv_sel, grid, 256, stream, args)) return -1;
+    }
+    long long nrows = 0;
+    if (dev_scan(st, stream, (long long*)d_rc.p, (long long*)d_base.p, nchunks,
+                 &nrows))
+        return -1;
+    if (nrows == 0) { empty_result(res, st->desc.sink == "mem"); return 0; }
+    DevBuf d_offs;
+    HIP_CHECK(d_offs.alloc(((size_t)nrows + 1) * 8));
+    {
+        void* args[] = {&d_in.p, (void*)&size, &nchunks, &d_qs.p, &d_base.p,
+                        &d_offs.p};
+        if (launch(st->k_csv_rows, grid, 256, stream, args)) return -1;
+    }
+    HIP_CHECK(hipEventSynchronize(ev1));
+    float ms = 0;
+    hipEventElapsedTime(&ms, ev0, ev1);
+    res->t_h2d_ms = ms;
+    (void)hipEventDestroy(ev0); (void)hipEventDestroy(ev1);
+
+    PayloadSrc psrc{csv_bytes, nullptr, 0};
+    return run_core(st, d_in.p, d_offs.p, nrows, first_global_row, size, res, psrc,
+                    stream);
 }
 
 extern "C" void tpx_result_free(tpx_result* res) {

@@ -484,6 +484,154 @@ __device__ __forceinline__ char* tpx_csv_cell_write(char* w, const tstr s) {
     return w;
 }
 
+// ---- CSV input: row-boundary detection + cell split ------------------------------
+//
+// Replaces the reference's chunked CSV reading (CSVReader.cc:390; chunk-boundary
+// row-start detection utils/src/CSVUtils.cc:1494 findLineStart) and the generated
+// row parser's scan (CSVParseRowGenerator.cc SSE4.2 spanner :355). RFC-4180 rule:
+// a '\n' is a row boundary iff the number of '"' before it is even (every quote
+// toggles in/out of a quoted cell; an escaped "" toggles twice). Per-chunk quote
+// counts + an exclusive scan give each chunk's start parity exactly — a
+// parallel-exact restatement of the reference's sequential state machine.
+
+#define TPX_CSV_CHUNK 4096
+
+extern "C" __global__ void tpx_csv_chunk_stats(const char* __restrict__ data,
+                                               long long size, long long nchunks,
+                                               long long* __restrict__ q,
+                                               long long* __restrict__ c0,
+                                               long long* __restrict__ c1) {
+    long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long c = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+         c < nchunks; c += stride) {
+        long long a = c * TPX_CSV_CHUNK;
+        long long b = a + TPX_CSV_CHUNK < size ? a + TPX_CSV_CHUNK : size;
+        long long qq = 0, e0 = 0, e1 = 0;
+        for (long long i = a; i < b; ++i) {
+            char ch = data[i];
+            if (ch == '"') ++qq;
+            else if (ch == '\n') { if (qq & 1) ++e1; else ++e0; }
+        }
+        q[c] = qq; c0[c] = e0; c1[c] = e1;
+    }
+}
+
+extern "C" __global__ void tpx_csv_select_counts(const long long* __restrict__ qscan,
+                                                 const long long* __restrict__ c0,
+                                                 const long long* __restrict__ c1,
+                                                 long long* __restrict__ rc,
+                                                 long long nchunks) {
+    long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long c = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+         c < nchunks; c += stride)
+        rc[c] = (qscan[c] & 1) ? c1[c] : c0[c];
+}
+
+extern "C" __global__ void tpx_csv_emit_rows(const char* __restrict__ data,
+                                             long long size, long long nchunks,
+                                             const long long* __restrict__ qscan,
+                                             const long long* __restrict__ base,
+                                             long long* __restrict__ row_offs) {
+    long long stride = (long long)gridDim.x * blockDim.x;
+    long long tid0 = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (tid0 == 0) row_offs[0] = 0;
+    for (long long c = tid0; c < nchunks; c += stride) {
+        long long a = c * TPX_CSV_CHUNK;
+        long long b = a + TPX_CSV_CHUNK < size ? a + TPX_CSV_CHUNK : size;
+        long long p = qscan[c] & 1;
+        long long qq = 0, idx = base[c];
+        for (long long i = a; i < b; ++i) {
+            char ch = data[i];
+            if (ch == '"') ++qq;
+            else if (ch == '\n' && (((qq + p) & 1) == 0)) row_offs[1 + idx++] = i + 1;
+        }
+    }
+}
+
+// one CSV cell starting at p; returns pointer to next cell start. *more = a
+// delimiter was consumed (another cell follows). flags: 1 quoted, 2 contains ""
+// escapes (diverted to host), 4 structurally bad (unterminated quote / junk after
+// closing quote — reference DOUBLEQUOTEERROR class)
+struct tpx_cell { const char* p; long long n; int flags; };
+
+__device__ __forceinline__ const char* tpx_csv_next_cell(const char* p,
+                                                         const char* end,
+                                                         tpx_cell* c, bool* more) {
+    c->flags = 0;
+    *more = false;
+    if (p < end && *p == '"') {
+        const char* s = p + 1;
+        const char* q = s;
+        bool esc = false;
+        while (q < end) {
+            if (*q == '"') {
+                if (q + 1 < end && q[1] == '"') { esc = true; q += 2; continue; }
+                break;
+            }
+            ++q;
+        }
+        if (q >= end) { c->p = p; c->n = end - p; c->flags = 4; return end; }
+        c->p = s; c->n = q - s; c->flags = 1 | (esc ? 2 : 0);
+        ++q;
+        if (q < end && *q != ',') c->flags |= 4;
+        while (q < end && *q != ',') ++q;
+        if (q < end) { *more = true; ++q; }
+        return q;
+    }
+    const char* q = p;
+    while (q < end && *q != ',') ++q;
+    c->p = p; c->n = q - p;
+    if (q < end) { *more = true; ++q; }
+    return q;
+}
+
+// typed cell parse (cells path semantics: python-whitespace trim + fast_atoX,
+// CellSourceTaskBuilder + Runtime.cc:319 wrappers)
+__device__ __forceinline__ int tpx_cell_i64(const tpx_cell& c, long long* out) {
+    const char* a = c.p;
+    const char* b = c.p + c.n;
+    while (a < b && tpx_is_pyws(*a)) ++a;
+    while (b > a && tpx_is_pyws(*(b - 1))) --b;
+    return tpx_fast_atoi64(a, b, out);
+}
+
+__device__ __forceinline__ int tpx_cell_f64(const tpx_cell& c, double* out) {
+    const char* a = c.p;
+    const char* b = c.p + c.n;
+    while (a < b && tpx_is_pyws(*a)) ++a;
+    while (b > a && tpx_is_pyws(*(b - 1))) --b;
+    return tpx_fast_atod(a, b, out);
+}
+
+// fast_atob semantics (StringUtils.cc:186): case-insensitive
+// true/t/yes/y/1 | false/f/no/n/0
+#define EC_BOOLPARSE_ 54
+__device__ __forceinline__ int tpx_cell_bool(const tpx_cell& c, bool* out) {
+    const char* a = c.p;
+    const char* b = c.p + c.n;
+    while (a < b && tpx_is_pyws(*a)) ++a;
+    while (b > a && tpx_is_pyws(*(b - 1))) --b;
+    long long n = b - a;
+    char buf[8];
+    if (n < 1 || n > 5) return EC_BOOLPARSE_;
+    for (long long i = 0; i < n; ++i) {
+        char ch = a[i];
+        buf[i] = (ch >= 'A' && ch <= 'Z') ? ch + 32 : ch;
+    }
+    auto eq = [&](const char* s, long long l) {
+        if (n != l) return false;
+        for (long long i = 0; i < l; ++i) if (buf[i] != s[i]) return false;
+        return true;
+    };
+    if (eq("true", 4) || eq("t", 1) || eq("yes", 3) || eq("y", 1) || eq("1", 1)) {
+        *out = true; return EC_OK;
+    }
+    if (eq("false", 5) || eq("f", 1) || eq("no", 2) || eq("n", 1) || eq("0", 1)) {
+        *out = false; return EC_OK;
+    }
+    return EC_BOOLPARSE_;
+}
+
 // ---- fixed kernels ---------------------------------------------------------------
 
 // exclusive-scan building block: per-block scan of ITEMS_PER_BLOCK i64 items.
