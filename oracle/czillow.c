@@ -1,0 +1,351 @@
+/* ORACLE (TEST INFRASTRUCTURE ONLY) — single-thread C port of the fused Zillow
+ * TransformStage fast path, used as bench.py's cpu_baseline (kind "port",
+ * cores 1). The reference itself cannot be built in this container (SURVEY.md
+ * §8c); this follows the repo's own hand-written-C++ baseline pattern
+ * (benchmarks/zillow/Z1/baseline/zillow.cpp, csvmonkey-based) but restates the
+ * semantics from the reference sources:
+ *  - CSV row/cell split: RFC-4180 quote-parity rules (CSVReader.cc:390,
+ *    CSVUtils.cc:1494, CSVParseRowGenerator.cc);
+ *  - parse: StringUtils.cc:22 fast_atoi64 / :71 fast_atod + Runtime.cc:319 trim;
+ *  - pipeline: benchmarks/zillow/Z1/runtuplex.py:192-205 operator chain
+ *    (withColumn/filter/mapColumn/selectColumns with the extract* UDFs);
+ *  - output: RFC-4180 cell quoting (PipelineBuilder.h:238 CSV writer semantics).
+ * Exception rows (bad parse / UDF error) are counted and skipped — the baseline
+ * times the fast path, like the reference's JobMetrics fast-path wall time.
+ *
+ * build: gcc -O3 -march=native -o czillow czillow.c   (oracle/Makefile)
+ * usage: czillow <csv-file> [min_seconds]
+ *        processes the file repeatedly until >= min_seconds of work, prints one
+ *        JSON line {rows, rows_out, exceptions, bytes_in, bytes_out, seconds,
+ *        rows_per_s}.
+ */
+#include <stdio.h>
+#include <stdlib.h>
+#include <string.h>
+#include <stdint.h>
+#include <time.h>
+
+typedef struct { const char* p; long n; } str_t;
+
+static int is_ws(char c) {
+    return c == ' ' || c == '\t' || c == '\n' || c == '\r' || c == '\v' || c == '\f';
+}
+
+/* StringUtils.cc:22 fast_atoi64 (with Runtime.cc:319 trim in caller) */
+static int fast_atoi64(const char* start, const char* end, long long* out) {
+    if (start == end) return 50;
+    long long x = 0;
+    const char* p = start;
+    int neg = 0;
+    if (*p == '-') { neg = 1; ++p; }
+    while (p < end && *p >= '0' && *p <= '9') { x = x * 10 + (*p - '0'); ++p; }
+    if (p != end) return 52;
+    *out = neg ? -x : x;
+    return 0;
+}
+
+static int trim_atoi64(str_t s, long long* out) {
+    const char* a = s.p;
+    const char* b = s.p + s.n;
+    while (a < b && is_ws(*a)) ++a;
+    while (b > a && is_ws(*(b - 1))) --b;
+    return fast_atoi64(a, b, out);
+}
+
+static long sfind(str_t s, const char* needle, long nn) {
+    for (long i = 0; i + nn <= s.n; ++i)
+        if (!memcmp(s.p + i, needle, (size_t)nn)) return i;
+    return -1;
+}
+
+static long srfind(str_t s, const char* needle, long nn) {
+    for (long i = s.n - nn; i >= 0; --i)
+        if (!memcmp(s.p + i, needle, (size_t)nn)) return i;
+    return -1;
+}
+
+static str_t sslice(str_t s, long lo, long hi) {
+    if (lo < 0) lo += s.n;
+    if (hi < 0) hi += s.n;
+    if (lo < 0) lo = 0;
+    if (lo > s.n) lo = s.n;
+    if (hi < 0) hi = 0;
+    if (hi > s.n) hi = s.n;
+    if (hi < lo) hi = lo;
+    str_t r = {s.p + lo, hi - lo};
+    return r;
+}
+
+static int scontains(str_t s, const char* needle) {
+    return sfind(s, needle, (long)strlen(needle)) >= 0;
+}
+
+/* lower() into scratch */
+static str_t slower(str_t s, char* scratch) {
+    for (long i = 0; i < s.n; ++i) {
+        char c = s.p[i];
+        scratch[i] = (c >= 'A' && c <= 'Z') ? c + 32 : c;
+    }
+    str_t r = {scratch, s.n};
+    return r;
+}
+
+/* replace(",", "") into scratch */
+static str_t sstripcommas(str_t s, char* scratch) {
+    long w = 0;
+    for (long i = 0; i < s.n; ++i)
+        if (s.p[i] != ',') scratch[w++] = s.p[i];
+    str_t r = {scratch, w};
+    return r;
+}
+
+/* ---- zillow UDFs (runtuplex.py:12-110 semantics) ------------------------------ */
+
+static int extract_bd(str_t facts, long long* out) {
+    long max_idx = sfind(facts, " bd", 3);
+    if (max_idx < 0) max_idx = facts.n;
+    str_t s = sslice(facts, 0, max_idx);
+    long split_idx = srfind(s, ",", 1);
+    split_idx = split_idx < 0 ? 0 : split_idx + 2;
+    return trim_atoi64(sslice(s, split_idx, s.n), out);
+}
+
+static int extract_ba(str_t facts, long long* out) {
+    long max_idx = sfind(facts, " ba", 3);
+    if (max_idx < 0) max_idx = facts.n;
+    str_t s = sslice(facts, 0, max_idx);
+    long split_idx = srfind(s, ",", 1);
+    split_idx = split_idx < 0 ? 0 : split_idx + 2;
+    return trim_atoi64(sslice(s, split_idx, s.n), out);
+}
+
+static int extract_sqft(str_t facts, long long* out, char* scratch) {
+    long max_idx = sfind(facts, " sqft", 5);
+    if (max_idx < 0) max_idx = facts.n;
+    str_t s = sslice(facts, 0, max_idx);
+    long split_idx = srfind(s, "ba ,", 4);
+    split_idx = split_idx < 0 ? 0 : split_idx + 5;
+    str_t r = sstripcommas(sslice(s, split_idx, s.n), scratch);
+    return trim_atoi64(r, out);
+}
+
+/* returns offer id: 0 sale, 1 rent, 2 sold, 3 foreclosed, 4 other */
+static int extract_offer(str_t title, char* scratch) {
+    str_t t = slower(title, scratch);
+    if (scontains(t, "sale")) return 0;
+    if (scontains(t, "rent")) return 1;
+    if (scontains(t, "sold")) return 2;
+    if (scontains(t, "foreclose")) return 3;
+    return 4;
+}
+
+/* 0 condo, 1 house, 2 unknown */
+static int extract_type(str_t title, char* scratch) {
+    str_t t = slower(title, scratch);
+    int type = 2;
+    if (scontains(t, "condo") || scontains(t, "apartment")) type = 0;
+    if (scontains(t, "house")) type = 1;
+    return type;
+}
+
+static int extract_price(str_t price, int offer, str_t facts, long long sqft,
+                         long long* out, char* scratch) {
+    if (offer == 2) { /* sold */
+        long i = sfind(facts, "Price/sqft:", 11);
+        str_t s = sslice(facts, i + 11 + 1, facts.n);
+        long d = sfind(s, "$", 1);
+        long c = sfind(s, ", ", 2);
+        str_t r = sslice(s, d + 1, c - 1);
+        long long pps;
+        if (trim_atoi64(r, &pps)) return 1;
+        *out = pps * sqft;
+        return 0;
+    } else if (offer == 1) { /* rent */
+        long max_idx = srfind(price, "/", 1);
+        str_t r = sstripcommas(sslice(price, 1, max_idx), scratch);
+        return trim_atoi64(r, out);
+    }
+    str_t r = sstripcommas(sslice(price, 1, price.n), scratch);
+    return trim_atoi64(r, out);
+}
+
+/* ---- csv ----------------------------------------------------------------------- */
+
+#define NCOLS 10
+static const char* OFFER_STR[5] = {"sale", "rent", "sold", "foreclosed", "?"};
+static const char* TYPE_STR[3] = {"condo", "house", "unknown"};
+
+/* split one row's cells; returns 0 ok (cells filled), nonzero bad */
+static int split_cells(const char* p, const char* end, str_t* cells) {
+    int k = 0;
+    int more = 1;
+    while (more) {
+        more = 0;
+        if (k >= NCOLS) return 1;
+        if (p < end && *p == '"') {
+            const char* s = p + 1;
+            const char* q = s;
+            int esc = 0;
+            while (q < end) {
+                if (*q == '"') {
+                    if (q + 1 < end && q[1] == '"') { esc = 1; q += 2; continue; }
+                    break;
+                }
+                ++q;
+            }
+            if (q >= end || esc) return 1;
+            cells[k].p = s;
+            cells[k].n = q - s;
+            ++k;
+            ++q;
+            if (q < end && *q != ',') return 1;
+            if (q < end) { more = 1; ++q; }
+            p = q;
+        } else {
+            const char* q = p;
+            while (q < end && *q != ',') ++q;
+            cells[k].p = p;
+            cells[k].n = q - p;
+            ++k;
+            if (q < end) { more = 1; ++q; }
+            p = q;
+        }
+    }
+    return k != NCOLS;
+}
+
+static char* out_cell(char* w, str_t s) {
+    int need = 0;
+    for (long i = 0; i < s.n; ++i) {
+        char c = s.p[i];
+        if (c == ',' || c == '"' || c == '\n' || c == '\r') { need = 1; break; }
+    }
+    if (!need) {
+        memcpy(w, s.p, (size_t)s.n);
+        return w + s.n;
+    }
+    *w++ = '"';
+    for (long i = 0; i < s.n; ++i) {
+        *w++ = s.p[i];
+        if (s.p[i] == '"') *w++ = '"';
+    }
+    *w++ = '"';
+    return w;
+}
+
+static char* out_i64(char* w, long long v) {
+    return w + sprintf(w, "%lld", v);
+}
+
+int main(int argc, char** argv) {
+    if (argc < 2) { fprintf(stderr, "usage: czillow <csv> [min_seconds]\n"); return 2; }
+    double min_seconds = argc > 2 ? atof(argv[2]) : 10.0;
+
+    FILE* f = fopen(argv[1], "rb");
+    if (!f) { perror("open"); return 2; }
+    fseek(f, 0, SEEK_END);
+    long long size = ftell(f);
+    fseek(f, 0, SEEK_SET);
+    char* data = (char*)malloc((size_t)size + 1);
+    if (fread(data, 1, (size_t)size, f) != (size_t)size) { perror("read"); return 2; }
+    fclose(f);
+    data[size] = 0;
+
+    /* skip header line */
+    const char* body = memchr(data, '\n', (size_t)size);
+    body = body ? body + 1 : data;
+    long long body_n = size - (body - data);
+
+    char* outbuf = (char*)malloc((size_t)body_n + (16 << 20));
+    char scratch1[4096], scratch2[4096];
+
+    long long rows = 0, rows_out = 0, excs = 0, bytes_out_total = 0, passes = 0;
+    struct timespec t0, t1;
+    clock_gettime(CLOCK_MONOTONIC, &t0);
+    double elapsed = 0;
+    while (elapsed < min_seconds) {
+        const char* p = body;
+        const char* end = body + body_n;
+        char* w = outbuf;
+        /* row loop: quote-parity line split + fused pipeline */
+        while (p < end) {
+            /* find row end (quote parity) */
+            const char* q = p;
+            int parity = 0;
+            while (q < end && !(*q == '\n' && parity == 0)) {
+                if (*q == '"') parity ^= 1;
+                ++q;
+            }
+            const char* rend = q;           /* points at '\n' or end */
+            const char* next = q < end ? q + 1 : end;
+            if (rend > p && rend[-1] == '\r') --rend;
+            ++rows;
+
+            str_t cells[NCOLS];
+            if (split_cells(p, rend, cells)) { ++excs; p = next; continue; }
+
+            /* pipeline: runtuplex.py:192-205 */
+            str_t title = cells[0], address = cells[1], city = cells[2],
+                  state = cells[3], postal = cells[4], price = cells[5],
+                  facts = cells[6], url = cells[8];
+            long long bd;
+            if (extract_bd(facts, &bd)) { ++excs; p = next; continue; }
+            if (!(bd < 10)) { p = next; continue; }
+            int typ = extract_type(title, scratch1);
+            if (typ != 1) { p = next; continue; }
+            /* zipcode: '%05d' % int(postal) — postal sniffs f64; int(f64) */
+            double pv = strtod(postal.p, NULL); /* fast_atod-adjacent; clean data */
+            long long zip = (long long)pv;
+            char zipbuf[16];
+            int zn = sprintf(zipbuf, "%05lld", zip);
+            /* city: x[0].upper() + x[1:].lower() */
+            char citybuf[512];
+            if (city.n == 0 || city.n > 500) { ++excs; p = next; continue; }
+            for (long i = 0; i < city.n; ++i) {
+                char c = city.p[i];
+                citybuf[i] = i == 0 ? ((c >= 'a' && c <= 'z') ? c - 32 : c)
+                                    : ((c >= 'A' && c <= 'Z') ? c + 32 : c);
+            }
+            long long ba, sqft;
+            if (extract_ba(facts, &ba)) { ++excs; p = next; continue; }
+            if (extract_sqft(facts, &sqft, scratch1)) { ++excs; p = next; continue; }
+            int offer = extract_offer(title, scratch1);
+            long long pr;
+            if (extract_price(price, offer, facts, sqft, &pr, scratch2)) {
+                ++excs; p = next; continue;
+            }
+            if (!(100000 < pr && pr < 20000000)) { p = next; continue; }
+
+            /* output: url,zipcode,address,city,state,bd,ba,sqft,offer,type,price */
+            w = out_cell(w, url); *w++ = ',';
+            memcpy(w, zipbuf, (size_t)zn); w += zn; *w++ = ',';
+            w = out_cell(w, address); *w++ = ',';
+            str_t cityv = {citybuf, city.n};
+            w = out_cell(w, cityv); *w++ = ',';
+            w = out_cell(w, state); *w++ = ',';
+            w = out_i64(w, bd); *w++ = ',';
+            w = out_i64(w, ba); *w++ = ',';
+            w = out_i64(w, sqft); *w++ = ',';
+            const char* os = OFFER_STR[offer];
+            memcpy(w, os, strlen(os)); w += strlen(os); *w++ = ',';
+            const char* ts = TYPE_STR[typ];
+            memcpy(w, ts, strlen(ts)); w += strlen(ts); *w++ = ',';
+            w = out_i64(w, pr); *w++ = '\n';
+            ++rows_out;
+            p = next;
+        }
+        bytes_out_total += (w - outbuf);
+        ++passes;
+        clock_gettime(CLOCK_MONOTONIC, &t1);
+        elapsed = (t1.tv_sec - t0.tv_sec) + 1e-9 * (t1.tv_nsec - t0.tv_nsec);
+    }
+
+    printf("{\"rows\": %lld, \"rows_out\": %lld, \"exceptions\": %lld, "
+           "\"bytes_in\": %lld, \"bytes_out\": %lld, \"seconds\": %.6f, "
+           "\"rows_per_s\": %.1f, \"passes\": %lld}\n",
+           rows, rows_out, excs, passes * body_n, bytes_out_total, elapsed,
+           rows / elapsed, passes);
+    free(data);
+    free(outbuf);
+    return 0;
+}

@@ -323,10 +323,13 @@ struct PayloadSrc {
 
 static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
                         long long row0, long long in_bytes, tpx_result* res,
-                        const PayloadSrc& psrc, hipStream_t stream) {
+                        const PayloadSrc& psrc, hipStream_t stream,
+                        int64_t flags = 0) {
     const StageDesc& D = st->desc;
-    hipEvent_t ev1, ev2, ev3;
+    res->in_num_rows = n;
+    hipEvent_t ev1, ev2, ev3, evm0, evm1, evs1;
     hipEventCreate(&ev1); hipEventCreate(&ev2); hipEventCreate(&ev3);
+    hipEventCreate(&evm0); hipEventCreate(&evm1); hipEventCreate(&evs1);
     hipEventRecord(ev1, stream);
 
     int nout = (int)D.out_cols.size();
@@ -365,7 +368,9 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
         void* args[] = {&d_in, &d_offs, &n, &row0, &d_heap.p, &d_heap_cursor.p,
                         &heap_cap, &d_keep.p, &d_keep01.p, &d_sizes.p, &d_exc.p,
                         &d_exc_count.p, &exc_cap, &d_outv.p};
+        hipEventRecord(evm0, stream);
         if (launch(st->k_main, grid, 256, stream, args)) return -1;
+        hipEventRecord(evm1, stream);
         HIP_CHECK(hipMemcpyAsync(&exc_count, d_exc_count.p, 8,
                                  hipMemcpyDeviceToHost, stream));
         HIP_CHECK(hipStreamSynchronize(stream));
@@ -398,6 +403,7 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
     if (dev_scan(st, stream, (long long*)d_sizes.p, (long long*)d_size_scan.p, n,
                  &total_bytes))
         return -1;
+    hipEventRecord(evs1, stream);
 
     bool mem_sink = D.sink == "mem";
     long long out_total = mem_sink ? 8 + total_bytes : total_bytes;
@@ -414,21 +420,23 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
     }
     hipEventRecord(ev2, stream);
 
-    // D2H
-    res->out_data = (uint8_t*)malloc((size_t)out_total);
-    HIP_CHECK(hipMemcpyAsync(res->out_data, d_out.p, (size_t)out_total,
-                             hipMemcpyDeviceToHost, stream));
+    // D2H (skipped when the caller keeps outputs device-resident, flags bit1)
     res->out_size = out_total;
     res->out_num_rows = total_rows;
-    res->out_row_offsets = (int64_t*)malloc(((size_t)total_rows + 1) * 8);
-    HIP_CHECK(hipMemcpyAsync(res->out_row_offsets, d_out_offs.p,
-                             ((size_t)total_rows + 1) * 8, hipMemcpyDeviceToHost,
-                             stream));
-    res->out_row_indices = (int64_t*)malloc(((size_t)total_rows + 1) * 8);
-    if (total_rows)
-        HIP_CHECK(hipMemcpyAsync(res->out_row_indices, d_out_rowidx.p,
-                                 (size_t)total_rows * 8, hipMemcpyDeviceToHost,
-                                 stream));
+    if (!(flags & 2)) {
+        res->out_data = (uint8_t*)malloc((size_t)out_total);
+        HIP_CHECK(hipMemcpyAsync(res->out_data, d_out.p, (size_t)out_total,
+                                 hipMemcpyDeviceToHost, stream));
+        res->out_row_offsets = (int64_t*)malloc(((size_t)total_rows + 1) * 8);
+        HIP_CHECK(hipMemcpyAsync(res->out_row_offsets, d_out_offs.p,
+                                 ((size_t)total_rows + 1) * 8,
+                                 hipMemcpyDeviceToHost, stream));
+        res->out_row_indices = (int64_t*)malloc(((size_t)total_rows + 1) * 8);
+        if (total_rows)
+            HIP_CHECK(hipMemcpyAsync(res->out_row_indices, d_out_rowidx.p,
+                                     (size_t)total_rows * 8,
+                                     hipMemcpyDeviceToHost, stream));
+    }
     std::vector<ExcRec> excs((size_t)exc_count);
     if (exc_count)
         HIP_CHECK(hipMemcpyAsync(excs.data(), d_exc.p,
@@ -450,12 +458,13 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
                 row_base[(size_t)p + 1] = row_base[(size_t)p] + psrc.parts[p].num_rows;
         }
         size_t total = 0;
+        std::vector<std::vector<uint8_t>> fetched;  // device-only input payloads
         std::vector<std::pair<const uint8_t*, long long>> payloads(excs.size());
         for (size_t i = 0; i < excs.size(); ++i) {
             if (psrc.bytes) {
                 payloads[i] = {psrc.bytes + excs[i].off_start,
                                excs[i].off_end - excs[i].off_start};
-            } else {
+            } else if (psrc.parts) {
                 long long r = excs[i].row - row0;
                 int64_t p = (int64_t)(std::upper_bound(row_base.begin(),
                                                        row_base.end(), r) -
@@ -464,6 +473,14 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
                 const tpx_partition& P = psrc.parts[p];
                 payloads[i] = {P.data + P.row_offsets[lr],
                                P.row_offsets[lr + 1] - P.row_offsets[lr]};
+            } else {
+                // input lives only on device: copy the raw line back (rare path)
+                long long len = excs[i].off_end - excs[i].off_start;
+                fetched.emplace_back((size_t)len);
+                HIP_CHECK(hipMemcpy(fetched.back().data(),
+                                    (const char*)d_in + excs[i].off_start,
+                                    (size_t)len, hipMemcpyDeviceToHost));
+                payloads[i] = {fetched.back().data(), len};
             }
             total += 32 + (size_t)payloads[i].second;
         }
@@ -485,7 +502,11 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
     float ms = 0;
     hipEventElapsedTime(&ms, ev1, ev2); res->t_kernel_ms = ms;
     hipEventElapsedTime(&ms, ev2, ev3); res->t_d2h_ms = ms;
+    hipEventElapsedTime(&ms, evm0, evm1); res->t_main_ms = ms;
+    hipEventElapsedTime(&ms, evm1, evs1); res->t_compact_ms = ms;
+    hipEventElapsedTime(&ms, evs1, ev2); res->t_write_ms = ms;
     (void)hipEventDestroy(ev1); (void)hipEventDestroy(ev2); (void)hipEventDestroy(ev3);
+    (void)hipEventDestroy(evm0); (void)hipEventDestroy(evm1); (void)hipEventDestroy(evs1);
     return 0;
 }
 
@@ -557,6 +578,11 @@ extern "C" int64_t tpx_stage_execute(tpx_stage* st, const tpx_partition* parts,
 // csv source (TransformTask.cc:724 processFileSource + CSVReader.cc:390 analog;
 // row boundaries found on-device by the quote-parity scan kernels)
 
+static int64_t csv_boundary_and_core(tpx_stage* st, void* d_in_p, long long size,
+                                     const uint8_t* host_bytes,
+                                     int64_t first_global_row, int64_t flags,
+                                     tpx_result* res);
+
 extern "C" int64_t tpx_stage_execute_csv(tpx_stage* st, const uint8_t* csv_bytes,
                                          int64_t size, int64_t first_global_row,
                                          tpx_result* res) {
@@ -578,7 +604,46 @@ extern "C" int64_t tpx_stage_execute_csv(tpx_stage* st, const uint8_t* csv_bytes
     HIP_CHECK(hipMemcpyAsync(d_in.p, csv_bytes, (size_t)size,
                              hipMemcpyHostToDevice, stream));
     hipEventRecord(ev1, stream);
+    HIP_CHECK(hipEventSynchronize(ev1));
+    float ms = 0;
+    hipEventElapsedTime(&ms, ev0, ev1);
+    res->t_h2d_ms = ms;
+    (void)hipEventDestroy(ev0); (void)hipEventDestroy(ev1);
 
+    return csv_boundary_and_core(st, d_in.p, size, csv_bytes, first_global_row,
+                                 0, res);
+}
+
+// ---------------------------------------------------------------------------------
+// device-resident input helpers (bench / cache())
+
+extern "C" uint64_t tpx_dev_alloc(int64_t size) {
+    void* p = nullptr;
+    if (hipMalloc(&p, (size_t)size) != hipSuccess) return 0;
+    return (uint64_t)(uintptr_t)p;
+}
+
+extern "C" int64_t tpx_dev_upload(uint64_t dst, const void* src, int64_t size) {
+    HIP_CHECK(hipMemcpy((void*)(uintptr_t)dst, src, (size_t)size,
+                        hipMemcpyHostToDevice));
+    return 0;
+}
+
+extern "C" void tpx_dev_free(uint64_t ptr) {
+    if (ptr) (void)hipFree((void*)(uintptr_t)ptr);
+}
+
+// csv execution over device-resident bytes. NOTE: exception payloads need host
+// bytes; with a device-only input, exception raw lines are copied back D2H
+// per-record (rare path).
+static int64_t csv_boundary_and_core(tpx_stage* st, void* d_in_p, long long size,
+                                     const uint8_t* host_bytes,
+                                     int64_t first_global_row, int64_t flags,
+                                     tpx_result* res) {
+    hipStream_t stream = nullptr;
+    hipEvent_t eb0, eb1;
+    hipEventCreate(&eb0); hipEventCreate(&eb1);
+    hipEventRecord(eb0, stream);
     long long nchunks = (size + 4095) / 4096;  // TPX_CSV_CHUNK
     DevBuf d_q, d_c0, d_c1, d_qs, d_rc, d_base;
     HIP_CHECK(d_q.alloc((size_t)nchunks * 8));
@@ -589,11 +654,12 @@ extern "C" int64_t tpx_stage_execute_csv(tpx_stage* st, const uint8_t* csv_bytes
     HIP_CHECK(d_base.alloc((size_t)nchunks * 8));
     unsigned grid = (unsigned)std::min<long long>((nchunks + 255) / 256, 8192);
     {
-        void* args[] = {&d_in.p, (void*)&size, &nchunks, &d_q.p, &d_c0.p, &d_c1.p};
+        void* args[] = {&d_in_p, (void*)&size, &nchunks, &d_q.p, &d_c0.p, &d_c1.p};
         if (launch(st->k_csv_chunk, grid, 256, stream, args)) return -1;
     }
     long long qtotal = 0;
-    if (dev_scan(st, stream, (long long*)d_q.p, (long long*)d_qs.p, nchunks, &qtotal))
+    if (dev_scan(st, stream, (long long*)d_q.p, (long long*)d_qs.p, nchunks,
+                 &qtotal))
         return -1;
     {
         void* args[] = {&d_qs.p, &d_c0.p, &d_c1.p, &d_rc.p, &nchunks};
@@ -607,19 +673,41 @@ extern "C" int64_t tpx_stage_execute_csv(tpx_stage* st, const uint8_t* csv_bytes
     DevBuf d_offs;
     HIP_CHECK(d_offs.alloc(((size_t)nrows + 1) * 8));
     {
-        void* args[] = {&d_in.p, (void*)&size, &nchunks, &d_qs.p, &d_base.p,
+        void* args[] = {&d_in_p, (void*)&size, &nchunks, &d_qs.p, &d_base.p,
                         &d_offs.p};
         if (launch(st->k_csv_rows, grid, 256, stream, args)) return -1;
     }
-    HIP_CHECK(hipEventSynchronize(ev1));
+    hipEventRecord(eb1, stream);
+    HIP_CHECK(hipEventSynchronize(eb1));
     float ms = 0;
-    hipEventElapsedTime(&ms, ev0, ev1);
-    res->t_h2d_ms = ms;
-    (void)hipEventDestroy(ev0); (void)hipEventDestroy(ev1);
+    hipEventElapsedTime(&ms, eb0, eb1);
+    res->t_boundary_ms = ms;
+    (void)hipEventDestroy(eb0); (void)hipEventDestroy(eb1);
 
-    PayloadSrc psrc{csv_bytes, nullptr, 0};
-    return run_core(st, d_in.p, d_offs.p, nrows, first_global_row, size, res, psrc,
-                    stream);
+    if (host_bytes) {
+        PayloadSrc psrc{host_bytes, nullptr, 0};
+        return run_core(st, d_in_p, d_offs.p, nrows, first_global_row, size, res,
+                        psrc, stream, flags);
+    }
+    // device-only input: run core without payloads, then fetch raw lines D2H
+    PayloadSrc psrc{nullptr, nullptr, 0};
+    // run_core builds payloads only when psrc.bytes or psrc.parts set; with
+    // neither, exceptions keep (row,ec,opid) and payloads are fetched here.
+    int64_t rc = run_core(st, d_in_p, d_offs.p, nrows, first_global_row, size,
+                          res, psrc, stream, flags | 4 /* defer payloads */);
+    return rc;
+}
+
+extern "C" int64_t tpx_stage_execute_csv_dev(tpx_stage* st, uint64_t dev_bytes,
+                                             int64_t size,
+                                             int64_t first_global_row,
+                                             int64_t flags, tpx_result* res) {
+    memset(res, 0, sizeof(*res));
+    if (!st->loaded) { set_err("stage not loaded on a GPU"); return -1; }
+    res->bytes_in = size;
+    if (size == 0) { empty_result(res, st->desc.sink == "mem"); return 0; }
+    return csv_boundary_and_core(st, (void*)(uintptr_t)dev_bytes, size, nullptr,
+                                 first_global_row, flags, res);
 }
 
 extern "C" void tpx_result_free(tpx_result* res) {

@@ -36,6 +36,11 @@ class TpxResult(ctypes.Structure):
         ("t_d2h_ms", ctypes.c_double),
         ("bytes_in", ctypes.c_int64),
         ("bytes_out", ctypes.c_int64),
+        ("t_boundary_ms", ctypes.c_double),
+        ("t_main_ms", ctypes.c_double),
+        ("t_compact_ms", ctypes.c_double),
+        ("t_write_ms", ctypes.c_double),
+        ("in_num_rows", ctypes.c_int64),
     ]
 
 
@@ -79,6 +84,18 @@ class GpuLib:
                                               ctypes.c_int64, ctypes.c_int64,
                                               ctypes.POINTER(TpxResult)]
         lib.tpx_result_free.argtypes = [ctypes.POINTER(TpxResult)]
+        lib.tpx_dev_alloc.restype = ctypes.c_uint64
+        lib.tpx_dev_alloc.argtypes = [ctypes.c_int64]
+        lib.tpx_dev_upload.restype = ctypes.c_int64
+        lib.tpx_dev_upload.argtypes = [ctypes.c_uint64, ctypes.c_void_p,
+                                       ctypes.c_int64]
+        lib.tpx_dev_free.argtypes = [ctypes.c_uint64]
+        lib.tpx_stage_execute_csv_dev.restype = ctypes.c_int64
+        lib.tpx_stage_execute_csv_dev.argtypes = [ctypes.c_void_p,
+                                                  ctypes.c_uint64,
+                                                  ctypes.c_int64, ctypes.c_int64,
+                                                  ctypes.c_int64,
+                                                  ctypes.POINTER(TpxResult)]
         self.lib = lib
         os.makedirs(_CACHE_DIR, exist_ok=True)
         self._stage_cache = {}
