@@ -661,20 +661,15 @@ class StageCodegen:
         L.append("    const char* rend = (const char*)in_data + in_offs[i+1];")
         L.append("    if (rend > rp && rend[-1] == '\\n') --rend;")
         L.append("    if (rend > rp && rend[-1] == '\\r') --rend;")
-        L.append("    tpx_cell cells[%d];" % nc)
-        L.append("    { bool m = true; int k = 0;")
-        L.append("      const char* cur = rp;")
-        L.append("      while (k < %d) { cur = tpx_csv_next_cell(cur, rend, &cells[k], &m);"
-                 " ++k; if (!m) break; }" % nc)
-        L.append("      if (k < %d) prc = %d;          // CSV_UNDERRUN" % (nc, 20))
-        L.append("      else if (m) prc = %d;          // CSV_OVERRUN" % 21)
-        L.append("      int badf = 0;")
-        L.append("      for (int j = 0; j < k; ++j) badf |= cells[j].flags;")
-        L.append("      if (!prc && (badf & 6)) prc = %d;  // BADPARSE (escapes -> host)" % 70)
-        L.append("    }")
+        # straight-line per-cell scan: one named local per cell — a dynamically
+        # indexed cells[] array spills to scratch (288 B/thread measured) and
+        # turns every access into an HBM round trip
+        L.append("    const char* cur = rp;")
+        L.append("    bool m = true, avail = true;")
+        L.append("    int badf = 0;")
 
         def null_check(idx):
-            cv = "tstr{cells[%d].p, cells[%d].n}" % (idx, idx)
+            cv = "tstr{cl%d.p, cl%d.n}" % (idx, idx)
             checks = " || ".join("tpx_streq(%s, %s)" % (cv, self.lit(nv))
                                  for nv in null_values) or "false"
             return checks
@@ -682,30 +677,36 @@ class StageCodegen:
         for idx, t in enumerate(in_types):
             base = T.deopt(t)
             opt = T.is_opt(t)
+            L.append("    tpx_cell cl%d{rp, 0, 0};" % idx)
+            L.append("    if (!prc) {")
+            L.append("      if (!avail) prc = %d;  // CSV_UNDERRUN" % 20)
+            L.append("      else { cur = tpx_csv_next_cell(cur, rend, &cl%d, &m);"
+                     " avail = m; badf |= cl%d.flags; }" % (idx, idx))
+            L.append("    }")
             if opt:
                 L.append("    bool c%d_n = !prc && (%s);" % (idx, null_check(idx)))
+            guard = ("!prc && !c%d_n" % idx) if opt else "!prc"
             if base == T.STR:
-                L.append("    tstr c%d{cells[%d].p, cells[%d].n};" % (idx, idx, idx))
+                L.append("    tstr c%d{cl%d.p, cl%d.n};" % (idx, idx, idx))
                 L.append("    if (prc) c%d = tstr{rp, 0};" % idx)
                 if opt:
                     L.append("    if (c%d_n) c%d = tstr{rp, 0};" % (idx, idx))
             elif base == T.I64:
                 L.append("    long long c%d = 0;" % idx)
-                guard = ("!prc && !c%d_n" % idx) if opt else "!prc"
-                L.append("    if (%s && tpx_cell_i64(cells[%d], &c%d) != 0) prc = %d;"
+                L.append("    if (%s && tpx_cell_i64(cl%d, &c%d) != 0) prc = %d;"
                          % (guard, idx, idx, 70))
             elif base == T.F64:
                 L.append("    double c%d = 0.0;" % idx)
-                guard = ("!prc && !c%d_n" % idx) if opt else "!prc"
-                L.append("    if (%s && tpx_cell_f64(cells[%d], &c%d) != 0) prc = %d;"
+                L.append("    if (%s && tpx_cell_f64(cl%d, &c%d) != 0) prc = %d;"
                          % (guard, idx, idx, 70))
             elif base == T.BOOL:
                 L.append("    bool c%d = false;" % idx)
-                guard = ("!prc && !c%d_n" % idx) if opt else "!prc"
-                L.append("    if (%s && tpx_cell_bool(cells[%d], &c%d) != 0) prc = %d;"
+                L.append("    if (%s && tpx_cell_bool(cl%d, &c%d) != 0) prc = %d;"
                          % (guard, idx, idx, 70))
             else:
                 raise CodegenError("csv input type %r" % (t,))
+        L.append("    if (!prc && avail) prc = %d;  // CSV_OVERRUN" % 21)
+        L.append("    if (!prc && (badf & 6)) prc = %d;  // BADPARSE (escapes/structure -> host)" % 70)
         return L
 
     def _store_columnar(self, out_types):

@@ -148,9 +148,8 @@ static bool compile_hiprtc(const char* src, std::vector<char>* code, std::string
         *log = "hiprtcCreateProgram failed";
         return false;
     }
-    const char* opts[] = {"--offload-arch=gfx950", "-O3", "-std=c++17",
-                          "-ffast-math"};
-    hiprtcResult rc = hiprtcCompileProgram(prog, 4, opts);
+    const char* opts[] = {"--offload-arch=gfx950", "-O3", "-std=c++17"};
+    hiprtcResult rc = hiprtcCompileProgram(prog, 3, opts);
     size_t log_size = 0;
     hiprtcGetProgramLogSize(prog, &log_size);
     if (log_size > 1) {
@@ -253,6 +252,74 @@ struct DevBuf {
     hipError_t alloc(size_t n) { return hipMalloc(&p, n ? n : 8); }
 };
 
+// Per-device workspace arena, reused across executes (the device partition
+// manager's scratch — replaces per-call hipMalloc churn, which measured ~70 ms
+// of a 110 ms step). Single-threaded per device, like the reference's
+// one-task-per-thread discipline. Pointers stay valid until the NEXT execute.
+struct DevArena {
+    std::vector<std::pair<void*, size_t>> blocks;
+    size_t cur = 0, off = 0, used = 0;
+    void* take(size_t n) {
+        n = (n + 255) & ~(size_t)255;
+        if (!n) n = 256;
+        used += n;
+        while (cur < blocks.size()) {
+            if (off + n <= blocks[cur].second) {
+                void* p = (char*)blocks[cur].first + off;
+                off += n;
+                return p;
+            }
+            ++cur;
+            off = 0;
+        }
+        size_t bs = std::max(n, (size_t)256 << 20);
+        void* p = nullptr;
+        if (hipMalloc(&p, bs) != hipSuccess) {
+            set_err("arena hipMalloc failed");
+            return nullptr;
+        }
+        blocks.emplace_back(p, bs);
+        cur = blocks.size() - 1;
+        off = n;
+        return p;
+    }
+    void begin() {
+        if (blocks.size() > 1) {  // consolidate to one block at the high-water mark
+            size_t want = used + (used >> 2);
+            (void)hipDeviceSynchronize();
+            for (auto& b : blocks) (void)hipFree(b.first);
+            blocks.clear();
+            void* p = nullptr;
+            if (hipMalloc(&p, want) == hipSuccess) blocks.emplace_back(p, want);
+        }
+        cur = 0;
+        off = 0;
+        used = 0;
+    }
+};
+static DevArena g_arena[64];
+
+// persistent per-device string heap / exception buffer / counters
+struct Persist {
+    void* heap = nullptr;
+    size_t heap_cap = 0;
+    void* exc = nullptr;
+    size_t exc_cap_recs = 0;
+    void* counters = nullptr;  // [heap_cursor, exc_count]
+};
+static Persist g_persist[64];
+
+static int cur_device() {
+    int d = 0;
+    (void)hipGetDevice(&d);
+    return d;
+}
+
+#define ARENA_TAKE(var, n)                                                   \
+    void* var = g_arena[dev].take((size_t)(n));                              \
+    if (!var) return -1;
+
+
 static int launch(hipFunction_t f, unsigned grid, unsigned block, hipStream_t s,
                   void** args) {
     hipError_t e = hipModuleLaunchKernel(f, grid, 1, 1, block, 1, 1, 0, s, args,
@@ -263,44 +330,6 @@ static int launch(hipFunction_t f, unsigned grid, unsigned block, hipStream_t s,
 
 // exclusive scan of device i64 array (n elements), in -> out; returns total via
 // host copy of (out[n-1] + in[n-1]).
-static int dev_scan(tpx_stage* st, hipStream_t stream, long long* d_in,
-                    long long* d_out, long long n, long long* total) {
-    const long long BLOCK = 2048;  // TPX_SCAN_BLOCK
-    long long nblocks = (n + BLOCK - 1) / BLOCK;
-    if (nblocks == 0) { *total = 0; return 0; }
-    DevBuf sums;
-    HIP_CHECK(sums.alloc((size_t)nblocks * 8));
-    void* a1[] = {&d_in, &d_out, &sums.p, &n};
-    if (launch(st->k_scan_block, (unsigned)nblocks, 256, stream, a1)) return -1;
-    if (nblocks > 1) {
-        DevBuf sums_scan;
-        HIP_CHECK(sums_scan.alloc((size_t)nblocks * 8));
-        long long dummy;
-        if (dev_scan(st, stream, (long long*)sums.p, (long long*)sums_scan.p,
-                     nblocks, &dummy))
-            return -1;
-        void* a2[] = {&d_out, &sums_scan.p, &n};
-        if (launch(st->k_scan_add, (unsigned)nblocks, 256, stream, a2)) return -1;
-        // total = last out + last in
-        long long last_out = 0, last_in = 0;
-        HIP_CHECK(hipMemcpyAsync(&last_out, d_out + (n - 1), 8,
-                                 hipMemcpyDeviceToHost, stream));
-        HIP_CHECK(hipMemcpyAsync(&last_in, d_in + (n - 1), 8,
-                                 hipMemcpyDeviceToHost, stream));
-        HIP_CHECK(hipStreamSynchronize(stream));
-        *total = last_out + last_in;
-    } else {
-        long long last_out = 0, last_in = 0;
-        HIP_CHECK(hipMemcpyAsync(&last_out, d_out + (n - 1), 8,
-                                 hipMemcpyDeviceToHost, stream));
-        HIP_CHECK(hipMemcpyAsync(&last_in, d_in + (n - 1), 8,
-                                 hipMemcpyDeviceToHost, stream));
-        HIP_CHECK(hipStreamSynchronize(stream));
-        *total = last_out + last_in;
-    }
-    return 0;
-}
-
 static size_t col_slot_bytes(const ColDesc& c, long long n, int slot) {
     if (slot == 0) return (size_t)n * 8;                       // value / ptr
     if (slot == 1) return c.kind == K_STR ? (size_t)n * 4 : 0; // len
@@ -310,6 +339,38 @@ static size_t col_slot_bytes(const ColDesc& c, long long n, int slot) {
 struct ExcRec {
     long long row, ec, opid, off_start, off_end;  // offsets into the input bytes
 };
+
+// exclusive scan of device i64 array (n elements), in -> out; total = out[n-1] +
+// in[n-1]. Temporaries come from the workspace arena.
+static int dev_scan(tpx_stage* st, hipStream_t stream, long long* d_in,
+                    long long* d_out, long long n, long long* total) {
+    int dev = cur_device();
+    const long long BLOCK = 2048;  // TPX_SCAN_BLOCK
+    long long nblocks = (n + BLOCK - 1) / BLOCK;
+    if (nblocks == 0) { *total = 0; return 0; }
+    ARENA_TAKE(sums, (size_t)nblocks * 8);
+    void* in_p = d_in;
+    void* out_p = d_out;
+    void* a1[] = {&in_p, &out_p, &sums, &n};
+    if (launch(st->k_scan_block, (unsigned)nblocks, 256, stream, a1)) return -1;
+    if (nblocks > 1) {
+        ARENA_TAKE(sums_scan, (size_t)nblocks * 8);
+        long long dummy;
+        if (dev_scan(st, stream, (long long*)sums, (long long*)sums_scan,
+                     nblocks, &dummy))
+            return -1;
+        void* a2[] = {&out_p, &sums_scan, &n};
+        if (launch(st->k_scan_add, (unsigned)nblocks, 256, stream, a2)) return -1;
+    }
+    long long last_out = 0, last_in = 0;
+    HIP_CHECK(hipMemcpyAsync(&last_out, d_out + (n - 1), 8,
+                             hipMemcpyDeviceToHost, stream));
+    HIP_CHECK(hipMemcpyAsync(&last_in, d_in + (n - 1), 8, hipMemcpyDeviceToHost,
+                             stream));
+    HIP_CHECK(hipStreamSynchronize(stream));
+    *total = last_out + last_in;
+    return 0;
+}
 
 // ---------------------------------------------------------------------------------
 // shared stage-execution core (mem + csv sources share the fixed kernel signature:
@@ -326,95 +387,116 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
                         const PayloadSrc& psrc, hipStream_t stream,
                         int64_t flags = 0) {
     const StageDesc& D = st->desc;
+    int dev = cur_device();
+    Persist& P = g_persist[dev];
     res->in_num_rows = n;
     hipEvent_t ev1, ev2, ev3, evm0, evm1, evs1;
     hipEventCreate(&ev1); hipEventCreate(&ev2); hipEventCreate(&ev3);
     hipEventCreate(&evm0); hipEventCreate(&evm1); hipEventCreate(&evs1);
     hipEventRecord(ev1, stream);
 
+    // output columnar buffers (3 slots per column) from the arena
     int nout = (int)D.out_cols.size();
-    std::vector<DevBuf> colbufs((size_t)nout * 3);
     std::vector<void*> outv((size_t)nout * 3, nullptr);
     for (int k = 0; k < nout; ++k)
         for (int s = 0; s < 3; ++s) {
             size_t b = col_slot_bytes(D.out_cols[(size_t)k], n, s);
             if (b) {
-                HIP_CHECK(colbufs[(size_t)k * 3 + s].alloc(b));
-                outv[(size_t)k * 3 + s] = colbufs[(size_t)k * 3 + s].p;
+                outv[(size_t)k * 3 + s] = g_arena[dev].take(b);
+                if (!outv[(size_t)k * 3 + s]) return -1;
             }
         }
-    DevBuf d_outv;
-    HIP_CHECK(d_outv.alloc(outv.size() * sizeof(void*) + 8));
-    HIP_CHECK(hipMemcpyAsync(d_outv.p, outv.data(), outv.size() * sizeof(void*),
+    ARENA_TAKE(d_outv, outv.size() * sizeof(void*) + 8);
+    HIP_CHECK(hipMemcpyAsync(d_outv, outv.data(), outv.size() * sizeof(void*),
                              hipMemcpyHostToDevice, stream));
 
-    DevBuf d_keep, d_keep01, d_sizes, d_exc, d_exc_count, d_heap, d_heap_cursor;
-    HIP_CHECK(d_keep.alloc((size_t)n));
-    HIP_CHECK(d_keep01.alloc((size_t)n * 8));
-    HIP_CHECK(d_sizes.alloc((size_t)n * 8));
-    unsigned long long exc_cap = (unsigned long long)std::min<long long>(n, 1 << 20);
-    HIP_CHECK(d_exc.alloc((size_t)exc_cap * sizeof(ExcRec)));
-    HIP_CHECK(d_exc_count.alloc(8));
-    unsigned long long heap_cap =
-        (unsigned long long)std::max<long long>(2 * in_bytes + (16 << 20), 1 << 20);
-    HIP_CHECK(d_heap.alloc(heap_cap));
-    HIP_CHECK(d_heap_cursor.alloc(8));
+    ARENA_TAKE(d_keep, (size_t)n);
+    ARENA_TAKE(d_keep01, (size_t)n * 8);
+    ARENA_TAKE(d_sizes, (size_t)n * 8);
 
+    // persistent heap / exception buffer / counters
+    if (!P.counters)
+        HIP_CHECK(hipMalloc(&P.counters, 16));
+    unsigned long long exc_cap =
+        (unsigned long long)std::min<long long>(std::max<long long>(n, 1024),
+                                                1 << 20);
+    if (P.exc_cap_recs < exc_cap) {
+        if (P.exc) (void)hipFree(P.exc);
+        HIP_CHECK(hipMalloc(&P.exc, (size_t)exc_cap * sizeof(ExcRec)));
+        P.exc_cap_recs = exc_cap;
+    }
+    exc_cap = P.exc_cap_recs;
+    unsigned long long heap_cap =
+        (unsigned long long)std::max<long long>(in_bytes + (in_bytes >> 1) +
+                                                (16 << 20), 1 << 20);
+    if (P.heap_cap < heap_cap) {
+        if (P.heap) (void)hipFree(P.heap);
+        HIP_CHECK(hipMalloc(&P.heap, heap_cap));
+        P.heap_cap = heap_cap;
+    }
+    heap_cap = P.heap_cap;
+
+    void* d_heap_cursor = P.counters;
+    void* d_exc_count = (char*)P.counters + 8;
     unsigned long long exc_count = 0;
     for (int attempt = 0;; ++attempt) {
-        HIP_CHECK(hipMemsetAsync(d_exc_count.p, 0, 8, stream));
-        HIP_CHECK(hipMemsetAsync(d_heap_cursor.p, 0, 8, stream));
+        HIP_CHECK(hipMemsetAsync(P.counters, 0, 16, stream));
         unsigned grid = (unsigned)std::min<long long>((n + 255) / 256, 8192);
-        void* args[] = {&d_in, &d_offs, &n, &row0, &d_heap.p, &d_heap_cursor.p,
-                        &heap_cap, &d_keep.p, &d_keep01.p, &d_sizes.p, &d_exc.p,
-                        &d_exc_count.p, &exc_cap, &d_outv.p};
+        void* args[] = {&d_in, &d_offs, &n, &row0, &P.heap, &d_heap_cursor,
+                        &heap_cap, &d_keep, &d_keep01, &d_sizes, &P.exc,
+                        &d_exc_count, &exc_cap, &d_outv};
         hipEventRecord(evm0, stream);
         if (launch(st->k_main, grid, 256, stream, args)) return -1;
         hipEventRecord(evm1, stream);
-        HIP_CHECK(hipMemcpyAsync(&exc_count, d_exc_count.p, 8,
-                                 hipMemcpyDeviceToHost, stream));
+        unsigned long long counters[2] = {0, 0};
+        HIP_CHECK(hipMemcpyAsync(counters, P.counters, 16, hipMemcpyDeviceToHost,
+                                 stream));
         HIP_CHECK(hipStreamSynchronize(stream));
-        unsigned long long heap_used = 0;
-        HIP_CHECK(hipMemcpy(&heap_used, d_heap_cursor.p, 8, hipMemcpyDeviceToHost));
+        unsigned long long heap_used = counters[0];
+        exc_count = counters[1];
         bool heap_overflow = heap_used > heap_cap;
         bool exc_overflow = exc_count > exc_cap;
         if (!heap_overflow && !exc_overflow) break;
         if (attempt >= 3) { set_err("retry limit (heap/exc overflow)"); return -1; }
         if (heap_overflow) {
-            hipFree(d_heap.p); d_heap.p = nullptr;
-            heap_cap *= 4;
-            HIP_CHECK(d_heap.alloc(heap_cap));
+            (void)hipFree(P.heap);
+            P.heap = nullptr;
+            P.heap_cap = 0;
+            heap_cap = heap_used + (heap_used >> 1) + (16 << 20);
+            HIP_CHECK(hipMalloc(&P.heap, heap_cap));
+            P.heap_cap = heap_cap;
         }
         if (exc_overflow) {
-            hipFree(d_exc.p); d_exc.p = nullptr;
+            (void)hipFree(P.exc);
+            P.exc = nullptr;
+            P.exc_cap_recs = 0;
             exc_cap = exc_count + 1024;
-            HIP_CHECK(d_exc.alloc((size_t)exc_cap * sizeof(ExcRec)));
+            HIP_CHECK(hipMalloc(&P.exc, (size_t)exc_cap * sizeof(ExcRec)));
+            P.exc_cap_recs = exc_cap;
         }
     }
 
     // compaction scans
-    DevBuf d_keep_scan, d_size_scan;
-    HIP_CHECK(d_keep_scan.alloc((size_t)n * 8));
-    HIP_CHECK(d_size_scan.alloc((size_t)n * 8));
+    ARENA_TAKE(d_keep_scan, (size_t)n * 8);
+    ARENA_TAKE(d_size_scan, (size_t)n * 8);
     long long total_rows = 0, total_bytes = 0;
-    if (dev_scan(st, stream, (long long*)d_keep01.p, (long long*)d_keep_scan.p, n,
+    if (dev_scan(st, stream, (long long*)d_keep01, (long long*)d_keep_scan, n,
                  &total_rows))
         return -1;
-    if (dev_scan(st, stream, (long long*)d_sizes.p, (long long*)d_size_scan.p, n,
+    if (dev_scan(st, stream, (long long*)d_sizes, (long long*)d_size_scan, n,
                  &total_bytes))
         return -1;
     hipEventRecord(evs1, stream);
 
     bool mem_sink = D.sink == "mem";
     long long out_total = mem_sink ? 8 + total_bytes : total_bytes;
-    DevBuf d_out, d_out_offs, d_out_rowidx;
-    HIP_CHECK(d_out.alloc((size_t)out_total));
-    HIP_CHECK(d_out_offs.alloc(((size_t)total_rows + 1) * 8));
-    HIP_CHECK(d_out_rowidx.alloc(((size_t)total_rows + 1) * 8));
+    ARENA_TAKE(d_out, (size_t)out_total + 8);
+    ARENA_TAKE(d_out_offs, ((size_t)total_rows + 1) * 8);
+    ARENA_TAKE(d_out_rowidx, ((size_t)total_rows + 1) * 8);
     {
         unsigned grid = (unsigned)std::min<long long>((n + 255) / 256, 8192);
-        void* args[] = {&d_keep.p, &d_keep_scan.p, &d_size_scan.p, &n, &row0,
-                        &d_outv.p, &d_out.p, &d_out_offs.p, &d_out_rowidx.p,
+        void* args[] = {&d_keep, &d_keep_scan, &d_size_scan, &n, &row0,
+                        &d_outv, &d_out, &d_out_offs, &d_out_rowidx,
                         &total_rows, &total_bytes};
         if (launch(st->k_write, grid, 256, stream, args)) return -1;
     }
@@ -425,21 +507,21 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
     res->out_num_rows = total_rows;
     if (!(flags & 2)) {
         res->out_data = (uint8_t*)malloc((size_t)out_total);
-        HIP_CHECK(hipMemcpyAsync(res->out_data, d_out.p, (size_t)out_total,
+        HIP_CHECK(hipMemcpyAsync(res->out_data, d_out, (size_t)out_total,
                                  hipMemcpyDeviceToHost, stream));
         res->out_row_offsets = (int64_t*)malloc(((size_t)total_rows + 1) * 8);
-        HIP_CHECK(hipMemcpyAsync(res->out_row_offsets, d_out_offs.p,
+        HIP_CHECK(hipMemcpyAsync(res->out_row_offsets, d_out_offs,
                                  ((size_t)total_rows + 1) * 8,
                                  hipMemcpyDeviceToHost, stream));
         res->out_row_indices = (int64_t*)malloc(((size_t)total_rows + 1) * 8);
         if (total_rows)
-            HIP_CHECK(hipMemcpyAsync(res->out_row_indices, d_out_rowidx.p,
+            HIP_CHECK(hipMemcpyAsync(res->out_row_indices, d_out_rowidx,
                                      (size_t)total_rows * 8,
                                      hipMemcpyDeviceToHost, stream));
     }
     std::vector<ExcRec> excs((size_t)exc_count);
     if (exc_count)
-        HIP_CHECK(hipMemcpyAsync(excs.data(), d_exc.p,
+        HIP_CHECK(hipMemcpyAsync(excs.data(), P.exc,
                                  (size_t)exc_count * sizeof(ExcRec),
                                  hipMemcpyDeviceToHost, stream));
     HIP_CHECK(hipStreamSynchronize(stream));
@@ -451,14 +533,15 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
         std::sort(excs.begin(), excs.end(),
                   [](const ExcRec& a, const ExcRec& b) { return a.row < b.row; });
         std::vector<long long> row_base;
-        if (!psrc.bytes) {
+        if (psrc.parts) {
             row_base.resize((size_t)psrc.n_parts + 1);
             row_base[0] = 0;
-            for (int64_t p = 0; p < psrc.n_parts; ++p)
-                row_base[(size_t)p + 1] = row_base[(size_t)p] + psrc.parts[p].num_rows;
+            for (int64_t pp = 0; pp < psrc.n_parts; ++pp)
+                row_base[(size_t)pp + 1] =
+                    row_base[(size_t)pp] + psrc.parts[pp].num_rows;
         }
         size_t total = 0;
-        std::vector<std::vector<uint8_t>> fetched;  // device-only input payloads
+        std::vector<std::vector<uint8_t>> fetched;
         std::vector<std::pair<const uint8_t*, long long>> payloads(excs.size());
         for (size_t i = 0; i < excs.size(); ++i) {
             if (psrc.bytes) {
@@ -466,15 +549,15 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
                                excs[i].off_end - excs[i].off_start};
             } else if (psrc.parts) {
                 long long r = excs[i].row - row0;
-                int64_t p = (int64_t)(std::upper_bound(row_base.begin(),
-                                                       row_base.end(), r) -
-                                      row_base.begin()) - 1;
-                long long lr = r - row_base[(size_t)p];
-                const tpx_partition& P = psrc.parts[p];
-                payloads[i] = {P.data + P.row_offsets[lr],
-                               P.row_offsets[lr + 1] - P.row_offsets[lr]};
+                int64_t pp = (int64_t)(std::upper_bound(row_base.begin(),
+                                                        row_base.end(), r) -
+                                       row_base.begin()) - 1;
+                long long lr = r - row_base[(size_t)pp];
+                const tpx_partition& PP = psrc.parts[pp];
+                payloads[i] = {PP.data + PP.row_offsets[lr],
+                               PP.row_offsets[lr + 1] - PP.row_offsets[lr]};
             } else {
-                // input lives only on device: copy the raw line back (rare path)
+                // device-only input: copy the raw line back (rare path)
                 long long len = excs[i].off_end - excs[i].off_start;
                 fetched.emplace_back((size_t)len);
                 HIP_CHECK(hipMemcpy(fetched.back().data(),
@@ -532,6 +615,8 @@ extern "C" int64_t tpx_stage_execute(tpx_stage* st, const tpx_partition* parts,
     memset(res, 0, sizeof(*res));
     if (!st->loaded) { set_err("stage not loaded on a GPU"); return -1; }
     hipStream_t stream = nullptr;
+    int dev = cur_device();
+    g_arena[dev].begin();
 
     long long n = 0, in_bytes = 0;
     for (int64_t p = 0; p < n_parts; ++p) {
@@ -544,15 +629,14 @@ extern "C" int64_t tpx_stage_execute(tpx_stage* st, const tpx_partition* parts,
     hipEvent_t ev0, ev1;
     hipEventCreate(&ev0); hipEventCreate(&ev1);
     hipEventRecord(ev0, stream);
-    DevBuf d_in, d_offs;
-    HIP_CHECK(d_in.alloc((size_t)in_bytes));
-    HIP_CHECK(d_offs.alloc((size_t)(n + 1) * 8));
+    ARENA_TAKE(d_in, (size_t)in_bytes);
+    ARENA_TAKE(d_offs, (size_t)(n + 1) * 8);
     {
         std::vector<long long> offs((size_t)n + 1);
         long long byte_base = 0, row_base = 0;
         for (int64_t p = 0; p < n_parts; ++p) {
             const tpx_partition& P = parts[p];
-            HIP_CHECK(hipMemcpyAsync((char*)d_in.p + byte_base, P.data,
+            HIP_CHECK(hipMemcpyAsync((char*)d_in + byte_base, P.data,
                                      (size_t)P.size, hipMemcpyHostToDevice, stream));
             for (long long r = 0; r < P.num_rows; ++r)
                 offs[(size_t)(row_base + r)] = byte_base + P.row_offsets[r];
@@ -560,8 +644,9 @@ extern "C" int64_t tpx_stage_execute(tpx_stage* st, const tpx_partition* parts,
             row_base += P.num_rows;
         }
         offs[(size_t)n] = byte_base;
-        HIP_CHECK(hipMemcpyAsync(d_offs.p, offs.data(), ((size_t)n + 1) * 8,
+        HIP_CHECK(hipMemcpyAsync(d_offs, offs.data(), ((size_t)n + 1) * 8,
                                  hipMemcpyHostToDevice, stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
     }
     hipEventRecord(ev1, stream);
     HIP_CHECK(hipEventSynchronize(ev1));
@@ -571,7 +656,7 @@ extern "C" int64_t tpx_stage_execute(tpx_stage* st, const tpx_partition* parts,
     (void)hipEventDestroy(ev0); (void)hipEventDestroy(ev1);
 
     PayloadSrc psrc{nullptr, parts, n_parts};
-    return run_core(st, d_in.p, d_offs.p, n, 0, in_bytes, res, psrc, stream);
+    return run_core(st, d_in, d_offs, n, 0, in_bytes, res, psrc, stream);
 }
 
 // ---------------------------------------------------------------------------------
@@ -581,7 +666,54 @@ extern "C" int64_t tpx_stage_execute(tpx_stage* st, const tpx_partition* parts,
 static int64_t csv_boundary_and_core(tpx_stage* st, void* d_in_p, long long size,
                                      const uint8_t* host_bytes,
                                      int64_t first_global_row, int64_t flags,
-                                     tpx_result* res);
+                                     tpx_result* res) {
+    hipStream_t stream = nullptr;
+    int dev = cur_device();
+    hipEvent_t eb0, eb1;
+    hipEventCreate(&eb0); hipEventCreate(&eb1);
+    hipEventRecord(eb0, stream);
+    long long nchunks = (size + 4095) / 4096;  // TPX_CSV_CHUNK
+    ARENA_TAKE(d_q, (size_t)nchunks * 8);
+    ARENA_TAKE(d_c0, (size_t)nchunks * 8);
+    ARENA_TAKE(d_c1, (size_t)nchunks * 8);
+    ARENA_TAKE(d_qs, (size_t)nchunks * 8);
+    ARENA_TAKE(d_rc, (size_t)nchunks * 8);
+    ARENA_TAKE(d_base, (size_t)nchunks * 8);
+    // wave-cooperative kernels: one wave per chunk, 4 waves per 256-thread block
+    unsigned grid_w = (unsigned)std::min<long long>((nchunks + 3) / 4, 8192);
+    unsigned grid_t = (unsigned)std::min<long long>((nchunks + 255) / 256, 8192);
+    {
+        void* args[] = {&d_in_p, (void*)&size, &nchunks, &d_q, &d_c0, &d_c1};
+        if (launch(st->k_csv_chunk, grid_w, 256, stream, args)) return -1;
+    }
+    long long qtotal = 0;
+    if (dev_scan(st, stream, (long long*)d_q, (long long*)d_qs, nchunks, &qtotal))
+        return -1;
+    {
+        void* args[] = {&d_qs, &d_c0, &d_c1, &d_rc, &nchunks};
+        if (launch(st->k_csv_sel, grid_t, 256, stream, args)) return -1;
+    }
+    long long nrows = 0;
+    if (dev_scan(st, stream, (long long*)d_rc, (long long*)d_base, nchunks,
+                 &nrows))
+        return -1;
+    if (nrows == 0) { empty_result(res, st->desc.sink == "mem"); return 0; }
+    ARENA_TAKE(d_offs, ((size_t)nrows + 1) * 8);
+    {
+        void* args[] = {&d_in_p, (void*)&size, &nchunks, &d_qs, &d_base, &d_offs};
+        if (launch(st->k_csv_rows, grid_w, 256, stream, args)) return -1;
+    }
+    hipEventRecord(eb1, stream);
+    HIP_CHECK(hipEventSynchronize(eb1));
+    float ms = 0;
+    hipEventElapsedTime(&ms, eb0, eb1);
+    res->t_boundary_ms = ms;
+    (void)hipEventDestroy(eb0); (void)hipEventDestroy(eb1);
+
+    PayloadSrc psrc{host_bytes, nullptr, 0};
+    return run_core(st, d_in_p, d_offs, nrows, first_global_row, size, res, psrc,
+                    stream, flags);
+}
 
 extern "C" int64_t tpx_stage_execute_csv(tpx_stage* st, const uint8_t* csv_bytes,
                                          int64_t size, int64_t first_global_row,
@@ -593,15 +725,16 @@ extern "C" int64_t tpx_stage_execute_csv(tpx_stage* st, const uint8_t* csv_bytes
         return -1;
     }
     hipStream_t stream = nullptr;
+    int dev = cur_device();
+    g_arena[dev].begin();
     res->bytes_in = size;
     if (size == 0) { empty_result(res, st->desc.sink == "mem"); return 0; }
 
     hipEvent_t ev0, ev1;
     hipEventCreate(&ev0); hipEventCreate(&ev1);
     hipEventRecord(ev0, stream);
-    DevBuf d_in;
-    HIP_CHECK(d_in.alloc((size_t)size));
-    HIP_CHECK(hipMemcpyAsync(d_in.p, csv_bytes, (size_t)size,
+    ARENA_TAKE(d_in, (size_t)size);
+    HIP_CHECK(hipMemcpyAsync(d_in, csv_bytes, (size_t)size,
                              hipMemcpyHostToDevice, stream));
     hipEventRecord(ev1, stream);
     HIP_CHECK(hipEventSynchronize(ev1));
@@ -610,8 +743,8 @@ extern "C" int64_t tpx_stage_execute_csv(tpx_stage* st, const uint8_t* csv_bytes
     res->t_h2d_ms = ms;
     (void)hipEventDestroy(ev0); (void)hipEventDestroy(ev1);
 
-    return csv_boundary_and_core(st, d_in.p, size, csv_bytes, first_global_row,
-                                 0, res);
+    return csv_boundary_and_core(st, d_in, size, csv_bytes, first_global_row, 0,
+                                 res);
 }
 
 // ---------------------------------------------------------------------------------
@@ -633,71 +766,6 @@ extern "C" void tpx_dev_free(uint64_t ptr) {
     if (ptr) (void)hipFree((void*)(uintptr_t)ptr);
 }
 
-// csv execution over device-resident bytes. NOTE: exception payloads need host
-// bytes; with a device-only input, exception raw lines are copied back D2H
-// per-record (rare path).
-static int64_t csv_boundary_and_core(tpx_stage* st, void* d_in_p, long long size,
-                                     const uint8_t* host_bytes,
-                                     int64_t first_global_row, int64_t flags,
-                                     tpx_result* res) {
-    hipStream_t stream = nullptr;
-    hipEvent_t eb0, eb1;
-    hipEventCreate(&eb0); hipEventCreate(&eb1);
-    hipEventRecord(eb0, stream);
-    long long nchunks = (size + 4095) / 4096;  // TPX_CSV_CHUNK
-    DevBuf d_q, d_c0, d_c1, d_qs, d_rc, d_base;
-    HIP_CHECK(d_q.alloc((size_t)nchunks * 8));
-    HIP_CHECK(d_c0.alloc((size_t)nchunks * 8));
-    HIP_CHECK(d_c1.alloc((size_t)nchunks * 8));
-    HIP_CHECK(d_qs.alloc((size_t)nchunks * 8));
-    HIP_CHECK(d_rc.alloc((size_t)nchunks * 8));
-    HIP_CHECK(d_base.alloc((size_t)nchunks * 8));
-    unsigned grid = (unsigned)std::min<long long>((nchunks + 255) / 256, 8192);
-    {
-        void* args[] = {&d_in_p, (void*)&size, &nchunks, &d_q.p, &d_c0.p, &d_c1.p};
-        if (launch(st->k_csv_chunk, grid, 256, stream, args)) return -1;
-    }
-    long long qtotal = 0;
-    if (dev_scan(st, stream, (long long*)d_q.p, (long long*)d_qs.p, nchunks,
-                 &qtotal))
-        return -1;
-    {
-        void* args[] = {&d_qs.p, &d_c0.p, &d_c1.p, &d_rc.p, &nchunks};
-        if (launch(st->k_csv_sel, grid, 256, stream, args)) return -1;
-    }
-    long long nrows = 0;
-    if (dev_scan(st, stream, (long long*)d_rc.p, (long long*)d_base.p, nchunks,
-                 &nrows))
-        return -1;
-    if (nrows == 0) { empty_result(res, st->desc.sink == "mem"); return 0; }
-    DevBuf d_offs;
-    HIP_CHECK(d_offs.alloc(((size_t)nrows + 1) * 8));
-    {
-        void* args[] = {&d_in_p, (void*)&size, &nchunks, &d_qs.p, &d_base.p,
-                        &d_offs.p};
-        if (launch(st->k_csv_rows, grid, 256, stream, args)) return -1;
-    }
-    hipEventRecord(eb1, stream);
-    HIP_CHECK(hipEventSynchronize(eb1));
-    float ms = 0;
-    hipEventElapsedTime(&ms, eb0, eb1);
-    res->t_boundary_ms = ms;
-    (void)hipEventDestroy(eb0); (void)hipEventDestroy(eb1);
-
-    if (host_bytes) {
-        PayloadSrc psrc{host_bytes, nullptr, 0};
-        return run_core(st, d_in_p, d_offs.p, nrows, first_global_row, size, res,
-                        psrc, stream, flags);
-    }
-    // device-only input: run core without payloads, then fetch raw lines D2H
-    PayloadSrc psrc{nullptr, nullptr, 0};
-    // run_core builds payloads only when psrc.bytes or psrc.parts set; with
-    // neither, exceptions keep (row,ec,opid) and payloads are fetched here.
-    int64_t rc = run_core(st, d_in_p, d_offs.p, nrows, first_global_row, size,
-                          res, psrc, stream, flags | 4 /* defer payloads */);
-    return rc;
-}
-
 extern "C" int64_t tpx_stage_execute_csv_dev(tpx_stage* st, uint64_t dev_bytes,
                                              int64_t size,
                                              int64_t first_global_row,
@@ -705,6 +773,8 @@ extern "C" int64_t tpx_stage_execute_csv_dev(tpx_stage* st, uint64_t dev_bytes,
     memset(res, 0, sizeof(*res));
     if (!st->loaded) { set_err("stage not loaded on a GPU"); return -1; }
     res->bytes_in = size;
+    int dev = cur_device();
+    g_arena[dev].begin();
     if (size == 0) { empty_result(res, st->desc.sink == "mem"); return 0; }
     return csv_boundary_and_core(st, (void*)(uintptr_t)dev_bytes, size, nullptr,
                                  first_global_row, flags, res);

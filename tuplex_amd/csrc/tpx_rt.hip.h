@@ -495,24 +495,80 @@ __device__ __forceinline__ char* tpx_csv_cell_write(char* w, const tstr s) {
 // parallel-exact restatement of the reference's sequential state machine.
 
 #define TPX_CSV_CHUNK 4096
+#define TPX_CSV_LANE_BYTES 64  /* 4096 / 64 lanes */
 
-extern "C" __global__ void tpx_csv_chunk_stats(const char* __restrict__ data,
-                                               long long size, long long nchunks,
-                                               long long* __restrict__ q,
-                                               long long* __restrict__ c0,
-                                               long long* __restrict__ c1) {
-    long long stride = (long long)gridDim.x * blockDim.x;
-    for (long long c = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-         c < nchunks; c += stride) {
-        long long a = c * TPX_CSV_CHUNK;
-        long long b = a + TPX_CSV_CHUNK < size ? a + TPX_CSV_CHUNK : size;
-        long long qq = 0, e0 = 0, e1 = 0;
+// wave helpers: 64-lane exclusive scan / reduction via shfl
+__device__ __forceinline__ long long tpx_wave_exscan(long long v) {
+    long long x = v;
+    for (int o = 1; o < 64; o <<= 1) {
+        long long y = __shfl_up((long long)x, o);
+        if ((threadIdx.x & 63) >= o) x += y;
+    }
+    return x - v;  // exclusive
+}
+
+__device__ __forceinline__ long long tpx_wave_sum(long long v) {
+    for (int o = 32; o; o >>= 1) v += __shfl_down((long long)v, o);
+    return __shfl((long long)v, 0);
+}
+
+// per-lane scan of its 64-byte slice: counts quotes and newlines at even/odd
+// LOCAL quote parity. Coalesced uint4 loads when the slice is full.
+__device__ __forceinline__ void tpx_scan64(const char* __restrict__ data,
+                                           long long a, long long b,
+                                           long long* lq, long long* l0,
+                                           long long* l1) {
+    long long qq = 0, e0 = 0, e1 = 0;
+    if (b - a == TPX_CSV_LANE_BYTES && ((a & 15) == 0)) {
+        #pragma unroll
+        for (int v = 0; v < 4; ++v) {
+            uint4 w = *(const uint4*)(data + a + v * 16);
+            unsigned words[4] = {w.x, w.y, w.z, w.w};
+            #pragma unroll
+            for (int k = 0; k < 4; ++k)
+                #pragma unroll
+                for (int s = 0; s < 4; ++s) {
+                    unsigned ch = (words[k] >> (8 * s)) & 0xFF;
+                    if (ch == '"') ++qq;
+                    else if (ch == '\n') { if (qq & 1) ++e1; else ++e0; }
+                }
+        }
+    } else {
         for (long long i = a; i < b; ++i) {
             char ch = data[i];
             if (ch == '"') ++qq;
             else if (ch == '\n') { if (qq & 1) ++e1; else ++e0; }
         }
-        q[c] = qq; c0[c] = e0; c1[c] = e1;
+    }
+    *lq = qq; *l0 = e0; *l1 = e1;
+}
+
+// wave-cooperative chunk stats: one wave per 4 KB chunk, lane i owns bytes
+// [i*64, i*64+64); lane parity composed by a wave scan — parallel-exact
+// restatement of the sequential quote-parity state machine.
+extern "C" __global__ void tpx_csv_chunk_stats(const char* __restrict__ data,
+                                               long long size, long long nchunks,
+                                               long long* __restrict__ q,
+                                               long long* __restrict__ c0,
+                                               long long* __restrict__ c1) {
+    int lane = threadIdx.x & 63;
+    int wid = threadIdx.x >> 6;
+    int wpb = blockDim.x >> 6;
+    long long wstride = (long long)gridDim.x * wpb;
+    for (long long c = (long long)blockIdx.x * wpb + wid; c < nchunks;
+         c += wstride) {
+        long long a = c * TPX_CSV_CHUNK + (long long)lane * TPX_CSV_LANE_BYTES;
+        long long b = a + TPX_CSV_LANE_BYTES;
+        if (a > size) a = size;
+        if (b > size) b = size;
+        long long lq, l0, l1;
+        tpx_scan64(data, a, b, &lq, &l0, &l1);
+        long long pref = tpx_wave_exscan(lq);
+        bool odd = (pref & 1) != 0;
+        long long tot_q = tpx_wave_sum(lq);
+        long long tot0 = tpx_wave_sum(odd ? l1 : l0);
+        long long tot1 = tpx_wave_sum(odd ? l0 : l1);
+        if (lane == 0) { q[c] = tot_q; c0[c] = tot0; c1[c] = tot1; }
     }
 }
 
@@ -532,18 +588,30 @@ extern "C" __global__ void tpx_csv_emit_rows(const char* __restrict__ data,
                                              const long long* __restrict__ qscan,
                                              const long long* __restrict__ base,
                                              long long* __restrict__ row_offs) {
-    long long stride = (long long)gridDim.x * blockDim.x;
-    long long tid0 = (long long)blockIdx.x * blockDim.x + threadIdx.x;
-    if (tid0 == 0) row_offs[0] = 0;
-    for (long long c = tid0; c < nchunks; c += stride) {
-        long long a = c * TPX_CSV_CHUNK;
-        long long b = a + TPX_CSV_CHUNK < size ? a + TPX_CSV_CHUNK : size;
-        long long p = qscan[c] & 1;
-        long long qq = 0, idx = base[c];
+    int lane = threadIdx.x & 63;
+    int wid = threadIdx.x >> 6;
+    int wpb = blockDim.x >> 6;
+    long long wstride = (long long)gridDim.x * wpb;
+    if (blockIdx.x == 0 && threadIdx.x == 0) row_offs[0] = 0;
+    for (long long c = (long long)blockIdx.x * wpb + wid; c < nchunks;
+         c += wstride) {
+        long long a = c * TPX_CSV_CHUNK + (long long)lane * TPX_CSV_LANE_BYTES;
+        long long b = a + TPX_CSV_LANE_BYTES;
+        if (a > size) a = size;
+        if (b > size) b = size;
+        long long lq, l0, l1;
+        tpx_scan64(data, a, b, &lq, &l0, &l1);
+        long long pref_q = tpx_wave_exscan(lq);
+        long long start_par = (qscan[c] + pref_q) & 1;
+        long long sel = start_par ? l1 : l0;   // valid newlines in this lane
+        long long lane_base = base[c] + tpx_wave_exscan(sel);
+        // second pass: emit offsets for valid newlines
+        long long qq = 0, idx = lane_base;
         for (long long i = a; i < b; ++i) {
             char ch = data[i];
             if (ch == '"') ++qq;
-            else if (ch == '\n' && (((qq + p) & 1) == 0)) row_offs[1 + idx++] = i + 1;
+            else if (ch == '\n' && (((qq + start_par) & 1) == 0))
+                row_offs[1 + idx++] = i + 1;
         }
     }
 }
