@@ -74,7 +74,8 @@ class _Emitter:
         self.lines: List[str] = []
         self.indent = 1
         self.ctr = 0
-        self.scopes = [{}]  # memo stacks: id(node) -> (var, opt)
+        self.scopes = [{}]       # memo stacks: id(node) -> (var, opt)
+        self.key_scopes = [{}]   # structural CSE: (op, argvars, ...) -> (var, opt)
 
     def w(self, line):
         self.lines.append("    " * self.indent + line)
@@ -85,9 +86,11 @@ class _Emitter:
 
     def push(self):
         self.scopes.append({})
+        self.key_scopes.append({})
 
     def pop(self):
         self.scopes.pop()
+        self.key_scopes.pop()
 
     def lookup(self, node):
         for s in reversed(self.scopes):
@@ -97,6 +100,16 @@ class _Emitter:
 
     def memo(self, node, val):
         self.scopes[-1][id(node)] = val
+        return val
+
+    def lookup_key(self, key):
+        for s in reversed(self.key_scopes):
+            if key in s:
+                return s[key]
+        return None
+
+    def memo_key(self, key, val):
+        self.key_scopes[-1][key] = val
         return val
 
 
@@ -123,11 +136,29 @@ class StageCodegen:
         return "tstr{%s, %d}" % (self.lits[s], len(s.encode("utf-8")))
 
     # ---- TIR expression emission -------------------------------------------
+    # ops whose arguments must NOT be evaluated eagerly (control flow) or that
+    # need no CSE
+    _NONSTRICT = {"if", "and", "or", "const", "input", "mktuple"}
+
     def emit_expr(self, em: _Emitter, node, row_ctx, opid) -> Tuple[str, Optional[str]]:
-        """Returns (value_var, null_var or None)."""
+        """Returns (value_var, null_var or None). Structural CSE across operators:
+        the reference's per-stage fused LLVM function gets cross-operator CSE from
+        LLVM; here identical pure subexpressions (same op + same resolved operand
+        vars) are reused, scoped to the current branch."""
         hit = em.lookup(node)
         if hit is not None:
             return hit
+        if node["op"] not in self._NONSTRICT:
+            argvals = tuple(self.emit_expr(em, a, row_ctx, opid)
+                            for a in node["args"])
+            key = (node["op"], argvals, repr(node.get("v")), node.get("i"),
+                   node.get("w"), repr(node["t"]))
+            k_hit = em.lookup_key(key)
+            if k_hit is not None:
+                return em.memo(node, k_hit)
+            v = self._emit(em, node, row_ctx, opid)
+            em.memo_key(key, v)
+            return em.memo(node, v)
         v = self._emit(em, node, row_ctx, opid)
         return em.memo(node, v)
 
@@ -579,7 +610,7 @@ class StageCodegen:
         L.append("    long long* exc_buf, unsigned long long* exc_count,"
                  " unsigned long long exc_cap,")
         L.append("    void** outv) {")
-        L.append("  TpxHeap heap{heap_base, heap_cursor, heap_cap};")
+        L.append("  TpxHeap heap{heap_base, heap_cursor, heap_cap, nullptr, nullptr};")
         L.append("  long long stride = (long long)gridDim.x * blockDim.x;")
         L.append("  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;"
                  " i < n; i += stride) {")

@@ -426,9 +426,10 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
         P.exc_cap_recs = exc_cap;
     }
     exc_cap = P.exc_cap_recs;
+    // +thread-chunk slack: <=2048 blocks x 256 threads x TPX_HEAP_CHUNK(256 B)
     unsigned long long heap_cap =
         (unsigned long long)std::max<long long>(in_bytes + (in_bytes >> 1) +
-                                                (16 << 20), 1 << 20);
+                                                (160ll << 20), 1 << 20);
     if (P.heap_cap < heap_cap) {
         if (P.heap) (void)hipFree(P.heap);
         HIP_CHECK(hipMalloc(&P.heap, heap_cap));
@@ -441,7 +442,9 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
     unsigned long long exc_count = 0;
     for (int attempt = 0;; ++attempt) {
         HIP_CHECK(hipMemsetAsync(P.counters, 0, 16, stream));
-        unsigned grid = (unsigned)std::min<long long>((n + 255) / 256, 8192);
+        // 2048 blocks x 256 = exactly 8 blocks/CU on 256 CUs (full occupancy);
+        // grid-stride covers the rest; bounds per-thread heap-chunk waste
+        unsigned grid = (unsigned)std::min<long long>((n + 255) / 256, 2048);
         void* args[] = {&d_in, &d_offs, &n, &row0, &P.heap, &d_heap_cursor,
                         &heap_cap, &d_keep, &d_keep01, &d_sizes, &P.exc,
                         &d_exc_count, &exc_cap, &d_outv};

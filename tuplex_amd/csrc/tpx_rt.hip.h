@@ -44,17 +44,37 @@ struct tstr {
     long long n;
 };
 
+// String heap: per-thread chunked bump allocation (one global atomic per 256-B
+// chunk per thread instead of one per allocation — a single global cursor word
+// saturates at ~88 atomics/µs, MI355X_MICROARCH.md §dequeue). The rtmalloc-arena
+// analog (Runtime.cc:186), device-side.
+#define TPX_HEAP_CHUNK 256
+
 struct TpxHeap {
     char* base;
     unsigned long long* cursor;  // device-global bump cursor
     unsigned long long cap;
+    char* tl_cur;                // per-thread chunk (registers)
+    char* tl_end;
 };
 
 __device__ __forceinline__ char* tpx_alloc(TpxHeap& h, long long n) {
     if (n <= 0) return h.base;  // zero-size alloc: any valid pointer
-    unsigned long long off = atomicAdd(h.cursor, (unsigned long long)n);
-    if (off + (unsigned long long)n > h.cap) return nullptr;
-    return h.base + off;
+    if (n > TPX_HEAP_CHUNK) {
+        unsigned long long off = atomicAdd(h.cursor, (unsigned long long)n);
+        if (off + (unsigned long long)n > h.cap) return nullptr;
+        return h.base + off;
+    }
+    if (h.tl_cur + n > h.tl_end) {
+        unsigned long long off =
+            atomicAdd(h.cursor, (unsigned long long)TPX_HEAP_CHUNK);
+        if (off + TPX_HEAP_CHUNK > h.cap) return nullptr;
+        h.tl_cur = h.base + off;
+        h.tl_end = h.tl_cur + TPX_HEAP_CHUNK;
+    }
+    char* p = h.tl_cur;
+    h.tl_cur += n;
+    return p;
 }
 
 __device__ __forceinline__ bool tpx_is_ascii_byte(unsigned char c) { return c < 0x80; }
