@@ -86,6 +86,18 @@ __device__ __forceinline__ bool tpx_ascii(const tstr s) {
     return true;
 }
 
+// translate an LDS-staged string view back to its global-memory address before it
+// escapes the kernel (columnar string cells must outlive the LDS staging window).
+// rbase = wave_lds - span_start, so (p - rbase) is the byte offset in the input.
+__device__ __forceinline__ tstr tpx_to_global(tstr s, const char* rbase,
+                                              const unsigned char* gbase,
+                                              const char* lds_lo,
+                                              const char* lds_hi) {
+    if (s.p >= lds_lo && s.p < lds_hi)
+        return tstr{(const char*)gbase + (s.p - rbase), s.n};
+    return s;
+}
+
 // ---- python-semantics helpers ---------------------------------------------------
 
 __device__ __forceinline__ long long tpx_floordiv_i64(long long a, long long b, int* ec) {
