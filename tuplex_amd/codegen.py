@@ -633,8 +633,10 @@ class StageCodegen:
         L.append("    long long span_start = in_offs[r0] & ~15LL;  // align staging window")
         L.append("    long long span_end = in_offs[rhi];")
         L.append("    long long span = span_end - span_start;")
-        L.append("    const char* rbase;")
-        L.append("    if (span <= TPX_SPAN_CAP) {")
+        # NB: never form (wave_lds - span_start): an LDS generic pointer offset
+        # outside the aperture is UB once the addrspace-inference pass narrows it
+        L.append("    bool staged = span <= TPX_SPAN_CAP;")
+        L.append("    if (staged) {")
         L.append("      for (long long k = (long long)lane * 16; k < span; k += 64 * 16) {")
         L.append("        if (k + 16 <= span)")
         L.append("          *(uint4*)(wave_lds + k) = *(const uint4*)((const char*)in_data + span_start + k);")
@@ -642,9 +644,6 @@ class StageCodegen:
         L.append("          for (long long j = k; j < span; ++j)")
         L.append("            wave_lds[j] = ((const char*)in_data)[span_start + j];")
         L.append("      }")
-        L.append("      rbase = wave_lds - span_start;")
-        L.append("    } else {")
-        L.append("      rbase = (const char*)in_data;")
         L.append("    }")
         L.append("    long long i = r0 + lane;")
         L.append("    if (i >= rhi) continue;")
@@ -682,7 +681,9 @@ class StageCodegen:
         """Deserialize one reference-layout row (Serializer.cc:20-24) into typed
         locals c0..cN."""
         L = ["    long long prc = 0;  // no pre-parse errors on the mem path"]
-        L.append("    const unsigned char* row = (const unsigned char*)(rbase + in_offs[i]);")
+        L.append("    const unsigned char* row = staged"
+                 " ? (const unsigned char*)(wave_lds + (in_offs[i] - span_start))"
+                 " : in_data + in_offs[i];")
         n_opt = sum(1 for t in in_types if T.is_opt(t))
         bitmap_size = ((n_opt + 63) // 64) * 8 if n_opt else 0
         L.append("    // deserialize (bitmap %dB, %d slots)" % (bitmap_size, len(in_types)))
@@ -722,8 +723,10 @@ class StageCodegen:
         null_values = self.csv_info.get("null_values", [""])
         L = []
         L.append("    long long prc = 0;")
-        L.append("    const char* rp = rbase + in_offs[i];")
-        L.append("    const char* rend = rbase + in_offs[i+1];")
+        L.append("    const char* rp = staged ? wave_lds + (in_offs[i] - span_start)"
+                 " : (const char*)in_data + in_offs[i];")
+        L.append("    const char* rend = staged ? wave_lds + (in_offs[i+1] - span_start)"
+                 " : (const char*)in_data + in_offs[i+1];")
         L.append("    if (rend > rp && rend[-1] == '\\n') --rend;")
         L.append("    if (rend > rp && rend[-1] == '\\r') --rend;")
         # straight-line per-cell scan: one named local per cell — a dynamically
@@ -781,8 +784,9 @@ class StageCodegen:
         for k, t in enumerate(out_types):
             base = T.deopt(t)
             if base == T.STR:
-                L.append("    o.o%d = tpx_to_global(o.o%d, rbase, in_data, wave_lds,"
-                         " wave_lds + TPX_SPAN_CAP);" % (k, k))
+                L.append("    o.o%d = tpx_to_global(o.o%d, wave_lds,"
+                         " wave_lds + TPX_SPAN_CAP, in_data, span_start);"
+                         % (k, k))
                 L.append("    ((unsigned long long*)outv[%d])[i] = (unsigned long long)o.o%d.p;"
                          % (3 * k, k))
                 L.append("    ((int*)outv[%d])[i] = (int)o.o%d.n;" % (3 * k + 1, k))

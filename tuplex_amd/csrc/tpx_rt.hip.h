@@ -88,13 +88,13 @@ __device__ __forceinline__ bool tpx_ascii(const tstr s) {
 
 // translate an LDS-staged string view back to its global-memory address before it
 // escapes the kernel (columnar string cells must outlive the LDS staging window).
-// rbase = wave_lds - span_start, so (p - rbase) is the byte offset in the input.
-__device__ __forceinline__ tstr tpx_to_global(tstr s, const char* rbase,
+// (p - lds_lo) + span_start is the byte offset in the input.
+__device__ __forceinline__ tstr tpx_to_global(tstr s, const char* lds_lo,
+                                              const char* lds_hi,
                                               const unsigned char* gbase,
-                                              const char* lds_lo,
-                                              const char* lds_hi) {
+                                              long long span_start) {
     if (s.p >= lds_lo && s.p < lds_hi)
-        return tstr{(const char*)gbase + (s.p - rbase), s.n};
+        return tstr{(const char*)gbase + span_start + (s.p - lds_lo), s.n};
     return s;
 }
 
