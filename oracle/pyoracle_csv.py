@@ -47,7 +47,7 @@ def split_rows(data: bytes) -> List[bytes]:
     return rows
 
 
-def split_cells(line: bytes):
+def split_cells(line: bytes, delim: bytes = b","):
     end = len(line)
     while end > 0 and line[end - 1:end] in (b"\n", b"\r"):
         end -= 1
@@ -76,9 +76,9 @@ def split_cells(line: bytes):
             if esc:
                 flags |= 2
             q += 1
-            if q < end and line[q:q + 1] != b",":
+            if q < end and line[q:q + 1] != delim:
                 flags |= 4
-            while q < end and line[q:q + 1] != b",":
+            while q < end and line[q:q + 1] != delim:
                 q += 1
             if q < end:
                 more = True
@@ -86,7 +86,7 @@ def split_cells(line: bytes):
             p = q
         else:
             q = p
-            while q < end and line[q:q + 1] != b",":
+            while q < end and line[q:q + 1] != delim:
                 q += 1
             cells.append(line[p:q])
             if q < end:
@@ -127,8 +127,19 @@ def _try_bool(s: str):
     return None
 
 
-def sniff(sample: bytes, null_values, threshold, header, columns):
-    rows = [split_cells(r)[0] for r in split_rows(sample)]
+def sniff_delimiter(sample: bytes) -> bytes:
+    first = split_rows(sample)[0]
+    best, bestn = b",", -1
+    for cand in (b",", b";", b"|", b"\t"):
+        n = len(split_cells(first, cand)[0])
+        if n > bestn:
+            best, bestn = cand, n
+    return best
+
+
+def sniff(sample: bytes, null_values, threshold, header, columns,
+          delim: bytes = b","):
+    rows = [split_cells(r, delim)[0] for r in split_rows(sample)]
     rows = [r for r in rows if r]
     txt = [[c.decode("utf-8", "replace") for c in r] for r in rows]
 
@@ -239,12 +250,13 @@ class _BadParse(Exception):
 _BadParse.__name__ = "BadParseStringInput"
 
 
-def _replay_row(raw_line: bytes, col_types, null_values, row_ops, columns):
+def _replay_row(raw_line: bytes, col_types, null_values, row_ops, columns,
+                delim: str = ","):
     """Interpreter replay: full RFC-4180 parse + CPython conversions (matches the
     product's csvio.replay_csv_row rules)."""
     text = raw_line.decode("utf-8", "replace").rstrip("\n").rstrip("\r")
     try:
-        cells = next(_pycsv.reader(io.StringIO(text)))
+        cells = next(_pycsv.reader(io.StringIO(text), delimiter=delim))
     except (StopIteration, _pycsv.Error):
         return ("exc", _BadParse("unparseable"))
     if len(cells) != len(col_types):
@@ -273,7 +285,8 @@ def _replay_row(raw_line: bytes, col_types, null_values, row_ops, columns):
 
 
 def run_csv_pipeline(csv_bytes: bytes, ops, columns=None, header=None,
-                     null_values=None, threshold=0.9, sink="collect"):
+                     null_values=None, threshold=0.9, sink="collect",
+                     delimiter=None):
     """Full oracle CSV pipeline. sink='collect' -> values; 'csv' -> output text
     (bytes, incl. header line)."""
     if null_values is None:
@@ -282,8 +295,9 @@ def run_csv_pipeline(csv_bytes: bytes, ops, columns=None, header=None,
     nl = sample.rfind(b"\n")
     if nl >= 0:
         sample = sample[:nl + 1]
+    delim = delimiter.encode() if delimiter else sniff_delimiter(sample)
     has_header, names, col_types = sniff(sample, null_values, threshold, header,
-                                         columns)
+                                         columns, delim)
     data = csv_bytes
     if has_header:
         p = data.find(b"\n")
@@ -305,12 +319,13 @@ def run_csv_pipeline(csv_bytes: bytes, ops, columns=None, header=None,
         exc_counts[nm] = exc_counts.get(nm, 0) + 1
 
     for i, line in enumerate(split_rows(data)):
-        cells, flags = split_cells(line)
+        cells, flags = split_cells(line, delim)
         pr = _fast_parse_row(cells, flags, col_types, null_values)
         if pr[0] == "row":
             r = pyoracle.process_row(pr[1], row_ops, names, fast=True)
         else:
-            r = _replay_row(line, col_types, null_values, row_ops, names)
+            r = _replay_row(line, col_types, null_values, row_ops, names,
+                            delim.decode())
         if r[0] == "row":
             v = r[1]
             if isinstance(v, tuple) and len(v) == 1:

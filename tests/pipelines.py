@@ -137,7 +137,21 @@ PIPELINES = [
     ("filter_all", list(range(100)), None, [("filter", gt1000)]),
     ("filter_none", list(range(100)), None, [("filter", ge0)]),
     ("zillow_mem", make_zillow_rows(2000, seed=42), ZILLOW_COLS, zillow_ops()),
+    ("agg_sum_large", list(range(1, 100001)), None,
+     [("aggregate", agg_combine, agg_sum, 0)]),
 ]
+
+
+def _long_rows():
+    from tests.extra_data import make_long_rows
+    return make_long_rows(300, seed=9)
+
+
+def long_probe(x):
+    return (x.find("zzz"), len(x), x[:16])
+
+
+PIPELINES.append(("long_rows", _long_rows(), None, [("map", long_probe)]))
 
 
 def apply_ops(ds, ops):
@@ -166,6 +180,46 @@ def apply_ops(ds, ops):
     return ds
 
 
+def precompile_csv(verbose=False):
+    """Pre-compile the CSV-source stages (zillow/q6/flights/logs) so their hsacos
+    are in the travelling cache."""
+    from tests import extra_data as X
+    from tests.zillow_data import make_zillow_csv_bytes
+    from tuplex_amd import codegen, csvio, engine, plan
+
+    glib = engine.GpuLib.get()
+    jobs = []
+    zdata, _ = make_zillow_csv_bytes(2000, seed=42, dirty_frac=0.02)
+    jobs.append(("zillow_csv", zdata, None, None, None, zillow_ops(),
+                 ["mem", "csv"]))
+    jobs.append(("q6", X.make_lineitem_csv(2000), X.LINEITEM_COLS, False, "|",
+                 X.q6_ops(), ["mem"]))
+    jobs.append(("flights", X.make_flights_csv(500), None, None, None,
+                 X.flights_ops(), ["mem"]))
+    jobs.append(("logs", X.make_weblog_lines(2000), None, False, None,
+                 X.logs_ops(), ["mem"]))
+    n = 0
+    for name, data, columns, header, delimiter, ops, sinks in jobs:
+        sample = data[:1 << 20]
+        delim = delimiter.encode() if delimiter else csvio.sniff_delimiter(sample)
+        _h, names, col_types = csvio.sniff(sample, [""], 0.9, header, columns,
+                                           delim)
+        sp = plan.build_stage(col_types, names, ops)
+        if not sp.compilable:
+            if verbose:
+                print("skip:", name, sp.why_not_compilable)
+            continue
+        for sink in sinks:
+            src, desc = codegen.generate_stage(
+                sp, source="csv", sink=sink,
+                csv_info={"null_values": [""], "delimiter": delim.decode()})
+            glib.compile_stage(src, desc, compile_only=True)
+            n += 1
+            if verbose:
+                print("precompiled csv:", name, sink)
+    return n
+
+
 def precompile_all(verbose=False):
     """Generate + hipRTC-compile (compile-only) every pipeline's stage so the
     hsaco cache is warm. Works with no GPU."""
@@ -189,4 +243,5 @@ def precompile_all(verbose=False):
         n += 1
         if verbose:
             print("precompiled:", name)
+    n += precompile_csv(verbose=verbose)
     return n

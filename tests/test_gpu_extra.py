@@ -1,0 +1,91 @@
+"""GPU parity for the remaining BASELINE configs: TPC-H Q6 aggregate (configs[3]),
+flights-shaped wide schema w/ ~1% malformed rows (configs[2]), weblog string-split
+(configs[4]), plus span-cap fallback (long rows) and the GPU-reduced aggregates."""
+import os
+
+import pytest
+
+import tuplex_amd
+from oracle import pyoracle, pyoracle_csv
+from tests.pipelines import apply_ops
+from tests import extra_data as X
+
+pytestmark = pytest.mark.gpu
+
+
+def _write(tmp_path, data, name="d.csv"):
+    p = os.path.join(str(tmp_path), name)
+    with open(p, "wb") as f:
+        f.write(data)
+    return p
+
+
+def test_q6_aggregate_csv(tmp_path):
+    data = make = X.make_lineitem_csv(200000, seed=42)
+    path = _write(tmp_path, data)
+    ctx = tuplex_amd.Context()
+    ds = apply_ops(ctx.csv(path, columns=X.LINEITEM_COLS, header=False,
+                           delimiter="|"), X.q6_ops())
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    ref = pyoracle_csv.run_csv_pipeline(data, X.q6_ops(),
+                                        columns=X.LINEITEM_COLS, header=False,
+                                        delimiter="|")
+    assert len(got) == 1
+    r, g = ref["output"][0], got[0]
+    assert abs(g - r) <= 1e-9 * max(1.0, abs(r)), (g, r)
+    assert ds.exception_counts == ref["exception_counts"]
+    assert r != 0.0
+
+
+def test_agg_sum_count_exact_gpu():
+    """i64 aggregates are exact regardless of reduction order."""
+    ctx = tuplex_amd.Context()
+    data = list(range(1, 100001))
+    from tests.pipelines import agg_combine, agg_sum
+    ds = ctx.parallelize(data).aggregate(agg_combine, agg_sum, 0)
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu"
+    assert got == [sum(data)]
+
+
+def test_flights_wide_csv(tmp_path):
+    data = X.make_flights_csv(3000, seed=7, bad_frac=0.01)
+    path = _write(tmp_path, data)
+    ctx = tuplex_amd.Context()
+    ds = apply_ops(ctx.csv(path), X.flights_ops())
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    ref = pyoracle_csv.run_csv_pipeline(data, X.flights_ops())
+    assert got == ref["output"]
+    assert ds.exception_counts == ref["exception_counts"]
+    assert len(got) > 100
+
+
+def test_logs_split_csv(tmp_path):
+    data = X.make_weblog_lines(20000, seed=3, bad_frac=0.01)
+    path = _write(tmp_path, data, "access.log")
+    ctx = tuplex_amd.Context()
+    ds = apply_ops(ctx.csv(path, header=False), X.logs_ops())
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    ref = pyoracle_csv.run_csv_pipeline(data, X.logs_ops(), header=False)
+    assert got == ref["output"]
+    assert ds.exception_counts == ref["exception_counts"]
+    assert len(got) > 1000
+
+
+def test_long_rows_span_fallback():
+    """Rows far beyond the 16 KiB/wave LDS staging cap take the global-parse
+    path; results must be identical."""
+    data = X.make_long_rows(300, seed=9)
+
+    def probe(x):
+        return (x.find("zzz"), len(x), x[:16])
+
+    ctx = tuplex_amd.Context()
+    ds = ctx.parallelize(data).map(probe)
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu"
+    ref = pyoracle.run_pipeline(data, [("map", probe)])
+    assert got == ref["output"]

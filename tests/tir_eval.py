@@ -116,6 +116,8 @@ def _ev(n, row, rec):
         return str(v)
     if op == "abs":
         return abs(rec(a[0]))
+    if op == "splitget":
+        return rec(a[0]).split(rec(a[1]))[rec(a[2])]
     if op == "fmt_int":
         v = rec(a[0])
         return ("%0" + str(n["w"]) + "d") % v if n["w"] else "%d" % v

@@ -400,6 +400,14 @@ class StageCodegen:
             else:
                 em.w("double %s = fabs(%s);" % (r, x))
             return (r, None)
+        if op == "splitget":
+            x, _ = ev(a[0])
+            y, _ = ev(a[1])
+            z, _ = ev(a[2])
+            r = em.fresh("s")
+            em.w("tstr %s = tpx_splitget(%s, %s, %s, &_ec);" % (r, x, y, z))
+            self._check(em, opid)
+            return (r, None)
         if op == "fmt_int":
             x, _ = ev(a[0])
             r = em.fresh("s")
@@ -528,6 +536,10 @@ class StageCodegen:
                 pass
             else:
                 raise CodegenError("op %r not supported in codegen" % op.kind)
+        if getattr(sp, "agg_expr", None) is not None:
+            # per-row aggregate expr (the fold's rhs); device reduce sums it
+            v, nv = self.emit_expr(em, sp.agg_expr, rc, sp.agg_opid)
+            rc = [(v, sp.agg_expr["t"], nv)]
         return rc
 
     # ---- full source --------------------------------------------------------
@@ -535,7 +547,8 @@ class StageCodegen:
         """Returns (hip_source, stage_desc)."""
         sp = self.sp
         in_types = sp.input_types
-        out_types = sp.output_types
+        out_types = [sp.agg_type] if getattr(sp, "agg_expr", None) is not None \
+            else sp.output_types
 
         body = _Emitter()
         # inputs: c0..cN (+ c0_n null flags)
@@ -721,6 +734,9 @@ class StageCodegen:
         the reference's BADPARSE_STRING_INPUT rows)."""
         nc = len(in_types)
         null_values = self.csv_info.get("null_values", [""])
+        delim = self.csv_info.get("delimiter", ",")
+        assert len(delim) == 1
+        delim_c = "'\\t'" if delim == "\t" else "'%s'" % delim
         L = []
         L.append("    long long prc = 0;")
         L.append("    const char* rp = staged ? wave_lds + (in_offs[i] - span_start)"
@@ -748,8 +764,8 @@ class StageCodegen:
             L.append("    tpx_cell cl%d{rp, 0, 0};" % idx)
             L.append("    if (!prc) {")
             L.append("      if (!avail) prc = %d;  // CSV_UNDERRUN" % 20)
-            L.append("      else { cur = tpx_csv_next_cell(cur, rend, &cl%d, &m);"
-                     " avail = m; badf |= cl%d.flags; }" % (idx, idx))
+            L.append("      else { cur = tpx_csv_next_cell(cur, rend, &cl%d, &m, %s);"
+                     " avail = m; badf |= cl%d.flags; }" % (idx, delim_c, idx))
             L.append("    }")
             if opt:
                 L.append("    bool c%d_n = !prc && (%s);" % (idx, null_check(idx)))
@@ -969,6 +985,8 @@ class StageCodegen:
             return ("opt," if T.is_opt(t) else "") + T.deopt(t)
         lines = ["source=%s" % self.source, "sink=%s" % self.sink,
                  "nin=%d" % len(in_types), "nout=%d" % len(out_types)]
+        if getattr(self.sp, "agg_expr", None) is not None:
+            lines.append("agg=%s" % T.deopt(self.sp.agg_type))
         for i, t in enumerate(in_types):
             lines.append("in%d=%s" % (i, tdesc(t)))
         for i, t in enumerate(out_types):

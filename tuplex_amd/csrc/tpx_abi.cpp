@@ -73,6 +73,7 @@ struct ColDesc {
 struct StageDesc {
     std::string source;  // mem | csv
     std::string sink;    // mem | csv
+    std::string agg;     // "" | i64 | f64 — GPU-reducible aggregate fold
     std::vector<ColDesc> in_cols, out_cols;
 };
 
@@ -99,6 +100,7 @@ static bool parse_desc(const char* text, StageDesc* d) {
     }
     d->source = kv.count("source") ? kv["source"] : "mem";
     d->sink = kv.count("sink") ? kv["sink"] : "mem";
+    d->agg = kv.count("agg") ? kv["agg"] : "";
     int nin = atoi(kv["nin"].c_str());
     int nout = atoi(kv["nout"].c_str());
     for (int i = 0; i < nin; ++i) {
@@ -129,6 +131,8 @@ struct tpx_stage {
     hipFunction_t k_main = nullptr, k_write = nullptr;
     hipFunction_t k_scan_block = nullptr, k_scan_add = nullptr;
     hipFunction_t k_csv_chunk = nullptr, k_csv_sel = nullptr, k_csv_rows = nullptr;
+    hipFunction_t k_red_f64 = nullptr, k_red_f64_fin = nullptr;
+    hipFunction_t k_red_i64 = nullptr, k_red_i64_fin = nullptr;
     bool loaded = false;
 };
 
@@ -220,6 +224,10 @@ extern "C" tpx_stage* tpx_stage_compile(const char* hip_source,
         {"tpx_csv_chunk_stats", &st->k_csv_chunk, false},
         {"tpx_csv_select_counts", &st->k_csv_sel, false},
         {"tpx_csv_emit_rows", &st->k_csv_rows, false},
+        {"tpx_reduce_f64", &st->k_red_f64, false},
+        {"tpx_reduce_f64_final", &st->k_red_f64_fin, false},
+        {"tpx_reduce_i64", &st->k_red_i64, false},
+        {"tpx_reduce_i64_final", &st->k_red_i64_fin, false},
     };
     for (auto& e : lut) {
         hipError_t r = hipModuleGetFunction(e.fn, st->module, e.name);
@@ -479,6 +487,38 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
         }
     }
 
+    if (!D.agg.empty()) {
+        // GPU aggregate fold (Q6/count/sum): deterministic masked reduce of the
+        // per-row expr column; result = a 1-row partition [numRows=1][value]
+        bool is_f64 = D.agg == "f64";
+        hipFunction_t kr = is_f64 ? st->k_red_f64 : st->k_red_i64;
+        hipFunction_t krf = is_f64 ? st->k_red_f64_fin : st->k_red_i64_fin;
+        if (!kr || !krf) { set_err("reduce kernels missing"); return -1; }
+        long long nb = std::min<long long>((n + 2047) / 2048, 1024);
+        if (nb < 1) nb = 1;
+        ARENA_TAKE(d_partials, (size_t)nb * 8);
+        ARENA_TAKE(d_res8, 8);
+        void* vals = outv[0];
+        void* a1[] = {&vals, &d_keep, &n, &d_partials};
+        if (launch(kr, (unsigned)nb, 256, stream, a1)) return -1;
+        void* a2[] = {&d_partials, &nb, &d_res8};
+        if (launch(krf, 1, 256, stream, a2)) return -1;
+        hipEventRecord(evs1, stream);
+        hipEventRecord(ev2, stream);
+        long long bits = 0;
+        HIP_CHECK(hipMemcpyAsync(&bits, d_res8, 8, hipMemcpyDeviceToHost, stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
+        res->out_size = 16;
+        res->out_num_rows = 1;
+        res->out_data = (uint8_t*)malloc(16);
+        ((int64_t*)res->out_data)[0] = 1;
+        ((int64_t*)res->out_data)[1] = bits;
+        res->out_row_offsets = (int64_t*)malloc(16);
+        res->out_row_offsets[0] = 8;
+        res->out_row_offsets[1] = 16;
+        res->out_row_indices = (int64_t*)malloc(8);
+        res->out_row_indices[0] = 0;
+    } else {
     // compaction scans
     ARENA_TAKE(d_keep_scan, (size_t)n * 8);
     ARENA_TAKE(d_size_scan, (size_t)n * 8);
@@ -521,6 +561,8 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
             HIP_CHECK(hipMemcpyAsync(res->out_row_indices, d_out_rowidx,
                                      (size_t)total_rows * 8,
                                      hipMemcpyDeviceToHost, stream));
+    }
+    res->bytes_out = out_total;
     }
     std::vector<ExcRec> excs((size_t)exc_count);
     if (exc_count)
@@ -584,7 +626,7 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
         res->exc_size = (int64_t)total;
         res->exc_num_rows = (int64_t)exc_count;
     }
-    res->bytes_out = out_total;
+    if (!D.agg.empty()) res->bytes_out = 16;
     float ms = 0;
     hipEventElapsedTime(&ms, ev1, ev2); res->t_kernel_ms = ms;
     hipEventElapsedTime(&ms, ev2, ev3); res->t_d2h_ms = ms;

@@ -334,6 +334,41 @@ __device__ __forceinline__ tstr tpx_replace(TpxHeap& h, const tstr s, const tstr
     return tstr{d, outn};
 }
 
+// s.split(sep)[idx] fused — python list-index semantics (negative wraps over the
+// PART count; IndexError when out of range; ValueError on empty separator)
+__device__ __forceinline__ tstr tpx_splitget(const tstr s, const tstr sep,
+                                             long long idx, int* ec) {
+    if (sep.n == 0) { *ec = EC_VALUEERROR; return tstr{s.p, 0}; }
+    if (idx < 0) {
+        long long parts = 1;
+        for (long long i = 0; i + sep.n <= s.n;) {
+            bool m = true;
+            for (long long j = 0; j < sep.n; ++j)
+                if (s.p[i + j] != sep.p[j]) { m = false; break; }
+            if (m) { ++parts; i += sep.n; } else { ++i; }
+        }
+        idx += parts;
+        if (idx < 0) { *ec = EC_INDEXERROR; return tstr{s.p, 0}; }
+    }
+    long long k = 0, start = 0;
+    for (long long i = 0; i + sep.n <= s.n;) {
+        bool m = true;
+        for (long long j = 0; j < sep.n; ++j)
+            if (s.p[i + j] != sep.p[j]) { m = false; break; }
+        if (m) {
+            if (k == idx) return tstr{s.p + start, i - start};
+            ++k;
+            i += sep.n;
+            start = i;
+        } else {
+            ++i;
+        }
+    }
+    if (k == idx) return tstr{s.p + start, s.n - start};
+    *ec = EC_INDEXERROR;
+    return tstr{s.p, 0};
+}
+
 // ---- parse (StringUtils.cc:22 fast_atoi64 / :71 fast_atod, restated) ------------
 
 __device__ __forceinline__ int tpx_fast_atoi64(const char* start, const char* end,
@@ -656,7 +691,8 @@ struct tpx_cell { const char* p; long long n; int flags; };
 
 __device__ __forceinline__ const char* tpx_csv_next_cell(const char* p,
                                                          const char* end,
-                                                         tpx_cell* c, bool* more) {
+                                                         tpx_cell* c, bool* more,
+                                                         char delim) {
     c->flags = 0;
     *more = false;
     if (p < end && *p == '"') {
@@ -673,13 +709,13 @@ __device__ __forceinline__ const char* tpx_csv_next_cell(const char* p,
         if (q >= end) { c->p = p; c->n = end - p; c->flags = 4; return end; }
         c->p = s; c->n = q - s; c->flags = 1 | (esc ? 2 : 0);
         ++q;
-        if (q < end && *q != ',') c->flags |= 4;
-        while (q < end && *q != ',') ++q;
+        if (q < end && *q != delim) c->flags |= 4;
+        while (q < end && *q != delim) ++q;
         if (q < end) { *more = true; ++q; }
         return q;
     }
     const char* q = p;
-    while (q < end && *q != ',') ++q;
+    while (q < end && *q != delim) ++q;
     c->p = p; c->n = q - p;
     if (q < end) { *more = true; ++q; }
     return q;
@@ -733,6 +769,70 @@ __device__ __forceinline__ int tpx_cell_bool(const tpx_cell& c, bool* out) {
 }
 
 // ---- fixed kernels ---------------------------------------------------------------
+
+// ---- aggregate reduction (AggregateFunctions.cc fold -> deterministic device
+//      reduce: per-thread grid-stride partials in fixed index order, block tree,
+//      then a single-thread final pass — same result every run for a given n) ----
+
+#define TPX_RED_THREADS 256
+
+extern "C" __global__ void tpx_reduce_f64(const double* __restrict__ vals,
+                                          const unsigned char* __restrict__ keep,
+                                          long long n,
+                                          double* __restrict__ partials) {
+    __shared__ double sh[TPX_RED_THREADS];
+    long long stride = (long long)gridDim.x * blockDim.x;
+    double acc = 0.0;
+    for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += stride)
+        if (keep[i]) acc += vals[i];
+    sh[threadIdx.x] = acc;
+    __syncthreads();
+    for (int o = TPX_RED_THREADS / 2; o; o >>= 1) {
+        if ((int)threadIdx.x < o) sh[threadIdx.x] += sh[threadIdx.x + o];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) partials[blockIdx.x] = sh[0];
+}
+
+extern "C" __global__ void tpx_reduce_f64_final(const double* __restrict__ partials,
+                                                long long n,
+                                                double* __restrict__ out) {
+    if (blockIdx.x == 0 && threadIdx.x == 0) {
+        double s = 0.0;
+        for (long long i = 0; i < n; ++i) s += partials[i];
+        *out = s;
+    }
+}
+
+extern "C" __global__ void tpx_reduce_i64(const long long* __restrict__ vals,
+                                          const unsigned char* __restrict__ keep,
+                                          long long n,
+                                          long long* __restrict__ partials) {
+    __shared__ long long sh[TPX_RED_THREADS];
+    long long stride = (long long)gridDim.x * blockDim.x;
+    long long acc = 0;
+    for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += stride)
+        if (keep[i]) acc += vals[i];
+    sh[threadIdx.x] = acc;
+    __syncthreads();
+    for (int o = TPX_RED_THREADS / 2; o; o >>= 1) {
+        if ((int)threadIdx.x < o) sh[threadIdx.x] += sh[threadIdx.x + o];
+        __syncthreads();
+    }
+    if (threadIdx.x == 0) partials[blockIdx.x] = sh[0];
+}
+
+extern "C" __global__ void tpx_reduce_i64_final(const long long* __restrict__ partials,
+                                                long long n,
+                                                long long* __restrict__ out) {
+    if (blockIdx.x == 0 && threadIdx.x == 0) {
+        long long s = 0;
+        for (long long i = 0; i < n; ++i) s += partials[i];
+        *out = s;
+    }
+}
 
 // exclusive-scan building block: per-block scan of ITEMS_PER_BLOCK i64 items.
 // grid-wide composition happens host-side (scan block sums, then add pass).

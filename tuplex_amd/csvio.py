@@ -49,7 +49,7 @@ def split_rows(data: bytes) -> List[bytes]:
     return rows
 
 
-def split_cells(line: bytes):
+def split_cells(line: bytes, delim: bytes = b","):
     """Cell split matching tpx_csv_next_cell: returns (cells, flags) where each
     cell is bytes (quoted cells: inner bytes, '""' NOT unescaped) and flags has
     bit1=escaped, bit2=bad structure."""
@@ -81,9 +81,9 @@ def split_cells(line: bytes):
             if esc:
                 flags |= 2
             q += 1
-            if q < end and line[q:q + 1] != b",":
+            if q < end and line[q:q + 1] != delim:
                 flags |= 4
-            while q < end and line[q:q + 1] != b",":
+            while q < end and line[q:q + 1] != delim:
                 q += 1
             if q < end:
                 more = True
@@ -91,7 +91,7 @@ def split_cells(line: bytes):
             p = q
         else:
             q = p
-            while q < end and line[q:q + 1] != b",":
+            while q < end and line[q:q + 1] != delim:
                 q += 1
             cells.append(line[p:q])
             if q < end:
@@ -156,10 +156,23 @@ def _try_bool(s: str):
     return s.strip(_PYWS).lower() in _BOOL_STRS
 
 
+def sniff_delimiter(sample: bytes) -> bytes:
+    """Most frequent candidate separator in the first row, quotes respected
+    (ContextOptions.cc csv.separators [',', ';', '|', '\\t'])."""
+    first = split_rows(sample)[0]
+    best, bestn = b",", -1
+    for cand in (b",", b";", b"|", b"\t"):
+        n = len(split_cells(first, cand)[0])
+        if n > bestn:
+            best, bestn = cand, n
+    return best
+
+
 def sniff(sample: bytes, null_values: List[str], threshold: float,
-          header: Optional[bool], columns: Optional[List[str]]):
+          header: Optional[bool], columns: Optional[List[str]],
+          delim: bytes = b","):
     """Returns (has_header, column_names, column_types)."""
-    rows = [split_cells(r)[0] for r in split_rows(sample)]
+    rows = [split_cells(r, delim)[0] for r in split_rows(sample)]
     rows = [r for r in rows if r]
     if not rows:
         raise ValueError("empty csv sample")
@@ -225,7 +238,8 @@ def sniff(sample: bytes, null_values: List[str], threshold: float,
 
 # ---- replay of exception rows ----------------------------------------------------
 
-def replay_csv_row(raw_line: bytes, col_types, null_values, logical_ops, columns):
+def replay_csv_row(raw_line: bytes, col_types, null_values, logical_ops, columns,
+                   delim: str = ","):
     """Interpreter replay of a diverted CSV row (BADPARSE semantics,
     ResolveTask.cc:459 interpreter path with parse_cells). Full RFC-4180 parse
     (with unescaping) then CPython-typed conversion; structure/convert failures
@@ -233,7 +247,7 @@ def replay_csv_row(raw_line: bytes, col_types, null_values, logical_ops, columns
     text = raw_line.decode("utf-8", "replace")
     text = text.rstrip("\n").rstrip("\r")
     try:
-        cells = next(_pycsv.reader(io.StringIO(text)))
+        cells = next(_pycsv.reader(io.StringIO(text), delimiter=delim))
     except (StopIteration, _pycsv.Error):
         return ("exc", _BadParse("unparseable line"))
     if len(cells) != len(col_types):
@@ -301,9 +315,10 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
     nl = sample.rfind(b"\n")
     if nl >= 0:
         sample = sample[:nl + 1]
+    delim = (src.delimiter.encode() if src.delimiter else sniff_delimiter(sample))
     has_header, names, col_types = sniff(sample, src.null_values,
                                          opts.normalcase_threshold,
-                                         src.header, src.columns)
+                                         src.header, src.columns, delim)
 
     # assemble the GPU input: concat files, strip header lines, ensure trailing \n
     parts = []
@@ -321,7 +336,8 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
 
     if not sp.compilable:
         return _run_csv_fallback(out, data, col_types, src.null_values,
-                                 logical_ops, names, sink, sp.why_not_compilable)
+                                 logical_ops, names, sink,
+                                 sp.why_not_compilable, delim.decode())
 
     glib = GpuLib.get()
     if glib.device_count() == 0:
@@ -330,12 +346,14 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
     glib.lib.tpx_set_device(int(opts.get("tuplex.gpu.device", "0")))
 
     try:
-        csrc, desc = codegen.generate_stage(sp, source="csv", sink=sink_kind,
-                                            csv_info={"null_values":
-                                                      src.null_values})
+        csrc, desc = codegen.generate_stage(
+            sp, source="csv", sink=sink_kind,
+            csv_info={"null_values": src.null_values,
+                      "delimiter": delim.decode()})
     except codegen.CodegenError as e:
         return _run_csv_fallback(out, data, col_types, src.null_values,
-                                 logical_ops, names, sink, str(e))
+                                 logical_ops, names, sink, str(e),
+                                 delim.decode())
     stage = glib.compile_stage(csrc, desc)
 
     buf = (ctypes.c_uint8 * len(data)).from_buffer_copy(data)
@@ -361,7 +379,7 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
                 payload = eb[pos + 32:pos + 32 + size]
                 pos += 32 + size
                 r = replay_csv_row(payload, col_types, src.null_values,
-                                   logical_ops, names)
+                                   logical_ops, names, delim.decode())
                 if r[0] == "row":
                     replayed[row] = r[1]
                 elif r[0] == "exc":
@@ -370,14 +388,22 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
 
         if sink_kind == "mem":
             out_bytes = ctypes.string_at(res.out_data, res.out_size)
-            rows = rowfmt.deserialize_partition(out_bytes, T.tup(sp.output_types))
+            rows = rowfmt.deserialize_partition(out_bytes,
+                                                T.tup(sp.gpu_output_types))
+            agg = next((op for op in logical_ops if op[0] == "aggregate"), None)
+            if sp.agg_expr is not None:
+                _, combine_fn, agg_fn, initial = agg
+                acc = initial + rows[0][0]
+                for i in sorted(replayed):
+                    acc = agg_fn(acc, replayed[i])
+                out.rows = [acc]
+                return out
             idxs = [res.out_row_indices[i] for i in range(res.out_num_rows)]
             merged = {}
             for row, i in zip(rows, idxs):
                 merged[i] = row[0] if len(row) == 1 else row
             merged.update(replayed)
             out.rows = [merged[i] for i in sorted(merged)]
-            agg = next((op for op in logical_ops if op[0] == "aggregate"), None)
             if agg is not None:
                 _, combine_fn, agg_fn, initial = agg
                 a = initial
@@ -400,14 +426,14 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
 
 
 def _run_csv_fallback(out, data, col_types, null_values, logical_ops, names,
-                      sink, why):
+                      sink, why, delim=","):
     """Whole-stage interpreter fallback (non-compilable UDF) — the reference's
     fallback mode. Still semantically exact; slow by design."""
     out.mode = "fallback"
     out.fallback_reason = why
     rows_out = {}
     for i, line in enumerate(split_rows(data)):
-        r = replay_csv_row(line, col_types, null_values, logical_ops, names)
+        r = replay_csv_row(line, col_types, null_values, logical_ops, names, delim)
         if r[0] == "row":
             rows_out[i] = r[1]
         elif r[0] == "exc":

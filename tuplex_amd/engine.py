@@ -163,7 +163,7 @@ def execute_stage_mem(sp: plan.StageProgram, norm_rows: List[tuple],
     try:
         out = ExecResult()
         out_bytes = ctypes.string_at(res.out_data, res.out_size)
-        out_row_type = T.tup(sp.output_types)
+        out_row_type = T.tup(sp.gpu_output_types)
         out.rows = rowfmt.deserialize_partition(out_bytes, out_row_type)
         out.row_indices = [res.out_row_indices[i] for i in range(res.out_num_rows)] \
             if res.out_num_rows else []
@@ -206,6 +206,26 @@ def run_collect(data: List[Any], logical_ops: List[tuple],
 
     input_types = list(T.tuple_params(row_maj))
     sp = plan.build_stage(input_types, columns, logical_ops)
+
+    # GPU-reducible aggregate: the stage returns one partial; replayed rows are
+    # folded on top with the user's agg fn (LocalBackend.cc:1180-1207 combine)
+    if (sp.compilable and norm_rows and sp.agg_expr is not None):
+        er = execute_stage_mem(sp, norm_rows)
+        out.mode = "gpu"
+        out.metrics = er.metrics
+        opid_a, combine_fn, agg_fn, initial = sp.aggregate
+        acc = initial + er.rows[0][0]
+        replay = [(norm_idx[r], data[norm_idx[r]]) for (r, _, _) in er.exceptions]
+        replay += fallback
+        for i, v in sorted(replay):
+            r = resolve.replay_row(v, logical_ops, columns, scalar_input)
+            if r[0] == "row":
+                acc = agg_fn(acc, r[1])
+            elif r[0] == "exc":
+                name = type(r[1]).__name__
+                out.exception_counts[name] = out.exception_counts.get(name, 0) + 1
+        out.rows = [acc]
+        return out
 
     results = {}  # original index -> row value
     if sp.compilable and norm_rows:

@@ -41,8 +41,21 @@ class StageProgram:
         self.input_columns = list(input_columns) if input_columns else None
         self.ops: List[StageOp] = []
         self.aggregate = None  # (opid, combine_fn, agg_fn, initial)
+        # GPU-reducible aggregate (agg fn of shape `lambda a, x: a + expr(x)`,
+        # the Q6/count/sum pattern — AggregateFunctions.cc fold): per-row expr TIR
+        # + accumulator type. The device computes a deterministic fixed-order
+        # partial sum; host folds replayed rows and the initial value on top.
+        self.agg_expr = None
+        self.agg_type = None
         self.compilable = True
         self.why_not_compilable = None
+
+    @property
+    def gpu_output_types(self):
+        """Types of the partition the GPU returns (agg stages: 1 column)."""
+        if self.agg_expr is not None:
+            return [self.agg_type]
+        return self.output_types
 
     @property
     def output_types(self):
@@ -107,6 +120,7 @@ def build_stage(input_types, input_columns, logical_ops) -> StageProgram:
             continue
         if kind == "aggregate":
             sp.aggregate = (opid, entry[1], entry[2], entry[3])
+            _compile_aggregate(sp, opid, entry[2], entry[3], cur_types, cur_cols)
             continue
 
         op = StageOp(kind, opid)
@@ -180,6 +194,49 @@ def build_stage(input_types, input_columns, logical_ops) -> StageProgram:
         op.out_columns = list(cur_cols) if cur_cols else None
         sp.ops.append(op)
     return sp
+
+
+def _compile_aggregate(sp, opid, agg_fn, initial, cur_types, cur_cols):
+    """Detect the linear-fold shape `lambda a, x: a + expr(x)` and compile expr.
+    The reference emits agg_init/agg_combine functors (StageBuilder.cc:886-911);
+    on MI355X the per-row expr feeds a deterministic device reduction and the
+    final combine happens host-side (and over RCCL across GPUs)."""
+    if not isinstance(initial, (int, float)) or isinstance(initial, bool):
+        return
+    acc_t = T.F64 if isinstance(initial, float) else T.I64
+    try:
+        from .udf.compile import compile_agg_udf
+        node = compile_agg_udf(agg_fn, acc_t, list(cur_types),
+                               list(cur_cols) if cur_cols else None)
+    except UDFCompileError:
+        return
+    if node["op"] != "add":
+        return
+    lhs, rhs = node["args"]
+    if not (lhs["op"] == "input" and lhs["i"] == 0):
+        return
+
+    # rhs must not reference the accumulator; shift input indices down by one
+    def shift(n):
+        if n["op"] == "input":
+            if n["i"] == 0:
+                raise UDFCompileError("acc used beyond a + expr")
+            return dict(n, i=n["i"] - 1, args=[])
+        out = dict(n)
+        out["args"] = [shift(a) for a in n["args"]]
+        return out
+
+    try:
+        expr = shift(rhs)
+    except UDFCompileError:
+        return
+    if node["t"] not in (T.I64, T.F64) or expr["t"] not in (T.I64, T.F64, T.BOOL):
+        return
+    sp.agg_expr = expr if expr["t"] == node["t"] else \
+        {"op": "float_num" if node["t"] == T.F64 else "int_i64",
+         "t": node["t"], "args": [expr]}
+    sp.agg_type = node["t"]
+    sp.agg_opid = opid
 
 
 def _compile_into(sp, op, cur_types, cur_cols):

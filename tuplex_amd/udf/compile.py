@@ -81,10 +81,13 @@ class _Ret:
 
 
 class _Compiler:
-    def __init__(self, fn, input_types: List, columns: Optional[List[str]]):
+    def __init__(self, fn, input_types: List, columns: Optional[List[str]],
+                 acc_mode: bool = False):
         self.fn = fn
         self.input_types = input_types
         self.columns = columns
+        self.acc_mode = acc_mode   # aggregate fn: arg0 = accumulator (input 0),
+        self.row_base = 0          # arg1 = row over inputs row_base..
         self.globals = dict(getattr(fn, "__globals__", {}))
         closure = getattr(fn, "__closure__", None)
         if closure:
@@ -101,7 +104,15 @@ class _Compiler:
         if node.args.vararg or node.args.kwonlyargs or node.args.kwarg:
             raise UDFCompileError("varargs not supported")
         env: Dict[str, dict] = {}
-        if len(args) == 1:
+        if self.acc_mode:
+            if len(args) != 2:
+                raise UDFCompileError("aggregate UDF must take (acc, row)")
+            env[args[0].arg] = tir.inp(0, self.input_types[0])
+            self.row_param = args[1].arg
+            self.row_base = 1
+            if len(self.input_types) == 2:
+                env[args[1].arg] = tir.inp(1, self.input_types[1])
+        elif len(args) == 1:
             p = args[0].arg
             self.row_param = p  # x['col'] / x[i] row access always allowed
             if len(self.input_types) == 1:
@@ -305,17 +316,25 @@ class _Compiler:
                 key = e.slice.value
                 if not self.columns or key not in self.columns:
                     raise UDFCompileError("unknown column %r" % key)
-                i = self.columns.index(key)
+                i = self.row_base + self.columns.index(key)
                 return tir.inp(i, self.input_types[i])
             if not scalar_is_str:
                 if (isinstance(e.slice, ast.Constant)
                         and isinstance(e.slice.value, int)):
-                    key = e.slice.value
+                    key = self.row_base + e.slice.value
                     if not (0 <= key < len(self.input_types)):
                         raise UDFCompileError("column index out of range")
                     return tir.inp(key, self.input_types[key])
                 raise UDFCompileError("row subscript must be a constant")
             # else: fall through to string getitem/slice on the scalar binding
+        # s.split(sep)[i] -> fused splitget (logs pipelines)
+        if (isinstance(e.value, ast.Call) and isinstance(e.value.func, ast.Attribute)
+                and e.value.func.attr == "split" and len(e.value.args) == 1
+                and not isinstance(e.slice, ast.Slice)):
+            s = self._expr(e.value.func.value, env)
+            sep = self._expr(e.value.args[0], env)
+            idx = self._expr(e.slice, env)
+            return tir.splitget(s, sep, idx)
         obj = self._expr(e.value, env)
         if isinstance(e.slice, ast.Slice):
             if e.slice.step is not None:
@@ -332,3 +351,11 @@ def compile_udf(fn: Callable, input_types: List, columns: Optional[List[str]] = 
     Returns the root node (output type in node['t']). Raises UDFCompileError for
     anything outside the vocabulary -> caller falls back to interpreter mode."""
     return _Compiler(fn, input_types, columns).compile()
+
+
+def compile_agg_udf(fn: Callable, acc_type, row_types: List,
+                    columns: Optional[List[str]] = None):
+    """Compile an aggregate UDF `lambda a, x: ...` — input 0 is the accumulator,
+    inputs 1.. are the row columns (AggregateFunctions.cc functor shape)."""
+    return _Compiler(fn, [acc_type] + list(row_types), columns,
+                     acc_mode=True).compile()

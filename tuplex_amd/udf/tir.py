@@ -188,6 +188,15 @@ def strslice(s, lo, hi):
     return mk("slice", T.STR, args)
 
 
+def splitget(s, sep, idx):
+    """s.split(sep)[i] — fused (the logs-pipeline idiom). Part indexing is
+    byte-exact for UTF-8 (parts are substrings, not char indices)."""
+    s = _deopt_node(s)
+    if s["t"] != T.STR or sep["t"] != T.STR or idx["t"] != T.I64:
+        raise TirError("split needs (str, str)[int]")
+    return mk("splitget", T.STR, [s, sep, idx])
+
+
 def fmt(spec, arg):
     """'%05d' % x style formatting; spec like (width, zero_pad) for %d."""
     arg = _deopt_node(arg)
