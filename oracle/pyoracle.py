@@ -357,10 +357,17 @@ def run_pipeline(data: List[Any], ops: List[tuple], columns: Optional[List[str]]
         _, combine_fn, agg_fn, initial = agg
         a = initial
         for v in out:
-            a = agg_fn(a, v)
+            a = agg_fn(a, _agg_row(v, columns))
         out = [a]
 
     return {"output": out, "exception_counts": exc_counts}
+
+
+def _agg_row(v, columns):
+    """agg fn's row arg follows the same dict convention as other UDFs."""
+    if columns and isinstance(v, tuple):
+        return dict(zip(columns, v))
+    return v
 
 
 def process_row(value, row_ops, columns, fast):

@@ -261,7 +261,7 @@ def _replay_row(raw_line: bytes, col_types, null_values, row_ops, columns,
         return ("exc", _BadParse("unparseable"))
     if len(cells) != len(col_types):
         return ("exc", _BadParse("cellcount"))
-    vals = []
+    vals = []  # noqa: replay path
     for c, t in zip(cells, col_types):
         base = _deopt(t)
         if _is_opt(t) and c in null_values:
@@ -281,7 +281,8 @@ def _replay_row(raw_line: bytes, col_types, null_values, row_ops, columns,
                 vals.append(c)
         except ValueError as e:
             return ("exc", e)
-    return pyoracle.process_row(tuple(vals), row_ops, columns, fast=False)
+    val = vals[0] if len(col_types) == 1 else tuple(vals)
+    return pyoracle.process_row(val, row_ops, columns, fast=False)
 
 
 def run_csv_pipeline(csv_bytes: bytes, ops, columns=None, header=None,
@@ -322,7 +323,8 @@ def run_csv_pipeline(csv_bytes: bytes, ops, columns=None, header=None,
         cells, flags = split_cells(line, delim)
         pr = _fast_parse_row(cells, flags, col_types, null_values)
         if pr[0] == "row":
-            r = pyoracle.process_row(pr[1], row_ops, names, fast=True)
+            val = pr[1][0] if len(col_types) == 1 else pr[1]  # 1-col = scalar row
+            r = pyoracle.process_row(val, row_ops, names, fast=True)
         else:
             r = _replay_row(line, col_types, null_values, row_ops, names,
                             delim.decode())
@@ -341,7 +343,7 @@ def run_csv_pipeline(csv_bytes: bytes, ops, columns=None, header=None,
         _, combine_fn, agg_fn, initial = agg
         a = initial
         for v in rows:
-            a = agg_fn(a, v)
+            a = agg_fn(a, pyoracle._agg_row(v, names))
         rows = [a]
     if sink == "collect":
         result["output"] = rows

@@ -276,6 +276,9 @@ def replay_csv_row(raw_line: bytes, col_types, null_values, logical_ops, columns
                 vals.append(c)
         except ValueError as e:
             return ("exc", e)
+    if len(col_types) == 1:  # single-column datasets carry scalar rows
+        return _resolve.replay_row(vals[0], logical_ops, columns,
+                                   scalar_input=True)
     return _resolve.replay_row(tuple(vals), logical_ops, columns,
                                scalar_input=False)
 
@@ -392,10 +395,11 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
                                                 T.tup(sp.gpu_output_types))
             agg = next((op for op in logical_ops if op[0] == "aggregate"), None)
             if sp.agg_expr is not None:
+                from .engine import _agg_row
                 _, combine_fn, agg_fn, initial = agg
                 acc = initial + rows[0][0]
                 for i in sorted(replayed):
-                    acc = agg_fn(acc, replayed[i])
+                    acc = agg_fn(acc, _agg_row(replayed[i], sp.output_columns))
                 out.rows = [acc]
                 return out
             idxs = [res.out_row_indices[i] for i in range(res.out_num_rows)]
@@ -405,10 +409,11 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
             merged.update(replayed)
             out.rows = [merged[i] for i in sorted(merged)]
             if agg is not None:
+                from .engine import _agg_row
                 _, combine_fn, agg_fn, initial = agg
                 a = initial
                 for v in out.rows:
-                    a = agg_fn(a, v)
+                    a = agg_fn(a, _agg_row(v, sp.output_columns))
                 out.rows = [a]
         else:
             text = ctypes.string_at(res.out_data, res.out_size) if res.out_size \
@@ -442,10 +447,11 @@ def _run_csv_fallback(out, data, col_types, null_values, logical_ops, names,
     out.rows = [rows_out[i] for i in sorted(rows_out)]
     agg = next((op for op in logical_ops if op[0] == "aggregate"), None)
     if agg is not None:
+        from .engine import _agg_row
         _, combine_fn, agg_fn, initial = agg
         a = initial
         for v in out.rows:
-            a = agg_fn(a, v)
+            a = agg_fn(a, _agg_row(v, names))
         out.rows = [a]
     if sink is not None and sink[0] == "csv":
         raise NotImplementedError("tocsv via fallback: next slice")

@@ -217,10 +217,11 @@ def run_collect(data: List[Any], logical_ops: List[tuple],
         acc = initial + er.rows[0][0]
         replay = [(norm_idx[r], data[norm_idx[r]]) for (r, _, _) in er.exceptions]
         replay += fallback
+        agg_cols = sp.output_columns
         for i, v in sorted(replay):
             r = resolve.replay_row(v, logical_ops, columns, scalar_input)
             if r[0] == "row":
-                acc = agg_fn(acc, r[1])
+                acc = agg_fn(acc, _agg_row(r[1], agg_cols))
             elif r[0] == "exc":
                 name = type(r[1]).__name__
                 out.exception_counts[name] = out.exception_counts.get(name, 0) + 1
@@ -260,10 +261,17 @@ def run_collect(data: List[Any], logical_ops: List[tuple],
         _, combine_fn, agg_fn, initial = agg
         a = initial
         for v in merged:
-            a = agg_fn(a, v)
+            a = agg_fn(a, _agg_row(v, sp.output_columns))
         merged = [a]
     out.rows = merged
     return out
+
+
+def _agg_row(v, columns):
+    """agg fn row arg follows the same dict convention as other UDFs."""
+    if columns and isinstance(v, tuple):
+        return dict(zip(columns, v))
+    return v
 
 
 def _unwrap_row(row: tuple):
