@@ -88,30 +88,42 @@ def cpu_baseline_leg(body, seconds):
         return None
 
 
-def parse_pmc_summary(path):
-    """Optional: rocprofv3 PMC csv with FETCH_SIZE/WRITE_SIZE for tpx_stage_main.
+def parse_pmc_dbs(paths):
+    """HBM traffic per tpx_stage_main launch from rocprofv3 rocpd sqlite dbs
+    (separate FETCH_SIZE and WRITE_SIZE passes — they don't fit one TCC pass).
     gfx950: FETCH_SIZE reports half the bytes of wide coalesced reads
-    (MI355X_MICROARCH.md §HBM) -> traffic = (2*FETCH + WRITE) * 1024 per launch."""
+    (MI355X_MICROARCH.md §HBM) -> traffic = (2*FETCH + WRITE) * 1024 per launch.
+    Valid only when the PMC runs used the same workload config as this bench."""
+    import glob as _glob
+    import sqlite3
+    counters = {}
     try:
-        import csv as _csv
-        fetch, write, disp = 0.0, 0.0, 0
-        with open(path) as f:
-            for row in _csv.DictReader(f):
-                name = row.get("Kernel_Name", row.get("Name", ""))
-                if "tpx_stage_main" not in name:
+        for p in paths:
+            for dbp in _glob.glob(os.path.join(p, "**", "*_results.db"),
+                                  recursive=True) or [p]:
+                db = sqlite3.connect(dbp)
+                cur = db.cursor()
+                tabs = [r[0] for r in cur.execute(
+                    "SELECT name FROM sqlite_master WHERE type='table'")]
+                kd = [t for t in tabs if t.startswith("rocpd_kernel_dispatch")]
+                if not kd:
                     continue
-                cname = row.get("Counter_Name", "")
-                v = float(row.get("Counter_Value", row.get("Value", 0)))
-                if cname == "FETCH_SIZE":
-                    fetch += v
-                    disp += 1
-                elif cname == "WRITE_SIZE":
-                    write += v
-        if disp == 0:
+                u = kd[0].replace("rocpd_kernel_dispatch_", "")
+                q = ("SELECT pi.name, AVG(pe.value) "
+                     "FROM rocpd_pmc_event_%s pe "
+                     "JOIN rocpd_kernel_dispatch_%s k ON pe.event_id=k.event_id "
+                     "JOIN rocpd_info_kernel_symbol_%s ks ON k.kernel_id=ks.id "
+                     "JOIN rocpd_info_pmc_%s pi ON pe.pmc_id=pi.id "
+                     "WHERE ks.kernel_name LIKE 'tpx_stage_main%%' "
+                     "GROUP BY pi.name" % (u, u, u, u))
+                for name, avg in cur.execute(q):
+                    counters[name] = avg
+        if "FETCH_SIZE" not in counters and "WRITE_SIZE" not in counters:
             return None
-        return (2.0 * fetch + write) * 1024.0 / disp  # bytes per launch (avg)
+        return (2.0 * counters.get("FETCH_SIZE", 0.0)
+                + counters.get("WRITE_SIZE", 0.0)) * 1024.0
     except Exception as e:  # noqa: BLE001
-        log("pmc summary parse failed:", e)
+        log("pmc db parse failed:", e)
         return None
 
 
@@ -124,7 +136,9 @@ def main():
     ap.add_argument("--base-rows", type=int, default=150000)
     ap.add_argument("--dirty", type=float, default=0.0)
     ap.add_argument("--cpu-seconds", type=float, default=10.0)
-    ap.add_argument("--pmc-summary", type=str, default=None)
+    ap.add_argument("--pmc-db", type=str, nargs="*", default=None,
+                    help="rocprofv3 output dirs (FETCH_SIZE / WRITE_SIZE passes "
+                         "at THIS workload config) for roofline.traffic")
     ap.add_argument("--no-cpu-baseline", action="store_true")
     args = ap.parse_args()
 
@@ -227,7 +241,7 @@ def main():
     algo_bytes = last["bytes_in"] + last["bytes_out"]
     achieved = (sum(k["bytes_in"] + k["bytes_out"] for k in kstats) / t_kernel_s
                 if t_kernel_s > 0 else 0.0)
-    traffic = parse_pmc_summary(args.pmc_summary) if args.pmc_summary else None
+    traffic = parse_pmc_dbs(args.pmc_db) if args.pmc_db else None
     roofline = {
         "bound": "hbm",
         "achieved": achieved / 1e9,
