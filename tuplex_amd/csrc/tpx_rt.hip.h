@@ -79,10 +79,70 @@ __device__ __forceinline__ char* tpx_alloc(TpxHeap& h, long long n) {
 
 __device__ __forceinline__ bool tpx_is_ascii_byte(unsigned char c) { return c < 0x80; }
 
-// scan for non-ASCII; true if all ASCII
+// ---- SWAR 8-byte scanning (the scanning loops dominate this HBM/LDS-bound
+// path; byte-at-a-time dependent loads are issue/latency-bound) -----------------
+
+#define TPX_SWAR_ONE 0x0101010101010101ULL
+#define TPX_SWAR_HIGH 0x8080808080808080ULL
+
+// zero-byte detector: nonzero bits at 0x80 positions of bytes equal to zero
+__device__ __forceinline__ unsigned long long tpx_swar_zero(unsigned long long x) {
+    return (x - TPX_SWAR_ONE) & ~x & TPX_SWAR_HIGH;
+}
+
+// first index of byte c in p[0..n), or -1. Aligned 8-byte windows (head/tail
+// byte loops), exact match of the byte-loop semantics.
+__device__ __forceinline__ long long tpx_memchr(const char* p, long long n,
+                                                char c) {
+    unsigned long long pat = TPX_SWAR_ONE * (unsigned char)c;
+    long long i = 0;
+    while (i < n && (((unsigned long long)(p + i)) & 7)) {
+        if (p[i] == c) return i;
+        ++i;
+    }
+    for (; i + 8 <= n; i += 8) {
+        unsigned long long v = *(const unsigned long long*)(p + i);
+        unsigned long long hit = tpx_swar_zero(v ^ pat);
+        if (hit) return i + (__ffsll((long long)hit) - 1) / 8;
+    }
+    for (; i < n; ++i)
+        if (p[i] == c) return i;
+    return -1;
+}
+
+// first index where p[i]==c1 or p[i]==c2, or -1
+__device__ __forceinline__ long long tpx_memchr2(const char* p, long long n,
+                                                 char c1, char c2) {
+    unsigned long long pat1 = TPX_SWAR_ONE * (unsigned char)c1;
+    unsigned long long pat2 = TPX_SWAR_ONE * (unsigned char)c2;
+    long long i = 0;
+    while (i < n && (((unsigned long long)(p + i)) & 7)) {
+        if (p[i] == c1 || p[i] == c2) return i;
+        ++i;
+    }
+    for (; i + 8 <= n; i += 8) {
+        unsigned long long v = *(const unsigned long long*)(p + i);
+        unsigned long long hit = tpx_swar_zero(v ^ pat1) | tpx_swar_zero(v ^ pat2);
+        if (hit) return i + (__ffsll((long long)hit) - 1) / 8;
+    }
+    for (; i < n; ++i)
+        if (p[i] == c1 || p[i] == c2) return i;
+    return -1;
+}
+
+// scan for non-ASCII; true if all ASCII (SWAR high-bit sweep)
 __device__ __forceinline__ bool tpx_ascii(const tstr s) {
-    for (long long i = 0; i < s.n; ++i)
-        if ((unsigned char)s.p[i] >= 0x80) return false;
+    const char* p = s.p;
+    long long n = s.n;
+    long long i = 0;
+    while (i < n && (((unsigned long long)(p + i)) & 7)) {
+        if ((unsigned char)p[i] >= 0x80) return false;
+        ++i;
+    }
+    for (; i + 8 <= n; i += 8)
+        if (*(const unsigned long long*)(p + i) & TPX_SWAR_HIGH) return false;
+    for (; i < n; ++i)
+        if ((unsigned char)p[i] >= 0x80) return false;
     return true;
 }
 
@@ -143,11 +203,17 @@ __device__ __forceinline__ long long tpx_len(const tstr s, int* ec) {
 __device__ __forceinline__ long long tpx_find(const tstr s, const tstr needle, int* ec) {
     if (!tpx_ascii(s)) { *ec = EC_NCV; return 0; }
     if (needle.n == 0) return 0;
-    for (long long i = 0; i + needle.n <= s.n; ++i) {
+    long long i = 0;
+    long long last = s.n - needle.n;
+    while (i <= last) {
+        long long k = tpx_memchr(s.p + i, last - i + 1, needle.p[0]);
+        if (k < 0) return -1;
+        i += k;
         bool m = true;
-        for (long long j = 0; j < needle.n; ++j)
+        for (long long j = 1; j < needle.n; ++j)
             if (s.p[i + j] != needle.p[j]) { m = false; break; }
         if (m) return i;
+        ++i;
     }
     return -1;
 }
@@ -166,11 +232,17 @@ __device__ __forceinline__ long long tpx_rfind(const tstr s, const tstr needle, 
 
 __device__ __forceinline__ bool tpx_contains(const tstr s, const tstr needle) {
     if (needle.n == 0) return true;
-    for (long long i = 0; i + needle.n <= s.n; ++i) {
+    long long i = 0;
+    long long last = s.n - needle.n;
+    while (i <= last) {
+        long long k = tpx_memchr(s.p + i, last - i + 1, needle.p[0]);
+        if (k < 0) return false;
+        i += k;
         bool m = true;
-        for (long long j = 0; j < needle.n; ++j)
+        for (long long j = 1; j < needle.n; ++j)
             if (s.p[i + j] != needle.p[j]) { m = false; break; }
         if (m) return true;
+        ++i;
     }
     return false;
 }
@@ -672,13 +744,30 @@ extern "C" __global__ void tpx_csv_emit_rows(const char* __restrict__ data,
         long long start_par = (qscan[c] + pref_q) & 1;
         long long sel = start_par ? l1 : l0;   // valid newlines in this lane
         long long lane_base = base[c] + tpx_wave_exscan(sel);
-        // second pass: emit offsets for valid newlines
+        // second pass: emit offsets for valid newlines (vector loads when full)
         long long qq = 0, idx = lane_base;
-        for (long long i = a; i < b; ++i) {
-            char ch = data[i];
-            if (ch == '"') ++qq;
-            else if (ch == '\n' && (((qq + start_par) & 1) == 0))
-                row_offs[1 + idx++] = i + 1;
+        if (b - a == TPX_CSV_LANE_BYTES && ((a & 15) == 0)) {
+            #pragma unroll
+            for (int v = 0; v < 4; ++v) {
+                uint4 w = *(const uint4*)(data + a + v * 16);
+                unsigned words[4] = {w.x, w.y, w.z, w.w};
+                #pragma unroll
+                for (int k = 0; k < 4; ++k)
+                    #pragma unroll
+                    for (int s2 = 0; s2 < 4; ++s2) {
+                        unsigned ch = (words[k] >> (8 * s2)) & 0xFF;
+                        if (ch == '"') ++qq;
+                        else if (ch == '\n' && (((qq + start_par) & 1) == 0))
+                            row_offs[1 + idx++] = a + v * 16 + k * 4 + s2 + 1;
+                    }
+            }
+        } else {
+            for (long long i = a; i < b; ++i) {
+                char ch = data[i];
+                if (ch == '"') ++qq;
+                else if (ch == '\n' && (((qq + start_par) & 1) == 0))
+                    row_offs[1 + idx++] = i + 1;
+            }
         }
     }
 }
@@ -700,22 +789,23 @@ __device__ __forceinline__ const char* tpx_csv_next_cell(const char* p,
         const char* q = s;
         bool esc = false;
         while (q < end) {
-            if (*q == '"') {
-                if (q + 1 < end && q[1] == '"') { esc = true; q += 2; continue; }
-                break;
-            }
-            ++q;
+            long long k = tpx_memchr(q, end - q, '"');
+            if (k < 0) { q = end; break; }
+            q += k;
+            if (q + 1 < end && q[1] == '"') { esc = true; q += 2; continue; }
+            break;
         }
         if (q >= end) { c->p = p; c->n = end - p; c->flags = 4; return end; }
         c->p = s; c->n = q - s; c->flags = 1 | (esc ? 2 : 0);
         ++q;
         if (q < end && *q != delim) c->flags |= 4;
-        while (q < end && *q != delim) ++q;
+        long long kd = tpx_memchr(q, end - q, delim);
+        q = kd < 0 ? end : q + kd;
         if (q < end) { *more = true; ++q; }
         return q;
     }
-    const char* q = p;
-    while (q < end && *q != delim) ++q;
+    long long kd = tpx_memchr(p, end - p, delim);
+    const char* q = kd < 0 ? end : p + kd;
     c->p = p; c->n = q - p;
     if (q < end) { *more = true; ++q; }
     return q;
