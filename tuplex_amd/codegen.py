@@ -632,7 +632,7 @@ class StageCodegen:
         L.append("    long long* exc_buf, unsigned long long* exc_count,"
                  " unsigned long long exc_cap,")
         L.append("    void** outv) {")
-        L.append("  __shared__ char smem[4 * TPX_SPAN_CAP];  // 4 waves per 256-thread block")
+        L.append("  __shared__ char smem[2 * TPX_SPAN_CAP + 16];  // 2 waves per 128-thread block -> 5 blocks/CU")
         L.append("  TpxHeap heap{heap_base, heap_cursor, heap_cap, nullptr, nullptr};")
         L.append("  int lane = threadIdx.x & 63;")
         L.append("  int wid = threadIdx.x >> 6;")
@@ -915,8 +915,7 @@ class StageCodegen:
                          % (fixed_end - bitmap, 8 * k))
                 L.append("      %s = off | ((v%d.n + 1) << 32);" % (slot, k))
                 L.append("      char* d = (char*)(w + %d + var_off);" % (fixed_end + 8))
-                L.append("      for (long long j = 0; j < v%d.n; ++j) d[j] = v%d.p[j];"
-                         % (k, k))
+                L.append("      tpx_memcpy(d, v%d.p, v%d.n);" % (k, k))
                 L.append("      d[v%d.n] = 0;" % k)
                 L.append("      var_off += v%d.n + 1;" % k)
                 L.append("    }")

@@ -440,7 +440,7 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
                                                 (160ll << 20), 1 << 20);
     if (P.heap_cap < heap_cap) {
         if (P.heap) (void)hipFree(P.heap);
-        HIP_CHECK(hipMalloc(&P.heap, heap_cap));
+        HIP_CHECK(hipMalloc(&P.heap, heap_cap + 16));  // tpx_memcpy over-read pad
         P.heap_cap = heap_cap;
     }
     heap_cap = P.heap_cap;
@@ -450,14 +450,15 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
     unsigned long long exc_count = 0;
     for (int attempt = 0;; ++attempt) {
         HIP_CHECK(hipMemsetAsync(P.counters, 0, 16, stream));
-        // 2048 blocks x 256 = exactly 8 blocks/CU on 256 CUs (full occupancy);
-        // grid-stride covers the rest; bounds per-thread heap-chunk waste
-        unsigned grid = (unsigned)std::min<long long>((n + 255) / 256, 2048);
+        // main runs 128-thread blocks (2 waves x 16 KiB LDS staging = 32 KiB
+        // -> 5 blocks/CU = 10 waves/CU); each wave owns 64 rows
+        long long waves = (n + 63) / 64;
+        unsigned grid = (unsigned)std::min<long long>((waves + 1) / 2, 4096);
         void* args[] = {&d_in, &d_offs, &n, &row0, &P.heap, &d_heap_cursor,
                         &heap_cap, &d_keep, &d_keep01, &d_sizes, &P.exc,
                         &d_exc_count, &exc_cap, &d_outv};
         hipEventRecord(evm0, stream);
-        if (launch(st->k_main, grid, 256, stream, args)) return -1;
+        if (launch(st->k_main, grid, 128, stream, args)) return -1;
         hipEventRecord(evm1, stream);
         unsigned long long counters[2] = {0, 0};
         HIP_CHECK(hipMemcpyAsync(counters, P.counters, 16, hipMemcpyDeviceToHost,
@@ -797,7 +798,7 @@ extern "C" int64_t tpx_stage_execute_csv(tpx_stage* st, const uint8_t* csv_bytes
 
 extern "C" uint64_t tpx_dev_alloc(int64_t size) {
     void* p = nullptr;
-    if (hipMalloc(&p, (size_t)size) != hipSuccess) return 0;
+    if (hipMalloc(&p, (size_t)size + 16) != hipSuccess) return 0;  // memcpy pad
     return (uint64_t)(uintptr_t)p;
 }
 

@@ -146,6 +146,32 @@ __device__ __forceinline__ bool tpx_ascii(const tstr s) {
     return true;
 }
 
+// vectorized copy: align dest, funnel-shift unaligned source words. May read up
+// to 7 bytes past s+n (every device buffer carries >=16 B tail padding).
+__device__ __forceinline__ void tpx_memcpy(char* d, const char* s, long long n) {
+    if (n < 16) {
+        for (long long i = 0; i < n; ++i) d[i] = s[i];
+        return;
+    }
+    long long i = 0;
+    while (((unsigned long long)(d + i)) & 7) { d[i] = s[i]; ++i; }
+    long long k = ((unsigned long long)(s + i)) & 7;
+    const unsigned long long* sw =
+        (const unsigned long long*)(s + i - k);
+    if (k == 0) {
+        for (; i + 8 <= n; i += 8) *(unsigned long long*)(d + i) = *sw++;
+    } else {
+        unsigned long long prev = *sw++;
+        int sh = 8 * (int)k;
+        for (; i + 8 <= n; i += 8) {
+            unsigned long long cur = *sw++;
+            *(unsigned long long*)(d + i) = (prev >> sh) | (cur << (64 - sh));
+            prev = cur;
+        }
+    }
+    for (; i < n; ++i) d[i] = s[i];
+}
+
 // translate an LDS-staged string view back to its global-memory address before it
 // escapes the kernel (columnar string cells must outlive the LDS staging window).
 // (p - lds_lo) + span_start is the byte offset in the input.
@@ -359,8 +385,8 @@ __device__ __forceinline__ tstr tpx_strip(const tstr s, int* ec) {
 __device__ __forceinline__ tstr tpx_concat(TpxHeap& h, const tstr a, const tstr b, int* ec) {
     char* d = tpx_alloc(h, a.n + b.n);
     if (!d) { *ec = EC_MEMORYERROR; return tstr{a.p, 0}; }
-    for (long long i = 0; i < a.n; ++i) d[i] = a.p[i];
-    for (long long i = 0; i < b.n; ++i) d[a.n + i] = b.p[i];
+    tpx_memcpy(d, a.p, a.n);
+    tpx_memcpy(d + a.n, b.p, b.n);
     return tstr{d, a.n + b.n};
 }
 
@@ -618,7 +644,8 @@ __device__ __forceinline__ char* tpx_csv_cell_write(char* w, const tstr s) {
         }
         *w++ = '"';
     } else {
-        for (long long i = 0; i < s.n; ++i) *w++ = s.p[i];
+        tpx_memcpy(w, s.p, s.n);
+        w += s.n;
     }
     return w;
 }
