@@ -325,10 +325,13 @@ def run_pipeline(data: List[Any], ops: List[tuple], columns: Optional[List[str]]
 
     # aggregate stages handled as a fold at the end
     agg = None
+    aggby = None
     row_ops = []
     for op in ops:
         if op[0] == "aggregate":
             agg = op
+        elif op[0] == "aggregateByKey":
+            aggby = op
         else:
             row_ops.append(op)
 
@@ -359,8 +362,25 @@ def run_pipeline(data: List[Any], ops: List[tuple], columns: Optional[List[str]]
         for v in out:
             a = agg_fn(a, _agg_row(v, columns))
         out = [a]
+    elif aggby is not None:
+        out = aggregate_by_key(out, aggby, columns)
 
     return {"output": out, "exception_counts": exc_counts}
+
+
+def aggregate_by_key(rows, aggby, columns):
+    """hashmap-sink fold (hashmap.cc / createFinalHashmap LocalBackend.cc:2219).
+    Output order is parity-unpinned (multiset compare, test_aggregates.py:53)."""
+    _, combine_fn, agg_fn, initial, key_cols = aggby
+    key_idx = [columns.index(c) for c in key_cols]
+    table = {}
+    for v in rows:
+        row = v if isinstance(v, tuple) else (v,)
+        k = tuple(row[i] for i in key_idx)
+        acc = table.get(k, initial)
+        table[k] = agg_fn(acc, _agg_row(v, columns))
+    return [k + ((val,) if not isinstance(val, tuple) else val)
+            for k, val in table.items()]
 
 
 def _agg_row(v, columns):

@@ -305,10 +305,13 @@ def run_csv_pipeline(csv_bytes: bytes, ops, columns=None, header=None,
         data = data[p + 1:] if p >= 0 else b""
 
     agg = None
+    aggby = None
     row_ops = []
     for op in ops:
         if op[0] == "aggregate":
             agg = op
+        elif op[0] == "aggregateByKey":
+            aggby = op
         else:
             row_ops.append(op)
 
@@ -345,6 +348,8 @@ def run_csv_pipeline(csv_bytes: bytes, ops, columns=None, header=None,
         for v in rows:
             a = agg_fn(a, pyoracle._agg_row(v, names))
         rows = [a]
+    elif aggby is not None:
+        rows = pyoracle.aggregate_by_key(rows, aggby, names)
     if sink == "collect":
         result["output"] = rows
         return result

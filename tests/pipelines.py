@@ -63,6 +63,18 @@ def agg_sum(a, x):
     return a + x
 
 
+def aggby_comb(a, b):
+    return a + b
+
+
+def aggby_vol(a, x):
+    return a + x["volume"]
+
+
+def aggby_cnt(a, x):
+    return a + 1
+
+
 def lower_udf(x):
     return x.lower()
 
@@ -152,6 +164,21 @@ def long_probe(x):
 
 
 PIPELINES.append(("long_rows", _long_rows(), None, [("map", long_probe)]))
+
+
+def _by_key_rows(n=50000, seed=5):
+    rng = random.Random(seed)
+    return [(rng.randint(-3, 1000), float(rng.randint(-50, 50)) / 4.0)
+            for _ in range(n)]
+
+
+PIPELINES.append(("sum_by_key_small", [(0, 10.0), (1, 20.0), (0, -4.5)],
+                  ["id", "volume"],
+                  [("aggregateByKey", aggby_comb, aggby_vol, 0.0, ["id"])]))
+PIPELINES.append(("sum_by_key_large", _by_key_rows(), ["id", "volume"],
+                  [("aggregateByKey", aggby_comb, aggby_vol, 0.0, ["id"])]))
+PIPELINES.append(("count_by_key", _by_key_rows(8000, seed=11), ["id", "volume"],
+                  [("aggregateByKey", aggby_comb, aggby_cnt, 0, ["id"])]))
 
 
 def apply_ops(ds, ops):

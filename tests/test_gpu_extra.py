@@ -89,3 +89,26 @@ def test_long_rows_span_fallback():
     assert ds._last_outcome.mode == "gpu"
     ref = pyoracle.run_pipeline(data, [("map", probe)])
     assert got == ref["output"]
+
+
+def _sorted_rows(rows):
+    return sorted(rows)
+
+
+@pytest.mark.parametrize("name", ["sum_by_key_small", "sum_by_key_large",
+                                  "count_by_key"])
+def test_aggregate_by_key_gpu(name):
+    from tests.pipelines import PIPELINES
+    nm, data, columns, ops = [p for p in PIPELINES if p[0] == name][0]
+    ctx = tuplex_amd.Context()
+    from tests.pipelines import apply_ops as ap
+    ds = ap(ctx.parallelize(data, columns=columns), ops)
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    ref = pyoracle.run_pipeline(data, ops, columns=columns)
+    g, r = _sorted_rows(got), _sorted_rows(ref["output"])
+    assert len(g) == len(r)
+    for (gk, gv), (rk, rv) in zip(g, r):
+        assert gk == rk
+        assert abs(gv - rv) <= 1e-9 * max(1.0, abs(rv)), (gk, gv, rv)
+    assert ds.exception_counts == ref["exception_counts"]

@@ -134,6 +134,20 @@ def test_ref_int_matches_python_on_normal():
         assert pyoracle.ref_int(s) == int(s)
 
 
+def test_sum_by_key_golden():
+    """tuplex/python/tests/test_aggregates.py:40 test_sum_by_key literal
+    expectations (sorted compare — by-key order is parity-unpinned)."""
+    data = [(0, 10.0), (1, 20.0), (0, -4.5)]
+    r = pyoracle.run_pipeline(
+        data, [("aggregateByKey", lambda a, b: a + b,
+                lambda a, x: a + x["volume"], 0.0, ["id"])],
+        columns=["id", "volume"])
+    res = sorted(r["output"])
+    assert len(res) == 2
+    assert res[0][0] == 0 and res[1][0] == 1
+    assert abs(res[0][1] - 5.5) < 1e-9 and abs(res[1][1] - 20.0) < 1e-9
+
+
 def test_zillow_udfs_vs_cpython():
     """The reference pins zillow-UDF results by comparing the compiled path against a
     per-row CPython map (test/wrappers/WrapperTest.cc:468 extractPriceExample). Same

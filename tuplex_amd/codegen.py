@@ -537,9 +537,13 @@ class StageCodegen:
             else:
                 raise CodegenError("op %r not supported in codegen" % op.kind)
         if getattr(sp, "agg_expr", None) is not None:
-            # per-row aggregate expr (the fold's rhs); device reduce sums it
+            # per-row aggregate expr (the fold's rhs); device reduce sums it.
+            # by-key: emit (key, expr) — the hash-reduce kernel consumes both
             v, nv = self.emit_expr(em, sp.agg_expr, rc, sp.agg_opid)
-            rc = [(v, sp.agg_expr["t"], nv)]
+            if getattr(sp, "agg_key_idx", None) is not None:
+                rc = [rc[sp.agg_key_idx], (v, sp.agg_expr["t"], nv)]
+            else:
+                rc = [(v, sp.agg_expr["t"], nv)]
         return rc
 
     # ---- full source --------------------------------------------------------
@@ -547,7 +551,7 @@ class StageCodegen:
         """Returns (hip_source, stage_desc)."""
         sp = self.sp
         in_types = sp.input_types
-        out_types = [sp.agg_type] if getattr(sp, "agg_expr", None) is not None \
+        out_types = sp.gpu_output_types if getattr(sp, "agg_expr", None) is not None \
             else sp.output_types
 
         body = _Emitter()
@@ -985,7 +989,10 @@ class StageCodegen:
         lines = ["source=%s" % self.source, "sink=%s" % self.sink,
                  "nin=%d" % len(in_types), "nout=%d" % len(out_types)]
         if getattr(self.sp, "agg_expr", None) is not None:
-            lines.append("agg=%s" % T.deopt(self.sp.agg_type))
+            if getattr(self.sp, "agg_key_idx", None) is not None:
+                lines.append("aggby=%s" % T.deopt(self.sp.agg_type))
+            else:
+                lines.append("agg=%s" % T.deopt(self.sp.agg_type))
         for i, t in enumerate(in_types):
             lines.append("in%d=%s" % (i, tdesc(t)))
         for i, t in enumerate(out_types):

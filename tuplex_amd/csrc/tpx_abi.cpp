@@ -74,6 +74,7 @@ struct StageDesc {
     std::string source;  // mem | csv
     std::string sink;    // mem | csv
     std::string agg;     // "" | i64 | f64 — GPU-reducible aggregate fold
+    std::string aggby;   // "" | i64 | f64 — by-key hash-reduce (key col 0)
     std::vector<ColDesc> in_cols, out_cols;
 };
 
@@ -101,6 +102,7 @@ static bool parse_desc(const char* text, StageDesc* d) {
     d->source = kv.count("source") ? kv["source"] : "mem";
     d->sink = kv.count("sink") ? kv["sink"] : "mem";
     d->agg = kv.count("agg") ? kv["agg"] : "";
+    d->aggby = kv.count("aggby") ? kv["aggby"] : "";
     int nin = atoi(kv["nin"].c_str());
     int nout = atoi(kv["nout"].c_str());
     for (int i = 0; i < nin; ++i) {
@@ -133,6 +135,7 @@ struct tpx_stage {
     hipFunction_t k_csv_chunk = nullptr, k_csv_sel = nullptr, k_csv_rows = nullptr;
     hipFunction_t k_red_f64 = nullptr, k_red_f64_fin = nullptr;
     hipFunction_t k_red_i64 = nullptr, k_red_i64_fin = nullptr;
+    hipFunction_t k_hk_f64 = nullptr, k_hk_i64 = nullptr, k_hk_emit = nullptr;
     bool loaded = false;
 };
 
@@ -228,6 +231,9 @@ extern "C" tpx_stage* tpx_stage_compile(const char* hip_source,
         {"tpx_reduce_f64_final", &st->k_red_f64_fin, false},
         {"tpx_reduce_i64", &st->k_red_i64, false},
         {"tpx_reduce_i64_final", &st->k_red_i64_fin, false},
+        {"tpx_hashagg_f64", &st->k_hk_f64, false},
+        {"tpx_hashagg_i64", &st->k_hk_i64, false},
+        {"tpx_hashagg_emit", &st->k_hk_emit, false},
     };
     for (auto& e : lut) {
         hipError_t r = hipModuleGetFunction(e.fn, st->module, e.name);
@@ -488,7 +494,80 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
         }
     }
 
-    if (!D.agg.empty()) {
+    if (!D.aggby.empty()) {
+        // by-key hash-reduce (hashmap.cc analog); output order unpinned
+        hipFunction_t kfill = D.aggby == "f64" ? st->k_hk_f64 : st->k_hk_i64;
+        if (!kfill || !st->k_hk_emit) { set_err("hashagg kernels missing"); return -1; }
+        unsigned long long tsize = 1ull << 20;
+        unsigned long long used = 0;
+        void* tkeys = nullptr;
+        void* tvals = nullptr;
+        ARENA_TAKE(d_hk_state, 40);  // used, overflow, special_cnt, special_val, out_idx
+        for (int attempt = 0;; ++attempt) {
+            tkeys = g_arena[dev].take(tsize * 8);
+            tvals = g_arena[dev].take(tsize * 8);
+            if (!tkeys || !tvals) return -1;
+            HIP_CHECK(hipMemsetAsync(tkeys, 0xFF, tsize * 8, stream));
+            HIP_CHECK(hipMemsetAsync(tvals, 0, tsize * 8, stream));
+            HIP_CHECK(hipMemsetAsync(d_hk_state, 0, 40, stream));
+            unsigned long long tmask = tsize - 1;
+            void* keys = outv[0];
+            void* vals = outv[3];
+            void* d_used = d_hk_state;
+            void* d_ovf = (char*)d_hk_state + 8;
+            void* d_scnt = (char*)d_hk_state + 16;
+            void* d_sval = (char*)d_hk_state + 24;
+            unsigned grid2 = (unsigned)std::min<long long>((n + 255) / 256, 2048);
+            void* a1[] = {&keys, &vals, &d_keep, &n, &tkeys, &tvals, &tmask,
+                          &d_used, &d_ovf, &d_scnt, &d_sval};
+            if (launch(kfill, grid2, 256, stream, a1)) return -1;
+            unsigned long long st8[2] = {0, 0};
+            HIP_CHECK(hipMemcpyAsync(st8, d_hk_state, 16, hipMemcpyDeviceToHost,
+                                     stream));
+            HIP_CHECK(hipStreamSynchronize(stream));
+            used = st8[0];
+            bool ovf = st8[1] != 0 || used * 2 > tsize;
+            if (!ovf) break;
+            if (attempt >= 3 || tsize >= (1ull << 27)) {
+                set_err("aggregateByKey table overflow");
+                return -1;
+            }
+            tsize *= 8;
+        }
+        hipEventRecord(evs1, stream);
+        long long out_rows_cap = (long long)used + 1;
+        long long out_total = 8 + out_rows_cap * 16;
+        ARENA_TAKE(d_out, (size_t)out_total + 16);
+        void* d_oidx = (char*)d_hk_state + 32;
+        {
+            long long ts_ll = (long long)tsize;
+            void* d_scnt = (char*)d_hk_state + 16;
+            void* d_sval = (char*)d_hk_state + 24;
+            unsigned grid2 = (unsigned)std::min<long long>(((long long)tsize + 255) / 256, 4096);
+            void* a2[] = {&tkeys, &tvals, &ts_ll, &d_scnt, &d_sval, &d_out,
+                          &d_oidx};
+            if (launch(st->k_hk_emit, grid2, 256, stream, a2)) return -1;
+        }
+        hipEventRecord(ev2, stream);
+        unsigned long long nout_rows = 0;
+        HIP_CHECK(hipMemcpyAsync(&nout_rows, d_oidx, 8, hipMemcpyDeviceToHost,
+                                 stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
+        long long total = (long long)nout_rows;
+        res->out_size = 8 + total * 16;
+        res->out_num_rows = total;
+        res->out_data = (uint8_t*)malloc((size_t)res->out_size);
+        ((int64_t*)res->out_data)[0] = total;
+        if (total)
+            HIP_CHECK(hipMemcpy(res->out_data + 8, (char*)d_out + 8,
+                                (size_t)total * 16, hipMemcpyDeviceToHost));
+        res->out_row_offsets = (int64_t*)malloc(((size_t)total + 1) * 8);
+        res->out_row_indices = (int64_t*)malloc(((size_t)total + 1) * 8);
+        for (long long i = 0; i <= total; ++i)
+            res->out_row_offsets[i] = 8 + i * 16;
+        for (long long i = 0; i < total; ++i) res->out_row_indices[i] = i;
+        res->bytes_out = res->out_size;
+    } else if (!D.agg.empty()) {
         // GPU aggregate fold (Q6/count/sum): deterministic masked reduce of the
         // per-row expr column; result = a 1-row partition [numRows=1][value]
         bool is_f64 = D.agg == "f64";
@@ -627,7 +706,7 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
         res->exc_size = (int64_t)total;
         res->exc_num_rows = (int64_t)exc_count;
     }
-    if (!D.agg.empty()) res->bytes_out = 16;
+    if (!D.agg.empty() && D.aggby.empty()) res->bytes_out = 16;
     float ms = 0;
     hipEventElapsedTime(&ms, ev1, ev2); res->t_kernel_ms = ms;
     hipEventElapsedTime(&ms, ev2, ev3); res->t_d2h_ms = ms;

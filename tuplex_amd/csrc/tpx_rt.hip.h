@@ -951,6 +951,143 @@ extern "C" __global__ void tpx_reduce_i64_final(const long long* __restrict__ pa
     }
 }
 
+// ---- aggregateByKey hash-reduce (hashmap.cc / int_hashmap.cc analog) ----------
+// Open-addressing table, EMPTY key = -1 (table memset 0xFF); real key -1 goes to
+// a dedicated special slot. Output order is parity-unpinned.
+
+#define TPX_HK_EMPTY (-1LL)
+#define TPX_HK_MAXPROBE 1024
+
+__device__ __forceinline__ unsigned long long tpx_hash_i64(long long k) {
+    unsigned long long x = (unsigned long long)k;
+    x ^= x >> 33;
+    x *= 0xff51afd7ed558ccdULL;  // murmur3 finalizer (mix quality only)
+    x ^= x >> 33;
+    x *= 0xc4ceb9fe1a85ec53ULL;
+    x ^= x >> 33;
+    return x;
+}
+
+extern "C" __global__ void tpx_hashagg_f64(const long long* __restrict__ keys,
+                                           const double* __restrict__ vals,
+                                           const unsigned char* __restrict__ keep,
+                                           long long n, long long* tkeys,
+                                           double* tvals,
+                                           unsigned long long tmask,
+                                           unsigned long long* used,
+                                           int* overflow,
+                                           unsigned long long* special_cnt,
+                                           double* special_val) {
+    long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        if (!keep[i]) continue;
+        long long k = keys[i];
+        double v = vals[i];
+        if (k == TPX_HK_EMPTY) {
+            atomicAdd(special_cnt, 1ULL);
+            atomicAdd(special_val, v);
+            continue;
+        }
+        unsigned long long h = tpx_hash_i64(k) & tmask;
+        int probe = 0;
+        for (; probe < TPX_HK_MAXPROBE; ++probe) {
+            long long cur = tkeys[h];
+            if (cur == k) { atomicAdd(&tvals[h], v); break; }
+            if (cur == TPX_HK_EMPTY) {
+                unsigned long long prev = atomicCAS(
+                    (unsigned long long*)&tkeys[h],
+                    (unsigned long long)TPX_HK_EMPTY, (unsigned long long)k);
+                if (prev == (unsigned long long)TPX_HK_EMPTY) {
+                    atomicAdd(used, 1ULL);
+                    atomicAdd(&tvals[h], v);
+                    break;
+                }
+                if ((long long)prev == k) { atomicAdd(&tvals[h], v); break; }
+            }
+            h = (h + 1) & tmask;
+        }
+        if (probe == TPX_HK_MAXPROBE) *overflow = 1;
+    }
+}
+
+extern "C" __global__ void tpx_hashagg_i64(const long long* __restrict__ keys,
+                                           const long long* __restrict__ vals,
+                                           const unsigned char* __restrict__ keep,
+                                           long long n, long long* tkeys,
+                                           long long* tvals,
+                                           unsigned long long tmask,
+                                           unsigned long long* used,
+                                           int* overflow,
+                                           unsigned long long* special_cnt,
+                                           long long* special_val) {
+    long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += stride) {
+        if (!keep[i]) continue;
+        long long k = keys[i];
+        long long v = vals[i];
+        if (k == TPX_HK_EMPTY) {
+            atomicAdd(special_cnt, 1ULL);
+            atomicAdd((unsigned long long*)special_val, (unsigned long long)v);
+            continue;
+        }
+        unsigned long long h = tpx_hash_i64(k) & tmask;
+        int probe = 0;
+        for (; probe < TPX_HK_MAXPROBE; ++probe) {
+            long long cur = tkeys[h];
+            if (cur == k) {
+                atomicAdd((unsigned long long*)&tvals[h], (unsigned long long)v);
+                break;
+            }
+            if (cur == TPX_HK_EMPTY) {
+                unsigned long long prev = atomicCAS(
+                    (unsigned long long*)&tkeys[h],
+                    (unsigned long long)TPX_HK_EMPTY, (unsigned long long)k);
+                if (prev == (unsigned long long)TPX_HK_EMPTY) {
+                    atomicAdd(used, 1ULL);
+                    atomicAdd((unsigned long long*)&tvals[h],
+                              (unsigned long long)v);
+                    break;
+                }
+                if ((long long)prev == k) {
+                    atomicAdd((unsigned long long*)&tvals[h],
+                              (unsigned long long)v);
+                    break;
+                }
+            }
+            h = (h + 1) & tmask;
+        }
+        if (probe == TPX_HK_MAXPROBE) *overflow = 1;
+    }
+}
+
+// emit table entries as a [numRows][key,val] partition body (16 B rows); the
+// numRows header is patched host-side from *out_idx
+extern "C" __global__ void tpx_hashagg_emit(const long long* __restrict__ tkeys,
+                                            const long long* __restrict__ tvals,
+                                            long long tsize,
+                                            const unsigned long long* special_cnt,
+                                            const long long* special_val,
+                                            unsigned char* out,
+                                            unsigned long long* out_idx) {
+    long long stride = (long long)gridDim.x * blockDim.x;
+    long long tid0 = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+    if (tid0 == 0 && *special_cnt > 0) {
+        unsigned long long j = atomicAdd(out_idx, 1ULL);
+        long long* w = (long long*)(out + 8 + j * 16);
+        w[0] = TPX_HK_EMPTY;
+        w[1] = *special_val;
+    }
+    for (long long i = tid0; i < tsize; i += stride) {
+        if (tkeys[i] == TPX_HK_EMPTY) continue;
+        unsigned long long j = atomicAdd(out_idx, 1ULL);
+        long long* w = (long long*)(out + 8 + j * 16);
+        w[0] = tkeys[i];
+        w[1] = tvals[i];
+    }
+}
+
 // exclusive-scan building block: per-block scan of ITEMS_PER_BLOCK i64 items.
 // grid-wide composition happens host-side (scan block sums, then add pass).
 #define TPX_SCAN_THREADS 256

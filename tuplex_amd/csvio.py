@@ -393,14 +393,28 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
             out_bytes = ctypes.string_at(res.out_data, res.out_size)
             rows = rowfmt.deserialize_partition(out_bytes,
                                                 T.tup(sp.gpu_output_types))
-            agg = next((op for op in logical_ops if op[0] == "aggregate"), None)
+            agg = next((op for op in logical_ops
+                        if op[0] in ("aggregate", "aggregateByKey")), None)
             if sp.agg_expr is not None:
                 from .engine import _agg_row
-                _, combine_fn, agg_fn, initial = agg
-                acc = initial + rows[0][0]
-                for i in sorted(replayed):
-                    acc = agg_fn(acc, _agg_row(replayed[i], sp.output_columns))
-                out.rows = [acc]
+                if sp.agg_key_idx is not None:
+                    _, combine_fn, agg_fn, initial, key_cols = agg[:5]
+                    table = {row[0]: initial + row[1] for row in rows}
+                    ki = sp.output_columns.index(key_cols[0])
+                    for i in sorted(replayed):
+                        v = replayed[i]
+                        rt = v if isinstance(v, tuple) else (v,)
+                        k = rt[ki]
+                        table[k] = agg_fn(table.get(k, initial),
+                                          _agg_row(v, sp.output_columns))
+                    out.rows = [(k, val) for k, val in table.items()]
+                else:
+                    _, combine_fn, agg_fn, initial = agg
+                    acc = initial + rows[0][0]
+                    for i in sorted(replayed):
+                        acc = agg_fn(acc, _agg_row(replayed[i],
+                                                   sp.output_columns))
+                    out.rows = [acc]
                 return out
             idxs = [res.out_row_indices[i] for i in range(res.out_num_rows)]
             merged = {}

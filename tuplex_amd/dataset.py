@@ -54,8 +54,12 @@ class DataSet:
                   initial_value: Any) -> "DataSet":
         return self._chain(("aggregate", combine, aggregate, initial_value))
 
-    def aggregateByKey(self, combine, aggregate, initial_value, key_columns):
-        raise NotImplementedError("aggregateByKey: next slice (SURVEY.md §7.6)")
+    def aggregateByKey(self, combine, aggregate, initial_value,
+                       key_columns) -> "DataSet":
+        if not isinstance(key_columns, list):
+            key_columns = [key_columns]
+        return self._chain(("aggregateByKey", combine, aggregate, initial_value,
+                            key_columns))
 
     def unique(self) -> "DataSet":
         raise NotImplementedError("unique: next slice")

@@ -207,25 +207,42 @@ def run_collect(data: List[Any], logical_ops: List[tuple],
     input_types = list(T.tuple_params(row_maj))
     sp = plan.build_stage(input_types, columns, logical_ops)
 
-    # GPU-reducible aggregate: the stage returns one partial; replayed rows are
-    # folded on top with the user's agg fn (LocalBackend.cc:1180-1207 combine)
+    # GPU-reducible aggregate: the stage returns partials; replayed rows are
+    # folded on top with the user's agg fn (LocalBackend.cc:1180-1207 combine;
+    # by-key: createFinalHashmap LocalBackend.cc:2219)
     if (sp.compilable and norm_rows and sp.agg_expr is not None):
         er = execute_stage_mem(sp, norm_rows)
         out.mode = "gpu"
         out.metrics = er.metrics
-        opid_a, combine_fn, agg_fn, initial = sp.aggregate
-        acc = initial + er.rows[0][0]
         replay = [(norm_idx[r], data[norm_idx[r]]) for (r, _, _) in er.exceptions]
         replay += fallback
         agg_cols = sp.output_columns
-        for i, v in sorted(replay):
-            r = resolve.replay_row(v, logical_ops, columns, scalar_input)
-            if r[0] == "row":
-                acc = agg_fn(acc, _agg_row(r[1], agg_cols))
-            elif r[0] == "exc":
-                name = type(r[1]).__name__
-                out.exception_counts[name] = out.exception_counts.get(name, 0) + 1
-        out.rows = [acc]
+
+        def replay_rows():
+            for i, v in sorted(replay):
+                r = resolve.replay_row(v, logical_ops, columns, scalar_input)
+                if r[0] == "row":
+                    yield r[1]
+                elif r[0] == "exc":
+                    name = type(r[1]).__name__
+                    out.exception_counts[name] = \
+                        out.exception_counts.get(name, 0) + 1
+
+        if sp.agg_key_idx is not None:
+            opid_a, combine_fn, agg_fn, initial, key_cols = sp.aggregate
+            table = {row[0]: initial + row[1] for row in er.rows}
+            ki = agg_cols.index(key_cols[0])
+            for row in replay_rows():
+                rt = row if isinstance(row, tuple) else (row,)
+                k = rt[ki]
+                table[k] = agg_fn(table.get(k, initial), _agg_row(row, agg_cols))
+            out.rows = [(k, v) for k, v in table.items()]
+        else:
+            opid_a, combine_fn, agg_fn, initial = sp.aggregate
+            acc = initial + er.rows[0][0]
+            for row in replay_rows():
+                acc = agg_fn(acc, _agg_row(row, agg_cols))
+            out.rows = [acc]
         return out
 
     results = {}  # original index -> row value

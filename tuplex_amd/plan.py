@@ -47,13 +47,20 @@ class StageProgram:
         # partial sum; host folds replayed rows and the initial value on top.
         self.agg_expr = None
         self.agg_type = None
+        # by-key variant (hashmap sink, hashmap.cc/int_hashmap.cc analog):
+        # single i64 key column -> device open-addressing hash-reduce
+        self.agg_key_idx = None
+        self.agg_key_type = None
         self.compilable = True
         self.why_not_compilable = None
 
     @property
     def gpu_output_types(self):
-        """Types of the partition the GPU returns (agg stages: 1 column)."""
+        """Types of the partition the GPU returns (agg stages: 1 column;
+        by-key: key + value)."""
         if self.agg_expr is not None:
+            if self.agg_key_idx is not None:
+                return [self.agg_key_type, self.agg_type]
             return [self.agg_type]
         return self.output_types
 
@@ -121,6 +128,25 @@ def build_stage(input_types, input_columns, logical_ops) -> StageProgram:
         if kind == "aggregate":
             sp.aggregate = (opid, entry[1], entry[2], entry[3])
             _compile_aggregate(sp, opid, entry[2], entry[3], cur_types, cur_cols)
+            continue
+        if kind == "aggregateByKey":
+            key_cols = entry[4]
+            sp.aggregate = (opid, entry[1], entry[2], entry[3], key_cols)
+            _compile_aggregate(sp, opid, entry[2], entry[3], cur_types, cur_cols)
+            if (sp.agg_expr is not None and len(key_cols) == 1 and cur_cols
+                    and key_cols[0] in cur_cols):
+                ki = cur_cols.index(key_cols[0])
+                kt = cur_types[ki]
+                if kt == T.I64:
+                    sp.agg_key_idx = ki
+                    sp.agg_key_type = kt
+                else:
+                    sp.agg_expr = None  # non-i64 key: interpreter fallback
+                    _fallback(sp, "aggregateByKey key type %r (i64 only on GPU "
+                              "this round)" % (kt,))
+            else:
+                sp.agg_expr = None
+                _fallback(sp, "aggregateByKey outside the GPU pattern")
             continue
 
         op = StageOp(kind, opid)

@@ -35,7 +35,8 @@ def replay_row(value: Any, logical_ops: List[tuple], columns: Optional[List[str]
     ("exc", exception)."""
     cur = value
     cols = list(columns) if columns else None
-    row_ops = [op for op in logical_ops if op[0] != "aggregate"]
+    row_ops = [op for op in logical_ops
+               if op[0] not in ("aggregate", "aggregateByKey")]
     k = 0
     while k < len(row_ops):
         op = row_ops[k]
