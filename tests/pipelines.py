@@ -172,13 +172,16 @@ def _by_key_rows(n=50000, seed=5):
             for _ in range(n)]
 
 
-PIPELINES.append(("sum_by_key_small", [(0, 10.0), (1, 20.0), (0, -4.5)],
-                  ["id", "volume"],
-                  [("aggregateByKey", aggby_comb, aggby_vol, 0.0, ["id"])]))
-PIPELINES.append(("sum_by_key_large", _by_key_rows(), ["id", "volume"],
-                  [("aggregateByKey", aggby_comb, aggby_vol, 0.0, ["id"])]))
-PIPELINES.append(("count_by_key", _by_key_rows(8000, seed=11), ["id", "volume"],
-                  [("aggregateByKey", aggby_comb, aggby_cnt, 0, ["id"])]))
+# by-key pipelines: output order is parity-unpinned -> dedicated multiset-compare
+# test (tests/test_gpu_extra.py), not the generic ordered parity test
+BYKEY_PIPELINES = [
+    ("sum_by_key_small", [(0, 10.0), (1, 20.0), (0, -4.5)], ["id", "volume"],
+     [("aggregateByKey", aggby_comb, aggby_vol, 0.0, ["id"])]),
+    ("sum_by_key_large", _by_key_rows(), ["id", "volume"],
+     [("aggregateByKey", aggby_comb, aggby_vol, 0.0, ["id"])]),
+    ("count_by_key", _by_key_rows(8000, seed=11), ["id", "volume"],
+     [("aggregateByKey", aggby_comb, aggby_cnt, 0, ["id"])]),
+]
 
 
 def apply_ops(ds, ops):
@@ -202,6 +205,8 @@ def apply_ops(ds, ops):
             ds = ds.ignore(op[1])
         elif kind == "aggregate":
             ds = ds.aggregate(op[1], op[2], op[3])
+        elif kind == "aggregateByKey":
+            ds = ds.aggregateByKey(op[1], op[2], op[3], op[4])
         else:
             raise ValueError(kind)
     return ds
@@ -257,7 +262,7 @@ def precompile_all(verbose=False):
     glib = engine.GpuLib.get()
     opts = Options()
     n = 0
-    for name, data, columns, ops in PIPELINES:
+    for name, data, columns, ops in PIPELINES + BYKEY_PIPELINES:
         maj = T.infer_majority_type(data, optional_threshold=opts.optional_threshold)
         row_maj = T.row_type_of(maj)
         sp = plan.build_stage(list(T.tuple_params(row_maj)), columns, ops)

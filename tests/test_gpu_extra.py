@@ -98,8 +98,8 @@ def _sorted_rows(rows):
 @pytest.mark.parametrize("name", ["sum_by_key_small", "sum_by_key_large",
                                   "count_by_key"])
 def test_aggregate_by_key_gpu(name):
-    from tests.pipelines import PIPELINES
-    nm, data, columns, ops = [p for p in PIPELINES if p[0] == name][0]
+    from tests.pipelines import BYKEY_PIPELINES
+    nm, data, columns, ops = [p for p in BYKEY_PIPELINES if p[0] == name][0]
     ctx = tuplex_amd.Context()
     from tests.pipelines import apply_ops as ap
     ds = ap(ctx.parallelize(data, columns=columns), ops)
