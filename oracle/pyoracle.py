@@ -326,12 +326,15 @@ def run_pipeline(data: List[Any], ops: List[tuple], columns: Optional[List[str]]
     # aggregate stages handled as a fold at the end
     agg = None
     aggby = None
+    uniq = False
     row_ops = []
     for op in ops:
         if op[0] == "aggregate":
             agg = op
         elif op[0] == "aggregateByKey":
             aggby = op
+        elif op[0] == "unique":
+            uniq = True
         else:
             row_ops.append(op)
 
@@ -364,6 +367,8 @@ def run_pipeline(data: List[Any], ops: List[tuple], columns: Optional[List[str]]
         out = [a]
     elif aggby is not None:
         out = aggregate_by_key(out, aggby, columns)
+    elif uniq:
+        out = list(dict.fromkeys(out))  # hashmap-sink dedup (order unpinned)
 
     return {"output": out, "exception_counts": exc_counts}
 

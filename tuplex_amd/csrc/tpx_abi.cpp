@@ -75,6 +75,7 @@ struct StageDesc {
     std::string sink;    // mem | csv
     std::string agg;     // "" | i64 | f64 — GPU-reducible aggregate fold
     std::string aggby;   // "" | i64 | f64 — by-key hash-reduce (key col 0)
+    int textmode = 0;    // text() source: rows split on every newline (no quotes)
     std::vector<ColDesc> in_cols, out_cols;
 };
 
@@ -103,6 +104,7 @@ static bool parse_desc(const char* text, StageDesc* d) {
     d->sink = kv.count("sink") ? kv["sink"] : "mem";
     d->agg = kv.count("agg") ? kv["agg"] : "";
     d->aggby = kv.count("aggby") ? kv["aggby"] : "";
+    d->textmode = kv.count("textmode") ? atoi(kv["textmode"].c_str()) : 0;
     int nin = atoi(kv["nin"].c_str());
     int nout = atoi(kv["nout"].c_str());
     for (int i = 0; i < nin; ++i) {
@@ -807,8 +809,10 @@ static int64_t csv_boundary_and_core(tpx_stage* st, void* d_in_p, long long size
     // wave-cooperative kernels: one wave per chunk, 4 waves per 256-thread block
     unsigned grid_w = (unsigned)std::min<long long>((nchunks + 3) / 4, 8192);
     unsigned grid_t = (unsigned)std::min<long long>((nchunks + 255) / 256, 8192);
+    int quotes_on = st->desc.textmode ? 0 : 1;
     {
-        void* args[] = {&d_in_p, (void*)&size, &nchunks, &d_q, &d_c0, &d_c1};
+        void* args[] = {&d_in_p, (void*)&size, &nchunks, &d_q, &d_c0, &d_c1,
+                        &quotes_on};
         if (launch(st->k_csv_chunk, grid_w, 256, stream, args)) return -1;
     }
     long long qtotal = 0;
@@ -825,7 +829,8 @@ static int64_t csv_boundary_and_core(tpx_stage* st, void* d_in_p, long long size
     if (nrows == 0) { empty_result(res, st->desc.sink == "mem"); return 0; }
     ARENA_TAKE(d_offs, ((size_t)nrows + 1) * 8);
     {
-        void* args[] = {&d_in_p, (void*)&size, &nchunks, &d_qs, &d_base, &d_offs};
+        void* args[] = {&d_in_p, (void*)&size, &nchunks, &d_qs, &d_base, &d_offs,
+                        &quotes_on};
         if (launch(st->k_csv_rows, grid_w, 256, stream, args)) return -1;
     }
     hipEventRecord(eb1, stream);

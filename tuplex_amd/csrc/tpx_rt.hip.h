@@ -682,6 +682,7 @@ __device__ __forceinline__ long long tpx_wave_sum(long long v) {
 // LOCAL quote parity. Coalesced uint4 loads when the slice is full.
 __device__ __forceinline__ void tpx_scan64(const char* __restrict__ data,
                                            long long a, long long b,
+                                           int quotes_on,
                                            long long* lq, long long* l0,
                                            long long* l1) {
     long long qq = 0, e0 = 0, e1 = 0;
@@ -695,14 +696,14 @@ __device__ __forceinline__ void tpx_scan64(const char* __restrict__ data,
                 #pragma unroll
                 for (int s = 0; s < 4; ++s) {
                     unsigned ch = (words[k] >> (8 * s)) & 0xFF;
-                    if (ch == '"') ++qq;
+                    if (quotes_on && ch == '"') ++qq;
                     else if (ch == '\n') { if (qq & 1) ++e1; else ++e0; }
                 }
         }
     } else {
         for (long long i = a; i < b; ++i) {
             char ch = data[i];
-            if (ch == '"') ++qq;
+            if (quotes_on && ch == '"') ++qq;
             else if (ch == '\n') { if (qq & 1) ++e1; else ++e0; }
         }
     }
@@ -716,7 +717,8 @@ extern "C" __global__ void tpx_csv_chunk_stats(const char* __restrict__ data,
                                                long long size, long long nchunks,
                                                long long* __restrict__ q,
                                                long long* __restrict__ c0,
-                                               long long* __restrict__ c1) {
+                                               long long* __restrict__ c1,
+                                               int quotes_on) {
     int lane = threadIdx.x & 63;
     int wid = threadIdx.x >> 6;
     int wpb = blockDim.x >> 6;
@@ -728,7 +730,7 @@ extern "C" __global__ void tpx_csv_chunk_stats(const char* __restrict__ data,
         if (a > size) a = size;
         if (b > size) b = size;
         long long lq, l0, l1;
-        tpx_scan64(data, a, b, &lq, &l0, &l1);
+        tpx_scan64(data, a, b, quotes_on, &lq, &l0, &l1);
         long long pref = tpx_wave_exscan(lq);
         bool odd = (pref & 1) != 0;
         long long tot_q = tpx_wave_sum(lq);
@@ -753,7 +755,8 @@ extern "C" __global__ void tpx_csv_emit_rows(const char* __restrict__ data,
                                              long long size, long long nchunks,
                                              const long long* __restrict__ qscan,
                                              const long long* __restrict__ base,
-                                             long long* __restrict__ row_offs) {
+                                             long long* __restrict__ row_offs,
+                                             int quotes_on) {
     int lane = threadIdx.x & 63;
     int wid = threadIdx.x >> 6;
     int wpb = blockDim.x >> 6;
@@ -766,7 +769,7 @@ extern "C" __global__ void tpx_csv_emit_rows(const char* __restrict__ data,
         if (a > size) a = size;
         if (b > size) b = size;
         long long lq, l0, l1;
-        tpx_scan64(data, a, b, &lq, &l0, &l1);
+        tpx_scan64(data, a, b, quotes_on, &lq, &l0, &l1);
         long long pref_q = tpx_wave_exscan(lq);
         long long start_par = (qscan[c] + pref_q) & 1;
         long long sel = start_par ? l1 : l0;   // valid newlines in this lane
@@ -783,7 +786,7 @@ extern "C" __global__ void tpx_csv_emit_rows(const char* __restrict__ data,
                     #pragma unroll
                     for (int s2 = 0; s2 < 4; ++s2) {
                         unsigned ch = (words[k] >> (8 * s2)) & 0xFF;
-                        if (ch == '"') ++qq;
+                        if (quotes_on && ch == '"') ++qq;
                         else if (ch == '\n' && (((qq + start_par) & 1) == 0))
                             row_offs[1 + idx++] = a + v * 16 + k * 4 + s2 + 1;
                     }
@@ -791,7 +794,7 @@ extern "C" __global__ void tpx_csv_emit_rows(const char* __restrict__ data,
         } else {
             for (long long i = a; i < b; ++i) {
                 char ch = data[i];
-                if (ch == '"') ++qq;
+                if (quotes_on && ch == '"') ++qq;
                 else if (ch == '\n' && (((qq + start_par) & 1) == 0))
                     row_offs[1 + idx++] = i + 1;
             }

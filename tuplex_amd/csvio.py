@@ -283,6 +283,19 @@ def replay_csv_row(raw_line: bytes, col_types, null_values, logical_ops, columns
                                scalar_input=False)
 
 
+def replay_text_row(raw_line: bytes, col_type, null_values, logical_ops):
+    """text() replay: the value IS the line (minus newline); null_values map
+    matching lines to None."""
+    v = raw_line.decode("utf-8", "replace")
+    if v.endswith("\n"):
+        v = v[:-1]
+    if v.endswith("\r"):
+        v = v[:-1]
+    if T.is_opt(col_type) and v in null_values:
+        v = None
+    return _resolve.replay_row(v, logical_ops, None, scalar_input=True)
+
+
 class _BadParse(Exception):
     """Internal marker for structurally bad CSV rows; surfaces as
     'BadParseStringInput' in exception_counts (BADPARSE_STRING_INPUT,
@@ -312,16 +325,25 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
         with open(p, "rb") as f:
             blobs.append(f.read())
 
-    # sniff on the first file's sample (FileInputOperator.cc:78 semantics)
-    sample = blobs[0][:max(256 << 10, 1 << 20)]
-    # cut sample at the last complete row
-    nl = sample.rfind(b"\n")
-    if nl >= 0:
-        sample = sample[:nl + 1]
-    delim = (src.delimiter.encode() if src.delimiter else sniff_delimiter(sample))
-    has_header, names, col_types = sniff(sample, src.null_values,
-                                         opts.normalcase_threshold,
-                                         src.header, src.columns, delim)
+    text_mode = bool(getattr(src, "text_mode", False))
+    if text_mode:
+        # text(): one str column, rows split on every newline, no sniffing
+        has_header = False
+        names = None
+        col_types = [T.opt(T.STR) if src.null_values else T.STR]
+        delim = b","  # unused in text mode
+    else:
+        # sniff on the first file's sample (FileInputOperator.cc:78 semantics)
+        sample = blobs[0][:max(256 << 10, 1 << 20)]
+        # cut sample at the last complete row
+        nl = sample.rfind(b"\n")
+        if nl >= 0:
+            sample = sample[:nl + 1]
+        delim = (src.delimiter.encode() if src.delimiter
+                 else sniff_delimiter(sample))
+        has_header, names, col_types = sniff(sample, src.null_values,
+                                             opts.normalcase_threshold,
+                                             src.header, src.columns, delim)
 
     # assemble the GPU input: concat files, strip header lines, ensure trailing \n
     parts = []
@@ -352,7 +374,8 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
         csrc, desc = codegen.generate_stage(
             sp, source="csv", sink=sink_kind,
             csv_info={"null_values": src.null_values,
-                      "delimiter": delim.decode()})
+                      "delimiter": delim.decode(),
+                      "text_mode": text_mode})
     except codegen.CodegenError as e:
         return _run_csv_fallback(out, data, col_types, src.null_values,
                                  logical_ops, names, sink, str(e),
@@ -381,8 +404,12 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
                 row, ecode, opid, size = _s.unpack_from("<4q", eb, pos)
                 payload = eb[pos + 32:pos + 32 + size]
                 pos += 32 + size
-                r = replay_csv_row(payload, col_types, src.null_values,
-                                   logical_ops, names, delim.decode())
+                if text_mode:
+                    r = replay_text_row(payload, col_types[0], src.null_values,
+                                        logical_ops)
+                else:
+                    r = replay_csv_row(payload, col_types, src.null_values,
+                                       logical_ops, names, delim.decode())
                 if r[0] == "row":
                     replayed[row] = r[1]
                 elif r[0] == "exc":

@@ -62,7 +62,9 @@ class DataSet:
                             key_columns))
 
     def unique(self) -> "DataSet":
-        raise NotImplementedError("unique: next slice")
+        """Distinct rows (hashtable sink like aggregateByKey; output order is
+        unpinned, as the reference's hashmap iteration order is)."""
+        return self._chain(("unique",))
 
     def join(self, *a, **kw):
         raise NotImplementedError("join is out of scope (SURVEY.md §2 OOS)")

@@ -228,7 +228,12 @@ def run_collect(data: List[Any], logical_ops: List[tuple],
                     out.exception_counts[name] = \
                         out.exception_counts.get(name, 0) + 1
 
-        if sp.agg_key_idx is not None:
+        if sp.agg_unique and sp.agg_key_idx is not None:
+            keys = {row[0] for row in er.rows}
+            for row in replay_rows():
+                keys.add(row if not isinstance(row, tuple) else row[0])
+            out.rows = list(keys)
+        elif sp.agg_key_idx is not None:
             opid_a, combine_fn, agg_fn, initial, key_cols = sp.aggregate
             table = {row[0]: initial + row[1] for row in er.rows}
             ki = agg_cols.index(key_cols[0])
@@ -273,6 +278,8 @@ def run_collect(data: List[Any], logical_ops: List[tuple],
             out.exception_counts[name] = out.exception_counts.get(name, 0) + 1
 
     merged = [results[i] for i in sorted(results)]
+    if any(op[0] == "unique" for op in logical_ops):
+        merged = list(dict.fromkeys(merged))
     agg = next((op for op in logical_ops if op[0] == "aggregate"), None)
     if agg is not None:
         _, combine_fn, agg_fn, initial = agg

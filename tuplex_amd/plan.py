@@ -51,6 +51,7 @@ class StageProgram:
         # single i64 key column -> device open-addressing hash-reduce
         self.agg_key_idx = None
         self.agg_key_type = None
+        self.agg_unique = False   # unique(): by-key count, keys-only output
         self.compilable = True
         self.why_not_compilable = None
 
@@ -128,6 +129,22 @@ def build_stage(input_types, input_columns, logical_ops) -> StageProgram:
         if kind == "aggregate":
             sp.aggregate = (opid, entry[1], entry[2], entry[3])
             _compile_aggregate(sp, opid, entry[2], entry[3], cur_types, cur_cols)
+            continue
+        if kind == "unique":
+            # single i64 column -> device hash-reduce (count, keys-only output);
+            # anything else -> interpreter fallback dedup
+            sp.aggregate = (opid, None, None, 0, None)
+            sp.agg_unique = True
+            if len(cur_types) == 1 and cur_types[0] == T.I64:
+                from .udf import tir
+                sp.agg_expr = tir.const(1)
+                sp.agg_type = T.I64
+                sp.agg_key_idx = 0
+                sp.agg_key_type = T.I64
+                sp.agg_opid = opid
+            else:
+                _fallback(sp, "unique() on %r (single i64 column on GPU this "
+                          "round)" % (cur_types,))
             continue
         if kind == "aggregateByKey":
             key_cols = entry[4]

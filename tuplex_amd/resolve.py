@@ -36,7 +36,7 @@ def replay_row(value: Any, logical_ops: List[tuple], columns: Optional[List[str]
     cur = value
     cols = list(columns) if columns else None
     row_ops = [op for op in logical_ops
-               if op[0] not in ("aggregate", "aggregateByKey")]
+               if op[0] not in ("aggregate", "aggregateByKey", "unique")]
     k = 0
     while k < len(row_ops):
         op = row_ops[k]
