@@ -155,3 +155,54 @@ def test_unique_gpu():
     assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
     ref = pyoracle.run_pipeline(data, [("unique",)])
     assert sorted(got) == sorted(ref["output"])
+
+
+def test_multi_partition_mem():
+    """partitionSize splitting -> multi-partition C-ABI batch (ContextOptions
+    partitionSize, LocalBackend PartitionGroup analog)."""
+    from tests.zillow_data import make_zillow_rows, ZILLOW_COLS
+    from tests.test_codegen_compile import zillow_ops
+    rows = make_zillow_rows(3000, seed=21, dirty_frac=0.02)
+    ctx = tuplex_amd.Context({"partitionSize": "64KB"})
+    from tests.pipelines import apply_ops as ap
+    ds = ap(ctx.parallelize(rows, columns=ZILLOW_COLS), zillow_ops())
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    ref = pyoracle.run_pipeline(rows, zillow_ops(), columns=ZILLOW_COLS)
+    assert got == ref["output"]
+    assert ds.exception_counts == ref["exception_counts"]
+
+
+def test_chunked_csv_collect_and_tocsv(tmp_path):
+    """inputSplitSize chunking with exact quote-parity boundaries: quoted
+    newlines must never split a row (CSVUtils.cc:1494 findLineStart purpose)."""
+    import random
+    from tests.zillow_data import make_zillow_csv_bytes
+    from tests.test_codegen_compile import zillow_ops
+    data, _ = make_zillow_csv_bytes(4000, seed=31, dirty_frac=0.02)
+    # sprinkle rows with quoted embedded newlines into a copy of the stream
+    rng = random.Random(4)
+    lines = data.split(b"\n")
+    header, body = lines[0], lines[1:-1]
+    for i in range(0, len(body), 97):
+        parts = body[i].split(b",")
+        parts[1] = b'"line1\nline2"'  # address cell with embedded newline
+        body[i] = b",".join(parts)
+    data = header + b"\n" + b"\n".join(body) + b"\n"
+    p = _write(tmp_path, data, "chunked.csv")
+
+    from tests.pipelines import apply_ops as ap
+    ctx = tuplex_amd.Context({"inputSplitSize": "128KB"})
+    ds = ap(ctx.csv(p), zillow_ops())
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    assert ctx.metrics.data.get("chunks", 1) > 2
+    ref = pyoracle_csv.run_csv_pipeline(data, zillow_ops())
+    assert got == ref["output"]
+    assert ds.exception_counts == ref["exception_counts"]
+
+    outp = os.path.join(str(tmp_path), "out.csv")
+    ds2 = ap(ctx.csv(p), zillow_ops())
+    ds2.tocsv(outp)
+    ref2 = pyoracle_csv.run_csv_pipeline(data, zillow_ops(), sink="csv")
+    assert open(outp, "rb").read() == ref2["csv_text"]

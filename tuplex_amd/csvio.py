@@ -296,6 +296,31 @@ def replay_text_row(raw_line: bytes, col_type, null_values, logical_ops):
     return _resolve.replay_row(v, logical_ops, None, scalar_input=True)
 
 
+def split_points(data: bytes, target: int):
+    """Row-aligned chunk boundaries every ~target bytes with EXACT quote-parity
+    (the reference's findLineStart purpose, CSVUtils.cc:1494: a ranged chunk must
+    start at a true row start; quoted newlines must not split a row)."""
+    points = [0]
+    pos = target
+    counted_to = 0
+    parity = 0
+    n = len(data)
+    while pos < n:
+        nl = data.find(b"\n", pos)
+        while nl >= 0:
+            parity = (parity + data.count(b'"', counted_to, nl)) & 1
+            counted_to = nl
+            if parity == 0:
+                break
+            nl = data.find(b"\n", nl + 1)
+        if nl < 0:
+            break
+        points.append(nl + 1)
+        pos = nl + 1 + target
+    points.append(n)
+    return points
+
+
 class _BadParse(Exception):
     """Internal marker for structurally bad CSV rows; surfaces as
     'BadParseStringInput' in exception_counts (BADPARSE_STRING_INPUT,
@@ -382,93 +407,126 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
                                  delim.decode())
     stage = glib.compile_stage(csrc, desc)
 
-    buf = (ctypes.c_uint8 * len(data)).from_buffer_copy(data)
-    res = TpxResult()
-    rc = glib.lib.tpx_stage_execute_csv(
-        stage, ctypes.cast(buf, ctypes.POINTER(ctypes.c_uint8)), len(data), 0,
-        ctypes.byref(res))
-    if rc != 0:
-        raise RuntimeError("csv stage execute failed: " + glib.err())
-    try:
-        out.mode = "gpu"
-        out.metrics = {"t_h2d_ms": res.t_h2d_ms, "t_kernel_ms": res.t_kernel_ms,
-                       "t_d2h_ms": res.t_d2h_ms, "bytes_in": res.bytes_in,
-                       "bytes_out": res.bytes_out}
-        # replay exceptions (payload = raw line)
-        import struct as _s
-        replayed = {}  # row -> ("row", value) outcomes that produced rows
-        if res.exc_num_rows:
-            eb = ctypes.string_at(res.exc_data, res.exc_size)
-            pos = 0
-            for _ in range(res.exc_num_rows):
-                row, ecode, opid, size = _s.unpack_from("<4q", eb, pos)
-                payload = eb[pos + 32:pos + 32 + size]
-                pos += 32 + size
-                if text_mode:
-                    r = replay_text_row(payload, col_types[0], src.null_values,
-                                        logical_ops)
-                else:
-                    r = replay_csv_row(payload, col_types, src.null_values,
-                                       logical_ops, names, delim.decode())
-                if r[0] == "row":
-                    replayed[row] = r[1]
-                elif r[0] == "exc":
-                    nm = type(r[1]).__name__
-                    out.exception_counts[nm] = out.exception_counts.get(nm, 0) + 1
+    # chunked execution at inputSplitSize boundaries (LocalBackend.cc:552-658:
+    # one task per 64 MB range; here one execute per range, global row indices)
+    split = max(opts.input_split_size, 64 << 10)
+    chunks = (split_points(data, split) if len(data) > split * 2
+              else [0, len(data)])
 
-        if sink_kind == "mem":
-            out_bytes = ctypes.string_at(res.out_data, res.out_size)
-            rows = rowfmt.deserialize_partition(out_bytes,
-                                                T.tup(sp.gpu_output_types))
-            agg = next((op for op in logical_ops
-                        if op[0] in ("aggregate", "aggregateByKey")), None)
-            if sp.agg_expr is not None:
-                from .engine import _agg_row
-                if sp.agg_key_idx is not None:
-                    _, combine_fn, agg_fn, initial, key_cols = agg[:5]
-                    table = {row[0]: initial + row[1] for row in rows}
-                    ki = sp.output_columns.index(key_cols[0])
-                    for i in sorted(replayed):
-                        v = replayed[i]
-                        rt = v if isinstance(v, tuple) else (v,)
-                        k = rt[ki]
-                        table[k] = agg_fn(table.get(k, initial),
-                                          _agg_row(v, sp.output_columns))
-                    out.rows = [(k, val) for k, val in table.items()]
-                else:
-                    _, combine_fn, agg_fn, initial = agg
-                    acc = initial + rows[0][0]
-                    for i in sorted(replayed):
-                        acc = agg_fn(acc, _agg_row(replayed[i],
-                                                   sp.output_columns))
-                    out.rows = [acc]
-                return out
-            idxs = [res.out_row_indices[i] for i in range(res.out_num_rows)]
-            merged = {}
-            for row, i in zip(rows, idxs):
-                merged[i] = row[0] if len(row) == 1 else row
-            merged.update(replayed)
-            out.rows = [merged[i] for i in sorted(merged)]
-            if agg is not None:
-                from .engine import _agg_row
+    out.mode = "gpu"
+    out.metrics = {"t_h2d_ms": 0.0, "t_kernel_ms": 0.0, "t_d2h_ms": 0.0,
+                   "bytes_in": 0, "bytes_out": 0, "chunks": len(chunks) - 1}
+    import struct as _s
+    replayed = {}      # global row -> replay output
+    all_rows = []      # (global_idx, row) for mem sink
+    text_parts = []    # (chunk_res snapshot) for csv sink
+    first_row = 0
+    for ci in range(len(chunks) - 1):
+        cdata = data[chunks[ci]:chunks[ci + 1]]
+        buf = (ctypes.c_uint8 * len(cdata)).from_buffer_copy(cdata)
+        res = TpxResult()
+        rc = glib.lib.tpx_stage_execute_csv(
+            stage, ctypes.cast(buf, ctypes.POINTER(ctypes.c_uint8)), len(cdata),
+            first_row, ctypes.byref(res))
+        if rc != 0:
+            raise RuntimeError("csv stage execute failed: " + glib.err())
+        try:
+            for k in ("t_h2d_ms", "t_kernel_ms", "t_d2h_ms"):
+                out.metrics[k] += getattr(res, k)
+            out.metrics["bytes_in"] += res.bytes_in
+            out.metrics["bytes_out"] += res.bytes_out
+            if res.exc_num_rows:
+                eb = ctypes.string_at(res.exc_data, res.exc_size)
+                pos = 0
+                for _ in range(res.exc_num_rows):
+                    row, ecode, opid, size = _s.unpack_from("<4q", eb, pos)
+                    payload = eb[pos + 32:pos + 32 + size]
+                    pos += 32 + size
+                    if text_mode:
+                        r = replay_text_row(payload, col_types[0],
+                                            src.null_values, logical_ops)
+                    else:
+                        r = replay_csv_row(payload, col_types, src.null_values,
+                                           logical_ops, names, delim.decode())
+                    if r[0] == "row":
+                        replayed[row] = r[1]
+                    elif r[0] == "exc":
+                        nm = type(r[1]).__name__
+                        out.exception_counts[nm] = \
+                            out.exception_counts.get(nm, 0) + 1
+            if sink_kind == "mem":
+                out_bytes = ctypes.string_at(res.out_data, res.out_size)
+                rws = rowfmt.deserialize_partition(out_bytes,
+                                                   T.tup(sp.gpu_output_types))
+                idxs = [res.out_row_indices[i] for i in range(res.out_num_rows)]
+                all_rows.extend(zip(idxs, rws))
+            else:
+                text = (ctypes.string_at(res.out_data, res.out_size)
+                        if res.out_size else b"")
+                n_out = res.out_num_rows
+                text_parts.append((
+                    text,
+                    [res.out_row_indices[i] for i in range(n_out)],
+                    [res.out_row_offsets[i] for i in range(n_out + 1)],
+                    first_row, first_row + res.in_num_rows))
+            first_row += res.in_num_rows
+        finally:
+            glib.lib.tpx_result_free(ctypes.byref(res))
+
+    # ---- merge across chunks (global row indices; ResolveTask.cc:878 order) ----
+    if sink_kind == "mem":
+        all_rows.sort(key=lambda t: t[0])
+        rows = [r for _, r in all_rows]
+        agg = next((op for op in logical_ops
+                    if op[0] in ("aggregate", "aggregateByKey")), None)
+        if sp.agg_expr is not None:
+            from .engine import _agg_row
+            if sp.agg_key_idx is not None:
+                _, combine_fn, agg_fn, initial, key_cols = agg[:5]
+                table = {}
+                for row in rows:  # per-chunk partials: sum per key, + initial once
+                    table[row[0]] = table.get(row[0], initial) + row[1]
+                ki = sp.output_columns.index(key_cols[0])
+                for i in sorted(replayed):
+                    v = replayed[i]
+                    rt = v if isinstance(v, tuple) else (v,)
+                    k = rt[ki]
+                    table[k] = agg_fn(table.get(k, initial),
+                                      _agg_row(v, sp.output_columns))
+                out.rows = [(k, val) for k, val in table.items()]
+            else:
                 _, combine_fn, agg_fn, initial = agg
-                a = initial
-                for v in out.rows:
-                    a = agg_fn(a, _agg_row(v, sp.output_columns))
-                out.rows = [a]
-        else:
-            text = ctypes.string_at(res.out_data, res.out_size) if res.out_size \
-                else b""
-            header_line = _format_csv_row(sp.output_columns or
-                                          ["column%d" % i
-                                           for i in range(len(sp.output_types))])
+                acc = initial + sum(r[0] for r in rows)  # per-chunk partials
+                for i in sorted(replayed):
+                    acc = agg_fn(acc, _agg_row(replayed[i], sp.output_columns))
+                out.rows = [acc]
+            return out
+        merged = {}
+        for i, row in all_rows:
+            merged[i] = row[0] if len(row) == 1 else row
+        merged.update(replayed)
+        out.rows = [merged[i] for i in sorted(merged)]
+        if agg is not None:
+            from .engine import _agg_row
+            _, combine_fn, agg_fn, initial = agg
+            a = initial
+            for v in out.rows:
+                a = agg_fn(a, _agg_row(v, sp.output_columns))
+            out.rows = [a]
+    else:
+        header_line = _format_csv_row(sp.output_columns or
+                                      ["column%d" % i
+                                       for i in range(len(sp.output_types))])
+        segs = [header_line]
+        for text, idxs, offs, row_lo, row_hi in text_parts:
             if replayed:
-                text = _merge_csv_text(text, res, replayed, sp.output_types)
-            _write_csv_output(sink[1], header_line + text)
-            out.rows = []
-        return out
-    finally:
-        glib.lib.tpx_result_free(ctypes.byref(res))
+                segs.append(_merge_csv_segments(text, idxs, offs, replayed,
+                                                row_lo, row_hi))
+            else:
+                segs.append(text)
+        _write_csv_output(sink[1], b"".join(segs))
+        out.rows = []
+    return out
 
 
 def _run_csv_fallback(out, data, col_types, null_values, logical_ops, names,
@@ -517,13 +575,13 @@ def _format_csv_row(vals) -> bytes:
     return b",".join(_format_cell(v) for v in vals) + b"\n"
 
 
-def _merge_csv_text(text: bytes, res, replayed, out_types) -> bytes:
-    """Ordered merge of GPU CSV text with host-resolved rows (the tocsv analog of
-    executeInOrder)."""
-    n = res.out_num_rows
-    idxs = [res.out_row_indices[i] for i in range(n)]
-    offs = [res.out_row_offsets[i] for i in range(n + 1)]
-    rep = sorted(replayed.items())
+def _merge_csv_segments(text: bytes, idxs, offs, replayed, row_lo,
+                        row_hi) -> bytes:
+    """Ordered merge of one chunk's GPU CSV text with host-resolved rows (the
+    tocsv analog of executeInOrder). `replayed` keys are global input row
+    indices; the chunk owns [row_lo, row_hi)."""
+    n = len(idxs)
+    rep = sorted((k, v) for k, v in replayed.items() if row_lo <= k < row_hi)
     segs = []
     gi = ri = 0
     while gi < n or ri < len(rep):

@@ -135,8 +135,10 @@ class ExecResult:
 
 
 def execute_stage_mem(sp: plan.StageProgram, norm_rows: List[tuple],
-                      device: int = 0) -> ExecResult:
-    """Run the fused stage on the GPU over serialized normal-case rows."""
+                      device: int = 0, partition_size: int = 32 << 20) -> ExecResult:
+    """Run the fused stage on the GPU over serialized normal-case rows, split
+    into partitionSize partitions (ContextOptions.cc:202; one tpx_partition per
+    arena, LocalBackend.cc:491 PartitionGroup analog)."""
     glib = GpuLib.get()
     if glib.device_count() == 0:
         raise RuntimeError("no HIP device visible — the normal-case path runs only "
@@ -150,14 +152,35 @@ def execute_stage_mem(sp: plan.StageProgram, norm_rows: List[tuple],
     src, desc = codegen.generate_stage(sp, source="mem", sink="mem")
     stage = glib.compile_stage(src, desc)
 
+    # split the serialized rows at partition_size boundaries (row-aligned)
     cbuf = (ctypes.c_uint8 * len(buf)).from_buffer_copy(buf)
-    coffs = (ctypes.c_int64 * len(offs))(*offs)
-    part = TpxPartition(
-        data=ctypes.cast(cbuf, ctypes.POINTER(ctypes.c_uint8)),
-        size=len(buf), num_rows=len(norm_rows),
-        row_offsets=ctypes.cast(coffs, ctypes.POINTER(ctypes.c_int64)))
+    parts_meta = []  # (byte_start, row_start, row_end)
+    row_start = 0
+    byte_start = 8
+    for r in range(len(norm_rows) + 1):
+        if r == len(norm_rows) or (offs[r] - byte_start >= partition_size
+                                   and r > row_start):
+            if r > row_start:
+                parts_meta.append((byte_start, row_start, r))
+            if r < len(norm_rows):
+                byte_start = offs[r]
+                row_start = r
+    keep_alive = []
+    parts = (TpxPartition * max(len(parts_meta), 1))()
+    for pi, (bs, rs, re_) in enumerate(parts_meta):
+        n_rows = re_ - rs
+        # per-partition offsets relative to the partition's data pointer
+        rel = (ctypes.c_int64 * (n_rows + 1))(
+            *[offs[r] - bs for r in range(rs, re_)], offs[re_] - bs)
+        keep_alive.append(rel)
+        parts[pi] = TpxPartition(
+            data=ctypes.cast(ctypes.byref(cbuf, bs),
+                             ctypes.POINTER(ctypes.c_uint8)),
+            size=offs[re_] - bs, num_rows=n_rows,
+            row_offsets=ctypes.cast(rel, ctypes.POINTER(ctypes.c_int64)))
     res = TpxResult()
-    rc = glib.lib.tpx_stage_execute(stage, ctypes.byref(part), 1, ctypes.byref(res))
+    rc = glib.lib.tpx_stage_execute(stage, parts, max(len(parts_meta), 1),
+                                    ctypes.byref(res))
     if rc != 0:
         raise RuntimeError("stage execute failed: " + glib.err())
     try:
@@ -211,7 +234,8 @@ def run_collect(data: List[Any], logical_ops: List[tuple],
     # folded on top with the user's agg fn (LocalBackend.cc:1180-1207 combine;
     # by-key: createFinalHashmap LocalBackend.cc:2219)
     if (sp.compilable and norm_rows and sp.agg_expr is not None):
-        er = execute_stage_mem(sp, norm_rows)
+        er = execute_stage_mem(sp, norm_rows,
+                               partition_size=options.partition_size)
         out.mode = "gpu"
         out.metrics = er.metrics
         replay = [(norm_idx[r], data[norm_idx[r]]) for (r, _, _) in er.exceptions]
@@ -252,7 +276,8 @@ def run_collect(data: List[Any], logical_ops: List[tuple],
 
     results = {}  # original index -> row value
     if sp.compilable and norm_rows:
-        er = execute_stage_mem(sp, norm_rows)
+        er = execute_stage_mem(sp, norm_rows,
+                               partition_size=options.partition_size)
         out.mode = "gpu"
         out.metrics = er.metrics
         # outputs carry their exact local row index (out_row_indices) -> map back
