@@ -664,10 +664,24 @@ class StageCodegen:
         L.append("    }")
         L.append("    long long i = r0 + lane;")
         L.append("    if (i >= rhi) continue;")
+        # duplicate the row body per pointer mode: in the staged branch every
+        # parse pointer provably derives from LDS, so address-space inference
+        # emits ds_read instead of flat loads
+        L.append("    if (staged) {")
+        L.extend("  " + ln for ln in self._row_body(in_types, out_types, lds=True))
+        L.append("    } else {")
+        L.extend("  " + ln for ln in self._row_body(in_types, out_types, lds=False))
+        L.append("    }")
+        L.append("  }")
+        L.append("}")
+        return "\n".join(L)
+
+    def _row_body(self, in_types, out_types, lds):
+        L = []
         if self.source == "csv":
-            L.extend(self._load_inputs_csv(in_types))
+            L.extend(self._load_inputs_csv(in_types, lds))
         else:
-            L.extend(self._load_inputs_mem(in_types))
+            L.extend(self._load_inputs_mem(in_types, lds))
         L.append("    Out o;")
         args = []
         for idx, t in enumerate(in_types):
@@ -689,18 +703,18 @@ class StageCodegen:
         L.append("    }")
         L.append("    if (!o.keep) { keep[i] = 0; keep01[i] = 0; sizes[i] = 0; continue; }")
         L.append("    keep[i] = 1; keep01[i] = 1;")
-        L.extend(self._store_columnar(out_types))
-        L.append("  }")
-        L.append("}")
-        return "\n".join(L)
+        L.extend(self._store_columnar(out_types, lds))
+        return L
 
-    def _load_inputs_mem(self, in_types):
+    def _load_inputs_mem(self, in_types, lds=True):
         """Deserialize one reference-layout row (Serializer.cc:20-24) into typed
         locals c0..cN."""
         L = ["    long long prc = 0;  // no pre-parse errors on the mem path"]
-        L.append("    const unsigned char* row = staged"
-                 " ? (const unsigned char*)(wave_lds + (in_offs[i] - span_start))"
-                 " : in_data + in_offs[i];")
+        if lds:
+            L.append("    const unsigned char* row ="
+                     " (const unsigned char*)(wave_lds + (in_offs[i] - span_start));")
+        else:
+            L.append("    const unsigned char* row = in_data + in_offs[i];")
         n_opt = sum(1 for t in in_types if T.is_opt(t))
         bitmap_size = ((n_opt + 63) // 64) * 8 if n_opt else 0
         L.append("    // deserialize (bitmap %dB, %d slots)" % (bitmap_size, len(in_types)))
@@ -731,13 +745,13 @@ class StageCodegen:
                 raise CodegenError("input type %r" % (t,))
         return L
 
-    def _load_inputs_csv(self, in_types):
+    def _load_inputs_csv(self, in_types, lds=True):
         """Single-pass cell split + typed parse specialised to the sniffed schema —
         the CSVParseRowGenerator.cc replacement. Structure/parse failures set prc
         (BADPARSE/UNDERRUN/OVERRUN; raw line becomes the exception payload, like
         the reference's BADPARSE_STRING_INPUT rows)."""
         if self.csv_info.get("text_mode"):
-            return self._load_inputs_text(in_types)
+            return self._load_inputs_text(in_types, lds)
         nc = len(in_types)
         null_values = self.csv_info.get("null_values", [""])
         delim = self.csv_info.get("delimiter", ",")
@@ -745,10 +759,12 @@ class StageCodegen:
         delim_c = "'\\t'" if delim == "\t" else "'%s'" % delim
         L = []
         L.append("    long long prc = 0;")
-        L.append("    const char* rp = staged ? wave_lds + (in_offs[i] - span_start)"
-                 " : (const char*)in_data + in_offs[i];")
-        L.append("    const char* rend = staged ? wave_lds + (in_offs[i+1] - span_start)"
-                 " : (const char*)in_data + in_offs[i+1];")
+        if lds:
+            L.append("    const char* rp = wave_lds + (in_offs[i] - span_start);")
+            L.append("    const char* rend = wave_lds + (in_offs[i+1] - span_start);")
+        else:
+            L.append("    const char* rp = (const char*)in_data + in_offs[i];")
+            L.append("    const char* rend = (const char*)in_data + in_offs[i+1];")
         L.append("    if (rend > rp && rend[-1] == '\\n') --rend;")
         L.append("    if (rend > rp && rend[-1] == '\\r') --rend;")
         # straight-line per-cell scan: one named local per cell — a dynamically
@@ -799,17 +815,19 @@ class StageCodegen:
         L.append("    if (!prc && (badf & 6)) prc = %d;  // BADPARSE (escapes/structure -> host)" % 70)
         return L
 
-    def _load_inputs_text(self, in_types):
+    def _load_inputs_text(self, in_types, lds=True):
         """text() source: each row is the raw line (minus newline); no cell
         split, no quoting (Context::text, core/src/Context.cc)."""
         assert len(in_types) == 1
         t = in_types[0]
         null_values = self.csv_info.get("null_values", [])
         L = ["    long long prc = 0;"]
-        L.append("    const char* rp = staged ? wave_lds + (in_offs[i] - span_start)"
-                 " : (const char*)in_data + in_offs[i];")
-        L.append("    const char* rend = staged ? wave_lds + (in_offs[i+1] - span_start)"
-                 " : (const char*)in_data + in_offs[i+1];")
+        if lds:
+            L.append("    const char* rp = wave_lds + (in_offs[i] - span_start);")
+            L.append("    const char* rend = wave_lds + (in_offs[i+1] - span_start);")
+        else:
+            L.append("    const char* rp = (const char*)in_data + in_offs[i];")
+            L.append("    const char* rend = (const char*)in_data + in_offs[i+1];")
         L.append("    if (rend > rp && rend[-1] == '\\n') --rend;")
         L.append("    if (rend > rp && rend[-1] == '\\r') --rend;")
         L.append("    tstr c0{rp, rend - rp};")
@@ -820,16 +838,17 @@ class StageCodegen:
             L.append("    if (c0_n) c0 = tstr{rp, 0};")
         return L
 
-    def _store_columnar(self, out_types):
+    def _store_columnar(self, out_types, lds=True):
         """Store Out o -> columnar arrays + per-row serialized size (mem sink) or
         csv text size (csv sink)."""
         L = []
         for k, t in enumerate(out_types):
             base = T.deopt(t)
             if base == T.STR:
-                L.append("    o.o%d = tpx_to_global(o.o%d, wave_lds,"
-                         " wave_lds + TPX_SPAN_CAP, in_data, span_start);"
-                         % (k, k))
+                if lds:
+                    L.append("    o.o%d = tpx_to_global(o.o%d, wave_lds,"
+                             " wave_lds + TPX_SPAN_CAP, in_data, span_start);"
+                             % (k, k))
                 L.append("    ((unsigned long long*)outv[%d])[i] = (unsigned long long)o.o%d.p;"
                          % (3 * k, k))
                 L.append("    ((int*)outv[%d])[i] = (int)o.o%d.n;" % (3 * k + 1, k))
