@@ -349,8 +349,6 @@ class StageCodegen:
         if op == "slice":
             x, _ = ev(a[0])
             lo_node, hi_node = a[1], a[2]
-            lo = "0, false" if lo_node["op"] == "const" and lo_node.get("v") is None \
-                else None
             r = em.fresh("s")
             parts = []
             for nd in (lo_node, hi_node):
