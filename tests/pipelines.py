@@ -107,6 +107,14 @@ def ge0(x):
     return x >= 0
 
 
+def ign_filter(x):
+    return (1 // x) < 5
+
+
+def renamed_use(x):
+    return x["n2"] * 2
+
+
 def _rand_ints(n=10000, seed=42):
     rng = random.Random(seed)
     return [rng.randint(-2**40, 2**40) for _ in range(n)]
@@ -148,6 +156,11 @@ PIPELINES = [
      [("map", int_udf)]),
     ("filter_all", list(range(100)), None, [("filter", gt1000)]),
     ("filter_none", list(range(100)), None, [("filter", ge0)]),
+    ("ignore_zerodiv", [1, 0, 2, 0, 3], None,
+     [("filter", ign_filter), ("ignore", ZeroDivisionError)]),
+    ("rename_column", [(1, 5), (2, 6)], ["n1", "n2"],
+     [("renameColumn", "n2", "n2x"), ("renameColumn", "n2x", "n2"),
+      ("map", renamed_use)]),
     ("zillow_mem", make_zillow_rows(2000, seed=42), ZILLOW_COLS, zillow_ops()),
     ("agg_sum_large", list(range(1, 100001)), None,
      [("aggregate", agg_combine, agg_sum, 0)]),
