@@ -469,7 +469,7 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
         if (launch(st->k_rlens, grid, 256, stream, a1)) return -1;
         HIP_CHECK(hipMemsetAsync(d_hist, 0, 512 * 8, stream));
         void* a2[] = {&d_lens, &n, &d_hist};
-        if (launch(st->k_bhist, grid, 256, stream, a2)) return -1;
+        if (launch(st->k_bhist, 256, 256, stream, a2)) return -1;
         unsigned long long hist[512];
         HIP_CHECK(hipMemcpyAsync(hist, d_hist, 512 * 8, hipMemcpyDeviceToHost,
                                  stream));
@@ -483,7 +483,7 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
         HIP_CHECK(hipMemcpyAsync(d_hist, hist, 512 * 8, hipMemcpyHostToDevice,
                                  stream));
         void* a3[] = {&d_lens, &n, &d_hist, &d_permbuf};
-        if (launch(st->k_bscat, grid, 256, stream, a3)) return -1;
+        if (launch(st->k_bscat, 256, 256, stream, a3)) return -1;
         d_perm = d_permbuf;
     }
 
