@@ -633,34 +633,18 @@ class StageCodegen:
         L.append("    long long* __restrict__ sizes,")
         L.append("    long long* exc_buf, unsigned long long* exc_count,"
                  " unsigned long long exc_cap,")
-        L.append("    void** outv, const long long* __restrict__ perm) {")
-        L.append("  __shared__ char smem[2 * 64 * TPX_LANE_BYTES + 16];  // 2 waves x 64 x 288B lane slots (contiguous-span staging uses the low 16 KiB)")
+        L.append("    void** outv) {")
+        L.append("  __shared__ char smem[2 * TPX_SPAN_CAP + 16];  // 2 waves per 128-thread block -> 5 blocks/CU")
         L.append("  TpxHeap heap{heap_base, heap_cursor, heap_cap, nullptr, nullptr};")
         L.append("  int lane = threadIdx.x & 63;")
         L.append("  int wid = threadIdx.x >> 6;")
-        L.append("  char* wave_lds = smem + wid * 64 * TPX_LANE_BYTES;")
+        L.append("  char* wave_lds = smem + wid * TPX_SPAN_CAP;")
         L.append("  long long wave_stride = (long long)gridDim.x * (blockDim.x >> 6);")
         L.append("  long long nwaves = (n + 63) >> 6;")
         L.append("  for (long long wb = (long long)blockIdx.x * (blockDim.x >> 6) + wid;"
                  " wb < nwaves; wb += wave_stride) {")
         L.append("    long long r0 = wb << 6;")
         L.append("    long long rhi = r0 + 64 < n ? r0 + 64 : n;")
-        L.append("    if (perm) {")
-        L.append("      // length-bucketed path: lane's row staged into its own")
-        L.append("      // 288-B LDS slot; outputs keyed by the ORIGINAL index i")
-        L.append("      long long i0 = r0 + lane;")
-        L.append("      if (i0 >= rhi) continue;")
-        L.append("      long long i = perm[i0];")
-        L.append("      long long g0 = in_offs[i] & ~15LL;")
-        L.append("      long long rl16 = in_offs[i+1] - g0;")
-        L.append("      char* lbase = wave_lds + lane * TPX_LANE_BYTES;")
-        L.append("      bool lstaged = rl16 <= TPX_LANE_BYTES;")
-        L.append("      if (lstaged)")
-        L.append("        for (long long k = 0; k < rl16; k += 16)")
-        L.append("          *(uint4*)(lbase + k) = *(const uint4*)((const char*)in_data + g0 + k);")
-        L.extend("  " + ln for ln in self._row_body(in_types, out_types, lds="lane"))
-        L.append("      continue;")
-        L.append("    }")
         L.append("    long long span_start = in_offs[r0] & ~15LL;  // align staging window")
         L.append("    long long span_end = in_offs[rhi];")
         L.append("    long long span = span_end - span_start;")
@@ -724,11 +708,7 @@ class StageCodegen:
         """Deserialize one reference-layout row (Serializer.cc:20-24) into typed
         locals c0..cN."""
         L = ["    long long prc = 0;  // no pre-parse errors on the mem path"]
-        if lds == "lane":
-            L.append("    const unsigned char* row = lstaged"
-                     " ? (const unsigned char*)(lbase + (in_offs[i] - g0))"
-                     " : in_data + in_offs[i];")
-        elif lds:
+        if lds:
             L.append("    const unsigned char* row ="
                      " (const unsigned char*)(wave_lds + (in_offs[i] - span_start));")
         else:
@@ -777,11 +757,7 @@ class StageCodegen:
         delim_c = "'\\t'" if delim == "\t" else "'%s'" % delim
         L = []
         L.append("    long long prc = 0;")
-        if lds == "lane":
-            L.append("    const char* rp = lstaged ? lbase + (in_offs[i] - g0)"
-                     " : (const char*)in_data + in_offs[i];")
-            L.append("    const char* rend = rp + (in_offs[i+1] - in_offs[i]);")
-        elif lds:
+        if lds:
             L.append("    const char* rp = wave_lds + (in_offs[i] - span_start);")
             L.append("    const char* rend = wave_lds + (in_offs[i+1] - span_start);")
         else:
@@ -844,11 +820,7 @@ class StageCodegen:
         t = in_types[0]
         null_values = self.csv_info.get("null_values", [])
         L = ["    long long prc = 0;"]
-        if lds == "lane":
-            L.append("    const char* rp = lstaged ? lbase + (in_offs[i] - g0)"
-                     " : (const char*)in_data + in_offs[i];")
-            L.append("    const char* rend = rp + (in_offs[i+1] - in_offs[i]);")
-        elif lds:
+        if lds:
             L.append("    const char* rp = wave_lds + (in_offs[i] - span_start);")
             L.append("    const char* rend = wave_lds + (in_offs[i+1] - span_start);")
         else:
@@ -871,10 +843,7 @@ class StageCodegen:
         for k, t in enumerate(out_types):
             base = T.deopt(t)
             if base == T.STR:
-                if lds == "lane":
-                    L.append("    o.o%d = tpx_lane_to_global(o.o%d, lbase,"
-                             " in_data, g0);" % (k, k))
-                elif lds:
+                if lds:
                     L.append("    o.o%d = tpx_to_global(o.o%d, wave_lds,"
                              " wave_lds + TPX_SPAN_CAP, in_data, span_start);"
                              % (k, k))

@@ -138,7 +138,6 @@ struct tpx_stage {
     hipFunction_t k_red_f64 = nullptr, k_red_f64_fin = nullptr;
     hipFunction_t k_red_i64 = nullptr, k_red_i64_fin = nullptr;
     hipFunction_t k_hk_f64 = nullptr, k_hk_i64 = nullptr, k_hk_emit = nullptr;
-    hipFunction_t k_rlens = nullptr, k_bhist = nullptr, k_bscat = nullptr;
     bool loaded = false;
 };
 
@@ -237,9 +236,6 @@ extern "C" tpx_stage* tpx_stage_compile(const char* hip_source,
         {"tpx_hashagg_f64", &st->k_hk_f64, false},
         {"tpx_hashagg_i64", &st->k_hk_i64, false},
         {"tpx_hashagg_emit", &st->k_hk_emit, false},
-        {"tpx_row_lens", &st->k_rlens, false},
-        {"tpx_bucket_hist", &st->k_bhist, false},
-        {"tpx_bucket_scatter", &st->k_bscat, false},
     };
     for (auto& e : lut) {
         hipError_t r = hipModuleGetFunction(e.fn, st->module, e.name);
@@ -457,36 +453,6 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
     }
     heap_cap = P.heap_cap;
 
-    // length-bucketed row permutation (divergence fix — see tpx_rt.hip.h):
-    // waves then process near-equal-length rows; outputs stay index-keyed
-    void* d_perm = nullptr;
-    if (n >= 65536 && st->k_rlens && st->k_bhist && st->k_bscat) {
-        ARENA_TAKE(d_lens, (size_t)n * 8);
-        ARENA_TAKE(d_permbuf, (size_t)n * 8);
-        ARENA_TAKE(d_hist, 512 * 8);
-        unsigned grid = (unsigned)std::min<long long>((n + 255) / 256, 2048);
-        void* a1[] = {&d_offs, &n, &d_lens};
-        if (launch(st->k_rlens, grid, 256, stream, a1)) return -1;
-        HIP_CHECK(hipMemsetAsync(d_hist, 0, 512 * 8, stream));
-        void* a2[] = {&d_lens, &n, &d_hist};
-        if (launch(st->k_bhist, 256, 256, stream, a2)) return -1;
-        unsigned long long hist[512];
-        HIP_CHECK(hipMemcpyAsync(hist, d_hist, 512 * 8, hipMemcpyDeviceToHost,
-                                 stream));
-        HIP_CHECK(hipStreamSynchronize(stream));
-        unsigned long long run = 0;
-        for (int b = 0; b < 512; ++b) {
-            unsigned long long c = hist[b];
-            hist[b] = run;
-            run += c;
-        }
-        HIP_CHECK(hipMemcpyAsync(d_hist, hist, 512 * 8, hipMemcpyHostToDevice,
-                                 stream));
-        void* a3[] = {&d_lens, &n, &d_hist, &d_permbuf};
-        if (launch(st->k_bscat, 256, 256, stream, a3)) return -1;
-        d_perm = d_permbuf;
-    }
-
     void* d_heap_cursor = P.counters;
     void* d_exc_count = (char*)P.counters + 8;
     unsigned long long exc_count = 0;
@@ -498,7 +464,7 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
         unsigned grid = (unsigned)std::min<long long>((waves + 1) / 2, 4096);
         void* args[] = {&d_in, &d_offs, &n, &row0, &P.heap, &d_heap_cursor,
                         &heap_cap, &d_keep, &d_keep01, &d_sizes, &P.exc,
-                        &d_exc_count, &exc_cap, &d_outv, &d_perm};
+                        &d_exc_count, &exc_cap, &d_outv};
         hipEventRecord(evm0, stream);
         if (launch(st->k_main, grid, 128, stream, args)) return -1;
         hipEventRecord(evm1, stream);

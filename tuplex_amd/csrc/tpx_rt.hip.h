@@ -21,8 +21,6 @@
 #endif
 
 #define TPX_WAVE 64
-#define TPX_NBUCKETS 512
-#define TPX_LANE_BYTES 288  /* per-lane LDS row slot in the permuted path */
 
 // ExceptionCodes.h:26 values (subset used on device)
 #define EC_OK 0
@@ -172,16 +170,6 @@ __device__ __forceinline__ void tpx_memcpy(char* d, const char* s, long long n) 
         }
     }
     for (; i < n; ++i) d[i] = s[i];
-}
-
-// per-lane variant for the permuted path: the lane's row was staged at lbase
-// from aligned global offset g0
-__device__ __forceinline__ tstr tpx_lane_to_global(tstr s, const char* lbase,
-                                                   const unsigned char* gbase,
-                                                   long long g0) {
-    if (s.p >= lbase && s.p < lbase + TPX_LANE_BYTES)
-        return tstr{(const char*)gbase + g0 + (s.p - lbase), s.n};
-    return s;
 }
 
 // translate an LDS-staged string view back to its global-memory address before it
@@ -963,75 +951,6 @@ extern "C" __global__ void tpx_reduce_i64_final(const long long* __restrict__ pa
         long long s = 0;
         for (long long i = 0; i < n; ++i) s += partials[i];
         *out = s;
-    }
-}
-
-// ---- length-bucketed row permutation --------------------------------------------
-// Variable-length rows make each wave execute the union of its 64 lanes'
-// data-dependent scan loops (measured 6.5x slowdown vs uniform rows). Bucketing
-// row indices by length (8-byte buckets) lets waves process near-equal-length
-// rows; outputs stay keyed by the ORIGINAL index, so compaction/order-merge are
-// unchanged.
-
-extern "C" __global__ void tpx_row_lens(const long long* __restrict__ offs,
-                                        long long n,
-                                        long long* __restrict__ lens) {
-    long long stride = (long long)gridDim.x * blockDim.x;
-    for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-         i += stride)
-        lens[i] = offs[i + 1] - offs[i];
-}
-
-extern "C" __global__ void tpx_bucket_hist(const long long* __restrict__ lens,
-                                           long long n,
-                                           unsigned long long* __restrict__ hist) {
-    // block-local LDS histogram first: zillow-like data concentrates in ~12
-    // buckets, and 5M global atomicAdds on a dozen words serialize (~11 ns each,
-    // MI355X_MICROARCH.md §dequeue — measured ~45 ms per step naive)
-    __shared__ unsigned int lh[TPX_NBUCKETS];
-    for (int t = threadIdx.x; t < TPX_NBUCKETS; t += blockDim.x) lh[t] = 0;
-    __syncthreads();
-    long long stride = (long long)gridDim.x * blockDim.x;
-    for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-         i += stride) {
-        long long b = lens[i] >> 3;
-        if (b > TPX_NBUCKETS - 1) b = TPX_NBUCKETS - 1;
-        atomicAdd(&lh[b], 1u);
-    }
-    __syncthreads();
-    for (int t = threadIdx.x; t < TPX_NBUCKETS; t += blockDim.x)
-        if (lh[t]) atomicAdd(&hist[t], (unsigned long long)lh[t]);
-}
-
-// each block owns a contiguous row range, counts locally, claims per-bucket base
-// offsets with ONE global atomic per (block, bucket), then places its rows
-extern "C" __global__ void tpx_bucket_scatter(const long long* __restrict__ lens,
-                                              long long n,
-                                              unsigned long long* __restrict__ cursors,
-                                              long long* __restrict__ perm) {
-    __shared__ unsigned int lh[TPX_NBUCKETS];
-    __shared__ unsigned long long lbase[TPX_NBUCKETS];
-    long long per = (n + gridDim.x - 1) / gridDim.x;
-    long long lo = (long long)blockIdx.x * per;
-    long long hi = lo + per < n ? lo + per : n;
-    for (int t = threadIdx.x; t < TPX_NBUCKETS; t += blockDim.x) lh[t] = 0;
-    __syncthreads();
-    for (long long i = lo + threadIdx.x; i < hi; i += blockDim.x) {
-        long long b = lens[i] >> 3;
-        if (b > TPX_NBUCKETS - 1) b = TPX_NBUCKETS - 1;
-        atomicAdd(&lh[b], 1u);
-    }
-    __syncthreads();
-    for (int t = threadIdx.x; t < TPX_NBUCKETS; t += blockDim.x)
-        lbase[t] = lh[t] ? atomicAdd(&cursors[t], (unsigned long long)lh[t]) : 0;
-    __syncthreads();
-    for (int t = threadIdx.x; t < TPX_NBUCKETS; t += blockDim.x) lh[t] = 0;
-    __syncthreads();
-    for (long long i = lo + threadIdx.x; i < hi; i += blockDim.x) {
-        long long b = lens[i] >> 3;
-        if (b > TPX_NBUCKETS - 1) b = TPX_NBUCKETS - 1;
-        unsigned p = atomicAdd(&lh[b], 1u);
-        perm[lbase[b] + p] = i;
     }
 }
 
