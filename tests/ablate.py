@@ -1,0 +1,61 @@
+"""Ablation probe (GPU box): how much of tpx_stage_main is CSV parse vs UDF chain.
+
+Runs the Zillow CSV through (A) a passthrough stage (parse + columnar store of all
+10 input columns, no UDFs) and (B) the full Z1 pipeline, device-resident, and
+prints per-kernel times. Usage: python tests/ablate.py [mb]
+"""
+import ctypes
+import sys
+
+sys.path.insert(0, ".")
+
+from bench import make_input  # noqa: E402
+from tests.test_codegen_compile import zillow_ops  # noqa: E402
+from tuplex_amd import codegen, csvio, plan  # noqa: E402
+from tuplex_amd.engine import GpuLib, TpxResult  # noqa: E402
+
+
+def run(stage, glib, dev, size, label):
+    for i in range(3):
+        res = TpxResult()
+        rc = glib.lib.tpx_stage_execute_csv_dev(stage, dev, size, 0, 2,
+                                                ctypes.byref(res))
+        assert rc == 0, glib.err()
+        if i == 2:
+            print("%-12s main %.3f ms  boundary %.3f  compact %.3f  write %.3f  "
+                  "rows %d out %d" % (label, res.t_main_ms, res.t_boundary_ms,
+                                      res.t_compact_ms, res.t_write_ms,
+                                      res.in_num_rows, res.out_num_rows))
+        glib.lib.tpx_result_free(ctypes.byref(res))
+
+
+def main():
+    mb = int(sys.argv[1]) if len(sys.argv) > 1 else 1024
+    header, body = make_input(150000, 42, 0.0, mb << 20)
+    glib = GpuLib.get()
+    assert glib.device_count() > 0
+    dev = glib.lib.tpx_dev_alloc(len(body))
+    buf = (ctypes.c_uint8 * len(body)).from_buffer_copy(body)
+    glib.lib.tpx_dev_upload(dev, buf, len(body))
+    del buf
+
+    sample = body[:1 << 20]
+    _h, names, col_types = csvio.sniff(sample, [""], 0.9, False, None, b",")
+
+    variants = [
+        ("passthrough", []),
+        ("udf_nofilter", [op for op in zillow_ops()
+                          if op[0] in ("withColumn", "mapColumn")]),
+        ("full", zillow_ops()),
+    ]
+    for label, ops in variants:
+        sp = plan.build_stage(col_types, names, ops)
+        assert sp.compilable, (label, sp.why_not_compilable)
+        src, desc = codegen.generate_stage(sp, source="csv", sink="mem",
+                                           csv_info={"null_values": [""]})
+        stage = glib.compile_stage(src, desc)
+        run(stage, glib, dev, len(body), label)
+
+
+if __name__ == "__main__":
+    main()
