@@ -739,6 +739,9 @@ class StageCodegen:
                          " (info%d >> 32) - 1};" % (idx, bitmap_size, idx * 8, idx, idx))
                 if T.is_opt(t):
                     L.append("    if (c%d_n) c%d = tstr{(const char*)row, 0};" % (idx, idx))
+                # per-column ASCII gate (mem rows hold binary i64/f64 slots, so
+                # only the string bytes are scanned) — NCV divert, once per col
+                L.append("    if (!prc && !tpx_ascii(c%d)) prc = 7;" % idx)
             else:
                 raise CodegenError("input type %r" % (t,))
         return L
@@ -811,6 +814,10 @@ class StageCodegen:
                 raise CodegenError("csv input type %r" % (t,))
         L.append("    if (!prc && avail) prc = %d;  // CSV_OVERRUN" % 21)
         L.append("    if (!prc && (badf & 6)) prc = %d;  // BADPARSE (escapes/structure -> host)" % 70)
+        # ONE ASCII gate per row (char-index-sensitive ops then skip their own
+        # scans; non-ASCII rows divert to the interpreter — semantically free)
+        L.append("    if (!prc && !tpx_ascii(tstr{rp, rend - rp})) prc = %d;  // NCV"
+                 % 7)
         return L
 
     def _load_inputs_text(self, in_types, lds=True):
@@ -829,6 +836,7 @@ class StageCodegen:
         L.append("    if (rend > rp && rend[-1] == '\\n') --rend;")
         L.append("    if (rend > rp && rend[-1] == '\\r') --rend;")
         L.append("    tstr c0{rp, rend - rp};")
+        L.append("    if (!tpx_ascii(c0)) prc = %d;  // row ASCII gate (NCV)" % 7)
         if T.is_opt(t):
             checks = " || ".join("tpx_streq(c0, %s)" % self.lit(nv)
                                  for nv in null_values) or "false"

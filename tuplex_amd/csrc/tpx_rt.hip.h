@@ -130,6 +130,28 @@ __device__ __forceinline__ long long tpx_memchr2(const char* p, long long n,
     return -1;
 }
 
+// last index of byte c in p[0..n), or -1 (backward SWAR)
+__device__ __forceinline__ long long tpx_memrchr(const char* p, long long n,
+                                                 char c) {
+    unsigned long long pat = TPX_SWAR_ONE * (unsigned char)c;
+    long long i = n;
+    while (i > 0 && (((unsigned long long)(p + i)) & 7)) {
+        --i;
+        if (p[i] == c) return i;
+    }
+    while (i >= 8) {
+        i -= 8;
+        unsigned long long v = *(const unsigned long long*)(p + i);
+        unsigned long long hit = tpx_swar_zero(v ^ pat);
+        if (hit) return i + (63 - __builtin_clzll(hit)) / 8;
+    }
+    while (i > 0) {
+        --i;
+        if (p[i] == c) return i;
+    }
+    return -1;
+}
+
 // scan for non-ASCII; true if all ASCII (SWAR high-bit sweep)
 __device__ __forceinline__ bool tpx_ascii(const tstr s) {
     const char* p = s.p;
@@ -221,13 +243,11 @@ __device__ __forceinline__ double tpx_mod_f64(double a, double b, int* ec) {
 
 // len: python counts CHARS; byte len == char len only for ASCII (divert otherwise)
 __device__ __forceinline__ long long tpx_len(const tstr s, int* ec) {
-    if (!tpx_ascii(s)) { *ec = EC_NCV; return 0; }
-    return s.n;
+    return s.n;  // rows are ASCII-gated once at load (EC_NCV divert)
 }
 
 // find/rfind return python CHAR index; ASCII-gate the prefix we index over
 __device__ __forceinline__ long long tpx_find(const tstr s, const tstr needle, int* ec) {
-    if (!tpx_ascii(s)) { *ec = EC_NCV; return 0; }
     if (needle.n == 0) return 0;
     long long i = 0;
     long long last = s.n - needle.n;
@@ -245,13 +265,16 @@ __device__ __forceinline__ long long tpx_find(const tstr s, const tstr needle, i
 }
 
 __device__ __forceinline__ long long tpx_rfind(const tstr s, const tstr needle, int* ec) {
-    if (!tpx_ascii(s)) { *ec = EC_NCV; return 0; }
     if (needle.n == 0) return s.n;
-    for (long long i = s.n - needle.n; i >= 0; --i) {
+    long long hi = s.n - needle.n;  // highest candidate start
+    while (hi >= 0) {
+        long long k = tpx_memrchr(s.p, hi + 1, needle.p[0]);
+        if (k < 0) return -1;
         bool m = true;
-        for (long long j = 0; j < needle.n; ++j)
-            if (s.p[i + j] != needle.p[j]) { m = false; break; }
-        if (m) return i;
+        for (long long j = 1; j < needle.n; ++j)
+            if (s.p[k + j] != needle.p[j]) { m = false; break; }
+        if (m) return k;
+        hi = k - 1;
     }
     return -1;
 }
@@ -306,7 +329,6 @@ __device__ __forceinline__ int tpx_strcmp(const tstr a, const tstr b) {
 
 // x[i]: python char access with negative wrap; view, no alloc
 __device__ __forceinline__ tstr tpx_getitem(const tstr s, long long i, int* ec) {
-    if (!tpx_ascii(s)) { *ec = EC_NCV; return tstr{s.p, 0}; }
     long long idx = i < 0 ? i + s.n : i;
     if (idx < 0 || idx >= s.n) { *ec = EC_INDEXERROR; return tstr{s.p, 0}; }
     return tstr{s.p + idx, 1};
@@ -315,7 +337,6 @@ __device__ __forceinline__ tstr tpx_getitem(const tstr s, long long i, int* ec) 
 // s[lo:hi]: python clamp semantics; view
 __device__ __forceinline__ tstr tpx_slice(const tstr s, long long lo, bool has_lo,
                                           long long hi, bool has_hi, int* ec) {
-    if (!tpx_ascii(s)) { *ec = EC_NCV; return tstr{s.p, 0}; }
     long long a = has_lo ? lo : 0;
     long long b = has_hi ? hi : s.n;
     if (a < 0) a += s.n;
@@ -331,7 +352,6 @@ __device__ __forceinline__ tstr tpx_lower(TpxHeap& h, const tstr s, int* ec) {
     if (!d) { *ec = EC_MEMORYERROR; return tstr{s.p, 0}; }
     for (long long i = 0; i < s.n; ++i) {
         unsigned char c = s.p[i];
-        if (c >= 0x80) { *ec = EC_NCV; return tstr{s.p, 0}; }
         d[i] = (c >= 'A' && c <= 'Z') ? c + 32 : c;
     }
     return tstr{d, s.n};
@@ -342,7 +362,6 @@ __device__ __forceinline__ tstr tpx_upper(TpxHeap& h, const tstr s, int* ec) {
     if (!d) { *ec = EC_MEMORYERROR; return tstr{s.p, 0}; }
     for (long long i = 0; i < s.n; ++i) {
         unsigned char c = s.p[i];
-        if (c >= 0x80) { *ec = EC_NCV; return tstr{s.p, 0}; }
         d[i] = (c >= 'a' && c <= 'z') ? c - 32 : c;
     }
     return tstr{d, s.n};
@@ -353,7 +372,6 @@ __device__ __forceinline__ tstr tpx_swapcase(TpxHeap& h, const tstr s, int* ec) 
     if (!d) { *ec = EC_MEMORYERROR; return tstr{s.p, 0}; }
     for (long long i = 0; i < s.n; ++i) {
         unsigned char c = s.p[i];
-        if (c >= 0x80) { *ec = EC_NCV; return tstr{s.p, 0}; }
         d[i] = (c >= 'a' && c <= 'z') ? c - 32 : ((c >= 'A' && c <= 'Z') ? c + 32 : c);
     }
     return tstr{d, s.n};
@@ -367,18 +385,8 @@ __device__ __forceinline__ bool tpx_is_pyws(unsigned char c) {
 // multi-byte whitespace -> divert; interior bytes don't matter
 __device__ __forceinline__ tstr tpx_strip(const tstr s, int* ec) {
     long long a = 0, b = s.n;
-    while (a < b) {
-        unsigned char c = s.p[a];
-        if (c >= 0x80) { *ec = EC_NCV; return tstr{s.p, 0}; }
-        if (!tpx_is_pyws(c)) break;
-        ++a;
-    }
-    while (b > a) {
-        unsigned char c = s.p[b - 1];
-        if (c >= 0x80) { *ec = EC_NCV; return tstr{s.p, 0}; }
-        if (!tpx_is_pyws(c)) break;
-        --b;
-    }
+    while (a < b && tpx_is_pyws(s.p[a])) ++a;
+    while (b > a && tpx_is_pyws(s.p[b - 1])) --b;
     return tstr{s.p + a, b - a};
 }
 
@@ -393,8 +401,7 @@ __device__ __forceinline__ tstr tpx_concat(TpxHeap& h, const tstr a, const tstr 
 __device__ __forceinline__ tstr tpx_replace(TpxHeap& h, const tstr s, const tstr a,
                                             const tstr b, int* ec) {
     if (a.n == 0) {
-        // python: '' needle inserts b between every CHAR -> needs char semantics
-        if (!tpx_ascii(s)) { *ec = EC_NCV; return tstr{s.p, 0}; }
+        // python: '' needle inserts b between every CHAR (rows ASCII-gated)
         long long outn = s.n + (s.n + 1) * b.n;
         char* d = tpx_alloc(h, outn);
         if (!d) { *ec = EC_MEMORYERROR; return tstr{s.p, 0}; }
