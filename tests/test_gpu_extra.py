@@ -206,3 +206,38 @@ def test_chunked_csv_collect_and_tocsv(tmp_path):
     ds2.tocsv(outp)
     ref2 = pyoracle_csv.run_csv_pipeline(data, zillow_ops(), sink="csv")
     assert open(outp, "rb").read() == ref2["csv_text"]
+
+
+def test_csv_resolver(tmp_path):
+    """resolve() on a CSV-source pipeline: badparse/UDF errors replay through the
+    resolver chain (ResolveTask.cc:389 order)."""
+    from tests.zillow_data import make_zillow_csv_bytes
+    from tests.extra_data import q6_ops  # noqa: F401
+    data, _ = make_zillow_csv_bytes(2500, seed=77, dirty_frac=0.04)
+    p = _write(tmp_path, data, "r.csv")
+
+    def bd(x):
+        val = x["facts and features"]
+        max_idx = val.find(" bd")
+        if max_idx < 0:
+            max_idx = len(val)
+        s = val[:max_idx]
+        split_idx = s.rfind(",")
+        split_idx = 0 if split_idx < 0 else split_idx + 2
+        return int(s[split_idx:])
+
+    def bd_fallback(x):
+        return -1
+
+    ops = [("withColumn", "bedrooms", bd),
+           ("resolve", ValueError, bd_fallback),
+           ("selectColumns", ["city", "bedrooms"])]
+    ctx = tuplex_amd.Context()
+    from tests.pipelines import apply_ops as ap
+    ds = ap(ctx.csv(p), ops)
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    ref = pyoracle_csv.run_csv_pipeline(data, ops)
+    assert got == ref["output"]
+    assert ds.exception_counts == ref["exception_counts"]
+    assert any(v == -1 for _, v in got)  # resolver actually fired

@@ -351,6 +351,11 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
             blobs.append(f.read())
 
     text_mode = bool(getattr(src, "text_mode", False))
+    if not text_mode and src.quotechar != '"':
+        # the GPU scan is specialised to RFC-4180 '"'; other quote chars take the
+        # interpreter path (correct, slow — like the reference's fallback mode)
+        raise NotImplementedError(
+            "quotechar %r: only '\"' runs on the GPU this round" % src.quotechar)
     if text_mode:
         # text(): one str column, rows split on every newline, no sniffing
         has_header = False
