@@ -645,10 +645,10 @@ class StageCodegen:
         L.append("    long long* exc_buf, unsigned long long* exc_count,"
                  " unsigned long long exc_cap,")
         L.append("    void** outv) {")
-        L.append("  __shared__ char smem[128 * TPX_CELL_SLOT + 16];"
-                 "  // 128-thread blocks, 16 KiB + memcpy pad")
+        L.append("  __shared__ char smem[128 * TPX_CELL_STRIDE + 16];"
+                 "  // 128-thread blocks, padded stride (bank-conflict fix)")
         L.append("  TpxHeap heap{heap_base, heap_cursor, heap_cap, nullptr, nullptr};")
-        L.append("  char* lane_slot = smem + threadIdx.x * TPX_CELL_SLOT;")
+        L.append("  char* lane_slot = smem + threadIdx.x * TPX_CELL_STRIDE;")
         L.append("  long long stride = (long long)gridDim.x * blockDim.x;")
         L.append("  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;"
                  " i < n; i += stride) {")

@@ -200,6 +200,10 @@ __device__ __forceinline__ void tpx_memcpy(char* d, const char* s, long long n) 
 // cells are ~45% of row bytes on Zillow-like schemas). Cells that don't fit
 // keep their global view (correctness identical, speed only).
 #define TPX_CELL_SLOT 128
+// slot STRIDE is padded by one bank (4 B): a 128-B stride puts every lane's
+// same-relative-offset access on one bank (64-way conflict, §G4); 132 B gives
+// (33*l mod 32) = distinct banks per 32-lane group
+#define TPX_CELL_STRIDE 132
 
 __device__ __forceinline__ tstr tpx_cell_cache(tstr s, char* slot, int* cur) {
     if (s.n > 0 && (long long)*cur + s.n <= TPX_CELL_SLOT) {
