@@ -46,18 +46,23 @@ def main():
     names = list(ZILLOW_COLS)
 
     variants = [
-        ("passthrough", []),
+        ("passthrough", [], None),
         ("udf_nofilter", [op for op in zillow_ops()
-                          if op[0] in ("withColumn", "mapColumn")]),
-        ("full", zillow_ops()),
+                          if op[0] in ("withColumn", "mapColumn")], None),
+        ("full", zillow_ops(), None),
+        ("full_noLDS", zillow_ops(), 16),   # span cap 16B -> all-global path,
+                                            # tiny smem -> full occupancy
     ]
-    for label, ops in variants:
+    saved_cap = codegen.StageCodegen.SPAN_CAP
+    for label, ops, cap in variants:
+        codegen.StageCodegen.SPAN_CAP = cap if cap else saved_cap
         sp = plan.build_stage(col_types, names, ops)
         assert sp.compilable, (label, sp.why_not_compilable)
         src, desc = codegen.generate_stage(sp, source="csv", sink="mem",
                                            csv_info={"null_values": [""]})
         stage = glib.compile_stage(src, desc)
         run(stage, glib, dev, len(body), label)
+    codegen.StageCodegen.SPAN_CAP = saved_cap
 
 
 if __name__ == "__main__":

@@ -124,9 +124,6 @@ class StageCodegen:
         self.csv_info = csv_info or {}  # {"null_values": [...]}
         self.lits = {}  # python str -> lit var name
         self.lit_defs = []
-        self.input_str_vars = set()   # stage-input str column vars ("c3")
-        self.hot_inputs = set()       # subset consumed by UDF expressions
-        self.cached_cells = []        # input col indices staged into LDS slots
 
     # ---- literals -----------------------------------------------------------
     def lit(self, s: str) -> str:
@@ -154,9 +151,6 @@ class StageCodegen:
         if node["op"] not in self._NONSTRICT:
             argvals = tuple(self.emit_expr(em, a, row_ctx, opid)
                             for a in node["args"])
-            for v, _nv in argvals:  # mark UDF-scanned input cells as LDS-hot
-                if v in self.input_str_vars:
-                    self.hot_inputs.add(v)
             key = (node["op"], argvals, repr(node.get("v")), node.get("i"),
                    node.get("w"), repr(node["t"]))
             k_hit = em.lookup_key(key)
@@ -560,8 +554,6 @@ class StageCodegen:
 
         body = _Emitter()
         # inputs: c0..cN (+ c0_n null flags)
-        self.input_str_vars = {"c%d" % i for i, t in enumerate(in_types)
-                               if T.deopt(t) == T.STR}
         self.row_ctx_initial = []
         for i, t in enumerate(in_types):
             nv = ("c%d_n" % i) if T.is_opt(t) else None
@@ -626,14 +618,11 @@ class StageCodegen:
     # rows through a thrashed L1/L2 (measured 7x HBM read amplification:
     # 2048 threads x ~200 B rows >> 32 KiB L1 per CU). Waves whose span exceeds
     # TPX_SPAN_CAP fall back to parsing from global memory (rare: long rows).
+    SPAN_CAP = 16384  # bytes per wave; 4 waves/block -> 64 KiB LDS, 2 blocks/CU
+
     def _main_kernel(self, in_types, out_types):
-        # Thread-per-row grid-stride loop; only the UDF-scanned ("hot") string
-        # cells are staged into a 128-B per-thread LDS slot (tpx_cell_cache).
-        # Whole-row LDS staging capped occupancy at 10 waves/CU; hot cells are
-        # ~45% of Zillow-like row bytes, so this keeps the LDS locality for the
-        # re-scanned cells at ~2x the occupancy. Cold cells (pass-through
-        # columns) are read once from global by the write kernel.
         L = []
+        L.append("#define TPX_SPAN_CAP %d" % self.SPAN_CAP)
         L.append('extern "C" __global__ void tpx_stage_main(')
         L.append("    const unsigned char* __restrict__ in_data,")
         L.append("    const long long* __restrict__ in_offs,")
@@ -645,26 +634,52 @@ class StageCodegen:
         L.append("    long long* exc_buf, unsigned long long* exc_count,"
                  " unsigned long long exc_cap,")
         L.append("    void** outv) {")
-        L.append("  __shared__ char smem[128 * TPX_CELL_STRIDE + 16];"
-                 "  // 128-thread blocks, padded stride (bank-conflict fix)")
+        L.append("  __shared__ char smem[2 * TPX_SPAN_CAP + 16];  // 2 waves per 128-thread block -> 5 blocks/CU")
         L.append("  TpxHeap heap{heap_base, heap_cursor, heap_cap, nullptr, nullptr};")
-        L.append("  char* lane_slot = smem + threadIdx.x * TPX_CELL_STRIDE;")
-        L.append("  long long stride = (long long)gridDim.x * blockDim.x;")
-        L.append("  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;"
-                 " i < n; i += stride) {")
-        L.append("    int lcur = 0;  // per-row slot cursor")
-        self.cached_cells = []
-        L.extend(self._row_body(in_types, out_types))
+        L.append("  int lane = threadIdx.x & 63;")
+        L.append("  int wid = threadIdx.x >> 6;")
+        L.append("  char* wave_lds = smem + wid * TPX_SPAN_CAP;")
+        L.append("  long long wave_stride = (long long)gridDim.x * (blockDim.x >> 6);")
+        L.append("  long long nwaves = (n + 63) >> 6;")
+        L.append("  for (long long wb = (long long)blockIdx.x * (blockDim.x >> 6) + wid;"
+                 " wb < nwaves; wb += wave_stride) {")
+        L.append("    long long r0 = wb << 6;")
+        L.append("    long long rhi = r0 + 64 < n ? r0 + 64 : n;")
+        L.append("    long long span_start = in_offs[r0] & ~15LL;  // align staging window")
+        L.append("    long long span_end = in_offs[rhi];")
+        L.append("    long long span = span_end - span_start;")
+        # NB: never form (wave_lds - span_start): an LDS generic pointer offset
+        # outside the aperture is UB once the addrspace-inference pass narrows it
+        L.append("    bool staged = span <= TPX_SPAN_CAP;")
+        L.append("    if (staged) {")
+        L.append("      for (long long k = (long long)lane * 16; k < span; k += 64 * 16) {")
+        L.append("        if (k + 16 <= span)")
+        L.append("          *(uint4*)(wave_lds + k) = *(const uint4*)((const char*)in_data + span_start + k);")
+        L.append("        else")
+        L.append("          for (long long j = k; j < span; ++j)")
+        L.append("            wave_lds[j] = ((const char*)in_data)[span_start + j];")
+        L.append("      }")
+        L.append("    }")
+        L.append("    long long i = r0 + lane;")
+        L.append("    if (i >= rhi) continue;")
+        # duplicate the row body per pointer mode: in the staged branch every
+        # parse pointer provably derives from LDS, so address-space inference
+        # emits ds_read instead of flat loads
+        L.append("    if (staged) {")
+        L.extend("  " + ln for ln in self._row_body(in_types, out_types, lds=True))
+        L.append("    } else {")
+        L.extend("  " + ln for ln in self._row_body(in_types, out_types, lds=False))
+        L.append("    }")
         L.append("  }")
         L.append("}")
         return "\n".join(L)
 
-    def _row_body(self, in_types, out_types):
+    def _row_body(self, in_types, out_types, lds):
         L = []
         if self.source == "csv":
-            L.extend(self._load_inputs_csv(in_types))
+            L.extend(self._load_inputs_csv(in_types, lds))
         else:
-            L.extend(self._load_inputs_mem(in_types))
+            L.extend(self._load_inputs_mem(in_types, lds))
         L.append("    Out o;")
         args = []
         for idx, t in enumerate(in_types):
@@ -686,14 +701,18 @@ class StageCodegen:
         L.append("    }")
         L.append("    if (!o.keep) { keep[i] = 0; keep01[i] = 0; sizes[i] = 0; continue; }")
         L.append("    keep[i] = 1; keep01[i] = 1;")
-        L.extend(self._store_columnar(out_types))
+        L.extend(self._store_columnar(out_types, lds))
         return L
 
-    def _load_inputs_mem(self, in_types):
+    def _load_inputs_mem(self, in_types, lds=True):
         """Deserialize one reference-layout row (Serializer.cc:20-24) into typed
         locals c0..cN."""
         L = ["    long long prc = 0;  // no pre-parse errors on the mem path"]
-        L.append("    const unsigned char* row = in_data + in_offs[i];")
+        if lds:
+            L.append("    const unsigned char* row ="
+                     " (const unsigned char*)(wave_lds + (in_offs[i] - span_start));")
+        else:
+            L.append("    const unsigned char* row = in_data + in_offs[i];")
         n_opt = sum(1 for t in in_types if T.is_opt(t))
         bitmap_size = ((n_opt + 63) // 64) * 8 if n_opt else 0
         L.append("    // deserialize (bitmap %dB, %d slots)" % (bitmap_size, len(in_types)))
@@ -723,22 +742,17 @@ class StageCodegen:
                 # per-column ASCII gate (mem rows hold binary i64/f64 slots, so
                 # only the string bytes are scanned) — NCV divert, once per col
                 L.append("    if (!prc && !tpx_ascii(c%d)) prc = 7;" % idx)
-                if ("c%d" % idx) in self.hot_inputs:
-                    L.append("    const char* cg%d = c%d.p;" % (idx, idx))
-                    L.append("    c%d = tpx_cell_cache(c%d, lane_slot, &lcur);"
-                             % (idx, idx))
-                    self.cached_cells.append(idx)
             else:
                 raise CodegenError("input type %r" % (t,))
         return L
 
-    def _load_inputs_csv(self, in_types):
+    def _load_inputs_csv(self, in_types, lds=True):
         """Single-pass cell split + typed parse specialised to the sniffed schema —
         the CSVParseRowGenerator.cc replacement. Structure/parse failures set prc
         (BADPARSE/UNDERRUN/OVERRUN; raw line becomes the exception payload, like
         the reference's BADPARSE_STRING_INPUT rows)."""
         if self.csv_info.get("text_mode"):
-            return self._load_inputs_text(in_types)
+            return self._load_inputs_text(in_types, lds)
         nc = len(in_types)
         null_values = self.csv_info.get("null_values", [""])
         delim = self.csv_info.get("delimiter", ",")
@@ -746,8 +760,12 @@ class StageCodegen:
         delim_c = "'\\t'" if delim == "\t" else "'%s'" % delim
         L = []
         L.append("    long long prc = 0;")
-        L.append("    const char* rp = (const char*)in_data + in_offs[i];")
-        L.append("    const char* rend = (const char*)in_data + in_offs[i+1];")
+        if lds:
+            L.append("    const char* rp = wave_lds + (in_offs[i] - span_start);")
+            L.append("    const char* rend = wave_lds + (in_offs[i+1] - span_start);")
+        else:
+            L.append("    const char* rp = (const char*)in_data + in_offs[i];")
+            L.append("    const char* rend = (const char*)in_data + in_offs[i+1];")
         L.append("    if (rend > rp && rend[-1] == '\\n') --rend;")
         L.append("    if (rend > rp && rend[-1] == '\\r') --rend;")
         # straight-line per-cell scan: one named local per cell — a dynamically
@@ -780,11 +798,6 @@ class StageCodegen:
                 L.append("    if (prc) c%d = tstr{rp, 0};" % idx)
                 if opt:
                     L.append("    if (c%d_n) c%d = tstr{rp, 0};" % (idx, idx))
-                if ("c%d" % idx) in self.hot_inputs:
-                    L.append("    const char* cg%d = c%d.p;" % (idx, idx))
-                    L.append("    c%d = tpx_cell_cache(c%d, lane_slot, &lcur);"
-                             % (idx, idx))
-                    self.cached_cells.append(idx)
             elif base == T.I64:
                 L.append("    long long c%d = 0;" % idx)
                 L.append("    if (%s && tpx_cell_i64(cl%d, &c%d) != 0) prc = %d;"
@@ -807,23 +820,23 @@ class StageCodegen:
                  % 7)
         return L
 
-    def _load_inputs_text(self, in_types):
+    def _load_inputs_text(self, in_types, lds=True):
         """text() source: each row is the raw line (minus newline); no cell
         split, no quoting (Context::text, core/src/Context.cc)."""
         assert len(in_types) == 1
         t = in_types[0]
         null_values = self.csv_info.get("null_values", [])
         L = ["    long long prc = 0;"]
-        L.append("    const char* rp = (const char*)in_data + in_offs[i];")
-        L.append("    const char* rend = (const char*)in_data + in_offs[i+1];")
+        if lds:
+            L.append("    const char* rp = wave_lds + (in_offs[i] - span_start);")
+            L.append("    const char* rend = wave_lds + (in_offs[i+1] - span_start);")
+        else:
+            L.append("    const char* rp = (const char*)in_data + in_offs[i];")
+            L.append("    const char* rend = (const char*)in_data + in_offs[i+1];")
         L.append("    if (rend > rp && rend[-1] == '\\n') --rend;")
         L.append("    if (rend > rp && rend[-1] == '\\r') --rend;")
         L.append("    tstr c0{rp, rend - rp};")
         L.append("    if (!tpx_ascii(c0)) prc = %d;  // row ASCII gate (NCV)" % 7)
-        if "c0" in self.hot_inputs:
-            L.append("    const char* cg0 = c0.p;")
-            L.append("    c0 = tpx_cell_cache(c0, lane_slot, &lcur);")
-            self.cached_cells.append(0)
         if T.is_opt(t):
             checks = " || ".join("tpx_streq(c0, %s)" % self.lit(nv)
                                  for nv in null_values) or "false"
@@ -831,17 +844,17 @@ class StageCodegen:
             L.append("    if (c0_n) c0 = tstr{rp, 0};")
         return L
 
-    def _store_columnar(self, out_types):
+    def _store_columnar(self, out_types, lds=True):
         """Store Out o -> columnar arrays + per-row serialized size (mem sink) or
-        csv text size (csv sink). String views derived from LDS-cached cells are
-        mapped back to their global addresses (tpx_cellfix)."""
+        csv text size (csv sink)."""
         L = []
         for k, t in enumerate(out_types):
             base = T.deopt(t)
             if base == T.STR:
-                for ci in self.cached_cells:
-                    L.append("    o.o%d = tpx_cellfix(o.o%d, c%d.p, cg%d, c%d.n);"
-                             % (k, k, ci, ci, ci))
+                if lds:
+                    L.append("    o.o%d = tpx_to_global(o.o%d, wave_lds,"
+                             " wave_lds + TPX_SPAN_CAP, in_data, span_start);"
+                             % (k, k))
                 L.append("    ((unsigned long long*)outv[%d])[i] = (unsigned long long)o.o%d.p;"
                          % (3 * k, k))
                 L.append("    ((int*)outv[%d])[i] = (int)o.o%d.n;" % (3 * k + 1, k))

@@ -458,9 +458,10 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
     unsigned long long exc_count = 0;
     for (int attempt = 0;; ++attempt) {
         HIP_CHECK(hipMemsetAsync(P.counters, 0, 16, stream));
-        // main: 128-thread blocks, 16 KiB LDS (per-thread hot-cell slots)
-        // -> 10 blocks/CU = 20 waves/CU; thread-per-row grid-stride
-        unsigned grid = (unsigned)std::min<long long>((n + 127) / 128, 2560);
+        // main runs 128-thread blocks (2 waves x 16 KiB LDS staging = 32 KiB
+        // -> 5 blocks/CU = 10 waves/CU); each wave owns 64 rows
+        long long waves = (n + 63) / 64;
+        unsigned grid = (unsigned)std::min<long long>((waves + 1) / 2, 4096);
         void* args[] = {&d_in, &d_offs, &n, &row0, &P.heap, &d_heap_cursor,
                         &heap_cap, &d_keep, &d_keep01, &d_sizes, &P.exc,
                         &d_exc_count, &exc_cap, &d_outv};

@@ -194,35 +194,6 @@ __device__ __forceinline__ void tpx_memcpy(char* d, const char* s, long long n) 
     for (; i < n; ++i) d[i] = s[i];
 }
 
-// ---- per-lane UDF-cell cache ---------------------------------------------------
-// Only the string cells the UDF chain actually SCANS are staged into a small
-// per-lane LDS slot (whole-row staging capped occupancy at 10 waves/CU; hot
-// cells are ~45% of row bytes on Zillow-like schemas). Cells that don't fit
-// keep their global view (correctness identical, speed only).
-#define TPX_CELL_SLOT 128
-// slot STRIDE is padded by one bank (4 B): a 128-B stride puts every lane's
-// same-relative-offset access on one bank (64-way conflict, §G4); 132 B gives
-// (33*l mod 32) = distinct banks per 32-lane group
-#define TPX_CELL_STRIDE 132
-
-__device__ __forceinline__ tstr tpx_cell_cache(tstr s, char* slot, int* cur) {
-    if (s.n > 0 && (long long)*cur + s.n <= TPX_CELL_SLOT) {
-        char* d = slot + *cur;
-        tpx_memcpy(d, s.p, s.n);  // alignment-safe; src buffers carry tail pad
-        *cur += (int)((s.n + 7) & ~7LL);
-        return tstr{d, s.n};
-    }
-    return s;
-}
-
-// map a view derived from a cached cell back to its global address
-__device__ __forceinline__ tstr tpx_cellfix(tstr s, const char* lp,
-                                            const char* gp, long long n) {
-    if (s.p >= lp && s.p <= lp + n)
-        return tstr{gp + (s.p - lp), s.n};
-    return s;
-}
-
 // translate an LDS-staged string view back to its global-memory address before it
 // escapes the kernel (columnar string cells must outlive the LDS staging window).
 // (p - lds_lo) + span_start is the byte offset in the input.
