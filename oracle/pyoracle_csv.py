@@ -287,7 +287,7 @@ def _replay_row(raw_line: bytes, col_types, null_values, row_ops, columns,
 
 def run_csv_pipeline(csv_bytes: bytes, ops, columns=None, header=None,
                      null_values=None, threshold=0.9, sink="collect",
-                     delimiter=None):
+                     delimiter=None, type_hints=None):
     """Full oracle CSV pipeline. sink='collect' -> values; 'csv' -> output text
     (bytes, incl. header line)."""
     if null_values is None:
@@ -299,6 +299,10 @@ def run_csv_pipeline(csv_bytes: bytes, ops, columns=None, header=None,
     delim = delimiter.encode() if delimiter else sniff_delimiter(sample)
     has_header, names, col_types = sniff(sample, null_values, threshold, header,
                                          columns, delim)
+    if type_hints:  # string lattice forms: "i64" / "opt_i64" / ...
+        for k, h in type_hints.items():
+            i = names.index(k) if isinstance(k, str) else int(k)
+            col_types[i] = ("opt", h[4:]) if h.startswith("opt_") else h
     data = csv_bytes
     if has_header:
         p = data.find(b"\n")

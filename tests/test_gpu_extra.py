@@ -241,3 +241,29 @@ def test_csv_resolver(tmp_path):
     assert got == ref["output"]
     assert ds.exception_counts == ref["exception_counts"]
     assert any(v == -1 for _, v in got)  # resolver actually fired
+
+
+def test_type_hints_and_custom_nulls(tmp_path):
+    """type_hints override sniffing; custom null_values make numeric columns
+    Option-typed with None semantics (context.py:288/:322)."""
+    from typing import Optional
+    lines = [b"a,b,c"] + [b"%d,NULL,x%d" % (i, i) if i % 5 == 0
+                          else b"%d,%d,x%d" % (i, i * 2, i)
+                          for i in range(2000)]
+    data = b"\n".join(lines) + b"\n"
+    p = _write(tmp_path, data, "h.csv")
+
+    def use(x):
+        return (x["a"], str(x["b"]), x["c"])
+
+    ctx = tuplex_amd.Context()
+    ds = ctx.csv(p, null_values=["NULL"],
+                 type_hints={"b": Optional[int]}).map(use)
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    ref = pyoracle_csv.run_csv_pipeline(data, [("map", use)],
+                                        null_values=["NULL"],
+                                        type_hints={"b": "opt_i64"})
+    assert got == ref["output"]
+    assert ds.exception_counts == ref["exception_counts"]
+    assert sum(1 for _, b, _c in got if b == "None") == 400

@@ -156,6 +156,36 @@ def _try_bool(s: str):
     return s.strip(_PYWS).lower() in _BOOL_STRS
 
 
+def apply_type_hints(col_types, names, hints):
+    """context.csv type_hints (context.py:288): per-column overrides keyed by
+    index or name; python types / typing.Optional map onto the lattice."""
+    if not hints:
+        return col_types
+    import typing
+
+    def to_t(h):
+        if h is int:
+            return T.I64
+        if h is float:
+            return T.F64
+        if h is bool:
+            return T.BOOL
+        if h is str:
+            return T.STR
+        origin = typing.get_origin(h)
+        if origin is typing.Union:
+            args = [a for a in typing.get_args(h) if a is not type(None)]
+            if len(args) == 1:
+                return T.opt(to_t(args[0]))
+        raise ValueError("unsupported type hint %r" % (h,))
+
+    out = list(col_types)
+    for k, h in hints.items():
+        i = names.index(k) if isinstance(k, str) else int(k)
+        out[i] = to_t(h)
+    return out
+
+
 def sniff_delimiter(sample: bytes) -> bytes:
     """Most frequent candidate separator in the first row, quotes respected
     (ContextOptions.cc csv.separators [',', ';', '|', '\\t'])."""
@@ -374,6 +404,8 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
         has_header, names, col_types = sniff(sample, src.null_values,
                                              opts.normalcase_threshold,
                                              src.header, src.columns, delim)
+        col_types = apply_type_hints(col_types, names,
+                                     getattr(src, "type_hints", None))
 
     # assemble the GPU input: concat files, strip header lines, ensure trailing \n
     parts = []
