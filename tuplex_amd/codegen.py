@@ -117,13 +117,22 @@ class StageCodegen:
     """Emits the full stage source. `row_ctx` maps TIR input index -> (var, type,
     nullvar or None) for the operator currently being emitted."""
 
-    def __init__(self, sp, source="mem", sink="mem", csv_info=None):
+    def __init__(self, sp, source="mem", sink="mem", csv_info=None,
+                 fusion_groups=None):
         self.sp = sp
         self.source = source
         self.sink = sink
         self.csv_info = csv_info or {}  # {"null_values": [...]}
         self.lits = {}  # python str -> lit var name
         self.lit_defs = []
+        # multi-needle scan fusion: pass 1 registers every const-needle
+        # find/contains per structurally-identified haystack; pass 2 emits ONE
+        # SWAR window scan per group (the serial dependent-load chains of
+        # repeated scans over the same short cell dominate the UDF phase —
+        # profiles/README.md ablation)
+        self.scan_registry = {}   # struct_key(haystack) -> [(op, needle), ...]
+        self.fusion_groups = fusion_groups  # None = pass 1 (collect)
+        self.fused_done = set()   # (scope-id tuple unnecessary; see emit site)
 
     # ---- literals -----------------------------------------------------------
     def lit(self, s: str) -> str:
@@ -134,6 +143,16 @@ class StageCodegen:
             self.lit_defs.append(
                 '__device__ const char %s[%d] = "%s";' % (name, len(b) + 1, _esc(s)))
         return "tstr{%s, %d}" % (self.lits[s], len(s.encode("utf-8")))
+
+    @staticmethod
+    def _struct_key(node):
+        """Var-name-independent identity of a TIR subtree (mirrors CSE
+        equivalence: same structure -> same emitted var)."""
+        out = [node["op"], repr(node.get("v")), node.get("i"), node.get("w"),
+               repr(node["t"])]
+        for a in node["args"]:
+            out.append(StageCodegen._struct_key(a))
+        return tuple(out)
 
     # ---- TIR expression emission -------------------------------------------
     # ops whose arguments must NOT be evaluated eagerly (control flow) or that
@@ -306,6 +325,12 @@ class StageCodegen:
             self._check(em, opid)
             return (r, None)
         if op in ("strfind", "strrfind"):
+            if op == "strfind" and a[1]["op"] == "const" and isinstance(
+                    a[1].get("v"), str) and a[1]["v"]:
+                self._register_scan(a[0], "strfind", a[1]["v"])
+                f = self._try_fuse(em, n, rc, opid)
+                if f is not None:
+                    return f
             x, _ = ev(a[0])
             y, _ = ev(a[1])
             r = em.fresh()
@@ -334,6 +359,12 @@ class StageCodegen:
             self._check(em, opid)
             return (r, None)
         if op in ("startswith", "endswith", "contains"):
+            if op == "contains" and a[1]["op"] == "const" and isinstance(
+                    a[1].get("v"), str) and a[1]["v"]:
+                self._register_scan(a[0], "contains", a[1]["v"])
+                f = self._try_fuse(em, n, rc, opid)
+                if f is not None:
+                    return f
             x, _ = ev(a[0])
             y, _ = ev(a[1])
             r = em.fresh("b")
@@ -419,6 +450,91 @@ class StageCodegen:
         if op == "mktuple":
             raise CodegenError("mktuple must be handled at operator level")
         raise CodegenError("unsupported TIR op %r" % op)
+
+    def _register_scan(self, hay_node, op, needle):
+        k = self._struct_key(hay_node)
+        lst = self.scan_registry.setdefault(k, [])
+        if (op, needle) not in lst:
+            lst.append((op, needle))
+
+    def _try_fuse(self, em, node, rc, opid):
+        """If this find/contains belongs to a fusion group (pass 2), emit the
+        whole group's fused SWAR scan at this point and memoize every member's
+        CSE key; returns this node's (var, None) or None (no fusion)."""
+        if not self.fusion_groups:
+            return None
+        hay = node["args"][0]
+        gk = self._struct_key(hay)
+        members = self.fusion_groups.get(gk)
+        if not members or len(members) < 2:
+            return None
+        hv, hn = self.emit_expr(em, hay, rc, opid)
+        # already fused in a visible scope? -> the CSE key lookup in emit_expr
+        # would have hit; reaching here means not yet emitted in-scope.
+        out = {}
+        tag = em.fresh("fg")
+        # member outputs
+        for mi, (mop, needle) in enumerate(members):
+            if mop == "strfind":
+                em.w("long long %s_%d = -1;" % (tag, mi))
+            else:
+                em.w("bool %s_%d = false;" % (tag, mi))
+        fcs = sorted({needle[0] for _, needle in members})
+        em.w("{  // fused multi-needle scan over %s (%d needles)"
+             % (hv, len(members)))
+        em.w("  const char* _p = %s.p; long long _n = %s.n;" % (hv, hv))
+        em.w("  unsigned _pend = %du;" % ((1 << len(members)) - 1))
+        em.w("  long long _i = 0;")
+
+        def checks(pos, indent):
+            for mi, (mop, needle) in enumerate(members):
+                nb = needle.encode()
+                cond = ["(_pend & %du)" % (1 << mi),
+                        "%s + %d <= _n" % (pos, len(nb)),
+                        "_ch == %d" % nb[0]]
+                for j, byte in enumerate(nb[1:], start=1):
+                    cond.append("_p[%s + %d] == %d" % (pos, j, byte))
+                body = ("%s_%d = %s; _pend &= ~%du;"
+                        % (tag, mi, pos if mop == "strfind" else "true",
+                           1 << mi))
+                em.w("%sif (%s) { %s }" % (indent, " && ".join(cond), body))
+
+        em.w("  while (_i < _n && _pend &&"
+             " (((unsigned long long)(_p + _i)) & 7)) {")
+        em.w("    int _ch = (unsigned char)_p[_i];")
+        checks("_i", "    ")
+        em.w("    ++_i;")
+        em.w("  }")
+        em.w("  for (; _i + 8 <= _n && _pend; _i += 8) {")
+        em.w("    unsigned long long _v = *(const unsigned long long*)(_p + _i);")
+        hits = " | ".join("tpx_swar_zero(_v ^ (TPX_SWAR_ONE * %du))" % ord(c)
+                          for c in fcs)
+        em.w("    unsigned long long _hit = %s;" % hits)
+        em.w("    if (_hit) {")
+        em.w("      for (int _b = 0; _b < 8; ++_b) {")
+        em.w("        long long _j = _i + _b;")
+        em.w("        int _ch = (unsigned char)_p[_j];")
+        checks("_j", "        ")
+        em.w("      }")
+        em.w("    }")
+        em.w("  }")
+        em.w("  for (; _i < _n && _pend; ++_i) {")
+        em.w("    int _ch = (unsigned char)_p[_i];")
+        checks("_i", "    ")
+        em.w("  }")
+        em.w("}")
+        # memoize every member under its CSE key (the exact shape emit_expr uses)
+        my_var = None
+        for mi, (mop, needle) in enumerate(members):
+            lit_expr = self.lit(needle)
+            argvals = ((hv, hn), (lit_expr, None))
+            t = T.I64 if mop == "strfind" else T.BOOL
+            key = (mop, argvals, repr(None), None, None, repr(t))
+            em.memo_key(key, ("%s_%d" % (tag, mi), None))
+            if (mop == node["op"] and needle == node["args"][1]["v"]):
+                my_var = ("%s_%d" % (tag, mi), None)
+        out = my_var
+        return out
 
     def _num(self, x, node, outt):
         if outt == T.F64 and T.deopt(node["t"]) in (T.I64, T.BOOL):
@@ -1051,4 +1167,10 @@ class StageCodegen:
 
 
 def generate_stage(sp, source="mem", sink="mem", csv_info=None):
-    return StageCodegen(sp, source, sink, csv_info).generate()
+    # pass 1: collect const-needle find/contains groups per haystack
+    cg1 = StageCodegen(sp, source, sink, csv_info)
+    cg1.generate()
+    groups = {k: v for k, v in cg1.scan_registry.items() if len(v) >= 2}
+    # pass 2: fresh instance (lits/emitter state are single-use)
+    return StageCodegen(sp, source, sink, csv_info,
+                        fusion_groups=groups or None).generate()
