@@ -263,7 +263,7 @@ int main(int argc, char** argv) {
     struct timespec t0, t1;
     clock_gettime(CLOCK_MONOTONIC, &t0);
     double elapsed = 0;
-    while (elapsed < min_seconds) {
+    do {  // at least one pass, then until min_seconds of work
         const char* p = body;
         const char* end = body + body_n;
         char* w = outbuf;
@@ -338,7 +338,7 @@ int main(int argc, char** argv) {
         ++passes;
         clock_gettime(CLOCK_MONOTONIC, &t1);
         elapsed = (t1.tv_sec - t0.tv_sec) + 1e-9 * (t1.tv_nsec - t0.tv_nsec);
-    }
+    } while (elapsed < min_seconds);
 
     printf("{\"rows\": %lld, \"rows_out\": %lld, \"exceptions\": %lld, "
            "\"bytes_in\": %lld, \"bytes_out\": %lld, \"seconds\": %.6f, "
