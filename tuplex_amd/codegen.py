@@ -130,7 +130,7 @@ class StageCodegen:
         # SWAR window scan per group (the serial dependent-load chains of
         # repeated scans over the same short cell dominate the UDF phase —
         # profiles/README.md ablation)
-        self.scan_registry = {}   # struct_key(haystack) -> [(op, needle), ...]
+        self.scan_registry = {}   # struct_key(haystack) -> [(op, needle, ci)..]
         self.fusion_groups = fusion_groups  # None = pass 1 (collect)
         self.fused_done = set()   # (scope-id tuple unnecessary; see emit site)
 
@@ -157,7 +157,11 @@ class StageCodegen:
     # ---- TIR expression emission -------------------------------------------
     # ops whose arguments must NOT be evaluated eagerly (control flow) or that
     # need no CSE
-    _NONSTRICT = {"if", "and", "or", "const", "input", "mktuple"}
+    # strfind/contains are non-strict so that contains(lower(s), lit) can be
+    # rewritten to a case-insensitive scan over s WITHOUT evaluating lower(s)
+    # (their CSE runs through structural scan keys in _scan_node instead)
+    _NONSTRICT = {"if", "and", "or", "const", "input", "mktuple",
+                  "strfind", "contains"}
 
     def emit_expr(self, em: _Emitter, node, row_ctx, opid) -> Tuple[str, Optional[str]]:
         """Returns (value_var, null_var or None). Structural CSE across operators:
@@ -327,8 +331,7 @@ class StageCodegen:
         if op in ("strfind", "strrfind"):
             if op == "strfind" and a[1]["op"] == "const" and isinstance(
                     a[1].get("v"), str) and a[1]["v"]:
-                self._register_scan(a[0], "strfind", a[1]["v"])
-                f = self._try_fuse(em, n, rc, opid)
+                f = self._scan_node(em, n, rc, opid)
                 if f is not None:
                     return f
             x, _ = ev(a[0])
@@ -361,8 +364,7 @@ class StageCodegen:
         if op in ("startswith", "endswith", "contains"):
             if op == "contains" and a[1]["op"] == "const" and isinstance(
                     a[1].get("v"), str) and a[1]["v"]:
-                self._register_scan(a[0], "contains", a[1]["v"])
-                f = self._try_fuse(em, n, rc, opid)
+                f = self._scan_node(em, n, rc, opid)
                 if f is not None:
                     return f
             x, _ = ev(a[0])
@@ -451,49 +453,93 @@ class StageCodegen:
             raise CodegenError("mktuple must be handled at operator level")
         raise CodegenError("unsupported TIR op %r" % op)
 
-    def _register_scan(self, hay_node, op, needle):
-        k = self._struct_key(hay_node)
-        lst = self.scan_registry.setdefault(k, [])
-        if (op, needle) not in lst:
-            lst.append((op, needle))
-
-    def _try_fuse(self, em, node, rc, opid):
-        """If this find/contains belongs to a fusion group (pass 2), emit the
-        whole group's fused SWAR scan at this point and memoize every member's
-        CSE key; returns this node's (var, None) or None (no fusion)."""
-        if not self.fusion_groups:
-            return None
+    @staticmethod
+    def _scan_shape(node):
+        """(inner_haystack_node, needle, ci) for a const-needle find/contains.
+        contains(lower(s), lit) / find(lower(s), lit) are rewritten to a
+        case-insensitive scan directly over s: rows are ASCII-gated, so
+        tolower is byte-local and length-preserving (indices coincide), and
+        the lower() allocation+copy is skipped entirely (and becomes lazy —
+        only branches that still reference it evaluate it)."""
         hay = node["args"][0]
-        gk = self._struct_key(hay)
-        members = self.fusion_groups.get(gk)
-        if not members or len(members) < 2:
-            return None
-        hv, hn = self.emit_expr(em, hay, rc, opid)
-        # already fused in a visible scope? -> the CSE key lookup in emit_expr
-        # would have hit; reaching here means not yet emitted in-scope.
-        out = {}
+        needle = node["args"][1]["v"]
+        ci = False
+        while hay["op"] == "lower":
+            hay = hay["args"][0]
+            ci = True
+        return hay, needle, ci
+
+    def _register_scan(self, node):
+        inner, needle, ci = self._scan_shape(node)
+        k = self._struct_key(inner)
+        lst = self.scan_registry.setdefault(k, [])
+        if (node["op"], needle, ci) not in lst:
+            lst.append((node["op"], needle, ci))
+
+    def _scan_node(self, em, node, rc, opid):
+        """Emit a const-needle find/contains through the scan machinery:
+        structural CSE (scan keys), case-insensitive rewrite, multi-needle
+        fusion when pass 2 knows the group. Always returns (var, None)."""
+        inner, needle, ci = self._scan_shape(node)
+        if not all(ord(c) < 128 for c in needle):
+            return None  # scan emitter is byte-indexed; plain path handles it
+        self._register_scan(node)
+        mk = ("scanm", node["op"], needle, ci, self._struct_key(inner))
+        hit = em.lookup_key(mk)
+        if hit is not None:
+            return hit
+        if ci and any("A" <= c <= "Z" for c in needle):
+            # lower(s) never contains an uppercase letter: constant result
+            v = ("-1LL", None) if node["op"] == "strfind" else ("false", None)
+            return em.memo_key(mk, v)
+        gk = self._struct_key(inner)
+        members = list((self.fusion_groups or {}).get(gk) or [])
+        members = [m for m in members
+                   if not (m[2] and any("A" <= c <= "Z" for c in m[1]))]
+        if (node["op"], needle, ci) not in members:
+            members.append((node["op"], needle, ci))
+        return self._emit_scan_group(em, inner, members, rc, opid,
+                                     (node["op"], needle, ci))
+
+    def _emit_scan_group(self, em, inner, members, rc, opid, want):
+        """ONE SWAR window scan over `inner` answering every (op, needle, ci)
+        in `members`; memoizes each under its scan key; returns `want`'s var.
+        ci member byte compare: needle is all non-uppercase; a letter byte b
+        matches c iff (c | 0x20) == b (exactly {b, toupper(b)} for ASCII);
+        non-letter bytes compare exact."""
+        hv, hn = self.emit_expr(em, inner, rc, opid)
         tag = em.fresh("fg")
-        # member outputs
-        for mi, (mop, needle) in enumerate(members):
+        for mi, (mop, needle, ci) in enumerate(members):
             if mop == "strfind":
                 em.w("long long %s_%d = -1;" % (tag, mi))
             else:
                 em.w("bool %s_%d = false;" % (tag, mi))
-        fcs = sorted({needle[0] for _, needle in members})
-        em.w("{  // fused multi-needle scan over %s (%d needles)"
-             % (hv, len(members)))
+        fcs = set()
+        for mop, needle, ci in members:
+            fcs.add(needle[0])
+            if ci and "a" <= needle[0] <= "z":
+                fcs.add(needle[0].upper())
+        fcs = sorted(fcs)
+        em.w("{  // fused scan over %s (%d needles)" % (hv, len(members)))
         em.w("  const char* _p = %s.p; long long _n = %s.n;" % (hv, hv))
         em.w("  unsigned _pend = %du;" % ((1 << len(members)) - 1))
         em.w("  long long _i = 0;")
 
+        def bcmp(expr, byte_ch, ci):
+            b = ord(byte_ch)
+            if ci and "a" <= byte_ch <= "z":
+                return "(%s | 0x20) == %d" % (expr, b)
+            return "%s == %d" % (expr, b)
+
         def checks(pos, indent):
-            for mi, (mop, needle) in enumerate(members):
-                nb = needle.encode()
+            for mi, (mop, needle, ci) in enumerate(members):
+                nb = needle
                 cond = ["(_pend & %du)" % (1 << mi),
-                        "%s + %d <= _n" % (pos, len(nb)),
-                        "_ch == %d" % nb[0]]
-                for j, byte in enumerate(nb[1:], start=1):
-                    cond.append("_p[%s + %d] == %d" % (pos, j, byte))
+                        "%s + %d <= _n" % (pos, len(nb.encode())),
+                        bcmp("_ch", nb[0], ci)]
+                for j, c in enumerate(nb[1:], start=1):
+                    cond.append(bcmp("(unsigned char)_p[%s + %d]" % (pos, j),
+                                     c, ci))
                 body = ("%s_%d = %s; _pend &= ~%du;"
                         % (tag, mi, pos if mop == "strfind" else "true",
                            1 << mi))
@@ -523,18 +569,14 @@ class StageCodegen:
         checks("_i", "    ")
         em.w("  }")
         em.w("}")
-        # memoize every member under its CSE key (the exact shape emit_expr uses)
         my_var = None
-        for mi, (mop, needle) in enumerate(members):
-            lit_expr = self.lit(needle)
-            argvals = ((hv, hn), (lit_expr, None))
-            t = T.I64 if mop == "strfind" else T.BOOL
-            key = (mop, argvals, repr(None), None, None, repr(t))
+        ik = self._struct_key(inner)
+        for mi, (mop, needle, ci) in enumerate(members):
+            key = ("scanm", mop, needle, ci, ik)
             em.memo_key(key, ("%s_%d" % (tag, mi), None))
-            if (mop == node["op"] and needle == node["args"][1]["v"]):
+            if (mop, needle, ci) == want:
                 my_var = ("%s_%d" % (tag, mi), None)
-        out = my_var
-        return out
+        return my_var
 
     def _num(self, x, node, outt):
         if outt == T.F64 and T.deopt(node["t"]) in (T.I64, T.BOOL):
