@@ -172,6 +172,9 @@ class StageCodegen:
         if hit is not None:
             return hit
         if node["op"] not in self._NONSTRICT:
+            pv = self._peephole(em, node, row_ctx, opid)
+            if pv is not None:
+                return em.memo(node, pv)
             argvals = tuple(self.emit_expr(em, a, row_ctx, opid)
                             for a in node["args"])
             key = (node["op"], argvals, repr(node.get("v")), node.get("i"),
@@ -184,6 +187,39 @@ class StageCodegen:
             return em.memo(node, v)
         v = self._emit(em, node, row_ctx, opid)
         return em.memo(node, v)
+
+    def _peephole(self, em, n, rc, opid):
+        """Pattern rewrites that must run BEFORE strict arg evaluation.
+        capitalize: x[0].upper() + x[1:].lower() -> tpx_capitalize_ix(x)
+        (one alloc + one pass instead of three; IndexError on empty x kept)."""
+        if n["op"] != "concat":
+            return None
+        a0, a1 = n["args"]
+        if not (a0["op"] == "upper" and a1["op"] == "lower"):
+            return None
+        g, sl = a0["args"][0], a1["args"][0]
+        if g["op"] != "getitem" or sl["op"] != "slice":
+            return None
+
+        def int_const(nd, val):
+            return (nd["op"] == "const" and not isinstance(nd.get("v"), bool)
+                    and nd.get("v") == val)
+
+        gi, lo, hi = g["args"][1], sl["args"][1], sl["args"][2]
+        if not (int_const(gi, 0) and int_const(lo, 1)
+                and hi["op"] == "const" and hi.get("v") is None):
+            return None
+        if self._struct_key(g["args"][0]) != self._struct_key(sl["args"][0]):
+            return None
+        key = ("capitalize", self._struct_key(g["args"][0]))
+        hit = em.lookup_key(key)
+        if hit is not None:
+            return hit
+        x, _ = self.emit_expr(em, g["args"][0], rc, opid)
+        r = em.fresh("s")
+        em.w("tstr %s = tpx_capitalize_ix(heap, %s, &_ec);" % (r, x))
+        self._check(em, opid)
+        return em.memo_key(key, (r, None))
 
     def _raise(self, em, ec_expr, opid):
         em.w("return ((long long)(%s)) | ((long long)%d << 32);" % (ec_expr, opid))
@@ -1077,7 +1113,8 @@ class StageCodegen:
         L.append("    long long n, long long row0, void** outv,")
         L.append("    unsigned char* __restrict__ out_data, long long* __restrict__ out_offs,")
         L.append("    long long* __restrict__ out_rowidx,")
-        L.append("    long long total_rows, long long total_bytes) {")
+        L.append("    long long total_rows, long long total_bytes,")
+        L.append("    const long long* __restrict__ kept_idx) {  // unused (mem sink)")
         L.append("  long long stride = (long long)gridDim.x * blockDim.x;")
         L.append("  long long tid0 = (long long)blockIdx.x * blockDim.x + threadIdx.x;")
         L.append("  if (tid0 == 0) {")
@@ -1143,8 +1180,17 @@ class StageCodegen:
         L.append("}")
         return "\n".join(L)
 
+    # CSV write: one lane per KEPT row (kept_idx from tpx_emit_kept), 64 kept
+    # rows per wave -> contiguous output span; format into LDS, then copy the
+    # span out wave-cooperatively with 8B chunks (the old one-thread-per-input
+    # -row version wrote ~31 B rows byte-by-byte to scattered global addresses
+    # with (1 - selectivity) of the lanes idle). Spans > TPX_WCAP (long rows)
+    # fall back to direct global formatting for that wave.
+    WRITE_CAP = 8192  # bytes per wave; 2 waves/block -> 16 KiB LDS
+
     def _write_kernel_csv(self, out_types):
         L = []
+        L.append("#define TPX_WCAP %d" % self.WRITE_CAP)
         L.append('extern "C" __global__ void tpx_stage_write(')
         L.append("    const unsigned char* __restrict__ keep,")
         L.append("    const long long* __restrict__ keep_scan,")
@@ -1152,15 +1198,33 @@ class StageCodegen:
         L.append("    long long n, long long row0, void** outv,")
         L.append("    unsigned char* __restrict__ out_data, long long* __restrict__ out_offs,")
         L.append("    long long* __restrict__ out_rowidx,")
-        L.append("    long long total_rows, long long total_bytes) {")
-        L.append("  long long stride = (long long)gridDim.x * blockDim.x;")
-        L.append("  long long tid0 = (long long)blockIdx.x * blockDim.x + threadIdx.x;")
-        L.append("  if (tid0 == 0) out_offs[total_rows] = total_bytes;")
-        L.append("  for (long long i = tid0; i < n; i += stride) {")
-        L.append("    if (!keep[i]) continue;")
-        L.append("    char* w = (char*)out_data + size_scan[i];")
-        L.append("    out_offs[keep_scan[i]] = size_scan[i];")
-        L.append("    out_rowidx[keep_scan[i]] = row0 + i;")
+        L.append("    long long total_rows, long long total_bytes,")
+        L.append("    const long long* __restrict__ kept_idx) {")
+        L.append("  __shared__ char wsmem[2 * TPX_WCAP + 16];  // 128-thread blocks")
+        L.append("  int lane = threadIdx.x & 63;")
+        L.append("  int wid = threadIdx.x >> 6;")
+        L.append("  char* wave_lds = wsmem + wid * TPX_WCAP;")
+        L.append("  if (blockIdx.x == 0 && threadIdx.x == 0)"
+                 " out_offs[total_rows] = total_bytes;")
+        L.append("  long long wave_stride = (long long)gridDim.x * (blockDim.x >> 6);")
+        L.append("  long long nwaves = (total_rows + 63) >> 6;")
+        L.append("  for (long long wb = (long long)blockIdx.x * (blockDim.x >> 6) + wid;"
+                 " wb < nwaves; wb += wave_stride) {")
+        L.append("    long long k0 = wb << 6;")
+        L.append("    long long khi = k0 + 64 < total_rows ? k0 + 64 : total_rows;")
+        L.append("    long long span_start = size_scan[kept_idx[k0]];")
+        L.append("    long long span_end = khi < total_rows ?"
+                 " size_scan[kept_idx[khi]] : total_bytes;")
+        L.append("    long long span = span_end - span_start;")
+        L.append("    bool staged = span <= TPX_WCAP;")
+        L.append("    long long k = k0 + lane;")
+        L.append("    if (k < khi) {")
+        L.append("    long long i = kept_idx[k];")
+        L.append("    long long my_start = size_scan[i];")
+        L.append("    char* w = staged ? (wave_lds + (my_start - span_start))"
+                 " : ((char*)out_data + my_start);")
+        L.append("    out_offs[k] = my_start;")
+        L.append("    out_rowidx[k] = row0 + i;")
         for k, t in enumerate(out_types):
             base = T.deopt(t)
             if k:
@@ -1185,6 +1249,16 @@ class StageCodegen:
             else:
                 raise CodegenError("csv sink for %r" % (t,))
         L.append("    *w++ = '\\n';")
+        L.append("    }")
+        L.append("    if (staged) {")
+        L.append("      __builtin_amdgcn_wave_barrier();")
+        L.append("      char* dst = (char*)out_data + span_start;")
+        L.append("      for (long long b = (long long)lane * 8; b + 8 <= span;"
+                 " b += 64 * 8)")
+        L.append("        __builtin_memcpy(dst + b, wave_lds + b, 8);")
+        L.append("      for (long long b = (span & ~7LL) + lane; b < span; b += 64)")
+        L.append("        dst[b] = wave_lds[b];")
+        L.append("    }")
         L.append("  }")
         L.append("}")
         return "\n".join(L)

@@ -134,6 +134,7 @@ struct tpx_stage {
     hipModule_t module = nullptr;
     hipFunction_t k_main = nullptr, k_write = nullptr;
     hipFunction_t k_scan_block = nullptr, k_scan_add = nullptr;
+    hipFunction_t k_emit_kept = nullptr;
     hipFunction_t k_csv_chunk = nullptr, k_csv_sel = nullptr, k_csv_rows = nullptr;
     hipFunction_t k_red_f64 = nullptr, k_red_f64_fin = nullptr;
     hipFunction_t k_red_i64 = nullptr, k_red_i64_fin = nullptr;
@@ -226,6 +227,7 @@ extern "C" tpx_stage* tpx_stage_compile(const char* hip_source,
         {"tpx_stage_write", &st->k_write, true},
         {"tpx_scan_block", &st->k_scan_block, true},
         {"tpx_scan_add", &st->k_scan_add, true},
+        {"tpx_emit_kept", &st->k_emit_kept, true},
         {"tpx_csv_chunk_stats", &st->k_csv_chunk, false},
         {"tpx_csv_select_counts", &st->k_csv_sel, false},
         {"tpx_csv_emit_rows", &st->k_csv_rows, false},
@@ -611,6 +613,15 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
     if (dev_scan(st, stream, (long long*)d_sizes, (long long*)d_size_scan, n,
                  &total_bytes))
         return -1;
+    // kept-row index list: lets the write kernel run one lane per kept row
+    // with wave-contiguous output spans (LDS-staged cooperative stores)
+    ARENA_TAKE(d_kept_idx, ((size_t)total_rows + 1) * 8);
+    {
+        unsigned g2 = (unsigned)std::min<long long>((n + 255) / 256, 8192);
+        if (g2 < 1) g2 = 1;
+        void* ea[] = {&d_keep, &d_keep_scan, &n, &d_kept_idx};
+        if (launch(st->k_emit_kept, g2, 256, stream, ea)) return -1;
+    }
     hipEventRecord(evs1, stream);
 
     bool mem_sink = D.sink == "mem";
@@ -619,11 +630,15 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
     ARENA_TAKE(d_out_offs, ((size_t)total_rows + 1) * 8);
     ARENA_TAKE(d_out_rowidx, ((size_t)total_rows + 1) * 8);
     {
-        unsigned grid = (unsigned)std::min<long long>((n + 255) / 256, 8192);
+        // csv sink iterates kept rows (128-thread wave-staged); mem sink
+        // iterates input rows. Size the grid for whichever is larger.
+        long long work = std::max<long long>(n, total_rows);
+        unsigned grid = (unsigned)std::min<long long>((work + 127) / 128, 8192);
+        if (grid < 1) grid = 1;
         void* args[] = {&d_keep, &d_keep_scan, &d_size_scan, &n, &row0,
                         &d_outv, &d_out, &d_out_offs, &d_out_rowidx,
-                        &total_rows, &total_bytes};
-        if (launch(st->k_write, grid, 256, stream, args)) return -1;
+                        &total_rows, &total_bytes, &d_kept_idx};
+        if (launch(st->k_write, grid, 128, stream, args)) return -1;
     }
     hipEventRecord(ev2, stream);
 

@@ -367,6 +367,22 @@ __device__ __forceinline__ tstr tpx_upper(TpxHeap& h, const tstr s, int* ec) {
     return tstr{d, s.n};
 }
 
+// x[0].upper() + x[1:].lower() fused (codegen peephole): one alloc, one pass
+// instead of getitem+upper+slice+lower+concat (3 allocs, 3 passes). Empty x
+// raises EC_INDEXERROR exactly like the unfused x[0] (tpx_getitem above).
+__device__ __forceinline__ tstr tpx_capitalize_ix(TpxHeap& h, const tstr s, int* ec) {
+    if (s.n <= 0) { *ec = EC_INDEXERROR; return tstr{s.p, 0}; }
+    char* d = tpx_alloc(h, s.n);
+    if (!d) { *ec = EC_MEMORYERROR; return tstr{s.p, 0}; }
+    unsigned char c0 = s.p[0];
+    d[0] = (c0 >= 'a' && c0 <= 'z') ? c0 - 32 : c0;
+    for (long long i = 1; i < s.n; ++i) {
+        unsigned char c = s.p[i];
+        d[i] = (c >= 'A' && c <= 'Z') ? c + 32 : c;
+    }
+    return tstr{d, s.n};
+}
+
 __device__ __forceinline__ tstr tpx_swapcase(TpxHeap& h, const tstr s, int* ec) {
     char* d = tpx_alloc(h, s.n);
     if (!d) { *ec = EC_MEMORYERROR; return tstr{s.p, 0}; }
@@ -1103,6 +1119,20 @@ extern "C" __global__ void tpx_hashagg_emit(const long long* __restrict__ tkeys,
 #define TPX_SCAN_THREADS 256
 #define TPX_SCAN_ITEMS 8
 #define TPX_SCAN_BLOCK (TPX_SCAN_THREADS * TPX_SCAN_ITEMS)
+
+// scatter the input-row index of every kept row to kept_idx[output position]:
+// lets tpx_stage_write iterate over KEPT rows with full lanes (at selectivity
+// s only s of the lanes would otherwise do work) and gives each wave of 64
+// kept rows a contiguous output byte span it can stage through LDS.
+extern "C" __global__ void tpx_emit_kept(const unsigned char* __restrict__ keep,
+                                         const long long* __restrict__ keep_scan,
+                                         long long n,
+                                         long long* __restrict__ kept_idx) {
+    long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+         i += stride)
+        if (keep[i]) kept_idx[keep_scan[i]] = i;
+}
 
 extern "C" __global__ void tpx_scan_block(const long long* __restrict__ in,
                                           long long* __restrict__ out,
