@@ -1218,13 +1218,38 @@ class StageCodegen:
         L.append("    long long span = span_end - span_start;")
         L.append("    bool staged = span <= TPX_WCAP;")
         L.append("    long long k = k0 + lane;")
+        L.append("    long long i = k < khi ? kept_idx[k] : 0;")
+        L.append("    long long my_start = k < khi ? size_scan[i] : 0;")
         L.append("    if (k < khi) {")
-        L.append("    long long i = kept_idx[k];")
-        L.append("    long long my_start = size_scan[i];")
-        L.append("    char* w = staged ? (wave_lds + (my_start - span_start))"
-                 " : ((char*)out_data + my_start);")
-        L.append("    out_offs[k] = my_start;")
-        L.append("    out_rowidx[k] = row0 + i;")
+        L.append("      out_offs[k] = my_start;")
+        L.append("      out_rowidx[k] = row0 + i;")
+        L.append("    }")
+        # duplicate the format body per pointer mode: in the staged branch w
+        # provably derives from LDS, so addrspace inference emits ds_write
+        # instead of flat stores (same trick as tpx_stage_main's row body)
+        body = self._csv_format_body(out_types)
+        L.append("    if (staged) {")
+        L.append("      if (k < khi) {")
+        L.append("        char* w = wave_lds + (my_start - span_start);")
+        L.extend("    " + ln for ln in body)
+        L.append("      }")
+        L.append("      __builtin_amdgcn_wave_barrier();")
+        L.append("      char* dst = (char*)out_data + span_start;")
+        L.append("      for (long long b = (long long)lane * 8; b + 8 <= span;"
+                 " b += 64 * 8)")
+        L.append("        __builtin_memcpy(dst + b, wave_lds + b, 8);")
+        L.append("      for (long long b = (span & ~7LL) + lane; b < span; b += 64)")
+        L.append("        dst[b] = wave_lds[b];")
+        L.append("    } else if (k < khi) {")
+        L.append("      char* w = (char*)out_data + my_start;")
+        L.extend("  " + ln for ln in body)
+        L.append("    }")
+        L.append("  }")
+        L.append("}")
+        return "\n".join(L)
+
+    def _csv_format_body(self, out_types):
+        L = []
         for k, t in enumerate(out_types):
             base = T.deopt(t)
             if k:
@@ -1232,7 +1257,8 @@ class StageCodegen:
             if base == T.STR:
                 pre = ""
                 if T.is_opt(t):
-                    L.append("    if (!((const unsigned char*)outv[%d])[i]) {" % (3 * k + 2))
+                    L.append("    if (!((const unsigned char*)outv[%d])[i]) {"
+                             % (3 * k + 2))
                     pre = "  "
                 L.append(pre + "    tstr v%d{(const char*)((const unsigned long long*)outv[%d])[i],"
                          " (long long)((const int*)outv[%d])[i]};" % (k, 3 * k, 3 * k + 1))
@@ -1249,19 +1275,7 @@ class StageCodegen:
             else:
                 raise CodegenError("csv sink for %r" % (t,))
         L.append("    *w++ = '\\n';")
-        L.append("    }")
-        L.append("    if (staged) {")
-        L.append("      __builtin_amdgcn_wave_barrier();")
-        L.append("      char* dst = (char*)out_data + span_start;")
-        L.append("      for (long long b = (long long)lane * 8; b + 8 <= span;"
-                 " b += 64 * 8)")
-        L.append("        __builtin_memcpy(dst + b, wave_lds + b, 8);")
-        L.append("      for (long long b = (span & ~7LL) + lane; b < span; b += 64)")
-        L.append("        dst[b] = wave_lds[b];")
-        L.append("    }")
-        L.append("  }")
-        L.append("}")
-        return "\n".join(L)
+        return L
 
     def _desc(self, in_types, out_types):
         def tdesc(t):
