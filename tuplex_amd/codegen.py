@@ -1200,7 +1200,8 @@ class StageCodegen:
         L.append("    long long* __restrict__ out_rowidx,")
         L.append("    long long total_rows, long long total_bytes,")
         L.append("    const long long* __restrict__ kept_idx) {")
-        L.append("  __shared__ char wsmem[2 * TPX_WCAP + 16];  // 128-thread blocks")
+        L.append("  __shared__ __attribute__((aligned(16)))"
+                 " char wsmem[2 * TPX_WCAP + 16];  // 128-thread blocks")
         L.append("  int lane = threadIdx.x & 63;")
         L.append("  int wid = threadIdx.x >> 6;")
         L.append("  char* wave_lds = wsmem + wid * TPX_WCAP;")
@@ -1234,12 +1235,27 @@ class StageCodegen:
         L.extend("    " + ln for ln in body)
         L.append("      }")
         L.append("      __builtin_amdgcn_wave_barrier();")
+        # cooperative span copy with ALIGNED global 8B stores (dst + a0 is
+        # 8-aligned by construction) and aligned LDS 8B reads + a uniform
+        # funnel shift (a0 is the same for every lane of the wave). A plain
+        # unaligned memcpy lowers to byte stores (measured: no faster than the
+        # unstaged kernel).
         L.append("      char* dst = (char*)out_data + span_start;")
-        L.append("      for (long long b = (long long)lane * 8; b + 8 <= span;"
-                 " b += 64 * 8)")
-        L.append("        __builtin_memcpy(dst + b, wave_lds + b, 8);")
-        L.append("      for (long long b = (span & ~7LL) + lane; b < span; b += 64)")
-        L.append("        dst[b] = wave_lds[b];")
+        L.append("      long long a0 = (8 - (span_start & 7)) & 7;")
+        L.append("      if (lane < a0 && lane < span) dst[lane] = wave_lds[lane];")
+        L.append("      int sh = (int)(a0 & 7) * 8;")
+        L.append("      for (long long b = a0 + (long long)lane * 8; b + 8 <= span;"
+                 " b += 64 * 8) {")
+        L.append("        unsigned long long lo ="
+                 " *(const unsigned long long*)(wave_lds + (b - a0));")
+        L.append("        unsigned long long hi ="
+                 " *(const unsigned long long*)(wave_lds + (b - a0) + 8);")
+        L.append("        unsigned long long v = sh ? ((lo >> sh) | (hi << (64 - sh)))"
+                 " : lo;")
+        L.append("        *(unsigned long long*)(dst + b) = v;")
+        L.append("      }")
+        L.append("      long long t0 = span > a0 ? a0 + ((span - a0) & ~7LL) : span;")
+        L.append("      if (t0 + lane < span) dst[t0 + lane] = wave_lds[t0 + lane];")
         L.append("    } else if (k < khi) {")
         L.append("      char* w = (char*)out_data + my_start;")
         L.extend("  " + ln for ln in body)
