@@ -1180,12 +1180,12 @@ class StageCodegen:
         L.append("}")
         return "\n".join(L)
 
-    # CSV write: one lane per KEPT row (kept_idx from tpx_emit_kept), 64 kept
-    # rows per wave -> contiguous output span; format into LDS, then copy the
-    # span out wave-cooperatively with 8B chunks (the old one-thread-per-input
-    # -row version wrote ~31 B rows byte-by-byte to scattered global addresses
-    # with (1 - selectivity) of the lanes idle). Spans > TPX_WCAP (long rows)
-    # fall back to direct global formatting for that wave.
+    # CSV write: one lane per INPUT row (coalesced columnar loads; a kept-row
+    # gather was measured SLOWER — gather defeats load coalescing). The 64
+    # input rows of a wave own a CONTIGUOUS output span of ~selectivity*64*row
+    # bytes: active lanes format into LDS, then the wave copies the span out
+    # with aligned 8B stores. Spans > TPX_WCAP (long rows) fall back to direct
+    # global formatting for that wave.
     WRITE_CAP = 8192  # bytes per wave; 2 waves/block -> 16 KiB LDS
 
     def _write_kernel_csv(self, out_types):
@@ -1199,7 +1199,7 @@ class StageCodegen:
         L.append("    unsigned char* __restrict__ out_data, long long* __restrict__ out_offs,")
         L.append("    long long* __restrict__ out_rowidx,")
         L.append("    long long total_rows, long long total_bytes,")
-        L.append("    const long long* __restrict__ kept_idx) {")
+        L.append("    const long long* __restrict__ kept_idx) {  // unused")
         L.append("  __shared__ __attribute__((aligned(16)))"
                  " char wsmem[2 * TPX_WCAP + 16];  // 128-thread blocks")
         L.append("  int lane = threadIdx.x & 63;")
@@ -1208,29 +1208,29 @@ class StageCodegen:
         L.append("  if (blockIdx.x == 0 && threadIdx.x == 0)"
                  " out_offs[total_rows] = total_bytes;")
         L.append("  long long wave_stride = (long long)gridDim.x * (blockDim.x >> 6);")
-        L.append("  long long nwaves = (total_rows + 63) >> 6;")
+        L.append("  long long nwaves = (n + 63) >> 6;")
         L.append("  for (long long wb = (long long)blockIdx.x * (blockDim.x >> 6) + wid;"
                  " wb < nwaves; wb += wave_stride) {")
-        L.append("    long long k0 = wb << 6;")
-        L.append("    long long khi = k0 + 64 < total_rows ? k0 + 64 : total_rows;")
-        L.append("    long long span_start = size_scan[kept_idx[k0]];")
-        L.append("    long long span_end = khi < total_rows ?"
-                 " size_scan[kept_idx[khi]] : total_bytes;")
+        L.append("    long long r0 = wb << 6;")
+        L.append("    long long rhi = r0 + 64 < n ? r0 + 64 : n;")
+        L.append("    long long span_start = size_scan[r0];")
+        L.append("    long long span_end = r0 + 64 < n ? size_scan[r0 + 64]"
+                 " : total_bytes;")
         L.append("    long long span = span_end - span_start;")
         L.append("    bool staged = span <= TPX_WCAP;")
-        L.append("    long long k = k0 + lane;")
-        L.append("    long long i = k < khi ? kept_idx[k] : 0;")
-        L.append("    long long my_start = k < khi ? size_scan[i] : 0;")
-        L.append("    if (k < khi) {")
-        L.append("      out_offs[k] = my_start;")
-        L.append("      out_rowidx[k] = row0 + i;")
+        L.append("    long long i = r0 + lane;")
+        L.append("    bool active = i < rhi && keep[i];")
+        L.append("    long long my_start = active ? size_scan[i] : 0;")
+        L.append("    if (active) {")
+        L.append("      out_offs[keep_scan[i]] = my_start;")
+        L.append("      out_rowidx[keep_scan[i]] = row0 + i;")
         L.append("    }")
         # duplicate the format body per pointer mode: in the staged branch w
         # provably derives from LDS, so addrspace inference emits ds_write
         # instead of flat stores (same trick as tpx_stage_main's row body)
         body = self._csv_format_body(out_types)
         L.append("    if (staged) {")
-        L.append("      if (k < khi) {")
+        L.append("      if (active) {")
         L.append("        char* w = wave_lds + (my_start - span_start);")
         L.extend("    " + ln for ln in body)
         L.append("      }")
@@ -1256,7 +1256,7 @@ class StageCodegen:
         L.append("      }")
         L.append("      long long t0 = span > a0 ? a0 + ((span - a0) & ~7LL) : span;")
         L.append("      if (t0 + lane < span) dst[t0 + lane] = wave_lds[t0 + lane];")
-        L.append("    } else if (k < khi) {")
+        L.append("    } else if (active) {")
         L.append("      char* w = (char*)out_data + my_start;")
         L.extend("  " + ln for ln in body)
         L.append("    }")

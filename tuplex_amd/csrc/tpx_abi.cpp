@@ -613,15 +613,6 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
     if (dev_scan(st, stream, (long long*)d_sizes, (long long*)d_size_scan, n,
                  &total_bytes))
         return -1;
-    // kept-row index list: lets the write kernel run one lane per kept row
-    // with wave-contiguous output spans (LDS-staged cooperative stores)
-    ARENA_TAKE(d_kept_idx, ((size_t)total_rows + 1) * 8);
-    {
-        unsigned g2 = (unsigned)std::min<long long>((n + 255) / 256, 8192);
-        if (g2 < 1) g2 = 1;
-        void* ea[] = {&d_keep, &d_keep_scan, &n, &d_kept_idx};
-        if (launch(st->k_emit_kept, g2, 256, stream, ea)) return -1;
-    }
     hipEventRecord(evs1, stream);
 
     bool mem_sink = D.sink == "mem";
@@ -630,14 +621,12 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
     ARENA_TAKE(d_out_offs, ((size_t)total_rows + 1) * 8);
     ARENA_TAKE(d_out_rowidx, ((size_t)total_rows + 1) * 8);
     {
-        // csv sink iterates kept rows (128-thread wave-staged); mem sink
-        // iterates input rows. Size the grid for whichever is larger.
-        long long work = std::max<long long>(n, total_rows);
-        unsigned grid = (unsigned)std::min<long long>((work + 127) / 128, 8192);
+        unsigned grid = (unsigned)std::min<long long>((n + 127) / 128, 8192);
         if (grid < 1) grid = 1;
+        void* nullp = nullptr;  // kept_idx: unused by both write kernels
         void* args[] = {&d_keep, &d_keep_scan, &d_size_scan, &n, &row0,
                         &d_outv, &d_out, &d_out_offs, &d_out_rowidx,
-                        &total_rows, &total_bytes, &d_kept_idx};
+                        &total_rows, &total_bytes, &nullp};
         if (launch(st->k_write, grid, 128, stream, args)) return -1;
     }
     hipEventRecord(ev2, stream);
