@@ -1186,7 +1186,8 @@ class StageCodegen:
     # bytes: active lanes format into LDS, then the wave copies the span out
     # with aligned 8B stores. Spans > TPX_WCAP (long rows) fall back to direct
     # global formatting for that wave.
-    WRITE_CAP = 8192  # bytes per wave; 2 waves/block -> 16 KiB LDS
+    WRITE_CAP = 4096  # bytes per wave; 2 waves/block -> 8 KiB LDS (measured
+    #                   best vs 8192/2048/unstaged on the Z1 bench: wprobe.py)
 
     def _write_kernel_csv(self, out_types):
         L = []
