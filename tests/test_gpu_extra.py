@@ -267,3 +267,32 @@ def test_type_hints_and_custom_nulls(tmp_path):
     assert got == ref["output"]
     assert ds.exception_counts == ref["exception_counts"]
     assert sum(1 for _, b, _c in got if b == "None") == 400
+
+
+def test_chunk_pipelined_large_input(tmp_path):
+    """>=1M rows triggers the 4-chunk multi-stream pipeline in run_core
+    (tpx_abi.cpp): per-chunk scans/writes overlap later chunks' main kernel.
+    Input = 20 tiles of one 64k-row dirty base, so the expected output is the
+    single-tile oracle result repeated 20x in order (merge order, exception
+    replay and chunk-boundary stitching all checked)."""
+    from tests.test_codegen_compile import zillow_ops
+    from tests.zillow_data import make_zillow_csv_bytes
+
+    base, _ = make_zillow_csv_bytes(64000, seed=13, dirty_frac=0.01,
+                                    header=True)
+    header, body = base.split(b"\n", 1)
+    data = header + b"\n" + body * 20  # 1.28M rows
+    p = _write(tmp_path, data, "big.csv")
+
+    # one engine call for the whole file (default inputSplitSize would split
+    # into <1M-row chunks and never reach the C=4 path)
+    ctx = tuplex_amd.Context({"tuplex.inputSplitSize": "1GB"})
+    ds = apply_ops(ctx.csv(p), zillow_ops())
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+
+    ref = pyoracle_csv.run_csv_pipeline(base, zillow_ops())
+    assert len(got) == 20 * len(ref["output"])
+    assert got == ref["output"] * 20
+    for k, v in ref["exception_counts"].items():
+        assert ds.exception_counts.get(k, 0) == v * 20, (k, v)

@@ -323,9 +323,18 @@ struct Persist {
     size_t heap_cap = 0;
     void* exc = nullptr;
     size_t exc_cap_recs = 0;
-    void* counters = nullptr;  // [heap_cursor, exc_count]
+    void* counters = nullptr;  // [cursor_c at 16*c (c<4) ... exc_count at 80]
 };
 static Persist g_persist[64];
+
+// two extra per-device streams for chunk pipelining (scans/writes of chunk c
+// overlap the main kernel of chunk c+1)
+static hipStream_t g_cstream[64][2];
+static hipStream_t cstream(int dev, int i) {
+    if (!g_cstream[dev][i])
+        (void)hipStreamCreateWithFlags(&g_cstream[dev][i], hipStreamNonBlocking);
+    return g_cstream[dev][i];
+}
 
 static int cur_device() {
     int d = 0;
@@ -434,7 +443,7 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
 
     // persistent heap / exception buffer / counters
     if (!P.counters)
-        HIP_CHECK(hipMalloc(&P.counters, 16));
+        HIP_CHECK(hipMalloc(&P.counters, 96));
     unsigned long long exc_cap =
         (unsigned long long)std::min<long long>(std::max<long long>(n, 1024),
                                                 1 << 20);
@@ -444,10 +453,45 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
         P.exc_cap_recs = exc_cap;
     }
     exc_cap = P.exc_cap_recs;
-    // +thread-chunk slack: <=2048 blocks x 256 threads x TPX_HEAP_CHUNK(256 B)
+
+    // chunk pipelining: split the rows into C ranges; each chunk's compaction
+    // scans + write kernel run on a side stream and overlap the NEXT chunk's
+    // main kernel (the step is otherwise strictly sequential; main dominates).
+    // Aggregate sinks consume the full keep/outv arrays at once -> C=1.
+    bool mem_sink = D.sink == "mem";
+    const int C = (D.agg.empty() && D.aggby.empty() && n >= (1 << 20)) ? 4 : 1;
+    hipStream_t S[2] = {stream, stream};
+    if (C > 1) { S[0] = cstream(dev, 0); S[1] = cstream(dev, 1); }
+    long long cstart[5] = {0}, ccnt[4] = {0};
+    for (int c = 0; c < C; ++c) {
+        cstart[c] = n * c / C;
+        cstart[c + 1] = n * (c + 1) / C;
+        ccnt[c] = cstart[c + 1] - cstart[c];
+    }
+    // per-chunk biased outv pointer tables (chunk rows index columnar slot i
+    // locally); chunk 0 table == the unbiased one
+    void* d_outv_c[4] = {d_outv, nullptr, nullptr, nullptr};
+    for (int c = 1; c < C; ++c) {
+        std::vector<void*> ov(outv.size());
+        for (int k = 0; k < nout; ++k) {
+            const ColDesc& col = D.out_cols[(size_t)k];
+            ov[(size_t)k * 3 + 0] = (char*)outv[(size_t)k * 3 + 0] + cstart[c] * 8;
+            ov[(size_t)k * 3 + 1] = col.kind == K_STR
+                ? (char*)outv[(size_t)k * 3 + 1] + cstart[c] * 4 : nullptr;
+            ov[(size_t)k * 3 + 2] = col.opt
+                ? (char*)outv[(size_t)k * 3 + 2] + cstart[c] : nullptr;
+        }
+        d_outv_c[c] = g_arena[dev].take(ov.size() * sizeof(void*) + 8);
+        if (!d_outv_c[c]) return -1;
+        HIP_CHECK(hipMemcpyAsync(d_outv_c[c], ov.data(),
+                                 ov.size() * sizeof(void*),
+                                 hipMemcpyHostToDevice, stream));
+    }
+
+    // +thread-chunk slack per LAUNCH: <=2048 blocks x 256 threads x 256 B
     unsigned long long heap_cap =
         (unsigned long long)std::max<long long>(in_bytes + (in_bytes >> 1) +
-                                                (160ll << 20), 1 << 20);
+                                                C * (160ll << 20), 1 << 20);
     if (P.heap_cap < heap_cap) {
         if (P.heap) (void)hipFree(P.heap);
         HIP_CHECK(hipMalloc(&P.heap, heap_cap + 16));  // tpx_memcpy over-read pad
@@ -455,37 +499,120 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
     }
     heap_cap = P.heap_cap;
 
-    void* d_heap_cursor = P.counters;
-    void* d_exc_count = (char*)P.counters + 8;
+    void* d_exc_count = (char*)P.counters + 80;
     unsigned long long exc_count = 0;
+    // [m1, s0, s1, w0, w1] per chunk + 2 join events
+    std::vector<hipEvent_t> cev((size_t)C * 5 + 2);
+    for (auto& e : cev) hipEventCreate(&e);
+    long long total_rows = 0, total_bytes = 0;
+    long long chunk_rows[4] = {0}, chunk_bytes[4] = {0};
+    void* chunk_out[4] = {nullptr};
+    void* d_out = nullptr;
+    void* d_out_offs = nullptr;
+    void* d_out_rowidx = nullptr;
+    bool sink_done = false;
     for (int attempt = 0;; ++attempt) {
-        HIP_CHECK(hipMemsetAsync(P.counters, 0, 16, stream));
-        // main runs 128-thread blocks (2 waves x 16 KiB LDS staging = 32 KiB
-        // -> 5 blocks/CU = 10 waves/CU); each wave owns 64 rows
-        long long waves = (n + 63) / 64;
-        unsigned grid = (unsigned)std::min<long long>((waves + 1) / 2, 4096);
-        void* args[] = {&d_in, &d_offs, &n, &row0, &P.heap, &d_heap_cursor,
-                        &heap_cap, &d_keep, &d_keep01, &d_sizes, &P.exc,
-                        &d_exc_count, &exc_cap, &d_outv};
+        HIP_CHECK(hipMemsetAsync(P.counters, 0, 96, stream));
         hipEventRecord(evm0, stream);
-        if (launch(st->k_main, grid, 128, stream, args)) return -1;
-        hipEventRecord(evm1, stream);
-        unsigned long long counters[2] = {0, 0};
-        HIP_CHECK(hipMemcpyAsync(counters, P.counters, 16, hipMemcpyDeviceToHost,
-                                 stream));
+        unsigned long long hs = heap_cap / (unsigned long long)C & ~255ull;
+        for (int c = 0; c < C; ++c) {
+            hipStream_t sc = S[c & 1];
+            if (C > 1) hipStreamWaitEvent(sc, evm0, 0);
+            // main runs 128-thread blocks (2 waves x 16 KiB LDS staging =
+            // 32 KiB -> 5 blocks/CU = 10 waves/CU); each wave owns 64 rows
+            long long nc = ccnt[c];
+            long long waves = (nc + 63) / 64;
+            unsigned grid = (unsigned)std::min<long long>((waves + 1) / 2, 4096);
+            void* in_offs_c = (char*)d_offs + cstart[c] * 8;
+            long long row0_c = row0 + cstart[c];
+            void* heap_c = (char*)P.heap + (unsigned long long)c * hs;
+            void* cursor_c = (char*)P.counters + 16 * c;
+            void* keep_c = (char*)d_keep + cstart[c];
+            void* keep01_c = (char*)d_keep01 + cstart[c] * 8;
+            void* sizes_c = (char*)d_sizes + cstart[c] * 8;
+            void* args[] = {&d_in, &in_offs_c, &nc, &row0_c, &heap_c, &cursor_c,
+                            &hs, &keep_c, &keep01_c, &sizes_c, &P.exc,
+                            &d_exc_count, &exc_cap, &d_outv_c[c]};
+            if (launch(st->k_main, grid, 128, sc, args)) return -1;
+            hipEventRecord(cev[(size_t)c * 5], sc);
+        }
+        // C>1 (non-agg sink): pipelined per-chunk scans + writes launched
+        // OPTIMISTICALLY before the overflow check (overflowed rows are marked
+        // exceptions with keep=false, so the write reads no garbage; on the
+        // rare overflow the attempt is simply redone)
+        if (C > 1) {
+            total_rows = total_bytes = 0;
+            d_out_offs = g_arena[dev].take(((size_t)n + 1) * 8);
+            d_out_rowidx = g_arena[dev].take(((size_t)n + 1) * 8);
+            void* d_keep_scan = g_arena[dev].take((size_t)n * 8);
+            void* d_size_scan = g_arena[dev].take((size_t)n * 8);
+            if (!d_out_offs || !d_out_rowidx || !d_keep_scan || !d_size_scan)
+                return -1;
+            for (int c = 0; c < C; ++c) {
+                hipStream_t sc = S[c & 1];
+                void* keep01_c = (char*)d_keep01 + cstart[c] * 8;
+                void* sizes_c = (char*)d_sizes + cstart[c] * 8;
+                void* kscan_c = (char*)d_keep_scan + cstart[c] * 8;
+                void* sscan_c = (char*)d_size_scan + cstart[c] * 8;
+                long long rows_c = 0, bytes_c = 0;
+                hipEventRecord(cev[(size_t)c * 5 + 1], sc);
+                if (dev_scan(st, sc, (long long*)keep01_c,
+                             (long long*)kscan_c, ccnt[c], &rows_c))
+                    return -1;
+                if (dev_scan(st, sc, (long long*)sizes_c, (long long*)sscan_c,
+                             ccnt[c], &bytes_c))
+                    return -1;
+                hipEventRecord(cev[(size_t)c * 5 + 2], sc);
+                chunk_out[c] = g_arena[dev].take((size_t)bytes_c + 16);
+                if (!chunk_out[c]) return -1;
+                long long out_byte0 = (mem_sink ? 8 : 0) + total_bytes;
+                void* keep_c = (char*)d_keep + cstart[c];
+                void* offs_c = (char*)d_out_offs + total_rows * 8;
+                void* ridx_c = (char*)d_out_rowidx + total_rows * 8;
+                long long row0_c = row0 + cstart[c];
+                unsigned grid = (unsigned)std::min<long long>(
+                    (ccnt[c] + 127) / 128, 8192);
+                if (grid < 1) grid = 1;
+                void* args[] = {&keep_c, &kscan_c, &sscan_c, &ccnt[c], &row0_c,
+                                &d_outv_c[c], &chunk_out[c], &offs_c, &ridx_c,
+                                &rows_c, &bytes_c, &out_byte0};
+                hipEventRecord(cev[(size_t)c * 5 + 3], sc);
+                if (launch(st->k_write, grid, 128, sc, args)) return -1;
+                hipEventRecord(cev[(size_t)c * 5 + 4], sc);
+                chunk_rows[c] = rows_c;
+                chunk_bytes[c] = bytes_c;
+                total_rows += rows_c;
+                total_bytes += bytes_c;
+            }
+        }
+        // overflow check (all mains have completed: each chunk's scans synced
+        // its stream for C>1; explicit sync below covers C==1)
+        unsigned long long counters[12] = {0};
+        HIP_CHECK(hipMemcpyAsync(counters, P.counters, 96,
+                                 hipMemcpyDeviceToHost, stream));
         HIP_CHECK(hipStreamSynchronize(stream));
-        unsigned long long heap_used = counters[0];
-        exc_count = counters[1];
-        bool heap_overflow = heap_used > heap_cap;
+        if (C == 1) {
+            // the copy above raced the main kernel only through stream order
+            // (same stream) -> already correct
+        }
+        unsigned long long max_used = 0;
+        for (int c = 0; c < C; ++c)
+            max_used = std::max(max_used, counters[c * 2]);
+        exc_count = counters[10];
+        bool heap_overflow = max_used > hs;
         bool exc_overflow = exc_count > exc_cap;
-        if (!heap_overflow && !exc_overflow) break;
+        if (!heap_overflow && !exc_overflow) {
+            sink_done = C > 1;
+            break;
+        }
         if (attempt >= 3) { set_err("retry limit (heap/exc overflow)"); return -1; }
         if (heap_overflow) {
             (void)hipFree(P.heap);
             P.heap = nullptr;
             P.heap_cap = 0;
-            heap_cap = heap_used + (heap_used >> 1) + (16 << 20);
-            HIP_CHECK(hipMalloc(&P.heap, heap_cap));
+            heap_cap = (max_used + (max_used >> 1) + (16 << 20)) *
+                       (unsigned long long)C;
+            HIP_CHECK(hipMalloc(&P.heap, heap_cap + 16));
             P.heap_cap = heap_cap;
         }
         if (exc_overflow) {
@@ -603,32 +730,59 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
         res->out_row_indices = (int64_t*)malloc(8);
         res->out_row_indices[0] = 0;
     } else {
-    // compaction scans
-    ARENA_TAKE(d_keep_scan, (size_t)n * 8);
-    ARENA_TAKE(d_size_scan, (size_t)n * 8);
-    long long total_rows = 0, total_bytes = 0;
-    if (dev_scan(st, stream, (long long*)d_keep01, (long long*)d_keep_scan, n,
-                 &total_rows))
-        return -1;
-    if (dev_scan(st, stream, (long long*)d_sizes, (long long*)d_size_scan, n,
-                 &total_bytes))
-        return -1;
-    hipEventRecord(evs1, stream);
-
-    bool mem_sink = D.sink == "mem";
-    long long out_total = mem_sink ? 8 + total_bytes : total_bytes;
-    ARENA_TAKE(d_out, (size_t)out_total + 8);
-    ARENA_TAKE(d_out_offs, ((size_t)total_rows + 1) * 8);
-    ARENA_TAKE(d_out_rowidx, ((size_t)total_rows + 1) * 8);
-    {
+    long long out_total;
+    if (sink_done) {
+        // C>1: per-chunk writes already done; gather the chunk buffers into
+        // the final contiguous output
+        out_total = (mem_sink ? 8 : 0) + total_bytes;
+        d_out = g_arena[dev].take((size_t)out_total + 16);
+        if (!d_out) return -1;
+        long long acc = mem_sink ? 8 : 0;
+        for (int c = 0; c < C; ++c) {
+            if (chunk_bytes[c])
+                HIP_CHECK(hipMemcpyAsync((char*)d_out + acc, chunk_out[c],
+                                         (size_t)chunk_bytes[c],
+                                         hipMemcpyDeviceToDevice, S[c & 1]));
+            acc += chunk_bytes[c];
+        }
+        // join the side streams back into `stream`
+        hipEventRecord(cev[(size_t)C * 5], S[0]);
+        hipEventRecord(cev[(size_t)C * 5 + 1], S[1]);
+        hipStreamWaitEvent(stream, cev[(size_t)C * 5], 0);
+        hipStreamWaitEvent(stream, cev[(size_t)C * 5 + 1], 0);
+        hipEventRecord(evs1, stream);
+    } else {
+        // C==1: compaction scans + one write into the final buffer
+        ARENA_TAKE(d_keep_scan, (size_t)n * 8);
+        ARENA_TAKE(d_size_scan, (size_t)n * 8);
+        if (dev_scan(st, stream, (long long*)d_keep01, (long long*)d_keep_scan,
+                     n, &total_rows))
+            return -1;
+        if (dev_scan(st, stream, (long long*)d_sizes, (long long*)d_size_scan,
+                     n, &total_bytes))
+            return -1;
+        hipEventRecord(evs1, stream);
+        out_total = (mem_sink ? 8 : 0) + total_bytes;
+        d_out = g_arena[dev].take((size_t)out_total + 16);
+        d_out_offs = g_arena[dev].take(((size_t)total_rows + 1) * 8);
+        d_out_rowidx = g_arena[dev].take(((size_t)total_rows + 1) * 8);
+        if (!d_out || !d_out_offs || !d_out_rowidx) return -1;
         unsigned grid = (unsigned)std::min<long long>((n + 127) / 128, 8192);
         if (grid < 1) grid = 1;
-        void* nullp = nullptr;  // kept_idx: unused by both write kernels
+        void* out_base = (char*)d_out + (mem_sink ? 8 : 0);
+        long long out_byte0 = mem_sink ? 8 : 0;
         void* args[] = {&d_keep, &d_keep_scan, &d_size_scan, &n, &row0,
-                        &d_outv, &d_out, &d_out_offs, &d_out_rowidx,
-                        &total_rows, &total_bytes, &nullp};
+                        &d_outv, &out_base, &d_out_offs, &d_out_rowidx,
+                        &total_rows, &total_bytes, &out_byte0};
         if (launch(st->k_write, grid, 128, stream, args)) return -1;
     }
+    // header + sentinel are host-written (the kernels write only row entries)
+    long long sentinel = out_total;
+    HIP_CHECK(hipMemcpyAsync((char*)d_out_offs + total_rows * 8, &sentinel, 8,
+                             hipMemcpyHostToDevice, stream));
+    if (mem_sink)
+        HIP_CHECK(hipMemcpyAsync(d_out, &total_rows, 8, hipMemcpyHostToDevice,
+                                 stream));
     hipEventRecord(ev2, stream);
 
     // D2H (skipped when the caller keeps outputs device-resident, flags bit1)
@@ -716,11 +870,32 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
     float ms = 0;
     hipEventElapsedTime(&ms, ev1, ev2); res->t_kernel_ms = ms;
     hipEventElapsedTime(&ms, ev2, ev3); res->t_d2h_ms = ms;
-    hipEventElapsedTime(&ms, evm0, evm1); res->t_main_ms = ms;
-    hipEventElapsedTime(&ms, evm1, evs1); res->t_compact_ms = ms;
-    hipEventElapsedTime(&ms, evs1, ev2); res->t_write_ms = ms;
+    // main: shared start -> latest chunk-main end (spans both side streams)
+    float tmain = 0;
+    for (int c = 0; c < C; ++c) {
+        hipEventElapsedTime(&ms, evm0, cev[(size_t)c * 5]);
+        tmain = std::max(tmain, ms);
+    }
+    res->t_main_ms = tmain;
+    if (sink_done) {
+        float tc = 0, tw = 0;
+        for (int c = 0; c < C; ++c) {
+            hipEventElapsedTime(&ms, cev[(size_t)c * 5 + 1],
+                                cev[(size_t)c * 5 + 2]);
+            tc += ms;
+            hipEventElapsedTime(&ms, cev[(size_t)c * 5 + 3],
+                                cev[(size_t)c * 5 + 4]);
+            tw += ms;
+        }
+        res->t_compact_ms = tc;
+        res->t_write_ms = tw;
+    } else {
+        hipEventElapsedTime(&ms, cev[0], evs1); res->t_compact_ms = ms;
+        hipEventElapsedTime(&ms, evs1, ev2); res->t_write_ms = ms;
+    }
     (void)hipEventDestroy(ev1); (void)hipEventDestroy(ev2); (void)hipEventDestroy(ev3);
     (void)hipEventDestroy(evm0); (void)hipEventDestroy(evm1); (void)hipEventDestroy(evs1);
+    for (auto& e : cev) (void)hipEventDestroy(e);
     return 0;
 }
 
