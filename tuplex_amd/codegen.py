@@ -1114,17 +1114,16 @@ class StageCodegen:
         L.append("    unsigned char* __restrict__ out_data, long long* __restrict__ out_offs,")
         L.append("    long long* __restrict__ out_rowidx,")
         L.append("    long long total_rows, long long total_bytes,")
-        L.append("    const long long* __restrict__ kept_idx) {  // unused (mem sink)")
+        # out_data is the CHUNK-LOCAL base (past the 8B numRows header, which
+        # the host writes); out_byte0 is the row's global byte offset bias
+        # stored into out_offs (chunk pipelining: tpx_abi.cpp run_core)
+        L.append("    long long out_byte0) {")
         L.append("  long long stride = (long long)gridDim.x * blockDim.x;")
         L.append("  long long tid0 = (long long)blockIdx.x * blockDim.x + threadIdx.x;")
-        L.append("  if (tid0 == 0) {")
-        L.append("    *(long long*)out_data = total_rows;")
-        L.append("    out_offs[total_rows] = 8 + total_bytes;")
-        L.append("  }")
         L.append("  for (long long i = tid0; i < n; i += stride) {")
         L.append("    if (!keep[i]) continue;")
-        L.append("    unsigned char* w = out_data + 8 + size_scan[i];")
-        L.append("    out_offs[keep_scan[i]] = 8 + size_scan[i];")
+        L.append("    unsigned char* w = out_data + size_scan[i];")
+        L.append("    out_offs[keep_scan[i]] = out_byte0 + size_scan[i];")
         L.append("    out_rowidx[keep_scan[i]] = row0 + i;")
         # load columnar values
         for k, t in enumerate(out_types):
@@ -1200,14 +1199,14 @@ class StageCodegen:
         L.append("    unsigned char* __restrict__ out_data, long long* __restrict__ out_offs,")
         L.append("    long long* __restrict__ out_rowidx,")
         L.append("    long long total_rows, long long total_bytes,")
-        L.append("    const long long* __restrict__ kept_idx) {  // unused")
+        # out_data is the CHUNK-LOCAL base; out_byte0 is the global byte bias
+        # stored into out_offs; the sentinel out_offs[total] is host-written
+        L.append("    long long out_byte0) {")
         L.append("  __shared__ __attribute__((aligned(16)))"
                  " char wsmem[2 * TPX_WCAP + 16];  // 128-thread blocks")
         L.append("  int lane = threadIdx.x & 63;")
         L.append("  int wid = threadIdx.x >> 6;")
         L.append("  char* wave_lds = wsmem + wid * TPX_WCAP;")
-        L.append("  if (blockIdx.x == 0 && threadIdx.x == 0)"
-                 " out_offs[total_rows] = total_bytes;")
         L.append("  long long wave_stride = (long long)gridDim.x * (blockDim.x >> 6);")
         L.append("  long long nwaves = (n + 63) >> 6;")
         L.append("  for (long long wb = (long long)blockIdx.x * (blockDim.x >> 6) + wid;"
@@ -1223,7 +1222,7 @@ class StageCodegen:
         L.append("    bool active = i < rhi && keep[i];")
         L.append("    long long my_start = active ? size_scan[i] : 0;")
         L.append("    if (active) {")
-        L.append("      out_offs[keep_scan[i]] = my_start;")
+        L.append("      out_offs[keep_scan[i]] = out_byte0 + my_start;")
         L.append("      out_rowidx[keep_scan[i]] = row0 + i;")
         L.append("    }")
         # duplicate the format body per pointer mode: in the staged branch w
