@@ -443,7 +443,7 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
 
     // persistent heap / exception buffer / counters
     if (!P.counters)
-        HIP_CHECK(hipMalloc(&P.counters, 96));
+        HIP_CHECK(hipMalloc(&P.counters, 160));
     unsigned long long exc_cap =
         (unsigned long long)std::min<long long>(std::max<long long>(n, 1024),
                                                 1 << 20);
@@ -459,18 +459,41 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
     // main kernel (the step is otherwise strictly sequential; main dominates).
     // Aggregate sinks consume the full keep/outv arrays at once -> C=1.
     bool mem_sink = D.sink == "mem";
-    const int C = (D.agg.empty() && D.aggby.empty() && n >= (1 << 20)) ? 4 : 1;
+    int C = 1;
+    if (D.agg.empty() && D.aggby.empty() && n >= (1 << 20)) {
+        static int env_c = [] {
+            const char* e = getenv("TPX_CHUNKS");
+            int v = e ? atoi(e) : 4;
+            return v < 1 ? 1 : (v > 8 ? 8 : v);
+        }();
+        C = env_c;
+    }
     hipStream_t S[2] = {stream, stream};
     if (C > 1) { S[0] = cstream(dev, 0); S[1] = cstream(dev, 1); }
-    long long cstart[5] = {0}, ccnt[4] = {0};
-    for (int c = 0; c < C; ++c) {
-        cstart[c] = n * c / C;
-        cstart[c + 1] = n * (c + 1) / C;
-        ccnt[c] = cstart[c + 1] - cstart[c];
+    // chunk split: sizes proportional to ratio^c (ratio<1 -> later chunks
+    // smaller, shrinking the non-overlapped scan+write tail of the last chunk)
+    static double env_ratio = [] {
+        const char* e = getenv("TPX_CHUNK_RATIO");
+        double v = e ? atof(e) : 0.75;
+        return (v > 0.1 && v <= 1.0) ? v : 1.0;
+    }();
+    long long cstart[9] = {0};
+    long long ccnt[8] = {0};
+    {
+        double wsum = 0, w = 1;
+        for (int c = 0; c < C; ++c) { wsum += w; w *= env_ratio; }
+        double acc = 0;
+        w = 1;
+        for (int c = 0; c < C; ++c) {
+            acc += w;
+            cstart[c + 1] = c + 1 == C ? n : (long long)(n * acc / wsum);
+            w *= env_ratio;
+        }
+        for (int c = 0; c < C; ++c) ccnt[c] = cstart[c + 1] - cstart[c];
     }
     // per-chunk biased outv pointer tables (chunk rows index columnar slot i
     // locally); chunk 0 table == the unbiased one
-    void* d_outv_c[4] = {d_outv, nullptr, nullptr, nullptr};
+    void* d_outv_c[8] = {d_outv};
     for (int c = 1; c < C; ++c) {
         std::vector<void*> ov(outv.size());
         for (int k = 0; k < nout; ++k) {
@@ -499,20 +522,20 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
     }
     heap_cap = P.heap_cap;
 
-    void* d_exc_count = (char*)P.counters + 80;
+    void* d_exc_count = (char*)P.counters + 128;
     unsigned long long exc_count = 0;
     // [m1, s0, s1, w0, w1] per chunk + 2 join events
     std::vector<hipEvent_t> cev((size_t)C * 5 + 2);
     for (auto& e : cev) hipEventCreate(&e);
     long long total_rows = 0, total_bytes = 0;
-    long long chunk_rows[4] = {0}, chunk_bytes[4] = {0};
-    void* chunk_out[4] = {nullptr};
+    long long chunk_rows[8] = {0}, chunk_bytes[8] = {0};
+    void* chunk_out[8] = {nullptr};
     void* d_out = nullptr;
     void* d_out_offs = nullptr;
     void* d_out_rowidx = nullptr;
     bool sink_done = false;
     for (int attempt = 0;; ++attempt) {
-        HIP_CHECK(hipMemsetAsync(P.counters, 0, 96, stream));
+        HIP_CHECK(hipMemsetAsync(P.counters, 0, 160, stream));
         hipEventRecord(evm0, stream);
         unsigned long long hs = heap_cap / (unsigned long long)C & ~255ull;
         for (int c = 0; c < C; ++c) {
@@ -587,8 +610,8 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
         }
         // overflow check (all mains have completed: each chunk's scans synced
         // its stream for C>1; explicit sync below covers C==1)
-        unsigned long long counters[12] = {0};
-        HIP_CHECK(hipMemcpyAsync(counters, P.counters, 96,
+        unsigned long long counters[20] = {0};
+        HIP_CHECK(hipMemcpyAsync(counters, P.counters, 160,
                                  hipMemcpyDeviceToHost, stream));
         HIP_CHECK(hipStreamSynchronize(stream));
         if (C == 1) {
@@ -598,7 +621,7 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
         unsigned long long max_used = 0;
         for (int c = 0; c < C; ++c)
             max_used = std::max(max_used, counters[c * 2]);
-        exc_count = counters[10];
+        exc_count = counters[16];
         bool heap_overflow = max_used > hs;
         bool exc_overflow = exc_count > exc_cap;
         if (!heap_overflow && !exc_overflow) {
