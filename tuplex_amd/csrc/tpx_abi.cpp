@@ -135,6 +135,7 @@ struct tpx_stage {
     hipFunction_t k_main = nullptr, k_write = nullptr;
     hipFunction_t k_scan_block = nullptr, k_scan_add = nullptr;
     hipFunction_t k_emit_kept = nullptr;
+    hipFunction_t k_pair_total = nullptr;
     hipFunction_t k_csv_chunk = nullptr, k_csv_sel = nullptr, k_csv_rows = nullptr;
     hipFunction_t k_red_f64 = nullptr, k_red_f64_fin = nullptr;
     hipFunction_t k_red_i64 = nullptr, k_red_i64_fin = nullptr;
@@ -228,6 +229,7 @@ extern "C" tpx_stage* tpx_stage_compile(const char* hip_source,
         {"tpx_scan_block", &st->k_scan_block, true},
         {"tpx_scan_add", &st->k_scan_add, true},
         {"tpx_emit_kept", &st->k_emit_kept, true},
+        {"tpx_pair_total", &st->k_pair_total, true},
         {"tpx_csv_chunk_stats", &st->k_csv_chunk, false},
         {"tpx_csv_select_counts", &st->k_csv_sel, false},
         {"tpx_csv_emit_rows", &st->k_csv_rows, false},
@@ -330,6 +332,13 @@ static Persist g_persist[64];
 // two extra per-device streams for chunk pipelining (scans/writes of chunk c
 // overlap the main kernel of chunk c+1)
 static hipStream_t g_cstream[64][2];
+// small pinned staging area per device for tiny D2H reads (scan totals,
+// counters): pageable 8B copies cost a hidden staging hop each
+static void* g_pinned[64];
+static void* pinned(int dev) {
+    if (!g_pinned[dev]) (void)hipHostMalloc(&g_pinned[dev], 1024);
+    return g_pinned[dev];
+}
 static hipStream_t cstream(int dev, int i) {
     if (!g_cstream[dev][i])
         (void)hipStreamCreateWithFlags(&g_cstream[dev][i], hipStreamNonBlocking);
@@ -374,7 +383,7 @@ static int dev_scan(tpx_stage* st, hipStream_t stream, long long* d_in,
     int dev = cur_device();
     const long long BLOCK = 2048;  // TPX_SCAN_BLOCK
     long long nblocks = (n + BLOCK - 1) / BLOCK;
-    if (nblocks == 0) { *total = 0; return 0; }
+    if (nblocks == 0) { if (total) *total = 0; return 0; }
     ARENA_TAKE(sums, (size_t)nblocks * 8);
     void* in_p = d_in;
     void* out_p = d_out;
@@ -389,6 +398,7 @@ static int dev_scan(tpx_stage* st, hipStream_t stream, long long* d_in,
         void* a2[] = {&out_p, &sums_scan, &n};
         if (launch(st->k_scan_add, (unsigned)nblocks, 256, stream, a2)) return -1;
     }
+    if (!total) return 0;  // caller fetches totals itself (tpx_pair_total)
     long long last_out = 0, last_in = 0;
     HIP_CHECK(hipMemcpyAsync(&last_out, d_out + (n - 1), 8,
                              hipMemcpyDeviceToHost, stream));
@@ -494,21 +504,26 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
     // per-chunk biased outv pointer tables (chunk rows index columnar slot i
     // locally); chunk 0 table == the unbiased one
     void* d_outv_c[8] = {d_outv};
-    for (int c = 1; c < C; ++c) {
-        std::vector<void*> ov(outv.size());
-        for (int k = 0; k < nout; ++k) {
-            const ColDesc& col = D.out_cols[(size_t)k];
-            ov[(size_t)k * 3 + 0] = (char*)outv[(size_t)k * 3 + 0] + cstart[c] * 8;
-            ov[(size_t)k * 3 + 1] = col.kind == K_STR
-                ? (char*)outv[(size_t)k * 3 + 1] + cstart[c] * 4 : nullptr;
-            ov[(size_t)k * 3 + 2] = col.opt
-                ? (char*)outv[(size_t)k * 3 + 2] + cstart[c] : nullptr;
+    if (C > 1) {
+        std::vector<void*> ovall(outv.size() * (size_t)(C - 1));
+        for (int c = 1; c < C; ++c) {
+            void** ov = ovall.data() + outv.size() * (size_t)(c - 1);
+            for (int k = 0; k < nout; ++k) {
+                const ColDesc& col = D.out_cols[(size_t)k];
+                ov[(size_t)k * 3 + 0] =
+                    (char*)outv[(size_t)k * 3 + 0] + cstart[c] * 8;
+                ov[(size_t)k * 3 + 1] = col.kind == K_STR
+                    ? (char*)outv[(size_t)k * 3 + 1] + cstart[c] * 4 : nullptr;
+                ov[(size_t)k * 3 + 2] = col.opt
+                    ? (char*)outv[(size_t)k * 3 + 2] + cstart[c] : nullptr;
+            }
         }
-        d_outv_c[c] = g_arena[dev].take(ov.size() * sizeof(void*) + 8);
-        if (!d_outv_c[c]) return -1;
-        HIP_CHECK(hipMemcpyAsync(d_outv_c[c], ov.data(),
-                                 ov.size() * sizeof(void*),
+        void* blk = g_arena[dev].take(ovall.size() * sizeof(void*) + 8);
+        if (!blk) return -1;
+        HIP_CHECK(hipMemcpyAsync(blk, ovall.data(), ovall.size() * sizeof(void*),
                                  hipMemcpyHostToDevice, stream));
+        for (int c = 1; c < C; ++c)
+            d_outv_c[c] = (char*)blk + outv.size() * sizeof(void*) * (size_t)(c - 1);
     }
 
     // +thread-chunk slack per LAUNCH: <=2048 blocks x 256 threads x 256 B
@@ -580,11 +595,24 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
                 long long rows_c = 0, bytes_c = 0;
                 hipEventRecord(cev[(size_t)c * 5 + 1], sc);
                 if (dev_scan(st, sc, (long long*)keep01_c,
-                             (long long*)kscan_c, ccnt[c], &rows_c))
+                             (long long*)kscan_c, ccnt[c], nullptr))
                     return -1;
                 if (dev_scan(st, sc, (long long*)sizes_c, (long long*)sscan_c,
-                             ccnt[c], &bytes_c))
+                             ccnt[c], nullptr))
                     return -1;
+                {   // both totals with one tiny kernel + one pinned D2H
+                    void* slot = g_arena[dev].take(16);
+                    if (!slot) return -1;
+                    void* ta[] = {&keep01_c, &kscan_c, &ccnt[c], &sizes_c,
+                                  &sscan_c, &ccnt[c], &slot};
+                    if (launch(st->k_pair_total, 1, 64, sc, ta)) return -1;
+                    long long* host = (long long*)pinned(dev);
+                    HIP_CHECK(hipMemcpyAsync(host, slot, 16,
+                                             hipMemcpyDeviceToHost, sc));
+                    HIP_CHECK(hipStreamSynchronize(sc));
+                    rows_c = host[0];
+                    bytes_c = host[1];
+                }
                 hipEventRecord(cev[(size_t)c * 5 + 2], sc);
                 chunk_out[c] = g_arena[dev].take((size_t)bytes_c + 16);
                 if (!chunk_out[c]) return -1;
@@ -610,7 +638,8 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
         }
         // overflow check (all mains have completed: each chunk's scans synced
         // its stream for C>1; explicit sync below covers C==1)
-        unsigned long long counters[20] = {0};
+        unsigned long long* counters =
+            (unsigned long long*)((char*)pinned(dev) + 256);
         HIP_CHECK(hipMemcpyAsync(counters, P.counters, 160,
                                  hipMemcpyDeviceToHost, stream));
         HIP_CHECK(hipStreamSynchronize(stream));

@@ -1134,6 +1134,21 @@ extern "C" __global__ void tpx_emit_kept(const unsigned char* __restrict__ keep,
         if (keep[i]) kept_idx[keep_scan[i]] = i;
 }
 
+// one-thread helper: totals of two exclusive scans (in[n-1] + out[n-1]) into a
+// 16B slot, so the host fetches both chunk totals with ONE tiny D2H copy
+extern "C" __global__ void tpx_pair_total(const long long* __restrict__ a_in,
+                                          const long long* __restrict__ a_out,
+                                          long long na,
+                                          const long long* __restrict__ b_in,
+                                          const long long* __restrict__ b_out,
+                                          long long nb,
+                                          long long* __restrict__ slot) {
+    if (threadIdx.x == 0 && blockIdx.x == 0) {
+        slot[0] = na > 0 ? a_in[na - 1] + a_out[na - 1] : 0;
+        slot[1] = nb > 0 ? b_in[nb - 1] + b_out[nb - 1] : 0;
+    }
+}
+
 extern "C" __global__ void tpx_scan_block(const long long* __restrict__ in,
                                           long long* __restrict__ out,
                                           long long* __restrict__ block_sums,
