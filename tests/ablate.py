@@ -39,18 +39,20 @@ def main():
     glib.lib.tpx_dev_upload(dev, buf, len(body))
     del buf
 
-    sample = body[:1 << 20]
     from tests.zillow_data import ZILLOW_COLS
-    _h, names, col_types = csvio.sniff(sample, [""], 0.9, False,
-                                       ZILLOW_COLS, b",")
+    from tuplex_amd import ttypes as T
+    # EXACTLY the bench stage shape (explicit schema, csv sink)
+    col_types = [T.STR, T.STR, T.STR, T.STR, T.F64, T.STR, T.STR, T.STR,
+                 T.STR, T.STR]
     names = list(ZILLOW_COLS)
 
+    zo = zillow_ops()
     variants = [
         ("passthrough", [], None),
-        ("udf_nofilter", [op for op in zillow_ops()
+        ("udf_nofilter", [op for op in zo
                           if op[0] in ("withColumn", "mapColumn")], None),
-        ("full", zillow_ops(), None),
-        ("full_noLDS", zillow_ops(), 16),   # span cap 16B -> all-global path,
+        ("full", zo, None),
+        ("full_noLDS", zo, 16),             # span cap 16B -> all-global path,
                                             # tiny smem -> full occupancy
     ]
     saved_cap = codegen.StageCodegen.SPAN_CAP
@@ -58,7 +60,7 @@ def main():
         codegen.StageCodegen.SPAN_CAP = cap if cap else saved_cap
         sp = plan.build_stage(col_types, names, ops)
         assert sp.compilable, (label, sp.why_not_compilable)
-        src, desc = codegen.generate_stage(sp, source="csv", sink="mem",
+        src, desc = codegen.generate_stage(sp, source="csv", sink="csv",
                                            csv_info={"null_values": [""]})
         stage = glib.compile_stage(src, desc)
         run(stage, glib, dev, len(body), label)
