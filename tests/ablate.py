@@ -66,6 +66,23 @@ def main():
         run(stage, glib, dev, len(body), label)
     codegen.StageCodegen.SPAN_CAP = saved_cap
 
+    if len(sys.argv) > 2 and sys.argv[2] == "ops":
+        # cumulative per-op ablation: prefix of the Z1 chain, csv sink
+        prev = None
+        for k in range(len(zo) + 1):
+            ops = zo[:k]
+            # selectColumns only valid at the end; skip bare filter prefixes ok
+            sp = plan.build_stage(col_types, names, ops)
+            if not sp.compilable:
+                print("%2d ops: not compilable (%s)" % (k, sp.why_not_compilable))
+                continue
+            src, desc = codegen.generate_stage(sp, source="csv", sink="csv",
+                                               csv_info={"null_values": [""]})
+            stage = glib.compile_stage(src, desc)
+            run(stage, glib, dev, len(body),
+                "+%s" % (zo[k - 1][0] + ":" + str(zo[k - 1][1])[:18]
+                         if k else "none"))
+
 
 if __name__ == "__main__":
     main()
