@@ -55,15 +55,24 @@ def main():
         ("full_noLDS", zo, 16),             # span cap 16B -> all-global path,
                                             # tiny smem -> full occupancy
     ]
+    def build(ops):
+        sp = plan.build_stage(col_types, names, ops)
+        assert sp.compilable, sp.why_not_compilable
+        try:
+            src, desc = codegen.generate_stage(sp, source="csv", sink="csv",
+                                               csv_info={"null_values": [""]})
+            sink = "csv"
+        except codegen.CodegenError:   # f64 outputs: mem sink for the probe
+            src, desc = codegen.generate_stage(sp, source="csv", sink="mem",
+                                               csv_info={"null_values": [""]})
+            sink = "mem"
+        return glib.compile_stage(src, desc), sink
+
     saved_cap = codegen.StageCodegen.SPAN_CAP
     for label, ops, cap in variants:
         codegen.StageCodegen.SPAN_CAP = cap if cap else saved_cap
-        sp = plan.build_stage(col_types, names, ops)
-        assert sp.compilable, (label, sp.why_not_compilable)
-        src, desc = codegen.generate_stage(sp, source="csv", sink="csv",
-                                           csv_info={"null_values": [""]})
-        stage = glib.compile_stage(src, desc)
-        run(stage, glib, dev, len(body), label)
+        stage, sink = build(ops)
+        run(stage, glib, dev, len(body), label + ("[mem]" if sink == "mem" else ""))
     codegen.StageCodegen.SPAN_CAP = saved_cap
 
     if len(sys.argv) > 2 and sys.argv[2] == "ops":
@@ -72,16 +81,11 @@ def main():
         for k in range(len(zo) + 1):
             ops = zo[:k]
             # selectColumns only valid at the end; skip bare filter prefixes ok
-            sp = plan.build_stage(col_types, names, ops)
-            if not sp.compilable:
-                print("%2d ops: not compilable (%s)" % (k, sp.why_not_compilable))
-                continue
-            src, desc = codegen.generate_stage(sp, source="csv", sink="csv",
-                                               csv_info={"null_values": [""]})
-            stage = glib.compile_stage(src, desc)
+            stage, sink = build(ops)
             run(stage, glib, dev, len(body),
-                "+%s" % (zo[k - 1][0] + ":" + str(zo[k - 1][1])[:18]
-                         if k else "none"))
+                ("+%s" % (zo[k - 1][0] + ":" + getattr(zo[k - 1][-1], "__name__",
+                                                       str(zo[k - 1][1]))[:16])
+                 if k else "none") + ("[mem]" if sink == "mem" else ""))
 
 
 if __name__ == "__main__":
