@@ -191,7 +191,29 @@ class StageCodegen:
     def _peephole(self, em, n, rc, opid):
         """Pattern rewrites that must run BEFORE strict arg evaluation.
         capitalize: x[0].upper() + x[1:].lower() -> tpx_capitalize_ix(x)
-        (one alloc + one pass instead of three; IndexError on empty x kept)."""
+        (one alloc + one pass instead of three; IndexError on empty x kept).
+        int-drop: int(s.replace(c, '')) -> tpx_int_drop(s, c) (no alloc)."""
+        if n["op"] == "int_str":
+            rep = n["args"][0]
+            if (rep["op"] == "replace" and rep["args"][1]["op"] == "const"
+                    and isinstance(rep["args"][1].get("v"), str)
+                    and len(rep["args"][1]["v"]) == 1
+                    and ord(rep["args"][1]["v"]) < 128
+                    and rep["args"][2]["op"] == "const"
+                    and rep["args"][2].get("v") == ""):
+                key = ("int_drop", repr(rep["args"][1]["v"]),
+                       self._struct_key(rep["args"][0]))
+                hit = em.lookup_key(key)
+                if hit is not None:
+                    return hit
+                x, _ = self.emit_expr(em, rep["args"][0], rc, opid)
+                r = em.fresh()
+                em.w("long long %s = tpx_int_drop(heap, %s, %d, %s, &_ec);"
+                     % (r, x, ord(rep["args"][1]["v"]),
+                        self.lit(rep["args"][1]["v"])))
+                self._check(em, opid)
+                return em.memo_key(key, (r, None))
+            return None
         if n["op"] != "concat":
             return None
         a0, a1 = n["args"]

@@ -383,6 +383,30 @@ __device__ __forceinline__ tstr tpx_capitalize_ix(TpxHeap& h, const tstr s, int*
     return tstr{d, s.n};
 }
 
+// int(s.replace(ch, '')) fused (codegen peephole): for short cells the dropped
+// copy goes through a stack buffer (no heap alloc, no extra pass); the >31B
+// tail falls back to replace+parse. Exact by construction: the same bytes that
+// replace() would produce feed the same tpx_int_str.
+__device__ __forceinline__ tstr tpx_replace(TpxHeap& h, const tstr s, const tstr a,
+                                            const tstr b, int* ec);
+__device__ __forceinline__ long long tpx_int_str(const tstr s, int* ec);
+__device__ __forceinline__ long long tpx_int_drop(TpxHeap& h, const tstr s,
+                                                  char drop, const tstr drop_s,
+                                                  int* ec) {
+    if (s.n <= 31) {
+        char buf[32];
+        long long m = 0;
+        for (long long i = 0; i < s.n; ++i) {
+            char c = s.p[i];
+            if (c != drop) buf[m++] = c;
+        }
+        return tpx_int_str(tstr{buf, m}, ec);
+    }
+    tstr r = tpx_replace(h, s, drop_s, tstr{s.p, 0}, ec);
+    if (*ec) return 0;
+    return tpx_int_str(r, ec);
+}
+
 __device__ __forceinline__ tstr tpx_swapcase(TpxHeap& h, const tstr s, int* ec) {
     char* d = tpx_alloc(h, s.n);
     if (!d) { *ec = EC_MEMORYERROR; return tstr{s.p, 0}; }
@@ -639,10 +663,31 @@ __device__ __forceinline__ tstr tpx_str_none() { return tstr{"None", 4}; }
 // ---- CSV output cell (RFC-4180: quote iff cell contains delim/quote/CR/LF;
 //      '"' doubled) ---------------------------------------------------------------
 
+// SWAR: 8 bytes per step (called twice per output string — size pass in the
+// main kernel, then the write kernel — so the byte loop was ~2x70B of serial
+// dependent loads per kept row). Head/tail stay byte-wise: no over-read.
 __device__ __forceinline__ bool tpx_csv_needs_quote(const tstr s, long long* nquotes) {
     bool need = false;
     long long q = 0;
-    for (long long i = 0; i < s.n; ++i) {
+    long long i = 0;
+    while (i < s.n && (((unsigned long long)(s.p + i)) & 7)) {
+        char c = s.p[i];
+        if (c == '"') { ++q; need = true; }
+        else if (c == ',' || c == '\n' || c == '\r') need = true;
+        ++i;
+    }
+    for (; i + 8 <= s.n; i += 8) {
+        unsigned long long v = *(const unsigned long long*)(s.p + i);
+        unsigned long long hq =
+            tpx_swar_zero(v ^ (TPX_SWAR_ONE * (unsigned long long)'"'));
+        unsigned long long ho =
+            tpx_swar_zero(v ^ (TPX_SWAR_ONE * (unsigned long long)',')) |
+            tpx_swar_zero(v ^ (TPX_SWAR_ONE * (unsigned long long)'\n')) |
+            tpx_swar_zero(v ^ (TPX_SWAR_ONE * (unsigned long long)'\r'));
+        if (hq) { q += __builtin_popcountll(hq); need = true; }
+        if (ho) need = true;
+    }
+    for (; i < s.n; ++i) {
         char c = s.p[i];
         if (c == '"') { ++q; need = true; }
         else if (c == ',' || c == '\n' || c == '\r') need = true;
