@@ -839,7 +839,11 @@ class StageCodegen:
     def _main_kernel(self, in_types, out_types):
         L = []
         L.append("#define TPX_SPAN_CAP %d" % self.SPAN_CAP)
-        L.append('extern "C" __global__ void tpx_stage_main(')
+        # occupancy is LDS-limited to ~2.5 waves/SIMD (5 blocks x 32 KiB); tell
+        # the compiler (min 2 waves/SIMD) so it uses the real VGPR budget
+        # instead of spilling ~284 B/thread to scratch at a 64-VGPR target
+        L.append('extern "C" __global__ void __launch_bounds__(128, 2)'
+                 ' tpx_stage_main(')
         L.append("    const unsigned char* __restrict__ in_data,")
         L.append("    const long long* __restrict__ in_offs,")
         L.append("    long long n, long long row0,")
