@@ -840,10 +840,13 @@ class StageCodegen:
         L = []
         L.append("#define TPX_SPAN_CAP %d" % self.SPAN_CAP)
         # occupancy is LDS-limited to ~2.5 waves/SIMD (5 blocks x 32 KiB); tell
-        # the compiler (min 2 waves/SIMD) so it uses the real VGPR budget
-        # instead of spilling ~284 B/thread to scratch at a 64-VGPR target
-        L.append('extern "C" __global__ void __launch_bounds__(128, 2)'
-                 ' tpx_stage_main(')
+        # the compiler the real VGPR budget instead of spilling ~284 B/thread
+        # to scratch at a 64-VGPR default target (waves-per-SIMD tunable via
+        # TPX_MAIN_LB: 0 = no bound)
+        import os as _os
+        lb = int(_os.environ.get("TPX_MAIN_LB", "2"))
+        bound = " __launch_bounds__(128, %d)" % lb if lb else ""
+        L.append('extern "C" __global__ void%s tpx_stage_main(' % bound)
         L.append("    const unsigned char* __restrict__ in_data,")
         L.append("    const long long* __restrict__ in_offs,")
         L.append("    long long n, long long row0,")
