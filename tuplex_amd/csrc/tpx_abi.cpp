@@ -21,6 +21,7 @@
 
 #include <algorithm>
 #include <cstdio>
+#include <functional>
 #include <cstring>
 #include <fstream>
 #include <map>
@@ -419,10 +420,15 @@ struct PayloadSrc {
     int64_t n_parts;
 };
 
+// pre_chunk(row_lo, row_hi, stream): optional hook launched on a chunk's
+// stream BEFORE its main kernel — the csv path uses it to emit that chunk's
+// row offsets just-in-time, overlapping the previous chunk's main kernel.
+using PreChunkFn = std::function<int(long long, long long, hipStream_t)>;
+
 static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
                         long long row0, long long in_bytes, tpx_result* res,
                         const PayloadSrc& psrc, hipStream_t stream,
-                        int64_t flags = 0) {
+                        int64_t flags = 0, const PreChunkFn* pre_chunk = nullptr) {
     const StageDesc& D = st->desc;
     int dev = cur_device();
     Persist& P = g_persist[dev];
@@ -556,6 +562,9 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
         for (int c = 0; c < C; ++c) {
             hipStream_t sc = S[c & 1];
             if (C > 1) hipStreamWaitEvent(sc, evm0, 0);
+            if (pre_chunk && attempt == 0 &&
+                (*pre_chunk)(cstart[c], cstart[c + 1], sc))
+                return -1;
             // main runs 128-thread blocks (2 waves x 16 KiB LDS staging =
             // 32 KiB -> 5 blocks/CU = 10 waves/CU); each wave owns 64 rows
             long long nc = ccnt[c];
@@ -1059,21 +1068,25 @@ static int64_t csv_boundary_and_core(tpx_stage* st, void* d_in_p, long long size
         return -1;
     if (nrows == 0) { empty_result(res, st->desc.sink == "mem"); return 0; }
     ARENA_TAKE(d_offs, ((size_t)nrows + 1) * 8);
-    {
-        void* args[] = {&d_in_p, (void*)&size, &nchunks, &d_qs, &d_base, &d_offs,
-                        &quotes_on};
-        if (launch(st->k_csv_rows, grid_w, 256, stream, args)) return -1;
-    }
     hipEventRecord(eb1, stream);
     HIP_CHECK(hipEventSynchronize(eb1));
     float ms = 0;
     hipEventElapsedTime(&ms, eb0, eb1);
-    res->t_boundary_ms = ms;
+    res->t_boundary_ms = ms;  // stats+scans+select; per-range emits overlap main
     (void)hipEventDestroy(eb0); (void)hipEventDestroy(eb1);
+
+    // row offsets are emitted per main-kernel chunk, on that chunk's stream,
+    // so emit(range c) overlaps main(chunk c-1) (ranged tpx_csv_emit_rows)
+    PreChunkFn emit_range = [&](long long lo, long long hi,
+                                hipStream_t sc) -> int {
+        void* args[] = {&d_in_p, (void*)&size, &nchunks, &d_qs, &d_base,
+                        &d_offs, &quotes_on, &lo, &hi};
+        return launch(st->k_csv_rows, grid_w, 256, sc, args);
+    };
 
     PayloadSrc psrc{host_bytes, nullptr, 0};
     return run_core(st, d_in_p, d_offs, nrows, first_global_row, size, res, psrc,
-                    stream, flags);
+                    stream, flags, &emit_range);
 }
 
 extern "C" int64_t tpx_stage_execute_csv(tpx_stage* st, const uint8_t* csv_bytes,

@@ -819,19 +819,26 @@ extern "C" __global__ void tpx_csv_select_counts(const long long* __restrict__ q
         rc[c] = (qscan[c] & 1) ? c1[c] : c0[c];
 }
 
+// row_lo/row_hi restrict emission to byte-chunks owning rows in [row_lo,row_hi)
+// so the offsets for one main-kernel chunk can be emitted just-in-time on that
+// chunk's stream (overlapping the previous chunk's main); chunks straddling a
+// range edge are re-emitted by the neighbouring call — idempotent writes.
 extern "C" __global__ void tpx_csv_emit_rows(const char* __restrict__ data,
                                              long long size, long long nchunks,
                                              const long long* __restrict__ qscan,
                                              const long long* __restrict__ base,
                                              long long* __restrict__ row_offs,
-                                             int quotes_on) {
+                                             int quotes_on, long long row_lo,
+                                             long long row_hi) {
     int lane = threadIdx.x & 63;
     int wid = threadIdx.x >> 6;
     int wpb = blockDim.x >> 6;
     long long wstride = (long long)gridDim.x * wpb;
-    if (blockIdx.x == 0 && threadIdx.x == 0) row_offs[0] = 0;
+    if (blockIdx.x == 0 && threadIdx.x == 0 && row_lo == 0) row_offs[0] = 0;
     for (long long c = (long long)blockIdx.x * wpb + wid; c < nchunks;
          c += wstride) {
+        if (base[c] >= row_hi) continue;
+        if (c + 1 < nchunks && base[c + 1] <= row_lo) continue;
         long long a = c * TPX_CSV_CHUNK + (long long)lane * TPX_CSV_LANE_BYTES;
         long long b = a + TPX_CSV_LANE_BYTES;
         if (a > size) a = size;
