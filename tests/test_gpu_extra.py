@@ -296,3 +296,37 @@ def test_chunk_pipelined_large_input(tmp_path):
     assert got == ref["output"] * 20
     for k, v in ref["exception_counts"].items():
         assert ds.exception_counts.get(k, 0) == v * 20, (k, v)
+
+
+def test_peephole_edge_parity(tmp_path):
+    """Edge rows through the rewritten fast paths: case-insensitive scans
+    (contains(lower(s), lit)), int(s.replace(',','')) stack/fallback parse,
+    capitalize peephole IndexError on empty city, >31-char numeric cells."""
+    from tests.test_codegen_compile import zillow_ops
+
+    cols = "title,address,city,state,postal_code,price,facts and features,real estate provider,url,sales_date"
+    rows = [
+        # uppercase keywords: lower() must still match (CI scan)
+        b'House For SALE,1 A St,boston,MA,2125,"$1,250,000","3 bds , 2 ba , 1,500 sqft",X,u1,2020',
+        b'CONDO FOR RENT,2 B St,ny,NY,10001,"$2,000/mo","2 bds , 1 ba , 900 sqft",X,u2,2020',
+        # empty city -> x[0] IndexError -> interpreter replay
+        b'house for sale,3 C St,,MA,2125,"$500,000","1 bd , 1 ba , 700 sqft",X,u3,2020',
+        # >31-char digit run with commas (int_drop falls back to the
+        # replace+parse path; leading zeros keep the value inside i64)
+        b'house for sale,4 D St,salem,MA,1970,"$1,000,000","2 bds , 2 ba , 0,000,000,000,000,000,000,001,234,567 sqft",X,u4,2020',
+        # no offer keyword at all -> extractOffer returns lowered title
+        b'Mystery Listing,5 E St,lynn,MA,1901,"$750,000","4 bds , 3 ba , 2,000 sqft",X,u5,2020',
+        # mixed-case condo (extractType lowers via CI scan)
+        b'Nice CoNdO for Sale,6 F St,dover,NH,3820,"$600,000","3 bds , 2 ba , 1,600 sqft",X,u6,2020',
+    ] * 400  # enough rows to exercise real wave shapes
+    data = cols.encode() + b"\n" + b"\n".join(rows) + b"\n"
+    p = _write(tmp_path, data, "edge.csv")
+
+    ctx = tuplex_amd.Context()
+    ds = apply_ops(ctx.csv(p), zillow_ops())
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    ref = pyoracle_csv.run_csv_pipeline(data, zillow_ops())
+    assert got == ref["output"]
+    assert ds.exception_counts == ref["exception_counts"]
+    assert len(got) > 0
