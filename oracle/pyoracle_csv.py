@@ -206,16 +206,22 @@ def _is_opt(t):
     return isinstance(t, tuple) and t[0] == "opt"
 
 
-def _fast_parse_row(cells, flags, col_types, null_values):
-    """Normal-case typed parse. Returns ("row", tuple) or ("bad", ec_name)."""
+def _fast_parse_row(cells, flags, col_types, null_values, used=None):
+    """Normal-case typed parse. Returns ("row", tuple) or ("bad", ec_name).
+    used: projection-pushdown column set (LogicalOptimizer selectionPushdown /
+    CSVParseRowGenerator willBeSerialized): unused columns keep their raw cell
+    text and never fail a typed parse."""
     if flags & 6:
         return ("bad", "escape/structure")
     if len(cells) != len(col_types):
         return ("bad", "cellcount")
     vals = []
-    for c, t in zip(cells, col_types):
+    for ci, (c, t) in enumerate(zip(cells, col_types)):
         s = c.decode("utf-8", "replace")
         base = _deopt(t)
+        if used is not None and ci not in used:
+            vals.append(s)
+            continue
         if _is_opt(t) and s in null_values:
             vals.append(None)
             continue
@@ -251,7 +257,7 @@ _BadParse.__name__ = "BadParseStringInput"
 
 
 def _replay_row(raw_line: bytes, col_types, null_values, row_ops, columns,
-                delim: str = ","):
+                delim: str = ",", used=None):
     """Interpreter replay: full RFC-4180 parse + CPython conversions (matches the
     product's csvio.replay_csv_row rules)."""
     text = raw_line.decode("utf-8", "replace").rstrip("\n").rstrip("\r")
@@ -262,8 +268,11 @@ def _replay_row(raw_line: bytes, col_types, null_values, row_ops, columns,
     if len(cells) != len(col_types):
         return ("exc", _BadParse("cellcount"))
     vals = []  # noqa: replay path
-    for c, t in zip(cells, col_types):
+    for ci, (c, t) in enumerate(zip(cells, col_types)):
         base = _deopt(t)
+        if used is not None and ci not in used:
+            vals.append(c)
+            continue
         if _is_opt(t) and c in null_values:
             vals.append(None)
             continue
@@ -287,7 +296,7 @@ def _replay_row(raw_line: bytes, col_types, null_values, row_ops, columns,
 
 def run_csv_pipeline(csv_bytes: bytes, ops, columns=None, header=None,
                      null_values=None, threshold=0.9, sink="collect",
-                     delimiter=None, type_hints=None):
+                     delimiter=None, type_hints=None, used_cols=None):
     """Full oracle CSV pipeline. sink='collect' -> values; 'csv' -> output text
     (bytes, incl. header line)."""
     if null_values is None:
@@ -328,13 +337,14 @@ def run_csv_pipeline(csv_bytes: bytes, ops, columns=None, header=None,
 
     for i, line in enumerate(split_rows(data)):
         cells, flags = split_cells(line, delim)
-        pr = _fast_parse_row(cells, flags, col_types, null_values)
+        pr = _fast_parse_row(cells, flags, col_types, null_values,
+                             used=used_cols)
         if pr[0] == "row":
             val = pr[1][0] if len(col_types) == 1 else pr[1]  # 1-col = scalar row
             r = pyoracle.process_row(val, row_ops, names, fast=True)
         else:
             r = _replay_row(line, col_types, null_values, row_ops, names,
-                            delim.decode())
+                            delim.decode(), used=used_cols)
         if r[0] == "row":
             v = r[1]
             if isinstance(v, tuple) and len(v) == 1:

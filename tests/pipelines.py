@@ -304,3 +304,20 @@ def precompile_all(verbose=False):
             print("precompiled:", name)
     n += precompile_csv(verbose=verbose)
     return n
+
+
+def csv_used_cols(data, ops, columns=None, header=None, delimiter=b",",
+                  null_values=None):
+    """The stage's projection-pushdown column set (plan._used_source_columns)
+    for feeding the oracle's used_cols — the pushdown decision is optimizer
+    metadata (LogicalOptimizer selectionPushdown), an INPUT to the semantics,
+    like the delimiter."""
+    from tuplex_amd import csvio, plan
+    sample = data[:1 << 20]
+    nl = sample.rfind(b"\n")
+    if nl >= 0:
+        sample = sample[:nl + 1]
+    _h, names, col_types = csvio.sniff(sample, null_values or [""], 0.9,
+                                       header, columns, delimiter)
+    sp = plan.build_stage(col_types, names, ops)
+    return sp.used_source_cols

@@ -56,7 +56,10 @@ def test_flights_wide_csv(tmp_path):
     ds = apply_ops(ctx.csv(path), X.flights_ops())
     got = ds.collect()
     assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
-    ref = pyoracle_csv.run_csv_pipeline(data, X.flights_ops())
+    from tests.pipelines import csv_used_cols
+    ref = pyoracle_csv.run_csv_pipeline(
+        data, X.flights_ops(),
+        used_cols=csv_used_cols(data, X.flights_ops()))
     assert got == ref["output"]
     assert ds.exception_counts == ref["exception_counts"]
     assert len(got) > 100
@@ -330,3 +333,37 @@ def test_peephole_edge_parity(tmp_path):
     assert got == ref["output"]
     assert ds.exception_counts == ref["exception_counts"]
     assert len(got) > 0
+
+
+def test_projection_pushdown(tmp_path):
+    """Unused columns are cell-walked but not value-parsed: garbage in an
+    unparsed numeric column diverts NO row (LogicalOptimizer selectionPushdown;
+    CSVParseRowGenerator willBeSerialized). The same pipeline without pushdown
+    (oracle used_cols=None) WOULD divert those rows."""
+    from tests.pipelines import csv_used_cols
+
+    lines = [b"a,b,junk,c"]
+    for i in range(5000):
+        junk = b"NOT_A_NUMBER" if i % 97 == 0 else b"%d" % i  # rare: keeps the sniffed type i64
+        lines.append(b"%d,%d,%s,x%d" % (i, i * 2, junk, i))
+    data = b"\n".join(lines) + b"\n"
+    p = _write(tmp_path, data, "pd.csv")
+
+    def use(x):
+        return (x["a"] + x["b"], x["c"])
+
+    ops = [("map", use)]
+    ctx = tuplex_amd.Context()
+    ds = apply_ops(ctx.csv(p), ops)
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    assert ds.exception_counts == {}          # junk column never parsed
+    assert len(got) == 5000
+    used = csv_used_cols(data, ops)
+    assert sorted(used) == [0, 1, 3]
+    ref = pyoracle_csv.run_csv_pipeline(data, ops, used_cols=used)
+    assert got == ref["output"]
+    assert ref["exception_counts"] == {}
+    # counter-check: without pushdown the junk rows DO divert in the oracle
+    ref_full = pyoracle_csv.run_csv_pipeline(data, ops, used_cols=None)
+    assert sum(ref_full["exception_counts"].values()) > 0

@@ -1004,6 +1004,10 @@ class StageCodegen:
                                  for nv in null_values) or "false"
             return checks
 
+        # projection pushdown: unused columns are still cell-walked (structure
+        # + column-count errors keep reference semantics) but their typed
+        # parse / null check is skipped (plan._used_source_columns)
+        used = getattr(self.sp, "used_source_cols", None)
         for idx, t in enumerate(in_types):
             base = T.deopt(t)
             opt = T.is_opt(t)
@@ -1013,6 +1017,19 @@ class StageCodegen:
             L.append("      else { cur = tpx_csv_next_cell(cur, rend, &cl%d, &m, %s);"
                      " avail = m; badf |= cl%d.flags; }" % (idx, delim_c, idx))
             L.append("    }")
+            if used is not None and idx not in used:
+                if opt:
+                    L.append("    bool c%d_n = false;  // unused (pushdown)" % idx)
+                if base == T.STR:
+                    L.append("    tstr c%d{cl%d.p, cl%d.n};" % (idx, idx, idx))
+                    L.append("    if (prc) c%d = tstr{rp, 0};" % idx)
+                elif base == T.I64:
+                    L.append("    long long c%d = 0;  // unused (pushdown)" % idx)
+                elif base == T.F64:
+                    L.append("    double c%d = 0.0;  // unused (pushdown)" % idx)
+                else:
+                    L.append("    bool c%d = false;  // unused (pushdown)" % idx)
+                continue
             if opt:
                 L.append("    bool c%d_n = !prc && (%s);" % (idx, null_check(idx)))
             guard = ("!prc && !c%d_n" % idx) if opt else "!prc"

@@ -269,7 +269,7 @@ def sniff(sample: bytes, null_values: List[str], threshold: float,
 # ---- replay of exception rows ----------------------------------------------------
 
 def replay_csv_row(raw_line: bytes, col_types, null_values, logical_ops, columns,
-                   delim: str = ","):
+                   delim: str = ",", used=None):
     """Interpreter replay of a diverted CSV row (BADPARSE semantics,
     ResolveTask.cc:459 interpreter path with parse_cells). Full RFC-4180 parse
     (with unescaping) then CPython-typed conversion; structure/convert failures
@@ -284,8 +284,11 @@ def replay_csv_row(raw_line: bytes, col_types, null_values, logical_ops, columns
         return ("exc", _BadParse("cell count %d != %d" % (len(cells),
                                                           len(col_types))))
     vals = []
-    for c, t in zip(cells, col_types):
+    for ci, (c, t) in enumerate(zip(cells, col_types)):
         base = T.deopt(t)
+        if used is not None and ci not in used:
+            vals.append(c)  # pushdown: unused column stays unconverted
+            continue
         if T.is_opt(t) and c in null_values:
             vals.append(None)
             continue
@@ -484,7 +487,8 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
                                             src.null_values, logical_ops)
                     else:
                         r = replay_csv_row(payload, col_types, src.null_values,
-                                           logical_ops, names, delim.decode())
+                                           logical_ops, names, delim.decode(),
+                                           used=sp.used_source_cols)
                     if r[0] == "row":
                         replayed[row] = r[1]
                     elif r[0] == "exc":

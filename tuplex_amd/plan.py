@@ -52,6 +52,7 @@ class StageProgram:
         self.agg_key_idx = None
         self.agg_key_type = None
         self.agg_unique = False   # unique(): by-key count, keys-only output
+        self.used_source_cols = None  # projection pushdown (None = parse all)
         self.compilable = True
         self.why_not_compilable = None
 
@@ -236,7 +237,68 @@ def build_stage(input_types, input_columns, logical_ops) -> StageProgram:
         op.out_types = list(cur_types)
         op.out_columns = list(cur_cols) if cur_cols else None
         sp.ops.append(op)
+    sp.used_source_cols = _used_source_columns(sp) if sp.compilable else None
     return sp
+
+
+def _tir_input_idxs(node, acc):
+    if node is None:
+        return
+    if node["op"] == "input":
+        acc.add(node["i"])
+    for a in node["args"]:
+        _tir_input_idxs(a, acc)
+
+
+def _used_source_columns(sp):
+    """Backward column lineage: the set of SOURCE columns whose typed values the
+    stage consumes (selection/projection pushdown — reference:
+    core/src/logical/LogicalOptimizer.cc selectionPushdown via
+    UDF::getAccessedColumns; CSVParseRowGenerator.cc parses a cell's value only
+    when willBeSerialized). The CSV loader still walks every cell (structure +
+    column-count errors keep reference semantics) but skips the typed parse of
+    unused columns — a malformed value there diverts no row, exactly like the
+    reference with pushdown on."""
+    if sp.agg_unique:
+        return frozenset(range(len(sp.input_types)))
+    if sp.agg_expr is not None:
+        used = set()
+        _tir_input_idxs(sp.agg_expr, used)
+        if sp.agg_key_idx is not None:
+            used.add(sp.agg_key_idx)
+    else:
+        final = sp.ops[-1].out_types if sp.ops else sp.input_types
+        used = set(range(len(final)))
+    for op in reversed(sp.ops):
+        n_in = len(op.in_types)
+        if op.kind == "map":
+            if op.tir is None:
+                return frozenset(range(len(sp.input_types)))
+            used = set()
+            _tir_input_idxs(op.tir, used)
+        elif op.kind == "filter":
+            if op.tir is None:
+                return frozenset(range(len(sp.input_types)))
+            used = set(used)
+            _tir_input_idxs(op.tir, used)
+        elif op.kind == "mapColumn":
+            pass  # positions map identity; the UDF consumes only its column
+        elif op.kind == "withColumn":
+            cols = op.in_columns or ["column%d" % k for k in range(n_in)]
+            i = cols.index(op.col) if op.col in cols else n_in
+            fed = i in used
+            used = {p for p in used if p != i}
+            if fed:
+                if op.tir is None:
+                    return frozenset(range(len(sp.input_types)))
+                _tir_input_idxs(op.tir, used)
+        elif op.kind == "selectColumns":
+            used = {op.sel_idxs[p] for p in used}
+        elif op.kind == "renameColumn":
+            pass
+        else:
+            return frozenset(range(len(sp.input_types)))
+    return frozenset(used)
 
 
 def _compile_aggregate(sp, opid, agg_fn, initial, cur_types, cur_cols):
