@@ -147,6 +147,21 @@ int64_t tpx_stage_execute_csv_dev(tpx_stage* stage,
                                   int64_t first_global_row, int64_t flags,
                                   tpx_result* result);
 
+/* Execute over a COLUMNAR (Arrow-layout) input already resident on device —
+ * the ORC ingest path (reference: io/src/OrcTypes.cc + physical OrcReader;
+ * SURVEY.md §8f-2). `slots` holds 3 device pointers per input column:
+ *   [3k+0] i64/f64 value array (stride 8), bool array (stride 1), or for
+ *          strings the int64 offsets array (n_rows+1 entries);
+ *   [3k+1] string data base pointer (strings only, else NULL);
+ *   [3k+2] null mask (uint8, 1 = null; optional columns only, else NULL).
+ * Unused (projection-pushed-down) columns may pass NULL slots. Exception
+ * payloads are empty (no row bytes exist); the caller replays exception rows
+ * from its own copy of the table by row index. */
+int64_t tpx_stage_execute_col(tpx_stage* stage, void* const* slots,
+                              int64_t n_slots, int64_t n_rows,
+                              int64_t in_bytes, int64_t first_global_row,
+                              int64_t flags, tpx_result* result);
+
 /* Generated-source introspection (debug / judge). Returns the HIP source the stage
  * was compiled from (owned by the stage). */
 const char* tpx_stage_source(const tpx_stage* stage);

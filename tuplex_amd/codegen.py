@@ -837,6 +837,8 @@ class StageCodegen:
     SPAN_CAP = 16384  # bytes per wave; 4 waves/block -> 64 KiB LDS, 2 blocks/CU
 
     def _main_kernel(self, in_types, out_types):
+        if self.source == "col":
+            return self._main_kernel_col(in_types, out_types)
         L = []
         L.append("#define TPX_SPAN_CAP %d" % self.SPAN_CAP)
         # occupancy is LDS-limited to ~2.5 waves/SIMD (5 blocks x 32 KiB); tell
@@ -897,10 +899,36 @@ class StageCodegen:
         L.append("}")
         return "\n".join(L)
 
+    def _main_kernel_col(self, in_types, out_types):
+        """Columnar-source main kernel: typed coalesced loads, no LDS staging,
+        plain grid-stride over rows (same fixed signature; in_offs unused)."""
+        L = []
+        L.append('extern "C" __global__ void tpx_stage_main(')
+        L.append("    const unsigned char* __restrict__ in_data,")
+        L.append("    const long long* __restrict__ in_offs,")
+        L.append("    long long n, long long row0,")
+        L.append("    char* heap_base, unsigned long long* heap_cursor,"
+                 " unsigned long long heap_cap,")
+        L.append("    unsigned char* __restrict__ keep, long long* __restrict__ keep01,")
+        L.append("    long long* __restrict__ sizes,")
+        L.append("    long long* exc_buf, unsigned long long* exc_count,"
+                 " unsigned long long exc_cap,")
+        L.append("    void** outv) {")
+        L.append("  TpxHeap heap{heap_base, heap_cursor, heap_cap, nullptr, nullptr};")
+        L.append("  long long stride = (long long)gridDim.x * blockDim.x;")
+        L.append("  for (long long i = (long long)blockIdx.x * blockDim.x +"
+                 " threadIdx.x; i < n; i += stride) {")
+        L.extend(self._row_body(in_types, out_types, lds=False))
+        L.append("  }")
+        L.append("}")
+        return "\n".join(L)
+
     def _row_body(self, in_types, out_types, lds):
         L = []
         if self.source == "csv":
             L.extend(self._load_inputs_csv(in_types, lds))
+        elif self.source == "col":
+            L.extend(self._load_inputs_col(in_types))
         else:
             L.extend(self._load_inputs_mem(in_types, lds))
         L.append("    Out o;")
@@ -917,14 +945,70 @@ class StageCodegen:
         L.append("        exc_buf[e*5+0] = row0 + i;")
         L.append("        exc_buf[e*5+1] = rc & 0xFFFFFFFFLL;")
         L.append("        exc_buf[e*5+2] = rc >> 32;")
-        L.append("        exc_buf[e*5+3] = in_offs[i];")
-        L.append("        exc_buf[e*5+4] = in_offs[i+1];")
+        if self.source == "col":
+            # columnar source has no row bytes; the host replays from the
+            # original Arrow table by row index (orcio.py)
+            L.append("        exc_buf[e*5+3] = 0;")
+            L.append("        exc_buf[e*5+4] = 0;")
+        else:
+            L.append("        exc_buf[e*5+3] = in_offs[i];")
+            L.append("        exc_buf[e*5+4] = in_offs[i+1];")
         L.append("      }")
         L.append("      keep[i] = 0; keep01[i] = 0; sizes[i] = 0; continue;")
         L.append("    }")
         L.append("    if (!o.keep) { keep[i] = 0; keep01[i] = 0; sizes[i] = 0; continue; }")
         L.append("    keep[i] = 1; keep01[i] = 1;")
         L.extend(self._store_columnar(out_types, lds))
+        return L
+
+    def _load_inputs_col(self, in_types):
+        """Columnar (Arrow-layout) source — ORC ingest (io/src/OrcTypes.cc,
+        physical/OrcReader analog). in_data is a device table
+        of 3 slots per column: [values-or-offsets, string-data, null-mask];
+        strings are (offset[i], offset[i+1]) views into the data buffer. No
+        parse, no LDS staging — loads are already typed and coalesced."""
+        used = getattr(self.sp, "used_source_cols", None)
+        L = ["    long long prc = 0;"]
+        L.append("    const void* const* ct = (const void* const*)in_data;")
+        for idx, t in enumerate(in_types):
+            base = T.deopt(t)
+            opt = T.is_opt(t)
+            if used is not None and idx not in used:
+                if opt:
+                    L.append("    bool c%d_n = false;  // unused (pushdown)" % idx)
+                if base == T.STR:
+                    L.append("    tstr c%d{(const char*)in_data, 0};" % idx)
+                elif base == T.I64:
+                    L.append("    long long c%d = 0;" % idx)
+                elif base == T.F64:
+                    L.append("    double c%d = 0.0;" % idx)
+                else:
+                    L.append("    bool c%d = false;" % idx)
+                continue
+            if opt:
+                L.append("    bool c%d_n = ((const unsigned char*)ct[%d])[i]"
+                         " != 0;" % (idx, 3 * idx + 2))
+            if base == T.STR:
+                L.append("    long long c%d_o = ((const long long*)ct[%d])[i];"
+                         % (idx, 3 * idx))
+                L.append("    tstr c%d{(const char*)ct[%d] + c%d_o,"
+                         " ((const long long*)ct[%d])[i + 1] - c%d_o};"
+                         % (idx, 3 * idx + 1, idx, 3 * idx, idx))
+                # same per-column ASCII gate as the mem loader: non-ASCII rows
+                # divert (char-index-sensitive ops need it)
+                guard = ("!prc && !c%d_n" % idx) if opt else "!prc"
+                L.append("    if (%s && !tpx_ascii(c%d)) prc = 7;" % (guard, idx))
+            elif base == T.I64:
+                L.append("    long long c%d = ((const long long*)ct[%d])[i];"
+                         % (idx, 3 * idx))
+            elif base == T.F64:
+                L.append("    double c%d = ((const double*)ct[%d])[i];"
+                         % (idx, 3 * idx))
+            elif base == T.BOOL:
+                L.append("    bool c%d = ((const unsigned char*)ct[%d])[i]"
+                         " != 0;" % (idx, 3 * idx))
+            else:
+                raise CodegenError("columnar input type %r" % (t,))
         return L
 
     def _load_inputs_mem(self, in_types, lds=True):

@@ -476,7 +476,7 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
     // Aggregate sinks consume the full keep/outv arrays at once -> C=1.
     bool mem_sink = D.sink == "mem";
     int C = 1;
-    if (D.agg.empty() && D.aggby.empty() && n >= (1 << 20)) {
+    if (D.agg.empty() && D.aggby.empty() && n >= (1 << 20) && d_offs) {
         static int env_c = [] {
             const char* e = getenv("TPX_CHUNKS");
             int v = e ? atoi(e) : 4;
@@ -972,6 +972,31 @@ static void empty_result(tpx_result* res, bool mem_sink) {
     res->out_row_offsets = (int64_t*)malloc(8);
     res->out_row_offsets[0] = mem_sink ? 8 : 0;
     res->out_row_indices = (int64_t*)malloc(8);
+}
+
+// ---------------------------------------------------------------------------------
+// columnar source (ORC/Arrow ingest — io OrcReader analog): slots[] holds 3
+// DEVICE pointers per input column ([values-or-offsets, string-data,
+// null-mask]; unused/pushed-down columns null). No row bytes exist, so
+// exception payloads are empty and the host replays from the original table.
+
+extern "C" int64_t tpx_stage_execute_col(tpx_stage* st, void* const* slots,
+                                         int64_t n_slots, int64_t n_rows,
+                                         int64_t in_bytes, int64_t first_row,
+                                         int64_t flags, tpx_result* res) {
+    memset(res, 0, sizeof(*res));
+    if (!st->loaded) { set_err("stage not loaded on a GPU"); return -1; }
+    hipStream_t stream = nullptr;
+    int dev = cur_device();
+    g_arena[dev].begin();
+    res->bytes_in = in_bytes;
+    if (n_rows == 0) { empty_result(res, st->desc.sink == "mem"); return 0; }
+    ARENA_TAKE(d_tab, (size_t)n_slots * sizeof(void*) + 8);
+    HIP_CHECK(hipMemcpyAsync(d_tab, slots, (size_t)n_slots * sizeof(void*),
+                             hipMemcpyHostToDevice, stream));
+    PayloadSrc psrc{nullptr, nullptr, 0};
+    return run_core(st, d_tab, nullptr, n_rows, first_row, in_bytes, res, psrc,
+                    stream, flags);
 }
 
 // ---------------------------------------------------------------------------------

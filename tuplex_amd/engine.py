@@ -90,6 +90,13 @@ class GpuLib:
         lib.tpx_dev_upload.argtypes = [ctypes.c_uint64, ctypes.c_void_p,
                                        ctypes.c_int64]
         lib.tpx_dev_free.argtypes = [ctypes.c_uint64]
+        lib.tpx_stage_execute_col.restype = ctypes.c_int64
+        lib.tpx_stage_execute_col.argtypes = [ctypes.c_void_p,
+                                              ctypes.POINTER(ctypes.c_void_p),
+                                              ctypes.c_int64, ctypes.c_int64,
+                                              ctypes.c_int64, ctypes.c_int64,
+                                              ctypes.c_int64,
+                                              ctypes.POINTER(TpxResult)]
         lib.tpx_stage_execute_csv_dev.restype = ctypes.c_int64
         lib.tpx_stage_execute_csv_dev.argtypes = [ctypes.c_void_p,
                                                   ctypes.c_uint64,
@@ -200,6 +207,51 @@ def execute_stage_mem(sp: plan.StageProgram, norm_rows: List[tuple],
                 out.exceptions.append((row, ecode, opid))
         out.metrics = {
             "t_serialize_s": t_ser,
+            "t_h2d_ms": res.t_h2d_ms, "t_kernel_ms": res.t_kernel_ms,
+            "t_d2h_ms": res.t_d2h_ms,
+            "bytes_in": res.bytes_in, "bytes_out": res.bytes_out,
+        }
+        return out
+    finally:
+        glib.lib.tpx_result_free(ctypes.byref(res))
+
+
+def execute_stage_col(sp: plan.StageProgram, dev_slots, n_rows: int,
+                      in_bytes: int, device: int = 0) -> ExecResult:
+    """Run the fused stage over a DEVICE-resident columnar (Arrow-layout)
+    table — the ORC ingest path. dev_slots: 3 device
+    pointers per input column ([values-or-offsets, string-data, null-mask];
+    unused columns None)."""
+    glib = GpuLib.get()
+    if glib.device_count() == 0:
+        raise RuntimeError("no HIP device visible — the normal-case path runs only "
+                           "on GPU (no CPU fallback by design)")
+    glib.lib.tpx_set_device(device)
+    src, desc = codegen.generate_stage(sp, source="col", sink="mem")
+    stage = glib.compile_stage(src, desc)
+    arr = (ctypes.c_void_p * len(dev_slots))(
+        *[ctypes.c_void_p(p or 0) for p in dev_slots])
+    res = TpxResult()
+    rc = glib.lib.tpx_stage_execute_col(stage, arr, len(dev_slots), n_rows,
+                                        in_bytes, 0, 0, ctypes.byref(res))
+    if rc != 0:
+        raise RuntimeError("columnar stage execute failed: " + glib.err())
+    try:
+        out = ExecResult()
+        out_bytes = ctypes.string_at(res.out_data, res.out_size)
+        out_row_type = T.tup(sp.gpu_output_types)
+        out.rows = rowfmt.deserialize_partition(out_bytes, out_row_type)
+        out.row_indices = [res.out_row_indices[i] for i in range(res.out_num_rows)] \
+            if res.out_num_rows else []
+        if res.exc_num_rows:
+            import struct as _s
+            eb = ctypes.string_at(res.exc_data, res.exc_size)
+            pos = 0
+            for _ in range(res.exc_num_rows):
+                row, ecode, opid, size = _s.unpack_from("<4q", eb, pos)
+                pos += 32 + size
+                out.exceptions.append((row, ecode, opid))
+        out.metrics = {
             "t_h2d_ms": res.t_h2d_ms, "t_kernel_ms": res.t_kernel_ms,
             "t_d2h_ms": res.t_d2h_ms,
             "bytes_in": res.bytes_in, "bytes_out": res.bytes_out,
