@@ -150,3 +150,29 @@ def test_orc_multi_file_and_tocsv(tmp_path):
     with open(out, "rb") as f:
         lines = f.read().decode().strip().split("\n")
     assert len(lines) == 1 + len(ref["output"])  # header + rows, in order
+
+
+@pytest.mark.gpu
+def test_cache_and_toorc_roundtrip(tmp_path):
+    """cache() materializes (CacheOperator analog); toorc -> orc() round trip."""
+    tab = _mk_table(2000, with_unicode=False)
+    p = _write_orc(tmp_path, tab)
+    from tests.pipelines import apply_ops
+    ds = apply_ops(tuplex_amd.Context().orc(p), orc_ops())
+    cached = ds.cache()
+    assert cached._last_outcome.mode == "gpu"
+
+    def second(x):
+        return (x[0] * 10, x[1])
+
+    got = cached.map(second).collect()
+    ref0 = pyoracle.run_pipeline(_rows(tab), orc_ops(),
+                                 columns=list(tab.schema.names))
+    ref = pyoracle.run_pipeline(ref0["output"], [("map", second)])
+    assert got == ref["output"]
+
+    out = os.path.join(str(tmp_path), "o.orc")
+    cached.toorc(out)
+    back = paorc.read_table(out)
+    assert back.num_rows == len(ref0["output"])
+    assert _rows(back) == ref0["output"]

@@ -73,7 +73,18 @@ class DataSet:
         raise NotImplementedError("join is out of scope (SURVEY.md §2 OOS)")
 
     def cache(self, store_specialized: bool = True) -> "DataSet":
-        return self  # device-resident cache: later round
+        """Materialize the pipeline so far; downstream operators start from the
+        materialized rows instead of re-executing (CacheOperator,
+        core/src/logical/CacheOperator.cc — SURVEY.md §8f-4). Exception rows
+        are resolved AT the cache point (reference stores them for later
+        resolution; resolvers attached after cache() therefore cannot see
+        pre-cache exceptions here — divergence documented in DESIGN.md)."""
+        from .context import ParallelizeSource
+        outcome = self._context._execute(self)
+        ds = DataSet(self._context,
+                     ParallelizeSource(list(outcome.rows), self.columns))
+        ds._last_outcome = outcome
+        return ds
 
     # ---- actions ------------------------------------------------------------
     def collect(self) -> List[Any]:
@@ -95,8 +106,18 @@ class DataSet:
         outcome = self._context._execute(self, sink=("csv", path))
         self._last_outcome = outcome
 
-    def toorc(self, *a, **kw):
-        raise NotImplementedError("ORC output is out of scope this round")
+    def toorc(self, path: str, **kw) -> None:
+        """write output to ORC (io OrcWriter analog; Arrow writer on host —
+        the GPU produces the rows, pyarrow serializes the columnar file)."""
+        import pyarrow as pa
+        import pyarrow.orc as paorc
+        outcome = self._context._execute(self)
+        self._last_outcome = outcome
+        rows = [(v if isinstance(v, tuple) else (v,)) for v in outcome.rows]
+        ncols = len(rows[0]) if rows else len(self.columns or []) or 1
+        names = self.columns or ["column%d" % i for i in range(ncols)]
+        cols = {names[k]: [r[k] for r in rows] for k in range(ncols)}
+        paorc.write_table(pa.table(cols), path)
 
     # ---- introspection ------------------------------------------------------
     @property
