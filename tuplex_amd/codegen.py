@@ -1081,6 +1081,7 @@ class StageCodegen:
         L.append("    const char* cur = rp;")
         L.append("    bool m = true, avail = true;")
         L.append("    int badf = 0;")
+        L.append("    unsigned long long hib = 0;  // non-ASCII bits seen by the walk")
 
         def null_check(idx):
             cv = "tstr{cl%d.p, cl%d.n}" % (idx, idx)
@@ -1098,8 +1099,8 @@ class StageCodegen:
             L.append("    tpx_cell cl%d{rp, 0, 0};" % idx)
             L.append("    if (!prc) {")
             L.append("      if (!avail) prc = %d;  // CSV_UNDERRUN" % 20)
-            L.append("      else { cur = tpx_csv_next_cell(cur, rend, &cl%d, &m, %s);"
-                     " avail = m; badf |= cl%d.flags; }" % (idx, delim_c, idx))
+            L.append("      else { cur = tpx_csv_next_cell(cur, rend, &cl%d, &m, %s,"
+                     " &hib); avail = m; badf |= cl%d.flags; }" % (idx, delim_c, idx))
             L.append("    }")
             if used is not None and idx not in used:
                 if opt:
@@ -1138,10 +1139,10 @@ class StageCodegen:
                 raise CodegenError("csv input type %r" % (t,))
         L.append("    if (!prc && avail) prc = %d;  // CSV_OVERRUN" % 21)
         L.append("    if (!prc && (badf & 6)) prc = %d;  // BADPARSE (escapes/structure -> host)" % 70)
-        # ONE ASCII gate per row (char-index-sensitive ops then skip their own
-        # scans; non-ASCII rows divert to the interpreter — semantically free)
-        L.append("    if (!prc && !tpx_ascii(tstr{rp, rend - rp})) prc = %d;  // NCV"
-                 % 7)
+        # ONE ASCII gate per row, fused into the cell walk (memchr_hi
+        # accumulated the high bits of every scanned byte; rows that error out
+        # earlier divert regardless, so partial accumulation is equivalent)
+        L.append("    if (!prc && hib) prc = %d;  // NCV" % 7)
         return L
 
     def _load_inputs_text(self, in_types, lds=True):

@@ -110,6 +110,32 @@ __device__ __forceinline__ long long tpx_memchr(const char* p, long long n,
     return -1;
 }
 
+// tpx_memchr that also ORs the high (non-ASCII) bits of every byte it scans
+// into *hi — lets the CSV cell walk double as the per-row ASCII gate (the walk
+// already touches every row byte; a separate tpx_ascii pass re-reads the row)
+__device__ __forceinline__ long long tpx_memchr_hi(const char* p, long long n,
+                                                   char c,
+                                                   unsigned long long* hi) {
+    unsigned long long pat = TPX_SWAR_ONE * (unsigned char)c;
+    long long i = 0;
+    while (i < n && (((unsigned long long)(p + i)) & 7)) {
+        *hi |= (unsigned char)p[i] & 0x80u;
+        if (p[i] == c) return i;
+        ++i;
+    }
+    for (; i + 8 <= n; i += 8) {
+        unsigned long long v = *(const unsigned long long*)(p + i);
+        *hi |= v & TPX_SWAR_HIGH;
+        unsigned long long hit = tpx_swar_zero(v ^ pat);
+        if (hit) return i + (__ffsll((long long)hit) - 1) / 8;
+    }
+    for (; i < n; ++i) {
+        *hi |= (unsigned char)p[i] & 0x80u;
+        if (p[i] == c) return i;
+    }
+    return -1;
+}
+
 // first index where p[i]==c1 or p[i]==c2, or -1
 __device__ __forceinline__ long long tpx_memchr2(const char* p, long long n,
                                                  char c1, char c2) {
@@ -883,10 +909,13 @@ extern "C" __global__ void tpx_csv_emit_rows(const char* __restrict__ data,
 // closing quote — reference DOUBLEQUOTEERROR class)
 struct tpx_cell { const char* p; long long n; int flags; };
 
+// hi: accumulates non-ASCII high bits of every scanned byte (row ASCII gate
+// fused into the walk — delimiters/quotes themselves are always ASCII)
 __device__ __forceinline__ const char* tpx_csv_next_cell(const char* p,
                                                          const char* end,
                                                          tpx_cell* c, bool* more,
-                                                         char delim) {
+                                                         char delim,
+                                                         unsigned long long* hi) {
     c->flags = 0;
     *more = false;
     if (p < end && *p == '"') {
@@ -894,7 +923,7 @@ __device__ __forceinline__ const char* tpx_csv_next_cell(const char* p,
         const char* q = s;
         bool esc = false;
         while (q < end) {
-            long long k = tpx_memchr(q, end - q, '"');
+            long long k = tpx_memchr_hi(q, end - q, '"', hi);
             if (k < 0) { q = end; break; }
             q += k;
             if (q + 1 < end && q[1] == '"') { esc = true; q += 2; continue; }
@@ -904,12 +933,12 @@ __device__ __forceinline__ const char* tpx_csv_next_cell(const char* p,
         c->p = s; c->n = q - s; c->flags = 1 | (esc ? 2 : 0);
         ++q;
         if (q < end && *q != delim) c->flags |= 4;
-        long long kd = tpx_memchr(q, end - q, delim);
+        long long kd = tpx_memchr_hi(q, end - q, delim, hi);
         q = kd < 0 ? end : q + kd;
         if (q < end) { *more = true; ++q; }
         return q;
     }
-    long long kd = tpx_memchr(p, end - p, delim);
+    long long kd = tpx_memchr_hi(p, end - p, delim, hi);
     const char* q = kd < 0 ? end : p + kd;
     c->p = p; c->n = q - p;
     if (q < end) { *more = true; ++q; }
