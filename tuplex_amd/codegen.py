@@ -1173,6 +1173,11 @@ class StageCodegen:
         """Store Out o -> columnar arrays + per-row serialized size (mem sink) or
         csv text size (csv sink)."""
         L = []
+        # csv sizes FIRST, while string views still point into LDS (the quote
+        # scan then runs on ds_read instead of re-reading global memory; the
+        # bytes are identical either way)
+        if self.sink != "mem":
+            L.extend(self._csv_size(out_types))
         for k, t in enumerate(out_types):
             base = T.deopt(t)
             if base == T.STR:
@@ -1203,8 +1208,6 @@ class StageCodegen:
                     else:
                         L.append("    sz += o.o%d.n + 1;" % k)
             L.append("    sizes[i] = sz;")
-        else:
-            L.extend(self._csv_size(out_types))
         return L
 
     def _csv_size(self, out_types):
