@@ -957,7 +957,12 @@ class StageCodegen:
         L.append("      keep[i] = 0; keep01[i] = 0; sizes[i] = 0; continue;")
         L.append("    }")
         L.append("    if (!o.keep) { keep[i] = 0; keep01[i] = 0; sizes[i] = 0; continue; }")
-        L.append("    keep[i] = 1; keep01[i] = 1;")
+        if self.sink == "mem":
+            L.append("    keep[i] = 1; keep01[i] = 1;")
+        else:
+            # bit 1: no cell of this row needs quoting (write fast path)
+            L.append("    keep[i] = (unsigned char)(1 | (_anyq ? 0 : 2));"
+                     " keep01[i] = 1;")
         L.extend(self._store_columnar(out_types, lds))
         return L
 
@@ -1212,15 +1217,26 @@ class StageCodegen:
 
     def _csv_size(self, out_types):
         """RFC-4180 output size: quote a cell iff it contains delim/quote/CR/LF;
-        '"' doubles. Appends newline per row."""
+        \'"\' doubles. Appends newline per row. Also records whether ANY cell
+        needs quoting (_anyq): the write kernel then takes a no-rescan memcpy
+        fast path for the (common) fully-unquoted rows, via keep[i] bit 1."""
         L = ["    long long sz = %d;  // delimiters + newline" % len(out_types)]
+        L.append("    bool _anyq = false;")
         for k, t in enumerate(out_types):
             base = T.deopt(t)
             if base == T.STR:
+                pre = ""
                 if T.is_opt(t):
-                    L.append("    if (!o.o%d_n) sz += tpx_csv_cell_len(o.o%d);" % (k, k))
-                else:
-                    L.append("    sz += tpx_csv_cell_len(o.o%d);" % k)
+                    L.append("    if (!o.o%d_n) {" % k)
+                    pre = "  "
+                L.append(pre + "    { long long q%d;" % k)
+                L.append(pre + "      bool nq%d = tpx_csv_needs_quote(o.o%d,"
+                         " &q%d);" % (k, k, k))
+                L.append(pre + "      sz += nq%d ? o.o%d.n + q%d + 2 : o.o%d.n;"
+                         % (k, k, k, k))
+                L.append(pre + "      _anyq |= nq%d; }" % k)
+                if T.is_opt(t):
+                    L.append("    }")
             elif base == T.I64:
                 L.append("    sz += tpx_i64_digits(o.o%d);" % k)
             elif base == T.BOOL:
@@ -1365,6 +1381,7 @@ class StageCodegen:
         body = self._csv_format_body(out_types)
         L.append("    if (staged) {")
         L.append("      if (active) {")
+        L.append("        bool _noq = (keep[i] & 2) != 0;")
         L.append("        char* w = wave_lds + (my_start - span_start);")
         L.extend("    " + ln for ln in body)
         L.append("      }")
@@ -1391,6 +1408,7 @@ class StageCodegen:
         L.append("      long long t0 = span > a0 ? a0 + ((span - a0) & ~7LL) : span;")
         L.append("      if (t0 + lane < span) dst[t0 + lane] = wave_lds[t0 + lane];")
         L.append("    } else if (active) {")
+        L.append("      bool _noq = (keep[i] & 2) != 0;")
         L.append("      char* w = (char*)out_data + my_start;")
         L.extend("  " + ln for ln in body)
         L.append("    }")
@@ -1412,7 +1430,11 @@ class StageCodegen:
                     pre = "  "
                 L.append(pre + "    tstr v%d{(const char*)((const unsigned long long*)outv[%d])[i],"
                          " (long long)((const int*)outv[%d])[i]};" % (k, 3 * k, 3 * k + 1))
-                L.append(pre + "    w = tpx_csv_cell_write(w, v%d);" % k)
+                # keep[i] bit 1 (set by the main kernel's size pass): no cell
+                # of this row needs quoting -> straight copy, no rescan
+                L.append(pre + "    if (_noq) { tpx_memcpy(w, v%d.p, v%d.n);"
+                         " w += v%d.n; }" % (k, k, k))
+                L.append(pre + "    else w = tpx_csv_cell_write(w, v%d);" % k)
                 if T.is_opt(t):
                     L.append("    }")
             elif base == T.I64:
