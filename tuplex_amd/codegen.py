@@ -957,13 +957,14 @@ class StageCodegen:
         L.append("      keep[i] = 0; keep01[i] = 0; sizes[i] = 0; continue;")
         L.append("    }")
         L.append("    if (!o.keep) { keep[i] = 0; keep01[i] = 0; sizes[i] = 0; continue; }")
+        L.extend(self._store_columnar(out_types, lds))
         if self.sink == "mem":
             L.append("    keep[i] = 1; keep01[i] = 1;")
         else:
-            # bit 1: no cell of this row needs quoting (write fast path)
+            # bit 1: no cell of this row needs quoting (write fast path;
+            # _anyq computed by the size pass inside _store_columnar)
             L.append("    keep[i] = (unsigned char)(1 | (_anyq ? 0 : 2));"
                      " keep01[i] = 1;")
-        L.extend(self._store_columnar(out_types, lds))
         return L
 
     def _load_inputs_col(self, in_types):
