@@ -470,6 +470,8 @@ def _apply_op(op, cur, cols, fast):
         names = [cols[i] for i in idxs]
         out = tuple(row[i] for i in idxs)
         return (out if len(out) > 1 else out[0]), names, False
+    if kind == "join":
+        return _apply_join(op, cur, cols)
     if kind == "renameColumn":
         old, new = op[1], op[2]
         cols2 = [new if c == old else c for c in cols]
@@ -494,3 +496,40 @@ def _apply_resolver(op, resolver, cur, cols, fast):
         col = op[1]
         return _apply_op(("mapColumn", col, resolver), cur, cols, fast)
     raise ValueError("resolver after %r unsupported" % (kind,))
+
+
+_JOIN_TABLES = {}
+
+
+def _join_table(op):
+    t = _JOIN_TABLES.get(id(op))
+    if t is None:
+        rrows, rcols, rk = op[1], op[2], op[4]
+        rki = rcols.index(rk)
+        t = {r[rki]: r for r in rrows}
+        _JOIN_TABLES[id(op)] = t
+    return t
+
+
+def _apply_join(op, cur, cols):
+    """Inner/left hash join against a materialized build side (single-match:
+    build keys are unique — logical/JoinOperator.cc:164 output layout
+    | left cols except key | key | right cols except key |; the key keeps the
+    left name and never nulls; left join nulls the right columns)."""
+    _, rrows, rcols, lk, rk, how, lp, ls, rp, rs = op
+    row = _as_row(cur)
+    cols2 = list(cols) if cols else None
+    if not cols2 or lk not in cols2:
+        raise ValueError("join: unknown left key column %r" % lk)
+    lki = cols2.index(lk)
+    rki = rcols.index(rk)
+    key = row[lki]
+    m = None if key is None else _join_table(op).get(key)
+    if m is None and how == "inner":
+        return cur, cols2, True
+    right_vals = tuple((None if m is None else m[j])
+                      for j in range(len(rcols)) if j != rki)
+    out = tuple(v for i, v in enumerate(row) if i != lki) + (key,) + right_vals
+    out_cols = ([lp + c + ls for c in cols2 if c != lk] + [lp + lk + ls] +
+                [rp + c + rs for c in rcols if c != rk])
+    return out, out_cols, False

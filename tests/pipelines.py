@@ -224,6 +224,16 @@ def apply_ops(ds, ops):
             ds = ds.ignore(op[1])
         elif kind == "aggregate":
             ds = ds.aggregate(op[1], op[2], op[3])
+        elif kind == "join":
+            _, rrows, rcols, lk, rk, how, lp, ls, rp, rs = op
+            rds = ds._context.parallelize(
+                [r if len(r) > 1 else r[0] for r in rrows], columns=rcols)
+            pre = (lp, rp) if (lp or rp) else None
+            suf = (ls, rs) if (ls or rs) else None
+            if how == "left":
+                ds = ds.leftJoin(rds, lk, rk, prefixes=pre, suffixes=suf)
+            else:
+                ds = ds.join(rds, lk, rk, prefixes=pre, suffixes=suf)
         elif kind == "aggregateByKey":
             ds = ds.aggregateByKey(op[1], op[2], op[3], op[4])
         else:

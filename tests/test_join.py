@@ -1,0 +1,135 @@
+"""Hash join (SURVEY.md §8f-3 MVP): right side = unique-key build table
+embedded in the generated stage source; probe on device. Output layout per
+logical/JoinOperator.cc:164: | left cols except key | key | right cols except
+key |; leftJoin nulls the right columns; the key keeps the left name."""
+import pytest
+
+import tuplex_amd
+from oracle import pyoracle
+from tests.pipelines import apply_ops
+
+RIGHT = [(1, "one", 1.5), (2, "two", 2.5), (3, "three", None)]
+RCOLS = ["k", "label", "w"]
+
+
+def _jop(how="inner", pre=("", ""), suf=("", "")):
+    return ("join", RIGHT, RCOLS, "key", "k", how, pre[0], suf[0], pre[1],
+            suf[1])
+
+
+def test_join_plan_and_compile():
+    from tuplex_amd import codegen, plan
+    from tuplex_amd import ttypes as T
+    sp = plan.build_stage([T.I64, T.STR], ["key", "val"], [_jop()])
+    assert sp.compilable, sp.why_not_compilable
+    assert sp.output_columns == ["val", "key", "label", "w"]
+    assert sp.output_types == [T.STR, T.I64, T.STR, ("opt", T.F64)]
+    src, desc = codegen.generate_stage(sp, source="mem", sink="mem")
+    from tuplex_amd.engine import GpuLib
+    glib = GpuLib.get()
+    st = glib.lib.tpx_stage_compile(src.encode(), desc.encode(), b"", 1)
+    assert st, glib.err()
+    # left join: right columns become nullable
+    spl = plan.build_stage([T.I64, T.STR], ["key", "val"], [_jop("left")])
+    assert spl.output_types == [T.STR, T.I64, ("opt", T.STR), ("opt", T.F64)]
+
+
+def test_join_string_key_compile():
+    from tuplex_amd import codegen, plan
+    from tuplex_amd import ttypes as T
+    right = [("BOS", "Boston"), ("NYC", "New York")]
+    op = ("join", right, ["code", "city"], "ap", "code", "inner",
+          "", "", "", "")
+    sp = plan.build_stage([T.STR, T.I64], ["ap", "n"], [op])
+    assert sp.compilable, sp.why_not_compilable
+    src, desc = codegen.generate_stage(sp, source="mem", sink="mem")
+    assert "tpx_jhash_bytes" in src
+
+
+def test_join_duplicate_keys_raise():
+    from tuplex_amd import plan
+    from tuplex_amd import ttypes as T
+    bad = [(1, "a"), (1, "b")]
+    with pytest.raises(ValueError, match="duplicate"):
+        plan.build_stage([T.I64], ["key"],
+                         [("join", bad, ["k", "x"], "key", "k", "inner",
+                           "", "", "", "")])
+
+
+def test_join_oracle_semantics():
+    rows = [(1, "a"), (2, "b"), (5, "c")]
+    r = pyoracle.run_pipeline(rows, [_jop()], columns=["key", "val"])
+    assert r["output"] == [("a", 1, "one", 1.5), ("b", 2, "two", 2.5)]
+    r = pyoracle.run_pipeline(rows, [_jop("left")], columns=["key", "val"])
+    assert r["output"][2] == ("c", 5, None, None)
+    # prefixes/suffixes rename like the reference
+    r = pyoracle.run_pipeline(rows, [_jop(pre=("l_", "r_"))],
+                              columns=["key", "val"])
+    assert r["output"][0] == ("a", 1, "one", 1.5)
+
+
+def test_join_pushdown_lineage():
+    from tuplex_amd import plan
+    from tuplex_amd import ttypes as T
+    ops = [_jop(), ("selectColumns", ["label"])]
+    sp = plan.build_stage([T.I64, T.STR, T.STR], ["key", "val", "junk"], ops)
+    assert sp.compilable
+    # only the probe key is read from the source
+    assert sorted(sp.used_source_cols) == [0]
+
+
+@pytest.mark.gpu
+def test_join_gpu_mem_parity():
+    import random
+    rng = random.Random(11)
+    rows = [(rng.randint(0, 6), "v%d" % i) for i in range(20000)]
+    ctx = tuplex_amd.Context()
+    ds = apply_ops(ctx.parallelize(rows, columns=["key", "val"]), [_jop()])
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    ref = pyoracle.run_pipeline(rows, [_jop()], columns=["key", "val"])
+    assert got == ref["output"]
+    assert len(got) < len(rows)  # keys 0,4,5,6 miss
+
+    dsl = apply_ops(ctx.parallelize(rows, columns=["key", "val"]),
+                    [_jop("left")])
+    gotl = dsl.collect()
+    assert dsl._last_outcome.mode == "gpu"
+    refl = pyoracle.run_pipeline(rows, [_jop("left")], columns=["key", "val"])
+    assert gotl == refl["output"]
+    assert len(gotl) == len(rows)
+
+
+@pytest.mark.gpu
+def test_join_gpu_csv_string_key(tmp_path):
+    """flights-style: csv fact rows joined to a string-keyed dimension."""
+    import os
+    codes = ["BOS", "JFK", "LAX", "ORD", "SEA", "SFO"]
+    dim = [(c, "City-%s" % c, i * 10) for i, c in enumerate(codes)]
+    lines = [b"ap,delay"]
+    import random
+    rng = random.Random(3)
+    for i in range(30000):
+        ap = rng.choice(codes + ["XXX"])  # XXX misses the dimension
+        lines.append(b"%s,%d" % (ap.encode(), rng.randint(-10, 500)))
+    data = b"\n".join(lines) + b"\n"
+    p = os.path.join(str(tmp_path), "fl.csv")
+    with open(p, "wb") as f:
+        f.write(data)
+
+    jop = ("join", dim, ["code", "city", "rank"], "ap", "code", "inner",
+           "", "", "", "")
+
+    def late(x):
+        return x["delay"] > 60
+
+    ops = [("filter", late), jop, ("selectColumns", ["ap", "city", "delay"])]
+    ctx = tuplex_amd.Context()
+    ds = apply_ops(ctx.csv(p), ops)
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    from oracle import pyoracle_csv
+    ref = pyoracle_csv.run_csv_pipeline(data, ops)
+    assert got == ref["output"]
+    assert ds.exception_counts == ref["exception_counts"]
+    assert all(c.startswith("City-") for _, c, _d in got)

@@ -125,6 +125,7 @@ class StageCodegen:
         self.csv_info = csv_info or {}  # {"null_values": [...]}
         self.lits = {}  # python str -> lit var name
         self.lit_defs = []
+        self.join_defs = []   # embedded hash-join build tables (per join op)
         # multi-needle scan fusion: pass 1 registers every const-needle
         # find/contains per structurally-identified haystack; pass 2 emits ONE
         # SWAR window scan per group (the serial dependent-load chains of
@@ -187,6 +188,197 @@ class StageCodegen:
             return em.memo(node, v)
         v = self._emit(em, node, row_ctx, opid)
         return em.memo(node, v)
+
+    @staticmethod
+    def _jhash_i64(k):
+        m = (1 << 64) - 1
+        h = k & m
+        h ^= h >> 33
+        h = (h * 0xff51afd7ed558ccd) & m
+        h ^= h >> 33
+        h = (h * 0xc4ceb9fe1a85ec53) & m
+        h ^= h >> 33
+        return h
+
+    @staticmethod
+    def _jhash_bytes(b):
+        m = (1 << 64) - 1
+        h = 1469598103934665603
+        for c in b:
+            h = ((h ^ c) * 1099511628211) & m
+        return h
+
+    def _emit_join_table(self, op):
+        """Embed the (unique-key) build side of a hash join as __device__
+        const arrays in the generated source — small dimension tables live in
+        L2, no extra kernel argument or ABI change (HashJoinStage build side;
+        probe = open addressing, same layout as the host build here)."""
+        rrows, rcols, lki, rki, how = op.join
+        nm = "jt%d" % op.opid
+        n = len(rrows)
+        tsize = 8
+        while tsize < 2 * max(n, 1):
+            tsize <<= 1
+        str_key = isinstance(rrows[0][rki], str) if n else False
+        occ = [0] * tsize
+        slotrow = [0] * tsize
+        for ri, r in enumerate(rrows):
+            k = r[rki]
+            h = (self._jhash_bytes(k.encode()) if str_key
+                 else self._jhash_i64(int(k))) & (tsize - 1)
+            while occ[h]:
+                h = (h + 1) & (tsize - 1)
+            occ[h] = 1
+            slotrow[h] = ri
+        blob = bytearray()
+
+        def put(sval):
+            off = len(blob)
+            b = sval.encode()
+            blob.extend(b)
+            return off, len(b)
+
+        D = self.join_defs
+        D.append("__device__ const unsigned char %s_occ[%d] = {%s};"
+                 % (nm, tsize, ",".join(str(x) for x in occ)))
+        if str_key:
+            koff, klen = [], []
+            for s2 in range(tsize):
+                if occ[s2]:
+                    o2, l2 = put(rrows[slotrow[s2]][rki])
+                else:
+                    o2, l2 = 0, 0
+                koff.append(o2)
+                klen.append(l2)
+            D.append("__device__ const int %s_koff[%d] = {%s};"
+                     % (nm, tsize, ",".join(map(str, koff))))
+            D.append("__device__ const int %s_klen[%d] = {%s};"
+                     % (nm, tsize, ",".join(map(str, klen))))
+        else:
+            D.append("__device__ const long long %s_key[%d] = {%s};"
+                     % (nm, tsize,
+                        ",".join(str(int(rrows[slotrow[s2]][rki])) if occ[s2]
+                                 else "0" for s2 in range(tsize))))
+        out_info = []  # (j, ctail) for right cols except key
+        for j in range(len(rcols)):
+            if j == rki:
+                continue
+            vals = [rrows[slotrow[s2]][j] if occ[s2] else None
+                    for s2 in range(tsize)]
+            kindc = None
+            for v in vals:
+                if v is None:
+                    continue
+                kindc = ("str" if isinstance(v, str) else
+                         "bool" if isinstance(v, bool) else
+                         "f64" if isinstance(v, float) else "i64")
+                break
+            kindc = kindc or "i64"
+            has_null = any(occ[s2] and rrows[slotrow[s2]][j] is None
+                           for s2 in range(tsize))
+            if kindc == "str":
+                offs, lens = [], []
+                for v in vals:
+                    if isinstance(v, str):
+                        o2, l2 = put(v)
+                    else:
+                        o2, l2 = 0, 0
+                    offs.append(o2)
+                    lens.append(l2)
+                D.append("__device__ const int %s_c%d_off[%d] = {%s};"
+                         % (nm, j, tsize, ",".join(map(str, offs))))
+                D.append("__device__ const int %s_c%d_len[%d] = {%s};"
+                         % (nm, j, tsize, ",".join(map(str, lens))))
+            elif kindc == "f64":
+                D.append("__device__ const double %s_c%d[%d] = {%s};"
+                         % (nm, j, tsize,
+                            ",".join(repr(float(v)) if isinstance(v, (int, float))
+                                     and not isinstance(v, bool) else "0.0"
+                                     for v in vals)))
+            else:
+                D.append("__device__ const long long %s_c%d[%d] = {%s};"
+                         % (nm, j, tsize,
+                            ",".join(str(int(v)) if v is not None
+                                     and not isinstance(v, str) else "0"
+                                     for v in vals)))
+            if has_null:
+                D.append("__device__ const unsigned char %s_c%d_null[%d] = {%s};"
+                         % (nm, j, tsize,
+                            ",".join("1" if (occ[s2] and
+                                             rrows[slotrow[s2]][j] is None)
+                                     else "0" for s2 in range(tsize))))
+            out_info.append((j, kindc, has_null))
+        if blob:
+            # string blob as concatenated hex-escaped literals
+            parts = []
+            bs = bytes(blob)
+            for i2 in range(0, len(bs), 2000):
+                seg = bs[i2:i2 + 2000]
+                parts.append('"%s"' % "".join("\\x%02x" % c for c in seg))
+            D.append("__device__ const char %s_blob[%d] = %s;"
+                     % (nm, len(bs) + 1, "\n".join(parts)))
+        else:
+            D.append("__device__ const char %s_blob[1] = \"\";" % nm)
+        return nm, tsize, str_key, out_info
+
+    def _emit_join(self, em, op, rc):
+        """Probe an embedded build table; extend the row context with the
+        right columns (JoinOperator.cc:164 layout)."""
+        rrows, rcols, lki, rki, how = op.join
+        nm, tsize, str_key, out_info = self._emit_join_table(op)
+        kv, _kt, knv = rc[lki]
+        ji = em.fresh("ji")
+        em.w("int %s = -1;" % ji)
+        guard = ("if (!(%s)) " % knv) if knv else ""
+        em.w("%s{" % guard)
+        if str_key:
+            em.w("  unsigned long long _h = tpx_jhash_bytes(%s.p, %s.n);"
+                 % (kv, kv))
+        else:
+            em.w("  unsigned long long _h = tpx_hash_i64(%s);" % kv)
+        em.w("  for (unsigned _p = 0; _p < %du; ++_p) {" % tsize)
+        em.w("    unsigned _s = (unsigned)((_h + _p) & %du);" % (tsize - 1))
+        em.w("    if (!%s_occ[_s]) break;" % nm)
+        if str_key:
+            em.w("    if (%s_klen[_s] == (int)%s.n && tpx_streq(tstr{%s_blob +"
+                 " %s_koff[_s], %s_klen[_s]}, %s)) { %s = (int)_s; break; }"
+                 % (nm, kv, nm, nm, nm, kv, ji))
+        else:
+            em.w("    if (%s_key[_s] == %s) { %s = (int)_s; break; }"
+                 % (nm, kv, ji))
+        em.w("  }")
+        em.w("}")
+        if how == "inner":
+            em.w("if (%s < 0) { o.keep = false; return 0; }" % ji)
+        new_rc = [rc[i2] for i2 in range(len(rc)) if i2 != lki] + [rc[lki]]
+        left_join = how == "left"
+        for j, kindc, has_null in out_info:
+            nullable = left_join or has_null
+            nv = None
+            if nullable:
+                nv = em.fresh("jn")
+                if has_null:
+                    em.w("bool %s = %s < 0 || %s_c%d_null[%s < 0 ? 0 : %s];"
+                         % (nv, ji, nm, j, ji, ji))
+                else:
+                    em.w("bool %s = %s < 0;" % (nv, ji))
+            v = em.fresh("jv")
+            sel = "%s < 0 ? 0 : %s" % (ji, ji)
+            if kindc == "str":
+                em.w("tstr %s = {%s_blob + %s_c%d_off[%s], %s_c%d_len[%s]};"
+                     % (v, nm, nm, j, sel, nm, j, sel))
+                t = T.STR
+            elif kindc == "f64":
+                em.w("double %s = %s_c%d[%s];" % (v, nm, j, sel))
+                t = T.F64
+            elif kindc == "bool":
+                em.w("bool %s = %s_c%d[%s] != 0;" % (v, nm, j, sel))
+                t = T.BOOL
+            else:
+                em.w("long long %s = %s_c%d[%s];" % (v, nm, j, sel))
+                t = T.I64
+            new_rc.append((v, ("opt", t) if nullable else t, nv))
+        return new_rc
 
     def _peephole(self, em, n, rc, opid):
         """Pattern rewrites that must run BEFORE strict arg evaluation.
@@ -748,6 +940,8 @@ class StageCodegen:
                 rc = [rc[i] for i in op.sel_idxs]
             elif op.kind == "renameColumn":
                 pass
+            elif op.kind == "join":
+                rc = self._emit_join(em, op, rc)
             else:
                 raise CodegenError("op %r not supported in codegen" % op.kind)
         if getattr(sp, "agg_expr", None) is not None:
@@ -798,7 +992,7 @@ class StageCodegen:
             "// generated by tuplex_amd.codegen — stage %s" % sp.signature(),
             runtime_header(),
             "",
-        ] + self.lit_defs + [
+        ] + self.lit_defs + self.join_defs + [
             "",
             "struct Out {",
         ] + out_fields + [

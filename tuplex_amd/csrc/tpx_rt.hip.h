@@ -1065,6 +1065,17 @@ extern "C" __global__ void tpx_reduce_i64_final(const long long* __restrict__ pa
 #define TPX_HK_EMPTY (-1LL)
 #define TPX_HK_MAXPROBE 1024
 
+// fnv1a for hash-join string keys (build-side replica in codegen._jhash_bytes)
+__device__ __forceinline__ unsigned long long tpx_jhash_bytes(const char* p,
+                                                              long long n) {
+    unsigned long long h = 1469598103934665603ULL;
+    for (long long i = 0; i < n; ++i) {
+        h ^= (unsigned char)p[i];
+        h *= 1099511628211ULL;
+    }
+    return h;
+}
+
 __device__ __forceinline__ unsigned long long tpx_hash_i64(long long k) {
     unsigned long long x = (unsigned long long)k;
     x ^= x >> 33;

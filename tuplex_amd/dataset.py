@@ -66,11 +66,40 @@ class DataSet:
         unpinned, as the reference's hashmap iteration order is)."""
         return self._chain(("unique",))
 
-    def join(self, *a, **kw):
-        raise NotImplementedError("join is out of scope (SURVEY.md §2 OOS)")
+    def _join(self, dsRight, leftKeyColumn, rightKeyColumn, prefixes, suffixes,
+              how):
+        """Inner/left join, right side materialized as the hash-BUILD side
+        (reference: logical/JoinOperator.cc; output layout JoinOperator.cc:164:
+        | left cols except key | key (left name) | right cols except key |;
+        HashJoinStage probe — SURVEY.md §8f-3). This round the build side needs
+        UNIQUE non-null keys (dimension-table joins); duplicate keys raise."""
+        lp = ls = rp = rs = ""
+        if prefixes:
+            p = tuple(prefixes)
+            lp, rp = p[0] or "", p[1] or ""
+        if suffixes:
+            sfx = tuple(suffixes)
+            ls, rs = sfx[0] or "", sfx[1] or ""
+        rcols = dsRight.columns
+        if not rcols:
+            raise ValueError("join: right dataset needs named columns")
+        if not dsRight._ops and getattr(dsRight._source, "kind", "") == "mem":
+            rrows = list(dsRight._source.data)  # no pipeline: direct
+        else:
+            rrows = dsRight.collect()
+        rrows = [(v if isinstance(v, tuple) else (v,)) for v in rrows]
+        return self._chain(("join", rrows, list(rcols), leftKeyColumn,
+                            rightKeyColumn, how, lp, ls, rp, rs))
 
-    def leftJoin(self, *a, **kw):
-        raise NotImplementedError("join is out of scope (SURVEY.md §2 OOS)")
+    def join(self, dsRight, leftKeyColumn: str, rightKeyColumn: str,
+             prefixes=None, suffixes=None) -> "DataSet":
+        return self._join(dsRight, leftKeyColumn, rightKeyColumn, prefixes,
+                          suffixes, "inner")
+
+    def leftJoin(self, dsRight, leftKeyColumn: str, rightKeyColumn: str,
+                 prefixes=None, suffixes=None) -> "DataSet":
+        return self._join(dsRight, leftKeyColumn, rightKeyColumn, prefixes,
+                          suffixes, "left")
 
     def cache(self, store_specialized: bool = True) -> "DataSet":
         """Materialize the pipeline so far; downstream operators start from the
@@ -135,6 +164,12 @@ class DataSet:
             elif op[0] == "renameColumn":
                 if cols:
                     cols = [op[2] if c == op[1] else c for c in cols]
+            elif op[0] == "join":
+                _, _rr, rcols, lk, rk, _how, lp, ls, rp, rs = op
+                if cols:
+                    cols = ([lp + c + ls for c in cols if c != lk] +
+                            [lp + lk + ls] +
+                            [rp + c + rs for c in rcols if c != rk])
         return cols
 
     @property

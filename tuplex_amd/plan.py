@@ -232,6 +232,56 @@ def build_stage(input_types, input_columns, logical_ops) -> StageProgram:
             if cur_cols is None or entry[1] not in cur_cols:
                 raise ValueError("renameColumn: unknown column %r" % entry[1])
             cur_cols = [entry[2] if c == entry[1] else c for c in cur_cols]
+        elif kind == "join":
+            # ("join", rrows, rcols, lk, rk, how, lp, ls, rp, rs) — inner/left
+            # hash join, right = build side (JoinOperator.cc:164 layout:
+            # left-except-key | key | right-except-key; left join nulls the
+            # right columns, the key never nulls)
+            _, rrows, rcols, lk, rk, how, lp, ls, rp, rs = entry
+            if cur_cols is None or lk not in cur_cols:
+                raise ValueError("join: unknown left key column %r" % lk)
+            if rk not in rcols:
+                raise ValueError("join: unknown right key column %r" % rk)
+            lki = cur_cols.index(lk)
+            rki = rcols.index(rk)
+            keys = [r[rki] for r in rrows]
+            if any(k is None for k in keys):
+                raise ValueError("join: null keys in the build side are "
+                                 "unsupported this round")
+            if len(set(keys)) != len(keys):
+                raise ValueError("join: duplicate build-side keys are "
+                                 "unsupported this round (SURVEY.md §8f-3)")
+            # per-column right types over the materialized rows
+            rtypes = []
+            for j in range(len(rcols)):
+                vals = [r[j] for r in rrows]
+                t = T.infer_majority_type([(v,) for v in vals],
+                                          optional_threshold=1.0)
+                rtypes.append(T.tuple_params(T.row_type_of(t))[0])
+            lkt = T.deopt(cur_types[lki])
+            if T.deopt(rtypes[rki]) != lkt or lkt not in (T.I64, T.STR):
+                _fallback(sp, "join key type %r/%r (i64/str only on GPU)"
+                          % (cur_types[lki], rtypes[rki]))
+            if any(T.deopt(t) not in (T.I64, T.F64, T.BOOL, T.STR)
+                   for t in rtypes):
+                _fallback(sp, "join: unsupported right column type")
+            if len(rrows) > 65536 or sum(
+                    len(str(r[j]).encode()) for r in rrows
+                    for j in range(len(rcols))
+                    if isinstance(r[j], str)) > (4 << 20):
+                _fallback(sp, "join build side too large for the embedded "
+                          "table this round")
+            op.join = (rrows, rcols, lki, rki, how)
+            out_r = []
+            for j, t in enumerate(rtypes):
+                if j == rki:
+                    continue
+                out_r.append(T.opt(t) if how == "left" else t)
+            cur_types = ([t for i2, t in enumerate(cur_types) if i2 != lki]
+                         + [cur_types[lki]] + out_r)
+            cur_cols = ([lp + c + ls for c in cur_cols if c != lk]
+                        + [lp + lk + ls]
+                        + [rp + c + rs for c in rcols if c != rk])
         else:
             raise ValueError("unknown op %r" % (kind,))
         op.out_types = list(cur_types)
@@ -296,6 +346,18 @@ def _used_source_columns(sp):
             used = {op.sel_idxs[p] for p in used}
         elif op.kind == "renameColumn":
             pass
+        elif op.kind == "join":
+            # out: [left except key | key | right...]; right cols have no
+            # source dependency; the probe always reads the key
+            lki = op.join[2]
+            left_map = [i for i in range(n_in) if i != lki]
+            u = {lki}
+            for p in used:
+                if p < n_in - 1:
+                    u.add(left_map[p])
+                elif p == n_in - 1:
+                    u.add(lki)
+            used = u
         else:
             return frozenset(range(len(sp.input_types)))
     return frozenset(used)
