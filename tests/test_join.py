@@ -133,3 +133,38 @@ def test_join_gpu_csv_string_key(tmp_path):
     assert got == ref["output"]
     assert ds.exception_counts == ref["exception_counts"]
     assert all(c.startswith("City-") for _, c, _d in got)
+
+
+@pytest.mark.gpu
+def test_join_gpu_flights_pipeline(tmp_path):
+    """The flights benchmark shape (runtuplex.py:210+): parse the wide CSV,
+    derive columns, then join TWO string-keyed dimension tables (airports,
+    carriers) — the full pipeline the reference's HashJoinStage serves."""
+    import os
+    from tests import extra_data as X
+    from oracle import pyoracle_csv
+
+    data = X.make_flights_csv(4000, seed=9, bad_frac=0.01)
+    p = os.path.join(str(tmp_path), "fl.csv")
+    with open(p, "wb") as f:
+        f.write(data)
+
+    # fl_code lowercases the carrier (aa/dl survive fl_carrier); c3 is
+    # "City-N Airport" — the left join hits only a subset of cities
+    carriers = [("aa", "American"), ("dl", "Delta"), ("ua", "United")]
+    cities = [("City-%d Airport" % i, i * 3) for i in range(1, 150)]
+    j1 = ("join", carriers, ["code_c", "carrier_name"], "code", "code_c",
+          "inner", "", "", "", "")
+    j2 = ("join", cities, ["city_a", "tz"], "c3", "city_a",
+          "left", "", "", "", "")
+    ops = X.flights_ops() + [j1, j2]
+
+    ctx = tuplex_amd.Context()
+    ds = apply_ops(ctx.csv(p), ops)
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    ref = pyoracle_csv.run_csv_pipeline(data, ops)
+    assert got == ref["output"]
+    assert ds.exception_counts == ref["exception_counts"]
+    assert len(got) > 0
+    assert any(r[-1] is None for r in got) or all(r[-1] for r in got)
