@@ -502,13 +502,16 @@ _JOIN_TABLES = {}
 
 
 def _join_table(op):
-    t = _JOIN_TABLES.get(id(op))
-    if t is None:
+    # keyed by id() but the op itself is pinned in the entry: a freed tuple's
+    # address can be reused, which would silently serve the wrong build table
+    ent = _JOIN_TABLES.get(id(op))
+    if ent is None or ent[0] is not op:
         rrows, rcols, rk = op[1], op[2], op[4]
         rki = rcols.index(rk)
         t = {r[rki]: r for r in rrows}
-        _JOIN_TABLES[id(op)] = t
-    return t
+        _JOIN_TABLES[id(op)] = (op, t)
+        return t
+    return ent[1]
 
 
 def _apply_join(op, cur, cols):
