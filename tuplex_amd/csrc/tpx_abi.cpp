@@ -340,6 +340,23 @@ static void* pinned(int dev) {
     if (!g_pinned[dev]) (void)hipHostMalloc(&g_pinned[dev], 1024);
     return g_pinned[dev];
 }
+
+// per-device event pool: hipEventCreate/Destroy cost ~10 us each and run_core
+// uses ~28 events per step — pooled, that overhead disappears
+struct EventPool {
+    std::vector<hipEvent_t> evs;
+    size_t next = 0;
+    hipEvent_t get() {
+        if (next == evs.size()) {
+            hipEvent_t e = nullptr;
+            (void)hipEventCreate(&e);
+            evs.push_back(e);
+        }
+        return evs[next++];
+    }
+    void reset() { next = 0; }
+};
+static EventPool g_events[64];
 static hipStream_t cstream(int dev, int i) {
     if (!g_cstream[dev][i])
         (void)hipStreamCreateWithFlags(&g_cstream[dev][i], hipStreamNonBlocking);
@@ -433,9 +450,11 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
     int dev = cur_device();
     Persist& P = g_persist[dev];
     res->in_num_rows = n;
-    hipEvent_t ev1, ev2, ev3, evm0, evm1, evs1;
-    hipEventCreate(&ev1); hipEventCreate(&ev2); hipEventCreate(&ev3);
-    hipEventCreate(&evm0); hipEventCreate(&evm1); hipEventCreate(&evs1);
+    EventPool& EP = g_events[dev];
+    EP.reset();  // previous call's events are fully consumed by its return
+    hipEvent_t ev1 = EP.get(), ev2 = EP.get(), ev3 = EP.get();
+    hipEvent_t evm0 = EP.get(), evm1 = EP.get(), evs1 = EP.get();
+    (void)evm1;
     hipEventRecord(ev1, stream);
 
     // output columnar buffers (3 slots per column) from the arena
@@ -547,7 +566,7 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
     unsigned long long exc_count = 0;
     // [m1, s0, s1, w0, w1] per chunk + 2 join events
     std::vector<hipEvent_t> cev((size_t)C * 5 + 2);
-    for (auto& e : cev) hipEventCreate(&e);
+    for (auto& e : cev) e = EP.get();
     long long total_rows = 0, total_bytes = 0;
     long long chunk_rows[8] = {0}, chunk_bytes[8] = {0};
     void* chunk_out[8] = {nullptr};
@@ -954,9 +973,6 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
         hipEventElapsedTime(&ms, cev[0], evs1); res->t_compact_ms = ms;
         hipEventElapsedTime(&ms, evs1, ev2); res->t_write_ms = ms;
     }
-    (void)hipEventDestroy(ev1); (void)hipEventDestroy(ev2); (void)hipEventDestroy(ev3);
-    (void)hipEventDestroy(evm0); (void)hipEventDestroy(evm1); (void)hipEventDestroy(evs1);
-    for (auto& e : cev) (void)hipEventDestroy(e);
     return 0;
 }
 
