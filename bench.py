@@ -149,11 +149,12 @@ def main():
 
     dist = None
     torch = None
+    backend = os.environ.get("TPX_BENCH_BACKEND", "nccl")
     if world > 1:
         import torch  # noqa: F811
         import torch.distributed as dist  # noqa: F811
-        torch.cuda.set_device(local_rank)
-        dist.init_process_group("nccl")
+        torch.cuda.set_device(local_rank % max(torch.cuda.device_count(), 1))
+        dist.init_process_group(backend)
 
     from tuplex_amd.engine import GpuLib, TpxResult
     glib = GpuLib.get()
@@ -215,15 +216,16 @@ def main():
     t1 = time.perf_counter()
     wall = t1 - t0
     if dist:
-        tt = torch.tensor([wall], device="cuda")
+        red_dev = "cuda" if backend == "nccl" else "cpu"
+        tt = torch.tensor([wall], device=red_dev)
         dist.all_reduce(tt, op=dist.ReduceOp.MAX)
         wall = float(tt.item())
         rows_t = torch.tensor([float(sum(k["in_rows"] for k in kstats))],
-                              device="cuda")
+                              device=red_dev)
         dist.all_reduce(rows_t, op=dist.ReduceOp.SUM)
         total_rows = float(rows_t.item())
         bytes_t = torch.tensor([float(sum(k["bytes_in"] + k["bytes_out"]
-                                          for k in kstats))], device="cuda")
+                                          for k in kstats))], device=red_dev)
         dist.all_reduce(bytes_t, op=dist.ReduceOp.SUM)
         total_bytes = float(bytes_t.item())
     else:
