@@ -367,3 +367,31 @@ def test_projection_pushdown(tmp_path):
     # counter-check: without pushdown the junk rows DO divert in the oracle
     ref_full = pyoracle_csv.run_csv_pipeline(data, ops, used_cols=None)
     assert sum(ref_full["exception_counts"].values()) > 0
+
+
+def test_crlf_line_endings(tmp_path):
+    """Windows CRLF files: \r stripped from unquoted row ends; \r\n inside a
+    quoted cell is DATA (CSVUtils row-boundary semantics)."""
+    def use(x):
+        return (x["a"] + 1, x["b"])
+
+    data = b"a,b\r\n" + b"".join(b"%d,x%d\r\n" % (i, i) for i in range(5000))
+    p = _write(tmp_path, data, "crlf.csv")
+    ctx = tuplex_amd.Context()
+    ds = apply_ops(ctx.csv(p), [("map", use)])
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    ref = pyoracle_csv.run_csv_pipeline(data, [("map", use)])
+    assert got == ref["output"]
+    assert got[0] == (1, "x0")
+
+    dataq = b"a,b\r\n" + b"".join(
+        (b'%d,"v\r\n%d"\r\n' if i % 7 == 0 else b"%d,x%d\r\n") % (i, i)
+        for i in range(3000))
+    p2 = _write(tmp_path, dataq, "crlfq.csv")
+    ds2 = apply_ops(ctx.csv(p2), [("map", use)])
+    got2 = ds2.collect()
+    assert ds2._last_outcome.mode == "gpu", ds2._last_outcome.fallback_reason
+    ref2 = pyoracle_csv.run_csv_pipeline(dataq, [("map", use)])
+    assert got2 == ref2["output"]
+    assert got2[0][1] == "v\r\n0"
