@@ -1445,14 +1445,12 @@ class StageCodegen:
     def _write_kernel(self, out_types):
         if self.sink == "csv":
             return self._write_kernel_csv(out_types)
-        # mem sink: same wave-staged design as the csv writer (input-row
-        # iteration keeps columnar loads coalesced; each wave's 64 input rows
-        # own a contiguous output span staged in LDS, copied out with aligned
-        # 8B stores). Rows in the reference layout are NOT 8-aligned in the
-        # stream, so the staged body writes slots with byte-safe memcpy while
-        # the global fallback keeps unaligned typed stores.
+        n_opt = sum(1 for t in out_types if T.is_opt(t))
+        bitmap = ((n_opt + 63) // 64) * 8 if n_opt else 0
+        has_var = any(T.is_varlen(t) for t in out_types)
+        nf = len(out_types)
+        fixed_end = bitmap + 8 * nf
         L = []
-        L.append("#define TPX_MWCAP %d" % self.MEM_WRITE_CAP)
         L.append('extern "C" __global__ void tpx_stage_write(')
         L.append("    const unsigned char* __restrict__ keep,")
         L.append("    const long long* __restrict__ keep_scan,")
@@ -1461,88 +1459,18 @@ class StageCodegen:
         L.append("    unsigned char* __restrict__ out_data, long long* __restrict__ out_offs,")
         L.append("    long long* __restrict__ out_rowidx,")
         L.append("    long long total_rows, long long total_bytes,")
+        # out_data is the CHUNK-LOCAL base (past the 8B numRows header, which
+        # the host writes); out_byte0 is the row's global byte offset bias
+        # stored into out_offs (chunk pipelining: tpx_abi.cpp run_core)
         L.append("    long long out_byte0) {")
-        L.append("  __shared__ __attribute__((aligned(16)))"
-                 " char wsmem[2 * TPX_MWCAP + 16];  // 128-thread blocks")
-        L.append("  int lane = threadIdx.x & 63;")
-        L.append("  int wid = threadIdx.x >> 6;")
-        L.append("  char* wave_lds = wsmem + wid * TPX_MWCAP;")
-        L.append("  long long wave_stride = (long long)gridDim.x * (blockDim.x >> 6);")
-        L.append("  long long nwaves = (n + 63) >> 6;")
-        L.append("  for (long long wb = (long long)blockIdx.x * (blockDim.x >> 6) + wid;"
-                 " wb < nwaves; wb += wave_stride) {")
-        L.append("    long long r0 = wb << 6;")
-        L.append("    long long rhi = r0 + 64 < n ? r0 + 64 : n;")
-        L.append("    long long span_start = size_scan[r0];")
-        L.append("    long long span_end = r0 + 64 < n ? size_scan[r0 + 64]"
-                 " : total_bytes;")
-        L.append("    long long span = span_end - span_start;")
-        L.append("    bool staged = span <= TPX_MWCAP;")
-        L.append("    long long i = r0 + lane;")
-        L.append("    bool active = i < rhi && keep[i];")
-        L.append("    long long my_start = active ? size_scan[i] : 0;")
-        L.append("    if (active) {")
-        L.append("      out_offs[keep_scan[i]] = out_byte0 + my_start;")
-        L.append("      out_rowidx[keep_scan[i]] = row0 + i;")
-        L.append("    }")
-        L.append("    if (staged) {")
-        L.append("      if (active) {")
-        L.append("        char* w = wave_lds + (my_start - span_start);")
-        L.extend("    " + ln for ln in self._mem_format_body(out_types, True))
-        L.append("      }")
-        L.append("      __builtin_amdgcn_wave_barrier();")
-        L.extend(self._span_copy_lines())
-        L.append("    } else if (active) {")
-        L.append("      char* w = (char*)out_data + my_start;")
-        L.extend("  " + ln for ln in self._mem_format_body(out_types, False))
-        L.append("    }")
-        L.append("  }")
-        L.append("}")
-        return "\n".join(L)
-
-    MEM_WRITE_CAP = 16384  # bytes/wave (mem rows ~200 B, all kept worst case)
-
-    def _span_copy_lines(self):
-        """Cooperative span copy LDS -> global: aligned 8B global stores via a
-        uniform funnel shift (see the csv writer; identical here)."""
-        return [
-            "      char* dst = (char*)out_data + span_start;",
-            "      long long a0 = (8 - (span_start & 7)) & 7;",
-            "      if (lane < a0 && lane < span) dst[lane] = wave_lds[lane];",
-            "      int sh = (int)(a0 & 7) * 8;",
-            "      for (long long b = a0 + (long long)lane * 8; b + 8 <= span;"
-            " b += 64 * 8) {",
-            "        unsigned long long lo ="
-            " *(const unsigned long long*)(wave_lds + (b - a0));",
-            "        unsigned long long hi ="
-            " *(const unsigned long long*)(wave_lds + (b - a0) + 8);",
-            "        unsigned long long v = sh ? ((lo >> sh) | (hi << (64 - sh)))"
-            " : lo;",
-            "        *(unsigned long long*)(dst + b) = v;",
-            "      }",
-            "      long long t0 = span > a0 ? a0 + ((span - a0) & ~7LL) : span;",
-            "      if (t0 + lane < span) dst[t0 + lane] = wave_lds[t0 + lane];",
-        ]
-
-    def _mem_format_body(self, out_types, lds):
-        n_opt = sum(1 for t in out_types if T.is_opt(t))
-        bitmap = ((n_opt + 63) // 64) * 8 if n_opt else 0
-        has_var = any(T.is_varlen(t) for t in out_types)
-        nf = len(out_types)
-        fixed_end = bitmap + 8 * nf
-        if bitmap > 8:
-            raise CodegenError(">64 optional fields unsupported")
-        L = []
-
-        def st8(off_expr, val_expr):
-            if lds:  # LDS dest, rows unaligned in the stream -> byte-safe
-                L.append("    { long long _t8 = (long long)(%s);"
-                         " __builtin_memcpy(w + (%s), &_t8, 8); }"
-                         % (val_expr, off_expr))
-            else:    # global: HW-supported unaligned typed store
-                L.append("    *(long long*)(w + (%s)) = (long long)(%s);"
-                         % (off_expr, val_expr))
-
+        L.append("  long long stride = (long long)gridDim.x * blockDim.x;")
+        L.append("  long long tid0 = (long long)blockIdx.x * blockDim.x + threadIdx.x;")
+        L.append("  for (long long i = tid0; i < n; i += stride) {")
+        L.append("    if (!keep[i]) continue;")
+        L.append("    unsigned char* w = out_data + size_scan[i];")
+        L.append("    out_offs[keep_scan[i]] = out_byte0 + size_scan[i];")
+        L.append("    out_rowidx[keep_scan[i]] = row0 + i;")
+        # load columnar values
         for k, t in enumerate(out_types):
             base = T.deopt(t)
             if base == T.STR:
@@ -1555,6 +1483,7 @@ class StageCodegen:
             if T.is_opt(t):
                 L.append("    bool v%d_n = ((const unsigned char*)outv[%d])[i] != 0;"
                          % (k, 3 * k + 2))
+        # bitmap
         if bitmap:
             L.append("    unsigned long long bm = 0;")
             oc = 0
@@ -1562,45 +1491,38 @@ class StageCodegen:
                 if T.is_opt(t):
                     L.append("    if (v%d_n) bm |= 1ULL << %d;" % (k, oc))
                     oc += 1
-            st8("0", "bm")
+            L.append("    *(unsigned long long*)w = bm;")
+            if bitmap > 8:
+                raise CodegenError(">64 optional fields unsupported")
+        # slots + varlen
         L.append("    long long var_off = 0;  // within varlen region")
         for k, t in enumerate(out_types):
             base = T.deopt(t)
-            off = "%d" % (bitmap + 8 * k)
+            slot = "((long long*)(w + %d))[%d]" % (bitmap, k)
             if base == T.STR:
                 null_guard = ("v%d_n" % k) if T.is_opt(t) else "false"
-                L.append("    if (%s) {" % null_guard)
-                st8(off, "0")
-                L.append("    } else {")
-                L.append("      long long off%d = %d + 8 + var_off - %d;"
-                         % (k, fixed_end - bitmap, 8 * k))
-                st8(off, "off%d | ((v%d.n + 1) << 32)" % (k, k))
+                L.append("    if (%s) { %s = 0; } else {" % (null_guard, slot))
+                L.append("      long long off = %d + 8 + var_off - %d;"
+                         % (fixed_end - bitmap, 8 * k))
+                L.append("      %s = off | ((v%d.n + 1) << 32);" % (slot, k))
                 L.append("      char* d = (char*)(w + %d + var_off);" % (fixed_end + 8))
                 L.append("      tpx_memcpy(d, v%d.p, v%d.n);" % (k, k))
                 L.append("      d[v%d.n] = 0;" % k)
                 L.append("      var_off += v%d.n + 1;" % k)
                 L.append("    }")
             elif base == T.F64:
+                L.append("    %s = __double_as_longlong(v%d);" % (slot, k))
                 if T.is_opt(t):
-                    L.append("    if (v%d_n) {" % k)
-                    st8(off, "0")
-                    L.append("    } else {")
-                    st8(off, "__double_as_longlong(v%d)" % k)
-                    L.append("    }")
-                else:
-                    st8(off, "__double_as_longlong(v%d)" % k)
+                    L.append("    if (v%d_n) %s = 0;" % (k, slot))
             else:
+                L.append("    %s = v%d;" % (slot, k))
                 if T.is_opt(t):
-                    L.append("    if (v%d_n) {" % k)
-                    st8(off, "0")
-                    L.append("    } else {")
-                    st8(off, "v%d" % k)
-                    L.append("    }")
-                else:
-                    st8(off, "v%d" % k)
+                    L.append("    if (v%d_n) %s = 0;" % (k, slot))
         if has_var:
-            st8("%d" % fixed_end, "var_off")
-        return L
+            L.append("    *(long long*)(w + %d) = var_off;" % fixed_end)
+        L.append("  }")
+        L.append("}")
+        return "\n".join(L)
 
     # CSV write: one lane per INPUT row (coalesced columnar loads; a kept-row
     # gather was measured SLOWER — gather defeats load coalescing). The 64
