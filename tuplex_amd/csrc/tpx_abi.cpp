@@ -445,7 +445,8 @@ using PreChunkFn = std::function<int(long long, long long, hipStream_t)>;
 static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
                         long long row0, long long in_bytes, tpx_result* res,
                         const PayloadSrc& psrc, hipStream_t stream,
-                        int64_t flags = 0, const PreChunkFn* pre_chunk = nullptr) {
+                        int64_t flags = 0, const PreChunkFn* pre_chunk = nullptr,
+                        const std::vector<void*>* in_tab = nullptr) {
     const StageDesc& D = st->desc;
     int dev = cur_device();
     Persist& P = g_persist[dev];
@@ -495,7 +496,8 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
     // Aggregate sinks consume the full keep/outv arrays at once -> C=1.
     bool mem_sink = D.sink == "mem";
     int C = 1;
-    if (D.agg.empty() && D.aggby.empty() && n >= (1 << 20) && d_offs) {
+    if (D.agg.empty() && D.aggby.empty() && n >= (1 << 20) &&
+        (d_offs || in_tab)) {
         static int env_c = [] {
             const char* e = getenv("TPX_CHUNKS");
             int v = e ? atoi(e) : 4;
@@ -550,6 +552,32 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
         for (int c = 1; c < C; ++c)
             d_outv_c[c] = (char*)blk + outv.size() * sizeof(void*) * (size_t)(c - 1);
     }
+    // columnar source: per-chunk biased INPUT tables (same 3-slot layout as
+    // outputs; strings keep their unbiased data base, offsets/masks shift)
+    void* d_in_c[8] = {d_in};
+    if (C > 1 && in_tab) {
+        size_t ns = in_tab->size();
+        std::vector<void*> iall(ns * (size_t)(C - 1));
+        for (int c = 1; c < C; ++c) {
+            void** tv = iall.data() + ns * (size_t)(c - 1);
+            for (size_t k3 = 0; k3 < ns; k3 += 3) {
+                const ColDesc& col = D.in_cols[k3 / 3];
+                char* v0 = (char*)(*in_tab)[k3];
+                char* v1 = (char*)(*in_tab)[k3 + 1];
+                char* v2 = (char*)(*in_tab)[k3 + 2];
+                long long stride0 = col.kind == K_BOOL ? 1 : 8;
+                tv[k3] = v0 ? v0 + cstart[c] * stride0 : nullptr;
+                tv[k3 + 1] = v1;  // string data base: unbiased
+                tv[k3 + 2] = v2 ? v2 + cstart[c] : nullptr;
+            }
+        }
+        void* blk = g_arena[dev].take(iall.size() * sizeof(void*) + 8);
+        if (!blk) return -1;
+        HIP_CHECK(hipMemcpyAsync(blk, iall.data(), iall.size() * sizeof(void*),
+                                 hipMemcpyHostToDevice, stream));
+        for (int c = 1; c < C; ++c)
+            d_in_c[c] = (char*)blk + ns * sizeof(void*) * (size_t)(c - 1);
+    }
 
     // +thread-chunk slack per LAUNCH: <=2048 blocks x 256 threads x 256 B
     unsigned long long heap_cap =
@@ -589,16 +617,17 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
             long long nc = ccnt[c];
             long long waves = (nc + 63) / 64;
             unsigned grid = (unsigned)std::min<long long>((waves + 1) / 2, 4096);
-            void* in_offs_c = (char*)d_offs + cstart[c] * 8;
+            void* in_offs_c = d_offs ? (char*)d_offs + cstart[c] * 8 : nullptr;
+            void* d_in_use = in_tab ? d_in_c[c] : d_in;
             long long row0_c = row0 + cstart[c];
             void* heap_c = (char*)P.heap + (unsigned long long)c * hs;
             void* cursor_c = (char*)P.counters + 16 * c;
             void* keep_c = (char*)d_keep + cstart[c];
             void* keep01_c = (char*)d_keep01 + cstart[c] * 8;
             void* sizes_c = (char*)d_sizes + cstart[c] * 8;
-            void* args[] = {&d_in, &in_offs_c, &nc, &row0_c, &heap_c, &cursor_c,
-                            &hs, &keep_c, &keep01_c, &sizes_c, &P.exc,
-                            &d_exc_count, &exc_cap, &d_outv_c[c]};
+            void* args[] = {&d_in_use, &in_offs_c, &nc, &row0_c, &heap_c,
+                            &cursor_c, &hs, &keep_c, &keep01_c, &sizes_c,
+                            &P.exc, &d_exc_count, &exc_cap, &d_outv_c[c]};
             if (launch(st->k_main, grid, 128, sc, args)) return -1;
             hipEventRecord(cev[(size_t)c * 5], sc);
         }
@@ -1011,8 +1040,9 @@ extern "C" int64_t tpx_stage_execute_col(tpx_stage* st, void* const* slots,
     HIP_CHECK(hipMemcpyAsync(d_tab, slots, (size_t)n_slots * sizeof(void*),
                              hipMemcpyHostToDevice, stream));
     PayloadSrc psrc{nullptr, nullptr, 0};
+    std::vector<void*> ht(slots, slots + n_slots);
     return run_core(st, d_tab, nullptr, n_rows, first_row, in_bytes, res, psrc,
-                    stream, flags);
+                    stream, flags, nullptr, &ht);
 }
 
 // ---------------------------------------------------------------------------------
