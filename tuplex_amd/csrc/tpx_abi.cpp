@@ -135,7 +135,6 @@ struct tpx_stage {
     hipModule_t module = nullptr;
     hipFunction_t k_main = nullptr, k_write = nullptr;
     hipFunction_t k_scan_block = nullptr, k_scan_add = nullptr;
-    hipFunction_t k_emit_kept = nullptr;
     hipFunction_t k_pair_total = nullptr;
     hipFunction_t k_csv_chunk = nullptr, k_csv_sel = nullptr, k_csv_rows = nullptr;
     hipFunction_t k_red_f64 = nullptr, k_red_f64_fin = nullptr;
@@ -229,7 +228,6 @@ extern "C" tpx_stage* tpx_stage_compile(const char* hip_source,
         {"tpx_stage_write", &st->k_write, true},
         {"tpx_scan_block", &st->k_scan_block, true},
         {"tpx_scan_add", &st->k_scan_add, true},
-        {"tpx_emit_kept", &st->k_emit_kept, true},
         {"tpx_pair_total", &st->k_pair_total, true},
         {"tpx_csv_chunk_stats", &st->k_csv_chunk, false},
         {"tpx_csv_select_counts", &st->k_csv_sel, false},

@@ -1212,20 +1212,6 @@ extern "C" __global__ void tpx_hashagg_emit(const long long* __restrict__ tkeys,
 #define TPX_SCAN_ITEMS 8
 #define TPX_SCAN_BLOCK (TPX_SCAN_THREADS * TPX_SCAN_ITEMS)
 
-// scatter the input-row index of every kept row to kept_idx[output position]:
-// lets tpx_stage_write iterate over KEPT rows with full lanes (at selectivity
-// s only s of the lanes would otherwise do work) and gives each wave of 64
-// kept rows a contiguous output byte span it can stage through LDS.
-extern "C" __global__ void tpx_emit_kept(const unsigned char* __restrict__ keep,
-                                         const long long* __restrict__ keep_scan,
-                                         long long n,
-                                         long long* __restrict__ kept_idx) {
-    long long stride = (long long)gridDim.x * blockDim.x;
-    for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-         i += stride)
-        if (keep[i]) kept_idx[keep_scan[i]] = i;
-}
-
 // one-thread helper: totals of two exclusive scans (in[n-1] + out[n-1]) into a
 // 16B slot, so the host fetches both chunk totals with ONE tiny D2H copy
 extern "C" __global__ void tpx_pair_total(const long long* __restrict__ a_in,
