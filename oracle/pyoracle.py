@@ -354,6 +354,11 @@ def run_pipeline(data: List[Any], ops: List[tuple], columns: Optional[List[str]]
             results.append((idx, r[1]))
         elif r[0] == "exc":
             record_exc(r[1])
+        elif r[0] == "rows":
+            for v2 in r[1]:
+                results.append((idx, v2))
+            for e in r[2]:
+                record_exc(e)
 
     if merge_rows_in_order:
         results.sort(key=lambda t: t[0])
@@ -397,17 +402,36 @@ def _agg_row(v, columns):
 
 def process_row(value, row_ops, columns, fast):
     """Run one row through row_ops with the reference's dual-mode semantics.
-    Returns ("row", value) | ("drop",) | ("exc", exception).
+    Returns ("row", value) | ("drop",) | ("exc", exception) | ("rows",
+    [v...], [exc...]) (a join with duplicate build keys expands 1:N; each
+    joined row continues the remaining ops independently).
     fast=True: compiled-path int()/float() semantics; False: interpreter."""
-    cur = value
-    cols = list(columns) if columns else None
-    k = 0
+    return _process_from(value, list(columns) if columns else None, row_ops,
+                         0, fast)
+
+
+def _process_from(cur, cols, row_ops, k, fast):
     while k < len(row_ops):
         op = row_ops[k]
         kind = op[0]
         if kind in ("resolve", "ignore"):
             k += 1
             continue
+        if kind == "join":
+            matches = _join_matches(op, cur, cols)
+            if len(matches) > 1:
+                rows_out, excs = [], []
+                for m in matches:
+                    row2, cols2 = _join_build_row(op, cur, cols, m)
+                    r = _process_from(row2, cols2, row_ops, k + 1, fast)
+                    if r[0] == "row":
+                        rows_out.append(r[1])
+                    elif r[0] == "exc":
+                        excs.append(r[1])
+                    elif r[0] == "rows":
+                        rows_out.extend(r[1])
+                        excs.extend(r[2])
+                return ("rows", rows_out, excs)
         try:
             cur, cols, dropped = _apply_op(op, cur, cols, fast)
             if dropped:
@@ -508,31 +532,46 @@ def _join_table(op):
     if ent is None or ent[0] is not op:
         rrows, rcols, rk = op[1], op[2], op[4]
         rki = rcols.index(rk)
-        t = {r[rki]: r for r in rrows}
+        t = {}
+        for r in rrows:  # bucket lists in build order
+            t.setdefault(r[rki], []).append(r)
         _JOIN_TABLES[id(op)] = (op, t)
         return t
     return ent[1]
 
 
-def _apply_join(op, cur, cols):
-    """Inner/left hash join against a materialized build side (single-match:
-    build keys are unique — logical/JoinOperator.cc:164 output layout
-    | left cols except key | key | right cols except key |; the key keeps the
-    left name and never nulls; left join nulls the right columns)."""
+def _join_matches(op, cur, cols):
     _, rrows, rcols, lk, rk, how, lp, ls, rp, rs = op
     row = _as_row(cur)
     cols2 = list(cols) if cols else None
     if not cols2 or lk not in cols2:
         raise ValueError("join: unknown left key column %r" % lk)
+    key = row[cols2.index(lk)]
+    return [] if key is None else _join_table(op).get(key, [])
+
+
+def _join_build_row(op, cur, cols, m):
+    _, rrows, rcols, lk, rk, how, lp, ls, rp, rs = op
+    row = _as_row(cur)
+    cols2 = list(cols) if cols else None
     lki = cols2.index(lk)
     rki = rcols.index(rk)
-    key = row[lki]
-    m = None if key is None else _join_table(op).get(key)
-    if m is None and how == "inner":
-        return cur, cols2, True
     right_vals = tuple((None if m is None else m[j])
                       for j in range(len(rcols)) if j != rki)
-    out = tuple(v for i, v in enumerate(row) if i != lki) + (key,) + right_vals
+    out = tuple(v for i, v in enumerate(row) if i != lki) + (row[lki],) + \
+        right_vals
     out_cols = ([lp + c + ls for c in cols2 if c != lk] + [lp + lk + ls] +
                 [rp + c + rs for c in rcols if c != rk])
+    return out, out_cols
+
+
+def _apply_join(op, cur, cols):
+    """0/1-match step (multi-match forks handled in _process_from);
+    logical/JoinOperator.cc:164 layout."""
+    how = op[5]
+    matches = _join_matches(op, cur, cols)
+    if not matches and how == "inner":
+        return cur, cols, True
+    out, out_cols = _join_build_row(op, cur, cols,
+                                    matches[0] if matches else None)
     return out, out_cols, False

@@ -346,14 +346,20 @@ def run_csv_pipeline(csv_bytes: bytes, ops, columns=None, header=None,
             r = _replay_row(line, col_types, null_values, row_ops, names,
                             delim.decode(), used=used_cols)
         if r[0] == "row":
-            v = r[1]
-            if isinstance(v, tuple) and len(v) == 1:
-                v = v[0]
-            out_rows[i] = v
+            out_rows[i] = [r[1]]
         elif r[0] == "exc":
             record(r[1])
+        elif r[0] == "rows":  # 1:N join expansion
+            out_rows[i] = r[1]
+            for e in r[2]:
+                record(e)
 
-    rows = [out_rows[i] for i in sorted(out_rows)]
+    rows = []
+    for i in sorted(out_rows):
+        for v in out_rows[i]:
+            if isinstance(v, tuple) and len(v) == 1:
+                v = v[0]
+            rows.append(v)
     result = {"exception_counts": exc_counts, "columns": names,
               "col_types": col_types, "has_header": has_header}
     if agg is not None:

@@ -46,14 +46,40 @@ def test_join_string_key_compile():
     assert "tpx_jhash_bytes" in src
 
 
-def test_join_duplicate_keys_raise():
+def test_join_duplicate_keys_fallback_semantics():
+    """Duplicate build keys: 1:N expansion — interpreter path this round
+    (plan falls back); every match emits a row in BUILD order and each joined
+    row runs the remaining ops independently."""
     from tuplex_amd import plan
     from tuplex_amd import ttypes as T
-    bad = [(1, "a"), (1, "b")]
-    with pytest.raises(ValueError, match="duplicate"):
-        plan.build_stage([T.I64], ["key"],
-                         [("join", bad, ["k", "x"], "key", "k", "inner",
-                           "", "", "", "")])
+    dup = [(1, "a"), (1, "b"), (2, "c")]
+    jop = ("join", dup, ["k", "x"], "key", "k", "inner", "", "", "", "")
+    sp = plan.build_stage([T.I64, T.STR], ["key", "val"], [jop])
+    assert not sp.compilable
+    assert "duplicate" in sp.why_not_compilable
+
+    rows = [(1, "L"), (2, "M"), (9, "N")]
+    r = pyoracle.run_pipeline(rows, [jop], columns=["key", "val"])
+    assert r["output"] == [("L", 1, "a"), ("L", 1, "b"), ("M", 2, "c")]
+
+    def wide(x):
+        return (x["val"], x["x"].upper())
+
+    r2 = pyoracle.run_pipeline(rows, [jop, ("map", wide)],
+                               columns=["key", "val"])
+    assert r2["output"] == [("L", "A"), ("L", "B"), ("M", "C")]
+
+    # per-duplicate exceptions: a post-join op failing on ONE duplicate
+    # leaves the other outputs intact
+    def picky(x):
+        if x["x"] == "b":
+            raise ValueError("b")
+        return (x["val"], x["x"])
+
+    r3 = pyoracle.run_pipeline(rows, [jop, ("map", picky)],
+                               columns=["key", "val"])
+    assert r3["output"] == [("L", "a"), ("M", "c")]
+    assert r3["exception_counts"] == {"ValueError": 1}
 
 
 def test_join_oracle_semantics():

@@ -490,11 +490,17 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
                                            logical_ops, names, delim.decode(),
                                            used=sp.used_source_cols)
                     if r[0] == "row":
-                        replayed[row] = r[1]
+                        replayed[row] = [r[1]]
                     elif r[0] == "exc":
                         nm = type(r[1]).__name__
                         out.exception_counts[nm] = \
                             out.exception_counts.get(nm, 0) + 1
+                    elif r[0] == "rows":  # 1:N join expansion
+                        replayed[row] = r[1]
+                        for e in r[2]:
+                            nm = type(e).__name__
+                            out.exception_counts[nm] = \
+                                out.exception_counts.get(nm, 0) + 1
             if sink_kind == "mem":
                 out_bytes = ctypes.string_at(res.out_data, res.out_size)
                 rws = rowfmt.deserialize_partition(out_bytes,
@@ -529,24 +535,26 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
                     table[row[0]] = table.get(row[0], initial) + row[1]
                 ki = sp.output_columns.index(key_cols[0])
                 for i in sorted(replayed):
-                    v = replayed[i]
-                    rt = v if isinstance(v, tuple) else (v,)
-                    k = rt[ki]
-                    table[k] = agg_fn(table.get(k, initial),
-                                      _agg_row(v, sp.output_columns))
+                    for v in replayed[i]:
+                        rt = v if isinstance(v, tuple) else (v,)
+                        k = rt[ki]
+                        table[k] = agg_fn(table.get(k, initial),
+                                          _agg_row(v, sp.output_columns))
                 out.rows = [(k, val) for k, val in table.items()]
             else:
                 _, combine_fn, agg_fn, initial = agg
                 acc = initial + sum(r[0] for r in rows)  # per-chunk partials
                 for i in sorted(replayed):
-                    acc = agg_fn(acc, _agg_row(replayed[i], sp.output_columns))
+                    for v in replayed[i]:
+                        acc = agg_fn(acc, _agg_row(v, sp.output_columns))
                 out.rows = [acc]
             return out
         merged = {}
         for i, row in all_rows:
-            merged[i] = row[0] if len(row) == 1 else row
-        merged.update(replayed)
-        out.rows = [merged[i] for i in sorted(merged)]
+            merged.setdefault(i, []).append(row[0] if len(row) == 1 else row)
+        for i, lst in replayed.items():
+            merged[i] = list(lst)
+        out.rows = [v for i in sorted(merged) for v in merged[i]]
         if agg is not None:
             from .engine import _agg_row
             _, combine_fn, agg_fn, initial = agg
@@ -580,11 +588,16 @@ def _run_csv_fallback(out, data, col_types, null_values, logical_ops, names,
     for i, line in enumerate(split_rows(data)):
         r = replay_csv_row(line, col_types, null_values, logical_ops, names, delim)
         if r[0] == "row":
-            rows_out[i] = r[1]
+            rows_out[i] = [r[1]]
         elif r[0] == "exc":
             nm = type(r[1]).__name__
             out.exception_counts[nm] = out.exception_counts.get(nm, 0) + 1
-    out.rows = [rows_out[i] for i in sorted(rows_out)]
+        elif r[0] == "rows":  # 1:N join expansion
+            rows_out[i] = r[1]
+            for e in r[2]:
+                nm = type(e).__name__
+                out.exception_counts[nm] = out.exception_counts.get(nm, 0) + 1
+    out.rows = [v for i in sorted(rows_out) for v in rows_out[i]]
     agg = next((op for op in logical_ops if op[0] == "aggregate"), None)
     if agg is not None:
         from .engine import _agg_row
@@ -633,10 +646,10 @@ def _merge_csv_segments(text: bytes, idxs, offs, replayed, row_lo,
                 gi += 1
             segs.append(text[offs[start]:offs[gi]])
         else:
-            v = rep[ri][1]
+            for v in rep[ri][1]:
+                row = v if isinstance(v, tuple) else (v,)
+                segs.append(_format_csv_row(list(row)))
             ri += 1
-            row = v if isinstance(v, tuple) else (v,)
-            segs.append(_format_csv_row(list(row)))
     return b"".join(segs)
 
 

@@ -303,6 +303,13 @@ def run_collect(data: List[Any], logical_ops: List[tuple],
                     name = type(r[1]).__name__
                     out.exception_counts[name] = \
                         out.exception_counts.get(name, 0) + 1
+                elif r[0] == "rows":  # 1:N join expansion
+                    for v2 in r[1]:
+                        yield v2
+                    for e in r[2]:
+                        name = type(e).__name__
+                        out.exception_counts[name] = \
+                            out.exception_counts.get(name, 0) + 1
 
         if sp.agg_unique and sp.agg_key_idx is not None:
             keys = {row[0] for row in er.rows}
@@ -335,7 +342,7 @@ def run_collect(data: List[Any], logical_ops: List[tuple],
         # outputs carry their exact local row index (out_row_indices) -> map back
         # to original data indices for the in-order merge (ResolveTask.cc:878)
         for row, local in zip(er.rows, er.row_indices):
-            results[norm_idx[local]] = _unwrap_row(row)
+            results.setdefault(norm_idx[local], []).append(_unwrap_row(row))
         replay = [(norm_idx[r], data[norm_idx[r]]) for (r, _, _) in er.exceptions]
     elif not sp.compilable:
         out.mode = "fallback"
@@ -349,12 +356,18 @@ def run_collect(data: List[Any], logical_ops: List[tuple],
     for i, v in sorted(replay):
         r = resolve.replay_row(v, logical_ops, columns, scalar_input)
         if r[0] == "row":
-            results[i] = r[1]
+            results[i] = [r[1]]
         elif r[0] == "exc":
             name = type(r[1]).__name__
             out.exception_counts[name] = out.exception_counts.get(name, 0) + 1
+        elif r[0] == "rows":  # 1:N join expansion
+            results[i] = r[1]
+            for e in r[2]:
+                name = type(e).__name__
+                out.exception_counts[name] = \
+                    out.exception_counts.get(name, 0) + 1
 
-    merged = [results[i] for i in sorted(results)]
+    merged = [v for i in sorted(results) for v in results[i]]
     if any(op[0] == "unique" for op in logical_ops):
         merged = list(dict.fromkeys(merged))
     agg = next((op for op in logical_ops if op[0] == "aggregate"), None)

@@ -172,6 +172,13 @@ def run_orc(ctx, src, logical_ops, sink=None):
                     nm = type(r[1]).__name__
                     out.exception_counts[nm] = \
                         out.exception_counts.get(nm, 0) + 1
+                elif r[0] == "rows":
+                    for v2 in r[1]:
+                        yield v2
+                    for e in r[2]:
+                        nm = type(e).__name__
+                        out.exception_counts[nm] = \
+                            out.exception_counts.get(nm, 0) + 1
 
         if sp.agg_unique and sp.agg_key_idx is not None:
             keys = {row[0] for row in er.rows}
@@ -197,14 +204,19 @@ def run_orc(ctx, src, logical_ops, sink=None):
 
     results = {}
     for row, local in zip(er.rows, er.row_indices):
-        results[local] = _unwrap_row(row)
+        results.setdefault(local, []).append(_unwrap_row(row))
     for i, r in replay_iter([r for (r, _, _) in er.exceptions]):
         if r[0] == "row":
-            results[i] = r[1]
+            results[i] = [r[1]]
         elif r[0] == "exc":
             nm = type(r[1]).__name__
             out.exception_counts[nm] = out.exception_counts.get(nm, 0) + 1
-    merged = [results[i] for i in sorted(results)]
+        elif r[0] == "rows":
+            results[i] = r[1]
+            for e in r[2]:
+                nm = type(e).__name__
+                out.exception_counts[nm] = out.exception_counts.get(nm, 0) + 1
+    merged = [v for i in sorted(results) for v in results[i]]
     if any(op[0] == "unique" for op in logical_ops):
         merged = list(dict.fromkeys(merged))
     out.rows = merged
@@ -220,11 +232,16 @@ def _fallback_all(out, tab, names, logical_ops, why):
     for i, v in enumerate(rows):
         r = resolve.replay_row(v, logical_ops, names, scalar_input)
         if r[0] == "row":
-            results[i] = r[1]
+            results[i] = [r[1]]
         elif r[0] == "exc":
             nm = type(r[1]).__name__
             out.exception_counts[nm] = out.exception_counts.get(nm, 0) + 1
-    merged = [results[i] for i in sorted(results)]
+        elif r[0] == "rows":
+            results[i] = r[1]
+            for e in r[2]:
+                nm = type(e).__name__
+                out.exception_counts[nm] = out.exception_counts.get(nm, 0) + 1
+    merged = [v for i in sorted(results) for v in results[i]]
     if any(op[0] == "unique" for op in logical_ops):
         merged = list(dict.fromkeys(merged))
     agg = next((op for op in logical_ops if op[0] == "aggregate"), None)

@@ -249,8 +249,11 @@ def build_stage(input_types, input_columns, logical_ops) -> StageProgram:
                 raise ValueError("join: null keys in the build side are "
                                  "unsupported this round")
             if len(set(keys)) != len(keys):
-                raise ValueError("join: duplicate build-side keys are "
-                                 "unsupported this round (SURVEY.md §8f-3)")
+                # 1:N expansion breaks the one-output-per-input-row engine
+                # contract; the interpreter path forks rows correctly
+                # (resolve._replay_from) — GPU dup-join is round-2 work
+                _fallback(sp, "duplicate build-side join keys (1:N) run on "
+                          "the interpreter path this round")
             # per-column right types over the materialized rows
             rtypes = []
             for j in range(len(rcols)):
