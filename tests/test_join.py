@@ -194,3 +194,17 @@ def test_join_gpu_flights_pipeline(tmp_path):
     assert ds.exception_counts == ref["exception_counts"]
     assert len(got) > 0
     assert any(r[-1] is None for r in got) or all(r[-1] for r in got)
+
+
+def test_join_duplicate_keys_product_fallback():
+    """End-to-end product path: dup-key join -> whole-stage interpreter
+    fallback (CPU-runnable), output matches the oracle incl. 1:N expansion."""
+    dup = [(1, "a"), (1, "b"), (2, "c")]
+    rows = [(1, "L"), (2, "M"), (9, "N")]
+    ctx = tuplex_amd.Context()
+    rds = ctx.parallelize([r for r in dup], columns=["k", "x"])
+    ds = ctx.parallelize(rows, columns=["key", "val"]).join(rds, "key", "k")
+    got = ds.collect()
+    assert ds._last_outcome.mode == "fallback"
+    assert "duplicate" in ds._last_outcome.fallback_reason
+    assert got == [("L", 1, "a"), ("L", 1, "b"), ("M", 2, "c")]
