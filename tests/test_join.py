@@ -54,9 +54,18 @@ def test_join_duplicate_keys_fallback_semantics():
     from tuplex_amd import ttypes as T
     dup = [(1, "a"), (1, "b"), (2, "c")]
     jop = ("join", dup, ["k", "x"], "key", "k", "inner", "", "", "", "")
+    # TERMINAL dup join: GPU-capable (write kernel loops the bucket)
     sp = plan.build_stage([T.I64, T.STR], ["key", "val"], [jop])
-    assert not sp.compilable
-    assert "duplicate" in sp.why_not_compilable
+    assert sp.compilable and getattr(sp.ops[-1], "join_dup", False)
+    # mid-pipeline dup join: interpreter path
+
+    def postop(x):
+        return (x[0], x[2])
+
+    spm = plan.build_stage([T.I64, T.STR], ["key", "val"],
+                           [jop, ("map", postop)])
+    assert not spm.compilable
+    assert "duplicate" in spm.why_not_compilable
 
     rows = [(1, "L"), (2, "M"), (9, "N")]
     r = pyoracle.run_pipeline(rows, [jop], columns=["key", "val"])
@@ -201,9 +210,13 @@ def test_join_duplicate_keys_product_fallback():
     fallback (CPU-runnable), output matches the oracle incl. 1:N expansion."""
     dup = [(1, "a"), (1, "b"), (2, "c")]
     rows = [(1, "L"), (2, "M"), (9, "N")]
+    def tail(x):
+        return (x["val"], x["key"], x["x"])
+
     ctx = tuplex_amd.Context()
     rds = ctx.parallelize([r for r in dup], columns=["k", "x"])
-    ds = ctx.parallelize(rows, columns=["key", "val"]).join(rds, "key", "k")
+    ds = ctx.parallelize(rows, columns=["key", "val"]).join(
+        rds, "key", "k").map(tail)  # post-join op -> interpreter path
     got = ds.collect()
     assert ds._last_outcome.mode == "fallback"
     assert "duplicate" in ds._last_outcome.fallback_reason

@@ -248,12 +248,7 @@ def build_stage(input_types, input_columns, logical_ops) -> StageProgram:
             if any(k is None for k in keys):
                 raise ValueError("join: null keys in the build side are "
                                  "unsupported this round")
-            if len(set(keys)) != len(keys):
-                # 1:N expansion breaks the one-output-per-input-row engine
-                # contract; the interpreter path forks rows correctly
-                # (resolve._replay_from) — GPU dup-join is round-2 work
-                _fallback(sp, "duplicate build-side join keys (1:N) run on "
-                          "the interpreter path this round")
+            op._join_dup = len(set(keys)) != len(keys)
             # per-column right types over the materialized rows
             rtypes = []
             for j in range(len(rcols)):
@@ -290,6 +285,19 @@ def build_stage(input_types, input_columns, logical_ops) -> StageProgram:
         op.out_types = list(cur_types)
         op.out_columns = list(cur_cols) if cur_cols else None
         sp.ops.append(op)
+    # duplicate-key joins: the GPU engine supports the 1:N expansion only
+    # when the join is the TERMINAL operator (its write kernel loops the
+    # bucket; mid-pipeline expansion would break the one-slot-per-input-row
+    # columnar contract) — otherwise the interpreter path forks the rows
+    # (resolve._replay_from)
+    for i2, op in enumerate(sp.ops):
+        if op.kind == "join" and getattr(op, "_join_dup", False):
+            if (i2 != len(sp.ops) - 1 or sp.aggregate is not None
+                    or op.resolvers or op.ignores):
+                _fallback(sp, "duplicate build-side join keys (1:N) "
+                          "mid-pipeline: interpreter path this round")
+            else:
+                op.join_dup = True
     sp.used_source_cols = _used_source_columns(sp) if sp.compilable else None
     return sp
 
