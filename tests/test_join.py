@@ -221,3 +221,64 @@ def test_join_duplicate_keys_product_fallback():
     assert ds._last_outcome.mode == "fallback"
     assert "duplicate" in ds._last_outcome.fallback_reason
     assert got == [("L", 1, "a"), ("L", 1, "b"), ("M", 2, "c")]
+
+
+@pytest.mark.gpu
+def test_join_gpu_duplicate_keys():
+    """GPU 1:N: terminal dup-key join — keep01 carries bucket counts, the
+    write kernel loops the bucket; output order = input row order x build
+    order, exactly the oracle's fork order."""
+    import random
+    dup = [(1, "a", 10), (1, "b", None), (2, "c", 30), (7, "z", 5),
+           (1, "d", 40)]
+    rng = random.Random(5)
+    rows = []
+    for i in range(30000):
+        v = "v%d" % i
+        if i % 411 == 0:
+            v = "vé%d" % i  # non-ASCII -> divert -> multi-row replay fork
+        rows.append((rng.randint(0, 8), v))
+    jop = ("join", dup, ["k", "x", "w"], "key", "k", "inner",
+           "", "", "", "")
+    ctx = tuplex_amd.Context()
+    ds = apply_ops(ctx.parallelize(rows, columns=["key", "val"]), [jop])
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    ref = pyoracle.run_pipeline(rows, [jop], columns=["key", "val"])
+    assert got == ref["output"]
+    assert ds.exception_counts == ref["exception_counts"]
+    per_key1 = sum(1 for r in rows if r[0] == 1)
+    assert sum(1 for g in got if g[1] == 1) == 3 * per_key1
+
+    jopl = ("join", dup, ["k", "x", "w"], "key", "k", "left",
+            "", "", "", "")
+    dsl = apply_ops(ctx.parallelize(rows, columns=["key", "val"]), [jopl])
+    gotl = dsl.collect()
+    assert dsl._last_outcome.mode == "gpu"
+    refl = pyoracle.run_pipeline(rows, [jopl], columns=["key", "val"])
+    assert gotl == refl["output"]
+
+
+@pytest.mark.gpu
+def test_join_gpu_duplicate_keys_csv(tmp_path):
+    """Dup-key join fed from the CSV source (string keys)."""
+    import os
+    import random
+    dim = [("aa", "A1"), ("aa", "A2"), ("bb", "B1"), ("cc", "C1")]
+    rng = random.Random(6)
+    lines = [b"code,n"]
+    for i in range(20000):
+        lines.append(b"%s,%d" % (rng.choice([b"aa", b"bb", b"cc", b"xx"]), i))
+    data = b"\n".join(lines) + b"\n"
+    p = os.path.join(str(tmp_path), "d.csv")
+    with open(p, "wb") as f:
+        f.write(data)
+    jop = ("join", dim, ["k", "label"], "code", "k", "inner", "", "", "", "")
+    ctx = tuplex_amd.Context()
+    ds = apply_ops(ctx.csv(p), [jop])
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    from oracle import pyoracle_csv
+    ref = pyoracle_csv.run_csv_pipeline(data, [jop])
+    assert got == ref["output"]
+    assert ds.exception_counts == ref["exception_counts"]

@@ -214,22 +214,58 @@ class StageCodegen:
         L2, no extra kernel argument or ABI change (HashJoinStage build side;
         probe = open addressing, same layout as the host build here)."""
         rrows, rcols, lki, rki, how = op.join
+        dup = getattr(op, "join_dup", False)
         nm = "jt%d" % op.opid
-        n = len(rrows)
+        if dup:
+            # bucket the build rows by key (first-occurrence order, build
+            # order within — hashmap.cc chains); value arrays are FLAT in
+            # bucket order, the slot table stores (start, count)
+            buckets = {}
+            order = []
+            for r in rrows:
+                k = r[rki]
+                if k not in buckets:
+                    buckets[k] = []
+                    order.append(k)
+                buckets[k].append(r)
+            flat = [r for k in order for r in buckets[k]]
+            ukeys = order
+            n = len(ukeys)
+        else:
+            flat = rrows
+            n = len(rrows)
         tsize = 8
         while tsize < 2 * max(n, 1):
             tsize <<= 1
-        str_key = isinstance(rrows[0][rki], str) if n else False
+        str_key = isinstance(rrows[0][rki], str) if rrows else False
         occ = [0] * tsize
-        slotrow = [0] * tsize
-        for ri, r in enumerate(rrows):
-            k = r[rki]
-            h = (self._jhash_bytes(k.encode()) if str_key
-                 else self._jhash_i64(int(k))) & (tsize - 1)
-            while occ[h]:
-                h = (h + 1) & (tsize - 1)
-            occ[h] = 1
-            slotrow[h] = ri
+        slotrow = [0] * tsize   # slot -> build row (unique) / bucket index (dup)
+        jstart = [0] * tsize
+        jcnt = [0] * tsize
+        if dup:
+            pos = 0
+            starts = {}
+            for k in ukeys:
+                starts[k] = pos
+                pos += len(buckets[k])
+            for ui, k in enumerate(ukeys):
+                h = (self._jhash_bytes(k.encode()) if str_key
+                     else self._jhash_i64(int(k))) & (tsize - 1)
+                while occ[h]:
+                    h = (h + 1) & (tsize - 1)
+                occ[h] = 1
+                slotrow[h] = ui
+                jstart[h] = starts[k]
+                jcnt[h] = len(buckets[k])
+        else:
+            for ri, r in enumerate(rrows):
+                k = r[rki]
+                h = (self._jhash_bytes(k.encode()) if str_key
+                     else self._jhash_i64(int(k))) & (tsize - 1)
+                while occ[h]:
+                    h = (h + 1) & (tsize - 1)
+                occ[h] = 1
+                slotrow[h] = ri
         blob = bytearray()
 
         def put(sval):
@@ -241,11 +277,16 @@ class StageCodegen:
         D = self.join_defs
         D.append("__device__ const unsigned char %s_occ[%d] = {%s};"
                  % (nm, tsize, ",".join(str(x) for x in occ)))
+        def slot_key(s2):
+            if dup:
+                return flat[jstart[s2]][rki]
+            return rrows[slotrow[s2]][rki]
+
         if str_key:
             koff, klen = [], []
             for s2 in range(tsize):
                 if occ[s2]:
-                    o2, l2 = put(rrows[slotrow[s2]][rki])
+                    o2, l2 = put(slot_key(s2))
                 else:
                     o2, l2 = 0, 0
                 koff.append(o2)
@@ -257,14 +298,23 @@ class StageCodegen:
         else:
             D.append("__device__ const long long %s_key[%d] = {%s};"
                      % (nm, tsize,
-                        ",".join(str(int(rrows[slotrow[s2]][rki])) if occ[s2]
+                        ",".join(str(int(slot_key(s2))) if occ[s2]
                                  else "0" for s2 in range(tsize))))
+        if dup:
+            D.append("__device__ const int %s_jst[%d] = {%s};"
+                     % (nm, tsize, ",".join(map(str, jstart))))
+            D.append("__device__ const int %s_jct[%d] = {%s};"
+                     % (nm, tsize, ",".join(map(str, jcnt))))
         out_info = []  # (j, ctail) for right cols except key
+        vlen = max(len(flat), 1) if dup else tsize
         for j in range(len(rcols)):
             if j == rki:
                 continue
-            vals = [rrows[slotrow[s2]][j] if occ[s2] else None
-                    for s2 in range(tsize)]
+            if dup:
+                vals = [r[j] for r in flat] or [None]  # empty build side
+            else:
+                vals = [rrows[slotrow[s2]][j] if occ[s2] else None
+                        for s2 in range(tsize)]
             kindc = None
             for v in vals:
                 if v is None:
@@ -274,8 +324,11 @@ class StageCodegen:
                          "f64" if isinstance(v, float) else "i64")
                 break
             kindc = kindc or "i64"
-            has_null = any(occ[s2] and rrows[slotrow[s2]][j] is None
-                           for s2 in range(tsize))
+            if dup:
+                has_null = any(r[j] is None for r in flat)
+            else:
+                has_null = any(occ[s2] and rrows[slotrow[s2]][j] is None
+                               for s2 in range(tsize))
             if kindc == "str":
                 offs, lens = [], []
                 for v in vals:
@@ -286,27 +339,29 @@ class StageCodegen:
                     offs.append(o2)
                     lens.append(l2)
                 D.append("__device__ const int %s_c%d_off[%d] = {%s};"
-                         % (nm, j, tsize, ",".join(map(str, offs))))
+                         % (nm, j, vlen, ",".join(map(str, offs))))
                 D.append("__device__ const int %s_c%d_len[%d] = {%s};"
-                         % (nm, j, tsize, ",".join(map(str, lens))))
+                         % (nm, j, vlen, ",".join(map(str, lens))))
             elif kindc == "f64":
                 D.append("__device__ const double %s_c%d[%d] = {%s};"
-                         % (nm, j, tsize,
+                         % (nm, j, vlen,
                             ",".join(repr(float(v)) if isinstance(v, (int, float))
                                      and not isinstance(v, bool) else "0.0"
                                      for v in vals)))
             else:
                 D.append("__device__ const long long %s_c%d[%d] = {%s};"
-                         % (nm, j, tsize,
+                         % (nm, j, vlen,
                             ",".join(str(int(v)) if v is not None
                                      and not isinstance(v, str) else "0"
                                      for v in vals)))
             if has_null:
+                if dup:
+                    nl = ["1" if r[j] is None else "0" for r in flat]
+                else:
+                    nl = ["1" if (occ[s2] and rrows[slotrow[s2]][j] is None)
+                          else "0" for s2 in range(tsize)]
                 D.append("__device__ const unsigned char %s_c%d_null[%d] = {%s};"
-                         % (nm, j, tsize,
-                            ",".join("1" if (occ[s2] and
-                                             rrows[slotrow[s2]][j] is None)
-                                     else "0" for s2 in range(tsize))))
+                         % (nm, j, vlen, ",".join(nl)))
             out_info.append((j, kindc, has_null))
         if blob:
             # string blob as concatenated hex-escaped literals
@@ -319,13 +374,13 @@ class StageCodegen:
                      % (nm, len(bs) + 1, "\n".join(parts)))
         else:
             D.append("__device__ const char %s_blob[1] = \"\";" % nm)
-        return nm, tsize, str_key, out_info
+        return nm, tsize, str_key, out_info, dup
 
     def _emit_join(self, em, op, rc):
         """Probe an embedded build table; extend the row context with the
         right columns (JoinOperator.cc:164 layout)."""
         rrows, rcols, lki, rki, how = op.join
-        nm, tsize, str_key, out_info = self._emit_join_table(op)
+        nm, tsize, str_key, out_info, dup = self._emit_join_table(op)
         kv, _kt, knv = rc[lki]
         ji = em.fresh("ji")
         em.w("int %s = -1;" % ji)
@@ -350,6 +405,16 @@ class StageCodegen:
         em.w("}")
         if how == "inner":
             em.w("if (%s < 0) { o.keep = false; return 0; }" % ji)
+        if dup:
+            # terminal 1:N: hand the bucket to the write kernel via the hidden
+            # (start, count) columnar pair; left-join miss = (-1, 1)
+            sv = em.fresh("jbs")
+            cv = em.fresh("jbc")
+            em.w("int %s = %s < 0 ? -1 : %s_jst[%s];" % (sv, ji, nm, ji))
+            em.w("int %s = %s < 0 ? 1 : %s_jct[%s];" % (cv, ji, nm, ji))
+            self.jdup = {"nm": nm, "out_info": out_info, "how": how,
+                         "start_var": sv, "cnt_var": cv, "op": op}
+            return [rc[i2] for i2 in range(len(rc)) if i2 != lki] + [rc[lki]]
         new_rc = [rc[i2] for i2 in range(len(rc)) if i2 != lki] + [rc[lki]]
         left_join = how == "left"
         for j, kindc, has_null in out_info:
@@ -955,12 +1020,32 @@ class StageCodegen:
         return rc
 
     # ---- full source --------------------------------------------------------
+    def _dup_join_op(self):
+        """The terminal duplicate-key join op, if this stage has one."""
+        sp = self.sp
+        if sp.ops and sp.ops[-1].kind == "join" and \
+                getattr(sp.ops[-1], "join_dup", False):
+            return sp.ops[-1]
+        return None
+
     def generate(self) -> Tuple[str, str]:
         """Returns (hip_source, stage_desc)."""
         sp = self.sp
         in_types = sp.input_types
         out_types = sp.gpu_output_types if getattr(sp, "agg_expr", None) is not None \
             else sp.output_types
+
+        dup_op = self._dup_join_op()
+        if dup_op is not None and self.sink != "mem":
+            raise CodegenError("duplicate-key join with a %s sink: round 2"
+                               % self.sink)
+        # terminal dup join: the process stores LEFT columns + a hidden
+        # (bucket_start, count) pair; the write kernel loops the bucket and
+        # reads the right columns from the embedded table (1:N expansion)
+        store_types = out_types
+        if dup_op is not None:
+            n_leftout = len(dup_op.in_types)
+            store_types = list(out_types[:n_leftout]) + [T.I64, T.I64]
 
         body = _Emitter()
         # inputs: c0..cN (+ c0_n null flags)
@@ -972,22 +1057,30 @@ class StageCodegen:
 
         # write outputs into Out struct
         out_fields = []
-        for k, t in enumerate(out_types):
+        for k, t in enumerate(store_types):
             ct = _ctype(t)
             out_fields.append("    %s o%d;" % (ct, k))
             if T.is_opt(t):
                 out_fields.append("    bool o%d_n;" % k)
-        for k, ((v, t, nv), ot) in enumerate(zip(final_rc, out_types)):
+        n_rc = len(store_types) - (2 if dup_op is not None else 0)
+        for k, ((v, t, nv), ot) in enumerate(zip(final_rc[:n_rc],
+                                                 store_types[:n_rc])):
             cast = "(double)" if T.deopt(ot) == T.F64 and T.deopt(t) in (T.I64, T.BOOL) else ""
             body.w("o.o%d = %s(%s);" % (k, cast, v))
             if T.is_opt(ot):
                 body.w("o.o%d_n = %s;" % (k, nv if nv else "false"))
+        if dup_op is not None:
+            body.w("o.o%d = (long long)%s;" % (n_rc, self.jdup["start_var"]))
+            body.w("o.o%d = (long long)%s;" % (n_rc + 1, self.jdup["cnt_var"]))
         body.w("o.keep = true;")
         body.w("return 0;")
 
         # generate kernels BEFORE assembling (they may add string literals)
-        main_src = self._main_kernel(in_types, out_types)
-        write_src = self._write_kernel(out_types)
+        self.store_types = store_types
+        self.full_out_types = out_types
+        main_src = self._main_kernel(in_types, store_types)
+        write_src = self._write_kernel(out_types) if dup_op is None \
+            else self._write_kernel_mem_dup(out_types, store_types)
         src = [
             "// generated by tuplex_amd.codegen — stage %s" % sp.signature(),
             runtime_header(),
@@ -1007,7 +1100,7 @@ class StageCodegen:
             main_src,
             write_src,
         ]
-        desc = self._desc(in_types, out_types)
+        desc = self._desc(in_types, store_types)
         return "\n".join(src), desc
 
     def _process_signature(self, in_types):
@@ -1152,7 +1245,12 @@ class StageCodegen:
         L.append("    }")
         L.append("    if (!o.keep) { keep[i] = 0; keep01[i] = 0; sizes[i] = 0; continue; }")
         L.extend(self._store_columnar(out_types, lds))
-        if self.sink == "mem":
+        if self._dup_join_op() is not None:
+            # keep01 = bucket count (the compaction scan then yields 1:N
+            # output row offsets); left-join miss stored as count 1
+            L.append("    keep[i] = 1; keep01[i] = o.o%d;"
+                     % (len(out_types) - 1))
+        elif self.sink == "mem":
             L.append("    keep[i] = 1; keep01[i] = 1;")
         else:
             # bit 1: no cell of this row needs quoting (write fast path;
@@ -1395,7 +1493,43 @@ class StageCodegen:
             if T.is_opt(t):
                 L.append("    ((unsigned char*)outv[%d])[i] = o.o%d_n ? 1 : 0;"
                          % (3 * k + 2, k))
-        if self.sink == "mem":
+        dup_op = self._dup_join_op()
+        if self.sink == "mem" and dup_op is not None:
+            # 1:N rows: sizes[i] = count * (fixed + left varlen) + right
+            # varlen summed over the bucket (full-row layout, FULL out types)
+            full = self.full_out_types
+            n_opt = sum(1 for t in full if T.is_opt(t))
+            bitmap = ((n_opt + 63) // 64) * 8 if n_opt else 0
+            has_var = any(T.is_varlen(t) for t in full)
+            fixed = bitmap + 8 * len(full) + (8 if has_var else 0)
+            n_left = len(dup_op.in_types)
+            jd = self.jdup
+            nm = jd["nm"]
+            L.append("    long long lvar = 0;")
+            for k in range(n_left):
+                if T.is_varlen(full[k]):
+                    if T.is_opt(full[k]):
+                        L.append("    if (!o.o%d_n) lvar += o.o%d.n + 1;"
+                                 % (k, k))
+                    else:
+                        L.append("    lvar += o.o%d.n + 1;" % k)
+            L.append("    long long rvar = 0;")
+            L.append("    for (int _d = 0; _d < o.o%d; ++_d) {"
+                     % (len(out_types) - 1))
+            L.append("      int _fi = o.o%d < 0 ? 0 : (int)o.o%d + _d;"
+                     % (len(out_types) - 2, len(out_types) - 2))
+            L.append("      bool _miss = o.o%d < 0;" % (len(out_types) - 2))
+            for (j, kindc, has_null) in jd["out_info"]:
+                if kindc == "str":
+                    cond = "!_miss"
+                    if has_null:
+                        cond += " && !%s_c%d_null[_fi]" % (nm, j)
+                    L.append("      if (%s) rvar += %s_c%d_len[_fi] + 1;"
+                             % (cond, nm, j))
+            L.append("    }")
+            L.append("    sizes[i] = (long long)o.o%d * (%d + lvar) + rvar;"
+                     % (len(out_types) - 1, fixed))
+        elif self.sink == "mem":
             n_opt = sum(1 for t in out_types if T.is_opt(t))
             bitmap = ((n_opt + 63) // 64) * 8 if n_opt else 0
             has_var = any(T.is_varlen(t) for t in out_types)
@@ -1520,6 +1654,125 @@ class StageCodegen:
                     L.append("    if (v%d_n) %s = 0;" % (k, slot))
         if has_var:
             L.append("    *(long long*)(w + %d) = var_off;" % fixed_end)
+        L.append("  }")
+        L.append("}")
+        return "\n".join(L)
+
+    def _write_kernel_mem_dup(self, full_types, store_types):
+        """mem-sink writer for a TERMINAL duplicate-key join: one thread per
+        input row, looping its bucket — row d reads the right columns from the
+        embedded build table at (start + d); left columns repeat. Output rows
+        land at keep_scan[i] + d / size_scan[i] + running offset (keep01
+        carried the bucket counts, so the scans already yield 1:N offsets)."""
+        jd = self.jdup
+        nm = jd["nm"]
+        dup_op = jd["op"]
+        n_left = len(dup_op.in_types)
+        n_store = len(store_types)
+        full = full_types
+        n_opt = sum(1 for t in full if T.is_opt(t))
+        bitmap = ((n_opt + 63) // 64) * 8 if n_opt else 0
+        has_var = any(T.is_varlen(t) for t in full)
+        nf = len(full)
+        fixed_end = bitmap + 8 * nf
+        if bitmap > 8:
+            raise CodegenError(">64 optional fields unsupported")
+        # right-col info by FULL index: full[n_left + idx] <- out_info[idx]
+        rinfo = {n_left + i2: info for i2, info in enumerate(jd["out_info"])}
+        L = []
+        L.append('extern "C" __global__ void tpx_stage_write(')
+        L.append("    const unsigned char* __restrict__ keep,")
+        L.append("    const long long* __restrict__ keep_scan,")
+        L.append("    const long long* __restrict__ size_scan,")
+        L.append("    long long n, long long row0, void** outv,")
+        L.append("    unsigned char* __restrict__ out_data, long long* __restrict__ out_offs,")
+        L.append("    long long* __restrict__ out_rowidx,")
+        L.append("    long long total_rows, long long total_bytes,")
+        L.append("    long long out_byte0) {")
+        L.append("  long long stride = (long long)gridDim.x * blockDim.x;")
+        L.append("  long long tid0 = (long long)blockIdx.x * blockDim.x + threadIdx.x;")
+        L.append("  for (long long i = tid0; i < n; i += stride) {")
+        L.append("    if (!keep[i]) continue;")
+        L.append("    int jb_s = (int)((const long long*)outv[%d])[i];" % (3 * (n_store - 2)))
+        L.append("    int jb_c = (int)((const long long*)outv[%d])[i];" % (3 * (n_store - 1)))
+        # left values loaded once
+        for k in range(n_left):
+            t = full[k]
+            base = T.deopt(t)
+            if base == T.STR:
+                L.append("    tstr v%d{(const char*)((const unsigned long long*)outv[%d])[i],"
+                         " (long long)((const int*)outv[%d])[i]};" % (k, 3 * k, 3 * k + 1))
+            elif base == T.F64:
+                L.append("    double v%d = ((const double*)outv[%d])[i];" % (k, 3 * k))
+            else:
+                L.append("    long long v%d = ((const long long*)outv[%d])[i];" % (k, 3 * k))
+            if T.is_opt(t):
+                L.append("    bool v%d_n = ((const unsigned char*)outv[%d])[i] != 0;"
+                         % (k, 3 * k + 2))
+        L.append("    unsigned char* w = out_data + size_scan[i];")
+        L.append("    long long obase = out_byte0 + size_scan[i];")
+        L.append("    long long kbase = keep_scan[i];")
+        L.append("    for (int _d = 0; _d < jb_c; ++_d) {")
+        L.append("      bool _miss = jb_s < 0;")
+        L.append("      int _fi = _miss ? 0 : jb_s + _d;")
+        # per-dup right values
+        for k in range(n_left, nf):
+            j, kindc, has_null = rinfo[k]
+            t = full[k]
+            nullv = "_miss"
+            if has_null:
+                nullv = "(_miss || %s_c%d_null[_fi])" % (nm, j)
+            if kindc == "str":
+                L.append("      tstr v%d{%s_blob + %s_c%d_off[_fi],"
+                         " %s_c%d_len[_fi]};" % (k, nm, nm, j, nm, j))
+            elif kindc == "f64":
+                L.append("      double v%d = %s_c%d[_fi];" % (k, nm, j))
+            elif kindc == "bool":
+                L.append("      long long v%d = %s_c%d[_fi];" % (k, nm, j))
+            else:
+                L.append("      long long v%d = %s_c%d[_fi];" % (k, nm, j))
+            if T.is_opt(t):
+                L.append("      bool v%d_n = %s;" % (k, nullv))
+        # header entries for this output row
+        L.append("      out_offs[kbase + _d] = obase;")
+        L.append("      out_rowidx[kbase + _d] = row0 + i;")
+        if bitmap:
+            L.append("      unsigned long long bm = 0;")
+            oc = 0
+            for k, t in enumerate(full):
+                if T.is_opt(t):
+                    L.append("      if (v%d_n) bm |= 1ULL << %d;" % (k, oc))
+                    oc += 1
+            L.append("      *(unsigned long long*)w = bm;")
+        L.append("      long long var_off = 0;")
+        for k, t in enumerate(full):
+            base = T.deopt(t)
+            slot = "((long long*)(w + %d))[%d]" % (bitmap, k)
+            if base == T.STR:
+                null_guard = ("v%d_n" % k) if T.is_opt(t) else "false"
+                L.append("      if (%s) { %s = 0; } else {" % (null_guard, slot))
+                L.append("        long long off = %d + 8 + var_off - %d;"
+                         % (fixed_end - bitmap, 8 * k))
+                L.append("        %s = off | ((v%d.n + 1) << 32);" % (slot, k))
+                L.append("        char* d2 = (char*)(w + %d + var_off);"
+                         % (fixed_end + 8))
+                L.append("        tpx_memcpy(d2, v%d.p, v%d.n);" % (k, k))
+                L.append("        d2[v%d.n] = 0;" % k)
+                L.append("        var_off += v%d.n + 1;" % k)
+                L.append("      }")
+            elif base == T.F64:
+                L.append("      %s = __double_as_longlong(v%d);" % (slot, k))
+                if T.is_opt(t):
+                    L.append("      if (v%d_n) %s = 0;" % (k, slot))
+            else:
+                L.append("      %s = v%d;" % (slot, k))
+                if T.is_opt(t):
+                    L.append("      if (v%d_n) %s = 0;" % (k, slot))
+        if has_var:
+            L.append("      *(long long*)(w + %d) = var_off;" % fixed_end)
+        L.append("      long long rsz = %d + var_off;" % (fixed_end + (8 if has_var else 0)))
+        L.append("      w += rsz; obase += rsz;")
+        L.append("    }")
         L.append("  }")
         L.append("}")
         return "\n".join(L)
