@@ -275,10 +275,10 @@ def test_join_gpu_duplicate_keys_csv(tmp_path):
         f.write(data)
     jop = ("join", dim, ["k", "label"], "code", "k", "inner", "", "", "", "")
     ctx = tuplex_amd.Context()
-    ds = apply_ops(ctx.csv(p), [jop])
+    ds = apply_ops(ctx.csv(p, header=True), [jop])
     got = ds.collect()
     assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
     from oracle import pyoracle_csv
-    ref = pyoracle_csv.run_csv_pipeline(data, [jop])
+    ref = pyoracle_csv.run_csv_pipeline(data, [jop], header=True)
     assert got == ref["output"]
     assert ds.exception_counts == ref["exception_counts"]
