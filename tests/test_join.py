@@ -282,3 +282,32 @@ def test_join_gpu_duplicate_keys_csv(tmp_path):
     ref = pyoracle_csv.run_csv_pipeline(data, [jop], header=True)
     assert got == ref["output"]
     assert ds.exception_counts == ref["exception_counts"]
+
+
+@pytest.mark.gpu
+def test_join_gpu_duplicate_keys_tocsv(tmp_path):
+    """Dup-key join straight to the csv sink (file->file, 1:N rows in the
+    device-formatted output)."""
+    import os
+    import random
+    dim = [("aa", "A1", 1), ("aa", "A2", 2), ("bb", "B1", 3)]
+    rng = random.Random(8)
+    lines = [b"code,n"]
+    for i in range(15000):
+        lines.append(b"%s,%d" % (rng.choice([b"aa", b"bb", b"xx"]), i))
+    data = b"\n".join(lines) + b"\n"
+    p = os.path.join(str(tmp_path), "in.csv")
+    with open(p, "wb") as f:
+        f.write(data)
+    outp = os.path.join(str(tmp_path), "out.csv")
+    jop = ("join", dim, ["k", "label", "r"], "code", "k", "inner",
+           "", "", "", "")
+    ctx = tuplex_amd.Context()
+    ds = apply_ops(ctx.csv(p, header=True), [jop])
+    ds.tocsv(outp)
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    from oracle import pyoracle_csv
+    ref = pyoracle_csv.run_csv_pipeline(data, [jop], header=True, sink="csv")
+    with open(outp, "rb") as f:
+        got = f.read()
+    assert got == ref["csv_text"]

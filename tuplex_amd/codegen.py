@@ -1036,9 +1036,6 @@ class StageCodegen:
             else sp.output_types
 
         dup_op = self._dup_join_op()
-        if dup_op is not None and self.sink != "mem":
-            raise CodegenError("duplicate-key join with a %s sink: round 2"
-                               % self.sink)
         # terminal dup join: the process stores LEFT columns + a hidden
         # (bucket_start, count) pair; the write kernel loops the bucket and
         # reads the right columns from the embedded table (1:N expansion)
@@ -1079,8 +1076,12 @@ class StageCodegen:
         self.store_types = store_types
         self.full_out_types = out_types
         main_src = self._main_kernel(in_types, store_types)
-        write_src = self._write_kernel(out_types) if dup_op is None \
-            else self._write_kernel_mem_dup(out_types, store_types)
+        if dup_op is None:
+            write_src = self._write_kernel(out_types)
+        elif self.sink == "mem":
+            write_src = self._write_kernel_mem_dup(out_types, store_types)
+        else:
+            write_src = self._write_kernel_csv_dup(out_types, store_types)
         src = [
             "// generated by tuplex_amd.codegen — stage %s" % sp.signature(),
             runtime_header(),
@@ -1475,7 +1476,10 @@ class StageCodegen:
         # scan then runs on ds_read instead of re-reading global memory; the
         # bytes are identical either way)
         if self.sink != "mem":
-            L.extend(self._csv_size(out_types))
+            if self._dup_join_op() is not None:
+                L.extend(self._csv_size_dup(out_types))
+            else:
+                L.extend(self._csv_size(out_types))
         for k, t in enumerate(out_types):
             base = T.deopt(t)
             if base == T.STR:
@@ -1574,6 +1578,176 @@ class StageCodegen:
                 raise CodegenError("csv sink for %r not supported yet" % (t,))
         L.append("    sizes[i] = sz;")
         return L
+
+    def _csv_size_dup(self, store_types):
+        """csv-sink sizes for a terminal dup join: count * (delims + newline +
+        left cells) + right cells summed over the bucket."""
+        jd = self.jdup
+        nm = jd["nm"]
+        dup_op = jd["op"]
+        full = self.full_out_types
+        n_left = len(dup_op.in_types)
+        rinfo = {n_left + i2: info for i2, info in enumerate(jd["out_info"])}
+        L = ["    bool _anyq = false;"]
+        L.append("    long long lsz = %d;  // delimiters + newline" % len(full))
+        for k in range(n_left):
+            t = full[k]
+            base = T.deopt(t)
+            if base == T.STR:
+                pre = ""
+                if T.is_opt(t):
+                    L.append("    if (!o.o%d_n) {" % k)
+                    pre = "  "
+                L.append(pre + "    { long long q%d;" % k)
+                L.append(pre + "      bool nq%d = tpx_csv_needs_quote(o.o%d,"
+                         " &q%d);" % (k, k, k))
+                L.append(pre + "      lsz += nq%d ? o.o%d.n + q%d + 2 : o.o%d.n;"
+                         % (k, k, k, k))
+                L.append(pre + "      _anyq |= nq%d; }" % k)
+                if T.is_opt(t):
+                    L.append("    }")
+            elif base == T.I64:
+                L.append("    lsz += tpx_i64_digits(o.o%d);" % k)
+            elif base == T.BOOL:
+                L.append("    lsz += o.o%d ? 4 : 5;" % k)
+            else:
+                raise CodegenError("csv sink for %r not supported yet" % (t,))
+        L.append("    long long rsz = 0;")
+        cnt = "o.o%d" % (len(store_types) - 1)
+        st = "o.o%d" % (len(store_types) - 2)
+        L.append("    for (int _d = 0; _d < %s; ++_d) {" % cnt)
+        L.append("      int _fi = %s < 0 ? 0 : (int)%s + _d;" % (st, st))
+        L.append("      bool _miss = %s < 0;" % st)
+        for k in range(n_left, len(full)):
+            j, kindc, has_null = rinfo[k]
+            t = full[k]
+            if kindc == "str":
+                guard = "!_miss"
+                if has_null:
+                    guard += " && !%s_c%d_null[_fi]" % (nm, j)
+                if T.is_opt(t):
+                    L.append("      if (%s) {" % guard)
+                else:
+                    L.append("      {")
+                L.append("        tstr _rv{%s_blob + %s_c%d_off[_fi],"
+                         " %s_c%d_len[_fi]};" % (nm, nm, j, nm, j))
+                L.append("        long long _q; bool _nq ="
+                         " tpx_csv_needs_quote(_rv, &_q);")
+                L.append("        rsz += _nq ? _rv.n + _q + 2 : _rv.n;")
+                L.append("        _anyq |= _nq;")
+                L.append("      }")
+            elif kindc in ("i64", "bool"):
+                if T.is_opt(t):
+                    raise CodegenError("csv sink for %r not supported yet"
+                                       % (t,))
+                if kindc == "bool":
+                    L.append("      rsz += %s_c%d[_fi] ? 4 : 5;" % (nm, j))
+                else:
+                    L.append("      rsz += tpx_i64_digits(%s_c%d[_fi]);"
+                             % (nm, j))
+            else:
+                raise CodegenError("csv sink for joined %s column" % kindc)
+        L.append("    }")
+        L.append("    sizes[i] = (long long)%s * lsz + rsz;" % cnt)
+        return L
+
+    def _write_kernel_csv_dup(self, full_types, store_types):
+        """csv-sink writer for a terminal dup join: thread per input row,
+        loops the bucket (perf-uncritical path; no LDS staging)."""
+        jd = self.jdup
+        nm = jd["nm"]
+        dup_op = jd["op"]
+        n_left = len(dup_op.in_types)
+        n_store = len(store_types)
+        full = full_types
+        rinfo = {n_left + i2: info for i2, info in enumerate(jd["out_info"])}
+        L = []
+        L.append('extern "C" __global__ void tpx_stage_write(')
+        L.append("    const unsigned char* __restrict__ keep,")
+        L.append("    const long long* __restrict__ keep_scan,")
+        L.append("    const long long* __restrict__ size_scan,")
+        L.append("    long long n, long long row0, void** outv,")
+        L.append("    unsigned char* __restrict__ out_data, long long* __restrict__ out_offs,")
+        L.append("    long long* __restrict__ out_rowidx,")
+        L.append("    long long total_rows, long long total_bytes,")
+        L.append("    long long out_byte0) {")
+        L.append("  long long stride = (long long)gridDim.x * blockDim.x;")
+        L.append("  long long tid0 = (long long)blockIdx.x * blockDim.x + threadIdx.x;")
+        L.append("  for (long long i = tid0; i < n; i += stride) {")
+        L.append("    if (!keep[i]) continue;")
+        L.append("    int jb_s = (int)((const long long*)outv[%d])[i];" % (3 * (n_store - 2)))
+        L.append("    int jb_c = (int)((const long long*)outv[%d])[i];" % (3 * (n_store - 1)))
+        for k in range(n_left):
+            t = full[k]
+            base = T.deopt(t)
+            if base == T.STR:
+                L.append("    tstr v%d{(const char*)((const unsigned long long*)outv[%d])[i],"
+                         " (long long)((const int*)outv[%d])[i]};" % (k, 3 * k, 3 * k + 1))
+            elif base == T.F64:
+                raise CodegenError("csv sink for f64 not supported yet")
+            else:
+                L.append("    long long v%d = ((const long long*)outv[%d])[i];" % (k, 3 * k))
+            if T.is_opt(t):
+                L.append("    bool v%d_n = ((const unsigned char*)outv[%d])[i] != 0;"
+                         % (k, 3 * k + 2))
+        L.append("    char* w = (char*)out_data + size_scan[i];")
+        L.append("    long long obase = out_byte0 + size_scan[i];")
+        L.append("    long long kbase = keep_scan[i];")
+        L.append("    for (int _d = 0; _d < jb_c; ++_d) {")
+        L.append("      bool _miss = jb_s < 0;")
+        L.append("      int _fi = _miss ? 0 : jb_s + _d;")
+        L.append("      char* w0 = w;")
+        L.append("      out_offs[kbase + _d] = obase;")
+        L.append("      out_rowidx[kbase + _d] = row0 + i;")
+        for k in range(len(full)):
+            t = full[k]
+            base = T.deopt(t)
+            if k:
+                L.append("      *w++ = ',';")
+            if k < n_left:
+                if base == T.STR:
+                    pre = ""
+                    if T.is_opt(t):
+                        L.append("      if (!v%d_n) {" % k)
+                        pre = "  "
+                    L.append(pre + "      w = tpx_csv_cell_write(w, v%d);" % k)
+                    if T.is_opt(t):
+                        L.append("      }")
+                elif base == T.I64:
+                    L.append("      { long long dl = tpx_i64_digits(v%d);"
+                             " tpx_i64_write(w, v%d, dl); w += dl; }" % (k, k))
+                elif base == T.BOOL:
+                    L.append("      { const char* s2 = v%d ? \"True\" : \"False\";"
+                             " long long l2 = v%d ? 4 : 5;"
+                             " for (long long j2 = 0; j2 < l2; ++j2) w[j2] = s2[j2];"
+                             " w += l2; }" % (k, k))
+            else:
+                j, kindc, has_null = rinfo[k]
+                if kindc == "str":
+                    guard = "!_miss"
+                    if has_null:
+                        guard += " && !%s_c%d_null[_fi]" % (nm, j)
+                    L.append("      if (%s) {" % guard)
+                    L.append("        tstr _rv{%s_blob + %s_c%d_off[_fi],"
+                             " %s_c%d_len[_fi]};" % (nm, nm, j, nm, j))
+                    L.append("        w = tpx_csv_cell_write(w, _rv);")
+                    L.append("      }")
+                elif kindc == "bool":
+                    L.append("      { bool b2 = %s_c%d[_fi] != 0;"
+                             " const char* s2 = b2 ? \"True\" : \"False\";"
+                             " long long l2 = b2 ? 4 : 5;"
+                             " for (long long j2 = 0; j2 < l2; ++j2) w[j2] = s2[j2];"
+                             " w += l2; }" % (nm, j))
+                else:
+                    L.append("      { long long rv2 = %s_c%d[_fi];"
+                             " long long dl = tpx_i64_digits(rv2);"
+                             " tpx_i64_write(w, rv2, dl); w += dl; }" % (nm, j))
+        L.append("      *w++ = '\\n';")
+        L.append("      obase += (long long)(w - w0);")
+        L.append("    }")
+        L.append("  }")
+        L.append("}")
+        return "\n".join(L)
 
     # -- write kernel ----------------------------------------------------------
     def _write_kernel(self, out_types):
