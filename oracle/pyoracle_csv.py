@@ -448,6 +448,12 @@ def _output_columns(row_ops, names):
         elif op[0] == "renameColumn":
             if cols:
                 cols = [op[2] if c == op[1] else c for c in cols]
+        elif op[0] == "join":
+            _, _rr, rcols, lk, rk, _how, lp, ls, rp, rs = op
+            if cols:
+                cols = ([lp + c + ls for c in cols if c != lk] +
+                        [lp + lk + ls] +
+                        [rp + c + rs for c in rcols if c != rk])
     return cols or []
 
 
