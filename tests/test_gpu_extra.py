@@ -395,3 +395,42 @@ def test_crlf_line_endings(tmp_path):
     ref2 = pyoracle_csv.run_csv_pipeline(dataq, [("map", use)])
     assert got2 == ref2["output"]
     assert got2[0][1] == "v\r\n0"
+
+
+def agg_comb2(a, b):
+    return a + b
+
+
+def agg_by_code(a, x):
+    return a + x["n"]
+
+
+def test_aggregate_by_string_key_gpu():
+    """STRING-key aggregateByKey on device (string hash table; slots claim
+    (ptr,len), host merges duplicate slots by key bytes)."""
+    import random
+    rng = random.Random(17)
+    codes = ["aa", "bb", "cc", "dd", "", "longer-key-name"]
+    rows = [(rng.choice(codes), rng.randint(0, 100)) for _ in range(50000)]
+    ctx = tuplex_amd.Context()
+    ds = ctx.parallelize(rows, columns=["code", "n"]).aggregateByKey(
+        agg_comb2, agg_by_code, 0, ["code"])
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    ref = pyoracle.run_pipeline(
+        rows, [("aggregateByKey", agg_comb2, agg_by_code, 0, ["code"])],
+        columns=["code", "n"])
+    assert sorted(got) == sorted(ref["output"])
+    assert len(got) == len(codes)
+
+
+def test_unique_strings_gpu():
+    import random
+    rng = random.Random(19)
+    vals = ["s%d" % rng.randint(0, 300) for _ in range(40000)] + ["", "x"]
+    ctx = tuplex_amd.Context()
+    ds = ctx.parallelize(vals).unique()
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    ref = pyoracle.run_pipeline(vals, [("unique",)])
+    assert sorted(got) == sorted(ref["output"])

@@ -2079,6 +2079,8 @@ class StageCodegen:
         if getattr(self.sp, "agg_expr", None) is not None:
             if getattr(self.sp, "agg_key_idx", None) is not None:
                 lines.append("aggby=%s" % T.deopt(self.sp.agg_type))
+                if T.deopt(getattr(self.sp, "agg_key_type", None)) == T.STR:
+                    lines.append("aggkeystr=1")
             else:
                 lines.append("agg=%s" % T.deopt(self.sp.agg_type))
         if self.csv_info.get("text_mode"):

@@ -76,6 +76,7 @@ struct StageDesc {
     std::string sink;    // mem | csv
     std::string agg;     // "" | i64 | f64 — GPU-reducible aggregate fold
     std::string aggby;   // "" | i64 | f64 — by-key hash-reduce (key col 0)
+    int aggkeystr = 0;   // by-key key column is a string
     int textmode = 0;    // text() source: rows split on every newline (no quotes)
     std::vector<ColDesc> in_cols, out_cols;
 };
@@ -105,6 +106,7 @@ static bool parse_desc(const char* text, StageDesc* d) {
     d->sink = kv.count("sink") ? kv["sink"] : "mem";
     d->agg = kv.count("agg") ? kv["agg"] : "";
     d->aggby = kv.count("aggby") ? kv["aggby"] : "";
+    d->aggkeystr = kv.count("aggkeystr") ? atoi(kv["aggkeystr"].c_str()) : 0;
     d->textmode = kv.count("textmode") ? atoi(kv["textmode"].c_str()) : 0;
     int nin = atoi(kv["nin"].c_str());
     int nout = atoi(kv["nout"].c_str());
@@ -140,6 +142,8 @@ struct tpx_stage {
     hipFunction_t k_red_f64 = nullptr, k_red_f64_fin = nullptr;
     hipFunction_t k_red_i64 = nullptr, k_red_i64_fin = nullptr;
     hipFunction_t k_hk_f64 = nullptr, k_hk_i64 = nullptr, k_hk_emit = nullptr;
+    hipFunction_t k_hk_sf64 = nullptr, k_hk_si64 = nullptr,
+                  k_hk_semit = nullptr;
     bool loaded = false;
 };
 
@@ -239,6 +243,9 @@ extern "C" tpx_stage* tpx_stage_compile(const char* hip_source,
         {"tpx_hashagg_f64", &st->k_hk_f64, false},
         {"tpx_hashagg_i64", &st->k_hk_i64, false},
         {"tpx_hashagg_emit", &st->k_hk_emit, false},
+        {"tpx_hashagg_str_f64", &st->k_hk_sf64, false},
+        {"tpx_hashagg_str_i64", &st->k_hk_si64, false},
+        {"tpx_hashagg_str_emit", &st->k_hk_semit, false},
     };
     for (auto& e : lut) {
         hipError_t r = hipModuleGetFunction(e.fn, st->module, e.name);
@@ -732,7 +739,131 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
         }
     }
 
-    if (!D.aggby.empty()) {
+    if (!D.aggby.empty() && D.aggkeystr) {
+        // STRING-key hash-reduce: slots claim the first row's (ptr,len); a
+        // not-yet-visible len makes the prober move on, so the same string
+        // may land in several slots — the HOST merges the emitted pairs by
+        // key bytes (sum is commutative). Key bytes are fetched from device
+        // memory (views into in_data / the string heap, both live here).
+        bool is_f64 = D.aggby == "f64";
+        hipFunction_t kfill = is_f64 ? st->k_hk_sf64 : st->k_hk_si64;
+        if (!kfill || !st->k_hk_semit) {
+            set_err("string hashagg kernels missing");
+            return -1;
+        }
+        unsigned long long tsize = 1ull << 20;
+        unsigned long long used = 0;
+        void* tkeys = nullptr;
+        void* tlens = nullptr;
+        void* tvals = nullptr;
+        ARENA_TAKE(d_hk_state, 40);
+        for (int attempt = 0;; ++attempt) {
+            tkeys = g_arena[dev].take(tsize * 8);
+            tlens = g_arena[dev].take(tsize * 4);
+            tvals = g_arena[dev].take(tsize * 8);
+            if (!tkeys || !tlens || !tvals) return -1;
+            HIP_CHECK(hipMemsetAsync(tkeys, 0, tsize * 8, stream));
+            HIP_CHECK(hipMemsetAsync(tlens, 0, tsize * 4, stream));
+            HIP_CHECK(hipMemsetAsync(tvals, 0, tsize * 8, stream));
+            HIP_CHECK(hipMemsetAsync(d_hk_state, 0, 40, stream));
+            unsigned long long tmask = tsize - 1;
+            void* kptr = outv[0];
+            void* klen = outv[1];
+            void* vals = outv[3];
+            void* d_used = d_hk_state;
+            void* d_ovf = (char*)d_hk_state + 8;
+            unsigned grid2 = (unsigned)std::min<long long>((n + 255) / 256,
+                                                           2048);
+            void* a1[] = {&kptr, &klen, &vals, &d_keep, &n, &tkeys, &tlens,
+                          &tvals, &tmask, &d_used, &d_ovf};
+            if (launch(kfill, grid2, 256, stream, a1)) return -1;
+            unsigned long long st8[2] = {0, 0};
+            HIP_CHECK(hipMemcpyAsync(st8, d_hk_state, 16,
+                                     hipMemcpyDeviceToHost, stream));
+            HIP_CHECK(hipStreamSynchronize(stream));
+            used = st8[0];
+            bool ovf = st8[1] != 0 || used * 2 > tsize;
+            if (!ovf) break;
+            if (attempt >= 3 || tsize >= (1ull << 27)) {
+                set_err("aggregateByKey table overflow");
+                return -1;
+            }
+            tsize *= 8;
+        }
+        hipEventRecord(evs1, stream);
+        long long cap = (long long)used + 1;
+        ARENA_TAKE(d_tri, (size_t)cap * 24 + 16);
+        void* d_oidx = (char*)d_hk_state + 32;
+        {
+            long long ts_ll = (long long)tsize;
+            unsigned grid2 = (unsigned)std::min<long long>(
+                ((long long)tsize + 255) / 256, 4096);
+            void* a2[] = {&tkeys, &tlens, &tvals, &ts_ll, &d_tri, &d_oidx};
+            if (launch(st->k_hk_semit, grid2, 256, stream, a2)) return -1;
+        }
+        hipEventRecord(ev2, stream);
+        unsigned long long nout = 0;
+        HIP_CHECK(hipMemcpyAsync(&nout, d_oidx, 8, hipMemcpyDeviceToHost,
+                                 stream));
+        HIP_CHECK(hipStreamSynchronize(stream));
+        std::vector<long long> tri((size_t)nout * 3);
+        if (nout)
+            HIP_CHECK(hipMemcpy(tri.data(), d_tri, (size_t)nout * 24,
+                                hipMemcpyDeviceToHost));
+        // fetch key bytes + merge duplicate slots by content
+        std::map<std::string, long long> merged;
+        for (unsigned long long e = 0; e < nout; ++e) {
+            long long len = tri[e * 3 + 1];
+            std::string key((size_t)len, 0);
+            if (len)
+                HIP_CHECK(hipMemcpy(&key[0], (void*)(uintptr_t)tri[e * 3],
+                                    (size_t)len, hipMemcpyDeviceToHost));
+            long long vb = tri[e * 3 + 2];
+            auto it = merged.find(key);
+            if (it == merged.end()) {
+                merged[key] = vb;
+            } else if (is_f64) {
+                double acc;
+                memcpy(&acc, &it->second, 8);
+                double add;
+                memcpy(&add, &vb, 8);
+                acc += add;
+                memcpy(&it->second, &acc, 8);
+            } else {
+                it->second += vb;
+            }
+        }
+        // build the (str, val) output partition (Serializer.cc layout:
+        // 2 slots + varlen_total + bytes, str slot = off|size<<32, off from
+        // the slot's own address)
+        size_t total_bytes2 = 8;
+        for (auto& kvp : merged)
+            total_bytes2 += 24 + kvp.first.size() + 1;
+        res->out_size = (int64_t)total_bytes2;
+        res->out_num_rows = (int64_t)merged.size();
+        res->out_data = (uint8_t*)malloc(total_bytes2);
+        ((int64_t*)res->out_data)[0] = (int64_t)merged.size();
+        res->out_row_offsets = (int64_t*)malloc((merged.size() + 1) * 8);
+        res->out_row_indices = (int64_t*)malloc((merged.size() + 1) * 8);
+        {
+            uint8_t* w = res->out_data + 8;
+            long long ri = 0;
+            for (auto& kvp : merged) {
+                res->out_row_offsets[ri] = (int64_t)(w - res->out_data);
+                res->out_row_indices[ri] = ri;
+                long long len = (long long)kvp.first.size();
+                ((int64_t*)w)[0] = 24 | ((len + 1) << 32);
+                ((int64_t*)w)[1] = kvp.second;
+                ((int64_t*)w)[2] = len + 1;
+                memcpy(w + 24, kvp.first.data(), (size_t)len);
+                w[24 + len] = 0;
+                w += 24 + len + 1;
+                ++ri;
+            }
+            res->out_row_offsets[ri] = (int64_t)total_bytes2;
+        }
+        res->bytes_out = res->out_size;
+    } else if (!D.aggby.empty()) {
         // by-key hash-reduce (hashmap.cc analog); output order unpinned
         hipFunction_t kfill = D.aggby == "f64" ? st->k_hk_f64 : st->k_hk_i64;
         if (!kfill || !st->k_hk_emit) { set_err("hashagg kernels missing"); return -1; }

@@ -1086,6 +1086,89 @@ __device__ __forceinline__ unsigned long long tpx_hash_i64(long long k) {
     return x;
 }
 
+// string-key variant: the slot key is the (ptr,len) of the FIRST row that
+// claimed it (CAS on the pointer; len stored after the win). A reader that
+// sees a claimed slot whose len is not yet visible just probes on — the same
+// string may then occupy several slots, and the HOST merges the emitted
+// (key,val) pairs by string content (sum is commutative), so no cross-lane
+// spin is ever needed (a same-wave spin on a diverged writer would deadlock
+// under wave-lockstep execution).
+__device__ __forceinline__ bool tpx_hk_bytes_eq(const char* a, const char* b,
+                                                int n) {
+    for (int i = 0; i < n; ++i)
+        if (a[i] != b[i]) return false;
+    return true;
+}
+
+#define TPX_HK_SDEF(NAME, VT, ATOMIC)                                        \
+extern "C" __global__ void NAME(const unsigned long long* __restrict__ kptr, \
+                                const int* __restrict__ klen,                \
+                                const VT* __restrict__ vals,                 \
+                                const unsigned char* __restrict__ keep,      \
+                                long long n, unsigned long long* tkeys,      \
+                                int* tlens, VT* tvals,                       \
+                                unsigned long long tmask,                    \
+                                unsigned long long* used, int* overflow) {   \
+    long long stride = (long long)gridDim.x * blockDim.x;                    \
+    for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;     \
+         i < n; i += stride) {                                               \
+        if (!keep[i]) continue;                                              \
+        unsigned long long p = kptr[i];                                      \
+        int len = klen[i];                                                   \
+        VT val = vals[i];                                                    \
+        unsigned long long h = tpx_jhash_bytes((const char*)p, len) & tmask; \
+        int probe = 0;                                                       \
+        for (; probe < TPX_HK_MAXPROBE; ++probe) {                           \
+            unsigned long long cur = tkeys[h];                               \
+            if (cur == 0) {                                                  \
+                unsigned long long prev =                                    \
+                    atomicCAS(&tkeys[h], 0ULL, p);                           \
+                if (prev == 0) {                                             \
+                    tlens[h] = len + 1; /* +1: 0 means not-ready */          \
+                    __threadfence();                                         \
+                    atomicAdd(used, 1ULL);                                   \
+                    ATOMIC(&tvals[h], val);                                  \
+                    break;                                                   \
+                }                                                            \
+                cur = prev;                                                  \
+            }                                                                \
+            int cl = tlens[h];                                               \
+            if (cl == len + 1 &&                                             \
+                (cur == p || tpx_hk_bytes_eq((const char*)cur,               \
+                                             (const char*)p, len))) {        \
+                ATOMIC(&tvals[h], val);                                      \
+                break;                                                       \
+            }                                                                \
+            h = (h + 1) & tmask;                                             \
+        }                                                                    \
+        if (probe == TPX_HK_MAXPROBE) *overflow = 1;                         \
+    }                                                                        \
+}
+
+#define TPX_HK_ADD_F64(ptr, v) atomicAdd(ptr, v)
+#define TPX_HK_ADD_I64(ptr, v) atomicAdd((unsigned long long*)(ptr), \
+                                         (unsigned long long)(v))
+TPX_HK_SDEF(tpx_hashagg_str_f64, double, TPX_HK_ADD_F64)
+TPX_HK_SDEF(tpx_hashagg_str_i64, long long, TPX_HK_ADD_I64)
+
+// emit claimed slots as [ptr,len,valbits] triples (host fetches key bytes and
+// merges duplicate slots by string content)
+extern "C" __global__ void tpx_hashagg_str_emit(
+        const unsigned long long* __restrict__ tkeys,
+        const int* __restrict__ tlens,
+        const long long* __restrict__ tvals, long long tsize,
+        long long* __restrict__ out3, unsigned long long* out_idx) {
+    long long stride = (long long)gridDim.x * blockDim.x;
+    for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+         i < tsize; i += stride) {
+        if (!tkeys[i] || tlens[i] == 0) continue;
+        unsigned long long w = atomicAdd(out_idx, 1ULL);
+        out3[w * 3 + 0] = (long long)tkeys[i];
+        out3[w * 3 + 1] = (long long)(tlens[i] - 1);
+        out3[w * 3 + 2] = tvals[i];
+    }
+}
+
 extern "C" __global__ void tpx_hashagg_f64(const long long* __restrict__ keys,
                                            const double* __restrict__ vals,
                                            const unsigned char* __restrict__ keep,

@@ -136,16 +136,16 @@ def build_stage(input_types, input_columns, logical_ops) -> StageProgram:
             # anything else -> interpreter fallback dedup
             sp.aggregate = (opid, None, None, 0, None)
             sp.agg_unique = True
-            if len(cur_types) == 1 and cur_types[0] == T.I64:
+            if len(cur_types) == 1 and cur_types[0] in (T.I64, T.STR):
                 from .udf import tir
                 sp.agg_expr = tir.const(1)
                 sp.agg_type = T.I64
                 sp.agg_key_idx = 0
-                sp.agg_key_type = T.I64
+                sp.agg_key_type = cur_types[0]
                 sp.agg_opid = opid
             else:
-                _fallback(sp, "unique() on %r (single i64 column on GPU this "
-                          "round)" % (cur_types,))
+                _fallback(sp, "unique() on %r (single i64/str column on GPU "
+                          "this round)" % (cur_types,))
             continue
         if kind == "aggregateByKey":
             key_cols = entry[4]
@@ -155,12 +155,12 @@ def build_stage(input_types, input_columns, logical_ops) -> StageProgram:
                     and key_cols[0] in cur_cols):
                 ki = cur_cols.index(key_cols[0])
                 kt = cur_types[ki]
-                if kt == T.I64:
+                if kt in (T.I64, T.STR):
                     sp.agg_key_idx = ki
                     sp.agg_key_type = kt
                 else:
-                    sp.agg_expr = None  # non-i64 key: interpreter fallback
-                    _fallback(sp, "aggregateByKey key type %r (i64 only on GPU "
+                    sp.agg_expr = None  # opt/f64 key: interpreter fallback
+                    _fallback(sp, "aggregateByKey key type %r (i64/str on GPU "
                               "this round)" % (kt,))
             else:
                 sp.agg_expr = None
