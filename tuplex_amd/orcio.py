@@ -40,9 +40,6 @@ def _arrow_base(t):
     return None
 
 
-_CAST = {T.I64: None, T.F64: None, T.BOOL: None, T.STR: None}
-
-
 def _target_type(base):
     return {T.I64: pa.int64(), T.F64: pa.float64(), T.BOOL: pa.bool_(),
             T.STR: pa.large_string()}[base]
