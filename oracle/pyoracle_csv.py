@@ -376,6 +376,9 @@ def run_csv_pipeline(csv_bytes: bytes, ops, columns=None, header=None,
 
     # csv sink
     out_cols = _output_columns(row_ops, names)
+    if not out_cols and rows:
+        w = len(rows[0]) if isinstance(rows[0], tuple) else 1
+        out_cols = ["column%d" % i for i in range(w)]  # product default
     segs = [format_csv_row(out_cols)]
     for v in rows:
         row = v if isinstance(v, tuple) else (v,)
@@ -463,7 +466,8 @@ def format_cell(v) -> bytes:
     elif isinstance(v, bool):
         s = "True" if v else "False"
     elif isinstance(v, float):
-        s = repr(v)
+        s = "%f" % v  # the reference formats csv doubles with "%f"
+                      # (PipelineBuilder.cc:1413)
     else:
         s = str(v)
     if any(c in s for c in ',"\n\r'):

@@ -434,3 +434,40 @@ def test_unique_strings_gpu():
     assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
     ref = pyoracle.run_pipeline(vals, [("unique",)])
     assert sorted(got) == sorted(ref["output"])
+
+
+def fl_tocsv_use(x):
+    return (x["a"], x["b"] / (x["a"] + 1.0), x["c"])
+
+
+def test_f64_csv_sink_gpu(tmp_path):
+    """f64 columns in the csv sink: device %f formatting, bit-identical to
+    the host/oracle "%f" (PipelineBuilder.cc:1413); out-of-range values
+    divert and come back through the host formatter."""
+    import os
+    import random
+    rng = random.Random(23)
+    lines = [b"a,b,c"]
+    for i in range(20000):
+        if i % 997 == 0:
+            b = repr(rng.uniform(1e13, 1e300)).encode()  # diverts (too big)
+        elif i % 499 == 0:
+            b = b"0.0000005"
+        else:
+            b = repr(rng.uniform(-1e6, 1e6)).encode()
+        lines.append(b"%d,%s,x%d" % (i, b, i))
+    data = b"\n".join(lines) + b"\n"
+    p = os.path.join(str(tmp_path), "f.csv")
+    with open(p, "wb") as f:
+        f.write(data)
+    outp = os.path.join(str(tmp_path), "out.csv")
+    ctx = tuplex_amd.Context()
+    ds = apply_ops(ctx.csv(p), [("map", fl_tocsv_use)])
+    ds.tocsv(outp)
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    ref = pyoracle_csv.run_csv_pipeline(data, [("map", fl_tocsv_use)],
+                                        sink="csv")
+    with open(outp, "rb") as f:
+        got = f.read()
+    assert got == ref["csv_text"]
+    assert b"." in got.split(b"\n")[1]

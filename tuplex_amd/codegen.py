@@ -1574,6 +1574,29 @@ class StageCodegen:
                 L.append("    sz += tpx_i64_digits(o.o%d);" % k)
             elif base == T.BOOL:
                 L.append("    sz += o.o%d ? 4 : 5;" % k)
+            elif base == T.F64 and not T.is_opt(t):
+                # exact %f (PipelineBuilder.cc:1413); out-of-range/nan divert
+                # to the host formatter, which prints identical text
+                L.append("    unsigned long long fN%d; bool fg%d;" % (k, k))
+                L.append("    if (tpx_f64_csv_n(o.o%d, &fN%d, &fg%d)) {"
+                         % (k, k, k))
+                L.append("      unsigned long long e = atomicAdd(exc_count,"
+                         " 1ULL);")
+                L.append("      if (e < exc_cap) {")
+                L.append("        exc_buf[e*5+0] = row0 + i;")
+                L.append("        exc_buf[e*5+1] = 7;  // NCV-style divert")
+                L.append("        exc_buf[e*5+2] = 0;")
+                if self.source == "col":
+                    L.append("        exc_buf[e*5+3] = 0;")
+                    L.append("        exc_buf[e*5+4] = 0;")
+                else:
+                    L.append("        exc_buf[e*5+3] = in_offs[i];")
+                    L.append("        exc_buf[e*5+4] = in_offs[i+1];")
+                L.append("      }")
+                L.append("      keep[i] = 0; keep01[i] = 0; sizes[i] = 0;"
+                         " continue;")
+                L.append("    }")
+                L.append("    sz += tpx_f64_csv_len(fN%d, fg%d);" % (k, k))
             else:
                 raise CodegenError("csv sink for %r not supported yet" % (t,))
         L.append("    sizes[i] = sz;")
@@ -2066,6 +2089,12 @@ class StageCodegen:
                 L.append("    { bool v = ((const long long*)outv[%d])[i] != 0;" % (3 * k))
                 L.append("      const char* s = v ? \"True\" : \"False\"; long long l = v ? 4 : 5;")
                 L.append("      for (long long j = 0; j < l; ++j) w[j] = s[j]; w += l; }")
+            elif base == T.F64 and not T.is_opt(t):
+                L.append("    { double v = ((const double*)outv[%d])[i];" % (3 * k))
+                L.append("      unsigned long long fN; bool fg;")
+                L.append("      tpx_f64_csv_n(v, &fN, &fg);  // rows out of"
+                         " range were diverted by the size pass")
+                L.append("      w = tpx_f64_csv_write(w, fN, fg); }")
             else:
                 raise CodegenError("csv sink for %r" % (t,))
         L.append("    *w++ = '\\n';")

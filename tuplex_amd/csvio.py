@@ -617,7 +617,8 @@ def _format_cell(v) -> bytes:
     elif isinstance(v, bool):
         s = "True" if v else "False"
     elif isinstance(v, float):
-        s = repr(v)
+        s = "%f" % v  # the reference formats csv doubles with "%f"
+                      # (PipelineBuilder.cc:1413)
     else:
         s = str(v)
     if any(c in s for c in ',"\n\r'):
