@@ -342,6 +342,12 @@ def run_csv_pipeline(csv_bytes: bytes, ops, columns=None, header=None,
         if pr[0] == "row":
             val = pr[1][0] if len(col_types) == 1 else pr[1]  # 1-col = scalar row
             r = pyoracle.process_row(val, row_ops, names, fast=True)
+            if sink == "csv" and r[0] == "row" and not _f64_row_fits(r[1]):
+                # device %f covers |v| < ~9.2e12; larger/nan floats divert
+                # and REPARSE through CPython on replay (fast_atod's value
+                # may differ there — documented divergence)
+                r = _replay_row(line, col_types, null_values, row_ops, names,
+                                delim.decode(), used=used_cols)
         else:
             r = _replay_row(line, col_types, null_values, row_ops, names,
                             delim.decode(), used=used_cols)
@@ -458,6 +464,42 @@ def _output_columns(row_ops, names):
                         [lp + lk + ls] +
                         [rp + c + rs for c in rcols if c != rk])
     return cols or []
+
+
+def _f64_fits_device(v):
+    """Mirror of tpx_f64_csv_n's range: True iff the device %f path formats
+    this double (N = round(v*10^6) fits 63 bits; finite)."""
+    import math
+    import struct
+    if not isinstance(v, float):
+        return True
+    if math.isnan(v) or math.isinf(v):
+        return False
+    bits = struct.unpack("<Q", struct.pack("<d", v))[0]
+    exp = (bits >> 52) & 0x7FF
+    man = bits & ((1 << 52) - 1)
+    if exp == 0:
+        m, e = man, -1074
+    else:
+        m, e = man | (1 << 52), exp - 1075
+    M = m * 15625
+    k = e + 6
+    if k >= 0:
+        return k < 62 and (M << k) < (1 << 63)
+    k = -k
+    if k >= 69:
+        return True
+    N = M >> k
+    rem = M & ((1 << k) - 1)
+    half = 1 << (k - 1)
+    if rem > half or (rem == half and (N & 1)):
+        N += 1
+    return N < (1 << 63)
+
+
+def _f64_row_fits(v):
+    row = v if isinstance(v, tuple) else (v,)
+    return all(_f64_fits_device(c) for c in row)
 
 
 def format_cell(v) -> bytes:
