@@ -241,6 +241,73 @@ def apply_ops(ds, ops):
     return ds
 
 
+def precompile_extra(verbose=False):
+    """Compile-only coverage for the stage shapes outside PIPELINES: joins
+    (unique + duplicate keys, i64 + str), columnar (ORC) source, string-key
+    aggregateByKey/unique, f64 csv sink — catches codegen regressions in the
+    no-GPU build check and warms the travelling hsaco cache."""
+    from tuplex_amd import codegen, engine, plan
+    from tuplex_amd import ttypes as T
+
+    glib = engine.GpuLib.get()
+    RIGHT = [(1, "one", 1.5), (2, "two", 2.5), (3, "three", None)]
+    DUP = [(1, "a", 10), (1, "b", None), (2, "c", 30), (7, "z", 5),
+           (1, "d", 40)]
+    SDIM = [("aa", "A1", 1), ("aa", "A2", 2), ("bb", "B1", 3)]
+    specs = [
+        ("join_i64", [T.I64, T.STR], ["key", "val"],
+         [("join", RIGHT, ["k", "label", "w"], "key", "k", "inner",
+           "", "", "", "")], "mem", "mem"),
+        ("join_left", [T.I64, T.STR], ["key", "val"],
+         [("join", RIGHT, ["k", "label", "w"], "key", "k", "left",
+           "", "", "", "")], "mem", "mem"),
+        ("join_dup_mem", [T.I64, T.STR], ["key", "val"],
+         [("join", DUP, ["k", "x", "w"], "key", "k", "inner",
+           "", "", "", "")], "mem", "mem"),
+        ("join_dup_csv", [T.STR, T.I64], ["code", "n"],
+         [("join", SDIM, ["k", "label", "r"], "code", "k", "inner",
+           "", "", "", "")], "csv", "csv"),
+        ("aggby_str", [T.STR, T.I64], ["code", "n"],
+         [("aggregateByKey", agg_combine, aggby_str_fn, 0, ["code"])],
+         "mem", "mem"),
+        ("unique_str", [T.STR], ["s"], [("unique",)], "mem", "mem"),
+        ("f64_csv", [T.I64, T.F64, T.STR], ["a", "b", "c"],
+         [("map", f64csv_map)], "csv", "csv"),
+        ("orc_col", [T.I64, ("opt", T.I64), T.STR, T.F64, T.BOOL],
+         ["a", "b", "name", "price", "active"],
+         [("map", orc_map), ("filter", orc_keep)], "col", "mem"),
+    ]
+    n = 0
+    for name, in_types, cols, ops, source, sink in specs:
+        sp = plan.build_stage(in_types, cols, ops)
+        assert sp.compilable, (name, sp.why_not_compilable)
+        kw = {"csv_info": {"null_values": [""]}} if source == "csv" else {}
+        src, desc = codegen.generate_stage(sp, source=source, sink=sink, **kw)
+        st = glib.lib.tpx_stage_compile(src.encode(), desc.encode(),
+                                        engine._CACHE_DIR.encode(), 1)
+        assert st, (name, glib.err())
+        n += 1
+        if verbose:
+            print("  extra %s ok" % name)
+    return n
+
+
+def aggby_str_fn(a, x):
+    return a + x["n"]
+
+
+def f64csv_map(x):
+    return (x["a"], x["b"] / (x["a"] + 1.0), x["c"])
+
+
+def orc_map(x):
+    return (x["a"], x["name"].upper(), x["b"], x["price"] * 2.0)
+
+
+def orc_keep(x):
+    return x[0] % 2 == 0
+
+
 def precompile_csv(verbose=False):
     """Pre-compile the CSV-source stages (zillow/q6/flights/logs) so their hsacos
     are in the travelling cache."""
