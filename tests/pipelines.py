@@ -267,6 +267,9 @@ def precompile_extra(verbose=False):
         ("join_dup_csv", [T.STR, T.I64], ["code", "n"],
          [("join", SDIM, ["k", "label", "r"], "code", "k", "inner",
            "", "", "", "")], "csv", "csv"),
+        ("join_dup_optleft", [T.STR, ("opt", T.I64)], ["code", "m"],
+         [("join", SDIM, ["k", "label", "r"], "code", "k", "left",
+           "", "", "", "")], "mem", "mem"),
         ("aggby_str", [T.STR, T.I64], ["code", "n"],
          [("aggregateByKey", agg_combine, aggby_str_fn, 0, ["code"])],
          "mem", "mem"),
