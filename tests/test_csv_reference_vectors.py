@@ -53,3 +53,22 @@ def test_manycases_shared_semantics():
         ]:
             cells, flags = split(line, b",")
             assert [c.decode() for c in cells] == expect, line
+
+
+def test_integer_float_recognition_vectors():
+    """TestStringUtils.cc:26 IntegerRecognition / :44 FloatRecognition —
+    the sniffer's recognizers restate isIntegerString/isFloatString (empty
+    strings are handled by the null-value path, not the recognizers)."""
+    from tuplex_amd import csvio
+    vec = {
+        "100-200": (False, False), "0": (True, True), "0234": (True, True),
+        "-20": (True, True), "42": (True, True), "99999999": (True, True),
+        "10.5": (False, True), ".4": (False, True),
+        "Hello world": (False, False), "10e-6": (False, True),
+        "--10": (False, False), ".4e-30": (False, True),
+        ".4E90": (False, True), "..4": (False, False),
+        "0-30": (False, False),
+    }
+    for s, (is_int, is_float) in vec.items():
+        assert (csvio._try_i64(s) is not None) == is_int, s
+        assert bool(csvio.try_f64(s)) == is_float, s
