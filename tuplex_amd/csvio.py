@@ -427,7 +427,8 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
     if not sp.compilable:
         return _run_csv_fallback(out, data, col_types, src.null_values,
                                  logical_ops, names, sink,
-                                 sp.why_not_compilable, delim.decode())
+                                 sp.why_not_compilable, delim.decode(),
+                                 out_cols=sp.output_columns)
 
     glib = GpuLib.get()
     if glib.device_count() == 0:
@@ -444,7 +445,7 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
     except codegen.CodegenError as e:
         return _run_csv_fallback(out, data, col_types, src.null_values,
                                  logical_ops, names, sink, str(e),
-                                 delim.decode())
+                                 delim.decode(), out_cols=sp.output_columns)
     stage = glib.compile_stage(csrc, desc)
 
     # chunked execution at inputSplitSize boundaries (LocalBackend.cc:552-658:
@@ -524,6 +525,15 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
     if sink_kind == "mem":
         all_rows.sort(key=lambda t: t[0])
         rows = [r for _, r in all_rows]
+        if sp.agg_expr is not None and sp.agg_unique:
+            # unique(): per-chunk tables emit (key, count); merge = key union
+            # across chunks, plus replayed rows (createFinalHashmap analog)
+            keys = dict.fromkeys(row[0] for row in rows)
+            for i in sorted(replayed):
+                for v in replayed[i]:
+                    keys[v if not isinstance(v, tuple) else v[0]] = None
+            out.rows = list(keys)
+            return out
         agg = next((op for op in logical_ops
                     if op[0] in ("aggregate", "aggregateByKey")), None)
         if sp.agg_expr is not None:
@@ -554,14 +564,10 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
             merged.setdefault(i, []).append(row[0] if len(row) == 1 else row)
         for i, lst in replayed.items():
             merged[i] = list(lst)
-        out.rows = [v for i in sorted(merged) for v in merged[i]]
-        if agg is not None:
-            from .engine import _agg_row
-            _, combine_fn, agg_fn, initial = agg
-            a = initial
-            for v in out.rows:
-                a = agg_fn(a, _agg_row(v, sp.output_columns))
-            out.rows = [a]
+        from .engine import finalize_merged
+        out.rows = finalize_merged(
+            [v for i in sorted(merged) for v in merged[i]],
+            logical_ops, sp.output_columns)
     else:
         header_line = _format_csv_row(sp.output_columns or
                                       ["column%d" % i
@@ -579,9 +585,11 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
 
 
 def _run_csv_fallback(out, data, col_types, null_values, logical_ops, names,
-                      sink, why, delim=","):
+                      sink, why, delim=",", out_cols=None):
     """Whole-stage interpreter fallback (non-compilable UDF) — the reference's
-    fallback mode. Still semantically exact; slow by design."""
+    fallback mode. Still semantically exact; slow by design. The trailing
+    aggregate/aggregateByKey/unique fold over the PIPELINE's output columns
+    (post-rename/withColumn names), not the source names."""
     out.mode = "fallback"
     out.fallback_reason = why
     rows_out = {}
@@ -597,18 +605,25 @@ def _run_csv_fallback(out, data, col_types, null_values, logical_ops, names,
             for e in r[2]:
                 nm = type(e).__name__
                 out.exception_counts[nm] = out.exception_counts.get(nm, 0) + 1
-    out.rows = [v for i in sorted(rows_out) for v in rows_out[i]]
-    agg = next((op for op in logical_ops if op[0] == "aggregate"), None)
-    if agg is not None:
-        from .engine import _agg_row
-        _, combine_fn, agg_fn, initial = agg
-        a = initial
-        for v in out.rows:
-            a = agg_fn(a, _agg_row(v, names))
-        out.rows = [a]
+    from .engine import finalize_merged, output_columns_of, _unwrap_row
+    if out_cols is None:
+        out_cols = output_columns_of(names, logical_ops)
+    merged = [v if not isinstance(v, tuple) else _unwrap_row(v)
+              for i in sorted(rows_out) for v in rows_out[i]]
+    out.rows = finalize_merged(merged, logical_ops, out_cols)
     if sink is not None and sink[0] == "csv":
-        raise NotImplementedError("tocsv via fallback: next slice")
+        header_line = _format_csv_row(out_cols or
+                                      ["column%d" % i for i in
+                                       range(max((len(_as_tup(v))
+                                                  for v in out.rows), default=1))])
+        body = b"".join(_format_csv_row(list(_as_tup(v))) for v in out.rows)
+        _write_csv_output(sink[1], header_line + body)
+        out.rows = []
     return out
+
+
+def _as_tup(v):
+    return v if isinstance(v, tuple) else (v,)
 
 
 def _format_cell(v) -> bytes:

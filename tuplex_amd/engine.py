@@ -368,23 +368,59 @@ def run_collect(data: List[Any], logical_ops: List[tuple],
                     out.exception_counts.get(name, 0) + 1
 
     merged = [v for i in sorted(results) for v in results[i]]
-    if any(op[0] == "unique" for op in logical_ops):
-        merged = list(dict.fromkeys(merged))
-    agg = next((op for op in logical_ops if op[0] == "aggregate"), None)
-    if agg is not None:
-        _, combine_fn, agg_fn, initial = agg
-        a = initial
-        for v in merged:
-            a = agg_fn(a, _agg_row(v, sp.output_columns))
-        merged = [a]
-    out.rows = merged
+    out.rows = finalize_merged(merged, logical_ops, sp.output_columns)
     return out
+
+
+def finalize_merged(merged, logical_ops, output_columns):
+    """Host-side finalization of trailing unique/aggregate/aggregateByKey over
+    interpreter-merged rows (fallback paths and non-GPU-reducible aggregates).
+    Mirrors the GPU merge the engine does for reducible aggregates
+    (LocalBackend.cc:1180-1207 combine; :2219 createFinalHashmap for by-key).
+    `output_columns` are the PIPELINE's final columns (the agg UDF sees the
+    post-rename/withColumn names, not the source names)."""
+    for op in logical_ops:
+        kind = op[0]
+        if kind == "unique":
+            merged = list(dict.fromkeys(merged))
+        elif kind == "aggregate":
+            _, combine_fn, agg_fn, initial = op
+            a = initial
+            for v in merged:
+                a = agg_fn(a, _agg_row(v, output_columns))
+            merged = [a]
+        elif kind == "aggregateByKey":
+            _, combine_fn, agg_fn, initial, key_cols = op
+            kis = [output_columns.index(c) for c in key_cols]
+            table, order = {}, []
+            for v in merged:
+                rt = v if isinstance(v, tuple) else (v,)
+                k = rt[kis[0]] if len(kis) == 1 else tuple(rt[i] for i in kis)
+                if k not in table:
+                    order.append(k)
+                table[k] = agg_fn(table.get(k, initial),
+                                  _agg_row(v, output_columns))
+            merged = [(k, table[k]) if len(kis) == 1 else k + (table[k],)
+                      for k in order]
+    return merged
+
+
+def output_columns_of(columns, logical_ops):
+    """Final column names of a pipeline without compiling its UDFs (for
+    fallback paths that never built a StageProgram): threads names through
+    build_stage with an all-str schema (column threading is type-independent)."""
+    from . import ttypes as _T
+    try:
+        n = len(columns) if columns else 1
+        return plan.build_stage([_T.STR] * n, columns, logical_ops).output_columns
+    except Exception:  # noqa: BLE001 — name threading is best-effort here
+        return list(columns) if columns else None
 
 
 def _agg_row(v, columns):
     """agg fn row arg follows the same dict convention as other UDFs."""
     if columns and isinstance(v, tuple):
-        return dict(zip(columns, v))
+        return resolve.RowView(v, columns)
     return v
 
 

@@ -82,7 +82,8 @@ def run_orc(ctx, src, logical_ops, sink=None):
     sp = plan.build_stage(col_types, names, logical_ops)
     if not sp.compilable:
         return _fallback_all(out, tab, names, logical_ops,
-                             sp.why_not_compilable)
+                             sp.why_not_compilable,
+                             out_cols=sp.output_columns)
 
     glib = GpuLib.get()
     if glib.device_count() == 0:
@@ -214,13 +215,12 @@ def run_orc(ctx, src, logical_ops, sink=None):
                 nm = type(e).__name__
                 out.exception_counts[nm] = out.exception_counts.get(nm, 0) + 1
     merged = [v for i in sorted(results) for v in results[i]]
-    if any(op[0] == "unique" for op in logical_ops):
-        merged = list(dict.fromkeys(merged))
-    out.rows = merged
+    from .engine import finalize_merged
+    out.rows = finalize_merged(merged, logical_ops, sp.output_columns)
     return out
 
 
-def _fallback_all(out, tab, names, logical_ops, why):
+def _fallback_all(out, tab, names, logical_ops, why, out_cols=None):
     out.mode = "fallback"
     out.fallback_reason = why
     scalar_input = len(names) == 1
@@ -238,15 +238,9 @@ def _fallback_all(out, tab, names, logical_ops, why):
             for e in r[2]:
                 nm = type(e).__name__
                 out.exception_counts[nm] = out.exception_counts.get(nm, 0) + 1
-    merged = [v for i in sorted(results) for v in results[i]]
-    if any(op[0] == "unique" for op in logical_ops):
-        merged = list(dict.fromkeys(merged))
-    agg = next((op for op in logical_ops if op[0] == "aggregate"), None)
-    if agg is not None:
-        _, combine_fn, agg_fn, initial = agg
-        a = initial
-        for v in merged:
-            a = agg_fn(a, _agg_row(v, names))
-        merged = [a]
-    out.rows = merged
+    from .engine import finalize_merged, output_columns_of
+    if out_cols is None:
+        out_cols = output_columns_of(names, logical_ops)
+    out.rows = finalize_merged([v for i in sorted(results)
+                                for v in results[i]], logical_ops, out_cols)
     return out

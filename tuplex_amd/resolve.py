@@ -15,12 +15,31 @@ def _n_params(fn):
     return fn.__code__.co_argcount
 
 
+class RowView(dict):
+    """Row argument for 1-param UDFs over named rows: a dict keyed by column
+    name that ALSO answers positional integer (and slice) subscripts, because
+    the compiled path maps x[0] on a named multi-column row to column 0
+    (udf/compile.py subscript rule) — host replay must accept the same access
+    patterns or diverted rows become spurious exceptions."""
+
+    __slots__ = ("_values",)
+
+    def __init__(self, values, columns):
+        super().__init__(zip(columns, values))
+        self._values = tuple(values)
+
+    def __getitem__(self, k):
+        if isinstance(k, bool) or not isinstance(k, (int, slice)):
+            return dict.__getitem__(self, k)
+        return self._values[k]
+
+
 def _call_udf(fn, row, columns):
     if isinstance(row, tuple):
         if _n_params(fn) == len(row) and _n_params(fn) > 1:
             return fn(*row)
         if columns:
-            return fn(dict(zip(columns, row)))
+            return fn(RowView(row, columns))
         return fn(row)
     return fn(row)
 
