@@ -223,3 +223,36 @@ def test_compile_zillow_stage_csv_sink():
     sp = plan.build_stage(zillow_input_types(), ZILLOW_COLS, zillow_ops())
     assert sp.compilable, sp.why_not_compilable
     _compile_only(sp, source="mem", sink="csv")
+
+
+def test_needle_window_mask_math():
+    """CPU check of _emit_scan_group's register-window compare: for random
+    8-byte windows, ((w | M) & lenmask) == P must hold iff every needle byte
+    matches under the ci rule (letter byte b matches c iff (c|0x20)==b)."""
+    import random
+    from tuplex_amd.codegen import StageCodegen
+    rng = random.Random(5)
+    pool = "abzAZB Q,:/q3"
+    for _ in range(3000):
+        nlen = rng.randint(1, 8)
+        needle = "".join(rng.choice("abz :,q") for _ in range(nlen))
+        ci = rng.random() < 0.5
+        P, M, lm, k = StageCodegen._needle_window(needle, ci)
+        win = [rng.choice(pool) for _ in range(8)]
+        if rng.random() < 0.5:  # force a (possibly case-flipped) match
+            for j, c in enumerate(needle[:8]):
+                win[j] = c.upper() if (ci and rng.random() < 0.5 and
+                                       "a" <= c <= "z") else c
+        w = 0
+        for j, c in enumerate(win):
+            w |= ord(c) << (8 * j)
+        got = ((w | M) & lm) == P
+        exp = True
+        for j, c in enumerate(needle[:k]):
+            b = ord(win[j])
+            if ci and "a" <= c <= "z":
+                if (b | 0x20) != ord(c):
+                    exp = False
+            elif b != ord(c):
+                exp = False
+        assert got == exp, (needle, ci, win, hex(P), hex(M))

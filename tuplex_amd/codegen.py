@@ -816,11 +816,33 @@ class StageCodegen:
         return self._emit_scan_group(em, inner, members, rc, opid,
                                      (node["op"], needle, ci))
 
+    @staticmethod
+    def _needle_window(needle: str, ci: bool):
+        """(P, M, lenmask, k) for the register-window compare of the needle's
+        first k=min(8,len) bytes: match at byte offset o of window _w iff
+        ((_w | M) & lenmask) == P. ci letter bytes get 0x20 in M and the
+        lowercase byte in P (exactly {b, toupper(b)} for ASCII); everything
+        else compares exact."""
+        nb = needle.encode()
+        k = min(len(nb), 8)
+        P = M = 0
+        for j in range(k):
+            c = nb[j]
+            if ci and 0x61 <= c <= 0x7A:  # 'a'..'z'
+                M |= 0x20 << (8 * j)
+            P |= c << (8 * j)
+        lenmask = (1 << (8 * k)) - 1 if k < 8 else (1 << 64) - 1
+        return P & lenmask, M, lenmask, k
+
     def _emit_scan_group(self, em, inner, members, rc, opid, want):
         """ONE SWAR window scan over `inner` answering every (op, needle, ci)
         in `members`; memoizes each under its scan key; returns `want`'s var.
-        ci member byte compare: needle is all non-uppercase; a letter byte b
-        matches c iff (c | 0x20) == b (exactly {b, toupper(b)} for ASCII);
+        The main loop takes 32 B per iteration with FOUR INDEPENDENT 8-B loads
+        (dependent 8-B chains over LDS pay ~50 cycles per step); candidate
+        positions (first-char byte matches) are verified against the needle via
+        a register window built from the already-loaded words — no per-
+        candidate byte loads for needles <= 8 B. ci member compare: needle is
+        all non-uppercase; a letter byte b matches c iff (c | 0x20) == b;
         non-letter bytes compare exact."""
         hv, hn = self.emit_expr(em, inner, rc, opid)
         tag = em.fresh("fg")
@@ -829,12 +851,15 @@ class StageCodegen:
                 em.w("long long %s_%d = -1;" % (tag, mi))
             else:
                 em.w("bool %s_%d = false;" % (tag, mi))
-        fcs = set()
+        # distinct first-char detectors: ("ci", lower_byte) uses the
+        # bit5-insensitive compare (may over-match punctuation pairs; the
+        # window verification rejects those)
+        dets = []
         for mop, needle, ci in members:
-            fcs.add(needle[0])
-            if ci and "a" <= needle[0] <= "z":
-                fcs.add(needle[0].upper())
-        fcs = sorted(fcs)
+            c0 = needle[0]
+            d = ("ci", ord(c0)) if (ci and "a" <= c0 <= "z") else ("ex", ord(c0))
+            if d not in dets:
+                dets.append(d)
         em.w("{  // fused scan over %s (%d needles)" % (hv, len(members)))
         em.w("  const char* _p = %s.p; long long _n = %s.n;" % (hv, hv))
         em.w("  unsigned _pend = %du;" % ((1 << len(members)) - 1))
@@ -860,16 +885,61 @@ class StageCodegen:
                            1 << mi))
                 em.w("%sif (%s) { %s }" % (indent, " && ".join(cond), body))
 
+        def det_expr(v, d):
+            kind, b = d
+            if kind == "ci":
+                return ("tpx_swar_zero((%s | (TPX_SWAR_ONE * 0x20u)) ^"
+                        " (TPX_SWAR_ONE * %du))" % (v, b))
+            return "tpx_swar_zero(%s ^ (TPX_SWAR_ONE * %du))" % (v, b)
+
         em.w("  while (_i < _n && _pend &&"
              " (((unsigned long long)(_p + _i)) & 7)) {")
         em.w("    int _ch = (unsigned char)_p[_i];")
         checks("_i", "    ")
         em.w("    ++_i;")
         em.w("  }")
+        em.w("  for (; _i + 32 <= _n && _pend; _i += 32) {")
+        for k in range(5):
+            # _v4 (one word past the 32-B window) feeds candidate windows at
+            # the tail of _v3; buffers carry >=16 B tail slack so the 8-B
+            # overread at _i+32 <= _n stays in bounds
+            em.w("    unsigned long long _v%d ="
+                 " *(const unsigned long long*)(_p + _i + %d);" % (k, 8 * k))
+        for k in range(4):
+            em.w("    unsigned long long _c%d = %s;"
+                 % (k, " | ".join(det_expr("_v%d" % k, d) for d in dets)))
+        em.w("    if (_c0 | _c1 | _c2 | _c3) {")
+        for k in range(4):
+            em.w("      while (_c%d && _pend) {" % k)
+            em.w("        int _o = (__ffsll((long long)_c%d) - 1) >> 3;"
+                 " _c%d &= _c%d - 1;" % (k, k, k))
+            em.w("        long long _j = _i + %d + _o;" % (8 * k))
+            em.w("        int _sh = _o << 3;")
+            em.w("        unsigned long long _w = _sh ? ((_v%d >> _sh) |"
+                 " (_v%d << (64 - _sh))) : _v%d;" % (k, k + 1, k))
+            for mi, (mop, needle, ci) in enumerate(members):
+                nb = needle.encode()
+                P, M, lm, wk = self._needle_window(needle, ci)
+                cond = ["(_pend & %du)" % (1 << mi),
+                        "_j + %d <= _n" % len(nb)]
+                if M:
+                    cond.append("((_w | 0x%xULL) & 0x%xULL) == 0x%xULL"
+                                % (M, lm, P | M))
+                else:
+                    cond.append("(_w & 0x%xULL) == 0x%xULL" % (lm, P))
+                for j in range(8, len(nb)):  # >8-B needles: byte-verify tail
+                    cond.append(bcmp("(unsigned char)_p[_j + %d]" % j,
+                                     needle[j], ci))
+                body = ("%s_%d = %s; _pend &= ~%du;"
+                        % (tag, mi, "_j" if mop == "strfind" else "true",
+                           1 << mi))
+                em.w("        if (%s) { %s }" % (" && ".join(cond), body))
+            em.w("      }")
+        em.w("    }")
+        em.w("  }")
         em.w("  for (; _i + 8 <= _n && _pend; _i += 8) {")
         em.w("    unsigned long long _v = *(const unsigned long long*)(_p + _i);")
-        hits = " | ".join("tpx_swar_zero(_v ^ (TPX_SWAR_ONE * %du))" % ord(c)
-                          for c in fcs)
+        hits = " | ".join(det_expr("_v", d) for d in dets)
         em.w("    unsigned long long _hit = %s;" % hits)
         em.w("    if (_hit) {")
         em.w("      for (int _b = 0; _b < 8; ++_b) {")
@@ -1019,6 +1089,126 @@ class StageCodegen:
                 rc = [(v, sp.agg_expr["t"], nv)]
         return rc
 
+    # ---- quote-freedom provenance (csv sink) -------------------------------
+    # An output string cell needs the RFC-4180 quote scan only if it may
+    # contain ',', '"', CR or LF. Values derived from an input CSV cell whose
+    # walk saw none of those (tpx_cell flags bit 8 clear) provably need no
+    # quoting, so the size pass can skip its per-string SWAR scan entirely —
+    # the reference's generated CSV writer pays the same scan per cell
+    # (PipelineBuilder buildWithCSVRowWriter), so this is a pure GPU-side win.
+    # Lattice: True (never needs quoting) / False (unknown -> scan) /
+    # frozenset of source cell indices (clean iff all those cells were clean).
+
+    @staticmethod
+    def _qfree_and(a, b):
+        if a is False or b is False:
+            return False
+        if a is True:
+            return b
+        if b is True:
+            return a
+        return a | b
+
+    @staticmethod
+    def _qfree_static_str(s):
+        return not any(c in s for c in ',"\r\n')
+
+    def _qfree_node(self, n, env):
+        op = n["op"]
+        a = n["args"]
+        if op == "input":
+            return env[n["i"]]
+        if op == "const":
+            v = n.get("v")
+            if isinstance(v, str):
+                return self._qfree_static_str(v)
+            return True  # numeric/bool/None render as digit/letter literals
+        if op in ("slice", "getitem", "strip", "lower", "upper", "swapcase",
+                  "splitget"):
+            return self._qfree_node(a[0], env)
+        if op == "concat":
+            return self._qfree_and(self._qfree_node(a[0], env),
+                                   self._qfree_node(a[1], env))
+        if op == "replace":
+            rep = a[2]
+            rep_ok = (rep["op"] == "const" and isinstance(rep.get("v"), str)
+                      and self._qfree_static_str(rep["v"]))
+            return self._qfree_and(self._qfree_node(a[0], env),
+                                   True if rep_ok else False)
+        if op == "to_str":
+            from . import ttypes as _T
+            if _T.deopt(a[0]["t"]) == _T.STR:
+                return self._qfree_node(a[0], env)
+            return True  # int/bool/None texts are clean
+        if op == "fmt_int":
+            return True
+        if op == "if":
+            return self._qfree_and(self._qfree_node(a[1], env),
+                                   self._qfree_node(a[2], env))
+        return False
+
+    def _thread_qfree(self):
+        """Per-output-column qfree lattice (csv sink; None when untracked)."""
+        sp = self.sp
+        if getattr(sp, "agg_expr", None) is not None or \
+                self._dup_join_op() is not None:
+            return None
+        env = []
+        for i, t in enumerate(sp.input_types):
+            if T.deopt(t) == T.STR:
+                env.append(frozenset([i]) if self.source == "csv"
+                           and not self.csv_info.get("text_mode") else False)
+            else:
+                env.append(True)
+        for op in sp.ops:
+            if op.kind == "map":
+                if op.tir["op"] == "mktuple":
+                    env = [self._qfree_node(el, env) for el in op.tir["args"]]
+                else:
+                    env = [self._qfree_node(op.tir, env)]
+            elif op.kind == "filter":
+                pass
+            elif op.kind == "withColumn":
+                v = self._qfree_node(op.tir, env)
+                cols = op.in_columns or ["column%d" % i
+                                         for i in range(len(env))]
+                if op.col in cols:
+                    i = cols.index(op.col)
+                    env = env[:i] + [v] + env[i + 1:]
+                else:
+                    env = env + [v]
+            elif op.kind == "mapColumn":
+                i = op.in_columns.index(op.col)
+                env = env[:i] + [self._qfree_node(op.tir, [env[i]])] + \
+                    env[i + 1:]
+            elif op.kind == "selectColumns":
+                env = [env[i] for i in op.sel_idxs]
+            elif op.kind == "renameColumn":
+                pass
+            elif op.kind == "join":
+                rrows, rcols, lki, rki, how = op.join
+                right = []
+                for j in range(len(rcols)):
+                    if j == rki:
+                        continue
+                    right.append(all(not isinstance(r[j], str)
+                                     or self._qfree_static_str(r[j])
+                                     for r in rrows))
+                env = [env[i] for i in range(len(env)) if i != lki] + \
+                    [env[lki]] + right
+            else:
+                return None
+        return env
+
+    def _qfree_cond(self, q):
+        """C condition that the value is quote-free, or None (must scan)."""
+        if q is True:
+            return "true"
+        if q is False or q is None:
+            return None
+        # flags bit0 quoted | bit3(8) content-special: both clear == clean
+        return " && ".join("((cl%d.flags & 9) == 0)" % i for i in sorted(q))
+
     # ---- full source --------------------------------------------------------
     def _dup_join_op(self):
         """The terminal duplicate-key join op, if this stage has one."""
@@ -1122,7 +1312,11 @@ class StageCodegen:
     # rows through a thrashed L1/L2 (measured 7x HBM read amplification:
     # 2048 threads x ~200 B rows >> 32 KiB L1 per CU). Waves whose span exceeds
     # TPX_SPAN_CAP fall back to parsing from global memory (rare: long rows).
-    SPAN_CAP = 16384  # bytes per wave; 4 waves/block -> 64 KiB LDS, 2 blocks/CU
+    # bytes per wave: 2 waves/block stage 2*CAP+16 B of LDS; 16368 (not 16384)
+    # keeps the block at 32752 B so FIVE blocks fit the 160 KiB CU budget
+    # (5 x 32768 rounded = 163840) -> 10 waves/CU; at 16384 the +16 tail pushes
+    # the block over 32 KiB and occupancy drops to 4 blocks = 8 waves/CU
+    SPAN_CAP = 16368
 
     def _main_kernel(self, in_types, out_types):
         if self.source == "col":
@@ -1555,19 +1749,33 @@ class StageCodegen:
         fast path for the (common) fully-unquoted rows, via keep[i] bit 1."""
         L = ["    long long sz = %d;  // delimiters + newline" % len(out_types)]
         L.append("    bool _anyq = false;")
+        qfree = self._thread_qfree()
         for k, t in enumerate(out_types):
             base = T.deopt(t)
             if base == T.STR:
+                qc = self._qfree_cond(qfree[k]) if qfree is not None else None
                 pre = ""
                 if T.is_opt(t):
                     L.append("    if (!o.o%d_n) {" % k)
                     pre = "  "
-                L.append(pre + "    { long long q%d;" % k)
-                L.append(pre + "      bool nq%d = tpx_csv_needs_quote(o.o%d,"
-                         " &q%d);" % (k, k, k))
-                L.append(pre + "      sz += nq%d ? o.o%d.n + q%d + 2 : o.o%d.n;"
-                         % (k, k, k, k))
-                L.append(pre + "      _anyq |= nq%d; }" % k)
+                if qc == "true":
+                    # provenance: value can never need quoting -> no scan
+                    L.append(pre + "    sz += o.o%d.n;" % k)
+                elif qc is not None:
+                    L.append(pre + "    if (%s) { sz += o.o%d.n; }" % (qc, k))
+                    L.append(pre + "    else { long long q%d;" % k)
+                    L.append(pre + "      bool nq%d = tpx_csv_needs_quote(o.o%d,"
+                             " &q%d);" % (k, k, k))
+                    L.append(pre + "      sz += nq%d ? o.o%d.n + q%d + 2 : o.o%d.n;"
+                             % (k, k, k, k))
+                    L.append(pre + "      _anyq |= nq%d; }" % k)
+                else:
+                    L.append(pre + "    { long long q%d;" % k)
+                    L.append(pre + "      bool nq%d = tpx_csv_needs_quote(o.o%d,"
+                             " &q%d);" % (k, k, k))
+                    L.append(pre + "      sz += nq%d ? o.o%d.n + q%d + 2 : o.o%d.n;"
+                             % (k, k, k, k))
+                    L.append(pre + "      _anyq |= nq%d; }" % k)
                 if T.is_opt(t):
                     L.append("    }")
             elif base == T.I64:

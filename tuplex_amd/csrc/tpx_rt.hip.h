@@ -90,8 +90,11 @@ __device__ __forceinline__ unsigned long long tpx_swar_zero(unsigned long long x
     return (x - TPX_SWAR_ONE) & ~x & TPX_SWAR_HIGH;
 }
 
-// first index of byte c in p[0..n), or -1. Aligned 8-byte windows (head/tail
-// byte loops), exact match of the byte-loop semantics.
+// first index of byte c in p[0..n), or -1. Head byte loop to 8-alignment,
+// then 32-B iterations with FOUR INDEPENDENT 8-B loads (the scans run over
+// LDS/L1 where a dependent 8-B chain pays ~50 cycles per step; 4 loads in
+// flight cut the chain depth 4x), then an 8-B + byte tail. Semantics exactly
+// match the byte loop.
 __device__ __forceinline__ long long tpx_memchr(const char* p, long long n,
                                                 char c) {
     unsigned long long pat = TPX_SWAR_ONE * (unsigned char)c;
@@ -99,6 +102,22 @@ __device__ __forceinline__ long long tpx_memchr(const char* p, long long n,
     while (i < n && (((unsigned long long)(p + i)) & 7)) {
         if (p[i] == c) return i;
         ++i;
+    }
+    for (; i + 32 <= n; i += 32) {
+        unsigned long long v0 = *(const unsigned long long*)(p + i);
+        unsigned long long v1 = *(const unsigned long long*)(p + i + 8);
+        unsigned long long v2 = *(const unsigned long long*)(p + i + 16);
+        unsigned long long v3 = *(const unsigned long long*)(p + i + 24);
+        unsigned long long h0 = tpx_swar_zero(v0 ^ pat);
+        unsigned long long h1 = tpx_swar_zero(v1 ^ pat);
+        unsigned long long h2 = tpx_swar_zero(v2 ^ pat);
+        unsigned long long h3 = tpx_swar_zero(v3 ^ pat);
+        if (h0 | h1 | h2 | h3) {
+            if (h0) return i + (__ffsll((long long)h0) - 1) / 8;
+            if (h1) return i + 8 + (__ffsll((long long)h1) - 1) / 8;
+            if (h2) return i + 16 + (__ffsll((long long)h2) - 1) / 8;
+            return i + 24 + (__ffsll((long long)h3) - 1) / 8;
+        }
     }
     for (; i + 8 <= n; i += 8) {
         unsigned long long v = *(const unsigned long long*)(p + i);
@@ -122,6 +141,30 @@ __device__ __forceinline__ long long tpx_memchr_hi(const char* p, long long n,
         *hi |= (unsigned char)p[i] & 0x80u;
         if (p[i] == c) return i;
         ++i;
+    }
+    for (; i + 32 <= n; i += 32) {
+        unsigned long long v0 = *(const unsigned long long*)(p + i);
+        unsigned long long v1 = *(const unsigned long long*)(p + i + 8);
+        unsigned long long v2 = *(const unsigned long long*)(p + i + 16);
+        unsigned long long v3 = *(const unsigned long long*)(p + i + 24);
+        unsigned long long h0 = tpx_swar_zero(v0 ^ pat);
+        unsigned long long h1 = tpx_swar_zero(v1 ^ pat);
+        unsigned long long h2 = tpx_swar_zero(v2 ^ pat);
+        unsigned long long h3 = tpx_swar_zero(v3 ^ pat);
+        if (h0 | h1 | h2 | h3) {
+            // hi must cover exactly the bytes scanned up to the hit
+            if (h0) { *hi |= v0 & TPX_SWAR_HIGH;
+                      return i + (__ffsll((long long)h0) - 1) / 8; }
+            *hi |= v0 & TPX_SWAR_HIGH;
+            if (h1) { *hi |= v1 & TPX_SWAR_HIGH;
+                      return i + 8 + (__ffsll((long long)h1) - 1) / 8; }
+            *hi |= v1 & TPX_SWAR_HIGH;
+            if (h2) { *hi |= v2 & TPX_SWAR_HIGH;
+                      return i + 16 + (__ffsll((long long)h2) - 1) / 8; }
+            *hi |= (v2 | v3) & TPX_SWAR_HIGH;
+            return i + 24 + (__ffsll((long long)h3) - 1) / 8;
+        }
+        *hi |= (v0 | v1 | v2 | v3) & TPX_SWAR_HIGH;
     }
     for (; i + 8 <= n; i += 8) {
         unsigned long long v = *(const unsigned long long*)(p + i);
@@ -156,7 +199,8 @@ __device__ __forceinline__ long long tpx_memchr2(const char* p, long long n,
     return -1;
 }
 
-// last index of byte c in p[0..n), or -1 (backward SWAR)
+// last index of byte c in p[0..n), or -1 (backward SWAR; 32-B iterations with
+// four independent loads, same rationale as tpx_memchr)
 __device__ __forceinline__ long long tpx_memrchr(const char* p, long long n,
                                                  char c) {
     unsigned long long pat = TPX_SWAR_ONE * (unsigned char)c;
@@ -164,6 +208,23 @@ __device__ __forceinline__ long long tpx_memrchr(const char* p, long long n,
     while (i > 0 && (((unsigned long long)(p + i)) & 7)) {
         --i;
         if (p[i] == c) return i;
+    }
+    while (i >= 32) {
+        i -= 32;
+        unsigned long long v0 = *(const unsigned long long*)(p + i);
+        unsigned long long v1 = *(const unsigned long long*)(p + i + 8);
+        unsigned long long v2 = *(const unsigned long long*)(p + i + 16);
+        unsigned long long v3 = *(const unsigned long long*)(p + i + 24);
+        unsigned long long h0 = tpx_swar_zero(v0 ^ pat);
+        unsigned long long h1 = tpx_swar_zero(v1 ^ pat);
+        unsigned long long h2 = tpx_swar_zero(v2 ^ pat);
+        unsigned long long h3 = tpx_swar_zero(v3 ^ pat);
+        if (h0 | h1 | h2 | h3) {
+            if (h3) return i + 24 + (63 - __builtin_clzll(h3)) / 8;
+            if (h2) return i + 16 + (63 - __builtin_clzll(h2)) / 8;
+            if (h1) return i + 8 + (63 - __builtin_clzll(h1)) / 8;
+            return i + (63 - __builtin_clzll(h0)) / 8;
+        }
     }
     while (i >= 8) {
         i -= 8;
@@ -186,6 +247,13 @@ __device__ __forceinline__ bool tpx_ascii(const tstr s) {
     while (i < n && (((unsigned long long)(p + i)) & 7)) {
         if ((unsigned char)p[i] >= 0x80) return false;
         ++i;
+    }
+    for (; i + 32 <= n; i += 32) {
+        unsigned long long v0 = *(const unsigned long long*)(p + i);
+        unsigned long long v1 = *(const unsigned long long*)(p + i + 8);
+        unsigned long long v2 = *(const unsigned long long*)(p + i + 16);
+        unsigned long long v3 = *(const unsigned long long*)(p + i + 24);
+        if ((v0 | v1 | v2 | v3) & TPX_SWAR_HIGH) return false;
     }
     for (; i + 8 <= n; i += 8)
         if (*(const unsigned long long*)(p + i) & TPX_SWAR_HIGH) return false;
@@ -702,6 +770,32 @@ __device__ __forceinline__ bool tpx_csv_needs_quote(const tstr s, long long* nqu
         else if (c == ',' || c == '\n' || c == '\r') need = true;
         ++i;
     }
+    for (; i + 32 <= s.n; i += 32) {  // 4 independent loads per iteration
+        unsigned long long v0 = *(const unsigned long long*)(s.p + i);
+        unsigned long long v1 = *(const unsigned long long*)(s.p + i + 8);
+        unsigned long long v2 = *(const unsigned long long*)(s.p + i + 16);
+        unsigned long long v3 = *(const unsigned long long*)(s.p + i + 24);
+        unsigned long long hq =
+            tpx_swar_zero(v0 ^ (TPX_SWAR_ONE * (unsigned long long)'"')) |
+            tpx_swar_zero(v1 ^ (TPX_SWAR_ONE * (unsigned long long)'"')) >> 1 |
+            tpx_swar_zero(v2 ^ (TPX_SWAR_ONE * (unsigned long long)'"')) >> 2 |
+            tpx_swar_zero(v3 ^ (TPX_SWAR_ONE * (unsigned long long)'"')) >> 3;
+        unsigned long long ho = 0;
+        ho |= tpx_swar_zero(v0 ^ (TPX_SWAR_ONE * (unsigned long long)',')) |
+              tpx_swar_zero(v0 ^ (TPX_SWAR_ONE * (unsigned long long)'\n')) |
+              tpx_swar_zero(v0 ^ (TPX_SWAR_ONE * (unsigned long long)'\r'));
+        ho |= tpx_swar_zero(v1 ^ (TPX_SWAR_ONE * (unsigned long long)',')) |
+              tpx_swar_zero(v1 ^ (TPX_SWAR_ONE * (unsigned long long)'\n')) |
+              tpx_swar_zero(v1 ^ (TPX_SWAR_ONE * (unsigned long long)'\r'));
+        ho |= tpx_swar_zero(v2 ^ (TPX_SWAR_ONE * (unsigned long long)',')) |
+              tpx_swar_zero(v2 ^ (TPX_SWAR_ONE * (unsigned long long)'\n')) |
+              tpx_swar_zero(v2 ^ (TPX_SWAR_ONE * (unsigned long long)'\r'));
+        ho |= tpx_swar_zero(v3 ^ (TPX_SWAR_ONE * (unsigned long long)',')) |
+              tpx_swar_zero(v3 ^ (TPX_SWAR_ONE * (unsigned long long)'\n')) |
+              tpx_swar_zero(v3 ^ (TPX_SWAR_ONE * (unsigned long long)'\r'));
+        if (hq) { q += __builtin_popcountll(hq); need = true; }
+        if (ho) need = true;
+    }
     for (; i + 8 <= s.n; i += 8) {
         unsigned long long v = *(const unsigned long long*)(s.p + i);
         unsigned long long hq =
@@ -968,8 +1062,89 @@ extern "C" __global__ void tpx_csv_emit_rows(const char* __restrict__ data,
 // one CSV cell starting at p; returns pointer to next cell start. *more = a
 // delimiter was consumed (another cell follows). flags: 1 quoted, 2 contains ""
 // escapes (diverted to host), 4 structurally bad (unterminated quote / junk after
-// closing quote — reference DOUBLEQUOTEERROR class)
+// closing quote — reference DOUBLEQUOTEERROR class), 8 content may need CSV-
+// OUTPUT quoting (quoted cells always; unquoted cells iff a ','/'"'/'\r' byte
+// was seen in the content — lets the csv-sink size pass skip its quote scan
+// for values provably derived from clean input cells, codegen._thread_qfree)
 struct tpx_cell { const char* p; long long n; int flags; };
+
+// tpx_memchr_hi that additionally reports whether any output-special byte
+// ('"', '\r', and ',' when chk_comma — needed only when the source delimiter
+// is not ',') occurs among the scanned bytes BEFORE the hit. Word-granular:
+// in the hit word only bytes below the hit position count.
+__device__ __forceinline__ long long tpx_memchr_hi_spec(
+        const char* p, long long n, char c, unsigned long long* hi,
+        int chk_comma, int* spec) {
+    unsigned long long pat = TPX_SWAR_ONE * (unsigned char)c;
+    unsigned long long sp_acc = 0;
+    long long i = 0;
+    #define TPX_SPECW(v) \
+        (tpx_swar_zero((v) ^ (TPX_SWAR_ONE * (unsigned long long)'"')) | \
+         tpx_swar_zero((v) ^ (TPX_SWAR_ONE * (unsigned long long)'\r')) | \
+         (chk_comma ? tpx_swar_zero((v) ^ (TPX_SWAR_ONE * (unsigned long long)',')) : 0ULL))
+    #define TPX_SPECB(ch) \
+        ((ch) == '"' || (ch) == '\r' || (chk_comma && (ch) == ','))
+    while (i < n && (((unsigned long long)(p + i)) & 7)) {
+        unsigned char ch = p[i];
+        *hi |= ch & 0x80u;
+        if ((char)ch == c) { if (sp_acc) *spec = 1; return i; }
+        if (TPX_SPECB(ch)) sp_acc = 1;
+        ++i;
+    }
+    for (; i + 32 <= n; i += 32) {
+        unsigned long long v0 = *(const unsigned long long*)(p + i);
+        unsigned long long v1 = *(const unsigned long long*)(p + i + 8);
+        unsigned long long v2 = *(const unsigned long long*)(p + i + 16);
+        unsigned long long v3 = *(const unsigned long long*)(p + i + 24);
+        unsigned long long h0 = tpx_swar_zero(v0 ^ pat);
+        unsigned long long h1 = tpx_swar_zero(v1 ^ pat);
+        unsigned long long h2 = tpx_swar_zero(v2 ^ pat);
+        unsigned long long h3 = tpx_swar_zero(v3 ^ pat);
+        if (h0 | h1 | h2 | h3) {
+            unsigned long long hw, sw;
+            long long base;
+            if (h0) { hw = h0; sw = TPX_SPECW(v0); base = i;
+                      *hi |= v0 & TPX_SWAR_HIGH; }
+            else if (h1) { hw = h1; sw = TPX_SPECW(v1); base = i + 8;
+                           sp_acc |= TPX_SPECW(v0);
+                           *hi |= (v0 | v1) & TPX_SWAR_HIGH; }
+            else if (h2) { hw = h2; sw = TPX_SPECW(v2); base = i + 16;
+                           sp_acc |= TPX_SPECW(v0) | TPX_SPECW(v1);
+                           *hi |= (v0 | v1 | v2) & TPX_SWAR_HIGH; }
+            else { hw = h3; sw = TPX_SPECW(v3); base = i + 24;
+                   sp_acc |= TPX_SPECW(v0) | TPX_SPECW(v1) | TPX_SPECW(v2);
+                   *hi |= (v0 | v1 | v2 | v3) & TPX_SWAR_HIGH; }
+            long long b = (__ffsll((long long)hw) - 1) / 8;
+            sp_acc |= sw & ((b ? (1ULL << (8 * b)) : 1ULL) - 1);
+            if (sp_acc) *spec = 1;
+            return base + b;
+        }
+        sp_acc |= TPX_SPECW(v0) | TPX_SPECW(v1) | TPX_SPECW(v2) | TPX_SPECW(v3);
+        *hi |= (v0 | v1 | v2 | v3) & TPX_SWAR_HIGH;
+    }
+    for (; i + 8 <= n; i += 8) {
+        unsigned long long v = *(const unsigned long long*)(p + i);
+        *hi |= v & TPX_SWAR_HIGH;
+        unsigned long long hit = tpx_swar_zero(v ^ pat);
+        if (hit) {
+            long long b = (__ffsll((long long)hit) - 1) / 8;
+            sp_acc |= TPX_SPECW(v) & ((b ? (1ULL << (8 * b)) : 1ULL) - 1);
+            if (sp_acc) *spec = 1;
+            return i + b;
+        }
+        sp_acc |= TPX_SPECW(v);
+    }
+    for (; i < n; ++i) {
+        unsigned char ch = p[i];
+        *hi |= ch & 0x80u;
+        if ((char)ch == c) { if (sp_acc) *spec = 1; return i; }
+        if (TPX_SPECB(ch)) sp_acc = 1;
+    }
+    if (sp_acc) *spec = 1;
+    return -1;
+    #undef TPX_SPECW
+    #undef TPX_SPECB
+}
 
 // hi: accumulates non-ASCII high bits of every scanned byte (row ASCII gate
 // fused into the walk — delimiters/quotes themselves are always ASCII)
@@ -992,7 +1167,8 @@ __device__ __forceinline__ const char* tpx_csv_next_cell(const char* p,
             break;
         }
         if (q >= end) { c->p = p; c->n = end - p; c->flags = 4; return end; }
-        c->p = s; c->n = q - s; c->flags = 1 | (esc ? 2 : 0);
+        // bit 8: quoted content was quoted for a reason (or we don't track it)
+        c->p = s; c->n = q - s; c->flags = 1 | (esc ? 2 : 0) | 8;
         ++q;
         if (q < end && *q != delim) c->flags |= 4;
         long long kd = tpx_memchr_hi(q, end - q, delim, hi);
@@ -1000,9 +1176,12 @@ __device__ __forceinline__ const char* tpx_csv_next_cell(const char* p,
         if (q < end) { *more = true; ++q; }
         return q;
     }
-    long long kd = tpx_memchr_hi(p, end - p, delim, hi);
+    int spec = 0;
+    long long kd = tpx_memchr_hi_spec(p, end - p, delim, hi,
+                                      delim != ',', &spec);
     const char* q = kd < 0 ? end : p + kd;
     c->p = p; c->n = q - p;
+    if (spec) c->flags |= 8;
     if (q < end) { *more = true; ++q; }
     return q;
 }
