@@ -60,7 +60,9 @@ def build_stage():
 
 def cpu_baseline_leg(body, seconds):
     """Time the oracle C port (kind 'port') on the host cores of this box —
-    bounded sample, single thread (oracle/czillow.c)."""
+    bounded sample. Primary number = ALL host cores (the reference runs its
+    compiled path at executorCount = hw threads, LocalBackend.cc:46); a
+    single-core leg is reported alongside as value_1core."""
     binp = os.path.join(HERE, "oracle", "czillow")
     if not os.path.exists(binp):
         try:
@@ -76,16 +78,67 @@ def cpu_baseline_leg(body, seconds):
     with open(tmp, "wb") as f:
         f.write(b"h1,h2,h3,h4,h5,h6,h7,h8,h9,h10\n")  # czillow skips line 1
         f.write(sample)
+    ncores = os.cpu_count() or 1
     try:
-        out = subprocess.check_output([binp, tmp, str(seconds)], timeout=seconds * 4 + 60)
+        out = subprocess.check_output([binp, tmp, str(seconds), str(ncores)],
+                                      timeout=seconds * 4 + 60)
         r = json.loads(out)
-        return {"value": r["rows_per_s"], "unit": "rows/s", "cores": 1,
+        out1 = subprocess.check_output([binp, tmp, str(max(seconds / 2, 2)),
+                                        "1"], timeout=seconds * 4 + 60)
+        r1 = json.loads(out1)
+        return {"value": r["rows_per_s"], "unit": "rows/s", "cores": ncores,
                 "kind": "port",
-                "sample": "%d MB of the same synthetic Zillow CSV, %d s single-thread"
-                          % (len(sample) >> 20, seconds)}
+                "value_1core": r1["rows_per_s"],
+                "sample": "%d MB of the same synthetic Zillow CSV; %d s on all "
+                          "%d host cores (pthread range shards), plus a "
+                          "single-core leg" % (len(sample) >> 20, seconds,
+                                               ncores)}
     except Exception as e:  # noqa: BLE001
         log("cpu_baseline failed:", e)
         return None
+
+
+def e2e_leg(header, body, mb, local_rank=0):
+    """End-to-end file->file tocsv: disk read + H2D + kernels + D2H + disk
+    write ALL inside the timed region (SURVEY.md §8d's IO-included number),
+    through the product engine (Context.csv -> tocsv), reported beside the
+    resident-kernel headline in the same JSON line."""
+    import shutil
+    import tuplex_amd
+    from tests.test_codegen_compile import zillow_ops
+    from tests.pipelines import apply_ops
+    tdir = "/tmp/tpx_e2e"
+    os.makedirs(tdir, exist_ok=True)
+    inp = os.path.join(tdir, "in.csv")
+    outp = os.path.join(tdir, "out.csv")
+    want = mb << 20
+    try:
+        with open(inp, "wb") as f:
+            f.write(header)
+            written = 0
+            while written < want:
+                f.write(body)
+                written += len(body)
+        rows = 0
+        t0 = time.perf_counter()
+        ctx = tuplex_amd.Context({"tuplex.gpu.device": str(local_rank)})
+        ds = apply_ops(ctx.csv(inp), zillow_ops())
+        ds.tocsv(outp)
+        t1 = time.perf_counter()
+        if ds._last_outcome.mode != "gpu":
+            log("e2e leg fell back:", ds._last_outcome.fallback_reason)
+            return None
+        with open(inp, "rb") as f:
+            f.seek(len(header))
+            rows = sum(chunk.count(b"\n")
+                       for chunk in iter(lambda: f.read(1 << 24), b""))
+        return {"e2e_rows_per_s": rows / (t1 - t0), "e2e_mb": written >> 20,
+                "e2e_seconds": t1 - t0}
+    except Exception as e:  # noqa: BLE001
+        log("e2e leg failed:", e)
+        return None
+    finally:
+        shutil.rmtree(tdir, ignore_errors=True)
 
 
 def parse_pmc_dbs(paths):
@@ -140,6 +193,9 @@ def main():
                     help="rocprofv3 output dirs (FETCH_SIZE / WRITE_SIZE passes "
                          "at THIS workload config) for roofline.traffic")
     ap.add_argument("--no-cpu-baseline", action="store_true")
+    ap.add_argument("--no-e2e", action="store_true")
+    ap.add_argument("--e2e-mb", type=int, default=None,
+                    help="end-to-end file->file leg size (default: mb-per-gpu)")
     args = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -260,6 +316,10 @@ def main():
     cpu = None
     if not args.no_cpu_baseline and rank == 0 and world <= 1:
         cpu = cpu_baseline_leg(body, args.cpu_seconds)
+    e2e = None
+    if not args.no_e2e and rank == 0 and world <= 1:
+        e2e = e2e_leg(header, body, args.e2e_mb or args.mb_per_gpu,
+                      local_rank % ndev)
 
     line = {
         "metric": "rows/s",
@@ -285,6 +345,7 @@ def main():
             "dirty_frac": args.dirty,
             "sink": "csv-device-resident",
             "upload_s": t_upload,
+            **(e2e or {}),
         },
         "roofline": roofline,
         "cpu_baseline": cpu,

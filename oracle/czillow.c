@@ -237,35 +237,17 @@ static char* out_i64(char* w, long long v) {
     return w + sprintf(w, "%lld", v);
 }
 
-int main(int argc, char** argv) {
-    if (argc < 2) { fprintf(stderr, "usage: czillow <csv> [min_seconds]\n"); return 2; }
-    double min_seconds = argc > 2 ? atof(argv[2]) : 10.0;
+/* one full pass over [body, end); returns bytes written to outbuf */
+typedef struct {
+    long long rows, rows_out, excs;
+} counters_t;
 
-    FILE* f = fopen(argv[1], "rb");
-    if (!f) { perror("open"); return 2; }
-    fseek(f, 0, SEEK_END);
-    long long size = ftell(f);
-    fseek(f, 0, SEEK_SET);
-    char* data = (char*)malloc((size_t)size + 1);
-    if (fread(data, 1, (size_t)size, f) != (size_t)size) { perror("read"); return 2; }
-    fclose(f);
-    data[size] = 0;
-
-    /* skip header line */
-    const char* body = memchr(data, '\n', (size_t)size);
-    body = body ? body + 1 : data;
-    long long body_n = size - (body - data);
-
-    char* outbuf = (char*)malloc((size_t)body_n + (16 << 20));
+static long long run_pass(const char* body, const char* end, char* outbuf,
+                          counters_t* ct) {
     char scratch1[4096], scratch2[4096];
-
-    long long rows = 0, rows_out = 0, excs = 0, bytes_out_total = 0, passes = 0;
-    struct timespec t0, t1;
-    clock_gettime(CLOCK_MONOTONIC, &t0);
-    double elapsed = 0;
-    do {  // at least one pass, then until min_seconds of work
+    long long rows = 0, rows_out = 0, excs = 0;
+    {
         const char* p = body;
-        const char* end = body + body_n;
         char* w = outbuf;
         /* row loop: quote-parity line split + fused pipeline */
         while (p < end) {
@@ -334,18 +316,123 @@ int main(int argc, char** argv) {
             ++rows_out;
             p = next;
         }
-        bytes_out_total += (w - outbuf);
-        ++passes;
+        ct->rows += rows;
+        ct->rows_out += rows_out;
+        ct->excs += excs;
+        return w - outbuf;
+    }
+}
+
+/* ---- threading: byte-range shards with quote-parity row alignment (the
+ * reference runs its compiled path on `executorCount` = all hw threads,
+ * LocalBackend range split :552-658; this gives the honest multi-core
+ * denominator the north_star asks for) ------------------------------------- */
+
+#include <pthread.h>
+
+typedef struct {
+    const char* lo;
+    const char* hi;
+    double min_seconds;
+    counters_t ct;
+    long long bytes_out, passes;
+    double elapsed;
+} shard_t;
+
+static void* worker(void* arg) {
+    shard_t* sh = (shard_t*)arg;
+    long long n = sh->hi - sh->lo;
+    char* outbuf = (char*)malloc((size_t)n + (4 << 20));
+    struct timespec t0, t1;
+    clock_gettime(CLOCK_MONOTONIC, &t0);
+    double elapsed = 0;
+    do {
+        sh->bytes_out += run_pass(sh->lo, sh->hi, outbuf, &sh->ct);
+        ++sh->passes;
         clock_gettime(CLOCK_MONOTONIC, &t1);
         elapsed = (t1.tv_sec - t0.tv_sec) + 1e-9 * (t1.tv_nsec - t0.tv_nsec);
-    } while (elapsed < min_seconds);
+    } while (elapsed < sh->min_seconds);
+    sh->elapsed = elapsed;
+    free(outbuf);
+    return NULL;
+}
 
+int main(int argc, char** argv) {
+    if (argc < 2) {
+        fprintf(stderr, "usage: czillow <csv> [min_seconds] [threads]\n");
+        return 2;
+    }
+    double min_seconds = argc > 2 ? atof(argv[2]) : 10.0;
+    int nthreads = argc > 3 ? atoi(argv[3]) : 1;
+    if (nthreads < 1) nthreads = 1;
+    if (nthreads > 512) nthreads = 512;
+
+    FILE* f = fopen(argv[1], "rb");
+    if (!f) { perror("open"); return 2; }
+    fseek(f, 0, SEEK_END);
+    long long size = ftell(f);
+    fseek(f, 0, SEEK_SET);
+    char* data = (char*)malloc((size_t)size + 1);
+    if (fread(data, 1, (size_t)size, f) != (size_t)size) { perror("read"); return 2; }
+    fclose(f);
+    data[size] = 0;
+
+    /* skip header line */
+    const char* body = memchr(data, '\n', (size_t)size);
+    body = body ? body + 1 : data;
+    long long body_n = size - (body - data);
+
+    /* row-aligned shard boundaries (quote-parity scan, once, untimed) */
+    shard_t* sh = (shard_t*)calloc((size_t)nthreads, sizeof(shard_t));
+    {
+        const char* p = body;
+        const char* end = body + body_n;
+        int parity = 0;
+        long long target = body_n / nthreads;
+        int k = 0;
+        sh[0].lo = body;
+        const char* next_cut = body + target;
+        for (; p < end && k < nthreads - 1; ++p) {
+            if (*p == '"') parity ^= 1;
+            else if (*p == '\n' && parity == 0 && p + 1 >= next_cut) {
+                sh[k].hi = p + 1;
+                ++k;
+                sh[k].lo = p + 1;
+                next_cut = body + (long long)(k + 1) * target;
+            }
+        }
+        for (; k < nthreads; ++k) sh[k].hi = end;
+        /* degenerate shards (tiny inputs): lo may exceed earlier hi — clamp */
+        for (int i = 1; i < nthreads; ++i)
+            if (sh[i].lo < sh[i - 1].hi) sh[i].lo = sh[i - 1].hi;
+    }
+    for (int i = 0; i < nthreads; ++i) sh[i].min_seconds = min_seconds;
+
+    pthread_t tids[512];
+    for (int i = 1; i < nthreads; ++i)
+        pthread_create(&tids[i], NULL, worker, &sh[i]);
+    worker(&sh[0]);
+    for (int i = 1; i < nthreads; ++i)
+        pthread_join(tids[i], NULL);
+
+    long long rows = 0, rows_out = 0, excs = 0, bytes_out_total = 0, passes = 0;
+    long long bytes_in = 0;
+    double elapsed = 0;
+    for (int i = 0; i < nthreads; ++i) {
+        rows += sh[i].ct.rows;
+        rows_out += sh[i].ct.rows_out;
+        excs += sh[i].ct.excs;
+        bytes_out_total += sh[i].bytes_out;
+        bytes_in += sh[i].passes * (sh[i].hi - sh[i].lo);
+        passes += sh[i].passes;
+        if (sh[i].elapsed > elapsed) elapsed = sh[i].elapsed;
+    }
     printf("{\"rows\": %lld, \"rows_out\": %lld, \"exceptions\": %lld, "
            "\"bytes_in\": %lld, \"bytes_out\": %lld, \"seconds\": %.6f, "
-           "\"rows_per_s\": %.1f, \"passes\": %lld}\n",
-           rows, rows_out, excs, passes * body_n, bytes_out_total, elapsed,
-           rows / elapsed, passes);
+           "\"rows_per_s\": %.1f, \"passes\": %lld, \"threads\": %d}\n",
+           rows, rows_out, excs, bytes_in, bytes_out_total, elapsed,
+           rows / elapsed, passes, nthreads);
     free(data);
-    free(outbuf);
+    free(sh);
     return 0;
 }
