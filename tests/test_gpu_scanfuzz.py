@@ -57,7 +57,10 @@ def test_scan_fuzz_csv_source(tmp_path):
     with open(p, "w") as f:
         f.write("pre,h,post\n")
         for i, s in enumerate(data):
-            f.write("p%d,%s,%d\n" % (i, s, i))
+            # RFC-4180-quote cells containing the delimiter (the GPU parses
+            # unescaped quoted cells natively; bare commas would split the cell)
+            cell = '"%s"' % s if "," in s else s
+            f.write("p%d,%s,%d\n" % (i, cell, i))
     ctx = tuplex_amd.Context()
     ds = ctx.csv(p).map(scan_udf)
     got = ds.collect()

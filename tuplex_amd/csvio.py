@@ -450,7 +450,7 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
 
     # chunked execution at inputSplitSize boundaries (LocalBackend.cc:552-658:
     # one task per 64 MB range; here one execute per range, global row indices)
-    split = max(opts.input_split_size, 64 << 10)
+    split = max(opts.gpu_input_split_size, 64 << 10)
     chunks = (split_points(data, split) if len(data) > split * 2
               else [0, len(data)])
 
@@ -458,17 +458,21 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
     out.metrics = {"t_h2d_ms": 0.0, "t_kernel_ms": 0.0, "t_d2h_ms": 0.0,
                    "bytes_in": 0, "bytes_out": 0, "chunks": len(chunks) - 1}
     import struct as _s
+    import numpy as _np
     replayed = {}      # global row -> replay output
     all_rows = []      # (global_idx, row) for mem sink
     text_parts = []    # (chunk_res snapshot) for csv sink
     first_row = 0
+    # zero-copy input pointers: the C side only reads the chunk bytes during
+    # the call (it uploads them itself), so point straight into `data`
+    data_arr = _np.frombuffer(data, dtype=_np.uint8)
     for ci in range(len(chunks) - 1):
-        cdata = data[chunks[ci]:chunks[ci + 1]]
-        buf = (ctypes.c_uint8 * len(cdata)).from_buffer_copy(cdata)
+        clen = chunks[ci + 1] - chunks[ci]
+        cptr = ctypes.cast(data_arr.ctypes.data + chunks[ci],
+                           ctypes.POINTER(ctypes.c_uint8))
         res = TpxResult()
         rc = glib.lib.tpx_stage_execute_csv(
-            stage, ctypes.cast(buf, ctypes.POINTER(ctypes.c_uint8)), len(cdata),
-            first_row, ctypes.byref(res))
+            stage, cptr, clen, first_row, ctypes.byref(res))
         if rc != 0:
             raise RuntimeError("csv stage execute failed: " + glib.err())
         try:
@@ -506,17 +510,23 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
                 out_bytes = ctypes.string_at(res.out_data, res.out_size)
                 rws = rowfmt.deserialize_partition(out_bytes,
                                                    T.tup(sp.gpu_output_types))
-                idxs = [res.out_row_indices[i] for i in range(res.out_num_rows)]
+                idxs = _np.ctypeslib.as_array(
+                    res.out_row_indices, shape=(res.out_num_rows,)).tolist() \
+                    if res.out_num_rows else []
                 all_rows.extend(zip(idxs, rws))
             else:
+                # bulk numpy copies (a python list per row costs seconds at
+                # millions of kept rows — the e2e path is host-bound on this)
                 text = (ctypes.string_at(res.out_data, res.out_size)
                         if res.out_size else b"")
                 n_out = res.out_num_rows
-                text_parts.append((
-                    text,
-                    [res.out_row_indices[i] for i in range(n_out)],
-                    [res.out_row_offsets[i] for i in range(n_out + 1)],
-                    first_row, first_row + res.in_num_rows))
+                idxs = _np.ctypeslib.as_array(
+                    res.out_row_indices, shape=(n_out,)).copy() \
+                    if n_out else _np.empty(0, _np.int64)
+                offs = _np.ctypeslib.as_array(
+                    res.out_row_offsets, shape=(n_out + 1,)).copy()
+                text_parts.append((text, idxs, offs,
+                                   first_row, first_row + res.in_num_rows))
             first_row += res.in_num_rows
         finally:
             glib.lib.tpx_result_free(ctypes.byref(res))

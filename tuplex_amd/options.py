@@ -50,6 +50,10 @@ DEFAULTS = {
     "tuplex.gpu.deviceCount": "1",
     "tuplex.gpu.device": "0",
     "tuplex.gpu.stagingBufferSize": "256MB",
+    # GPU range granularity: the reference splits at 64MB for CPU-thread
+    # work distribution (ContextOptions.cc:227); one MI355X wants fewer,
+    # larger ranges (the C-ABI pipelines 4 sub-chunks on 2 streams per call)
+    "tuplex.gpu.inputSplitSize": "256MB",
     "tuplex.gpu.heapFactor": "2.0",
 }
 
@@ -96,6 +100,10 @@ class Options:
     @property
     def input_split_size(self) -> int:
         return parse_size(self.get("tuplex.inputSplitSize"))
+
+    @property
+    def gpu_input_split_size(self) -> int:
+        return parse_size(self.get("tuplex.gpu.inputSplitSize"))
 
     @property
     def merge_in_order(self) -> bool:
