@@ -1,7 +1,7 @@
 import sqlite3, glob, sys
 out = []
 for d in sys.argv[1:]:
-    for db_path in glob.glob(f'gpurun_out/{d}/runc/*_results.db'):
+    for db_path in sorted(glob.glob(f'gpurun_out/{d}/**/*_results.db', recursive=True) or glob.glob(f'{d}/**/*_results.db', recursive=True)):
         db = sqlite3.connect(db_path)
         cur = db.cursor()
         tabs = [r[0] for r in cur.execute("SELECT name FROM sqlite_master WHERE type='table'")]
