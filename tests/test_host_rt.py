@@ -1,0 +1,17 @@
+"""Host-compiled fuzz parity for the device runtime's scalar helpers: the
+mask-based cell walk (tpx_mwalk) vs the byte walk (tpx_csv_next_cell), on
+200k random rows including quotes, escapes, CR, high bytes, '|' delimiters
+and every alignment. Compiles tests/host_rt_test.cpp with g++ (TPX_HOST_TEST
+strips device-only kernels)."""
+import os
+import subprocess
+
+HERE = os.path.dirname(os.path.abspath(__file__))
+
+
+def test_mwalk_matches_byte_walk(tmp_path):
+    exe = os.path.join(str(tmp_path), "host_rt_test")
+    subprocess.check_call(["g++", "-O2", "-Wall", "-o", exe,
+                           os.path.join(HERE, "host_rt_test.cpp")])
+    out = subprocess.check_output([exe], timeout=300).decode()
+    assert out.strip().endswith("OK"), out

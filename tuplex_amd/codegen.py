@@ -1312,11 +1312,12 @@ class StageCodegen:
     # rows through a thrashed L1/L2 (measured 7x HBM read amplification:
     # 2048 threads x ~200 B rows >> 32 KiB L1 per CU). Waves whose span exceeds
     # TPX_SPAN_CAP fall back to parsing from global memory (rare: long rows).
-    # bytes per wave: 2 waves/block stage 2*CAP+16 B of LDS; 16368 (not 16384)
-    # keeps the block at 32752 B so FIVE blocks fit the 160 KiB CU budget
-    # (5 x 32768 rounded = 163840) -> 10 waves/CU; at 16384 the +16 tail pushes
-    # the block over 32 KiB and occupancy drops to 4 blocks = 8 waves/CU
-    SPAN_CAP = 16368
+    # bytes per wave: 2 waves/block stage 2*CAP+80 B of LDS; 16344 keeps the
+    # block at exactly 32768 B so FIVE blocks fit the 160 KiB CU budget ->
+    # 10 waves/CU (at 16384+pad the block tips over 32 KiB and occupancy
+    # drops to 4 blocks = 8 waves/CU). The 80-B tail absorbs the mask walk's
+    # aligned 64-B group overread past the last row of a wave's span.
+    SPAN_CAP = 16344
 
     def _main_kernel(self, in_types, out_types):
         if self.source == "col":
@@ -1341,7 +1342,7 @@ class StageCodegen:
         L.append("    long long* exc_buf, unsigned long long* exc_count,"
                  " unsigned long long exc_cap,")
         L.append("    void** outv) {")
-        L.append("  __shared__ char smem[2 * TPX_SPAN_CAP + 16];  // 2 waves per 128-thread block -> 5 blocks/CU")
+        L.append("  __shared__ char smem[2 * TPX_SPAN_CAP + 80];  // 2 waves per 128-thread block -> 5 blocks/CU")
         L.append("  TpxHeap heap{heap_base, heap_cursor, heap_cap, nullptr, nullptr};")
         L.append("  int lane = threadIdx.x & 63;")
         L.append("  int wid = threadIdx.x >> 6;")
@@ -1568,13 +1569,16 @@ class StageCodegen:
             L.append("    const char* rend = (const char*)in_data + in_offs[i+1];")
         L.append("    if (rend > rp && rend[-1] == '\\n') --rend;")
         L.append("    if (rend > rp && rend[-1] == '\\r') --rend;")
-        # straight-line per-cell scan: one named local per cell — a dynamically
-        # indexed cells[] array spills to scratch (288 B/thread measured) and
-        # turns every access into an HBM round trip
-        L.append("    const char* cur = rp;")
-        L.append("    bool m = true, avail = true;")
+        # straight-line per-cell walk over lazily-loaded 64-B group bitmasks
+        # (tpx_mwalk): one named local per cell — a dynamically indexed cells[]
+        # array spills to scratch (288 B/thread measured). The masks replace
+        # the serial per-cell memchr chain with 8-independent-load groups +
+        # ALU bit derivation; the ASCII gate and output-special (quote
+        # provenance) tracking ride along for free.
+        chk_comma = 1 if delim != "," else 0
+        L.append("    tpx_mwalk S; tpx_mw_init(S, rp, rend);")
+        L.append("    bool avail = true;")
         L.append("    int badf = 0;")
-        L.append("    unsigned long long hib = 0;  // non-ASCII bits seen by the walk")
 
         def null_check(idx):
             cv = "tstr{cl%d.p, cl%d.n}" % (idx, idx)
@@ -1592,8 +1596,9 @@ class StageCodegen:
             L.append("    tpx_cell cl%d{rp, 0, 0};" % idx)
             L.append("    if (!prc) {")
             L.append("      if (!avail) prc = %d;  // CSV_UNDERRUN" % 20)
-            L.append("      else { cur = tpx_csv_next_cell(cur, rend, &cl%d, &m, %s,"
-                     " &hib); avail = m; badf |= cl%d.flags; }" % (idx, delim_c, idx))
+            L.append("      else { tpx_mw_cell(S, &cl%d, %s, %d);"
+                     " avail = S.more; badf |= cl%d.flags; }"
+                     % (idx, delim_c, chk_comma, idx))
             L.append("    }")
             if used is not None and idx not in used:
                 if opt:
@@ -1632,10 +1637,10 @@ class StageCodegen:
                 raise CodegenError("csv input type %r" % (t,))
         L.append("    if (!prc && avail) prc = %d;  // CSV_OVERRUN" % 21)
         L.append("    if (!prc && (badf & 6)) prc = %d;  // BADPARSE (escapes/structure -> host)" % 70)
-        # ONE ASCII gate per row, fused into the cell walk (memchr_hi
-        # accumulated the high bits of every scanned byte; rows that error out
-        # earlier divert regardless, so partial accumulation is equivalent)
-        L.append("    if (!prc && hib) prc = %d;  // NCV" % 7)
+        # ONE ASCII gate per row, fused into the group-mask loads (rows that
+        # error out earlier divert regardless, so partial accumulation over
+        # the loaded groups is equivalent)
+        L.append("    if (!prc && S.hib) prc = %d;  // NCV" % 7)
         return L
 
     def _load_inputs_text(self, in_types, lds=True):

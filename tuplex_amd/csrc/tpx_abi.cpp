@@ -1307,7 +1307,7 @@ extern "C" int64_t tpx_stage_execute_csv(tpx_stage* st, const uint8_t* csv_bytes
     hipEvent_t ev0, ev1;
     hipEventCreate(&ev0); hipEventCreate(&ev1);
     hipEventRecord(ev0, stream);
-    ARENA_TAKE(d_in, (size_t)size);
+    ARENA_TAKE(d_in, (size_t)size + 80);  // mask-walk 64-B group overread slack
     HIP_CHECK(hipMemcpyAsync(d_in, csv_bytes, (size_t)size,
                              hipMemcpyHostToDevice, stream));
     hipEventRecord(ev1, stream);
@@ -1326,7 +1326,7 @@ extern "C" int64_t tpx_stage_execute_csv(tpx_stage* st, const uint8_t* csv_bytes
 
 extern "C" uint64_t tpx_dev_alloc(int64_t size) {
     void* p = nullptr;
-    if (hipMalloc(&p, (size_t)size + 16) != hipSuccess) return 0;  // memcpy pad
+    if (hipMalloc(&p, (size_t)size + 128) != hipSuccess) return 0;  // memcpy + mask-walk group overread pad
     return (uint64_t)(uintptr_t)p;
 }
 

@@ -913,6 +913,7 @@ __device__ __forceinline__ char* tpx_f64_csv_write(char* w,
 #define TPX_CSV_CHUNK 4096
 #define TPX_CSV_LANE_BYTES 64  /* 4096 / 64 lanes */
 
+#ifndef TPX_HOST_TEST  // device-only: wave helpers + boundary-scan kernels
 // wave helpers: 64-lane exclusive scan / reduction via shfl
 __device__ __forceinline__ long long tpx_wave_exscan(long long v) {
     long long x = v;
@@ -1059,6 +1060,8 @@ extern "C" __global__ void tpx_csv_emit_rows(const char* __restrict__ data,
     }
 }
 
+#endif  // TPX_HOST_TEST
+
 // one CSV cell starting at p; returns pointer to next cell start. *more = a
 // delimiter was consumed (another cell follows). flags: 1 quoted, 2 contains ""
 // escapes (diverted to host), 4 structurally bad (unterminated quote / junk after
@@ -1186,6 +1189,166 @@ __device__ __forceinline__ const char* tpx_csv_next_cell(const char* p,
     return q;
 }
 
+// ---- mask-based cell walk -------------------------------------------------------
+// The per-cell memchr walk is a serial dependent-load chain (~10 sequential
+// cells x ~50-cycle LDS loads per SWAR step). This walk instead classifies the
+// row's bytes in 64-BYTE GROUPS — 8 independent 8-B loads, then pure-ALU
+// bitmask derivation (movemask trick) — and consumes cell boundaries from the
+// group bitmasks. Groups load lazily and monotonically (each at most once per
+// row); semantics are EXACTLY tpx_csv_next_cell's, including flag bits and the
+// fused ASCII gate. Mirrors the reference CSVParseRowGenerator's SSE spanner
+// idea (:355) at CDNA4 width.
+
+struct tpx_grp { unsigned long long q, d, s; };
+
+struct tpx_mwalk {
+    const char* ab;   // 8-aligned base; bit i of a group mask = byte ab[64g+i]
+    int off, endb;    // row occupies bytes [off, endb) relative to ab
+    int g;            // loaded group index (-1 = none)
+    int pos;          // next cell start
+    bool more;
+    tpx_grp G;
+    unsigned long long hib;  // non-ASCII bits seen in loaded groups (row-ranged)
+};
+
+// gather the 0x80-position hit bits of one SWAR word into 8 contiguous bits
+// (byte j -> bit j); all 64 partial products land on distinct bit positions,
+// so the multiply is carry-free and exact
+__device__ __forceinline__ unsigned long long tpx_mm8(unsigned long long h) {
+    return ((h >> 7) * 0x0102040810204080ULL) >> 56;
+}
+
+__device__ __forceinline__ void tpx_mw_group(tpx_mwalk& S, int g, char delim,
+                                             int chk_comma) {
+    const unsigned long long* w =
+        (const unsigned long long*)(S.ab + ((long long)g << 6));
+    unsigned long long pq = TPX_SWAR_ONE * (unsigned long long)'"';
+    unsigned long long pd = TPX_SWAR_ONE * (unsigned char)delim;
+    unsigned long long pr = TPX_SWAR_ONE * (unsigned long long)'\r';
+    unsigned long long pc = TPX_SWAR_ONE * (unsigned long long)',';
+    unsigned long long q = 0, d = 0, s = 0, h = 0;
+    #pragma unroll
+    for (int k = 0; k < 8; ++k) {
+        unsigned long long v = w[k];
+        q |= tpx_mm8(tpx_swar_zero(v ^ pq)) << (8 * k);
+        d |= tpx_mm8(tpx_swar_zero(v ^ pd)) << (8 * k);
+        unsigned long long sv = tpx_swar_zero(v ^ pr);
+        if (chk_comma) sv |= tpx_swar_zero(v ^ pc);
+        s |= tpx_mm8(sv) << (8 * k);
+        h |= tpx_mm8(v & TPX_SWAR_HIGH) << (8 * k);
+    }
+    int base = g << 6;
+    int lo = S.off - base; if (lo < 0) lo = 0;          // lo in [0, 63]
+    int hi = S.endb - base; if (hi > 64) hi = 64;
+    unsigned long long rm = (hi <= lo) ? 0ULL :
+        ((hi >= 64 ? ~0ULL : ((1ULL << hi) - 1)) & ~((1ULL << lo) - 1));
+    S.G.q = q & rm;
+    S.G.d = d & rm;
+    S.G.s = (s | q) & rm;  // a '"' byte in unquoted content also forces quoting
+    S.hib |= h & rm;
+    S.g = g;
+}
+
+__device__ __forceinline__ void tpx_mw_init(tpx_mwalk& S, const char* rp,
+                                            const char* rend) {
+    S.ab = (const char*)((unsigned long long)rp & ~7ULL);
+    S.off = (int)(rp - S.ab);
+    S.endb = S.off + (int)(rend - rp);
+    S.g = -1;
+    S.pos = S.off;
+    S.more = true;
+    S.hib = 0;
+}
+
+__device__ __forceinline__ int tpx_mw_test_q(tpx_mwalk& S, int k, char delim,
+                                             int chk) {
+    if (k >= S.endb) return 0;
+    if ((k >> 6) != S.g) tpx_mw_group(S, k >> 6, delim, chk);
+    return (int)((S.G.q >> (k & 63)) & 1);
+}
+
+__device__ __forceinline__ int tpx_mw_test_d(tpx_mwalk& S, int k, char delim,
+                                             int chk) {
+    if (k >= S.endb) return 0;
+    if ((k >> 6) != S.g) tpx_mw_group(S, k >> 6, delim, chk);
+    return (int)((S.G.d >> (k & 63)) & 1);
+}
+
+// next delim bit >= k (or endb); ORs output-special bits in [k, result) into
+// *spec — the word-granular analog of tpx_memchr_hi_spec
+__device__ __forceinline__ int tpx_mw_next_d(tpx_mwalk& S, int k, char delim,
+                                             int chk, int* spec) {
+    unsigned long long sp_acc = 0;
+    for (;;) {
+        if (k >= S.endb) { if (sp_acc) *spec = 1; return S.endb; }
+        if ((k >> 6) != S.g) tpx_mw_group(S, k >> 6, delim, chk);
+        int local = k & 63;
+        unsigned long long ge = local ? (~0ULL << local) : ~0ULL;
+        unsigned long long d = S.G.d & ge;
+        unsigned long long s = S.G.s & ge;
+        if (d) {
+            int b = __ffsll((long long)d) - 1;
+            sp_acc |= s & ((b ? (1ULL << b) : 1ULL) - 1);
+            if (sp_acc) *spec = 1;
+            return (S.g << 6) + b;
+        }
+        sp_acc |= s;
+        k = (S.g + 1) << 6;
+    }
+}
+
+__device__ __forceinline__ int tpx_mw_next_q(tpx_mwalk& S, int k, char delim,
+                                             int chk) {
+    for (;;) {
+        if (k >= S.endb) return S.endb;
+        if ((k >> 6) != S.g) tpx_mw_group(S, k >> 6, delim, chk);
+        int local = k & 63;
+        unsigned long long ge = local ? (~0ULL << local) : ~0ULL;
+        unsigned long long q = S.G.q & ge;
+        if (q) return (S.g << 6) + (__ffsll((long long)q) - 1);
+        k = (S.g + 1) << 6;
+    }
+}
+
+// one cell via the group masks — flag/position semantics identical to
+// tpx_csv_next_cell (incl. bit 8 output-special tracking)
+__device__ __forceinline__ void tpx_mw_cell(tpx_mwalk& S, tpx_cell* c,
+                                            char delim, int chk) {
+    c->flags = 0;
+    S.more = false;
+    int pos = S.pos;
+    if (pos < S.endb && tpx_mw_test_q(S, pos, delim, chk)) {  // quoted cell
+        int q = pos + 1;
+        bool esc = false;
+        for (;;) {
+            int cq = tpx_mw_next_q(S, q, delim, chk);
+            if (cq >= S.endb) {  // unterminated
+                c->p = S.ab + pos; c->n = S.endb - pos; c->flags = 4;
+                S.pos = S.endb;
+                return;
+            }
+            if (cq + 1 < S.endb && tpx_mw_test_q(S, cq + 1, delim, chk)) {
+                esc = true; q = cq + 2; continue;
+            }
+            q = cq;
+            break;
+        }
+        c->p = S.ab + pos + 1; c->n = q - (pos + 1);
+        c->flags = 1 | (esc ? 2 : 0) | 8;
+        ++q;
+        if (q < S.endb && !tpx_mw_test_d(S, q, delim, chk)) c->flags |= 4;
+        int spec = 0;
+        int nd = tpx_mw_next_d(S, q, delim, chk, &spec);
+        if (nd < S.endb) { S.more = true; S.pos = nd + 1; } else S.pos = S.endb;
+        return;
+    }
+    int spec = 0;
+    int nd = tpx_mw_next_d(S, pos, delim, chk, &spec);
+    c->p = S.ab + pos; c->n = nd - pos;
+    if (spec) c->flags |= 8;
+    if (nd < S.endb) { S.more = true; S.pos = nd + 1; } else S.pos = S.endb;
+}
+
 // typed cell parse (cells path semantics: python-whitespace trim + fast_atoX,
 // CellSourceTaskBuilder + Runtime.cc:319 wrappers)
 __device__ __forceinline__ int tpx_cell_i64(const tpx_cell& c, long long* out) {
@@ -1233,6 +1396,7 @@ __device__ __forceinline__ int tpx_cell_bool(const tpx_cell& c, bool* out) {
     return EC_BOOLPARSE_;
 }
 
+#ifndef TPX_HOST_TEST  // device-only: reduction/hash/scan kernels
 // ---- fixed kernels ---------------------------------------------------------------
 
 // ---- aggregate reduction (AggregateFunctions.cc fold -> deterministic device
@@ -1597,3 +1761,4 @@ extern "C" __global__ void tpx_scan_add(long long* __restrict__ data,
         if (i < n) data[i] += add;
     }
 }
+#endif  // TPX_HOST_TEST
