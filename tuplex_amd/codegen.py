@@ -1370,14 +1370,12 @@ class StageCodegen:
         L.append("    }")
         L.append("    long long i = r0 + lane;")
         L.append("    if (i >= rhi) continue;")
-        # duplicate the row body per pointer mode: in the staged branch every
-        # parse pointer provably derives from LDS, so address-space inference
-        # emits ds_read instead of flat loads
-        L.append("    if (staged) {")
-        L.extend("  " + ln for ln in self._row_body(in_types, out_types, lds=True))
-        L.append("    } else {")
-        L.extend("  " + ln for ln in self._row_body(in_types, out_types, lds=False))
-        L.append("    }")
+        # ONE generic-pointer row body for both the staged (LDS) and overflow
+        # (global) paths: flat loads on LDS addresses were measured as fast as
+        # ds_read for this kernel (profiles/README.md, hot-cell slot ablation),
+        # and a single body halves the kernel size and the ~60 s hipRTC
+        # compile of big stages that the duplicated bodies caused
+        L.extend(self._row_body(in_types, out_types, lds="gen"))
         L.append("  }")
         L.append("}")
         return "\n".join(L)
@@ -1509,7 +1507,11 @@ class StageCodegen:
         """Deserialize one reference-layout row (Serializer.cc:20-24) into typed
         locals c0..cN."""
         L = ["    long long prc = 0;  // no pre-parse errors on the mem path"]
-        if lds:
+        if lds == "gen":
+            L.append("    const unsigned char* row = staged ? (const unsigned"
+                     " char*)(wave_lds + (in_offs[i] - span_start)) :"
+                     " (in_data + in_offs[i]);")
+        elif lds:
             L.append("    const unsigned char* row ="
                      " (const unsigned char*)(wave_lds + (in_offs[i] - span_start));")
         else:
@@ -1561,7 +1563,12 @@ class StageCodegen:
         delim_c = "'\\t'" if delim == "\t" else "'%s'" % delim
         L = []
         L.append("    long long prc = 0;")
-        if lds:
+        if lds == "gen":
+            L.append("    const char* rp = staged ? (wave_lds + (in_offs[i] -"
+                     " span_start)) : ((const char*)in_data + in_offs[i]);")
+            L.append("    const char* rend = staged ? (wave_lds + (in_offs[i+1]"
+                     " - span_start)) : ((const char*)in_data + in_offs[i+1]);")
+        elif lds:
             L.append("    const char* rp = wave_lds + (in_offs[i] - span_start);")
             L.append("    const char* rend = wave_lds + (in_offs[i+1] - span_start);")
         else:
@@ -1650,7 +1657,12 @@ class StageCodegen:
         t = in_types[0]
         null_values = self.csv_info.get("null_values", [])
         L = ["    long long prc = 0;"]
-        if lds:
+        if lds == "gen":
+            L.append("    const char* rp = staged ? (wave_lds + (in_offs[i] -"
+                     " span_start)) : ((const char*)in_data + in_offs[i]);")
+            L.append("    const char* rend = staged ? (wave_lds + (in_offs[i+1]"
+                     " - span_start)) : ((const char*)in_data + in_offs[i+1]);")
+        elif lds:
             L.append("    const char* rp = wave_lds + (in_offs[i] - span_start);")
             L.append("    const char* rend = wave_lds + (in_offs[i+1] - span_start);")
         else:
@@ -1682,7 +1694,11 @@ class StageCodegen:
         for k, t in enumerate(out_types):
             base = T.deopt(t)
             if base == T.STR:
-                if lds:
+                if lds == "gen":
+                    L.append("    if (staged) o.o%d = tpx_to_global(o.o%d,"
+                             " wave_lds, wave_lds + TPX_SPAN_CAP, in_data,"
+                             " span_start);" % (k, k))
+                elif lds:
                     L.append("    o.o%d = tpx_to_global(o.o%d, wave_lds,"
                              " wave_lds + TPX_SPAN_CAP, in_data, span_start);"
                              % (k, k))
@@ -2233,16 +2249,17 @@ class StageCodegen:
         L.append("      out_offs[keep_scan[i]] = out_byte0 + my_start;")
         L.append("      out_rowidx[keep_scan[i]] = row0 + i;")
         L.append("    }")
-        # duplicate the format body per pointer mode: in the staged branch w
-        # provably derives from LDS, so addrspace inference emits ds_write
-        # instead of flat stores (same trick as tpx_stage_main's row body)
+        # ONE generic-pointer format body for both the staged (LDS) and
+        # overflow (global) paths — flat stores handle either address space;
+        # halves the kernel and its hipRTC compile time
         body = self._csv_format_body(out_types)
+        L.append("    if (active) {")
+        L.append("      bool _noq = (keep[i] & 2) != 0;")
+        L.append("      char* w = staged ? (wave_lds + (my_start - span_start))"
+                 " : ((char*)out_data + my_start);")
+        L.extend("  " + ln for ln in body)
+        L.append("    }")
         L.append("    if (staged) {")
-        L.append("      if (active) {")
-        L.append("        bool _noq = (keep[i] & 2) != 0;")
-        L.append("        char* w = wave_lds + (my_start - span_start);")
-        L.extend("    " + ln for ln in body)
-        L.append("      }")
         L.append("      __builtin_amdgcn_wave_barrier();")
         # cooperative span copy with ALIGNED global 8B stores (dst + a0 is
         # 8-aligned by construction) and aligned LDS 8B reads + a uniform
@@ -2265,10 +2282,6 @@ class StageCodegen:
         L.append("      }")
         L.append("      long long t0 = span > a0 ? a0 + ((span - a0) & ~7LL) : span;")
         L.append("      if (t0 + lane < span) dst[t0 + lane] = wave_lds[t0 + lane];")
-        L.append("    } else if (active) {")
-        L.append("      bool _noq = (keep[i] & 2) != 0;")
-        L.append("      char* w = (char*)out_data + my_start;")
-        L.extend("  " + ln for ln in body)
         L.append("    }")
         L.append("  }")
         L.append("}")
