@@ -102,8 +102,9 @@ def test_orc_fallback_trailing_aggregate(tmp_path):
     p = os.path.join(str(tmp_path), "t.orc")
     paorc.write_table(tab, p)
     ctx = tuplex_amd.Context()
+    # single-column rows are SCALARS in UDFs (reference row semantics)
     ds = ctx.orc(p).mapColumn("a", _noncompilable_scalar) \
-        .aggregate(lambda a, b: max(a, b), lambda a, x: max(a, x["a"]), 0)
+        .aggregate(lambda a, b: max(a, b), lambda a, x: max(a, x), 0)
     got = ds.collect()
     assert ds._last_outcome.mode == "fallback"
     assert got == [9]

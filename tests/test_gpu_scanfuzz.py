@@ -92,8 +92,9 @@ def test_orc_trailing_nonreducible_aggregate(tmp_path):
     p = os.path.join(str(tmp_path), "t.orc")
     paorc.write_table(tab, p)
     ctx = tuplex_amd.Context()
+    # single-column rows are SCALARS in UDFs (reference row semantics)
     ds = ctx.orc(p).aggregate(lambda a, b: max(a, b),
-                              lambda a, x: max(a, x["a"]), 0)
+                              lambda a, x: max(a, x), 0)
     got = ds.collect()
     assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
     assert got == [99]

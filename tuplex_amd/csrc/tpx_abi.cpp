@@ -1196,7 +1196,7 @@ extern "C" int64_t tpx_stage_execute(tpx_stage* st, const tpx_partition* parts,
     hipEvent_t ev0, ev1;
     hipEventCreate(&ev0); hipEventCreate(&ev1);
     hipEventRecord(ev0, stream);
-    ARENA_TAKE(d_in, (size_t)in_bytes);
+    ARENA_TAKE(d_in, (size_t)in_bytes + 80);  // scan-window/group overread slack
     ARENA_TAKE(d_offs, (size_t)(n + 1) * 8);
     {
         std::vector<long long> offs((size_t)n + 1);

@@ -450,7 +450,11 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
 
     # chunked execution at inputSplitSize boundaries (LocalBackend.cc:552-658:
     # one task per 64 MB range; here one execute per range, global row indices)
-    split = max(opts.gpu_input_split_size, 64 << 10)
+    # an explicitly-set tuplex.inputSplitSize wins (reference semantics);
+    # otherwise the GPU default (bigger ranges, fewer sync points)
+    split = (opts.input_split_size if opts.is_set("tuplex.inputSplitSize")
+             else opts.gpu_input_split_size)
+    split = max(split, 64 << 10)
     chunks = (split_points(data, split) if len(data) > split * 2
               else [0, len(data)])
 

@@ -61,10 +61,16 @@ DEFAULTS = {
 class Options:
     def __init__(self, conf: Optional[dict] = None):
         self.store = dict(DEFAULTS)
+        self.explicit = set()
         if conf:
             for k, v in conf.items():
                 key = k if k.startswith("tuplex.") else "tuplex." + k
                 self.store[key] = self._fmt(v)
+                self.explicit.add(key)
+
+    def is_set(self, key) -> bool:
+        key = key if key.startswith("tuplex.") else "tuplex." + key
+        return key in getattr(self, "explicit", set())
 
     @staticmethod
     def _fmt(v):

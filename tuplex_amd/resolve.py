@@ -60,7 +60,22 @@ def replay_row(value: Any, logical_ops: List[tuple], columns: Optional[List[str]
     return _replay_from(value, list(columns) if columns else None, row_ops, 0)
 
 
+def _unwrap1(v):
+    """Arity-1 rows collect as scalars (the GPU merges unwrap 1-tuples the
+    same way; PythonDataSet.cc fast conversion does too)."""
+    return v[0] if isinstance(v, tuple) and len(v) == 1 else v
+
+
 def _replay_from(cur, cols, row_ops, k):
+    r = _replay_from_raw(cur, cols, row_ops, k)
+    if r[0] == "row":
+        return ("row", _unwrap1(r[1]))
+    if r[0] == "rows":
+        return ("rows", [_unwrap1(v) for v in r[1]], r[2])
+    return r
+
+
+def _replay_from_raw(cur, cols, row_ops, k):
     while k < len(row_ops):
         op = row_ops[k]
         kind = op[0]
