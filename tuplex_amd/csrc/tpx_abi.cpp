@@ -163,8 +163,12 @@ static bool compile_hiprtc(const char* src, std::vector<char>* code, std::string
         *log = "hiprtcCreateProgram failed";
         return false;
     }
-    const char* opts[] = {"--offload-arch=gfx950", "-O3", "-std=c++17"};
-    hiprtcResult rc = hiprtcCompileProgram(prog, 3, opts);
+    // -ffp-contract=off: the reference's compiled path (x86 SSE2, no FMA)
+    // and the oracle never contract a*b+c; fused multiply-add in fast_atod's
+    // digit accumulation shifts parsed doubles by an ulp and breaks bit parity
+    const char* opts[] = {"--offload-arch=gfx950", "-O3", "-std=c++17",
+                          "-ffp-contract=off"};
+    hiprtcResult rc = hiprtcCompileProgram(prog, 4, opts);
     size_t log_size = 0;
     hiprtcGetProgramLogSize(prog, &log_size);
     if (log_size > 1) {
