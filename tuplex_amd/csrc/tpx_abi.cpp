@@ -202,7 +202,9 @@ extern "C" tpx_stage* tpx_stage_compile(const char* hip_source,
     std::string cache_path;
     if (cache_dir && *cache_dir) {
         char name[64];
-        snprintf(name, sizeof name, "tpx_%016llx_%zu.hsaco",
+        // key = source hash + length + COMPILE-OPTION GENERATION (bump when
+        // hipRTC options change, or stale hsacos built under old options load)
+        snprintf(name, sizeof name, "tpx_%016llx_%zu_g2.hsaco",
                  (unsigned long long)fnv1a(hip_source, strlen(hip_source)),
                  strlen(hip_source));
         cache_path = std::string(cache_dir) + "/" + name;
