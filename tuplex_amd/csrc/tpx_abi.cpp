@@ -260,6 +260,8 @@ extern "C" tpx_stage* tpx_stage_compile(const char* hip_source,
             delete st;
             return nullptr;
         }
+        if (getenv("TPX_TRACE") && r == hipSuccess)
+            fprintf(stderr, "[tpx] kernel %s = %p\n", e.name, (void*)*e.fn);
     }
     st->loaded = true;
     return st;
@@ -387,8 +389,19 @@ static int cur_device() {
 
 static int launch(hipFunction_t f, unsigned grid, unsigned block, hipStream_t s,
                   void** args) {
+    static int trace = getenv("TPX_TRACE") ? 1 : 0;
+    if (trace) {
+        fprintf(stderr, "[tpx] launch f=%p grid=%u block=%u\n", (void*)f,
+                grid, block);
+        (void)hipDeviceSynchronize();
+    }
     hipError_t e = hipModuleLaunchKernel(f, grid, 1, 1, block, 1, 1, 0, s, args,
                                          nullptr);
+    if (trace) {
+        hipError_t e2 = hipDeviceSynchronize();
+        fprintf(stderr, "[tpx] done  f=%p rc=%d sync=%d\n", (void*)f, (int)e,
+                (int)e2);
+    }
     if (e != hipSuccess) { set_err(hipGetErrorString(e)); return -1; }
     return 0;
 }
