@@ -290,13 +290,25 @@ __device__ __forceinline__ void tpx_memcpy(char* d, const char* s, long long n) 
 
 // translate an LDS-staged string view back to its global-memory address before it
 // escapes the kernel (columnar string cells must outlive the LDS staging window).
-// (p - lds_lo) + span_start is the byte offset in the input.
+// (p - lds_lo) + span_start is the byte offset in the input. The LDS test uses
+// the HARDWARE aperture check (__builtin_amdgcn_is_shared), NOT a pointer
+// range compare: relational comparison of pointers with different provenance
+// is unspecified and the optimizer folds it once the row pointer is a
+// staged-or-global generic select (observed: untranslated LDS-aperture
+// pointers escaping to the write kernel -> APERTURE_VIOLATION).
 __device__ __forceinline__ tstr tpx_to_global(tstr s, const char* lds_lo,
                                               const char* lds_hi,
                                               const unsigned char* gbase,
                                               long long span_start) {
-    if (s.p >= lds_lo && s.p < lds_hi)
-        return tstr{(const char*)gbase + span_start + (s.p - lds_lo), s.n};
+#ifdef TPX_HOST_TEST
+    bool in_lds = s.p >= lds_lo && s.p < lds_hi;
+#else
+    bool in_lds = __builtin_amdgcn_is_shared((const void*)s.p);
+#endif
+    if (in_lds)
+        return tstr{(const char*)gbase + span_start +
+                        ((unsigned long long)s.p - (unsigned long long)lds_lo),
+                    s.n};
     return s;
 }
 
