@@ -1694,6 +1694,9 @@ class StageCodegen:
         for k, t in enumerate(out_types):
             base = T.deopt(t)
             if base == T.STR:
+                import os as _os
+                if int(_os.environ.get("TPX_WDBG", "0")) == 3:
+                    L.append("    o.o%d = tstr{(const char*)in_data, 0};" % k)
                 if lds == "gen":
                     L.append("    if (staged) o.o%d = tpx_to_global(o.o%d,"
                              " wave_lds, wave_lds + TPX_SPAN_CAP, in_data,"
@@ -2056,6 +2059,8 @@ class StageCodegen:
                 raise CodegenError(">64 optional fields unsupported")
         # slots + varlen
         L.append("    long long var_off = 0;  // within varlen region")
+        import os as _os
+        wdbg = int(_os.environ.get("TPX_WDBG", "0"))
         for k, t in enumerate(out_types):
             base = T.deopt(t)
             slot = "((long long*)(w + %d))[%d]" % (bitmap, k)
@@ -2066,8 +2071,13 @@ class StageCodegen:
                          % (fixed_end - bitmap, 8 * k))
                 L.append("      %s = off | ((v%d.n + 1) << 32);" % (slot, k))
                 L.append("      char* d = (char*)(w + %d + var_off);" % (fixed_end + 8))
-                L.append("      tpx_memcpy(d, v%d.p, v%d.n);" % (k, k))
-                L.append("      d[v%d.n] = 0;" % k)
+                if wdbg == 0:
+                    L.append("      tpx_memcpy(d, v%d.p, v%d.n);" % (k, k))
+                    L.append("      d[v%d.n] = 0;" % k)
+                elif wdbg == 1:   # skip string READS, keep layout writes
+                    L.append("      for (long long _z = 0; _z <= v%d.n; ++_z)"
+                             " d[_z] = 0;" % k)
+                # wdbg >= 2: no varlen writes at all
                 L.append("      var_off += v%d.n + 1;" % k)
                 L.append("    }")
             elif base == T.F64:
