@@ -1695,12 +1695,22 @@ class StageCodegen:
             base = T.deopt(t)
             if base == T.STR:
                 import os as _os
-                if int(_os.environ.get("TPX_WDBG", "0")) == 3:
+                _wd = int(_os.environ.get("TPX_WDBG", "0"))
+                if _wd == 3:
                     L.append("    o.o%d = tstr{(const char*)in_data, 0};" % k)
+                if _wd == 9 and k == 1:
+                    L.append("    o.o0 = (long long)o.o%d.p;  // DBG pre" % k)
                 if lds == "gen":
                     L.append("    if (staged) o.o%d = tpx_to_global(o.o%d,"
                              " wave_lds, wave_lds + TPX_SPAN_CAP, in_data,"
                              " span_start);" % (k, k))
+                    if _wd == 7 and k == 1:
+                        L.append("    o.o0 = (long long)o.o%d.p;  // DBG post"
+                                 % k)
+                    if _wd == 8 and k == 1:
+                        L.append("    o.o0 = (long long)wave_lds +"
+                                 " ((long long)(unsigned long long)in_data"
+                                 " << 1 >> 1 & 0);  // DBG lds base")
                 elif lds:
                     L.append("    o.o%d = tpx_to_global(o.o%d, wave_lds,"
                              " wave_lds + TPX_SPAN_CAP, in_data, span_start);"
@@ -2074,7 +2084,7 @@ class StageCodegen:
                 if wdbg == 0:
                     L.append("      tpx_memcpy(d, v%d.p, v%d.n);" % (k, k))
                     L.append("      d[v%d.n] = 0;" % k)
-                elif wdbg == 1:   # skip string READS, keep layout writes
+                elif wdbg in (1, 7, 8, 9):  # skip string READS, keep layout
                     L.append("      for (long long _z = 0; _z <= v%d.n; ++_z)"
                              " d[_z] = 0;" % k)
                 # wdbg >= 2: no varlen writes at all
