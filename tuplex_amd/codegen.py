@@ -1683,6 +1683,21 @@ class StageCodegen:
         """Store Out o -> columnar arrays + per-row serialized size (mem sink) or
         csv text size (csv sink)."""
         L = []
+        import os as _os2
+        _wd2 = int(_os2.environ.get("TPX_WDBG", "0"))
+        if _wd2 in (7, 8, 9, 10, 11) and lds == "gen":
+            if _wd2 == 7:
+                L.append("    { tstr _t = o.o1; if (staged) _t = tpx_to_global"
+                         "(_t, wave_lds, wave_lds + TPX_SPAN_CAP, in_data,"
+                         " span_start); o.o0 = (long long)_t.p; }  // DBG")
+            elif _wd2 == 9:
+                L.append("    o.o0 = (long long)o.o1.p;  // DBG pre-translate")
+            elif _wd2 == 8:
+                L.append("    o.o0 = (long long)wave_lds;  // DBG")
+            elif _wd2 == 10:
+                L.append("    o.o0 = (long long)in_data;  // DBG")
+            elif _wd2 == 11:
+                L.append("    o.o0 = staged ? span_start : -1;  // DBG")
         # csv sizes FIRST, while string views still point into LDS (the quote
         # scan then runs on ds_read instead of re-reading global memory; the
         # bytes are identical either way)
@@ -1698,19 +1713,10 @@ class StageCodegen:
                 _wd = int(_os.environ.get("TPX_WDBG", "0"))
                 if _wd == 3:
                     L.append("    o.o%d = tstr{(const char*)in_data, 0};" % k)
-                if _wd == 9 and k == 1:
-                    L.append("    o.o0 = (long long)o.o%d.p;  // DBG pre" % k)
                 if lds == "gen":
                     L.append("    if (staged) o.o%d = tpx_to_global(o.o%d,"
                              " wave_lds, wave_lds + TPX_SPAN_CAP, in_data,"
                              " span_start);" % (k, k))
-                    if _wd == 7 and k == 1:
-                        L.append("    o.o0 = (long long)o.o%d.p;  // DBG post"
-                                 % k)
-                    if _wd == 8 and k == 1:
-                        L.append("    o.o0 = (long long)wave_lds +"
-                                 " ((long long)(unsigned long long)in_data"
-                                 " << 1 >> 1 & 0);  // DBG lds base")
                 elif lds:
                     L.append("    o.o%d = tpx_to_global(o.o%d, wave_lds,"
                              " wave_lds + TPX_SPAN_CAP, in_data, span_start);"
