@@ -1508,9 +1508,11 @@ class StageCodegen:
         locals c0..cN."""
         L = ["    long long prc = 0;  // no pre-parse errors on the mem path"]
         if lds == "gen":
-            L.append("    const unsigned char* row = staged ? (const unsigned"
-                     " char*)(wave_lds + (in_offs[i] - span_start)) :"
-                     " (in_data + in_offs[i]);")
+            # u64-opaque select (see csv loader note)
+            L.append("    const unsigned char* row = (const unsigned char*)"
+                     "(staged ? (unsigned long long)(wave_lds + (in_offs[i] -"
+                     " span_start)) : (unsigned long long)(in_data +"
+                     " in_offs[i]));")
         elif lds:
             L.append("    const unsigned char* row ="
                      " (const unsigned char*)(wave_lds + (in_offs[i] - span_start));")
@@ -1564,10 +1566,18 @@ class StageCodegen:
         L = []
         L.append("    long long prc = 0;")
         if lds == "gen":
-            L.append("    const char* rp = staged ? (wave_lds + (in_offs[i] -"
-                     " span_start)) : ((const char*)in_data + in_offs[i]);")
-            L.append("    const char* rend = staged ? (wave_lds + (in_offs[i+1]"
-                     " - span_start)) : ((const char*)in_data + in_offs[i+1]);")
+            # the staged/global pointer select goes through u64: a mixed
+            # LDS/global pointer select lets InferAddressSpaces collapse the
+            # generic pointer (observed: 0xFFFFFFFF_xxxxxxx VAs escaping) —
+            # integer selects are opaque to it, flat loads handle either space
+            L.append("    const char* rp = (const char*)(staged ?"
+                     " (unsigned long long)(wave_lds + (in_offs[i] -"
+                     " span_start)) : (unsigned long long)((const char*)"
+                     "in_data + in_offs[i]));")
+            L.append("    const char* rend = (const char*)(staged ?"
+                     " (unsigned long long)(wave_lds + (in_offs[i+1] -"
+                     " span_start)) : (unsigned long long)((const char*)"
+                     "in_data + in_offs[i+1]));")
         elif lds:
             L.append("    const char* rp = wave_lds + (in_offs[i] - span_start);")
             L.append("    const char* rend = wave_lds + (in_offs[i+1] - span_start);")
@@ -1658,10 +1668,15 @@ class StageCodegen:
         null_values = self.csv_info.get("null_values", [])
         L = ["    long long prc = 0;"]
         if lds == "gen":
-            L.append("    const char* rp = staged ? (wave_lds + (in_offs[i] -"
-                     " span_start)) : ((const char*)in_data + in_offs[i]);")
-            L.append("    const char* rend = staged ? (wave_lds + (in_offs[i+1]"
-                     " - span_start)) : ((const char*)in_data + in_offs[i+1]);")
+            # u64-opaque select (see csv loader note)
+            L.append("    const char* rp = (const char*)(staged ?"
+                     " (unsigned long long)(wave_lds + (in_offs[i] -"
+                     " span_start)) : (unsigned long long)((const char*)"
+                     "in_data + in_offs[i]));")
+            L.append("    const char* rend = (const char*)(staged ?"
+                     " (unsigned long long)(wave_lds + (in_offs[i+1] -"
+                     " span_start)) : (unsigned long long)((const char*)"
+                     "in_data + in_offs[i+1]));")
         elif lds:
             L.append("    const char* rp = wave_lds + (in_offs[i] - span_start);")
             L.append("    const char* rend = wave_lds + (in_offs[i+1] - span_start);")
@@ -2281,8 +2296,9 @@ class StageCodegen:
         body = self._csv_format_body(out_types)
         L.append("    if (active) {")
         L.append("      bool _noq = (keep[i] & 2) != 0;")
-        L.append("      char* w = staged ? (wave_lds + (my_start - span_start))"
-                 " : ((char*)out_data + my_start);")
+        L.append("      char* w = (char*)(staged ? (unsigned long long)"
+                 "(wave_lds + (my_start - span_start)) : (unsigned long long)"
+                 "((char*)out_data + my_start));")
         L.extend("  " + ln for ln in body)
         L.append("    }")
         L.append("    if (staged) {")
