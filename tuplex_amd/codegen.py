@@ -1509,10 +1509,14 @@ class StageCodegen:
         L = ["    long long prc = 0;  // no pre-parse errors on the mem path"]
         if lds == "gen":
             # u64-opaque select (see csv loader note)
-            L.append("    const unsigned char* row = (const unsigned char*)"
-                     "(staged ? (unsigned long long)(wave_lds + (in_offs[i] -"
+            L.append("    unsigned long long _rwa = staged ?"
+                     " (unsigned long long)(wave_lds + (in_offs[i] -"
                      " span_start)) : (unsigned long long)(in_data +"
-                     " in_offs[i]));")
+                     " in_offs[i]);")
+            L.append('    asm volatile("" : "+v"(_rwa));'
+                     "  // opaque: block addrspace re-inference")
+            L.append("    const unsigned char* row ="
+                     " (const unsigned char*)_rwa;")
         elif lds:
             L.append("    const unsigned char* row ="
                      " (const unsigned char*)(wave_lds + (in_offs[i] - span_start));")
@@ -1570,14 +1574,18 @@ class StageCodegen:
             # LDS/global pointer select lets InferAddressSpaces collapse the
             # generic pointer (observed: 0xFFFFFFFF_xxxxxxx VAs escaping) —
             # integer selects are opaque to it, flat loads handle either space
-            L.append("    const char* rp = (const char*)(staged ?"
+            L.append("    unsigned long long _rpa = staged ?"
                      " (unsigned long long)(wave_lds + (in_offs[i] -"
                      " span_start)) : (unsigned long long)((const char*)"
-                     "in_data + in_offs[i]));")
-            L.append("    const char* rend = (const char*)(staged ?"
+                     "in_data + in_offs[i]);")
+            L.append("    unsigned long long _rea = staged ?"
                      " (unsigned long long)(wave_lds + (in_offs[i+1] -"
                      " span_start)) : (unsigned long long)((const char*)"
-                     "in_data + in_offs[i+1]));")
+                     "in_data + in_offs[i+1]);")
+            L.append('    asm volatile("" : "+v"(_rpa), "+v"(_rea));'
+                     "  // opaque: block addrspace re-inference")
+            L.append("    const char* rp = (const char*)_rpa;")
+            L.append("    const char* rend = (const char*)_rea;")
         elif lds:
             L.append("    const char* rp = wave_lds + (in_offs[i] - span_start);")
             L.append("    const char* rend = wave_lds + (in_offs[i+1] - span_start);")
@@ -1669,14 +1677,18 @@ class StageCodegen:
         L = ["    long long prc = 0;"]
         if lds == "gen":
             # u64-opaque select (see csv loader note)
-            L.append("    const char* rp = (const char*)(staged ?"
+            L.append("    unsigned long long _rpa = staged ?"
                      " (unsigned long long)(wave_lds + (in_offs[i] -"
                      " span_start)) : (unsigned long long)((const char*)"
-                     "in_data + in_offs[i]));")
-            L.append("    const char* rend = (const char*)(staged ?"
+                     "in_data + in_offs[i]);")
+            L.append("    unsigned long long _rea = staged ?"
                      " (unsigned long long)(wave_lds + (in_offs[i+1] -"
                      " span_start)) : (unsigned long long)((const char*)"
-                     "in_data + in_offs[i+1]));")
+                     "in_data + in_offs[i+1]);")
+            L.append('    asm volatile("" : "+v"(_rpa), "+v"(_rea));'
+                     "  // opaque: block addrspace re-inference")
+            L.append("    const char* rp = (const char*)_rpa;")
+            L.append("    const char* rend = (const char*)_rea;")
         elif lds:
             L.append("    const char* rp = wave_lds + (in_offs[i] - span_start);")
             L.append("    const char* rend = wave_lds + (in_offs[i+1] - span_start);")
@@ -2296,9 +2308,11 @@ class StageCodegen:
         body = self._csv_format_body(out_types)
         L.append("    if (active) {")
         L.append("      bool _noq = (keep[i] & 2) != 0;")
-        L.append("      char* w = (char*)(staged ? (unsigned long long)"
-                 "(wave_lds + (my_start - span_start)) : (unsigned long long)"
-                 "((char*)out_data + my_start));")
+        L.append("      unsigned long long _wa = staged ? (unsigned long"
+                 " long)(wave_lds + (my_start - span_start)) : (unsigned"
+                 " long long)((char*)out_data + my_start);")
+        L.append('      asm volatile("" : "+v"(_wa));')
+        L.append("      char* w = (char*)_wa;")
         L.extend("  " + ln for ln in body)
         L.append("    }")
         L.append("    if (staged) {")
