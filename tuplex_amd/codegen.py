@@ -1510,8 +1510,8 @@ class StageCodegen:
         if lds == "gen":
             # u64-opaque select (see csv loader note)
             L.append("    unsigned long long _rwa = staged ?"
-                     " (unsigned long long)(wave_lds + (in_offs[i] -"
-                     " span_start)) : (unsigned long long)(in_data +"
+                     " (unsigned long long)wave_lds + (unsigned long long)(in_offs[i]"
+                     " - span_start) : (unsigned long long)(in_data +"
                      " in_offs[i]);")
             L.append('    asm volatile("" : "+v"(_rwa));'
                      "  // opaque: block addrspace re-inference")
@@ -1575,12 +1575,12 @@ class StageCodegen:
             # generic pointer (observed: 0xFFFFFFFF_xxxxxxx VAs escaping) —
             # integer selects are opaque to it, flat loads handle either space
             L.append("    unsigned long long _rpa = staged ?"
-                     " (unsigned long long)(wave_lds + (in_offs[i] -"
-                     " span_start)) : (unsigned long long)((const char*)"
+                     " (unsigned long long)wave_lds + (unsigned long long)(in_offs[i]"
+                     " - span_start) : (unsigned long long)((const char*)"
                      "in_data + in_offs[i]);")
             L.append("    unsigned long long _rea = staged ?"
-                     " (unsigned long long)(wave_lds + (in_offs[i+1] -"
-                     " span_start)) : (unsigned long long)((const char*)"
+                     " (unsigned long long)wave_lds + (unsigned long long)(in_offs[i+1]"
+                     " - span_start) : (unsigned long long)((const char*)"
                      "in_data + in_offs[i+1]);")
             L.append('    asm volatile("" : "+v"(_rpa), "+v"(_rea));'
                      "  // opaque: block addrspace re-inference")
@@ -1678,12 +1678,12 @@ class StageCodegen:
         if lds == "gen":
             # u64-opaque select (see csv loader note)
             L.append("    unsigned long long _rpa = staged ?"
-                     " (unsigned long long)(wave_lds + (in_offs[i] -"
-                     " span_start)) : (unsigned long long)((const char*)"
+                     " (unsigned long long)wave_lds + (unsigned long long)(in_offs[i]"
+                     " - span_start) : (unsigned long long)((const char*)"
                      "in_data + in_offs[i]);")
             L.append("    unsigned long long _rea = staged ?"
-                     " (unsigned long long)(wave_lds + (in_offs[i+1] -"
-                     " span_start)) : (unsigned long long)((const char*)"
+                     " (unsigned long long)wave_lds + (unsigned long long)(in_offs[i+1]"
+                     " - span_start) : (unsigned long long)((const char*)"
                      "in_data + in_offs[i+1]);")
             L.append('    asm volatile("" : "+v"(_rpa), "+v"(_rea));'
                      "  // opaque: block addrspace re-inference")
@@ -2309,7 +2309,7 @@ class StageCodegen:
         L.append("    if (active) {")
         L.append("      bool _noq = (keep[i] & 2) != 0;")
         L.append("      unsigned long long _wa = staged ? (unsigned long"
-                 " long)(wave_lds + (my_start - span_start)) : (unsigned"
+                 " long)wave_lds + (unsigned long long)(my_start - span_start) : (unsigned"
                  " long long)((char*)out_data + my_start);")
         L.append('      asm volatile("" : "+v"(_wa));')
         L.append("      char* w = (char*)_wa;")
