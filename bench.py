@@ -98,11 +98,14 @@ def cpu_baseline_leg(body, seconds):
         return None
 
 
-def e2e_leg(header, body, mb, local_rank=0):
+def e2e_leg(header, body, mb, local_rank=0, dist=None, rank=0, world=1):
     """End-to-end file->file tocsv: disk read + H2D + kernels + D2H + disk
     write ALL inside the timed region (SURVEY.md §8d's IO-included number),
     through the product engine (Context.csv -> tocsv), reported beside the
-    resident-kernel headline in the same JSON line."""
+    resident-kernel headline in the same JSON line. At world>1 every rank runs
+    it together on ONE shared input file (total world*mb MB): the engine
+    itself shards chunks across ranks (csvio.run_csv dist path) and each rank
+    writes its own part file — this is the product multi-GPU path, measured."""
     import shutil
     import tuplex_amd
     from tests.test_codegen_compile import zillow_ops
@@ -110,16 +113,18 @@ def e2e_leg(header, body, mb, local_rank=0):
     tdir = "/tmp/tpx_e2e"
     os.makedirs(tdir, exist_ok=True)
     inp = os.path.join(tdir, "in.csv")
-    outp = os.path.join(tdir, "out.csv")
+    outp = os.path.join(tdir, "out.csv" if world == 1 else "out_parts")
     want = mb << 20
     try:
-        with open(inp, "wb") as f:
-            f.write(header)
-            written = 0
-            while written < want:
-                f.write(body)
-                written += len(body)
-        rows = 0
+        if rank == 0:
+            with open(inp, "wb") as f:
+                f.write(header)
+                written = 0
+                while written < want * world:
+                    f.write(body)
+                    written += len(body)
+        if dist:
+            dist.barrier()
         t0 = time.perf_counter()
         ctx = tuplex_amd.Context({"tuplex.gpu.device": str(local_rank)})
         ds = apply_ops(ctx.csv(inp), zillow_ops())
@@ -128,17 +133,27 @@ def e2e_leg(header, body, mb, local_rank=0):
         if ds._last_outcome.mode != "gpu":
             log("e2e leg fell back:", ds._last_outcome.fallback_reason)
             return None
+        wall = t1 - t0
+        if dist:
+            import torch
+            tt = torch.tensor([wall], device="cuda")
+            dist.all_reduce(tt, op=dist.ReduceOp.MAX)
+            wall = float(tt.item())
+        if rank != 0:
+            return None
         with open(inp, "rb") as f:
             f.seek(len(header))
             rows = sum(chunk.count(b"\n")
                        for chunk in iter(lambda: f.read(1 << 24), b""))
-        return {"e2e_rows_per_s": rows / (t1 - t0), "e2e_mb": written >> 20,
-                "e2e_seconds": t1 - t0}
+        return {"e2e_rows_per_s": rows / wall,
+                "e2e_mb": os.path.getsize(inp) >> 20,
+                "e2e_seconds": wall, "e2e_engine_ranks": world}
     except Exception as e:  # noqa: BLE001
         log("e2e leg failed:", e)
         return None
     finally:
-        shutil.rmtree(tdir, ignore_errors=True)
+        if rank == 0:
+            shutil.rmtree(tdir, ignore_errors=True)
 
 
 def parse_pmc_dbs(paths):
@@ -288,6 +303,12 @@ def main():
         total_rows = float(sum(k["in_rows"] for k in kstats))
         total_bytes = float(sum(k["bytes_in"] + k["bytes_out"] for k in kstats))
 
+    # e2e leg is collective at world>1 (engine-path sharding) — every rank
+    # must enter it before non-zero ranks exit
+    e2e = None
+    if not args.no_e2e:
+        e2e = e2e_leg(header, body, args.e2e_mb or args.mb_per_gpu,
+                      local_rank % ndev, dist=dist, rank=rank, world=world)
     if rank != 0:
         return
 
@@ -316,11 +337,6 @@ def main():
     cpu = None
     if not args.no_cpu_baseline and rank == 0 and world <= 1:
         cpu = cpu_baseline_leg(body, args.cpu_seconds)
-    e2e = None
-    if not args.no_e2e and rank == 0 and world <= 1:
-        e2e = e2e_leg(header, body, args.e2e_mb or args.mb_per_gpu,
-                      local_rank % ndev)
-
     line = {
         "metric": "rows/s",
         "value": value,

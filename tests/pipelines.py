@@ -241,6 +241,32 @@ def apply_ops(ds, ops):
     return ds
 
 
+# When set (a list), stage sources are collected instead of compiled so the
+# caller can fan the hipRTC compiles over a thread pool (hipRTC is
+# thread-safe; the hsaco cache publish is atomic rename) — build() uses this.
+_COLLECTOR = None
+
+
+def _stage_compile(glib, src, desc):
+    if _COLLECTOR is not None:
+        _COLLECTOR.append((src, desc))
+        return True
+    return glib.compile_stage(src, desc, compile_only=True)
+
+
+def compile_collected(jobs, workers=None, verbose=False):
+    """hipRTC-compile collected (src, desc) stage jobs in parallel."""
+    import os as _os
+    from concurrent.futures import ThreadPoolExecutor
+    from tuplex_amd import engine
+    glib = engine.GpuLib.get()
+    nw = workers or min(16, _os.cpu_count() or 8)
+    with ThreadPoolExecutor(max_workers=nw) as ex:
+        list(ex.map(lambda j: glib.compile_stage(j[0], j[1],
+                                                 compile_only=True), jobs))
+    return len(jobs)
+
+
 def precompile_extra(verbose=False):
     """Compile-only coverage for the stage shapes outside PIPELINES: joins
     (unique + duplicate keys, i64 + str), columnar (ORC) source, string-key
@@ -286,9 +312,7 @@ def precompile_extra(verbose=False):
         assert sp.compilable, (name, sp.why_not_compilable)
         kw = {"csv_info": {"null_values": [""]}} if source == "csv" else {}
         src, desc = codegen.generate_stage(sp, source=source, sink=sink, **kw)
-        st = glib.lib.tpx_stage_compile(src.encode(), desc.encode(),
-                                        engine._CACHE_DIR.encode(), 1)
-        assert st, (name, glib.err())
+        _stage_compile(glib, src, desc)
         n += 1
         if verbose:
             print("  extra %s ok" % name)
@@ -344,7 +368,7 @@ def precompile_csv(verbose=False):
             src, desc = codegen.generate_stage(
                 sp, source="csv", sink=sink,
                 csv_info={"null_values": [""], "delimiter": delim.decode()})
-            glib.compile_stage(src, desc, compile_only=True)
+            _stage_compile(glib, src, desc)
             n += 1
             if verbose:
                 print("precompiled csv:", name, sink)
@@ -354,7 +378,7 @@ def precompile_csv(verbose=False):
         src_t, desc_t = codegen.generate_stage(
             sp_t, source="csv", sink="mem",
             csv_info={"null_values": [], "text_mode": True})
-        glib.compile_stage(src_t, desc_t, compile_only=True)
+        _stage_compile(glib, src_t, desc_t)
         n += 1
     return n
 
@@ -378,7 +402,7 @@ def precompile_all(verbose=False):
                 print("skip (fallback):", name, sp.why_not_compilable)
             continue
         src, desc = codegen.generate_stage(sp, source="mem", sink="mem")
-        glib.compile_stage(src, desc, compile_only=True)
+        _stage_compile(glib, src, desc)
         n += 1
         if verbose:
             print("precompiled:", name)

@@ -50,3 +50,36 @@ def test_delimiter_sniff():
     assert csvio.sniff_delimiter(b"a|b|c\n1|2|3\n") == b"|"
     assert csvio.sniff_delimiter(b"a,b,c\n") == b","
     assert pyoracle_csv.sniff_delimiter(b"x\ty\tz\n") == b"\t"
+
+
+def test_parallel_resolver_matches_inline(tmp_path):
+    """presolve.ResolverPool replays diverted rows identically to the inline
+    loop (same shrunk results, same order)."""
+    from tuplex_amd import presolve
+    from tuplex_amd import csvio
+    from tuplex_amd import ttypes as T
+
+    def du(x):
+        return (x["a"] * 2, x["b"])
+
+    ops = [("map", du), ("filter", lambda x: x[0] % 3 != 0)]
+    col_types = [T.I64, T.STR]
+    names = ["a", "b"]
+    payloads = []
+    for i in range(300):
+        if i % 7 == 0:
+            payloads.append(b"notint,s%d\n" % i)
+        else:
+            payloads.append(b"%d,s%d\n" % (i, i))
+    inline = [presolve._shrink(
+        csvio.replay_csv_row(p, col_types, [""], ops, names, ","))
+        for p in payloads]
+    pool = presolve.ResolverPool(col_types, [""], ops, names, ",", None,
+                                 False, processes=2)
+    try:
+        got = pool.resolve(payloads)
+    finally:
+        pool.close()
+    assert got == inline
+    assert any(r[0] == "excname" for r in got)
+    assert any(r[0] == "row" for r in got)

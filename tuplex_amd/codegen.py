@@ -1370,12 +1370,23 @@ class StageCodegen:
         L.append("    }")
         L.append("    long long i = r0 + lane;")
         L.append("    if (i >= rhi) continue;")
-        # ONE generic-pointer row body for both the staged (LDS) and overflow
-        # (global) paths: flat loads on LDS addresses were measured as fast as
-        # ds_read for this kernel (profiles/README.md, hot-cell slot ablation),
-        # and a single body halves the kernel size and the ~60 s hipRTC
-        # compile of big stages that the duplicated bodies caused
-        L.extend(self._row_body(in_types, out_types, lds="gen"))
+        # Default: duplicate the row body per pointer mode — in the staged
+        # branch every parse pointer provably derives from LDS, so address
+        # space inference emits ds_read and no generic pointers exist. The
+        # single generic-pointer body (TPX_GEN_BODY=1) halves kernel size and
+        # hipRTC compile time but MISCOMPILES on hipRTC gfx950 (mem-path null
+        # VAs survived three opaque-pointer workarounds; see commits
+        # f347c0f/cee4b86/08e5bef) — kept only for offline investigation.
+        if _os.environ.get("TPX_GEN_BODY") == "1":
+            L.extend(self._row_body(in_types, out_types, lds="gen"))
+        else:
+            L.append("    if (staged) {")
+            L.extend("  " + ln
+                     for ln in self._row_body(in_types, out_types, lds=True))
+            L.append("    } else {")
+            L.extend("  " + ln
+                     for ln in self._row_body(in_types, out_types, lds=False))
+            L.append("    }")
         L.append("  }")
         L.append("}")
         return "\n".join(L)
@@ -2302,20 +2313,17 @@ class StageCodegen:
         L.append("      out_offs[keep_scan[i]] = out_byte0 + my_start;")
         L.append("      out_rowidx[keep_scan[i]] = row0 + i;")
         L.append("    }")
-        # ONE generic-pointer format body for both the staged (LDS) and
-        # overflow (global) paths — flat stores handle either address space;
-        # halves the kernel and its hipRTC compile time
+        # duplicate the format body per pointer mode: in the staged branch w
+        # provably derives from LDS, so addrspace inference emits ds_write
+        # instead of flat stores (same reasoning as tpx_stage_main's row body;
+        # the single generic body miscompiled on hipRTC gfx950)
         body = self._csv_format_body(out_types)
-        L.append("    if (active) {")
-        L.append("      bool _noq = (keep[i] & 2) != 0;")
-        L.append("      unsigned long long _wa = staged ? (unsigned long"
-                 " long)wave_lds + (unsigned long long)(my_start - span_start) : (unsigned"
-                 " long long)((char*)out_data + my_start);")
-        L.append('      asm volatile("" : "+v"(_wa));')
-        L.append("      char* w = (char*)_wa;")
-        L.extend("  " + ln for ln in body)
-        L.append("    }")
         L.append("    if (staged) {")
+        L.append("      if (active) {")
+        L.append("        bool _noq = (keep[i] & 2) != 0;")
+        L.append("        char* w = wave_lds + (my_start - span_start);")
+        L.extend("    " + ln for ln in body)
+        L.append("      }")
         L.append("      __builtin_amdgcn_wave_barrier();")
         # cooperative span copy with ALIGNED global 8B stores (dst + a0 is
         # 8-aligned by construction) and aligned LDS 8B reads + a uniform
@@ -2338,6 +2346,10 @@ class StageCodegen:
         L.append("      }")
         L.append("      long long t0 = span > a0 ? a0 + ((span - a0) & ~7LL) : span;")
         L.append("      if (t0 + lane < span) dst[t0 + lane] = wave_lds[t0 + lane];")
+        L.append("    } else if (active) {")
+        L.append("      bool _noq = (keep[i] & 2) != 0;")
+        L.append("      char* w = (char*)out_data + my_start;")
+        L.extend("  " + ln for ln in body)
         L.append("    }")
         L.append("  }")
         L.append("}")

@@ -21,6 +21,7 @@
 
 #include <algorithm>
 #include <cstdio>
+#include <unistd.h>
 #include <functional>
 #include <cstring>
 #include <fstream>
@@ -222,8 +223,16 @@ extern "C" tpx_stage* tpx_stage_compile(const char* hip_source,
             return nullptr;
         }
         if (!cache_path.empty()) {
-            std::ofstream f(cache_path, std::ios::binary);
-            f.write(st->code.data(), (std::streamsize)st->code.size());
+            // atomic publish: concurrent ranks compile the same stage (the
+            // multi-process engine path) — a torn cache file must never be
+            // observable, so write to a pid-suffixed temp and rename
+            std::string tmp = cache_path + "." + std::to_string(getpid());
+            {
+                std::ofstream f(tmp, std::ios::binary);
+                f.write(st->code.data(), (std::streamsize)st->code.size());
+            }
+            if (rename(tmp.c_str(), cache_path.c_str()) != 0)
+                unlink(tmp.c_str());
         }
     }
     if (flags & 1) return st;  // compile-only (no GPU present)

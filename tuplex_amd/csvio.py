@@ -28,6 +28,7 @@ from typing import List, Optional
 from . import codegen, plan, rowfmt
 from . import ttypes as T
 from . import resolve as _resolve
+from . import presolve
 from .engine import GpuLib, TpxResult, CollectOutcome
 
 
@@ -363,6 +364,82 @@ class _BadParse(Exception):
 _BadParse.__name__ = "BadParseStringInput"
 
 
+# ---- multi-GPU plumbing ----------------------------------------------------------
+#
+# The reference fans ranged tasks over executor threads inside LocalBackend
+# (LocalBackend.cc:491 createLoadAndTransformToMemoryTasks, :552-658 range
+# split). The MI355X-native analog has two shapes, both driven from here:
+#   * one PROCESS per GPU (torchrun; torch.distributed initialized): each rank
+#     takes chunks ci % world == rank and the final merge is a collective
+#     (RCCL over xGMI when the backend is nccl, gloo on CPU);
+#   * one THREAD per GPU inside a single process (plain `Context()` user with
+#     N visible devices): chunks are pulled from a shared queue by per-device
+#     worker threads (hipSetDevice is per-thread; the .so keeps per-device
+#     arenas/streams/modules), results merged in chunk order.
+# Chunk ordering keys are composite (chunk_index << 40) + local_row so a
+# worker never needs the row counts of chunks it didn't process.
+
+_CHUNK_SHIFT = 40
+
+
+def _dist():
+    """torch.distributed if it is initialized (bench/torchrun path), else None.
+    torch is deliberately NOT imported unless the caller runs distributed."""
+    import sys
+    if "torch" not in sys.modules:
+        return None
+    try:
+        import torch.distributed as dist
+        if dist.is_available() and dist.is_initialized():
+            return dist
+    except Exception:
+        pass
+    return None
+
+
+def _gather_objects(dist, obj):
+    """all_gather_object over the active backend; returns list of per-rank
+    objects in rank order."""
+    world = dist.get_world_size()
+    out = [None] * world
+    dist.all_gather_object(out, obj)
+    return out
+
+
+def _run_chunks_parallel(glib, csrc, desc, work, run_one, devices):
+    """Run `run_one(ci, stage)` for every ci in `work`, fanned over one worker
+    thread per device. Each thread pins its device and loads its own module
+    (hipModuleLoadData is per-device). Exceptions propagate."""
+    import queue as _q
+    import threading
+    q = _q.Queue()
+    for ci in work:
+        q.put(ci)
+    errs = []
+
+    def worker(dev):
+        try:
+            glib.lib.tpx_set_device(dev)
+            stage_d = glib.compile_stage(csrc, desc)
+            while True:
+                try:
+                    ci = q.get_nowait()
+                except _q.Empty:
+                    return
+                run_one(ci, stage_d)
+        except BaseException as e:  # noqa: BLE001 - surface to caller
+            errs.append(e)
+
+    threads = [threading.Thread(target=worker, args=(d,), daemon=True)
+               for d in devices]
+    for t in threads:
+        t.start()
+    for t in threads:
+        t.join()
+    if errs:
+        raise errs[0]
+
+
 # ---- main entry ------------------------------------------------------------------
 
 def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
@@ -463,53 +540,84 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
                    "bytes_in": 0, "bytes_out": 0, "chunks": len(chunks) - 1}
     import struct as _s
     import numpy as _np
-    replayed = {}      # global row -> replay output
-    all_rows = []      # (global_idx, row) for mem sink
-    text_parts = []    # (chunk_res snapshot) for csv sink
-    first_row = 0
+    import threading as _thr
+    nch = len(chunks) - 1
+    dist = _dist()
+    rank, world = (dist.get_rank(), dist.get_world_size()) if dist else (0, 1)
+    my_work = [ci for ci in range(nch) if ci % world == rank]
+    # per-chunk result slots, merged in chunk order below
+    chunk_rows = [None] * nch     # mem sink: list[(key,row)]
+    chunk_text = [None] * nch     # csv sink: (text, idxs, offs, lo, hi)
+    chunk_repl = [None] * nch     # {key: [rows]} host-replayed
+    chunk_excs = [None] * nch     # {exc_name: count}
+    mlock = _thr.Lock()
+    resolve_procs = int(opts.get("tuplex.gpu.resolveProcesses", "0") or "0")
+    rpool = [None]                # lazily-started parallel resolver pool
+    rpool_lock = _thr.Lock()
     # zero-copy input pointers: the C side only reads the chunk bytes during
     # the call (it uploads them itself), so point straight into `data`
     data_arr = _np.frombuffer(data, dtype=_np.uint8)
-    for ci in range(len(chunks) - 1):
+
+    def run_one(ci, stage_h):
+        base = ci << _CHUNK_SHIFT  # composite ordering key namespace
         clen = chunks[ci + 1] - chunks[ci]
         cptr = ctypes.cast(data_arr.ctypes.data + chunks[ci],
                            ctypes.POINTER(ctypes.c_uint8))
         res = TpxResult()
         rc = glib.lib.tpx_stage_execute_csv(
-            stage, cptr, clen, first_row, ctypes.byref(res))
+            stage_h, cptr, clen, base, ctypes.byref(res))
         if rc != 0:
             raise RuntimeError("csv stage execute failed: " + glib.err())
         try:
-            for k in ("t_h2d_ms", "t_kernel_ms", "t_d2h_ms"):
-                out.metrics[k] += getattr(res, k)
-            out.metrics["bytes_in"] += res.bytes_in
-            out.metrics["bytes_out"] += res.bytes_out
+            with mlock:
+                for k in ("t_h2d_ms", "t_kernel_ms", "t_d2h_ms"):
+                    out.metrics[k] += getattr(res, k)
+                out.metrics["bytes_in"] += res.bytes_in
+                out.metrics["bytes_out"] += res.bytes_out
+            repl, excs = {}, {}
             if res.exc_num_rows:
                 eb = ctypes.string_at(res.exc_data, res.exc_size)
                 pos = 0
+                rows_keys, payloads = [], []
                 for _ in range(res.exc_num_rows):
                     row, ecode, opid, size = _s.unpack_from("<4q", eb, pos)
-                    payload = eb[pos + 32:pos + 32 + size]
+                    payloads.append(eb[pos + 32:pos + 32 + size])
+                    rows_keys.append(row)
                     pos += 32 + size
-                    if text_mode:
-                        r = replay_text_row(payload, col_types[0],
-                                            src.null_values, logical_ops)
-                    else:
-                        r = replay_csv_row(payload, col_types, src.null_values,
-                                           logical_ops, names, delim.decode(),
-                                           used=sp.used_source_cols)
+                if len(payloads) >= presolve.MIN_POOL_ROWS and \
+                        resolve_procs != 1:
+                    # parallel host resolver (LocalBackend.cc:1254 slow-path
+                    # fan-out analog): replay on a process pool
+                    with rpool_lock:
+                        if rpool[0] is None:
+                            rpool[0] = presolve.ResolverPool(
+                                col_types, src.null_values, logical_ops,
+                                names, delim.decode(), sp.used_source_cols,
+                                text_mode, processes=resolve_procs)
+                    results = rpool[0].resolve(payloads)
+                else:
+                    results = []
+                    for payload in payloads:
+                        if text_mode:
+                            r = replay_text_row(payload, col_types[0],
+                                                src.null_values, logical_ops)
+                        else:
+                            r = replay_csv_row(payload, col_types,
+                                               src.null_values, logical_ops,
+                                               names, delim.decode(),
+                                               used=sp.used_source_cols)
+                        results.append(presolve._shrink(r))
+                for row, r in zip(rows_keys, results):
                     if r[0] == "row":
-                        replayed[row] = [r[1]]
-                    elif r[0] == "exc":
-                        nm = type(r[1]).__name__
-                        out.exception_counts[nm] = \
-                            out.exception_counts.get(nm, 0) + 1
+                        repl[row] = [r[1]]
+                    elif r[0] == "excname":
+                        excs[r[1]] = excs.get(r[1], 0) + 1
                     elif r[0] == "rows":  # 1:N join expansion
-                        replayed[row] = r[1]
-                        for e in r[2]:
-                            nm = type(e).__name__
-                            out.exception_counts[nm] = \
-                                out.exception_counts.get(nm, 0) + 1
+                        repl[row] = r[1]
+                        for nm in r[2]:
+                            excs[nm] = excs.get(nm, 0) + 1
+            chunk_repl[ci] = repl
+            chunk_excs[ci] = excs
             if sink_kind == "mem":
                 out_bytes = ctypes.string_at(res.out_data, res.out_size)
                 rws = rowfmt.deserialize_partition(out_bytes,
@@ -517,7 +625,7 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
                 idxs = _np.ctypeslib.as_array(
                     res.out_row_indices, shape=(res.out_num_rows,)).tolist() \
                     if res.out_num_rows else []
-                all_rows.extend(zip(idxs, rws))
+                chunk_rows[ci] = list(zip(idxs, rws))
             else:
                 # bulk numpy copies (a python list per row costs seconds at
                 # millions of kept rows — the e2e path is host-bound on this)
@@ -529,11 +637,87 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
                     if n_out else _np.empty(0, _np.int64)
                 offs = _np.ctypeslib.as_array(
                     res.out_row_offsets, shape=(n_out + 1,)).copy()
-                text_parts.append((text, idxs, offs,
-                                   first_row, first_row + res.in_num_rows))
-            first_row += res.in_num_rows
+                chunk_text[ci] = (text, idxs, offs, base,
+                                  base + res.in_num_rows)
         finally:
             glib.lib.tpx_result_free(ctypes.byref(res))
+
+    ndev = glib.device_count()
+    want = int(opts.get("tuplex.gpu.devices", "0") or "0")
+    use_devs = min(want or ndev, ndev, max(len(my_work), 1))
+    if world > 1 or use_devs <= 1:
+        # distributed: one device per rank, already pinned above
+        for ci in my_work:
+            run_one(ci, stage)
+    else:
+        dev0 = int(opts.get("tuplex.gpu.device", "0"))
+        devices = [(dev0 + k) % ndev for k in range(use_devs)]
+        _run_chunks_parallel(glib, csrc, desc, my_work, run_one, devices)
+        out.metrics["devices"] = use_devs
+    if rpool[0] is not None:
+        rpool[0].close()
+
+    replayed = {}
+    all_rows = []
+    text_parts = []
+    for ci in range(nch):
+        if chunk_repl[ci]:
+            replayed.update(chunk_repl[ci])
+        for nm, c in (chunk_excs[ci] or {}).items():
+            out.exception_counts[nm] = out.exception_counts.get(nm, 0) + c
+        if chunk_rows[ci]:
+            all_rows.extend(chunk_rows[ci])
+        if chunk_text[ci] is not None:
+            text_parts.append(chunk_text[ci])
+
+    if dist and world > 1:
+        out.metrics["world"] = world
+        if sink_kind == "mem":
+            # collective union of (key,row) lists + replays; every rank folds
+            # the identical union so collect() is replicated (the aggregate
+            # fast path below uses a true RCCL all_reduce instead)
+            agg0 = next((op for op in logical_ops
+                         if op[0] in ("aggregate", "aggregateByKey")), None)
+            # scalar additive aggregate: schema-determined, identical on every
+            # rank (same plan, same source) — safe to branch on collectively
+            scalar_add = (sp.agg_expr is not None and sp.agg_key_idx is None
+                          and not sp.agg_unique and agg0 is not None
+                          and len(sp.gpu_output_types) == 1
+                          and T.deopt(sp.gpu_output_types[0]) in (T.I64, T.F64))
+            if scalar_add:
+                # final combine: all_reduce of the local partial — RCCL over
+                # xGMI when the backend is nccl (device tensor), gloo/CPU
+                # otherwise; same code path either way so the gloo world-2
+                # test covers it (LocalBackend.cc:1180-1207 thread-combine
+                # analog). Replayed rows are data-dependent per rank, so agree
+                # on the path with one flag reduce first; any replay anywhere
+                # falls back to the object gather (host agg_fn fold).
+                import torch
+                is_int = T.deopt(sp.gpu_output_types[0]) == T.I64
+                dev = (torch.device("cuda", torch.cuda.current_device())
+                       if dist.get_backend() == "nccl" else
+                       torch.device("cpu"))
+                flag = torch.tensor([1 if replayed else 0], device=dev)
+                dist.all_reduce(flag)
+                if flag.item() == 0:
+                    local = sum(r[0] for _, r in all_rows)
+                    t = torch.tensor(
+                        [local],
+                        dtype=torch.int64 if is_int else torch.float64,
+                        device=dev)
+                    dist.all_reduce(t)
+                    _, _, agg_fn, initial = agg0
+                    out.rows = [t.item() + initial]
+                    return out
+            parts = _gather_objects(dist, (all_rows, replayed,
+                                           out.exception_counts))
+            all_rows, replayed, out.exception_counts = [], {}, {}
+            for pr, pp, pe in parts:
+                all_rows.extend(pr)
+                replayed.update(pp)
+                for nm, c in pe.items():
+                    out.exception_counts[nm] = \
+                        out.exception_counts.get(nm, 0) + c
 
     # ---- merge across chunks (global row indices; ResolveTask.cc:878 order) ----
     if sink_kind == "mem":
@@ -586,14 +770,36 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
         header_line = _format_csv_row(sp.output_columns or
                                       ["column%d" % i
                                        for i in range(len(sp.output_types))])
-        segs = [header_line]
+        segs = []  # (chunk_key, bytes) — key orders chunks across ranks
         for text, idxs, offs, row_lo, row_hi in text_parts:
+            ci = row_lo >> _CHUNK_SHIFT
             if replayed:
-                segs.append(_merge_csv_segments(text, idxs, offs, replayed,
-                                                row_lo, row_hi))
+                segs.append((ci, _merge_csv_segments(text, idxs, offs,
+                                                     replayed, row_lo,
+                                                     row_hi)))
             else:
-                segs.append(text)
-        _write_csv_output(sink[1], b"".join(segs))
+                segs.append((ci, text))
+        if dist and world > 1:
+            path = sink[1]
+            if path.endswith(".csv"):
+                # single output file: gather segments to rank 0, write in
+                # chunk order (executeInOrder analog across ranks)
+                parts = _gather_objects(dist, segs)
+                if rank == 0:
+                    allsegs = sorted(s for p in parts for s in p)
+                    _write_csv_output(path, header_line +
+                                      b"".join(b for _, b in allsegs))
+                dist.barrier()
+            else:
+                # directory sink: one part file per rank (the reference's
+                # one-part-per-task layout), each with the header
+                os.makedirs(path, exist_ok=True)
+                with open(os.path.join(path, "part%d.csv" % rank), "wb") as f:
+                    f.write(header_line + b"".join(b for _, b in segs))
+                dist.barrier()
+        else:
+            _write_csv_output(sink[1], header_line +
+                              b"".join(b for _, b in segs))
         out.rows = []
     return out
 
@@ -606,9 +812,16 @@ def _run_csv_fallback(out, data, col_types, null_values, logical_ops, names,
     (post-rename/withColumn names), not the source names."""
     out.mode = "fallback"
     out.fallback_reason = why
+    dist = _dist()
+    rank, world = (dist.get_rank(), dist.get_world_size()) if dist else (0, 1)
     rows_out = {}
-    for i, line in enumerate(split_rows(data)):
-        r = replay_csv_row(line, col_types, null_values, logical_ops, names, delim)
+    all_lines = split_rows(data)
+    # rank-sharded like the GPU path (contiguous blocks ~ reference ranges)
+    blk = (len(all_lines) + world - 1) // world if world > 1 else len(all_lines)
+    lo, hi = rank * blk, min((rank + 1) * blk, len(all_lines))
+    for i in range(lo, hi):
+        r = replay_csv_row(all_lines[i], col_types, null_values, logical_ops,
+                           names, delim)
         if r[0] == "row":
             rows_out[i] = [r[1]]
         elif r[0] == "exc":
@@ -619,6 +832,13 @@ def _run_csv_fallback(out, data, col_types, null_values, logical_ops, names,
             for e in r[2]:
                 nm = type(e).__name__
                 out.exception_counts[nm] = out.exception_counts.get(nm, 0) + 1
+    if dist and world > 1:
+        parts = _gather_objects(dist, (rows_out, out.exception_counts))
+        rows_out, out.exception_counts = {}, {}
+        for pr, pe in parts:
+            rows_out.update(pr)
+            for nm, c in pe.items():
+                out.exception_counts[nm] = out.exception_counts.get(nm, 0) + c
     from .engine import finalize_merged, output_columns_of, _unwrap_row
     if out_cols is None:
         out_cols = output_columns_of(names, logical_ops)
@@ -631,7 +851,10 @@ def _run_csv_fallback(out, data, col_types, null_values, logical_ops, names,
                                        range(max((len(_as_tup(v))
                                                   for v in out.rows), default=1))])
         body = b"".join(_format_csv_row(list(_as_tup(v))) for v in out.rows)
-        _write_csv_output(sink[1], header_line + body)
+        if rank == 0:
+            _write_csv_output(sink[1], header_line + body)
+        if dist and world > 1:
+            dist.barrier()
         out.rows = []
     return out
 

@@ -1,0 +1,84 @@
+"""Parallel host resolver — the process-pool analog of the reference's
+executor-thread slow path (LocalBackend.cc:1254 resolveViaSlowPath fans
+ResolveTasks over the WorkQueue). CPython replay is GIL-bound, so diverted
+rows are resolved on a spawn-based process pool instead of threads; the UDF
+chain (closures/lambdas) ships to the workers once via cloudpickle at pool
+init, payload batches stream after.
+
+Engaged only when a chunk diverts more than MIN_POOL_ROWS rows — below that
+the inline loop is faster than the IPC. `tuplex.gpu.resolveProcesses`
+(0 = auto, 1 = inline always) controls the pool width."""
+import multiprocessing as mp
+import os
+
+MIN_POOL_ROWS = 4096
+_BATCH = 2048
+
+# worker-side state, set once by the initializer
+_CTX = None
+
+
+def _init_worker(blob):
+    import cloudpickle
+    global _CTX
+    _CTX = cloudpickle.loads(blob)
+
+
+def _shrink(r):
+    """Reduce a replay result ("row"/"drop"/"exc"/"rows") to a picklable,
+    exception-free shape."""
+    if r[0] == "exc":
+        return ("excname", type(r[1]).__name__)
+    if r[0] == "rows":  # 1:N join expansion
+        return ("rows", r[1], [type(e).__name__ for e in r[2]])
+    return r  # ("row", v) or ("drop",)
+
+
+def _work_csv(payloads):
+    from . import csvio
+    col_types, null_values, logical_ops, names, delim, used, text_mode = _CTX
+    out = []
+    for p in payloads:
+        if text_mode:
+            r = csvio.replay_text_row(p, col_types[0], null_values,
+                                      logical_ops)
+        else:
+            r = csvio.replay_csv_row(p, col_types, null_values, logical_ops,
+                                     names, delim, used=used)
+        out.append(_shrink(r))
+    return out
+
+
+class ResolverPool:
+    """One pool per stage execution; lazily started on first large batch."""
+
+    def __init__(self, col_types, null_values, logical_ops, names, delim,
+                 used, text_mode, processes=0):
+        import cloudpickle
+        self._blob = cloudpickle.dumps(
+            (col_types, null_values, logical_ops, names, delim, used,
+             text_mode))
+        self._nproc = processes or max(2, (os.cpu_count() or 8) // 2)
+        self._pool = None
+
+    def _ensure(self):
+        if self._pool is None:
+            ctx = mp.get_context("spawn")  # no HIP state inherited
+            self._pool = ctx.Pool(self._nproc, initializer=_init_worker,
+                                  initargs=(self._blob,))
+        return self._pool
+
+    def resolve(self, payloads):
+        """payloads: list[bytes] -> list of shrunk results, same order."""
+        pool = self._ensure()
+        batches = [payloads[i:i + _BATCH]
+                   for i in range(0, len(payloads), _BATCH)]
+        outs = pool.map(_work_csv, batches,
+                        chunksize=max(1, len(batches) // (4 * self._nproc)))
+        return [r for b in outs for r in b]
+
+    def close(self):
+        if self._pool is not None:
+            self._pool.terminate()
+            self._pool.join()
+            self._pool = None
