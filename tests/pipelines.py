@@ -254,16 +254,24 @@ def _stage_compile(glib, src, desc):
     return glib.compile_stage(src, desc, compile_only=True)
 
 
-def compile_collected(jobs, workers=None, verbose=False):
-    """hipRTC-compile collected (src, desc) stage jobs in parallel."""
-    import os as _os
-    from concurrent.futures import ThreadPoolExecutor
+def _compile_one_job(job):
     from tuplex_amd import engine
     glib = engine.GpuLib.get()
-    nw = workers or min(16, _os.cpu_count() or 8)
-    with ThreadPoolExecutor(max_workers=nw) as ex:
-        list(ex.map(lambda j: glib.compile_stage(j[0], j[1],
-                                                 compile_only=True), jobs))
+    glib.compile_stage(job[0], job[1], compile_only=True)
+    return True
+
+
+def compile_collected(jobs, workers=None, verbose=False):
+    """hipRTC-compile collected (src, desc) stage jobs in parallel across
+    PROCESSES (hipRTC/comgr serializes compiles behind an internal lock, so
+    threads don't scale; each worker process gets its own hipRTC)."""
+    import os as _os
+    from concurrent.futures import ProcessPoolExecutor
+    import multiprocessing as _mp
+    nw = workers or min(8, _os.cpu_count() or 8)
+    ctx = _mp.get_context("spawn")
+    with ProcessPoolExecutor(max_workers=nw, mp_context=ctx) as ex:
+        list(ex.map(_compile_one_job, jobs))
     return len(jobs)
 
 
