@@ -36,8 +36,12 @@ static unsigned long long rnd(void) {
 }
 
 int main(void) {
-    const char* alpha = "ab,\"\r\n x9\xc3\x00";  // incl. delim/quote/CR/LF/high
-    int alpha_n = 11;
+    // incl. delim/quote/CR/LF/high AND the bytes one above each special
+    // (',' -> '-', '"' -> '#', '\r' -> 0x0e, '|' -> '}'): the classic SWAR
+    // zero-detect falsely flags value+1 bytes after a true match — positional
+    // masks must be exact (caught live on flights ",-471.04")
+    const char* alpha = "ab,\"\r\n x9\xc3\x00-#\x0e|}";
+    int alpha_n = 16;
     char buf[4096 + 128];
     int fails = 0;
     for (int iter = 0; iter < 200000; ++iter) {

@@ -29,21 +29,36 @@ def _make_input(path):
         f.write("".join(lines))
 
 
+def _du_map(x):
+    return (x["a"] * 2, x["b"] + 1, x["name"])
+
+
+def _du_keep(x):
+    return x[0] % 5 != 0
+
+
+def _du_ab(x):
+    return (x["a"], x["b"])
+
+
+def _comb(a, b):
+    return a + b
+
+
+def _sumb(a, x):
+    return a + x[1]
+
+
 def _pipeline(inp, out_csv):
     import tuplex_amd
     conf = {"tuplex.inputSplitSize": "64KB"}  # force many chunks
-    ds = (tuplex_amd.Context(conf).csv(inp)
-          .map(lambda x: (x["a"] * 2, x["b"] + 1, x["name"]))
-          .filter(lambda x: x[0] % 5 != 0))
+    ds = tuplex_amd.Context(conf).csv(inp).map(_du_map).filter(_du_keep)
     rows = ds.collect()
     assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
-    (tuplex_amd.Context(conf).csv(inp)
-     .map(lambda x: (x["a"] * 2, x["b"] + 1, x["name"]))
-     .filter(lambda x: x[0] % 5 != 0)
-     .tocsv(out_csv))
-    agg_ds = (tuplex_amd.Context(conf).csv(inp)
-              .map(lambda x: (x["a"], x["b"]))
-              .aggregate(lambda a, b: a + b, lambda a, x: a + x[1], 0))
+    tuplex_amd.Context(conf).csv(inp).map(_du_map).filter(_du_keep) \
+        .tocsv(out_csv)
+    agg_ds = tuplex_amd.Context(conf).csv(inp).map(_du_ab) \
+        .aggregate(_comb, _sumb, 0)
     agg = agg_ds.collect()
     with open(out_csv, "rb") as f:
         return rows, f.read(), agg

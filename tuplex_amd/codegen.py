@@ -134,6 +134,17 @@ class StageCodegen:
         self.scan_registry = {}   # struct_key(haystack) -> [(op, needle, ci)..]
         self.fusion_groups = fusion_groups  # None = pass 1 (collect)
         self.fused_done = set()   # (scope-id tuple unnecessary; see emit site)
+        # split parse/UDF kernels (round-2 restructure): the CSV cell walk +
+        # typed parse runs in an LDS-staged tpx_stage_parse that writes a
+        # dense columnar "cell park" (typed values; string cells copied to a
+        # per-wave bump-allocated byte park); tpx_stage_main then runs the
+        # UDF chain as a plain grid-stride kernel over coalesced columns at
+        # FULL occupancy (no LDS), where the fused kernel was measured
+        # latency-bound at 10 waves/CU (profiles/README.md SQ analysis).
+        import os as _os
+        self.split = (source == "csv"
+                      and not (csv_info or {}).get("text_mode")
+                      and _os.environ.get("TPX_SPLIT", "1") != "0")
 
     # ---- literals -----------------------------------------------------------
     def lit(self, s: str) -> str:
@@ -1206,6 +1217,13 @@ class StageCodegen:
             return "true"
         if q is False or q is None:
             return None
+        if self.split:
+            # split mode: the parse kernel folded (flags & 9) per string cell
+            # into the per-row dirty mask (cells >= 64 untracked -> scan)
+            if any(i >= 64 for i in q):
+                return None
+            return " && ".join("(((_dirty >> %d) & 1) == 0)" % i
+                               for i in sorted(q))
         # flags bit0 quoted | bit3(8) content-special: both clear == clean
         return " && ".join("((cl%d.flags & 9) == 0)" % i for i in sorted(q))
 
@@ -1265,7 +1283,12 @@ class StageCodegen:
         # generate kernels BEFORE assembling (they may add string literals)
         self.store_types = store_types
         self.full_out_types = out_types
-        main_src = self._main_kernel(in_types, store_types)
+        if self.split:
+            main_src = ("#define TPX_SPAN_CAP %d\n" % self.SPAN_CAP +
+                        self._parse_kernel(in_types) + "\n\n" +
+                        self._main_kernel_split(in_types, store_types))
+        else:
+            main_src = self._main_kernel(in_types, store_types)
         if dup_op is None:
             write_src = self._write_kernel(out_types)
         elif self.sink == "mem":
@@ -1391,6 +1414,204 @@ class StageCodegen:
         L.append("}")
         return "\n".join(L)
 
+    # -- split parse/UDF kernels (csv source) --------------------------------
+    # tpx_stage_parse: the LDS-staged cell walk + typed parse, writing a dense
+    # columnar "cell park": typed values for numeric columns, and string cell
+    # bytes copied into a per-wave bump-allocated byte park (each wave
+    # reserves one contiguous region for its 64 rows, so the UDF kernel's
+    # waves — the same 64-row grouping — read a ~13 KB window per wave).
+    # tpx_stage_main then runs the UDF chain as a plain grid-stride kernel
+    # over the park at full occupancy: the fused kernel was measured
+    # latency-bound (53-60% SQ WAIT_ANY at its LDS-capped 10 waves/CU).
+
+    def _parse_kernel(self, in_types):
+        used = getattr(self.sp, "used_source_cols", None)
+        if used is None:
+            used = set(range(len(in_types)))
+        L = []
+        L.append('extern "C" __global__ void __launch_bounds__(128, 2)'
+                 ' tpx_stage_parse(')
+        L.append("    const unsigned char* __restrict__ in_data,")
+        L.append("    const long long* __restrict__ in_offs,")
+        L.append("    long long n, long long row0,")
+        L.append("    char* __restrict__ strbuf,"
+                 " unsigned long long* __restrict__ str_cursor,")
+        L.append("    void** pv, long long* __restrict__ prc_out,")
+        L.append("    unsigned long long* __restrict__ dirty_out) {")
+        L.append("  __shared__ char smem[2 * TPX_SPAN_CAP + 80];")
+        L.append("  int lane = threadIdx.x & 63;")
+        L.append("  int wid = threadIdx.x >> 6;")
+        L.append("  char* wave_lds = smem + wid * TPX_SPAN_CAP;")
+        L.append("  long long wave_stride = (long long)gridDim.x * (blockDim.x >> 6);")
+        L.append("  long long nwaves = (n + 63) >> 6;")
+        L.append("  for (long long wb = (long long)blockIdx.x * (blockDim.x >> 6) + wid;"
+                 " wb < nwaves; wb += wave_stride) {")
+        L.append("    long long r0 = wb << 6;")
+        L.append("    long long rhi = r0 + 64 < n ? r0 + 64 : n;")
+        L.append("    long long span_start = in_offs[r0] & ~15LL;")
+        L.append("    long long span_end = in_offs[rhi];")
+        L.append("    long long span = span_end - span_start;")
+        L.append("    bool staged = span <= TPX_SPAN_CAP;")
+        L.append("    if (staged) {")
+        L.append("      for (long long k = (long long)lane * 16; k < span; k += 64 * 16) {")
+        L.append("        if (k + 16 <= span)")
+        L.append("          *(uint4*)(wave_lds + k) = *(const uint4*)((const char*)in_data + span_start + k);")
+        L.append("        else")
+        L.append("          for (long long j = k; j < span; ++j)")
+        L.append("            wave_lds[j] = ((const char*)in_data)[span_start + j];")
+        L.append("      }")
+        L.append("    }")
+        # inactive lanes redundantly parse row r0 (always a valid row) so the
+        # whole wave can join the park prefix sum; their stores are masked
+        L.append("    long long i = r0 + lane;")
+        L.append("    bool _act = i < rhi;")
+        L.append("    if (!_act) i = r0;")
+        body_store = self._store_parsed(in_types, used)
+        L.append("    if (staged) {")
+        L.extend("  " + ln for ln in
+                 self._load_inputs_csv(in_types, lds=True) + body_store)
+        L.append("    } else {")
+        L.extend("  " + ln for ln in
+                 self._load_inputs_csv(in_types, lds=False) + body_store)
+        L.append("    }")
+        L.append("  }")
+        L.append("}")
+        return "\n".join(L)
+
+    def _store_parsed(self, in_types, used):
+        """Park one parsed row: wave-cooperative bump reservation of the
+        string park, then typed/value stores. Runs inside the (dual-body)
+        parse kernel with c0..cN, cl0..clN, prc in scope."""
+        str_used = [i for i, t in enumerate(in_types)
+                    if i in used and T.deopt(t) == T.STR]
+        L = []
+        L.append("    long long _sb = 0;")
+        if str_used:
+            L.append("    if (_act && !prc) _sb = 8 %s;"
+                     % "".join(" + c%d.n" % i for i in str_used))
+        # wave-exclusive prefix sum (shfl ladder) + one atomic per wave
+        L.append("    unsigned long long _v = (unsigned long long)_sb;")
+        L.append("    for (int _s = 1; _s < 64; _s <<= 1) {")
+        L.append("      unsigned long long _u = __shfl_up(_v, _s);")
+        L.append("      if (lane >= _s) _v += _u;")
+        L.append("    }")
+        L.append("    unsigned long long _off = _v - (unsigned long long)_sb;")
+        L.append("    unsigned long long _tot = __shfl(_v, 63);")
+        L.append("    unsigned long long _base = 0;")
+        L.append("    if (lane == 0 && _tot) _base = atomicAdd(str_cursor, _tot);")
+        L.append("    _base = __shfl(_base, 0);")
+        L.append("    char* _dst = strbuf + _base + _off;")
+        for idx, t in enumerate(in_types):
+            if idx not in used:
+                continue
+            base = T.deopt(t)
+            opt = T.is_opt(t)
+            if base == T.STR:
+                L.append("    if (_act) {")
+                L.append("      ((unsigned long long*)pv[%d])[i] ="
+                         " (unsigned long long)_dst;" % (3 * idx))
+                L.append("      ((int*)pv[%d])[i] = prc ? 0 : (int)c%d.n;"
+                         % (3 * idx + 1, idx))
+                L.append("      if (!prc) {")
+                L.append("        const char* _s = c%d.p; long long _n = c%d.n;"
+                         % (idx, idx))
+                L.append("        long long _k = 0;")
+                L.append("        for (; _k + 8 <= _n; _k += 8)")
+                L.append("          *(unsigned long long*)(_dst + _k) ="
+                         " *(const unsigned long long*)(_s + _k);")
+                L.append("        for (; _k < _n; ++_k) _dst[_k] = _s[_k];")
+                L.append("        _dst += _n;")
+                L.append("      }")
+                L.append("    }")
+            elif base == T.F64:
+                L.append("    if (_act) ((double*)pv[%d])[i] = c%d;"
+                         % (3 * idx, idx))
+            else:  # I64 / BOOL both park as i64
+                L.append("    if (_act) ((long long*)pv[%d])[i] ="
+                         " (long long)c%d;" % (3 * idx, idx))
+            if opt:
+                L.append("    if (_act) ((unsigned char*)pv[%d])[i] ="
+                         " c%d_n ? 1 : 0;" % (3 * idx + 2, idx))
+        dirty_bits = ["((unsigned long long)((cl%d.flags & 9) != 0) << %d)"
+                      % (i, i) for i in str_used if i < 64]
+        L.append("    if (_act) {")
+        L.append("      prc_out[i] = prc;")
+        L.append("      dirty_out[i] = %s;"
+                 % (" | ".join(dirty_bits) if dirty_bits else "0"))
+        L.append("    }")
+        return L
+
+    def _main_kernel_split(self, in_types, out_types):
+        """UDF-phase kernel over the parsed cell park: plain grid-stride, no
+        LDS, full occupancy. Same fixed signature as the fused main (in_data
+        carries the park slot table; in_offs still indexes the raw chunk
+        bytes for exception payload ranges)."""
+        L = []
+        L.append('extern "C" __global__ void tpx_stage_main(')
+        L.append("    const unsigned char* __restrict__ in_data,")
+        L.append("    const long long* __restrict__ in_offs,")
+        L.append("    long long n, long long row0,")
+        L.append("    char* heap_base, unsigned long long* heap_cursor,"
+                 " unsigned long long heap_cap,")
+        L.append("    unsigned char* __restrict__ keep, long long* __restrict__ keep01,")
+        L.append("    long long* __restrict__ sizes,")
+        L.append("    long long* exc_buf, unsigned long long* exc_count,"
+                 " unsigned long long exc_cap,")
+        L.append("    void** outv) {")
+        L.append("  TpxHeap heap{heap_base, heap_cursor, heap_cap, nullptr, nullptr};")
+        L.append("  long long stride = (long long)gridDim.x * blockDim.x;")
+        L.append("  for (long long i = (long long)blockIdx.x * blockDim.x +"
+                 " threadIdx.x; i < n; i += stride) {")
+        L.extend(self._row_body(in_types, out_types, lds="park"))
+        L.append("  }")
+        L.append("}")
+        return "\n".join(L)
+
+    def _load_parsed(self, in_types):
+        """Row loader for the UDF-phase kernel: typed coalesced loads from the
+        cell park (strings are absolute (ptr,len) views into the byte park)."""
+        used = getattr(self.sp, "used_source_cols", None)
+        nin = len(in_types)
+        L = []
+        L.append("    const void* const* ct = (const void* const*)in_data;")
+        L.append("    long long prc = ((const long long*)ct[%d])[i];" % (3 * nin))
+        L.append("    unsigned long long _dirty ="
+                 " ((const unsigned long long*)ct[%d])[i];" % (3 * nin + 1))
+        L.append("    (void)_dirty;")
+        for idx, t in enumerate(in_types):
+            base = T.deopt(t)
+            opt = T.is_opt(t)
+            if used is not None and idx not in used:
+                if opt:
+                    L.append("    bool c%d_n = false;  // unused (pushdown)" % idx)
+                if base == T.STR:
+                    L.append("    tstr c%d{(const char*)in_data, 0};" % idx)
+                elif base == T.I64:
+                    L.append("    long long c%d = 0;" % idx)
+                elif base == T.F64:
+                    L.append("    double c%d = 0.0;" % idx)
+                else:
+                    L.append("    bool c%d = false;" % idx)
+                continue
+            if opt:
+                L.append("    bool c%d_n = ((const unsigned char*)ct[%d])[i]"
+                         " != 0;" % (idx, 3 * idx + 2))
+            if base == T.STR:
+                L.append("    tstr c%d{(const char*)"
+                         "((const unsigned long long*)ct[%d])[i],"
+                         " (long long)((const int*)ct[%d])[i]};"
+                         % (idx, 3 * idx, 3 * idx + 1))
+            elif base == T.I64:
+                L.append("    long long c%d = ((const long long*)ct[%d])[i];"
+                         % (idx, 3 * idx))
+            elif base == T.F64:
+                L.append("    double c%d = ((const double*)ct[%d])[i];"
+                         % (idx, 3 * idx))
+            else:
+                L.append("    bool c%d = ((const long long*)ct[%d])[i] != 0;"
+                         % (idx, 3 * idx))
+        return L
+
     def _main_kernel_col(self, in_types, out_types):
         """Columnar-source main kernel: typed coalesced loads, no LDS staging,
         plain grid-stride over rows (same fixed signature; in_offs unused)."""
@@ -1417,7 +1638,9 @@ class StageCodegen:
 
     def _row_body(self, in_types, out_types, lds):
         L = []
-        if self.source == "csv":
+        if lds == "park":
+            L.extend(self._load_parsed(in_types))
+        elif self.source == "csv":
             L.extend(self._load_inputs_csv(in_types, lds))
         elif self.source == "col":
             L.extend(self._load_inputs_col(in_types))
@@ -1755,7 +1978,7 @@ class StageCodegen:
                     L.append("    if (staged) o.o%d = tpx_to_global(o.o%d,"
                              " wave_lds, wave_lds + TPX_SPAN_CAP, in_data,"
                              " span_start);" % (k, k))
-                elif lds:
+                elif lds and lds != "park":
                     L.append("    o.o%d = tpx_to_global(o.o%d, wave_lds,"
                              " wave_lds + TPX_SPAN_CAP, in_data, span_start);"
                              % (k, k))
@@ -2408,6 +2631,12 @@ class StageCodegen:
                 lines.append("agg=%s" % T.deopt(self.sp.agg_type))
         if self.csv_info.get("text_mode"):
             lines.append("textmode=1")
+        if self.split:
+            lines.append("split=1")
+            used = getattr(self.sp, "used_source_cols", None)
+            if used is None:
+                used = range(len(in_types))
+            lines.append("used=%s" % ",".join(str(i) for i in sorted(used)))
         for i, t in enumerate(in_types):
             lines.append("in%d=%s" % (i, tdesc(t)))
         for i, t in enumerate(out_types):

@@ -87,7 +87,13 @@ __device__ __forceinline__ bool tpx_is_ascii_byte(unsigned char c) { return c < 
 
 // zero-byte detector: nonzero bits at 0x80 positions of bytes equal to zero
 __device__ __forceinline__ unsigned long long tpx_swar_zero(unsigned long long x) {
-    return (x - TPX_SWAR_ONE) & ~x & TPX_SWAR_HIGH;
+    // EXACT per-byte zero detect: (x|HIGH) - ONE never borrows across bytes,
+    // so every byte's high bit is independent. The classic
+    // (x-ONE) & ~x & HIGH is only first-match exact: a borrow from a true
+    // zero byte falsely flags a following 0x01 byte (ex: scanning for ','
+    // flags the '-' of ",-471.04" — cell walk and quote-parity masks use
+    // these bits POSITIONALLY, so they must be exact).
+    return ~(x | ((x | TPX_SWAR_HIGH) - TPX_SWAR_ONE)) & TPX_SWAR_HIGH;
 }
 
 // first index of byte c in p[0..n), or -1. Head byte loop to 8-alignment,
