@@ -79,6 +79,9 @@ struct StageDesc {
     std::string aggby;   // "" | i64 | f64 — by-key hash-reduce (key col 0)
     int aggkeystr = 0;   // by-key key column is a string
     int textmode = 0;    // text() source: rows split on every newline (no quotes)
+    int split = 0;       // split parse/UDF kernels (csv): k_parse fills the
+                         // cell park, k_main runs grid-stride over it
+    std::vector<int> used;  // park columns (projection pushdown)
     std::vector<ColDesc> in_cols, out_cols;
 };
 
@@ -109,6 +112,13 @@ static bool parse_desc(const char* text, StageDesc* d) {
     d->aggby = kv.count("aggby") ? kv["aggby"] : "";
     d->aggkeystr = kv.count("aggkeystr") ? atoi(kv["aggkeystr"].c_str()) : 0;
     d->textmode = kv.count("textmode") ? atoi(kv["textmode"].c_str()) : 0;
+    d->split = kv.count("split") ? atoi(kv["split"].c_str()) : 0;
+    if (kv.count("used")) {
+        std::istringstream us(kv["used"]);
+        std::string tok;
+        while (std::getline(us, tok, ','))
+            if (!tok.empty()) d->used.push_back(atoi(tok.c_str()));
+    }
     int nin = atoi(kv["nin"].c_str());
     int nout = atoi(kv["nout"].c_str());
     for (int i = 0; i < nin; ++i) {
@@ -137,6 +147,7 @@ struct tpx_stage {
     std::vector<char> code;  // hsaco
     hipModule_t module = nullptr;
     hipFunction_t k_main = nullptr, k_write = nullptr;
+    hipFunction_t k_parse = nullptr;  // split mode (desc.split)
     hipFunction_t k_scan_block = nullptr, k_scan_add = nullptr;
     hipFunction_t k_pair_total = nullptr;
     hipFunction_t k_csv_chunk = nullptr, k_csv_sel = nullptr, k_csv_rows = nullptr;
@@ -244,6 +255,7 @@ extern "C" tpx_stage* tpx_stage_compile(const char* hip_source,
     }
     struct { const char* name; hipFunction_t* fn; bool required; } lut[] = {
         {"tpx_stage_main", &st->k_main, true},
+        {"tpx_stage_parse", &st->k_parse, false},
         {"tpx_stage_write", &st->k_write, true},
         {"tpx_scan_block", &st->k_scan_block, true},
         {"tpx_scan_add", &st->k_scan_add, true},
@@ -612,6 +624,64 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
             d_in_c[c] = (char*)blk + ns * sizeof(void*) * (size_t)(c - 1);
     }
 
+    // split mode (D.split): allocate the cell park — typed value arrays per
+    // used input column, a string byte park with one bump cursor, and per-row
+    // prc/dirty arrays; the k_main table gets them as slots [3*nin, 3*nin+1].
+    bool split = D.split && d_offs && st->k_parse != nullptr;
+    void* d_park_c[8] = {nullptr};
+    void* d_strbuf = nullptr;
+    void* d_strcur = nullptr;
+    void* d_prc = nullptr;
+    void* d_dirty = nullptr;
+    if (split) {
+        int nin = (int)D.in_cols.size();
+        std::vector<void*> parkv((size_t)nin * 3 + 2, nullptr);
+        for (int k : D.used) {
+            const ColDesc& col = D.in_cols[(size_t)k];
+            parkv[(size_t)k * 3] = g_arena[dev].take((size_t)n * 8);
+            if (!parkv[(size_t)k * 3]) return -1;
+            if (col.kind == K_STR) {
+                parkv[(size_t)k * 3 + 1] = g_arena[dev].take((size_t)n * 4);
+                if (!parkv[(size_t)k * 3 + 1]) return -1;
+            }
+            if (col.opt) {
+                parkv[(size_t)k * 3 + 2] = g_arena[dev].take((size_t)n);
+                if (!parkv[(size_t)k * 3 + 2]) return -1;
+            }
+        }
+        d_prc = g_arena[dev].take((size_t)n * 8);
+        d_dirty = g_arena[dev].take((size_t)n * 8);
+        // park cells <= raw row bytes; +8 pad per row + SWAR overread slack
+        d_strbuf = g_arena[dev].take((size_t)in_bytes + (size_t)n * 8 + 4096);
+        d_strcur = g_arena[dev].take(256);
+        if (!d_prc || !d_dirty || !d_strbuf || !d_strcur) return -1;
+        HIP_CHECK(hipMemsetAsync(d_strcur, 0, 8, stream));
+        parkv[(size_t)nin * 3] = d_prc;
+        parkv[(size_t)nin * 3 + 1] = d_dirty;
+        std::vector<void*> pall(parkv.size() * (size_t)C);
+        for (int c = 0; c < C; ++c) {
+            void** tv = pall.data() + parkv.size() * (size_t)c;
+            for (int k = 0; k < nin; ++k) {
+                char* v0 = (char*)parkv[(size_t)k * 3];
+                char* v1 = (char*)parkv[(size_t)k * 3 + 1];
+                char* v2 = (char*)parkv[(size_t)k * 3 + 2];
+                tv[k * 3] = v0 ? v0 + cstart[c] * 8 : nullptr;
+                tv[k * 3 + 1] = v1 ? v1 + cstart[c] * 4 : nullptr;
+                tv[k * 3 + 2] = v2 ? v2 + cstart[c] : nullptr;
+            }
+            tv[nin * 3] = (char*)d_prc + cstart[c] * 8;
+            tv[nin * 3 + 1] = (char*)d_dirty + cstart[c] * 8;
+        }
+        void* blk = g_arena[dev].take(pall.size() * sizeof(void*) + 8);
+        if (!blk) return -1;
+        HIP_CHECK(hipMemcpyAsync(blk, pall.data(),
+                                 pall.size() * sizeof(void*),
+                                 hipMemcpyHostToDevice, stream));
+        for (int c = 0; c < C; ++c)
+            d_park_c[c] = (char*)blk +
+                          parkv.size() * sizeof(void*) * (size_t)c;
+    }
+
     // +thread-chunk slack per LAUNCH: <=2048 blocks x 256 threads x 256 B
     unsigned long long heap_cap =
         (unsigned long long)std::max<long long>(in_bytes + (in_bytes >> 1) +
@@ -658,10 +728,28 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
             void* keep_c = (char*)d_keep + cstart[c];
             void* keep01_c = (char*)d_keep01 + cstart[c] * 8;
             void* sizes_c = (char*)d_sizes + cstart[c] * 8;
+            if (split) {
+                if (attempt == 0) {  // park is attempt-invariant
+                    void* prc_c = (char*)d_prc + cstart[c] * 8;
+                    void* dirty_c = (char*)d_dirty + cstart[c] * 8;
+                    void* pargs[] = {&d_in, &in_offs_c, &nc, &row0_c,
+                                     &d_strbuf, &d_strcur, &d_park_c[c],
+                                     &prc_c, &dirty_c};
+                    if (launch(st->k_parse, grid, 128, sc, pargs)) return -1;
+                }
+                d_in_use = d_park_c[c];
+            }
             void* args[] = {&d_in_use, &in_offs_c, &nc, &row0_c, &heap_c,
                             &cursor_c, &hs, &keep_c, &keep01_c, &sizes_c,
                             &P.exc, &d_exc_count, &exc_cap, &d_outv_c[c]};
-            if (launch(st->k_main, grid, 128, sc, args)) return -1;
+            if (split) {
+                unsigned g2 = (unsigned)std::min<long long>(
+                    (nc + 255) / 256, 4096);
+                if (g2 < 1) g2 = 1;
+                if (launch(st->k_main, g2, 256, sc, args)) return -1;
+            } else {
+                if (launch(st->k_main, grid, 128, sc, args)) return -1;
+            }
             hipEventRecord(cev[(size_t)c * 5], sc);
         }
         // C>1 (non-agg sink): pipelined per-chunk scans + writes launched
