@@ -145,6 +145,12 @@ class StageCodegen:
         self.split = (source == "csv"
                       and not (csv_info or {}).get("text_mode")
                       and _os.environ.get("TPX_SPLIT", "1") != "0")
+        # park copy mode: 0 (default) = zero-copy park — string cells stay
+        # (ptr,len) views into the device input bytes (the parse kernel's
+        # passthrough cost was measured at 2.3 ms/GB; byte-copying cells into
+        # a dense park cost more than the UDF kernel's occupancy win
+        # returned); 1 = dense byte park (kept for experiments)
+        self.park_copy = _os.environ.get("TPX_PARK_COPY", "0") == "1"
 
     # ---- literals -----------------------------------------------------------
     def lit(self, s: str) -> str:
@@ -1479,11 +1485,13 @@ class StageCodegen:
         return "\n".join(L)
 
     def _store_parsed(self, in_types, used):
-        """Park one parsed row: wave-cooperative bump reservation of the
-        string park, then typed/value stores. Runs inside the (dual-body)
-        parse kernel with c0..cN, cl0..clN, prc in scope."""
+        """Park one parsed row: typed/value stores (+ wave-cooperative bump
+        reservation of the byte park in park_copy mode). Runs inside the
+        (dual-body) parse kernel with c0..cN, cl0..clN, prc in scope."""
         str_used = [i for i, t in enumerate(in_types)
                     if i in used and T.deopt(t) == T.STR]
+        if not self.park_copy:
+            return self._store_parsed_nocopy(in_types, used, str_used)
         L = []
         L.append("    long long _sb = 0;")
         if str_used:
@@ -1522,6 +1530,46 @@ class StageCodegen:
                 L.append("        for (; _k < _n; ++_k) _dst[_k] = _s[_k];")
                 L.append("        _dst += _n;")
                 L.append("      }")
+                L.append("    }")
+            elif base == T.F64:
+                L.append("    if (_act) ((double*)pv[%d])[i] = c%d;"
+                         % (3 * idx, idx))
+            else:  # I64 / BOOL both park as i64
+                L.append("    if (_act) ((long long*)pv[%d])[i] ="
+                         " (long long)c%d;" % (3 * idx, idx))
+            if opt:
+                L.append("    if (_act) ((unsigned char*)pv[%d])[i] ="
+                         " c%d_n ? 1 : 0;" % (3 * idx + 2, idx))
+        dirty_bits = ["((unsigned long long)((cl%d.flags & 9) != 0) << %d)"
+                      % (i, i) for i in str_used if i < 64]
+        L.append("    if (_act) {")
+        L.append("      prc_out[i] = prc;")
+        L.append("      dirty_out[i] = %s;"
+                 % (" | ".join(dirty_bits) if dirty_bits else "0"))
+        L.append("    }")
+        return L
+
+    def _store_parsed_nocopy(self, in_types, used, str_used):
+        """Zero-copy park: string cells stored as (global ptr, len) views into
+        the device input bytes — staged-branch LDS pointers translate via
+        tpx_to_global. No byte park, no cursor."""
+        L = []
+        for idx, t in enumerate(in_types):
+            if idx not in used:
+                continue
+            base = T.deopt(t)
+            opt = T.is_opt(t)
+            if base == T.STR:
+                L.append("    if (_act) {")
+                L.append("      tstr _g%d = c%d;" % (idx, idx))
+                if True:  # staged-branch pointers live in LDS
+                    L.append("      _g%d = tpx_to_global(_g%d, wave_lds,"
+                             " wave_lds + TPX_SPAN_CAP, in_data,"
+                             " span_start);" % (idx, idx))
+                L.append("      ((unsigned long long*)pv[%d])[i] ="
+                         " (unsigned long long)_g%d.p;" % (3 * idx, idx))
+                L.append("      ((int*)pv[%d])[i] = prc ? 0 : (int)_g%d.n;"
+                         % (3 * idx + 1, idx))
                 L.append("    }")
             elif base == T.F64:
                 L.append("    if (_act) ((double*)pv[%d])[i] = c%d;"
@@ -2633,6 +2681,8 @@ class StageCodegen:
             lines.append("textmode=1")
         if self.split:
             lines.append("split=1")
+            if self.park_copy:
+                lines.append("parkcopy=1")
             used = getattr(self.sp, "used_source_cols", None)
             if used is None:
                 used = range(len(in_types))

@@ -81,6 +81,7 @@ struct StageDesc {
     int textmode = 0;    // text() source: rows split on every newline (no quotes)
     int split = 0;       // split parse/UDF kernels (csv): k_parse fills the
                          // cell park, k_main runs grid-stride over it
+    int parkcopy = 0;    // 1 = dense byte park (default: zero-copy views)
     std::vector<int> used;  // park columns (projection pushdown)
     std::vector<ColDesc> in_cols, out_cols;
 };
@@ -113,6 +114,7 @@ static bool parse_desc(const char* text, StageDesc* d) {
     d->aggkeystr = kv.count("aggkeystr") ? atoi(kv["aggkeystr"].c_str()) : 0;
     d->textmode = kv.count("textmode") ? atoi(kv["textmode"].c_str()) : 0;
     d->split = kv.count("split") ? atoi(kv["split"].c_str()) : 0;
+    d->parkcopy = kv.count("parkcopy") ? atoi(kv["parkcopy"].c_str()) : 0;
     if (kv.count("used")) {
         std::istringstream us(kv["used"]);
         std::string tok;
@@ -651,11 +653,15 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
         }
         d_prc = g_arena[dev].take((size_t)n * 8);
         d_dirty = g_arena[dev].take((size_t)n * 8);
-        // park cells <= raw row bytes; +8 pad per row + SWAR overread slack
-        d_strbuf = g_arena[dev].take((size_t)in_bytes + (size_t)n * 8 + 4096);
-        d_strcur = g_arena[dev].take(256);
-        if (!d_prc || !d_dirty || !d_strbuf || !d_strcur) return -1;
-        HIP_CHECK(hipMemsetAsync(d_strcur, 0, 8, stream));
+        if (!d_prc || !d_dirty) return -1;
+        if (D.parkcopy) {
+            // park cells <= raw row bytes; +8/row pad + SWAR overread slack
+            d_strbuf = g_arena[dev].take((size_t)in_bytes + (size_t)n * 8 +
+                                         4096);
+            d_strcur = g_arena[dev].take(256);
+            if (!d_strbuf || !d_strcur) return -1;
+            HIP_CHECK(hipMemsetAsync(d_strcur, 0, 8, stream));
+        }
         parkv[(size_t)nin * 3] = d_prc;
         parkv[(size_t)nin * 3 + 1] = d_dirty;
         std::vector<void*> pall(parkv.size() * (size_t)C);

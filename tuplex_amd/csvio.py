@@ -330,23 +330,40 @@ def replay_text_row(raw_line: bytes, col_type, null_values, logical_ops):
     return _resolve.replay_row(v, logical_ops, None, scalar_input=True)
 
 
-def split_points(data: bytes, target: int):
+def split_points(data, target: int):
     """Row-aligned chunk boundaries every ~target bytes with EXACT quote-parity
     (the reference's findLineStart purpose, CSVUtils.cc:1494: a ranged chunk must
-    start at a true row start; quoted newlines must not split a row)."""
-    points = [0]
-    pos = target
-    counted_to = 0
-    parity = 0
+    start at a true row start; quoted newlines must not split a row).
+
+    Quote parity is established from per-block quote counts computed with
+    numpy (SIMD, ~10 GB/s) instead of bytes.count over ever-growing ranges —
+    the old form cost ~1.3 s/GB on the e2e file->file path. `data` may be any
+    buffer (bytes or mmap)."""
+    import numpy as _np
     n = len(data)
+    points = [0]
+    if n == 0:
+        return [0, 0]
+    arr = _np.frombuffer(data, dtype=_np.uint8)
+    B = 1 << 22
+    nb = (n + B - 1) // B
+    qc = [0] * (nb + 1)  # qc[b] = quotes in data[:b*B]
+    total_q = 0
+    for b in range(nb):
+        total_q += int(_np.count_nonzero(arr[b * B:(b + 1) * B] == 34))
+        qc[b + 1] = total_q
+
+    def parity_before(p):
+        b = p >> 22
+        return (qc[b] + int(_np.count_nonzero(arr[b << 22:p] == 34))) & 1
+
+    find = data.find
+    pos = target
     while pos < n:
-        nl = data.find(b"\n", pos)
-        while nl >= 0:
-            parity = (parity + data.count(b'"', counted_to, nl)) & 1
-            counted_to = nl
-            if parity == 0:
-                break
-            nl = data.find(b"\n", nl + 1)
+        nl = find(b"\n", pos)
+        if total_q:
+            while nl >= 0 and parity_before(nl):
+                nl = find(b"\n", nl + 1)
         if nl < 0:
             break
         points.append(nl + 1)
@@ -474,7 +491,7 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
         delim = b","  # unused in text mode
     else:
         # sniff on the first file's sample (FileInputOperator.cc:78 semantics)
-        sample = blobs[0][:max(256 << 10, 1 << 20)]
+        sample = bytes(blobs[0][:256 << 10])
         # cut sample at the last complete row
         nl = sample.rfind(b"\n")
         if nl >= 0:
@@ -532,18 +549,27 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
     split = (opts.input_split_size if opts.is_set("tuplex.inputSplitSize")
              else opts.gpu_input_split_size)
     split = max(split, 64 << 10)
-    chunks = (split_points(data, split) if len(data) > split * 2
-              else [0, len(data)])
+
+    import struct as _s
+    import numpy as _np
+    import threading as _thr
+    dist = _dist()
+    rank, world = (dist.get_rank(), dist.get_world_size()) if dist else (0, 1)
+    ndev = glib.device_count()
+    want = int(opts.get("tuplex.gpu.devices", "0") or "0")
+    from .options import parse_size
+    # resident fast path (1 rank, 1 device): skip host chunking entirely —
+    # one H2D, one execute; the boundary scan runs device-side
+    use_resident = (world == 1 and min(want or ndev, ndev) <= 1 and
+                    len(data) <= parse_size(
+                        opts.get("tuplex.gpu.residentMaxSize", "24GB")))
+    chunks = ([0, len(data)] if use_resident or len(data) <= split * 2
+              else split_points(data, split))
 
     out.mode = "gpu"
     out.metrics = {"t_h2d_ms": 0.0, "t_kernel_ms": 0.0, "t_d2h_ms": 0.0,
                    "bytes_in": 0, "bytes_out": 0, "chunks": len(chunks) - 1}
-    import struct as _s
-    import numpy as _np
-    import threading as _thr
     nch = len(chunks) - 1
-    dist = _dist()
-    rank, world = (dist.get_rank(), dist.get_world_size()) if dist else (0, 1)
     my_work = [ci for ci in range(nch) if ci % world == rank]
     # per-chunk result slots, merged in chunk order below
     chunk_rows = [None] * nch     # mem sink: list[(key,row)]
@@ -561,11 +587,15 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
     def run_one(ci, stage_h):
         base = ci << _CHUNK_SHIFT  # composite ordering key namespace
         clen = chunks[ci + 1] - chunks[ci]
-        cptr = ctypes.cast(data_arr.ctypes.data + chunks[ci],
-                           ctypes.POINTER(ctypes.c_uint8))
         res = TpxResult()
-        rc = glib.lib.tpx_stage_execute_csv(
-            stage_h, cptr, clen, base, ctypes.byref(res))
+        if resident_dev:
+            rc = glib.lib.tpx_stage_execute_csv_dev(
+                stage_h, resident_dev, clen, base, 0, ctypes.byref(res))
+        else:
+            cptr = ctypes.cast(data_arr.ctypes.data + chunks[ci],
+                               ctypes.POINTER(ctypes.c_uint8))
+            rc = glib.lib.tpx_stage_execute_csv(
+                stage_h, cptr, clen, base, ctypes.byref(res))
         if rc != 0:
             raise RuntimeError("csv stage execute failed: " + glib.err())
         try:
@@ -642,10 +672,26 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
         finally:
             glib.lib.tpx_result_free(ctypes.byref(res))
 
-    ndev = glib.device_count()
-    want = int(opts.get("tuplex.gpu.devices", "0") or "0")
     use_devs = min(want or ndev, ndev, max(len(my_work), 1))
-    if world > 1 or use_devs <= 1:
+    resident_dev = 0
+    if use_resident:
+        resident_dev = glib.lib.tpx_dev_alloc(len(data))
+        if resident_dev:
+            import time as _time
+            _t0 = _time.perf_counter()
+            rc = glib.lib.tpx_dev_upload(
+                resident_dev, data_arr.ctypes.data_as(ctypes.c_void_p),
+                len(data))
+            out.metrics["t_h2d_ms"] += (_time.perf_counter() - _t0) * 1e3
+            if rc != 0:
+                glib.lib.tpx_dev_free(resident_dev)
+                resident_dev = 0
+    if resident_dev:
+        try:
+            run_one(0, stage)
+        finally:
+            glib.lib.tpx_dev_free(resident_dev)
+    elif world > 1 or use_devs <= 1:
         # distributed: one device per rank, already pinned above
         for ci in my_work:
             run_one(ci, stage)
