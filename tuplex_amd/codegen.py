@@ -1594,8 +1594,14 @@ class StageCodegen:
         LDS, full occupancy. Same fixed signature as the fused main (in_data
         carries the park slot table; in_offs still indexes the raw chunk
         bytes for exception payload ranges)."""
+        import os as _os
+        # waves/SIMD bound: without it the compiler targets max occupancy at
+        # 64 VGPRs and spills ~272 B/thread (measured); 4 waves/SIMD buys a
+        # 128-VGPR budget at 16 waves/CU (fused kernel needed 104)
+        lb = int(_os.environ.get("TPX_SPLIT_LB", "4"))
+        bound = " __launch_bounds__(256, %d)" % lb if lb else ""
         L = []
-        L.append('extern "C" __global__ void tpx_stage_main(')
+        L.append('extern "C" __global__ void%s tpx_stage_main(' % bound)
         L.append("    const unsigned char* __restrict__ in_data,")
         L.append("    const long long* __restrict__ in_offs,")
         L.append("    long long n, long long row0,")
