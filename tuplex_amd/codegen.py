@@ -1595,10 +1595,11 @@ class StageCodegen:
         carries the park slot table; in_offs still indexes the raw chunk
         bytes for exception payload ranges)."""
         import os as _os
-        # waves/SIMD bound: without it the compiler targets max occupancy at
-        # 64 VGPRs and spills ~272 B/thread (measured); 4 waves/SIMD buys a
-        # 128-VGPR budget at 16 waves/CU (fused kernel needed 104)
-        lb = int(_os.environ.get("TPX_SPLIT_LB", "4"))
+        # waves/SIMD bound: the UDF-phase kernel genuinely needs ~198 VGPRs
+        # (whole UDF chain + csv-size logic live ranges); bounding tighter
+        # (128 at lb=4) spilled 272 B/thread and lost more than the
+        # occupancy won. lb=2 -> no spills at 8 waves/CU.
+        lb = int(_os.environ.get("TPX_SPLIT_LB", "2"))
         bound = " __launch_bounds__(256, %d)" % lb if lb else ""
         L = []
         L.append('extern "C" __global__ void%s tpx_stage_main(' % bound)

@@ -262,11 +262,19 @@ def execute_stage_col(sp: plan.StageProgram, dev_slots, n_rows: int,
 
 
 def run_collect(data: List[Any], logical_ops: List[tuple],
-                columns: Optional[List[str]], options) -> "CollectOutcome":
+                columns: Optional[List[str]], options,
+                keep_exceptions=False, keep_keys=False) -> "CollectOutcome":
     """Full dual-mode collect: majority-type inference, GPU normal case,
     interpreter replay of fallback + exception rows, in-order merge
-    (mirrors LocalBackend.cc:963-1085 + ResolveTask.cc:878)."""
+    (mirrors LocalBackend.cc:963-1085 + ResolveTask.cc:878).
+
+    keep_exceptions: rows that stay exceptions are STORED on the outcome
+    (out.pending + out.pending_replayer) instead of counted — the cache()
+    path (CacheOperator stores exceptions for later resolution). Implies
+    keep_keys. keep_keys: out.row_keys aligns each output row with its
+    original input index (ordered-merge key)."""
     out = CollectOutcome()
+    keep_keys = keep_keys or keep_exceptions
     maj = T.infer_majority_type(data, optional_threshold=options.optional_threshold)
     row_maj = T.row_type_of(maj)
     scalar_input = not T.is_tuple(maj)
@@ -358,6 +366,9 @@ def run_collect(data: List[Any], logical_ops: List[tuple],
         if r[0] == "row":
             results[i] = [r[1]]
         elif r[0] == "exc":
+            if keep_exceptions:
+                out.pending.append((i, v))
+                continue
             name = type(r[1]).__name__
             out.exception_counts[name] = out.exception_counts.get(name, 0) + 1
         elif r[0] == "rows":  # 1:N join expansion
@@ -368,7 +379,77 @@ def run_collect(data: List[Any], logical_ops: List[tuple],
                     out.exception_counts.get(name, 0) + 1
 
     merged = [v for i in sorted(results) for v in results[i]]
+    if keep_keys:
+        out.row_keys = [i for i in sorted(results) for _ in results[i]]
+    if keep_exceptions:
+        def _mem_replayer(value, ops, used=None):
+            return resolve.replay_row(value, ops, columns, scalar_input)
+        out.pending_replayer = _mem_replayer
     out.rows = finalize_merged(merged, logical_ops, sp.output_columns)
+    if out.row_keys is not None and len(out.rows) != len(out.row_keys):
+        out.row_keys = None  # a trailing agg/unique consumed row identity
+    return out
+
+
+_AGG_KINDS = ("aggregate", "aggregateByKey", "unique")
+
+
+def run_cached(context, src, post_ops, keep_exceptions=False):
+    """Execute downstream ops over a cached dataset (CacheOperator.cc
+    semantics): the materialized normal-case rows run through the engine,
+    and the STORED exception rows (src.pending) are replayed through the
+    full pre+post op chain — so resolvers attached after cache() see
+    pre-cache exceptions, exactly like the reference."""
+    row_ops = [op for op in post_ops if op[0] not in _AGG_KINDS]
+    out2 = run_collect(src.rows, row_ops, src.columns, context.options_obj,
+                       keep_keys=True)
+    out = CollectOutcome()
+    out.mode = out2.mode
+    out.fallback_reason = out2.fallback_reason
+    out.metrics = out2.metrics
+    out.exception_counts = dict(out2.exception_counts)
+
+    pairs = []
+    if out2.row_keys is not None:
+        for k2, row in zip(out2.row_keys, out2.rows):
+            pairs.append((src.row_keys[k2], row))
+    else:  # defensive: row identity lost (should not happen: no aggs here)
+        for j, row in enumerate(out2.rows):
+            pairs.append((j, row))
+
+    full_ops = list(src.pre_ops) + list(post_ops)
+    for key, payload in src.pending:
+        r = src.replayer(payload, full_ops)
+        if r[0] == "row":
+            pairs.append((key, r[1]))
+        elif r[0] == "rows":
+            for v in r[1]:
+                pairs.append((key, v))
+            for e in r[2]:
+                nm = type(e).__name__
+                out.exception_counts[nm] = out.exception_counts.get(nm, 0) + 1
+        elif r[0] == "exc":
+            if keep_exceptions:
+                out.pending.append((key, payload))
+            else:
+                nm = type(r[1]).__name__
+                out.exception_counts[nm] = \
+                    out.exception_counts.get(nm, 0) + 1
+    if keep_exceptions:
+        # cache-of-cache: a later consumer replays with its own (pre2+post2)
+        # chain, which by construction is our post_ops+...; prepend OUR
+        # pre-cache ops so the payload replays through the full lineage
+        pre1 = list(src.pre_ops)
+        rep = src.replayer
+        out.pending_replayer = \
+            lambda payload, ops, used=None: rep(payload, pre1 + list(ops))
+    pairs.sort(key=lambda t: t[0])
+    merged = [_unwrap_row(v) if isinstance(v, tuple) else v for _, v in pairs]
+    out_cols = output_columns_of(src.columns, post_ops)
+    out.rows = finalize_merged(merged, post_ops, out_cols)
+    if keep_exceptions:
+        out.row_keys = [k for k, _ in pairs] \
+            if len(out.rows) == len(pairs) else None
     return out
 
 
@@ -435,3 +516,9 @@ class CollectOutcome:
         self.mode = None
         self.fallback_reason = None
         self.metrics = {}
+        # cache()-path extras (CacheOperator semantics: exceptions stored,
+        # not resolved): pending = [(order_key, payload)], replayer replays a
+        # payload through an op chain; row_keys aligns rows with order keys
+        self.pending = []
+        self.pending_replayer = None
+        self.row_keys = None
