@@ -141,10 +141,16 @@ class StageCodegen:
         # UDF chain as a plain grid-stride kernel over coalesced columns at
         # FULL occupancy (no LDS), where the fused kernel was measured
         # latency-bound at 10 waves/CU (profiles/README.md SQ analysis).
+        # Split parse/UDF kernels (TPX_SPLIT=1): measured SLOWER than the
+        # fused kernel on Zillow Z1 (689-727 vs 805 M rows/s, profiles/
+        # README.md round-2 ledger) — the UDF-phase kernel needs ~198 VGPRs
+        # (whole chain live), capping it at 8 waves/CU vs the fused kernel's
+        # 10, and the parse kernel alone costs 0.77 ms/chunk (walk-bound).
+        # Kept as an env-gated experiment; default stays fused.
         import os as _os
         self.split = (source == "csv"
                       and not (csv_info or {}).get("text_mode")
-                      and _os.environ.get("TPX_SPLIT", "1") != "0")
+                      and _os.environ.get("TPX_SPLIT", "0") == "1")
         # park copy mode: 0 (default) = zero-copy park — string cells stay
         # (ptr,len) views into the device input bytes (the parse kernel's
         # passthrough cost was measured at 2.3 ms/GB; byte-copying cells into

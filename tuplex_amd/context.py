@@ -34,6 +34,22 @@ class CsvSource:
         self.text_mode = text_mode
 
 
+class CachedSource:
+    """cache() materialization (CacheOperator.cc analog): normal-case rows
+    plus the STORED exception rows (pending payloads + a replayer through the
+    original pre-cache op chain) — resolvers attached after cache() resolve
+    pre-cache exceptions, like the reference."""
+    kind = "cached"
+
+    def __init__(self, rows, row_keys, pending, replayer, pre_ops, columns):
+        self.rows = rows
+        self.row_keys = row_keys
+        self.pending = pending
+        self.replayer = replayer
+        self.pre_ops = pre_ops
+        self.columns = columns
+
+
 class OrcSource:
     kind = "orc"
 
@@ -102,22 +118,28 @@ class Context:
         return ""
 
     # ---- execution ----------------------------------------------------------
-    def _execute(self, ds: DataSet, sink=None):
+    def _execute(self, ds: DataSet, sink=None, keep_exceptions=False):
         src = ds._source
         if src.kind == "mem":
             outcome = engine.run_collect(src.data, ds._ops, src.columns,
-                                         self.options_obj)
+                                         self.options_obj,
+                                         keep_exceptions=keep_exceptions)
+        elif src.kind == "cached":
+            outcome = engine.run_cached(self, src, ds._ops,
+                                        keep_exceptions=keep_exceptions)
         elif src.kind == "orc":
             from . import orcio
             outcome = orcio.run_orc(self, src, ds._ops, sink)
         else:
             from . import csvio
-            outcome = csvio.run_csv(self, src, ds._ops, sink)
+            outcome = csvio.run_csv(self, src, ds._ops, sink,
+                                    keep_exceptions=keep_exceptions)
         self.metrics.data.update({
             "mode": outcome.mode,
             **{k: v for k, v in outcome.metrics.items()},
         })
-        if sink is not None and sink[0] == "csv" and src.kind in ("mem", "orc"):
+        if sink is not None and sink[0] == "csv" and src.kind in ("mem", "orc",
+                                                                   "cached"):
             # mem-source tocsv: GPU collect + host CSV formatting (the graded
             # file->file config is csv-source tocsv, which formats on device;
             # parallelize->tocsv is API-completeness, not a bench path)

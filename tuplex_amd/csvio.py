@@ -459,7 +459,8 @@ def _run_chunks_parallel(glib, csrc, desc, work, run_one, devices):
 
 # ---- main entry ------------------------------------------------------------------
 
-def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
+def run_csv(context, src, logical_ops, sink=None,
+            keep_exceptions=False) -> CollectOutcome:
     opts = context.options_obj
     out = CollectOutcome()
 
@@ -522,7 +523,8 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
         return _run_csv_fallback(out, data, col_types, src.null_values,
                                  logical_ops, names, sink,
                                  sp.why_not_compilable, delim.decode(),
-                                 out_cols=sp.output_columns)
+                                 out_cols=sp.output_columns,
+                                 keep_exceptions=keep_exceptions)
 
     glib = GpuLib.get()
     if glib.device_count() == 0:
@@ -539,7 +541,8 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
     except codegen.CodegenError as e:
         return _run_csv_fallback(out, data, col_types, src.null_values,
                                  logical_ops, names, sink, str(e),
-                                 delim.decode(), out_cols=sp.output_columns)
+                                 delim.decode(), out_cols=sp.output_columns,
+                                 keep_exceptions=keep_exceptions)
     stage = glib.compile_stage(csrc, desc)
 
     # chunked execution at inputSplitSize boundaries (LocalBackend.cc:552-658:
@@ -576,6 +579,7 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
     chunk_text = [None] * nch     # csv sink: (text, idxs, offs, lo, hi)
     chunk_repl = [None] * nch     # {key: [rows]} host-replayed
     chunk_excs = [None] * nch     # {exc_name: count}
+    chunk_pend = [None] * nch     # keep_exceptions: [(key, raw payload)]
     mlock = _thr.Lock()
     resolve_procs = int(opts.get("tuplex.gpu.resolveProcesses", "0") or "0")
     rpool = [None]                # lazily-started parallel resolver pool
@@ -637,15 +641,21 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
                                                names, delim.decode(),
                                                used=sp.used_source_cols)
                         results.append(presolve._shrink(r))
-                for row, r in zip(rows_keys, results):
+                pend = []
+                for (row, r), payload in zip(zip(rows_keys, results),
+                                             payloads):
                     if r[0] == "row":
                         repl[row] = [r[1]]
                     elif r[0] == "excname":
-                        excs[r[1]] = excs.get(r[1], 0) + 1
+                        if keep_exceptions:
+                            pend.append((row, payload))
+                        else:
+                            excs[r[1]] = excs.get(r[1], 0) + 1
                     elif r[0] == "rows":  # 1:N join expansion
                         repl[row] = r[1]
                         for nm in r[2]:
                             excs[nm] = excs.get(nm, 0) + 1
+                chunk_pend[ci] = pend
             chunk_repl[ci] = repl
             chunk_excs[ci] = excs
             if sink_kind == "mem":
@@ -715,6 +725,26 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
             all_rows.extend(chunk_rows[ci])
         if chunk_text[ci] is not None:
             text_parts.append(chunk_text[ci])
+        if chunk_pend[ci]:
+            out.pending.extend(chunk_pend[ci])
+    if keep_exceptions:
+        _nv, _ct, _nm, _dl = src.null_values, col_types, names, delim.decode()
+        _tm = text_mode
+
+        def _csv_replayer(payload, ops, _memo={}):
+            ck = (len(ops), id(ops[-1]) if ops else 0)
+            used2 = _memo.get(ck, -1)
+            if used2 == -1:
+                try:
+                    used2 = plan.build_stage(_ct, _nm, ops).used_source_cols
+                except Exception:  # noqa: BLE001 - fallback: parse all cells
+                    used2 = None
+                _memo[ck] = used2
+            if _tm:
+                return replay_text_row(payload, _ct[0], _nv, ops)
+            return replay_csv_row(payload, _ct, _nv, ops, _nm, _dl,
+                                  used=used2)
+        out.pending_replayer = _csv_replayer
 
     if dist and world > 1:
         out.metrics["world"] = world
@@ -812,6 +842,10 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
         out.rows = finalize_merged(
             [v for i in sorted(merged) for v in merged[i]],
             logical_ops, sp.output_columns)
+        if keep_exceptions:
+            out.row_keys = [i for i in sorted(merged) for _ in merged[i]]
+            if len(out.row_keys) != len(out.rows):
+                out.row_keys = None  # trailing agg consumed row identity
     else:
         header_line = _format_csv_row(sp.output_columns or
                                       ["column%d" % i
@@ -851,7 +885,8 @@ def run_csv(context, src, logical_ops, sink=None) -> CollectOutcome:
 
 
 def _run_csv_fallback(out, data, col_types, null_values, logical_ops, names,
-                      sink, why, delim=",", out_cols=None):
+                      sink, why, delim=",", out_cols=None,
+                      keep_exceptions=False):
     """Whole-stage interpreter fallback (non-compilable UDF) — the reference's
     fallback mode. Still semantically exact; slow by design. The trailing
     aggregate/aggregateByKey/unique fold over the PIPELINE's output columns
@@ -871,6 +906,9 @@ def _run_csv_fallback(out, data, col_types, null_values, logical_ops, names,
         if r[0] == "row":
             rows_out[i] = [r[1]]
         elif r[0] == "exc":
+            if keep_exceptions:
+                out.pending.append((i, all_lines[i]))
+                continue
             nm = type(r[1]).__name__
             out.exception_counts[nm] = out.exception_counts.get(nm, 0) + 1
         elif r[0] == "rows":  # 1:N join expansion
@@ -891,6 +929,15 @@ def _run_csv_fallback(out, data, col_types, null_values, logical_ops, names,
     merged = [v if not isinstance(v, tuple) else _unwrap_row(v)
               for i in sorted(rows_out) for v in rows_out[i]]
     out.rows = finalize_merged(merged, logical_ops, out_cols)
+    if keep_exceptions:
+        out.row_keys = [i for i in sorted(rows_out) for _ in rows_out[i]]
+        if len(out.row_keys) != len(out.rows):
+            out.row_keys = None
+
+        def _fb_replayer(payload, ops, _ct=col_types, _nv=null_values,
+                         _nm=names, _dl=delim):
+            return replay_csv_row(payload, _ct, _nv, ops, _nm, _dl)
+        out.pending_replayer = _fb_replayer
     if sink is not None and sink[0] == "csv":
         header_line = _format_csv_row(out_cols or
                                       ["column%d" % i for i in

@@ -102,16 +102,26 @@ class DataSet:
                           suffixes, "left")
 
     def cache(self, store_specialized: bool = True) -> "DataSet":
-        """Materialize the pipeline so far; downstream operators start from the
-        materialized rows instead of re-executing (CacheOperator,
+        """Materialize the pipeline so far; downstream operators start from
+        the materialized rows instead of re-executing (CacheOperator,
         core/src/logical/CacheOperator.cc — SURVEY.md §8f-4). Exception rows
-        are resolved AT the cache point (reference stores them for later
-        resolution; resolvers attached after cache() therefore cannot see
-        pre-cache exceptions here — divergence documented in DESIGN.md)."""
-        from .context import ParallelizeSource
-        outcome = self._context._execute(self)
-        ds = DataSet(self._context,
-                     ParallelizeSource(list(outcome.rows), self.columns))
+        are STORED at the cache point, not resolved — resolvers attached
+        after cache() replay them through the original op chain, matching
+        the reference's deferred resolution. Pipelines ending in
+        aggregate/unique consume row identity, so those cache the final
+        values (nothing left to resolve into)."""
+        from .context import CachedSource, ParallelizeSource
+        has_agg = any(op[0] in ("aggregate", "aggregateByKey", "unique")
+                      for op in self._ops)
+        outcome = self._context._execute(self, keep_exceptions=not has_agg)
+        if (has_agg or outcome.row_keys is None
+                or outcome.pending_replayer is None):
+            src = ParallelizeSource(list(outcome.rows), self.columns)
+        else:
+            src = CachedSource(list(outcome.rows), outcome.row_keys,
+                               outcome.pending, outcome.pending_replayer,
+                               list(self._ops), self.columns)
+        ds = DataSet(self._context, src)
         ds._last_outcome = outcome
         return ds
 

@@ -275,6 +275,20 @@ def run_collect(data: List[Any], logical_ops: List[tuple],
     original input index (ordered-merge key)."""
     out = CollectOutcome()
     keep_keys = keep_keys or keep_exceptions
+    if not logical_ops:
+        # pure materialization (collect of an empty pipeline / cached rows):
+        # nothing to compute — pass rows through (the reference round-trips
+        # partitions; observable rows are identical)
+        out.mode = "identity"
+        out.rows = list(data)
+        if keep_keys:
+            out.row_keys = list(range(len(data)))
+        if keep_exceptions:
+            out.pending_replayer = \
+                lambda payload, ops, used=None: resolve.replay_row(
+                    payload, ops, columns,
+                    not (data and isinstance(data[0], tuple)))
+        return out
     maj = T.infer_majority_type(data, optional_threshold=options.optional_threshold)
     row_maj = T.row_type_of(maj)
     scalar_input = not T.is_tuple(maj)
@@ -401,8 +415,21 @@ def run_cached(context, src, post_ops, keep_exceptions=False):
     full pre+post op chain — so resolvers attached after cache() see
     pre-cache exceptions, exactly like the reference."""
     row_ops = [op for op in post_ops if op[0] not in _AGG_KINDS]
-    out2 = run_collect(src.rows, row_ops, src.columns, context.options_obj,
-                       keep_keys=True)
+    # leading resolve/ignore handlers target PRE-cache exceptions (replayed
+    # below through the full chain); the materialized normal rows have no
+    # exception to resolve, so strip them for the row run
+    lead = 0
+    while lead < len(row_ops) and row_ops[lead][0] in ("resolve", "ignore"):
+        lead += 1
+    row_run_ops = row_ops[lead:]
+    if row_run_ops:
+        out2 = run_collect(src.rows, row_run_ops, src.columns,
+                           context.options_obj, keep_keys=True)
+    else:  # identity over cached rows — no engine pass needed
+        out2 = CollectOutcome()
+        out2.mode = "gpu"
+        out2.rows = list(src.rows)
+        out2.row_keys = list(range(len(src.rows)))
     out = CollectOutcome()
     out.mode = out2.mode
     out.fallback_reason = out2.fallback_reason
