@@ -311,3 +311,36 @@ def test_join_gpu_duplicate_keys_tocsv(tmp_path):
     with open(outp, "rb") as f:
         got = f.read()
     assert got == ref["csv_text"]
+
+
+def _post_join_map(x):
+    return (x["key"], x["x"], (x["w"] or 0) + 1)
+
+
+def _post_join_keep(x):
+    return x[2] != 31
+
+
+@pytest.mark.gpu
+def test_join_gpu_duplicate_keys_mid_pipeline():
+    """Mid-pipeline dup-key join (HashJoinStage.cc:473 case): the engine
+    splits the pipeline at the join (find_dup_join_split) so BOTH halves run
+    compiled on the GPU — stage 1 ends in the terminal 1:N expansion, stage 2
+    maps/filters the expanded rows from the materialized partitions. Output
+    identical to the oracle's forked replay."""
+    import random
+    dup = [(1, "a", 10), (1, "b", None), (2, "c", 30), (7, "z", 5),
+           (1, "d", 40)]
+    rng = random.Random(5)
+    rows = [(rng.randint(0, 8), "v%d" % i) for i in range(30000)]
+    jop = ("join", dup, ["k", "x", "w"], "key", "k", "inner",
+           "", "", "", "")
+    ops = [jop, ("map", _post_join_map), ("filter", _post_join_keep)]
+    ctx = tuplex_amd.Context()
+    ds = apply_ops(ctx.parallelize(rows, columns=["key", "val"]), ops)
+    got = ds.collect()
+    assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    ref = pyoracle.run_pipeline(rows, ops, columns=["key", "val"])
+    assert got == ref["output"]
+    assert ds.exception_counts == ref["exception_counts"]
+    assert len(got) > 10000

@@ -120,6 +120,31 @@ class Context:
     # ---- execution ----------------------------------------------------------
     def _execute(self, ds: DataSet, sink=None, keep_exceptions=False):
         src = ds._source
+        # mid-pipeline duplicate-key join: split the pipeline at the join
+        # (PhysicalPlan stage-split analog) so stage 1 runs the GPU terminal
+        # 1:N expansion and stage 2 continues compiled from the materialized
+        # rows; exceptions stay deferred through the CachedSource replayer
+        if src.kind in ("mem", "csv", "cached"):
+            from . import plan as _plan
+            cut = _plan.find_dup_join_split(ds._ops)
+            if cut is not None:
+                try:
+                    no_gpu = engine.GpuLib.get().device_count() == 0
+                except RuntimeError:
+                    no_gpu = True
+                if no_gpu:  # CPU box: the interpreter fallback handles 1:N
+                    cut = None
+            if cut is not None:
+                ds1 = DataSet(self, src, list(ds._ops[:cut + 1]))
+                out1 = self._execute(ds1, keep_exceptions=True)
+                if (out1.row_keys is not None
+                        and out1.pending_replayer is not None):
+                    csrc = CachedSource(list(out1.rows), out1.row_keys,
+                                        out1.pending, out1.pending_replayer,
+                                        list(ds._ops[:cut + 1]), ds1.columns)
+                    ds2 = DataSet(self, csrc, list(ds._ops[cut + 1:]))
+                    return self._execute(ds2, sink=sink,
+                                         keep_exceptions=keep_exceptions)
         if src.kind == "mem":
             outcome = engine.run_collect(src.data, ds._ops, src.columns,
                                          self.options_obj,

@@ -429,3 +429,24 @@ def _fallback(sp, why):
     sp.compilable = False
     if sp.why_not_compilable is None:
         sp.why_not_compilable = why
+
+
+def find_dup_join_split(logical_ops):
+    """Index of the first duplicate-build-key join that is NOT eligible for
+    the terminal-join GPU expansion (it is mid-pipeline), or None. The engine
+    splits the pipeline there — the PhysicalPlan stage-split analog — so both
+    halves run compiled on the GPU: stage 1 ends in the terminal 1:N join
+    (bucket-looping write kernel), stage 2 continues from the materialized
+    rows via the CachedSource machinery (exceptions stay deferred)."""
+    for i, entry in enumerate(logical_ops):
+        if entry[0] != "join" or i == len(logical_ops) - 1:
+            continue
+        _, rrows, rcols, lk, rk = entry[:5]
+        try:
+            rki = rcols.index(rk)
+            keys = [r[rki] for r in rrows]
+        except Exception:  # malformed join op -> let build_stage report it
+            continue
+        if len(set(keys)) != len(keys):
+            return i
+    return None
