@@ -145,8 +145,10 @@ def e2e_leg(header, body, mb, local_rank=0, dist=None, rank=0, world=1):
             f.seek(len(header))
             rows = sum(chunk.count(b"\n")
                        for chunk in iter(lambda: f.read(1 << 24), b""))
+        sz = os.path.getsize(inp)
         return {"e2e_rows_per_s": rows / wall,
-                "e2e_mb": os.path.getsize(inp) >> 20,
+                "e2e_mb": sz >> 20,
+                "e2e_gb_per_s": sz / wall / 1e9,
                 "e2e_seconds": wall, "e2e_engine_ranks": world}
     except Exception as e:  # noqa: BLE001
         log("e2e leg failed:", e)
