@@ -126,19 +126,31 @@ def e2e_leg(header, body, mb, local_rank=0, dist=None, rank=0, world=1):
         if dist:
             dist.barrier()
         t0 = time.perf_counter()
-        ctx = tuplex_amd.Context({"tuplex.gpu.device": str(local_rank)})
-        ds = apply_ops(ctx.csv(inp), zillow_ops())
-        ds.tocsv(outp)
+        ok = 1
+        try:
+            ctx = tuplex_amd.Context({"tuplex.gpu.device": str(local_rank)})
+            ds = apply_ops(ctx.csv(inp), zillow_ops())
+            ds.tocsv(outp)
+            if ds._last_outcome.mode != "gpu":
+                log("e2e leg fell back:", ds._last_outcome.fallback_reason)
+                ok = 0
+        except Exception as e:  # noqa: BLE001 — stay collective on failure
+            log("e2e leg failed:", e)
+            ok = 0
         t1 = time.perf_counter()
-        if ds._last_outcome.mode != "gpu":
-            log("e2e leg fell back:", ds._last_outcome.fallback_reason)
-            return None
         wall = t1 - t0
         if dist:
+            # agree on success + MAX wall collectively so no rank bails out
+            # of a pending collective (deadlock) when another rank failed
             import torch
-            tt = torch.tensor([wall], device="cuda")
-            dist.all_reduce(tt, op=dist.ReduceOp.MAX)
-            wall = float(tt.item())
+            red_dev = ("cuda" if dist.get_backend() == "nccl" else "cpu")
+            t = torch.tensor([float(ok), wall], device=red_dev)
+            dist.all_reduce(t[:1], op=dist.ReduceOp.MIN)
+            dist.all_reduce(t[1:], op=dist.ReduceOp.MAX)
+            ok = int(t[0].item())
+            wall = float(t[1].item())
+        if not ok:
+            return None
         if rank != 0:
             return None
         with open(inp, "rb") as f:
