@@ -391,6 +391,44 @@ def precompile_csv(verbose=False):
     return n
 
 
+def precompile_tests(verbose=False):
+    """Exact stage sources of the newer GPU tests/probes (dist engine, cache
+    compiled path, mid-pipeline dup join stage 2) so fresh boxes never
+    hipRTC-compile during the suite."""
+    from tuplex_amd import codegen, engine, plan
+    from tuplex_amd import ttypes as T
+    glib = engine.GpuLib.get()
+    n = 0
+    # test_gpu_dist stages (csv source, 3 cols)
+    from tests.test_gpu_dist import _du_map, _du_keep, _du_ab, _comb, _sumb
+    for ops, sinks in ((([("map", _du_map), ("filter", _du_keep)]),
+                        ("mem", "csv")),
+                       (([("map", _du_ab),
+                          ("aggregate", _comb, _sumb, 0)]), ("mem",))):
+        sp = plan.build_stage([T.I64, T.I64, T.STR], ["a", "b", "name"], ops)
+        if not sp.compilable:
+            continue
+        for sink in sinks:
+            src, desc = codegen.generate_stage(
+                sp, source="csv", sink=sink,
+                csv_info={"null_values": [""], "delimiter": ","})
+            _stage_compile(glib, src, desc)
+            n += 1
+    # test_cache_cpu gpu stages (mem source)
+    from tests.test_cache_cpu import _gdiv, _gres
+    for in_types, cols, ops in (
+            ([T.I64, T.I64], None, [("map", _gdiv)]),
+            ([T.I64, T.I64], None,
+             [("map", _gdiv), ("resolve", ZeroDivisionError, _gres)]),
+    ):
+        sp = plan.build_stage(in_types, cols, ops)
+        if sp.compilable:
+            src, desc = codegen.generate_stage(sp, source="mem", sink="mem")
+            _stage_compile(glib, src, desc)
+            n += 1
+    return n
+
+
 def precompile_all(verbose=False):
     """Generate + hipRTC-compile (compile-only) every pipeline's stage so the
     hsaco cache is warm. Works with no GPU."""
@@ -415,6 +453,7 @@ def precompile_all(verbose=False):
         if verbose:
             print("precompiled:", name)
     n += precompile_csv(verbose=verbose)
+    n += precompile_tests(verbose=verbose)
     return n
 
 

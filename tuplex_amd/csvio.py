@@ -577,6 +577,7 @@ def run_csv(context, src, logical_ops, sink=None,
     # resident fast path (1 rank, 1 device): skip host chunking entirely —
     # one H2D, one execute; the boundary scan runs device-side
     use_resident = (world == 1 and min(want or ndev, ndev) <= 1 and
+                    not opts.is_set("tuplex.inputSplitSize") and
                     len(data) <= parse_size(
                         opts.get("tuplex.gpu.residentMaxSize", "24GB")))
     chunks = ([0, len(data)] if use_resident or len(data) <= split * 2
