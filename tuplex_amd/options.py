@@ -55,7 +55,7 @@ DEFAULTS = {
     # larger ranges (the C-ABI pipelines 4 sub-chunks on 2 streams per call)
     "tuplex.gpu.inputSplitSize": "256MB",
     "tuplex.gpu.heapFactor": "2.0",
-    "tuplex.gpu.devices": "0",            # in-process device fan-out (0 = all visible)
+    "tuplex.gpu.devices": "1",            # in-process device fan-out (N>1 opt-in; 0 = all visible)
     "tuplex.gpu.resolveProcesses": "0",   # parallel host resolver width (0 = auto, 1 = inline)
 }
 
