@@ -36,20 +36,22 @@ def main():
              time.perf_counter() - t0), flush=True)
 
     for use_procs in ([procs] if procs else [1, 0]):
-        ctx = tuplex_amd.Context(
-            {"tuplex.gpu.resolveProcesses": str(use_procs)})
-        ds = apply_ops(ctx.csv(path), zillow_ops())
-        t0 = time.perf_counter()
-        ds.tocsv("/tmp/dirtyprobe_out.csv")
-        dt = time.perf_counter() - t0
-        assert ds._last_outcome.mode == "gpu", \
-            ds._last_outcome.fallback_reason
-        exc = ds._last_outcome.exception_counts
         label = "serial" if use_procs == 1 else ("pool(%s)" % (use_procs or
                                                                "auto"))
-        print("%s resolver: %.2f s  %.1f M rows/s whole-job  (exc %s)"
-              % (label, dt, n_rows / dt / 1e6,
-                 dict(list(exc.items())[:3])), flush=True)
+        for it in range(2):  # 2nd iteration = warm persistent pool
+            ctx = tuplex_amd.Context(
+                {"tuplex.gpu.resolveProcesses": str(use_procs)})
+            ds = apply_ops(ctx.csv(path), zillow_ops())
+            t0 = time.perf_counter()
+            ds.tocsv("/tmp/dirtyprobe_out.csv")
+            dt = time.perf_counter() - t0
+            assert ds._last_outcome.mode == "gpu", \
+                ds._last_outcome.fallback_reason
+            exc = ds._last_outcome.exception_counts
+            print("%s resolver (pass %d): %.2f s  %.1f M rows/s whole-job"
+                  "  (exc %s)"
+                  % (label, it, dt, n_rows / dt / 1e6,
+                     dict(list(exc.items())[:3])), flush=True)
     os.unlink(path)
 
 

@@ -426,6 +426,30 @@ def precompile_tests(verbose=False):
             src, desc = codegen.generate_stage(sp, source="mem", sink="mem")
             _stage_compile(glib, src, desc)
             n += 1
+    # flights + 2 joins (the 110-col stage: ~13 min of cold hipRTC)
+    from tests import extra_data as X
+    from tests.test_join import FL_J1, FL_J2, _post_join_map, _post_join_keep
+    from tuplex_amd import csvio
+    data = X.make_flights_csv(500)
+    sample = data[:1 << 18]
+    _h, names, col_types = csvio.sniff(sample, [""], 0.9, None, None, b",")
+    sp = plan.build_stage(col_types, names,
+                          X.flights_ops() + [FL_J1, FL_J2])
+    if sp.compilable:
+        src, desc = codegen.generate_stage(
+            sp, source="csv", sink="mem",
+            csv_info={"null_values": [""], "delimiter": ","})
+        _stage_compile(glib, src, desc)
+        n += 1
+    # mid-pipeline dup-join stage 2 (map+filter over the joined rows)
+    sp = plan.build_stage([T.STR, T.I64, T.STR, ("opt", T.I64)],
+                          ["val", "key", "x", "w"],
+                          [("map", _post_join_map),
+                           ("filter", _post_join_keep)])
+    if sp.compilable:
+        src, desc = codegen.generate_stage(sp, source="mem", sink="mem")
+        _stage_compile(glib, src, desc)
+        n += 1
     return n
 
 

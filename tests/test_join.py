@@ -170,6 +170,16 @@ def test_join_gpu_csv_string_key(tmp_path):
     assert all(c.startswith("City-") for _, c, _d in got)
 
 
+# module-level so tests/pipelines.precompile_tests can prebuild the exact
+# stage (this 110-col + 2-join stage costs ~13 min of hipRTC on a cold box)
+FL_CARRIERS = [("aa", "American"), ("dl", "Delta"), ("ua", "United")]
+FL_CITIES = [("City-%d Airport" % i, i * 3) for i in range(1, 150)]
+FL_J1 = ("join", FL_CARRIERS, ["code_c", "carrier_name"], "code", "code_c",
+         "inner", "", "", "", "")
+FL_J2 = ("join", FL_CITIES, ["city_a", "tz"], "c3", "city_a",
+         "left", "", "", "", "")
+
+
 @pytest.mark.gpu
 def test_join_gpu_flights_pipeline(tmp_path):
     """The flights benchmark shape (runtuplex.py:210+): parse the wide CSV,
@@ -184,15 +194,7 @@ def test_join_gpu_flights_pipeline(tmp_path):
     with open(p, "wb") as f:
         f.write(data)
 
-    # fl_code lowercases the carrier (aa/dl survive fl_carrier); c3 is
-    # "City-N Airport" — the left join hits only a subset of cities
-    carriers = [("aa", "American"), ("dl", "Delta"), ("ua", "United")]
-    cities = [("City-%d Airport" % i, i * 3) for i in range(1, 150)]
-    j1 = ("join", carriers, ["code_c", "carrier_name"], "code", "code_c",
-          "inner", "", "", "", "")
-    j2 = ("join", cities, ["city_a", "tz"], "c3", "city_a",
-          "left", "", "", "", "")
-    ops = X.flights_ops() + [j1, j2]
+    ops = X.flights_ops() + [FL_J1, FL_J2]
 
     ctx = tuplex_amd.Context()
     ds = apply_ops(ctx.csv(p), ops)
@@ -314,7 +316,8 @@ def test_join_gpu_duplicate_keys_tocsv(tmp_path):
 
 
 def _post_join_map(x):
-    return (x["key"], x["x"], (x["w"] or 0) + 1)
+    return (x["key"], x["x"],
+            (x["w"] if x["w"] is not None else 0) + 1)
 
 
 def _post_join_keep(x):

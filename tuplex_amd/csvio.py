@@ -638,7 +638,7 @@ def run_csv(context, src, logical_ops, sink=None,
                     # fan-out analog): replay on a process pool
                     with rpool_lock:
                         if rpool[0] is None:
-                            rpool[0] = presolve.ResolverPool(
+                            rpool[0] = presolve.get_pool(
                                 col_types, src.null_values, logical_ops,
                                 names, delim.decode(), sp.used_source_cols,
                                 text_mode, processes=resolve_procs)
@@ -724,8 +724,7 @@ def run_csv(context, src, logical_ops, sink=None,
         devices = [(dev0 + k) % ndev for k in range(use_devs)]
         _run_chunks_parallel(glib, csrc, desc, my_work, run_one, devices)
         out.metrics["devices"] = use_devs
-    if rpool[0] is not None:
-        rpool[0].close()
+    # pools persist across executions (presolve._POOLS; atexit shutdown)
 
     replayed = {}
     all_rows = []
