@@ -674,8 +674,13 @@ def run_csv(context, src, logical_ops, sink=None,
             chunk_excs[ci] = excs
             if sink_kind == "mem":
                 out_bytes = ctypes.string_at(res.out_data, res.out_size)
+                _offs = (_np.ctypeslib.as_array(
+                    res.out_row_offsets,
+                    shape=(res.out_num_rows,)).copy()
+                    if res.out_num_rows else None)
                 rws = rowfmt.deserialize_partition(out_bytes,
-                                                   T.tup(sp.gpu_output_types))
+                                                   T.tup(sp.gpu_output_types),
+                                                   offsets=_offs)
                 idxs = _np.ctypeslib.as_array(
                     res.out_row_indices, shape=(res.out_num_rows,)).tolist() \
                     if res.out_num_rows else []

@@ -194,7 +194,10 @@ def execute_stage_mem(sp: plan.StageProgram, norm_rows: List[tuple],
         out = ExecResult()
         out_bytes = ctypes.string_at(res.out_data, res.out_size)
         out_row_type = T.tup(sp.gpu_output_types)
-        out.rows = rowfmt.deserialize_partition(out_bytes, out_row_type)
+        _offs = ([res.out_row_offsets[i] for i in range(res.out_num_rows)]
+                 if res.out_num_rows else None)
+        out.rows = rowfmt.deserialize_partition(out_bytes, out_row_type,
+                                                offsets=_offs)
         out.row_indices = [res.out_row_indices[i] for i in range(res.out_num_rows)] \
             if res.out_num_rows else []
         if res.exc_num_rows:
@@ -240,7 +243,10 @@ def execute_stage_col(sp: plan.StageProgram, dev_slots, n_rows: int,
         out = ExecResult()
         out_bytes = ctypes.string_at(res.out_data, res.out_size)
         out_row_type = T.tup(sp.gpu_output_types)
-        out.rows = rowfmt.deserialize_partition(out_bytes, out_row_type)
+        _offs = ([res.out_row_offsets[i] for i in range(res.out_num_rows)]
+                 if res.out_num_rows else None)
+        out.rows = rowfmt.deserialize_partition(out_bytes, out_row_type,
+                                                offsets=_offs)
         out.row_indices = [res.out_row_indices[i] for i in range(res.out_num_rows)] \
             if res.out_num_rows else []
         if res.exc_num_rows:
