@@ -154,7 +154,8 @@ class Context:
                                         keep_exceptions=keep_exceptions)
         elif src.kind == "orc":
             from . import orcio
-            outcome = orcio.run_orc(self, src, ds._ops, sink)
+            outcome = orcio.run_orc(self, src, ds._ops, sink,
+                                    keep_exceptions=keep_exceptions)
         else:
             from . import csvio
             outcome = csvio.run_csv(self, src, ds._ops, sink,

@@ -54,7 +54,7 @@ def _table_rows(tab, names, lo=0, hi=None):
     return list(zip(*cols))
 
 
-def run_orc(ctx, src, logical_ops, sink=None):
+def run_orc(ctx, src, logical_ops, sink=None, keep_exceptions=False):
     out = CollectOutcome()
     if pa is None:
         raise RuntimeError("pyarrow is required for the ORC source")
@@ -207,6 +207,9 @@ def run_orc(ctx, src, logical_ops, sink=None):
         if r[0] == "row":
             results[i] = [r[1]]
         elif r[0] == "exc":
+            if keep_exceptions:
+                out.pending.append((i, _table_rows(tab, names, i, i + 1)[0]))
+                continue
             nm = type(r[1]).__name__
             out.exception_counts[nm] = out.exception_counts.get(nm, 0) + 1
         elif r[0] == "rows":
@@ -217,6 +220,15 @@ def run_orc(ctx, src, logical_ops, sink=None):
     merged = [v for i in sorted(results) for v in results[i]]
     from .engine import finalize_merged
     out.rows = finalize_merged(merged, logical_ops, sp.output_columns)
+    if keep_exceptions:
+        out.row_keys = [i for i in sorted(results) for _ in results[i]]
+        if len(out.row_keys) != len(out.rows):
+            out.row_keys = None
+
+        def _orc_replayer(payload, ops, _names=names,
+                          _scalar=scalar_input):
+            return resolve.replay_row(payload, ops, _names, _scalar)
+        out.pending_replayer = _orc_replayer
     return out
 
 
