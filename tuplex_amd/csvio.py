@@ -878,6 +878,13 @@ def run_csv(context, src, logical_ops, sink=None,
             else:
                 segs.append((ci, text))
         if dist and world > 1:
+            # job-level exception counts are the union over ranks
+            parts_e = _gather_objects(dist, dict(out.exception_counts))
+            out.exception_counts = {}
+            for pe in parts_e:
+                for nm, c in pe.items():
+                    out.exception_counts[nm] = \
+                        out.exception_counts.get(nm, 0) + c
             path = sink[1]
             if path.endswith(".csv"):
                 # single output file: gather segments to rank 0, write in
