@@ -1462,6 +1462,18 @@ extern "C" void tpx_dev_free(uint64_t ptr) {
     if (ptr) (void)hipFree((void*)(uintptr_t)ptr);
 }
 
+// pinned host staging for streamed ingestion (file -> pinned ring -> DMA):
+// pageable hipMemcpy runs ~6 GB/s, pinned ~20+ GB/s, and reads can overlap
+extern "C" uint64_t tpx_pinned_alloc(int64_t size) {
+    void* p = nullptr;
+    if (hipHostMalloc(&p, (size_t)size) != hipSuccess) return 0;
+    return (uint64_t)(uintptr_t)p;
+}
+
+extern "C" void tpx_pinned_free(uint64_t ptr) {
+    if (ptr) (void)hipHostFree((void*)(uintptr_t)ptr);
+}
+
 extern "C" int64_t tpx_stage_execute_csv_dev(tpx_stage* st, uint64_t dev_bytes,
                                              int64_t size,
                                              int64_t first_global_row,

@@ -470,6 +470,81 @@ def _run_chunks_parallel(glib, csrc, desc, work, run_one, devices):
         raise errs[0]
 
 
+def _stream_files_to_device(glib, fmeta, dev, total_len, metrics):
+    """Stream the input files straight to HBM through a pinned double buffer:
+    file reads land in pinned memory (readinto) while an uploader thread DMAs
+    the previous block (pageable hipMemcpy runs ~6 GB/s; pinned ~20 GB/s and
+    the read overlaps it). The host never holds the whole input. Returns
+    False on any failure — caller falls back to the host-buffered path."""
+    import queue as _q
+    import threading as _thr
+    import time as _time
+    B = 64 << 20
+    pin = [glib.lib.tpx_pinned_alloc(B), glib.lib.tpx_pinned_alloc(B)]
+    if not (pin[0] and pin[1]):
+        for p in pin:
+            if p:
+                glib.lib.tpx_pinned_free(p)
+        return False
+    views = [(ctypes.c_char * B).from_address(p) for p in pin]
+    jobs = _q.Queue(maxsize=2)
+    free_slots = _q.Queue()
+    free_slots.put(0)
+    free_slots.put(1)
+    err = []
+
+    def uploader():
+        while True:
+            item = jobs.get()
+            if item is None:
+                return
+            bi, doff, n = item
+            rc = glib.lib.tpx_dev_upload(dev + doff, pin[bi], n)
+            if rc != 0:
+                err.append(glib.err())
+                return
+            free_slots.put(bi)
+
+    th = _thr.Thread(target=uploader, daemon=True)
+    th.start()
+    t0 = _time.perf_counter()
+    ok = False
+    try:
+        off = 0
+        for path, hdr, ln, needs_nl in fmeta:
+            with open(path, "rb") as f:
+                f.seek(hdr)
+                rem = ln
+                while rem > 0 and not err:
+                    bi = free_slots.get(timeout=120)
+                    n = min(B, rem)
+                    got = f.readinto(memoryview(views[bi])[:n])
+                    if got != n:  # file changed underneath us
+                        return False
+                    jobs.put((bi, off, n))
+                    off += n
+                    rem -= n
+            if needs_nl and not err:
+                bi = free_slots.get(timeout=120)
+                views[bi][0] = b"\n"
+                jobs.put((bi, off, 1))
+                off += 1
+        ok = not err and off == total_len
+        return ok
+    except Exception:  # noqa: BLE001 - caller falls back
+        return False
+    finally:
+        try:
+            jobs.put(None, timeout=120)
+        except Exception:  # noqa: BLE001 - uploader already dead, queue full
+            pass
+        th.join(timeout=120)
+        for p in pin:
+            glib.lib.tpx_pinned_free(p)
+        if ok:
+            metrics["t_h2d_ms"] += (_time.perf_counter() - t0) * 1e3
+
+
 # ---- main entry ------------------------------------------------------------------
 
 def run_csv(context, src, logical_ops, sink=None,
@@ -486,17 +561,17 @@ def run_csv(context, src, logical_ops, sink=None,
     if not paths:
         raise FileNotFoundError(src.pattern)
 
-    blobs = []
-    for p in paths:
-        with open(p, "rb") as f:
-            blobs.append(f.read())
-
     text_mode = bool(getattr(src, "text_mode", False))
     if not text_mode and src.quotechar != '"':
         # the GPU scan is specialised to RFC-4180 '"'; other quote chars take the
         # interpreter path (correct, slow — like the reference's fallback mode)
         raise NotImplementedError(
             "quotechar %r: only '\"' runs on the GPU this round" % src.quotechar)
+    # sniff from a bounded sample; files are NOT read eagerly — the resident
+    # path streams them straight to the device (pinned ring, below) and the
+    # ranged/fallback paths load lazily via _load_data()
+    with open(paths[0], "rb") as _f0:
+        sample = _f0.read(256 << 10)
     if text_mode:
         # text(): one str column, rows split on every newline, no sniffing
         has_header = False
@@ -505,8 +580,6 @@ def run_csv(context, src, logical_ops, sink=None,
         delim = b","  # unused in text mode
     else:
         # sniff on the first file's sample (FileInputOperator.cc:78 semantics)
-        sample = bytes(blobs[0][:256 << 10])
-        # cut sample at the last complete row
         nl = sample.rfind(b"\n")
         if nl >= 0:
             sample = sample[:nl + 1]
@@ -518,23 +591,46 @@ def run_csv(context, src, logical_ops, sink=None,
         col_types = apply_type_hints(col_types, names,
                                      getattr(src, "type_hints", None))
 
-    # assemble the GPU input: concat files, strip header lines, ensure trailing \n
-    parts = []
-    for i, b in enumerate(blobs):
+    # per-file metadata: header skip, payload length, missing trailing \n
+    fmeta = []  # (path, data_off, data_len, needs_nl)
+    for p in paths:
+        size = os.path.getsize(p)
+        hdr = 0
         if has_header:
-            p = b.find(b"\n")
-            b = b[p + 1:] if p >= 0 else b""
-        if b and not b.endswith(b"\n"):
-            b += b"\n"
-        parts.append(b)
-    data = b"".join(parts) if len(parts) > 1 else parts[0]
+            with open(p, "rb") as f:
+                head = f.read(256 << 10)
+            hnl = head.find(b"\n")
+            hdr = hnl + 1 if hnl >= 0 else size
+        needs_nl = False
+        if size - hdr > 0:
+            with open(p, "rb") as f:
+                f.seek(size - 1)
+                needs_nl = f.read(1) != b"\n"
+        fmeta.append((p, hdr, size - hdr, needs_nl))
+    total_len = sum(ln + (1 if nn else 0) for _, _, ln, nn in fmeta)
+    data = None  # lazily materialized host copy (ranged / fallback paths)
+
+    def _load_data():
+        nonlocal data
+        if data is None:
+            parts = []
+            for p, hdr, ln, nn in fmeta:
+                with open(p, "rb") as f:
+                    f.seek(hdr)
+                    b = f.read()
+                if nn:
+                    b += b"\n"
+                parts.append(b)
+            data = (b"".join(parts) if len(parts) != 1
+                    else (parts[0] if parts else b""))
+        return data
 
     sp = plan.build_stage(col_types, names, logical_ops)
     sink_kind = "csv" if (sink is not None and sink[0] == "csv") else "mem"
 
     if not sp.compilable:
-        return _run_csv_fallback(out, data, col_types, src.null_values,
-                                 logical_ops, names, sink,
+        return _run_csv_fallback(out, _load_data(), col_types,
+                                 src.null_values, logical_ops, names, sink,
                                  sp.why_not_compilable, delim.decode(),
                                  out_cols=sp.output_columns,
                                  keep_exceptions=keep_exceptions)
@@ -552,9 +648,10 @@ def run_csv(context, src, logical_ops, sink=None,
                       "delimiter": delim.decode(),
                       "text_mode": text_mode})
     except codegen.CodegenError as e:
-        return _run_csv_fallback(out, data, col_types, src.null_values,
-                                 logical_ops, names, sink, str(e),
-                                 delim.decode(), out_cols=sp.output_columns,
+        return _run_csv_fallback(out, _load_data(), col_types,
+                                 src.null_values, logical_ops, names, sink,
+                                 str(e), delim.decode(),
+                                 out_cols=sp.output_columns,
                                  keep_exceptions=keep_exceptions)
     stage = glib.compile_stage(csrc, desc)
 
@@ -578,10 +675,14 @@ def run_csv(context, src, logical_ops, sink=None,
     # one H2D, one execute; the boundary scan runs device-side
     use_resident = (world == 1 and min(want or ndev, ndev) <= 1 and
                     not opts.is_set("tuplex.inputSplitSize") and
-                    len(data) <= parse_size(
+                    total_len <= parse_size(
                         opts.get("tuplex.gpu.residentMaxSize", "24GB")))
-    chunks = ([0, len(data)] if use_resident or len(data) <= split * 2
-              else split_points(data, split))
+    if use_resident:
+        chunks = [0, total_len]
+    else:
+        _load_data()
+        chunks = ([0, total_len] if total_len <= split * 2
+                  else split_points(data, split))
 
     out.mode = "gpu"
     out.metrics = {"t_h2d_ms": 0.0, "t_kernel_ms": 0.0, "t_d2h_ms": 0.0,
@@ -600,7 +701,8 @@ def run_csv(context, src, logical_ops, sink=None,
     rpool_lock = _thr.Lock()
     # zero-copy input pointers: the C side only reads the chunk bytes during
     # the call (it uploads them itself), so point straight into `data`
-    data_arr = _np.frombuffer(data, dtype=_np.uint8)
+    data_arr = (_np.frombuffer(data, dtype=_np.uint8)
+                if data is not None else None)
 
     def run_one(ci, stage_h):
         base = ci << _CHUNK_SHIFT  # composite ordering key namespace
@@ -704,17 +806,15 @@ def run_csv(context, src, logical_ops, sink=None,
     use_devs = min(want or ndev, ndev, max(len(my_work), 1))
     resident_dev = 0
     if use_resident:
-        resident_dev = glib.lib.tpx_dev_alloc(len(data))
-        if resident_dev:
-            import time as _time
-            _t0 = _time.perf_counter()
-            rc = glib.lib.tpx_dev_upload(
-                resident_dev, data_arr.ctypes.data_as(ctypes.c_void_p),
-                len(data))
-            out.metrics["t_h2d_ms"] += (_time.perf_counter() - _t0) * 1e3
-            if rc != 0:
-                glib.lib.tpx_dev_free(resident_dev)
-                resident_dev = 0
+        resident_dev = glib.lib.tpx_dev_alloc(total_len)
+        if resident_dev and not _stream_files_to_device(
+                glib, fmeta, resident_dev, total_len, out.metrics):
+            glib.lib.tpx_dev_free(resident_dev)
+            resident_dev = 0
+        if not resident_dev:
+            # stream/alloc failed: fall back to the host-buffered ranged path
+            _load_data()
+            data_arr = _np.frombuffer(data, dtype=_np.uint8)
     if resident_dev:
         try:
             run_one(0, stage)

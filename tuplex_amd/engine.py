@@ -90,6 +90,9 @@ class GpuLib:
         lib.tpx_dev_upload.argtypes = [ctypes.c_uint64, ctypes.c_void_p,
                                        ctypes.c_int64]
         lib.tpx_dev_free.argtypes = [ctypes.c_uint64]
+        lib.tpx_pinned_alloc.restype = ctypes.c_uint64
+        lib.tpx_pinned_alloc.argtypes = [ctypes.c_int64]
+        lib.tpx_pinned_free.argtypes = [ctypes.c_uint64]
         lib.tpx_stage_execute_col.restype = ctypes.c_int64
         lib.tpx_stage_execute_col.argtypes = [ctypes.c_void_p,
                                               ctypes.POINTER(ctypes.c_void_p),
