@@ -1003,8 +1003,8 @@ def run_csv(context, src, logical_ops, sink=None,
                     f.write(header_line + b"".join(b for _, b in segs))
                 dist.barrier()
         else:
-            _write_csv_output(sink[1], header_line +
-                              b"".join(b for _, b in segs))
+            _write_csv_output(sink[1],
+                              [header_line] + [b for _, b in segs])
         out.rows = []
     return out
 
@@ -1124,12 +1124,17 @@ def _merge_csv_segments(text: bytes, idxs, offs, replayed, row_lo,
     return b"".join(segs)
 
 
-def _write_csv_output(path: str, content: bytes):
+def _write_csv_output(path: str, content):
+    """content: bytes or an iterable of byte segments (streamed write — no
+    whole-output host buffer)."""
     if path.endswith(".csv"):
         os.makedirs(os.path.dirname(path) or ".", exist_ok=True)
-        with open(path, "wb") as f:
-            f.write(content)
+        target = path
     else:
         os.makedirs(path, exist_ok=True)
-        with open(os.path.join(path, "part0.csv"), "wb") as f:
+        target = os.path.join(path, "part0.csv")
+    with open(target, "wb") as f:
+        if isinstance(content, (bytes, bytearray)):
             f.write(content)
+        else:
+            f.writelines(content)
