@@ -561,17 +561,27 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
         double v = e ? atof(e) : 1.0;  // measured: equal split beats 0.75
         return (v > 0.1 && v <= 1.0) ? v : 1.0;
     }();
+    // tail split: equal chunks, but the LAST one halved into two — the
+    // final chunk's compaction+write tail is the only non-overlapped part
+    // of the step, and a half-size tail halves it (TPX_TAIL_SPLIT=0 off)
+    static int env_tail = [] {
+        const char* e = getenv("TPX_TAIL_SPLIT");
+        return e ? atoi(e) : 1;
+    }();
+    if (env_tail && C > 1 && C < 8) ++C;
     long long cstart[9] = {0};
     long long ccnt[8] = {0};
     {
+        int CW = (env_tail && C > 2) ? C - 1 : C;  // weight units: last 2 = half
         double wsum = 0, w = 1;
-        for (int c = 0; c < C; ++c) { wsum += w; w *= env_ratio; }
+        for (int c = 0; c < CW; ++c) { wsum += w; w *= env_ratio; }
         double acc = 0;
         w = 1;
         for (int c = 0; c < C; ++c) {
-            acc += w;
+            bool half = env_tail && C > 2 && c >= C - 2;
+            acc += half ? w * 0.5 : w;
             cstart[c + 1] = c + 1 == C ? n : (long long)(n * acc / wsum);
-            w *= env_ratio;
+            if (!half) w *= env_ratio;
         }
         for (int c = 0; c < C; ++c) ccnt[c] = cstart[c + 1] - cstart[c];
     }
