@@ -561,12 +561,13 @@ static int64_t run_core(tpx_stage* st, void* d_in, void* d_offs, long long n,
         double v = e ? atof(e) : 1.0;  // measured: equal split beats 0.75
         return (v > 0.1 && v <= 1.0) ? v : 1.0;
     }();
-    // tail split: equal chunks, but the LAST one halved into two — the
-    // final chunk's compaction+write tail is the only non-overlapped part
-    // of the step, and a half-size tail halves it (TPX_TAIL_SPLIT=0 off)
+    // tail split: equal chunks, but the LAST one halved into two. Measured
+    // NEGATIVE on Zillow Z1 (772 vs 804 M rows/s: the extra launch + lower
+    // wave efficiency at 5 chunks cost more than the halved tail saved) —
+    // default off, kept for other shapes (TPX_TAIL_SPLIT=1).
     static int env_tail = [] {
         const char* e = getenv("TPX_TAIL_SPLIT");
-        return e ? atoi(e) : 1;
+        return e ? atoi(e) : 0;
     }();
     if (env_tail && C > 1 && C < 8) ++C;
     long long cstart[9] = {0};
