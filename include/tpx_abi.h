@@ -139,6 +139,12 @@ uint64_t tpx_dev_alloc(int64_t size);               /* returns device VA or 0 */
 int64_t  tpx_dev_upload(uint64_t dst, const void* src, int64_t size);
 void     tpx_dev_free(uint64_t ptr);
 
+/* Pinned host staging for streamed ingestion (file -> pinned ring -> DMA,
+ * the read-buffer role of CSVReader.cc:390). Host pointer as u64, 0 on
+ * failure. */
+uint64_t tpx_pinned_alloc(int64_t size);
+void     tpx_pinned_free(uint64_t ptr);
+
 /* Execute over CSV bytes already resident on device. `flags` bit1 (value 2):
  * leave output partitions on device (report sizes/counts only — the device
  * partition manager analog of memory-sink partitions, Partition.h:38). */
