@@ -471,3 +471,35 @@ def test_f64_csv_sink_gpu(tmp_path):
         got = f.read()
     assert got == ref["csv_text"]
     assert b"." in got.split(b"\n")[1]
+
+
+@pytest.mark.gpu
+def test_parallel_resolver_pool_engaged(tmp_path):
+    """>=MIN_POOL_ROWS diverted rows engage the persistent process pool
+    (presolve); output must be identical to the serial resolver's."""
+    import os
+    from tuplex_amd import presolve
+    rows = []
+    # ~5% dirty keeps the i64 sniff (>= normalcase threshold) while the
+    # divert count stays above MIN_POOL_ROWS
+    n = presolve.MIN_POOL_ROWS * 25
+    for i in range(n):
+        if i % 20 == 0:
+            rows.append("notanint,%d\n" % i)   # diverts; replay raises
+        else:
+            rows.append("%d,x%d\n" % (i, i))   # clean
+    p = os.path.join(str(tmp_path), "d.csv")
+    with open(p, "w") as f:
+        f.write("a,b\n" + "".join(rows))
+
+    def du(x):
+        return (x["a"] * 2, x["b"])
+
+    outs = {}
+    for procs in ("1", "2"):
+        ctx = tuplex_amd.Context({"tuplex.gpu.resolveProcesses": procs})
+        ds = apply_ops(ctx.csv(p), [("map", du)])
+        outs[procs] = (ds.collect(), dict(ds._last_outcome.exception_counts))
+        assert ds._last_outcome.mode == "gpu", ds._last_outcome.fallback_reason
+    assert outs["1"] == outs["2"]
+    assert outs["1"][1].get("ValueError", 0) > presolve.MIN_POOL_ROWS // 2
