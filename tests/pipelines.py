@@ -115,6 +115,34 @@ def renamed_use(x):
     return x["n2"] * 2
 
 
+def sq_then(x):
+    return x * x
+
+
+def gt10(x):
+    return x > 10
+
+
+def or235(x):
+    return x == 2 or x == 3 or x == 5
+
+
+def cube(x):
+    return x * x * x
+
+
+def chain24(x):
+    return 2 < x <= 4
+
+
+def addf(x):
+    return x + 10.7
+
+
+def idivneg(x):
+    return x // -6
+
+
 def _rand_ints(n=10000, seed=42):
     rng = random.Random(seed)
     return [rng.randint(-2**40, 2**40) for _ in range(n)]
@@ -164,6 +192,16 @@ PIPELINES = [
     ("zillow_mem", make_zillow_rows(2000, seed=42), ZILLOW_COLS, zillow_ops()),
     ("agg_sum_large", list(range(1, 100001)), None,
      [("aggregate", agg_combine, agg_sum, 0)]),
+    # shapes pinned by tests/golden (reference test_filter/test_arithmetic)
+    ("filter_squares", [1, 2, 3, 4, 5], None,
+     [("map", sq_then), ("filter", gt10)]),
+    ("filter_bool_or_cubes", [1, 2, 3, 4, 5], None,
+     [("filter", or235), ("map", cube)]),
+    ("filter_chained_compare", [1, 2, 3, 4, 5], None,
+     [("filter", chain24)]),
+    ("arith_add_float", [1, 2, 4], None, [("map", addf)]),
+    ("arith_idiv_negdiv", [-10, -9, -8, -7, -6, -5], None,
+     [("map", idivneg)]),
 ]
 
 

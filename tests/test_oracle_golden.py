@@ -177,3 +177,38 @@ def test_zillow_udfs_vs_cpython():
     expect = [extractPrice(dict(zip(cols, row))) for row in rows]
     assert r["output"] == expect
     assert r["exception_counts"] == {}
+
+
+def test_filter_squares():
+    _run("filter_squares", [("map", lambda x: x * x),
+                            ("filter", lambda x: x > 10)])
+
+
+def test_filter_bool_or_cubes():
+    _run("filter_bool_or_cubes",
+         [("filter", lambda x: x == 2 or x == 3 or x == 5),
+          ("map", lambda x: x * x * x)])
+
+
+def test_filter_chained_compare():
+    _run("filter_chained_compare", [("filter", lambda x: 2 < x <= 4)])
+
+
+def test_filter_all_empty():
+    _run("filter_all_empty", [("filter", lambda x: x > 10)])
+
+
+def test_arith_add_float():
+    _run("arith_add_float", [("map", lambda x: x + 10.7)])
+
+
+def test_arith_unary_neg():
+    _run("arith_unary_neg", [("map", lambda x: -x)])
+
+
+def test_arith_idiv_pos():
+    _run("arith_idiv_pos", [("map", lambda x: x // 7)])
+
+
+def test_arith_idiv_negdiv():
+    _run("arith_idiv_negdiv", [("map", lambda x: x // -6)])
