@@ -43,3 +43,47 @@ def test_rowfmt_roundtrip(data):
     back = rowfmt.deserialize_partition(bytes(buf), rt)
     assert back == rows
     assert len(offs) == len(rows) + 1
+
+
+def test_fast_deserializer_matches_scalar_random():
+    """_deserialize_fast must agree with the scalar row loop on random
+    schemas/rows (incl. optionals, empty/unicode strings, extreme ints) both
+    with and without explicit row offsets."""
+    import random
+    from tuplex_amd import rowfmt
+    from tuplex_amd import ttypes as T
+    rng = random.Random(11)
+    bases = [T.I64, T.F64, T.BOOL, T.STR]
+    for trial in range(12):
+        nf = rng.randint(1, 6)
+        ps = []
+        for _ in range(nf):
+            b = rng.choice(bases)
+            ps.append(T.opt(b) if rng.random() < 0.4 else b)
+        rt = T.tup(ps)
+        rows = []
+        for i in range(300):
+            row = []
+            for p in ps:
+                if T.is_opt(p) and rng.random() < 0.25:
+                    row.append(None)
+                    continue
+                b = T.deopt(p)
+                if b == T.I64:
+                    row.append(rng.randint(-2**63, 2**63 - 1))
+                elif b == T.F64:
+                    row.append(rng.choice([0.0, -0.0, 1e300, -1.5,
+                                           float("inf"), 3.14159]))
+                elif b == T.BOOL:
+                    row.append(bool(rng.getrandbits(1)))
+                else:
+                    row.append(rng.choice(["", "a", "héllo", "x" * 50,
+                                           "q,\"\n"]))
+            rows.append(tuple(row))
+        buf, offs = rowfmt.serialize_partition(rows, rt)
+        scalar = [rowfmt.deserialize_row(buf, o, rt)[0] for o in offs[:-1]]
+        fast1 = rowfmt._deserialize_fast(buf, rt, len(rows), offs)
+        fast2 = rowfmt._deserialize_fast(buf, rt, len(rows), None)
+        assert fast1 is not None and fast2 is not None
+        assert list(map(tuple, fast1)) == scalar, trial
+        assert list(map(tuple, fast2)) == scalar, trial
