@@ -156,3 +156,15 @@ def test_cache_gpu_post_stage():
               .filter(lambda x: x[0] % 3 == 0))
     got = cached.collect()
     assert got == want
+
+
+def test_cache_tocsv_matches_uncached(tmp_path):
+    import os
+    ctx = tuplex_amd.Context()
+    p1 = os.path.join(str(tmp_path), "a.csv")
+    p2 = os.path.join(str(tmp_path), "b.csv")
+    plain = ctx.parallelize(DATA).map(_div).resolve(ZeroDivisionError, _res)
+    plain.tocsv(p1)
+    (ctx.parallelize(DATA).map(_div).cache()
+     .resolve(ZeroDivisionError, _res).tocsv(p2))
+    assert open(p1, "rb").read() == open(p2, "rb").read()
