@@ -143,6 +143,10 @@ def idivneg(x):
     return x // -6
 
 
+def strdup(a, b):
+    return a * b
+
+
 def _rand_ints(n=10000, seed=42):
     rng = random.Random(seed)
     return [rng.randint(-2**40, 2**40) for _ in range(n)]
@@ -202,6 +206,11 @@ PIPELINES = [
     ("arith_add_float", [1, 2, 4], None, [("map", addf)]),
     ("arith_idiv_negdiv", [-10, -9, -8, -7, -6, -5], None,
      [("map", idivneg)]),
+    # test_strings.py:34 test_duplication (str*int and int*str repetition)
+    ("strings_dup_str_int", [("negative", -2), ("zero", 0), ("hello", 1),
+                             ("goodbye", 5)], None, [("map", strdup)]),
+    ("strings_dup_int_str", [(-2, "negative"), (0, "zero"), (1, "hello"),
+                             (6, "foo")], None, [("map", strdup)]),
 ]
 
 

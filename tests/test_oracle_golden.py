@@ -212,3 +212,15 @@ def test_arith_idiv_pos():
 
 def test_arith_idiv_negdiv():
     _run("arith_idiv_negdiv", [("map", lambda x: x // -6)])
+
+
+def test_strings_concat():
+    _run("strings_concat", [("map", lambda a, b: a + b)])
+
+
+def test_strings_dup_str_int():
+    _run("strings_dup_str_int", [("map", lambda a, b: a * b)])
+
+
+def test_strings_dup_int_str():
+    _run("strings_dup_int_str", [("map", lambda a, b: a * b)])

@@ -61,6 +61,8 @@ def _ev(n, row, rec):
         return x % y
     if op == "concat":
         return rec(a[0]) + rec(a[1])
+    if op == "strmul":
+        return rec(a[0]) * int(rec(a[1]))
     if op in ("lt", "le", "gt", "ge", "eq", "ne", "strlt", "strle", "strgt",
               "strge", "streq", "strne", "opteq", "optne"):
         x, y = rec(a[0]), rec(a[1])

@@ -558,6 +558,14 @@ class StageCodegen:
                 em.w("if (%s) return ((long long)%d) | ((long long)%d << 32);"
                      % (nv, 129, opid))  # EC_TYPEERROR
             return (v, None)
+        if op == "strmul":
+            x, _ = ev(a[0])
+            y, _ = ev(a[1])
+            r = em.fresh("s")
+            em.w("tstr %s = tpx_strmul(heap, %s, (long long)%s, &_ec);"
+                 % (r, x, y))
+            self._check(em, opid)
+            return (r, None)
         if op in ("add", "sub", "mul"):
             x, _ = ev(a[0])
             y, _ = ev(a[1])
@@ -1152,6 +1160,8 @@ class StageCodegen:
         if op == "concat":
             return self._qfree_and(self._qfree_node(a[0], env),
                                    self._qfree_node(a[1], env))
+        if op == "strmul":
+            return self._qfree_node(a[0], env)
         if op == "replace":
             rep = a[2]
             rep_ok = (rep["op"] == "const" and isinstance(rep.get("v"), str)

@@ -459,6 +459,19 @@ __device__ __forceinline__ tstr tpx_slice(const tstr s, long long lo, bool has_l
     return tstr{s.p + a, b - a};
 }
 
+// str * int repetition (python sequence semantics: n <= 0 -> "")
+__device__ __forceinline__ tstr tpx_strmul(TpxHeap& h, const tstr s,
+                                           long long n, int* ec) {
+    if (n <= 0 || s.n == 0) return tstr{s.p, 0};
+    long long total = s.n * n;
+    char* d = tpx_alloc(h, total);
+    if (!d) { *ec = EC_MEMORYERROR; return tstr{s.p, 0}; }
+    for (long long r = 0; r < n; ++r)
+        for (long long i = 0; i < s.n; ++i)
+            d[r * s.n + i] = s.p[i];
+    return tstr{d, total};
+}
+
 __device__ __forceinline__ tstr tpx_lower(TpxHeap& h, const tstr s, int* ec) {
     char* d = tpx_alloc(h, s.n);
     if (!d) { *ec = EC_MEMORYERROR; return tstr{s.p, 0}; }

@@ -63,6 +63,12 @@ def binop(op, a, b):
             return mk("concat", T.STR, [a, b])
         if op == "mod" and a["t"] == T.STR:
             raise TirError("%-format must go through fmt()")
+        if op == "mul":
+            # str * int / int * str repetition (python sequence semantics)
+            if a["t"] == T.STR and b["t"] in (T.I64, T.BOOL):
+                return mk("strmul", T.STR, [a, b])
+            if b["t"] == T.STR and a["t"] in (T.I64, T.BOOL):
+                return mk("strmul", T.STR, [b, a])
         nt = _numt(a, b)
         if nt is None:
             raise TirError("unsupported operand types for %s: %r %r" % (op, a["t"], b["t"]))
