@@ -147,6 +147,14 @@ def strdup(a, b):
     return a * b
 
 
+def center3(x):
+    return x.center(3)
+
+
+def center4f(x, y):
+    return x.center(4, y)
+
+
 def _rand_ints(n=10000, seed=42):
     rng = random.Random(seed)
     return [rng.randint(-2**40, 2**40) for _ in range(n)]
@@ -211,6 +219,12 @@ PIPELINES = [
                              ("goodbye", 5)], None, [("map", strdup)]),
     ("strings_dup_int_str", [(-2, "negative"), (0, "zero"), (1, "hello"),
                              (6, "foo")], None, [("map", strdup)]),
+    # test_strings.py:207 test_strcenter (CPython left-bias quirk)
+    ("strings_center", ["", "a", "ab", "abc", "abcd"], None,
+     [("map", center3)]),
+    ("strings_center_fill", [("", "|"), ("a", ","), ("ab", "+"),
+                             ("abc", "2"), ("abcd", "%"), ("abcde", "*")],
+     None, [("map", center4f)]),
 ]
 
 

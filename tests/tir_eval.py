@@ -63,6 +63,9 @@ def _ev(n, row, rec):
         return rec(a[0]) + rec(a[1])
     if op == "strmul":
         return rec(a[0]) * int(rec(a[1]))
+    if op == "center":
+        f = rec(a[2]) if len(a) == 3 else " "
+        return rec(a[0]).center(int(rec(a[1])), f)
     if op in ("lt", "le", "gt", "ge", "eq", "ne", "strlt", "strle", "strgt",
               "strge", "streq", "strne", "opteq", "optne"):
         x, y = rec(a[0]), rec(a[1])

@@ -558,6 +558,17 @@ class StageCodegen:
                 em.w("if (%s) return ((long long)%d) | ((long long)%d << 32);"
                      % (nv, 129, opid))  # EC_TYPEERROR
             return (v, None)
+        if op == "center":
+            x, _ = ev(a[0])
+            w, _ = ev(a[1])
+            f = None
+            if len(a) == 3:
+                f, _ = ev(a[2])
+            r = em.fresh("s")
+            em.w("tstr %s = tpx_center(heap, %s, (long long)%s, %s, &_ec);"
+                 % (r, x, w, f if f else 'tstr{" ", 1}'))
+            self._check(em, opid)
+            return (r, None)
         if op == "strmul":
             x, _ = ev(a[0])
             y, _ = ev(a[1])
@@ -1160,6 +1171,17 @@ class StageCodegen:
         if op == "concat":
             return self._qfree_and(self._qfree_node(a[0], env),
                                    self._qfree_node(a[1], env))
+        if op == "center":
+            x, _ = ev(a[0])
+            w, _ = ev(a[1])
+            f = None
+            if len(a) == 3:
+                f, _ = ev(a[2])
+            r = em.fresh("s")
+            em.w("tstr %s = tpx_center(heap, %s, (long long)%s, %s, &_ec);"
+                 % (r, x, w, f if f else 'tstr{" ", 1}'))
+            self._check(em, opid)
+            return (r, None)
         if op == "strmul":
             return self._qfree_node(a[0], env)
         if op == "replace":

@@ -459,6 +459,25 @@ __device__ __forceinline__ tstr tpx_slice(const tstr s, long long lo, bool has_l
     return tstr{s.p + a, b - a};
 }
 
+// s.center(width, fill) — CPython's left-bias quirk
+// (left = marg/2 + (marg & width & 1)); fill must be exactly one char
+// (TypeError otherwise, like CPython / strCenter StringFunctions.cc:224)
+__device__ __forceinline__ tstr tpx_center(TpxHeap& h, const tstr s,
+                                           long long width, const tstr fill,
+                                           int* ec) {
+    if (fill.n != 1) { *ec = EC_TYPEERROR; return tstr{s.p, 0}; }
+    if (s.n >= width) return s;
+    long long marg = width - s.n;
+    long long left = marg / 2 + (marg & width & 1);
+    char* d = tpx_alloc(h, width);
+    if (!d) { *ec = EC_MEMORYERROR; return tstr{s.p, 0}; }
+    char fc = fill.p[0];
+    for (long long i = 0; i < left; ++i) d[i] = fc;
+    for (long long i = 0; i < s.n; ++i) d[left + i] = s.p[i];
+    for (long long i = left + s.n; i < width; ++i) d[i] = fc;
+    return tstr{d, width};
+}
+
 // str * int repetition (python sequence semantics: n <= 0 -> "")
 __device__ __forceinline__ tstr tpx_strmul(TpxHeap& h, const tstr s,
                                            long long n, int* ec) {

@@ -289,7 +289,7 @@ class _Compiler:
             meth = e.func.attr
             args = [self._expr(a, env) for a in e.args]
             if meth in ("find", "rfind", "lower", "upper", "strip", "replace",
-                        "startswith", "endswith", "swapcase"):
+                        "startswith", "endswith", "swapcase", "center"):
                 return tir.call(meth if meth not in ("find", "rfind") else meth,
                                 [obj] + args)
             raise UDFCompileError("unsupported method .%s" % meth)

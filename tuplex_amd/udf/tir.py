@@ -139,6 +139,16 @@ def call(name, args):
     if name == "replace":
         _need(a, (T.STR, T.STR, T.STR))
         return mk("replace", T.STR, a)
+    if name == "center":
+        # s.center(width[, fill]) — FunctionRegistry strCenter
+        # (StringFunctions.cc:224); bool widths coerce like CPython
+        if len(a) not in (2, 3) or a[0]["t"] != T.STR:
+            raise TirError("center args")
+        if a[1]["t"] not in (T.I64, T.BOOL):
+            raise TirError("center width must be int")
+        if len(a) == 3 and a[2]["t"] != T.STR:
+            raise TirError("center fill must be str")
+        return mk("center", T.STR, a)
     if name in ("startswith", "endswith"):
         _need(a, (T.STR, T.STR))
         return mk(name, T.BOOL, a)
