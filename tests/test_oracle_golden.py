@@ -224,3 +224,11 @@ def test_strings_dup_str_int():
 
 def test_strings_dup_int_str():
     _run("strings_dup_int_str", [("map", lambda a, b: a * b)])
+
+
+def test_strings_center3():
+    _run("strings_center3", [("map", lambda x: x.center(3))])
+
+
+def test_strings_center4_fill():
+    _run("strings_center4_fill", [("map", lambda x, y: x.center(4, y))])
