@@ -232,3 +232,19 @@ def test_strings_center3():
 
 def test_strings_center4_fill():
     _run("strings_center4_fill", [("map", lambda x, y: x.center(4, y))])
+
+
+def test_nulls_eq_none_mixed():
+    _run("nulls_eq_none_mixed", [("map", lambda x: x == None)])  # noqa: E711
+
+
+def test_nulls_ne_none_mixed():
+    _run("nulls_ne_none_mixed", [("map", lambda x: x != None)])  # noqa: E711
+
+
+def test_is_bool_false():
+    _run("is_bool_false", [("map", lambda x: x is False)])
+
+
+def test_is_none_opt_bool():
+    _run("is_none_opt_bool", [("map", lambda x: x is not None)])
